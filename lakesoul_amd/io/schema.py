@@ -1,0 +1,237 @@
+"""Table schema model + Spark-compatible JSON serde.
+
+The reference stores ``table_info.table_schema`` as Spark-style schema JSON
+(serialized from Arrow by ``rust/lakesoul-common/src/ser/arrow_java.rs``).
+We emit/parse the same structure: ``{"type":"struct","fields":[{"name":...,
+"type":<spark type name>,"nullable":...,"metadata":{}}]}``.
+"""
+
+from __future__ import annotations
+
+import json
+from dataclasses import dataclass, field as dc_field
+from typing import List, Optional, Sequence, Tuple, Union
+
+# canonical dtype strings used across the engine
+_CANONICAL = {
+    "bool": "bool",
+    "boolean": "bool",
+    "int8": "int8",
+    "byte": "int8",
+    "int16": "int16",
+    "short": "int16",
+    "int32": "int32",
+    "int": "int32",
+    "integer": "int32",
+    "int64": "int64",
+    "long": "int64",
+    "float16": "float16",
+    "half": "float16",
+    "float32": "float32",
+    "float": "float32",
+    "float64": "float64",
+    "double": "float64",
+    "string": "string",
+    "utf8": "string",
+    "str": "string",
+    "binary": "binary",
+    "bytes": "binary",
+    "date32": "date32",
+    "date": "date32",
+    "timestamp[us]": "timestamp[us]",
+    "timestamp": "timestamp[us]",
+    "timestamp[ms]": "timestamp[ms]",
+    "timestamp[ns]": "timestamp[ns]",
+}
+
+_TO_SPARK = {
+    "bool": "boolean",
+    "int8": "byte",
+    "int16": "short",
+    "int32": "integer",
+    "int64": "long",
+    "float16": "half",
+    "float32": "float",
+    "float64": "double",
+    "string": "string",
+    "binary": "binary",
+    "date32": "date",
+    "timestamp[us]": "timestamp",
+    "timestamp[ms]": "timestamp_millis",
+    "timestamp[ns]": "timestamp_nanos",
+}
+_FROM_SPARK = {v: k for k, v in _TO_SPARK.items()}
+
+FIXED_WIDTH_BYTES = {
+    "bool": 1,
+    "int8": 1,
+    "int16": 2,
+    "int32": 4,
+    "int64": 8,
+    "float16": 2,
+    "float32": 4,
+    "float64": 8,
+    "date32": 4,
+    "timestamp[us]": 8,
+    "timestamp[ms]": 8,
+    "timestamp[ns]": 8,
+}
+
+
+def canonical_dtype(dt: str) -> str:
+    try:
+        return _CANONICAL[dt.lower()]
+    except KeyError:
+        raise TypeError(f"unsupported dtype: {dt}")
+
+
+@dataclass
+class Field:
+    name: str
+    dtype: str
+    nullable: bool = True
+
+    def __post_init__(self):
+        self.dtype = canonical_dtype(self.dtype)
+
+    @property
+    def is_fixed_width(self) -> bool:
+        return self.dtype in FIXED_WIDTH_BYTES
+
+
+@dataclass
+class Schema:
+    fields: List[Field] = dc_field(default_factory=list)
+
+    def names(self) -> List[str]:
+        return [f.name for f in self.fields]
+
+    def field(self, name: str) -> Field:
+        for f in self.fields:
+            if f.name == name:
+                return f
+        raise KeyError(name)
+
+    def index(self, name: str) -> int:
+        for i, f in enumerate(self.fields):
+            if f.name == name:
+                return i
+        raise KeyError(name)
+
+    def select(self, names: Sequence[str]) -> "Schema":
+        return Schema([self.field(n) for n in names])
+
+    def __len__(self):
+        return len(self.fields)
+
+    def __iter__(self):
+        return iter(self.fields)
+
+    def __eq__(self, other):
+        return isinstance(other, Schema) and self.fields == other.fields
+
+
+SchemaLike = Union[Schema, Sequence[Tuple[str, str]], "object"]
+
+
+def normalize_schema(s: SchemaLike) -> Schema:
+    if isinstance(s, Schema):
+        return s
+    # pyarrow.Schema duck-typing
+    if hasattr(s, "types") and hasattr(s, "names"):
+        fields = []
+        for f in s:  # type: ignore
+            fields.append(Field(f.name, _arrow_type_to_dtype(f.type), f.nullable))
+        return Schema(fields)
+    # sequence of (name, dtype[, nullable])
+    out = []
+    for item in s:  # type: ignore
+        if len(item) == 2:
+            out.append(Field(item[0], item[1]))
+        else:
+            out.append(Field(item[0], item[1], item[2]))
+    return Schema(out)
+
+
+def _arrow_type_to_dtype(t) -> str:
+    import pyarrow as pa
+
+    mapping = {
+        pa.bool_(): "bool",
+        pa.int8(): "int8",
+        pa.int16(): "int16",
+        pa.int32(): "int32",
+        pa.int64(): "int64",
+        pa.float16(): "float16",
+        pa.float32(): "float32",
+        pa.float64(): "float64",
+        pa.string(): "string",
+        pa.large_string(): "string",
+        pa.binary(): "binary",
+        pa.large_binary(): "binary",
+        pa.date32(): "date32",
+    }
+    if t in mapping:
+        return mapping[t]
+    import pyarrow.types as pt
+
+    if pt.is_timestamp(t):
+        return f"timestamp[{t.unit}]"
+    raise TypeError(f"unsupported arrow type {t}")
+
+
+def dtype_to_arrow(dtype: str):
+    import pyarrow as pa
+
+    mapping = {
+        "bool": pa.bool_(),
+        "int8": pa.int8(),
+        "int16": pa.int16(),
+        "int32": pa.int32(),
+        "int64": pa.int64(),
+        "float16": pa.float16(),
+        "float32": pa.float32(),
+        "float64": pa.float64(),
+        "string": pa.string(),
+        "binary": pa.binary(),
+        "date32": pa.date32(),
+        "timestamp[us]": pa.timestamp("us"),
+        "timestamp[ms]": pa.timestamp("ms"),
+        "timestamp[ns]": pa.timestamp("ns"),
+    }
+    return mapping[dtype]
+
+
+def schema_to_arrow(schema: Schema):
+    import pyarrow as pa
+
+    return pa.schema(
+        [pa.field(f.name, dtype_to_arrow(f.dtype), nullable=f.nullable) for f in schema]
+    )
+
+
+def schema_to_json(schema: Schema) -> str:
+    return json.dumps(
+        {
+            "type": "struct",
+            "fields": [
+                {
+                    "name": f.name,
+                    "type": _TO_SPARK[f.dtype],
+                    "nullable": f.nullable,
+                    "metadata": {},
+                }
+                for f in schema
+            ],
+        }
+    )
+
+
+def schema_from_json(s: str) -> Schema:
+    d = json.loads(s)
+    fields = []
+    for f in d.get("fields", []):
+        t = f["type"]
+        dtype = _FROM_SPARK.get(t, t)
+        fields.append(Field(f["name"], dtype, bool(f.get("nullable", True))))
+    return Schema(fields)

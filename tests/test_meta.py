@@ -1,0 +1,257 @@
+"""Metadata layer tests: schema, MVCC commit protocol, two-phase commit,
+snapshot/time-travel/incremental queries, compaction trigger rule."""
+
+import threading
+import time
+
+import pytest
+
+from lakesoul_amd.meta.client import MetaClient
+from lakesoul_amd.meta.entities import (
+    CommitOp,
+    DataCommitInfo,
+    DataFileOp,
+    FileOp,
+    MetaInfo,
+    PartitionInfo,
+    TableInfo,
+)
+from lakesoul_amd.meta.store import CommitConflictError, SqliteMetaStore
+
+
+def _mk_table(client, name="t1", partitions=";id"):
+    info = TableInfo(
+        table_id=TableInfo.new_table_id(),
+        table_name=name,
+        table_path=f"/tmp/{name}",
+        table_schema="{}",
+        properties='{"hashBucketNum": "4"}',
+        partitions=partitions,
+    )
+    client.create_table(info)
+    return info
+
+
+def _commit_files(client, info, paths, desc="-5", op=CommitOp.MergeCommit):
+    dci = DataCommitInfo(
+        table_id=info.table_id,
+        partition_desc=desc,
+        file_ops=[DataFileOp(p, FileOp.add, size=100) for p in paths],
+        commit_op=op,
+    )
+    client.store.insert_data_commit_info(dci)
+    client.commit_data(
+        MetaInfo(
+            table_info=info,
+            list_partition=[
+                PartitionInfo(
+                    table_id=info.table_id,
+                    partition_desc=desc,
+                    snapshot=[dci.commit_id],
+                    commit_op=op,
+                )
+            ],
+        ),
+        op,
+    )
+    return dci
+
+
+def test_create_and_lookup_table(meta_store):
+    client = MetaClient(meta_store)
+    info = _mk_table(client)
+    assert client.get_table_info_by_name("t1").table_id == info.table_id
+    assert client.get_table_info_by_path("/tmp/t1").table_id == info.table_id
+    assert info.primary_keys() == ["id"]
+    assert info.hash_bucket_num() == 4
+    client.drop_table(info.table_id)
+    assert client.get_table_info_by_name("t1") is None
+
+
+def test_append_merge_versions_and_snapshot_accumulation(meta_store):
+    client = MetaClient(meta_store)
+    info = _mk_table(client)
+    c1 = _commit_files(client, info, ["f1.parquet"])
+    c2 = _commit_files(client, info, ["f2.parquet"])
+    part = meta_store.get_latest_partition_info(info.table_id, "-5")
+    assert part.version == 1
+    assert part.snapshot == [c1.commit_id, c2.commit_id]
+    files = client.files_for_partition(info.table_id, "-5")
+    assert [f.path for f in files] == ["f1.parquet", "f2.parquet"]
+    # commits flipped to committed (two-phase, meta_init.sql:78)
+    assert meta_store.get_data_commit_info(info.table_id, "-5", c1.commit_id).committed
+
+
+def test_compaction_replaces_snapshot(meta_store):
+    client = MetaClient(meta_store)
+    info = _mk_table(client)
+    _commit_files(client, info, ["f1.parquet"])
+    _commit_files(client, info, ["f2.parquet"])
+    compact = _commit_files(
+        client, info, ["compactdir/c1.parquet"], op=CommitOp.CompactionCommit
+    )
+    files = client.files_for_partition(info.table_id, "-5")
+    assert [f.path for f in files] == ["compactdir/c1.parquet"]
+    part = meta_store.get_latest_partition_info(info.table_id, "-5")
+    assert part.version == 2
+    assert part.commit_op is CommitOp.CompactionCommit
+
+
+def test_delete_commit_clears_snapshot(meta_store):
+    client = MetaClient(meta_store)
+    info = _mk_table(client)
+    _commit_files(client, info, ["f1.parquet"])
+    client.commit_data(
+        MetaInfo(
+            table_info=info,
+            list_partition=[
+                PartitionInfo(info.table_id, "-5", snapshot=[], commit_op=CommitOp.DeleteCommit)
+            ],
+        ),
+        CommitOp.DeleteCommit,
+    )
+    assert client.files_for_partition(info.table_id, "-5") == []
+
+
+def test_time_travel_by_version_and_timestamp(meta_store):
+    client = MetaClient(meta_store)
+    info = _mk_table(client)
+    _commit_files(client, info, ["f1.parquet"])
+    t_after_v0 = int(time.time() * 1000)
+    time.sleep(0.01)
+    _commit_files(client, info, ["f2.parquet"])
+    v0_files = client.files_for_partition(info.table_id, "-5", version=0)
+    assert [f.path for f in v0_files] == ["f1.parquet"]
+    ts_files = client.files_for_partition(info.table_id, "-5", timestamp_ms=t_after_v0)
+    assert [f.path for f in ts_files] == ["f1.parquet"]
+
+
+def test_incremental_files(meta_store):
+    client = MetaClient(meta_store)
+    info = _mk_table(client)
+    _commit_files(client, info, ["f1.parquet"])
+    _commit_files(client, info, ["f2.parquet"])
+    _commit_files(client, info, ["f3.parquet"])
+    inc = client.incremental_files(info.table_id, "-5", 0, 2)
+    assert [f.path for f in inc] == ["f2.parquet", "f3.parquet"]
+
+
+def test_file_del_ops_drop_files(meta_store):
+    client = MetaClient(meta_store)
+    info = _mk_table(client)
+    c1 = _commit_files(client, info, ["f1.parquet", "f2.parquet"])
+    dci = DataCommitInfo(
+        table_id=info.table_id,
+        partition_desc="-5",
+        file_ops=[
+            DataFileOp("f1.parquet", FileOp.del_),
+            DataFileOp("f3.parquet", FileOp.add),
+        ],
+        commit_op=CommitOp.UpdateCommit,
+    )
+    client.store.insert_data_commit_info(dci)
+    client.commit_data(
+        MetaInfo(
+            table_info=info,
+            list_partition=[
+                PartitionInfo(
+                    info.table_id,
+                    "-5",
+                    snapshot=[c1.commit_id, dci.commit_id],
+                    commit_op=CommitOp.UpdateCommit,
+                )
+            ],
+        ),
+        CommitOp.UpdateCommit,
+    )
+    files = client.files_for_partition(info.table_id, "-5")
+    assert sorted(f.path for f in files) == ["f2.parquet", "f3.parquet"]
+
+
+def test_concurrent_commits_mvcc_retry(meta_store):
+    """Two threads commit concurrently; both must land (versions 0 and 1)."""
+    client = MetaClient(meta_store)
+    info = _mk_table(client)
+    errs = []
+
+    def worker(i):
+        try:
+            local = MetaClient(SqliteMetaStore(meta_store.path))
+            _commit_files(local, info, [f"f{i}.parquet"])
+        except Exception as e:  # pragma: no cover
+            errs.append(e)
+
+    ts = [threading.Thread(target=worker, args=(i,)) for i in range(4)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join()
+    assert not errs
+    part = meta_store.get_latest_partition_info(info.table_id, "-5")
+    assert part.version == 3
+    assert len(part.snapshot) == 4
+
+
+def test_conflict_error_when_version_taken(meta_store):
+    client = MetaClient(meta_store)
+    info = _mk_table(client)
+    p = PartitionInfo(info.table_id, "-5", version=0, snapshot=["x"])
+    meta_store.transaction_insert_partition_info([p])
+    with pytest.raises(CommitConflictError):
+        meta_store.transaction_insert_partition_info(
+            [PartitionInfo(info.table_id, "-5", version=0, snapshot=["y"])]
+        )
+
+
+def test_compaction_trigger_rule(meta_store):
+    client = MetaClient(meta_store)
+    info = _mk_table(client)
+    for i in range(10):
+        _commit_files(client, info, [f"f{i}.parquet"])
+        expected = i >= 9  # versions 0..9 -> trigger at version >= 10? v9 -> False
+    # after 10 commits latest version is 9 -> not yet (rule: version >= 10)
+    assert not client.compaction_needed(info.table_id, "-5")
+    _commit_files(client, info, ["f10.parquet"])  # version 10
+    assert client.compaction_needed(info.table_id, "-5")
+    _commit_files(client, info, ["c.parquet"], op=CommitOp.CompactionCommit)
+    assert not client.compaction_needed(info.table_id, "-5")
+    # 9 more deltas -> not yet; 10th -> trigger
+    for i in range(9):
+        _commit_files(client, info, [f"g{i}.parquet"])
+    assert not client.compaction_needed(info.table_id, "-5")
+    _commit_files(client, info, ["g9.parquet"])
+    assert client.compaction_needed(info.table_id, "-5")
+
+
+def test_rollback(meta_store):
+    client = MetaClient(meta_store)
+    info = _mk_table(client)
+    _commit_files(client, info, ["f1.parquet"])
+    _commit_files(client, info, ["f2.parquet"])
+    client.rollback_partition(info.table_id, "-5", 0)
+    files = client.files_for_partition(info.table_id, "-5")
+    assert [f.path for f in files] == ["f1.parquet"]
+
+
+def test_schema_json_roundtrip():
+    from lakesoul_amd.io.schema import Field, Schema, schema_from_json, schema_to_json
+
+    s = Schema(
+        [
+            Field("id", "int64", False),
+            Field("name", "string"),
+            Field("score", "float64"),
+            Field("flag", "bool"),
+        ]
+    )
+    j = schema_to_json(s)
+    assert '"type": "long"' in j or '"type":"long"' in j
+    s2 = schema_from_json(j)
+    assert s2 == s
+
+
+def test_namespace_ops(meta_store):
+    client = MetaClient(meta_store)
+    assert "default" in client.list_namespaces()
+    client.create_namespace("ns2")
+    assert "ns2" in client.list_namespaces()
