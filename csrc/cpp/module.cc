@@ -1,0 +1,478 @@
+// lakesoul_amd._cpp — host-side native core: parquet read/write, spark
+// murmur3, CPU decode fallbacks. GPU kernels live in lakesoul_amd._hip.
+#include <torch/extension.h>
+
+#include <atomic>
+#include <cstring>
+#include <memory>
+#include <mutex>
+#include <thread>
+#include <unordered_map>
+
+#include "murmur3.h"
+#include "parquet_file.h"
+
+namespace py = pybind11;
+using namespace lakesoul;
+
+// ---------------------------------------------------------------------- //
+// dtype mapping: canonical dtype string <-> parquet physical/logical
+// ---------------------------------------------------------------------- //
+
+struct DtypeInfo {
+  int32_t physical;
+  int32_t converted;
+  LogicalTag logical;
+  int32_t bit_width;
+};
+
+static DtypeInfo dtype_info(const std::string& dt) {
+  if (dt == "bool") return {PT_BOOLEAN, CV_NONE, LogicalTag::NONE, 0};
+  if (dt == "int8") return {PT_INT32, CV_INT_8, LogicalTag::INT, 8};
+  if (dt == "int16") return {PT_INT32, CV_INT_16, LogicalTag::INT, 16};
+  if (dt == "int32") return {PT_INT32, CV_NONE, LogicalTag::NONE, 0};
+  if (dt == "int64") return {PT_INT64, CV_NONE, LogicalTag::NONE, 0};
+  if (dt == "float32") return {PT_FLOAT, CV_NONE, LogicalTag::NONE, 0};
+  if (dt == "float64") return {PT_DOUBLE, CV_NONE, LogicalTag::NONE, 0};
+  if (dt == "string") return {PT_BYTE_ARRAY, CV_UTF8, LogicalTag::STRING, 0};
+  if (dt == "binary") return {PT_BYTE_ARRAY, CV_NONE, LogicalTag::NONE, 0};
+  if (dt == "date32") return {PT_INT32, CV_DATE, LogicalTag::DATE, 0};
+  if (dt == "timestamp[ms]")
+    return {PT_INT64, CV_TIMESTAMP_MILLIS, LogicalTag::TIMESTAMP_MILLIS, 0};
+  if (dt == "timestamp[us]")
+    return {PT_INT64, CV_TIMESTAMP_MICROS, LogicalTag::TIMESTAMP_MICROS, 0};
+  if (dt == "timestamp[ns]")
+    return {PT_INT64, CV_NONE, LogicalTag::TIMESTAMP_NANOS, 0};
+  throw std::runtime_error("unsupported dtype: " + dt);
+}
+
+static std::string dtype_name(const ColumnDesc& c) {
+  switch (c.physical) {
+    case PT_BOOLEAN:
+      return "bool";
+    case PT_INT32:
+      if (c.converted == CV_INT_8 || (c.logical == LogicalTag::INT && c.int_bit_width == 8))
+        return "int8";
+      if (c.converted == CV_INT_16 || (c.logical == LogicalTag::INT && c.int_bit_width == 16))
+        return "int16";
+      if (c.converted == CV_DATE || c.logical == LogicalTag::DATE) return "date32";
+      return "int32";
+    case PT_INT64:
+      if (c.converted == CV_TIMESTAMP_MILLIS || c.logical == LogicalTag::TIMESTAMP_MILLIS)
+        return "timestamp[ms]";
+      if (c.converted == CV_TIMESTAMP_MICROS || c.logical == LogicalTag::TIMESTAMP_MICROS)
+        return "timestamp[us]";
+      if (c.logical == LogicalTag::TIMESTAMP_NANOS) return "timestamp[ns]";
+      return "int64";
+    case PT_FLOAT:
+      return "float32";
+    case PT_DOUBLE:
+      return "float64";
+    case PT_BYTE_ARRAY:
+      if (c.converted == CV_UTF8 || c.logical == LogicalTag::STRING) return "string";
+      return "binary";
+    default:
+      return "unsupported";
+  }
+}
+
+// ---------------------------------------------------------------------- //
+// writer binding
+// ---------------------------------------------------------------------- //
+
+static int64_t write_parquet(
+    const std::string& path, const std::vector<std::string>& names,
+    const std::vector<std::string>& dtypes, const std::vector<torch::Tensor>& columns,
+    const std::vector<c10::optional<torch::Tensor>>& offsets,
+    const std::vector<c10::optional<torch::Tensor>>& validity,
+    const std::vector<bool>& nullable, int64_t row_group_size, int64_t codec,
+    int64_t level) {
+  size_t ncol = names.size();
+  TORCH_CHECK(dtypes.size() == ncol && columns.size() == ncol);
+  std::vector<ColumnDesc> descs(ncol);
+  std::vector<ColumnData> data(ncol);
+  int64_t num_rows = -1;
+  std::vector<torch::Tensor> keep;  // hold contiguous refs
+
+  for (size_t i = 0; i < ncol; i++) {
+    DtypeInfo di = dtype_info(dtypes[i]);
+    descs[i].name = names[i];
+    descs[i].physical = di.physical;
+    descs[i].converted = di.converted;
+    descs[i].logical = di.logical;
+    descs[i].int_bit_width = di.bit_width;
+    descs[i].nullable = nullable[i];
+
+    torch::Tensor col = columns[i].contiguous().cpu();
+    keep.push_back(col);
+    if (di.physical == PT_BYTE_ARRAY) {
+      TORCH_CHECK(offsets[i].has_value(), "byte_array column needs offsets");
+      torch::Tensor off = offsets[i]->contiguous().cpu().to(torch::kInt32);
+      keep.push_back(off);
+      data[i].offsets = off.data_ptr<int32_t>();
+      data[i].bytes = (const uint8_t*)col.data_ptr();
+      int64_t n = off.numel() - 1;
+      TORCH_CHECK(num_rows < 0 || n == num_rows, "row count mismatch");
+      num_rows = n;
+    } else {
+      // dtype check: int8/int16 must be passed widened to int32
+      data[i].data = (const uint8_t*)col.data_ptr();
+      int64_t n = col.numel();
+      TORCH_CHECK(num_rows < 0 || n == num_rows, "row count mismatch");
+      num_rows = n;
+      int es = physical_elem_size(di.physical);
+      TORCH_CHECK(col.element_size() == es, "column ", names[i],
+                  " element size ", col.element_size(), " != physical ", es);
+    }
+    if (validity[i].has_value()) {
+      torch::Tensor v = validity[i]->contiguous().cpu().to(torch::kUInt8);
+      keep.push_back(v);
+      data[i].validity = v.data_ptr<uint8_t>();
+    }
+  }
+  if (num_rows < 0) num_rows = 0;
+
+  int64_t size;
+  {
+    py::gil_scoped_release rel;
+    ParquetWriter w(path, descs, (int)codec, (int)level, row_group_size);
+    w.write_row_group(data, num_rows);
+    size = w.close();
+  }
+  return size;
+}
+
+// ---------------------------------------------------------------------- //
+// reader bindings
+// ---------------------------------------------------------------------- //
+
+static std::unordered_map<int64_t, std::shared_ptr<ParquetFile>> g_files;
+static std::mutex g_files_mu;
+static std::atomic<int64_t> g_next_handle{1};
+
+static int64_t open_parquet(const std::string& path) {
+  auto f = std::make_shared<ParquetFile>(path);
+  std::lock_guard<std::mutex> lk(g_files_mu);
+  int64_t h = g_next_handle++;
+  g_files[h] = f;
+  return h;
+}
+
+static std::shared_ptr<ParquetFile> get_file(int64_t h) {
+  std::lock_guard<std::mutex> lk(g_files_mu);
+  auto it = g_files.find(h);
+  TORCH_CHECK(it != g_files.end(), "bad parquet handle");
+  return it->second;
+}
+
+static void close_parquet(int64_t h) {
+  std::lock_guard<std::mutex> lk(g_files_mu);
+  g_files.erase(h);
+}
+
+static py::dict parquet_meta(int64_t h) {
+  auto f = get_file(h);
+  py::dict d;
+  d["num_rows"] = f->num_rows();
+  d["num_row_groups"] = (int64_t)f->num_row_groups();
+  py::list cols;
+  for (auto& c : f->columns()) {
+    py::dict cd;
+    cd["name"] = c.name;
+    cd["dtype"] = dtype_name(c);
+    cd["nullable"] = c.nullable;
+    cols.append(cd);
+  }
+  d["columns"] = cols;
+  py::list rgs;
+  for (auto& g : f->meta().row_groups) {
+    py::dict gd;
+    gd["num_rows"] = g.num_rows;
+    py::list cstats;
+    for (auto& cm : g.columns) {
+      py::dict sd;
+      sd["num_values"] = cm.num_values;
+      sd["null_count"] = cm.stats.null_count;
+      if (cm.stats.has_min_max) {
+        sd["min"] = py::bytes(cm.stats.min_value);
+        sd["max"] = py::bytes(cm.stats.max_value);
+      }
+      cstats.append(sd);
+    }
+    gd["columns"] = cstats;
+    rgs.append(gd);
+  }
+  d["row_groups"] = rgs;
+  return d;
+}
+
+static torch::Tensor vec_to_tensor_u8(std::vector<uint8_t>&& v) {
+  auto t = torch::empty({(int64_t)v.size()}, torch::kUInt8);
+  std::memcpy(t.data_ptr(), v.data(), v.size());
+  return t;
+}
+
+static torch::Tensor vec_to_tensor_i32(std::vector<int32_t>&& v) {
+  auto t = torch::empty({(int64_t)v.size()}, torch::kInt32);
+  std::memcpy(t.data_ptr(), v.data(), v.size() * 4);
+  return t;
+}
+
+// raw chunk for GPU decode
+static py::dict chunk_to_dict_raw(ParquetFile::ChunkData&& ch) {
+  py::dict d;
+  d["physical"] = ch.physical;
+  d["num_values"] = ch.num_values;
+  d["null_count"] = ch.null_count;
+  d["is_dict"] = ch.is_dict;
+  d["dict_num_values"] = ch.dict_num_values;
+  d["values"] = vec_to_tensor_u8(std::move(ch.values));
+  d["validity"] = vec_to_tensor_u8(std::move(ch.validity));
+  d["dict"] = vec_to_tensor_u8(std::move(ch.dict));
+  // idx pages as int64 [n,5]
+  auto pt = torch::empty({(int64_t)ch.idx_pages.size(), 5}, torch::kInt64);
+  auto pa = pt.accessor<int64_t, 2>();
+  for (size_t i = 0; i < ch.idx_pages.size(); i++) {
+    pa[i][0] = ch.idx_pages[i].out_off;
+    pa[i][1] = ch.idx_pages[i].n;
+    pa[i][2] = ch.idx_pages[i].payload_off;
+    pa[i][3] = ch.idx_pages[i].payload_len;
+    pa[i][4] = ch.idx_pages[i].bit_width;
+  }
+  d["idx_pages"] = pt;
+  return d;
+}
+
+static py::dict read_chunk_raw(int64_t h, int64_t rg, int64_t col) {
+  auto f = get_file(h);
+  ParquetFile::ChunkData ch;
+  {
+    py::gil_scoped_release rel;
+    ch = f->read_chunk((size_t)rg, (size_t)col);
+  }
+  return chunk_to_dict_raw(std::move(ch));
+}
+
+static py::dict decoded_to_dict(DecodedColumn&& dc) {
+  py::dict d;
+  d["num_values"] = dc.num_values;
+  d["data"] = vec_to_tensor_u8(std::move(dc.data));
+  d["offsets"] = vec_to_tensor_i32(std::move(dc.offsets));
+  d["bytes"] = vec_to_tensor_u8(std::move(dc.bytes));
+  d["validity"] = vec_to_tensor_u8(std::move(dc.validity));
+  return d;
+}
+
+static py::dict read_chunk_cpu(int64_t h, int64_t rg, int64_t col) {
+  auto f = get_file(h);
+  DecodedColumn dc;
+  {
+    py::gil_scoped_release rel;
+    auto ch = f->read_chunk((size_t)rg, (size_t)col);
+    dc = decode_chunk_cpu(ch);
+  }
+  return decoded_to_dict(std::move(dc));
+}
+
+// batch: parallel host decode across (rg,col) pairs
+static py::list read_chunks_cpu_batch(int64_t h,
+                                      const std::vector<std::pair<int64_t, int64_t>>& rc,
+                                      int64_t nthreads) {
+  auto f = get_file(h);
+  std::vector<DecodedColumn> out(rc.size());
+  {
+    py::gil_scoped_release rel;
+    std::atomic<size_t> next{0};
+    int nt = (int)std::min<int64_t>(nthreads > 0 ? nthreads : 8, (int64_t)rc.size());
+    std::vector<std::thread> threads;
+    std::string err;
+    std::mutex err_mu;
+    for (int t = 0; t < nt; t++) {
+      threads.emplace_back([&]() {
+        while (true) {
+          size_t i = next.fetch_add(1);
+          if (i >= rc.size()) break;
+          try {
+            auto ch = f->read_chunk((size_t)rc[i].first, (size_t)rc[i].second);
+            out[i] = decode_chunk_cpu(ch);
+          } catch (std::exception& e) {
+            std::lock_guard<std::mutex> lk(err_mu);
+            err = e.what();
+          }
+        }
+      });
+    }
+    for (auto& th : threads) th.join();
+    if (!err.empty()) throw std::runtime_error(err);
+  }
+  py::list result;
+  for (auto& dc : out) result.append(decoded_to_dict(std::move(dc)));
+  return result;
+}
+
+static py::list read_chunks_raw_batch(int64_t h,
+                                      const std::vector<std::pair<int64_t, int64_t>>& rc,
+                                      int64_t nthreads) {
+  auto f = get_file(h);
+  std::vector<ParquetFile::ChunkData> out(rc.size());
+  {
+    py::gil_scoped_release rel;
+    std::atomic<size_t> next{0};
+    int nt = (int)std::min<int64_t>(nthreads > 0 ? nthreads : 8, (int64_t)rc.size());
+    std::vector<std::thread> threads;
+    std::string err;
+    std::mutex err_mu;
+    for (int t = 0; t < nt; t++) {
+      threads.emplace_back([&]() {
+        while (true) {
+          size_t i = next.fetch_add(1);
+          if (i >= rc.size()) break;
+          try {
+            out[i] = f->read_chunk((size_t)rc[i].first, (size_t)rc[i].second);
+          } catch (std::exception& e) {
+            std::lock_guard<std::mutex> lk(err_mu);
+            err = e.what();
+          }
+        }
+      });
+    }
+    for (auto& th : threads) th.join();
+    if (!err.empty()) throw std::runtime_error(err);
+  }
+  py::list result;
+  for (auto& ch : out) result.append(chunk_to_dict_raw(std::move(ch)));
+  return result;
+}
+
+// ---------------------------------------------------------------------- //
+// murmur3 (CPU)
+// ---------------------------------------------------------------------- //
+
+// Hash fixed-width columns with seed chaining; returns int64 tensor of u32
+// hashes. columns: CPU tensors (bool/int8/16/32/64/float32/64). Strings:
+// pass (offsets,bytes) via hash_string_column step.
+static torch::Tensor hash_columns_cpu(
+    const std::vector<torch::Tensor>& columns,
+    const std::vector<c10::optional<torch::Tensor>>& validity) {
+  TORCH_CHECK(!columns.empty());
+  int64_t n = columns[0].numel();
+  auto out = torch::empty({n}, torch::kInt64);
+  int64_t* hp = out.data_ptr<int64_t>();
+  std::vector<torch::Tensor> cols;
+  for (auto& c : columns) cols.push_back(c.contiguous().cpu());
+
+  for (size_t ci = 0; ci < cols.size(); ci++) {
+    torch::Tensor c = cols[ci];
+    const uint8_t* vmask = nullptr;
+    torch::Tensor vt;
+    if (validity[ci].has_value()) {
+      vt = validity[ci]->contiguous().cpu().to(torch::kUInt8);
+      vmask = vt.data_ptr<uint8_t>();
+    }
+    auto st = c.scalar_type();
+    at::parallel_for(0, n, 4096, [&](int64_t b, int64_t e) {
+      for (int64_t i = b; i < e; i++) {
+        uint32_t seed = ci == 0 ? kHashSeed : (uint32_t)hp[i];
+        if (vmask && !vmask[i]) {
+          if (ci == 0) hp[i] = 0;
+          continue;
+        }
+        uint32_t hv;
+        switch (st) {
+          case torch::kBool:
+            hv = spark_hash_u32((uint32_t)c.data_ptr<bool>()[i], seed);
+            break;
+          case torch::kInt8:
+            hv = spark_hash_u32((uint32_t)(int32_t)c.data_ptr<int8_t>()[i], seed);
+            break;
+          case torch::kInt16:
+            hv = spark_hash_u32((uint32_t)(int32_t)c.data_ptr<int16_t>()[i], seed);
+            break;
+          case torch::kInt32:
+            hv = spark_hash_u32((uint32_t)c.data_ptr<int32_t>()[i], seed);
+            break;
+          case torch::kInt64:
+            hv = spark_hash_u64((uint64_t)c.data_ptr<int64_t>()[i], seed);
+            break;
+          case torch::kFloat:
+            hv = spark_hash_f32(c.data_ptr<float>()[i], seed);
+            break;
+          case torch::kDouble:
+            hv = spark_hash_f64(c.data_ptr<double>()[i], seed);
+            break;
+          default:
+            hv = 0;  // checked below
+        }
+        hp[i] = (int64_t)hv;
+      }
+    });
+    TORCH_CHECK(st == torch::kBool || st == torch::kInt8 || st == torch::kInt16 ||
+                    st == torch::kInt32 || st == torch::kInt64 ||
+                    st == torch::kFloat || st == torch::kDouble,
+                "unsupported dtype for murmur3");
+  }
+  return out;
+}
+
+// string column step: chain into existing hashes (or start, col_index==0)
+static torch::Tensor hash_string_column_cpu(torch::Tensor offsets,
+                                            torch::Tensor bytes,
+                                            c10::optional<torch::Tensor> validity,
+                                            c10::optional<torch::Tensor> prev,
+                                            bool first_column) {
+  auto off = offsets.contiguous().cpu().to(torch::kInt32);
+  auto by = bytes.contiguous().cpu();
+  int64_t n = off.numel() - 1;
+  auto out = prev.has_value() ? prev->clone() : torch::zeros({n}, torch::kInt64);
+  int64_t* hp = out.data_ptr<int64_t>();
+  const int32_t* op = off.data_ptr<int32_t>();
+  const uint8_t* bp = (const uint8_t*)by.data_ptr();
+  const uint8_t* vmask = nullptr;
+  torch::Tensor vt;
+  if (validity.has_value()) {
+    vt = validity->contiguous().cpu().to(torch::kUInt8);
+    vmask = vt.data_ptr<uint8_t>();
+  }
+  at::parallel_for(0, n, 4096, [&](int64_t b, int64_t e) {
+    for (int64_t i = b; i < e; i++) {
+      if (vmask && !vmask[i]) {
+        if (first_column) hp[i] = 0;
+        continue;
+      }
+      uint32_t seed = first_column ? kHashSeed : (uint32_t)hp[i];
+      hp[i] = (int64_t)spark_hash_bytes(bp + op[i], op[i + 1] - op[i], seed);
+    }
+  });
+  return out;
+}
+
+static torch::Tensor bucket_ids_from_hashes(torch::Tensor hashes, int64_t nbuckets) {
+  auto h = hashes.contiguous().cpu();
+  int64_t n = h.numel();
+  auto out = torch::empty({n}, torch::kInt32);
+  const int64_t* hp = h.data_ptr<int64_t>();
+  int32_t* op = out.data_ptr<int32_t>();
+  for (int64_t i = 0; i < n; i++)
+    op[i] = (int32_t)((uint32_t)hp[i] % (uint32_t)nbuckets);
+  return out;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "lakesoul_amd host-native core";
+  m.def("write_parquet", &write_parquet, "write a parquet file");
+  m.def("open_parquet", &open_parquet);
+  m.def("close_parquet", &close_parquet);
+  m.def("parquet_meta", &parquet_meta);
+  m.def("read_chunk_raw", &read_chunk_raw);
+  m.def("read_chunk_cpu", &read_chunk_cpu);
+  m.def("read_chunks_cpu_batch", &read_chunks_cpu_batch);
+  m.def("read_chunks_raw_batch", &read_chunks_raw_batch);
+  m.def("hash_columns_cpu", &hash_columns_cpu);
+  m.def("hash_string_column_cpu", &hash_string_column_cpu);
+  m.def("bucket_ids_from_hashes", &bucket_ids_from_hashes);
+  m.def("spark_hash_bytes", [](py::bytes b, int64_t seed) {
+    std::string s = b;
+    return (int64_t)spark_hash_bytes((const uint8_t*)s.data(), (int64_t)s.size(),
+                                     (uint32_t)seed);
+  });
+}

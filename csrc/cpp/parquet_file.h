@@ -1,0 +1,652 @@
+// Parquet file writer + reader (host side).
+//
+// MI355X-native replacement for the reference's arrow-rs `parquet`
+// dependency (wired at rust/lakesoul-io/src/writer/mod.rs:53,
+// file_format.rs:19-23). Defaults follow the reference's measured config:
+// zstd(1), dictionary off, row-group <= 250k rows (writer/mod.rs:224-245).
+//
+// The reader separates host work (footer/page walk, decompression,
+// def-level decode) from value decode: PLAIN fixed-width payloads are
+// returned as contiguous buffers ready for H2D + GPU cast/scatter, and
+// dictionary-index pages are returned raw with run descriptors for the
+// HIP expansion kernel (csrc/hip/decode.hip). decode_chunk_cpu() is the
+// CPU fallback used in non-GPU tests.
+#pragma once
+
+#include <fcntl.h>
+#include <sys/mman.h>
+#include <sys/stat.h>
+#include <unistd.h>
+
+#include <algorithm>
+#include <cstdint>
+#include <cstdio>
+#include <cstring>
+#include <memory>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+#include "compress.h"
+#include "parquet_types.h"
+#include "rle.h"
+
+namespace lakesoul {
+
+// ---------------------------------------------------------------------- //
+// column description (writer input / reader output)
+// ---------------------------------------------------------------------- //
+
+struct ColumnDesc {
+  std::string name;
+  int32_t physical = PT_INT64;
+  bool nullable = true;
+  int32_t converted = CV_NONE;
+  LogicalTag logical = LogicalTag::NONE;
+  int32_t int_bit_width = 0;
+  bool int_signed = true;
+};
+
+inline int physical_elem_size(int32_t pt) {
+  switch (pt) {
+    case PT_INT32:
+    case PT_FLOAT:
+      return 4;
+    case PT_INT64:
+    case PT_DOUBLE:
+      return 8;
+    case PT_BOOLEAN:
+      return 1;  // in-memory byte-per-value
+    default:
+      return -1;  // variable
+  }
+}
+
+// ---------------------------------------------------------------------- //
+// writer
+// ---------------------------------------------------------------------- //
+
+struct ColumnData {
+  // fixed-width: `data` points at n*elem values (bool: byte per value)
+  // byte_array: `offsets` (n+1 int32) + `bytes`
+  const uint8_t* data = nullptr;
+  const int32_t* offsets = nullptr;
+  const uint8_t* bytes = nullptr;
+  const uint8_t* validity = nullptr;  // byte per row, 1=valid; null=all valid
+};
+
+class ParquetWriter {
+ public:
+  ParquetWriter(const std::string& path, std::vector<ColumnDesc> cols,
+                int codec = CODEC_ZSTD, int level = 1,
+                int64_t max_row_group_rows = 250000)
+      : cols_(std::move(cols)),
+        codec_(codec),
+        level_(level),
+        max_rg_rows_(max_row_group_rows) {
+    f_ = std::fopen(path.c_str(), "wb");
+    if (!f_) throw std::runtime_error("cannot open for write: " + path);
+    const char magic[4] = {'P', 'A', 'R', '1'};
+    fwrite_all(magic, 4);
+  }
+
+  ~ParquetWriter() {
+    if (f_) std::fclose(f_);
+  }
+
+  // write one row group (caller may pass more rows; we split internally)
+  void write_row_group(const std::vector<ColumnData>& data, int64_t num_rows) {
+    int64_t off = 0;
+    while (off < num_rows) {
+      int64_t n = std::min(max_rg_rows_, num_rows - off);
+      write_one_row_group(data, off, n);
+      off += n;
+    }
+  }
+
+  int64_t close() {
+    FileMetaData fm;
+    fm.version = 2;
+    fm.created_by = "lakesoul_amd 0.1.0";
+    SchemaElement root;
+    root.name = "schema";
+    root.type = -1;
+    root.repetition = REP_REQUIRED;
+    root.num_children = (int32_t)cols_.size();
+    fm.schema.push_back(root);
+    for (auto& c : cols_) {
+      SchemaElement e;
+      e.name = c.name;
+      e.type = c.physical;
+      e.repetition = c.nullable ? REP_OPTIONAL : REP_REQUIRED;
+      e.converted = c.converted;
+      e.logical = c.logical;
+      e.int_bit_width = c.int_bit_width;
+      e.int_signed = c.int_signed;
+      fm.schema.push_back(e);
+    }
+    fm.num_rows = total_rows_;
+    fm.row_groups = row_groups_;
+    auto meta = serialize_file_meta(fm);
+    fwrite_all(meta.data(), meta.size());
+    uint32_t len = (uint32_t)meta.size();
+    fwrite_all(&len, 4);
+    const char magic[4] = {'P', 'A', 'R', '1'};
+    fwrite_all(magic, 4);
+    int64_t size = pos_;
+    std::fclose(f_);
+    f_ = nullptr;
+    return size;
+  }
+
+ private:
+  void fwrite_all(const void* p, size_t n) {
+    if (std::fwrite(p, 1, n, f_) != n) throw std::runtime_error("write failed");
+    pos_ += (int64_t)n;
+  }
+
+  void write_one_row_group(const std::vector<ColumnData>& data, int64_t row_off,
+                           int64_t n) {
+    RowGroup rg;
+    rg.num_rows = n;
+    for (size_t ci = 0; ci < cols_.size(); ci++) {
+      const ColumnDesc& cd = cols_[ci];
+      const ColumnData& col = data[ci];
+
+      // ---- assemble page payload (def levels + PLAIN values) ----
+      std::vector<uint8_t> payload;
+      const uint8_t* validity =
+          cd.nullable && col.validity ? col.validity + row_off : nullptr;
+      int64_t null_count = 0;
+      if (cd.nullable) {
+        std::vector<uint8_t> levels = encode_def_levels(validity, n);
+        uint32_t lv_len = (uint32_t)levels.size();
+        payload.insert(payload.end(), (uint8_t*)&lv_len, (uint8_t*)&lv_len + 4);
+        payload.insert(payload.end(), levels.begin(), levels.end());
+        if (validity)
+          for (int64_t i = 0; i < n; i++) null_count += validity[i] ? 0 : 1;
+      }
+
+      Statistics stats;
+      stats.null_count = cd.nullable ? null_count : 0;
+      append_plain_values(cd, col, row_off, n, validity, payload, stats);
+
+      // ---- compress ----
+      std::vector<uint8_t> compressed;
+      const uint8_t* body = payload.data();
+      size_t body_n = payload.size();
+      if (codec_ == CODEC_ZSTD) {
+        compressed = zstd_compress(payload.data(), payload.size(), level_);
+        body = compressed.data();
+        body_n = compressed.size();
+      } else if (codec_ != CODEC_UNCOMPRESSED) {
+        throw std::runtime_error("writer supports zstd/uncompressed only");
+      }
+
+      // ---- page header + emit ----
+      PageHeader ph;
+      ph.type = PAGE_DATA;
+      ph.uncompressed_size = (int32_t)payload.size();
+      ph.compressed_size = (int32_t)body_n;
+      ph.num_values = (int32_t)n;
+      ph.encoding = ENC_PLAIN;
+      ph.def_encoding = ENC_RLE;
+      ph.rep_encoding = ENC_RLE;
+      auto ph_bytes = serialize_page_header(ph);
+
+      ColumnMeta cm;
+      cm.type = cd.physical;
+      cm.encodings = {ENC_PLAIN, ENC_RLE};
+      cm.path_in_schema = {cd.name};
+      cm.codec = codec_;
+      cm.num_values = n;
+      cm.data_page_offset = pos_;
+      cm.total_uncompressed_size = (int64_t)(ph_bytes.size() + payload.size());
+      cm.total_compressed_size = (int64_t)(ph_bytes.size() + body_n);
+      cm.stats = stats;
+      rg.columns.push_back(cm);
+      rg.total_byte_size += cm.total_uncompressed_size;
+
+      fwrite_all(ph_bytes.data(), ph_bytes.size());
+      fwrite_all(body, body_n);
+    }
+    row_groups_.push_back(std::move(rg));
+    total_rows_ += n;
+  }
+
+  template <typename T>
+  void minmax_update(const uint8_t* p, Statistics& s) {
+    T v;
+    std::memcpy(&v, p, sizeof(T));
+    T mn, mx;
+    if (!s.has_min_max) {
+      mn = mx = v;
+      s.has_min_max = true;
+    } else {
+      std::memcpy(&mn, s.min_value.data(), sizeof(T));
+      std::memcpy(&mx, s.max_value.data(), sizeof(T));
+      if (v < mn) mn = v;
+      if (v > mx) mx = v;
+    }
+    s.min_value.assign((char*)&mn, sizeof(T));
+    s.max_value.assign((char*)&mx, sizeof(T));
+  }
+
+  void append_plain_values(const ColumnDesc& cd, const ColumnData& col,
+                           int64_t row_off, int64_t n, const uint8_t* validity,
+                           std::vector<uint8_t>& payload, Statistics& stats) {
+    if (cd.physical == PT_BYTE_ARRAY) {
+      const int32_t* offs = col.offsets + row_off;
+      for (int64_t i = 0; i < n; i++) {
+        if (validity && !validity[i]) continue;
+        uint32_t len = (uint32_t)(offs[i + 1] - offs[i]);
+        payload.insert(payload.end(), (uint8_t*)&len, (uint8_t*)&len + 4);
+        payload.insert(payload.end(), col.bytes + offs[i],
+                       col.bytes + offs[i + 1]);
+        // lexicographic min/max
+        std::string v((const char*)(col.bytes + offs[i]), len);
+        if (!stats.has_min_max) {
+          stats.min_value = stats.max_value = v;
+          stats.has_min_max = true;
+        } else {
+          if (v < stats.min_value) stats.min_value = v;
+          if (v > stats.max_value) stats.max_value = v;
+        }
+      }
+      return;
+    }
+    if (cd.physical == PT_BOOLEAN) {
+      // bit-pack non-null byte values LSB-first
+      uint8_t cur = 0;
+      int nb = 0;
+      const uint8_t* d = col.data + row_off;
+      for (int64_t i = 0; i < n; i++) {
+        if (validity && !validity[i]) continue;
+        if (d[i]) cur |= (uint8_t)(1 << nb);
+        if (++nb == 8) {
+          payload.push_back(cur);
+          cur = 0;
+          nb = 0;
+        }
+      }
+      if (nb) payload.push_back(cur);
+      return;
+    }
+    int es = physical_elem_size(cd.physical);
+    const uint8_t* d = col.data + row_off * es;
+    if (!validity) {
+      payload.insert(payload.end(), d, d + n * es);
+      for (int64_t i = 0; i < n; i++) {
+        const uint8_t* p = d + i * es;
+        switch (cd.physical) {
+          case PT_INT32: minmax_update<int32_t>(p, stats); break;
+          case PT_INT64: minmax_update<int64_t>(p, stats); break;
+          case PT_FLOAT: minmax_update<float>(p, stats); break;
+          case PT_DOUBLE: minmax_update<double>(p, stats); break;
+        }
+      }
+    } else {
+      for (int64_t i = 0; i < n; i++) {
+        if (!validity[i]) continue;
+        const uint8_t* p = d + i * es;
+        payload.insert(payload.end(), p, p + es);
+        switch (cd.physical) {
+          case PT_INT32: minmax_update<int32_t>(p, stats); break;
+          case PT_INT64: minmax_update<int64_t>(p, stats); break;
+          case PT_FLOAT: minmax_update<float>(p, stats); break;
+          case PT_DOUBLE: minmax_update<double>(p, stats); break;
+        }
+      }
+    }
+  }
+
+  std::FILE* f_ = nullptr;
+  std::vector<ColumnDesc> cols_;
+  int codec_;
+  int level_;
+  int64_t max_rg_rows_;
+  int64_t pos_ = 0;
+  int64_t total_rows_ = 0;
+  std::vector<RowGroup> row_groups_;
+};
+
+// ---------------------------------------------------------------------- //
+// reader
+// ---------------------------------------------------------------------- //
+
+class ParquetFile {
+ public:
+  explicit ParquetFile(const std::string& path) : path_(path) {
+    fd_ = ::open(path.c_str(), O_RDONLY);
+    if (fd_ < 0) throw std::runtime_error("cannot open " + path);
+    struct stat st;
+    if (fstat(fd_, &st) != 0) throw std::runtime_error("fstat failed " + path);
+    size_ = (size_t)st.st_size;
+    if (size_ < 12) throw std::runtime_error("file too small: " + path);
+    map_ = (uint8_t*)mmap(nullptr, size_, PROT_READ, MAP_PRIVATE, fd_, 0);
+    if (map_ == MAP_FAILED) throw std::runtime_error("mmap failed " + path);
+    if (std::memcmp(map_, "PAR1", 4) != 0 ||
+        std::memcmp(map_ + size_ - 4, "PAR1", 4) != 0)
+      throw std::runtime_error("not a parquet file: " + path);
+    uint32_t meta_len;
+    std::memcpy(&meta_len, map_ + size_ - 8, 4);
+    if (meta_len + 12 > size_) throw std::runtime_error("bad footer length");
+    meta_ = parse_file_meta(map_ + size_ - 8 - meta_len, meta_len);
+    // build leaf column list (flat schemas: children of root)
+    build_columns();
+  }
+
+  ~ParquetFile() {
+    if (map_ && map_ != MAP_FAILED) munmap(map_, size_);
+    if (fd_ >= 0) ::close(fd_);
+  }
+
+  const FileMetaData& meta() const { return meta_; }
+  const std::vector<ColumnDesc>& columns() const { return cols_; }
+  const std::string& path() const { return path_; }
+  int64_t num_rows() const { return meta_.num_rows; }
+  size_t num_row_groups() const { return meta_.row_groups.size(); }
+
+  int column_index(const std::string& name) const {
+    for (size_t i = 0; i < cols_.size(); i++)
+      if (cols_[i].name == name) return (int)i;
+    return -1;
+  }
+
+  struct IdxPage {
+    int64_t out_off;
+    int64_t n;            // non-null values in page
+    int64_t payload_off;  // offset into ChunkData.values
+    int64_t payload_len;
+    int32_t bit_width;
+  };
+
+  struct ChunkData {
+    int32_t physical = 0;
+    int64_t num_values = 0;  // rows in chunk
+    int64_t null_count = 0;
+    bool is_dict = false;
+    std::vector<uint8_t> values;    // PLAIN payload (dense non-null) or idx payloads
+    std::vector<uint8_t> validity;  // byte/value, empty if no nulls present
+    std::vector<uint8_t> dict;      // PLAIN dictionary payload
+    int64_t dict_num_values = 0;
+    std::vector<IdxPage> idx_pages;
+  };
+
+  ChunkData read_chunk(size_t rg, size_t col) const {
+    const RowGroup& g = meta_.row_groups.at(rg);
+    const ColumnMeta& cm = g.columns.at(col);
+    const ColumnDesc& cd = cols_.at(col);
+    ChunkData out;
+    out.physical = cd.physical;
+    out.num_values = cm.num_values;
+
+    int64_t off = cm.dictionary_page_offset >= 0
+                      ? cm.dictionary_page_offset
+                      : cm.data_page_offset;
+    // some writers put data_page_offset < dictionary_page_offset wrongly;
+    // start at the min positive
+    if (cm.dictionary_page_offset >= 0 &&
+        cm.data_page_offset < cm.dictionary_page_offset)
+      off = cm.data_page_offset;
+    int64_t end = off + cm.total_compressed_size;
+    int64_t values_seen = 0;
+    bool any_null_page = false;
+    out.validity.assign((size_t)cm.num_values, 1);
+
+    while (off < end && values_seen < cm.num_values) {
+      ThriftReader r(map_ + off, (size_t)(end - off));
+      const uint8_t* hdr_start = map_ + off;
+      PageHeader ph = parse_page_header(r);
+      int64_t hdr_len = (int64_t)r.consumed(hdr_start);
+      const uint8_t* body = map_ + off + hdr_len;
+      off += hdr_len + ph.compressed_size;
+
+      if (ph.type == PAGE_DICTIONARY) {
+        out.dict.resize(ph.uncompressed_size);
+        decompress_into(cm.codec, out.dict.data(), out.dict.size(), body,
+                        ph.compressed_size);
+        out.dict_num_values = ph.dict_num_values;
+        continue;
+      }
+      if (ph.type != PAGE_DATA && ph.type != PAGE_DATA_V2) continue;
+
+      std::vector<uint8_t> page;
+      const uint8_t* vals;
+      size_t vals_len;
+      int64_t nv = ph.num_values;
+      int64_t page_nulls = 0;
+
+      if (ph.type == PAGE_DATA) {
+        page.resize(ph.uncompressed_size);
+        decompress_into(cm.codec, page.data(), page.size(), body,
+                        ph.compressed_size);
+        const uint8_t* p = page.data();
+        size_t rem = page.size();
+        if (cd.nullable) {
+          if (ph.def_encoding != ENC_RLE)
+            throw std::runtime_error("unsupported def-level encoding");
+          uint32_t lv_len;
+          std::memcpy(&lv_len, p, 4);
+          std::vector<uint8_t> levels((size_t)nv);
+          rle_decode<uint8_t>(p + 4, lv_len, 1, nv, levels.data());
+          for (int64_t i = 0; i < nv; i++) {
+            if (!levels[i]) {
+              out.validity[values_seen + i] = 0;
+              page_nulls++;
+            }
+          }
+          p += 4 + lv_len;
+          rem -= 4 + lv_len;
+        }
+        vals = p;
+        vals_len = rem;
+      } else {  // DATA_PAGE_V2: levels uncompressed, values possibly compressed
+        int64_t lv = ph.rep_levels_byte_length + ph.def_levels_byte_length;
+        if (cd.nullable && ph.def_levels_byte_length > 0) {
+          std::vector<uint8_t> levels((size_t)nv);
+          rle_decode<uint8_t>(body + ph.rep_levels_byte_length,
+                              ph.def_levels_byte_length, 1, nv, levels.data());
+          for (int64_t i = 0; i < nv; i++) {
+            if (!levels[i]) {
+              out.validity[values_seen + i] = 0;
+              page_nulls++;
+            }
+          }
+        }
+        size_t comp_vals = (size_t)(ph.compressed_size - lv);
+        size_t uncomp_vals = (size_t)(ph.uncompressed_size - lv);
+        page.resize(uncomp_vals);
+        if (ph.v2_is_compressed && cm.codec != CODEC_UNCOMPRESSED) {
+          decompress_into(cm.codec, page.data(), uncomp_vals, body + lv,
+                          comp_vals);
+        } else {
+          std::memcpy(page.data(), body + lv, uncomp_vals);
+        }
+        vals = page.data();
+        vals_len = uncomp_vals;
+      }
+      if (page_nulls) any_null_page = true;
+      out.null_count += page_nulls;
+      int64_t nonnull = nv - page_nulls;
+
+      if (ph.encoding == ENC_PLAIN) {
+        if (cd.physical == PT_BOOLEAN) {
+          // unpack bits to bytes here (cheap, host)
+          size_t base = out.values.size();
+          out.values.resize(base + (size_t)nonnull);
+          for (int64_t i = 0; i < nonnull; i++)
+            out.values[base + i] = (vals[i >> 3] >> (i & 7)) & 1;
+        } else {
+          out.values.insert(out.values.end(), vals, vals + vals_len);
+        }
+      } else if (ph.encoding == ENC_RLE_DICTIONARY ||
+                 ph.encoding == ENC_PLAIN_DICTIONARY) {
+        out.is_dict = true;
+        int bw = vals[0];
+        IdxPage ip;
+        ip.out_off = values_seen;  // row offset (incl. nulls); indices are dense
+        ip.n = nonnull;
+        ip.payload_off = (int64_t)out.values.size();
+        ip.payload_len = (int64_t)(vals_len - 1);
+        ip.bit_width = bw;
+        out.idx_pages.push_back(ip);
+        out.values.insert(out.values.end(), vals + 1, vals + vals_len);
+      } else {
+        throw std::runtime_error("unsupported data encoding " +
+                                 std::to_string(ph.encoding));
+      }
+      values_seen += nv;
+    }
+    if (!any_null_page && out.null_count == 0) out.validity.clear();
+    return out;
+  }
+
+ private:
+  void build_columns() {
+    if (meta_.schema.empty()) throw std::runtime_error("empty schema");
+    // flat schema: every element after root with num_children==0
+    for (size_t i = 1; i < meta_.schema.size(); i++) {
+      const SchemaElement& e = meta_.schema[i];
+      if (e.num_children > 0)
+        throw std::runtime_error("nested schemas not supported yet: " + e.name);
+      ColumnDesc c;
+      c.name = e.name;
+      c.physical = e.type;
+      c.nullable = e.repetition == REP_OPTIONAL;
+      c.converted = e.converted;
+      c.logical = e.logical;
+      c.int_bit_width = e.int_bit_width;
+      c.int_signed = e.int_signed;
+      cols_.push_back(c);
+    }
+  }
+
+  std::string path_;
+  int fd_ = -1;
+  uint8_t* map_ = nullptr;
+  size_t size_ = 0;
+  FileMetaData meta_;
+  std::vector<ColumnDesc> cols_;
+};
+
+// ---------------------------------------------------------------------- //
+// CPU decode of a ChunkData into a full-length typed column
+// ---------------------------------------------------------------------- //
+
+struct DecodedColumn {
+  std::vector<uint8_t> data;      // elem_size * num_values (fixed width)
+  std::vector<int32_t> offsets;   // byte_array
+  std::vector<uint8_t> bytes;     // byte_array
+  std::vector<uint8_t> validity;  // empty if no nulls
+  int64_t num_values = 0;
+};
+
+inline DecodedColumn decode_chunk_cpu(const ParquetFile::ChunkData& ch) {
+  DecodedColumn out;
+  out.num_values = ch.num_values;
+  out.validity = ch.validity;
+  const uint8_t* validity = ch.validity.empty() ? nullptr : ch.validity.data();
+
+  // 1) materialize dense (non-null) values
+  std::vector<uint8_t> dense;         // fixed width
+  std::vector<int32_t> dense_offs;    // byte_array
+  std::vector<uint8_t> dense_bytes;
+  int es = physical_elem_size(ch.physical);
+  int64_t nonnull = ch.num_values - ch.null_count;
+
+  if (!ch.is_dict) {
+    if (ch.physical == PT_BYTE_ARRAY) {
+      dense_offs.reserve(nonnull + 1);
+      dense_offs.push_back(0);
+      const uint8_t* p = ch.values.data();
+      const uint8_t* endp = p + ch.values.size();
+      while (p + 4 <= endp && (int64_t)dense_offs.size() <= nonnull) {
+        uint32_t len;
+        std::memcpy(&len, p, 4);
+        p += 4;
+        dense_bytes.insert(dense_bytes.end(), p, p + len);
+        p += len;
+        dense_offs.push_back((int32_t)dense_bytes.size());
+      }
+    } else {
+      dense = ch.values;
+    }
+  } else {
+    // dict: decode indices then gather
+    std::vector<int32_t> idx((size_t)nonnull);
+    for (auto& ip : ch.idx_pages) {
+      // out_off counts rows incl. nulls; compute dense offset by counting
+      // valid rows before out_off
+      int64_t dense_off = 0;
+      if (validity) {
+        for (int64_t i = 0; i < ip.out_off; i++) dense_off += validity[i];
+      } else {
+        dense_off = ip.out_off;
+      }
+      rle_decode<int32_t>(ch.values.data() + ip.payload_off,
+                          (size_t)ip.payload_len, ip.bit_width, ip.n,
+                          idx.data() + dense_off);
+    }
+    // parse dict values
+    if (ch.physical == PT_BYTE_ARRAY) {
+      std::vector<int32_t> doffs;
+      std::vector<uint8_t> dbytes;
+      doffs.push_back(0);
+      const uint8_t* p = ch.dict.data();
+      for (int64_t i = 0; i < ch.dict_num_values; i++) {
+        uint32_t len;
+        std::memcpy(&len, p, 4);
+        p += 4;
+        dbytes.insert(dbytes.end(), p, p + len);
+        p += len;
+        doffs.push_back((int32_t)dbytes.size());
+      }
+      dense_offs.push_back(0);
+      for (int64_t i = 0; i < nonnull; i++) {
+        int32_t k = idx[i];
+        dense_bytes.insert(dense_bytes.end(), dbytes.data() + doffs[k],
+                           dbytes.data() + doffs[k + 1]);
+        dense_offs.push_back((int32_t)dense_bytes.size());
+      }
+    } else {
+      int des = es;
+      dense.resize((size_t)nonnull * des);
+      for (int64_t i = 0; i < nonnull; i++)
+        std::memcpy(dense.data() + i * des, ch.dict.data() + idx[i] * des, des);
+    }
+  }
+
+  // 2) scatter through validity to full length
+  if (ch.physical == PT_BYTE_ARRAY) {
+    out.offsets.resize(ch.num_values + 1);
+    out.bytes = std::move(dense_bytes);
+    int64_t di = 0;
+    out.offsets[0] = 0;
+    for (int64_t i = 0; i < ch.num_values; i++) {
+      if (!validity || validity[i]) {
+        out.offsets[i + 1] = dense_offs[di + 1];
+        di++;
+      } else {
+        out.offsets[i + 1] = out.offsets[i];
+      }
+    }
+  } else {
+    out.data.resize((size_t)ch.num_values * es, 0);
+    if (!validity) {
+      std::memcpy(out.data.data(), dense.data(),
+                  std::min(dense.size(), out.data.size()));
+    } else {
+      int64_t di = 0;
+      for (int64_t i = 0; i < ch.num_values; i++) {
+        if (validity[i]) {
+          std::memcpy(out.data.data() + i * es, dense.data() + di * es, es);
+          di++;
+        }
+      }
+    }
+  }
+  return out;
+}
+
+}  // namespace lakesoul

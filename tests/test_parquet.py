@@ -1,0 +1,216 @@
+"""Parquet interop: our native writer/reader vs pyarrow (golden oracle).
+
+This is the writer x reader consistency matrix idea from the reference's
+``python/tests/compat``, with pyarrow standing in as the
+foreign engine.
+"""
+
+import numpy as np
+import pyarrow as pa
+import pyarrow.parquet as pq
+import pytest
+import torch
+
+from lakesoul_amd.ops import cpp
+
+
+def _write_ours(path, names, dtypes, columns, offsets=None, validity=None,
+                nullable=None, rg=250000, codec=6, level=1):
+    n = len(names)
+    offsets = offsets or [None] * n
+    validity = validity or [None] * n
+    if nullable is None:
+        nullable = [v is not None for v in validity]
+    return cpp().write_parquet(
+        str(path), names, dtypes, columns, offsets, validity, nullable, rg, codec, level
+    )
+
+
+def _read_ours_cpu(path):
+    """Read a parquet file fully via the native CPU path -> dict of numpy."""
+    h = cpp().open_parquet(str(path))
+    try:
+        meta = cpp().parquet_meta(h)
+        cols = {c["name"]: c for c in meta["columns"]}
+        out = {}
+        for ci, cinfo in enumerate(meta["columns"]):
+            parts = []
+            masks = []
+            str_parts = []
+            for rg in range(meta["num_row_groups"]):
+                d = cpp().read_chunk_cpu(h, rg, ci)
+                nv = d["num_values"]
+                dt = cinfo["dtype"]
+                if dt in ("string", "binary"):
+                    offs = d["offsets"].numpy()
+                    bys = d["bytes"].numpy().tobytes()
+                    vals = [bys[offs[i]:offs[i + 1]] for i in range(nv)]
+                    str_parts.extend(vals)
+                else:
+                    npdt = {
+                        "bool": np.uint8,
+                        "int8": np.int32,
+                        "int16": np.int32,
+                        "int32": np.int32,
+                        "int64": np.int64,
+                        "float32": np.float32,
+                        "float64": np.float64,
+                        "date32": np.int32,
+                        "timestamp[us]": np.int64,
+                        "timestamp[ms]": np.int64,
+                    }[dt]
+                    parts.append(d["data"].numpy().view(npdt))
+                v = d["validity"].numpy()
+                masks.append(v if len(v) else np.ones(nv, dtype=np.uint8))
+            name = cinfo["name"]
+            if cinfo["dtype"] in ("string", "binary"):
+                out[name] = (str_parts, np.concatenate(masks))
+            else:
+                out[name] = (np.concatenate(parts), np.concatenate(masks))
+        return out, meta
+    finally:
+        cpp().close_parquet(h)
+
+
+def test_roundtrip_ours_to_pyarrow(tmp_path):
+    n = 10000
+    rng = np.random.default_rng(0)
+    ids = np.arange(n, dtype=np.int64)
+    vals = rng.normal(size=n)
+    f32 = rng.normal(size=n).astype(np.float32)
+    i32 = rng.integers(-1000, 1000, n, dtype=np.int32)
+    path = tmp_path / "ours.parquet"
+    size = _write_ours(
+        path,
+        ["id", "v", "f", "k"],
+        ["int64", "float64", "float32", "int32"],
+        [torch.from_numpy(ids), torch.from_numpy(vals), torch.from_numpy(f32), torch.from_numpy(i32)],
+        nullable=[False, False, False, False],
+    )
+    assert size > 0
+    t = pq.read_table(str(path))
+    assert t.num_rows == n
+    np.testing.assert_array_equal(t["id"].to_numpy(), ids)
+    np.testing.assert_array_equal(t["v"].to_numpy(), vals)
+    np.testing.assert_array_equal(t["f"].to_numpy(), f32)
+    np.testing.assert_array_equal(t["k"].to_numpy(), i32)
+
+
+def test_roundtrip_with_nulls_and_strings(tmp_path):
+    n = 5000
+    rng = np.random.default_rng(1)
+    ids = np.arange(n, dtype=np.int64)
+    vals = rng.normal(size=n)
+    mask = (rng.random(n) > 0.3).astype(np.uint8)
+    strings = [f"row-{i}" if i % 3 else "" for i in range(n)]
+    joined = "".join(strings).encode()
+    offs = np.zeros(n + 1, dtype=np.int32)
+    offs[1:] = np.cumsum([len(s.encode()) for s in strings]).astype(np.int32)
+    path = tmp_path / "ours2.parquet"
+    _write_ours(
+        path,
+        ["id", "v", "s"],
+        ["int64", "float64", "string"],
+        [
+            torch.from_numpy(ids),
+            torch.from_numpy(vals),
+            torch.from_numpy(np.frombuffer(joined, dtype=np.uint8).copy()),
+        ],
+        offsets=[None, None, torch.from_numpy(offs)],
+        validity=[None, torch.from_numpy(mask), None],
+        nullable=[False, True, False],
+    )
+    t = pq.read_table(str(path))
+    assert t.num_rows == n
+    np.testing.assert_array_equal(t["id"].to_numpy(), ids)
+    got_v = t["v"].to_pylist()
+    for i in range(n):
+        if mask[i]:
+            assert got_v[i] == pytest.approx(vals[i])
+        else:
+            assert got_v[i] is None
+    assert t["s"].to_pylist() == strings
+
+
+def test_roundtrip_bool_and_small_ints(tmp_path):
+    n = 1000
+    rng = np.random.default_rng(2)
+    flags = (rng.random(n) > 0.5).astype(np.uint8)
+    i16 = rng.integers(-300, 300, n, dtype=np.int16).astype(np.int32)
+    path = tmp_path / "ours3.parquet"
+    _write_ours(
+        path,
+        ["flag", "small"],
+        ["bool", "int16"],
+        [torch.from_numpy(flags), torch.from_numpy(i16)],
+        nullable=[False, False],
+    )
+    t = pq.read_table(str(path))
+    np.testing.assert_array_equal(t["flag"].to_numpy(), flags.astype(bool))
+    assert t["small"].type == pa.int16()
+    np.testing.assert_array_equal(t["small"].to_numpy().astype(np.int32), i16)
+
+
+@pytest.mark.parametrize("compression", ["zstd", "snappy", "none"])
+@pytest.mark.parametrize("dictionary", [True, False])
+def test_read_pyarrow_files(tmp_path, compression, dictionary):
+    n = 20000
+    rng = np.random.default_rng(3)
+    tbl = pa.table(
+        {
+            "id": pa.array(np.arange(n, dtype=np.int64)),
+            "v": pa.array(rng.normal(size=n)),
+            "cat": pa.array((np.arange(n) % 50).astype(np.int32)),
+            "s": pa.array([f"name_{i % 100}" for i in range(n)]),
+        }
+    )
+    path = tmp_path / "pa.parquet"
+    pq.write_table(
+        tbl,
+        str(path),
+        compression=compression,
+        use_dictionary=dictionary,
+        row_group_size=7000,
+    )
+    data, meta = _read_ours_cpu(path)
+    assert meta["num_rows"] == n
+    np.testing.assert_array_equal(data["id"][0], np.arange(n, dtype=np.int64))
+    np.testing.assert_allclose(data["v"][0], tbl["v"].to_numpy())
+    np.testing.assert_array_equal(data["cat"][0], tbl["cat"].to_numpy())
+    assert [b.decode() for b in data["s"][0]] == tbl["s"].to_pylist()
+
+
+def test_read_pyarrow_with_nulls(tmp_path):
+    n = 9999
+    rng = np.random.default_rng(4)
+    raw = rng.normal(size=n)
+    vals = [None if rng.random() < 0.25 else float(raw[i]) for i in range(n)]
+    tbl = pa.table({"id": np.arange(n, dtype=np.int64), "v": pa.array(vals, type=pa.float64())})
+    path = tmp_path / "pa_nulls.parquet"
+    pq.write_table(tbl, str(path), compression="zstd", use_dictionary=False, row_group_size=4000)
+    data, meta = _read_ours_cpu(path)
+    got, mask = data["v"]
+    for i in range(n):
+        if vals[i] is None:
+            assert mask[i] == 0
+        else:
+            assert mask[i] == 1
+            assert got[i] == pytest.approx(vals[i])
+
+
+def test_multiple_row_groups_and_stats(tmp_path):
+    n = 600000  # > 2 row groups at 250k
+    ids = np.arange(n, dtype=np.int64)
+    path = tmp_path / "rg.parquet"
+    _write_ours(path, ["id"], ["int64"], [torch.from_numpy(ids)], nullable=[False])
+    h = cpp().open_parquet(str(path))
+    meta = cpp().parquet_meta(h)
+    assert meta["num_row_groups"] == 3
+    s0 = meta["row_groups"][0]["columns"][0]
+    assert np.frombuffer(s0["min"], dtype=np.int64)[0] == 0
+    assert np.frombuffer(s0["max"], dtype=np.int64)[0] == 249999
+    cpp().close_parquet(h)
+    # pyarrow agrees
+    pf = pq.ParquetFile(str(path))
+    assert pf.metadata.num_row_groups == 3
+    assert pf.metadata.row_group(0).column(0).statistics.min == 0
