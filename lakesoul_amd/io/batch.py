@@ -1,0 +1,305 @@
+"""Column batch container — the in-memory currency of the engine.
+
+Fixed-width columns are torch tensors (CPU or HBM-resident on GPU);
+string/binary columns are (offsets int32, bytes uint8) pairs; NULLs are a
+uint8 validity mask. The GPU reader materializes these directly in HBM;
+``to_arrow``/``from_any`` bridge to pyarrow/pandas/numpy at the edges
+(replacing the reference's Arrow C Data FFI plane).
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Sequence
+
+import numpy as np
+import torch
+
+from .schema import FIXED_WIDTH_BYTES, Field as LsField, Schema
+
+_TORCH_DTYPE = {
+    "bool": torch.uint8,
+    "int8": torch.int8,
+    "int16": torch.int16,
+    "int32": torch.int32,
+    "int64": torch.int64,
+    "float16": torch.float16,
+    "float32": torch.float32,
+    "float64": torch.float64,
+    "date32": torch.int32,
+    "timestamp[us]": torch.int64,
+    "timestamp[ms]": torch.int64,
+    "timestamp[ns]": torch.int64,
+}
+
+_NP_DTYPE = {
+    "bool": np.uint8,
+    "int8": np.int8,
+    "int16": np.int16,
+    "int32": np.int32,
+    "int64": np.int64,
+    "float32": np.float32,
+    "float64": np.float64,
+    "date32": np.int32,
+    "timestamp[us]": np.int64,
+    "timestamp[ms]": np.int64,
+    "timestamp[ns]": np.int64,
+}
+
+
+def torch_dtype_for(dtype: str):
+    return _TORCH_DTYPE[dtype]
+
+
+@dataclass
+class Column:
+    dtype: str
+    data: Optional[torch.Tensor] = None      # fixed-width values
+    offsets: Optional[torch.Tensor] = None   # int32 [n+1], string/binary
+    bytes_: Optional[torch.Tensor] = None    # uint8, string/binary
+    validity: Optional[torch.Tensor] = None  # uint8 [n], 1=valid; None=all valid
+
+    @property
+    def is_string(self) -> bool:
+        return self.dtype in ("string", "binary")
+
+    def __len__(self) -> int:
+        if self.is_string:
+            return int(self.offsets.numel()) - 1
+        return int(self.data.numel())
+
+    def to_numpy(self):
+        if self.is_string:
+            raise TypeError("string column: use offsets/bytes")
+        return self.data.cpu().numpy()
+
+    def to_device(self, device) -> "Column":
+        def mv(t):
+            return None if t is None else t.to(device, non_blocking=True)
+
+        return Column(self.dtype, mv(self.data), mv(self.offsets), mv(self.bytes_), mv(self.validity))
+
+    def take(self, idx: torch.Tensor) -> "Column":
+        """Gather rows by index tensor (same device)."""
+        if self.is_string:
+            offs = self.offsets
+            lens = offs[1:] - offs[:-1]
+            new_lens = lens[idx]
+            new_offs = torch.zeros(idx.numel() + 1, dtype=torch.int32, device=offs.device)
+            torch.cumsum(new_lens, 0, out=new_offs[1:].view(-1))
+            # gather bytes (CPU loop-free path via numpy for now)
+            if offs.device.type == "cpu":
+                src_off = offs.numpy()
+                src_b = self.bytes_.numpy()
+                sel = idx.numpy()
+                out = np.empty(int(new_offs[-1]), dtype=np.uint8)
+                no = new_offs.numpy()
+                for i, si in enumerate(sel):
+                    out[no[i]:no[i + 1]] = src_b[src_off[si]:src_off[si + 1]]
+                nb = torch.from_numpy(out)
+            else:
+                from ..ops import hip
+
+                nb = hip().gather_strings(self.bytes_, self.offsets.to(torch.int64), idx.to(torch.int64), new_offs.to(torch.int64))
+            v = None if self.validity is None else self.validity[idx]
+            return Column(self.dtype, None, new_offs, nb, v)
+        v = None if self.validity is None else self.validity[idx]
+        return Column(self.dtype, self.data[idx], None, None, v)
+
+
+@dataclass
+class Batch:
+    schema: Schema
+    columns: Dict[str, Column] = field(default_factory=dict)
+
+    @property
+    def num_rows(self) -> int:
+        for c in self.columns.values():
+            return len(c)
+        return 0
+
+    def column(self, name: str) -> Column:
+        return self.columns[name]
+
+    def to_device(self, device) -> "Batch":
+        return Batch(self.schema, {k: v.to_device(device) for k, v in self.columns.items()})
+
+    def take(self, idx: torch.Tensor) -> "Batch":
+        return Batch(self.schema, {k: v.take(idx) for k, v in self.columns.items()})
+
+    # ------------------------------------------------------------------ #
+
+    @classmethod
+    def from_any(cls, data, schema: Schema) -> "Batch":
+        """Build from pyarrow Table/RecordBatch, pandas DataFrame, or a
+        dict of numpy arrays / torch tensors / lists."""
+        mod = type(data).__module__
+        if mod.startswith("pyarrow"):
+            return cls.from_arrow(data, schema)
+        if mod.startswith("pandas"):
+            return cls.from_dict({c: data[c].to_numpy() for c in data.columns}, schema)
+        if isinstance(data, dict):
+            return cls.from_dict(data, schema)
+        raise TypeError(f"unsupported data type {type(data)}")
+
+    @classmethod
+    def from_dict(cls, d: dict, schema: Schema) -> "Batch":
+        cols = {}
+        for f in schema:
+            if f.name not in d:
+                raise KeyError(f"missing column {f.name}")
+            v = d[f.name]
+            if f.is_fixed_width:
+                if isinstance(v, torch.Tensor):
+                    t = v
+                elif isinstance(v, np.ndarray):
+                    if v.dtype == np.bool_:
+                        v = v.astype(np.uint8)
+                    t = torch.from_numpy(np.ascontiguousarray(v))
+                else:
+                    t = torch.from_numpy(np.asarray(v, dtype=_NP_DTYPE[f.dtype]))
+                t = t.to(_TORCH_DTYPE[f.dtype])
+                cols[f.name] = Column(f.dtype, data=t)
+            else:
+                if isinstance(v, tuple) and len(v) in (2, 3):
+                    offs, by = v[0], v[1]
+                    val = v[2] if len(v) == 3 else None
+                    cols[f.name] = Column(
+                        f.dtype,
+                        offsets=torch.as_tensor(offs, dtype=torch.int32),
+                        bytes_=torch.as_tensor(by, dtype=torch.uint8),
+                        validity=None if val is None else torch.as_tensor(val, dtype=torch.uint8),
+                    )
+                else:
+                    # list of str/bytes/None
+                    items = list(v)
+                    validity = None
+                    if any(x is None for x in items):
+                        validity = torch.tensor(
+                            [0 if x is None else 1 for x in items], dtype=torch.uint8
+                        )
+                    enc = [
+                        b"" if x is None else (x.encode() if isinstance(x, str) else bytes(x))
+                        for x in items
+                    ]
+                    offs = np.zeros(len(enc) + 1, dtype=np.int32)
+                    offs[1:] = np.cumsum([len(e) for e in enc])
+                    cols[f.name] = Column(
+                        f.dtype,
+                        offsets=torch.from_numpy(offs),
+                        bytes_=torch.from_numpy(
+                            np.frombuffer(b"".join(enc), dtype=np.uint8).copy()
+                        ),
+                        validity=validity,
+                    )
+        return cls(schema, cols)
+
+    @classmethod
+    def from_arrow(cls, t, schema: Schema) -> "Batch":
+        import pyarrow as pa
+
+        if isinstance(t, pa.RecordBatch):
+            t = pa.Table.from_batches([t])
+        d = {}
+        for f in schema:
+            col = t.column(f.name)
+            arr = col.combine_chunks() if hasattr(col, "combine_chunks") else col
+            if f.is_fixed_width:
+                np_arr = arr.to_numpy(zero_copy_only=False)
+                validity = None
+                if arr.null_count:
+                    validity = (~np.asarray(arr.is_null())).astype(np.uint8)
+                    np_arr = np.nan_to_num(np_arr) if np_arr.dtype.kind == "f" else np_arr
+                    if np_arr.dtype == object:
+                        np_arr = np.where(validity, np_arr, 0).astype(_NP_DTYPE[f.dtype])
+                if np_arr.dtype == object or (np_arr.dtype.kind == "f" and f.dtype.startswith("int")):
+                    np_arr = np.asarray(
+                        [0 if x is None else x for x in arr.to_pylist()],
+                        dtype=_NP_DTYPE[f.dtype],
+                    )
+                b = cls.from_dict({f.name: np_arr.astype(_NP_DTYPE[f.dtype])}, Schema([f]))
+                c = b.columns[f.name]
+                if validity is not None:
+                    c.validity = torch.from_numpy(validity)
+                d[f.name] = c
+            else:
+                d[f.name] = None  # placeholder, handled below
+                items = arr.to_pylist()
+                b = cls.from_dict({f.name: items}, Schema([f]))
+                d[f.name] = b.columns[f.name]
+        return cls(schema, d)
+
+    def to_arrow(self):
+        import pyarrow as pa
+
+        from .schema import dtype_to_arrow
+
+        arrays = []
+        for f in self.schema:
+            c = self.columns[f.name]
+            if f.is_fixed_width:
+                np_arr = c.data.cpu().numpy()
+                if f.dtype == "bool":
+                    np_arr = np_arr.astype(bool)
+                mask = None
+                if c.validity is not None:
+                    mask = ~(c.validity.cpu().numpy().astype(bool))
+                arrays.append(pa.array(np_arr, type=dtype_to_arrow(f.dtype), from_pandas=False, mask=mask))
+            else:
+                offs = c.offsets.cpu().numpy()
+                bys = c.bytes_.cpu().numpy().tobytes()
+                items = []
+                val = c.validity.cpu().numpy() if c.validity is not None else None
+                for i in range(len(c)):
+                    if val is not None and not val[i]:
+                        items.append(None)
+                    else:
+                        raw = bys[offs[i]:offs[i + 1]]
+                        items.append(raw.decode() if f.dtype == "string" else raw)
+                arrays.append(pa.array(items, type=dtype_to_arrow(f.dtype)))
+        return pa.table(dict(zip(self.schema.names(), arrays)))
+
+
+def concat_batches(batches: List[Batch]) -> Batch:
+    if len(batches) == 1:
+        return batches[0]
+    schema = batches[0].schema
+    cols = {}
+    for f in schema:
+        cs = [b.columns[f.name] for b in batches]
+        n_total = sum(len(c) for c in cs)
+        if f.is_fixed_width:
+            data = torch.cat([c.data for c in cs])
+            validity = None
+            if any(c.validity is not None for c in cs):
+                validity = torch.cat(
+                    [
+                        c.validity
+                        if c.validity is not None
+                        else torch.ones(len(c), dtype=torch.uint8, device=data.device)
+                        for c in cs
+                    ]
+                )
+            cols[f.name] = Column(f.dtype, data=data, validity=validity)
+        else:
+            bytes_ = torch.cat([c.bytes_ for c in cs])
+            offs = torch.zeros(n_total + 1, dtype=torch.int32, device=bytes_.device)
+            pos = 0
+            base = 0
+            for c in cs:
+                n = len(c)
+                offs[pos + 1 : pos + n + 1] = c.offsets[1:] + base
+                base += int(c.offsets[-1])
+                pos += n
+            validity = None
+            if any(c.validity is not None for c in cs):
+                validity = torch.cat(
+                    [
+                        c.validity
+                        if c.validity is not None
+                        else torch.ones(len(c), dtype=torch.uint8, device=bytes_.device)
+                        for c in cs
+                    ]
+                )
+            cols[f.name] = Column(f.dtype, offsets=offs, bytes_=bytes_, validity=validity)
+    return Batch(schema, cols)

@@ -1,0 +1,271 @@
+"""End-to-end table tests on the CPU path: write / upsert / MOR scan /
+merge operators / CDC / compaction / time travel / sharding."""
+
+import numpy as np
+import pandas as pd
+import pytest
+
+from lakesoul_amd.io.schema import Field, Schema
+
+
+def _mk_pk_table(catalog, name="t", buckets=4, extra=(), props=None):
+    fields = [Field("id", "int64", False), Field("v", "float64"), Field("s", "string")]
+    fields += list(extra)
+    return catalog.create_table(
+        name,
+        Schema(fields),
+        primary_keys=["id"],
+        hash_bucket_num=buckets,
+        properties=props,
+    )
+
+
+def _df(tbl, **kwargs):
+    df = tbl.to_pandas(**kwargs)
+    return df.sort_values("id").reset_index(drop=True)
+
+
+def test_write_scan_roundtrip_no_pk(catalog):
+    t = catalog.create_table(
+        "plain", Schema([Field("a", "int64"), Field("b", "float64")])
+    )
+    data = {"a": np.arange(100, dtype=np.int64), "b": np.ones(100)}
+    t.write(data)
+    df = t.to_pandas()
+    assert len(df) == 100
+    np.testing.assert_array_equal(np.sort(df["a"].to_numpy()), np.arange(100))
+    assert t.scan().count() == 100
+
+
+def test_upsert_use_last(catalog):
+    t = _mk_pk_table(catalog, "up1")
+    n = 10000
+    rng = np.random.default_rng(0)
+    base = {
+        "id": np.arange(n, dtype=np.int64),
+        "v": rng.normal(size=n),
+        "s": [f"s{i}" for i in range(n)],
+    }
+    t.upsert(base)
+    # upsert 2000 overlapping keys + 500 new
+    up_ids = np.concatenate([rng.choice(n, 2000, replace=False), np.arange(n, n + 500)])
+    up = {
+        "id": up_ids.astype(np.int64),
+        "v": np.full(len(up_ids), 99.0),
+        "s": [f"u{i}" for i in range(len(up_ids))],
+    }
+    t.upsert(up)
+
+    df = _df(t)
+    assert len(df) == n + 500
+    # expected via pandas
+    expect = pd.concat(
+        [pd.DataFrame(base), pd.DataFrame(up)], ignore_index=True
+    ).groupby("id", as_index=False).last().sort_values("id").reset_index(drop=True)
+    np.testing.assert_array_equal(df["id"].to_numpy(), expect["id"].to_numpy())
+    np.testing.assert_allclose(df["v"].to_numpy(), expect["v"].to_numpy())
+    assert df["s"].tolist() == expect["s"].tolist()
+
+
+def test_multiple_upserts_many_files(catalog):
+    t = _mk_pk_table(catalog, "up2", buckets=2)
+    n = 2000
+    state = {}
+    rng = np.random.default_rng(1)
+    ids0 = np.arange(n, dtype=np.int64)
+    t.upsert({"id": ids0, "v": np.zeros(n), "s": ["a"] * n})
+    for i in ids0:
+        state[int(i)] = 0.0
+    for it in range(5):
+        ids = rng.choice(n, 300, replace=False).astype(np.int64)
+        vals = np.full(300, float(it + 1))
+        t.upsert({"id": ids, "v": vals, "s": [f"it{it}"] * 300})
+        for i in ids:
+            state[int(i)] = float(it + 1)
+    df = _df(t)
+    assert len(df) == n
+    expect_v = np.array([state[i] for i in df["id"]])
+    np.testing.assert_allclose(df["v"].to_numpy(), expect_v)
+
+
+def test_merge_operator_sum_all(catalog):
+    t = catalog.create_table(
+        "sums",
+        Schema([Field("id", "int64", False), Field("cnt", "int64")]),
+        primary_keys=["id"],
+        hash_bucket_num=2,
+        properties={"merge_op.cnt": "SumAll"},
+    )
+    t.upsert({"id": np.array([1, 2, 3], dtype=np.int64), "cnt": np.array([10, 20, 30], dtype=np.int64)})
+    t.upsert({"id": np.array([2, 3, 4], dtype=np.int64), "cnt": np.array([1, 2, 3], dtype=np.int64)})
+    df = _df(t)
+    got = dict(zip(df["id"], df["cnt"]))
+    assert got == {1: 10, 2: 21, 3: 32, 4: 3}
+
+
+def test_merge_operator_use_last_not_null(catalog):
+    t = catalog.create_table(
+        "ulnn",
+        Schema([Field("id", "int64", False), Field("x", "float64")]),
+        primary_keys=["id"],
+        properties={"merge_op.x": "UseLastNotNull"},
+    )
+    t.upsert({"id": np.array([1, 2], dtype=np.int64), "x": np.array([1.5, 2.5])})
+    import pyarrow as pa
+
+    t.upsert(pa.table({"id": pa.array([1, 2], pa.int64()), "x": pa.array([None, 7.5], pa.float64())}))
+    df = _df(t)
+    assert df["x"].tolist() == [1.5, 7.5]
+
+
+def test_partial_column_upsert(catalog):
+    t = catalog.create_table(
+        "partial",
+        Schema([Field("id", "int64", False), Field("a", "float64"), Field("b", "float64")]),
+        primary_keys=["id"],
+    )
+    t.upsert({"id": np.array([1, 2], dtype=np.int64), "a": np.array([1.0, 2.0]), "b": np.array([10.0, 20.0])})
+    # partial upsert: only column a — b keeps old values
+    t.upsert({"id": np.array([1, 2], dtype=np.int64), "a": np.array([5.0, 6.0])})
+    df = _df(t)
+    assert df["a"].tolist() == [5.0, 6.0]
+    assert df["b"].tolist() == [10.0, 20.0]
+
+
+def test_cdc_delete_rows_dropped(catalog):
+    t = catalog.create_table(
+        "cdc",
+        Schema([Field("id", "int64", False), Field("v", "float64"), Field("rowKinds", "string")]),
+        primary_keys=["id"],
+        properties={"lakesoul_cdc_change_column": "rowKinds"},
+    )
+    t.upsert({"id": np.array([1, 2, 3], dtype=np.int64), "v": np.ones(3), "rowKinds": ["insert"] * 3})
+    t.upsert({"id": np.array([2], dtype=np.int64), "v": np.array([0.0]), "rowKinds": ["delete"]})
+    df = _df(t)
+    assert df["id"].tolist() == [1, 3]
+
+
+def test_compaction_preserves_data(catalog, tmp_path):
+    t = _mk_pk_table(catalog, "comp", buckets=2)
+    n = 3000
+    rng = np.random.default_rng(2)
+    t.upsert({"id": np.arange(n, dtype=np.int64), "v": np.zeros(n), "s": ["x"] * n})
+    for it in range(4):
+        ids = rng.choice(n, 200, replace=False).astype(np.int64)
+        t.upsert({"id": ids, "v": np.full(200, it + 1.0), "s": ["y"] * 200})
+    before = _df(t)
+    files_before = t.files()
+    t.compaction()
+    after = _df(t)
+    files_after = t.files()
+    pd.testing.assert_frame_equal(before, after)
+    assert len(files_after) < len(files_before)
+    assert all("compactdir" in f.path for f in files_after)
+    # scan after further upsert on top of compacted base
+    t.upsert({"id": np.array([0], dtype=np.int64), "v": np.array([123.0]), "s": ["z"]})
+    df = _df(t)
+    assert df.loc[df["id"] == 0, "v"].iloc[0] == 123.0
+
+
+def test_time_travel(catalog):
+    t = _mk_pk_table(catalog, "tt", buckets=1)
+    t.upsert({"id": np.array([1], dtype=np.int64), "v": np.array([1.0]), "s": ["a"]})
+    t.upsert({"id": np.array([1], dtype=np.int64), "v": np.array([2.0]), "s": ["b"]})
+    assert _df(t)["v"].tolist() == [2.0]
+    assert _df(t, version=0)["v"].tolist() == [1.0]
+
+
+def test_bucket_pruning_point_lookup(catalog):
+    t = _mk_pk_table(catalog, "prune", buckets=8)
+    n = 5000
+    t.upsert({"id": np.arange(n, dtype=np.int64), "v": np.arange(n, dtype=np.float64), "s": ["x"] * n})
+    scan = t.scan(filters=[("id", "==", 1234)])
+    units = scan.plan()
+    assert len(units) == 1  # only the matching bucket
+    df = scan.to_arrow().to_pandas()
+    assert len(df) == 1 and df["v"].iloc[0] == 1234.0
+
+
+def test_shard_disjoint_cover(catalog):
+    t = _mk_pk_table(catalog, "shard", buckets=8)
+    n = 4000
+    t.upsert({"id": np.arange(n, dtype=np.int64), "v": np.zeros(n), "s": ["x"] * n})
+    ids = []
+    for rank in range(4):
+        df = t.scan().shard(rank, 4).to_arrow().to_pandas()
+        ids.append(df["id"].to_numpy())
+    allids = np.sort(np.concatenate(ids))
+    np.testing.assert_array_equal(allids, np.arange(n))
+
+
+def test_string_pk(catalog):
+    t = catalog.create_table(
+        "spk",
+        Schema([Field("k", "string", False), Field("v", "int64")]),
+        primary_keys=["k"],
+        hash_bucket_num=4,
+    )
+    t.upsert({"k": [f"key{i}" for i in range(100)], "v": np.arange(100, dtype=np.int64)})
+    t.upsert({"k": ["key5", "key50", "zzz"], "v": np.array([500, 5000, 1], dtype=np.int64)})
+    df = t.to_pandas().sort_values("k").reset_index(drop=True)
+    assert len(df) == 101
+    got = dict(zip(df["k"], df["v"]))
+    assert got["key5"] == 500 and got["key50"] == 5000 and got["zzz"] == 1
+    assert got["key6"] == 6
+
+
+def test_multi_pk(catalog):
+    t = catalog.create_table(
+        "mpk",
+        Schema([Field("a", "int64", False), Field("b", "int32", False), Field("v", "float64")]),
+        primary_keys=["a", "b"],
+        hash_bucket_num=4,
+    )
+    t.upsert({"a": np.array([1, 1, 2], dtype=np.int64), "b": np.array([1, 2, 1], dtype=np.int32), "v": np.array([1.0, 2.0, 3.0])})
+    t.upsert({"a": np.array([1], dtype=np.int64), "b": np.array([2], dtype=np.int32), "v": np.array([9.0])})
+    df = t.to_pandas().sort_values(["a", "b"]).reset_index(drop=True)
+    assert df["v"].tolist() == [1.0, 9.0, 3.0]
+
+
+def test_range_partitions(catalog):
+    t = catalog.create_table(
+        "ranged",
+        Schema([Field("dt", "string", False), Field("id", "int64", False), Field("v", "float64")]),
+        primary_keys=["id"],
+        range_partitions=["dt"],
+        hash_bucket_num=2,
+    )
+    t.upsert(
+        {
+            "dt": ["2024-01-01"] * 3 + ["2024-01-02"] * 2,
+            "id": np.arange(5, dtype=np.int64),
+            "v": np.arange(5, dtype=np.float64),
+        }
+    )
+    descs = t.partition_descs()
+    assert sorted(descs) == ["dt=2024-01-01", "dt=2024-01-02"]
+    df = t.to_pandas().sort_values("id")
+    assert len(df) == 5
+    assert df["dt"].tolist() == ["2024-01-01"] * 3 + ["2024-01-02"] * 2
+    # partition-pruned scan
+    df1 = t.scan(partitions=["dt=2024-01-02"]).to_arrow().to_pandas()
+    assert sorted(df1["id"].tolist()) == [3, 4]
+
+
+def test_incremental_read(catalog):
+    t = _mk_pk_table(catalog, "incr", buckets=1)
+    t.upsert({"id": np.array([1, 2], dtype=np.int64), "v": np.array([1.0, 2.0]), "s": ["a", "b"]})
+    t.upsert({"id": np.array([3], dtype=np.int64), "v": np.array([3.0]), "s": ["c"]})
+    t.upsert({"id": np.array([4], dtype=np.int64), "v": np.array([4.0]), "s": ["d"]})
+    from lakesoul_amd.io.reader import LakeSoulScan
+
+    scan = LakeSoulScan(t, incremental=(0, 2))
+    df = scan.to_arrow().to_pandas().sort_values("id")
+    assert df["id"].tolist() == [3, 4]
+
+
+def test_delete_partition(catalog):
+    t = _mk_pk_table(catalog, "delp", buckets=1)
+    t.upsert({"id": np.array([1], dtype=np.int64), "v": np.array([1.0]), "s": ["a"]})
+    t.delete_partition("-5")
+    assert t.scan().count() == 0
