@@ -344,6 +344,45 @@ static py::list read_chunks_raw_batch(int64_t h,
   return result;
 }
 
+// prep dictionary-index RLE runs for the GPU expansion kernel:
+// idx_pages int64 [p,5] = {row_off, n_nonnull, payload_off, payload_len, bw}
+// returns (payload padded +8 bytes, runs int64 [m,5] =
+//          {dense_out_off, n, is_literal, value_or_abs_bitoff, bit_width},
+//          total_dense_n)
+static py::tuple prep_rle_runs(torch::Tensor values, torch::Tensor idx_pages) {
+  auto v = values.contiguous();
+  auto pages = idx_pages.contiguous();
+  const uint8_t* base = v.data_ptr<uint8_t>();
+  auto pa = pages.accessor<int64_t, 2>();
+  std::vector<RleRun> all;
+  int64_t dense_off = 0;
+  std::vector<int64_t> bws;
+  for (int64_t p = 0; p < pages.size(0); p++) {
+    int64_t n = pa[p][1], off = pa[p][2], len = pa[p][3];
+    int bw = (int)pa[p][4];
+    std::vector<RleRun> runs;
+    parse_rle_runs(base + off, (size_t)len, bw, n, off * 8, runs);
+    for (auto& r : runs) {
+      r.out_off += dense_off;
+      all.push_back(r);
+      bws.push_back(bw);
+    }
+    dense_off += n;
+  }
+  auto runs_t = torch::empty({(int64_t)all.size(), 5}, torch::kInt64);
+  auto ra = runs_t.accessor<int64_t, 2>();
+  for (size_t i = 0; i < all.size(); i++) {
+    ra[i][0] = all[i].out_off;
+    ra[i][1] = all[i].n;
+    ra[i][2] = all[i].is_literal;
+    ra[i][3] = all[i].is_literal ? all[i].bit_off : (int64_t)all[i].value;
+    ra[i][4] = bws[i];
+  }
+  auto padded = torch::zeros({v.numel() + 8}, torch::kUInt8);
+  std::memcpy(padded.data_ptr(), base, v.numel());
+  return py::make_tuple(padded, runs_t, dense_off);
+}
+
 // ---------------------------------------------------------------------- //
 // murmur3 (CPU)
 // ---------------------------------------------------------------------- //
@@ -466,6 +505,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("read_chunk_raw", &read_chunk_raw);
   m.def("read_chunk_cpu", &read_chunk_cpu);
   m.def("read_chunks_cpu_batch", &read_chunks_cpu_batch);
+  m.def("prep_rle_runs", &prep_rle_runs);
   m.def("read_chunks_raw_batch", &read_chunks_raw_batch);
   m.def("hash_columns_cpu", &hash_columns_cpu);
   m.def("hash_string_column_cpu", &hash_string_column_cpu);

@@ -16,6 +16,7 @@ def cpp():
     global _cpp
     if _cpp is None:
         try:
+            import torch  # noqa: F401  (loads libc10 et al. for the ext)
             from lakesoul_amd import _cpp as mod  # type: ignore
         except ImportError as e:
             raise ImportError(
@@ -30,6 +31,7 @@ def hip():
     global _hip
     if _hip is None:
         try:
+            import torch  # noqa: F401  (loads libc10/libtorch_hip for the ext)
             from lakesoul_amd import _hip as mod  # type: ignore
         except ImportError as e:
             raise ImportError(
