@@ -281,10 +281,10 @@ static std::vector<torch::Tensor> segmented_sum(torch::Tensor vals, torch::Tenso
 static torch::Tensor segmented_last(torch::Tensor grp, torch::Tensor contrib,
                                     torch::Tensor validity, int64_t ngroups, int64_t n) {
   CHECK_GPU(grp);
-  auto out = torch::full({ngroups}, -1, grp.options());
+  auto out = torch::zeros({ngroups}, grp.options());
   launch_segmented_last(grp.data_ptr<int64_t>(), opt_u8(contrib), opt_u8(validity),
                         out.data_ptr<int64_t>(), n, cur_stream());
-  return out;
+  return out - 1;  // -1 = no contributing row
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

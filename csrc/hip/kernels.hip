@@ -388,18 +388,20 @@ __global__ void segmented_sum_kernel(const T* __restrict__ vals,
   }
 }
 
-// last contributing (and optionally non-null) row index per group
+// last contributing (and optionally non-null) row index per group.
+// out_idx is zero-initialized and stores (i+1) — unsigned atomicMax can't
+// start from -1 (0xFF..F would already be maximal). 0 means "no row".
 __global__ void segmented_last_kernel(const int64_t* __restrict__ group_of_row,
                                       const uint8_t* __restrict__ contrib,
                                       const uint8_t* __restrict__ validity,
-                                      int64_t* __restrict__ out_idx,  // init -1
+                                      int64_t* __restrict__ out_idx,  // init 0
                                       int64_t n) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     if (contrib && !contrib[i]) continue;
     if (validity && !validity[i]) continue;
     atomicMax((unsigned long long*)&out_idx[group_of_row[i]],
-              (unsigned long long)i);
+              (unsigned long long)(i + 1));
   }
 }
 
