@@ -11,6 +11,7 @@
 
 #include "murmur3.h"
 #include "parquet_file.h"
+#include "read_unit.h"
 
 namespace py = pybind11;
 using namespace lakesoul;
@@ -369,7 +370,7 @@ static py::tuple prep_rle_runs(torch::Tensor values, torch::Tensor idx_pages) {
     }
     dense_off += n;
   }
-  auto runs_t = torch::empty({(int64_t)all.size(), 5}, torch::kInt64);
+  auto runs_t = torch::empty({(int64_t)all.size(), 6}, torch::kInt64);
   auto ra = runs_t.accessor<int64_t, 2>();
   for (size_t i = 0; i < all.size(); i++) {
     ra[i][0] = all[i].out_off;
@@ -377,10 +378,78 @@ static py::tuple prep_rle_runs(torch::Tensor values, torch::Tensor idx_pages) {
     ra[i][2] = all[i].is_literal;
     ra[i][3] = all[i].is_literal ? all[i].bit_off : (int64_t)all[i].value;
     ra[i][4] = bws[i];
+    ra[i][5] = 0;
   }
   auto padded = torch::zeros({v.numel() + 8}, torch::kUInt8);
   std::memcpy(padded.data_ptr(), base, v.numel());
   return py::make_tuple(padded, runs_t, dense_off);
+}
+
+// read all files of a scan unit in one call; big contiguous buffers for
+// single-H2D GPU decode (see read_unit.h). ``pin`` requests pinned host
+// memory for the big buffers (GPU boxes).
+static py::dict read_unit_raw_py(const std::vector<std::string>& paths,
+                                 const std::vector<std::string>& names,
+                                 int64_t nthreads, bool pin) {
+  UnitData ud;
+  {
+    py::gil_scoped_release rel;
+    ud = read_unit_raw(paths, names, (int)nthreads);
+  }
+  auto mk_u8 = [&](std::vector<uint8_t>& v) {
+    auto opts = torch::TensorOptions().dtype(torch::kUInt8);
+    torch::Tensor t;
+    if (pin) {
+      try {
+        t = torch::empty({(int64_t)v.size()}, opts.pinned_memory(true));
+      } catch (...) {
+        t = torch::empty({(int64_t)v.size()}, opts);
+      }
+    } else {
+      t = torch::empty({(int64_t)v.size()}, opts);
+    }
+    if (!v.empty()) std::memcpy(t.data_ptr(), v.data(), v.size());
+    return t;
+  };
+  auto mk_i64 = [&](std::vector<int64_t>& v) {
+    auto t = torch::empty({(int64_t)v.size()}, torch::kInt64);
+    if (!v.empty()) std::memcpy(t.data_ptr(), v.data(), v.size() * 8);
+    return t;
+  };
+  py::dict d;
+  d["values"] = mk_u8(ud.values);
+  d["validity"] = mk_u8(ud.validity);
+  d["dicts"] = mk_u8(ud.dicts);
+  d["runs"] = mk_i64(ud.runs);
+  d["soffs"] = mk_i64(ud.soffs);
+  py::list frows;
+  for (auto r : ud.file_rows) frows.append(r);
+  d["file_rows"] = frows;
+  py::list cols;
+  for (auto& c : ud.cols) {
+    py::dict cd;
+    cd["file_idx"] = c.file_idx;
+    cd["name"] = c.name;
+    cd["present"] = c.present;
+    cd["is_string"] = c.is_string;
+    cd["is_dict"] = c.is_dict;
+    cd["num_values"] = c.num_values;
+    cd["null_count"] = c.null_count;
+    cd["val_off"] = c.val_off;
+    cd["val_len"] = c.val_len;
+    cd["validity_off"] = c.validity_off;
+    cd["dict_off"] = c.dict_off;
+    cd["dict_len"] = c.dict_len;
+    cd["run_off"] = c.run_off;
+    cd["run_cnt"] = c.run_cnt;
+    cd["dense_n"] = c.dense_n;
+    cd["soff_off"] = c.soff_off;
+    cd["sbytes_off"] = c.sbytes_off;
+    cd["sbytes_len"] = c.sbytes_len;
+    cols.append(cd);
+  }
+  d["cols"] = cols;
+  return d;
 }
 
 // ---------------------------------------------------------------------- //
@@ -506,6 +575,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("read_chunk_cpu", &read_chunk_cpu);
   m.def("read_chunks_cpu_batch", &read_chunks_cpu_batch);
   m.def("prep_rle_runs", &prep_rle_runs);
+  m.def("read_unit_raw", &read_unit_raw_py);
   m.def("read_chunks_raw_batch", &read_chunks_raw_batch);
   m.def("hash_columns_cpu", &hash_columns_cpu);
   m.def("hash_string_column_cpu", &hash_string_column_cpu);

@@ -96,7 +96,9 @@ __global__ void bucket_ids_kernel(const int64_t* __restrict__ hashes,
 // RLE / bit-packed hybrid expansion (dictionary indices, levels)
 // ===================================================================== //
 
-// runs: int64 [nruns][5] = {out_off, n, is_literal, value_or_bitoff, bit_width}
+// runs: int64 [nruns][6] =
+//   {out_off, n, is_literal, value_or_bitoff, bit_width, dict_elem_bias}
+// repeat-run values arrive pre-biased; literal unpacks get bias added here.
 // payload must be padded by >=8 bytes past the last literal bit.
 __global__ void rle_expand_kernel(const uint8_t* __restrict__ payload,
                                   const int64_t* __restrict__ runs,
@@ -108,10 +110,10 @@ __global__ void rle_expand_kernel(const uint8_t* __restrict__ payload,
     int64_t lo = 0, hi = nruns - 1;
     while (lo < hi) {
       int64_t mid = (lo + hi + 1) >> 1;
-      if (runs[mid * 5] <= i) lo = mid;
+      if (runs[mid * 6] <= i) lo = mid;
       else hi = mid - 1;
     }
-    const int64_t* r = runs + lo * 5;
+    const int64_t* r = runs + lo * 6;
     int64_t k = i - r[0];
     if (r[2]) {  // literal bit-packed group
       int bit_width = (int)r[4];
@@ -119,7 +121,7 @@ __global__ void rle_expand_kernel(const uint8_t* __restrict__ payload,
       uint64_t bitpos = (uint64_t)r[3] + (uint64_t)k * bit_width;
       uint64_t w;
       __builtin_memcpy(&w, payload + (bitpos >> 3), 8);
-      out[i] = (int32_t)((w >> (bitpos & 7)) & mask);
+      out[i] = (int32_t)(((w >> (bitpos & 7)) & mask) + (uint64_t)r[5]);
     } else {
       out[i] = (int32_t)r[3];
     }

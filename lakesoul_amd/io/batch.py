@@ -85,7 +85,7 @@ class Column:
             offs = self.offsets
             lens = offs[1:] - offs[:-1]
             new_lens = lens[idx]
-            new_offs = torch.zeros(idx.numel() + 1, dtype=torch.int32, device=offs.device)
+            new_offs = torch.zeros(idx.numel() + 1, dtype=offs.dtype, device=offs.device)
             torch.cumsum(new_lens, 0, out=new_offs[1:].view(-1))
             # gather bytes (CPU loop-free path via numpy for now)
             if offs.device.type == "cpu":
@@ -283,7 +283,8 @@ def concat_batches(batches: List[Batch]) -> Batch:
             cols[f.name] = Column(f.dtype, data=data, validity=validity)
         else:
             bytes_ = torch.cat([c.bytes_ for c in cs])
-            offs = torch.zeros(n_total + 1, dtype=torch.int32, device=bytes_.device)
+            odt = cs[0].offsets.dtype
+            offs = torch.zeros(n_total + 1, dtype=odt, device=bytes_.device)
             pos = 0
             base = 0
             for c in cs:
