@@ -75,30 +75,43 @@ inline UnitData read_unit_raw(const std::vector<std::string>& paths,
     std::vector<int> col_idx;  // -1 if absent
   };
   std::vector<FileData> files(nfiles);
+  int nt_req = nthreads > 0 ? nthreads : (int)std::thread::hardware_concurrency();
+  if (nt_req < 1) nt_req = 1;
   {
+    // open + footer parse (serial — cheap), then one task per
+    // (file, column, row-group) chunk so a single large base file still
+    // decompresses across every core
+    struct Task {
+      size_t fi, c, rg;
+    };
+    std::vector<Task> tasks;
+    for (size_t i = 0; i < nfiles; i++) {
+      FileData& fd = files[i];
+      fd.f = std::make_unique<ParquetFile>(paths[i]);
+      fd.chunks.resize(names.size());
+      fd.col_idx.resize(names.size());
+      size_t nrg = fd.f->num_row_groups();
+      for (size_t c = 0; c < names.size(); c++) {
+        int ci = fd.f->column_index(names[c]);
+        fd.col_idx[c] = ci;
+        if (ci < 0) continue;
+        fd.chunks[c].resize(nrg);
+        for (size_t rg = 0; rg < nrg; rg++) tasks.push_back({i, c, rg});
+      }
+    }
     std::atomic<size_t> next{0};
     std::string err;
     std::mutex err_mu;
-    int nt = nthreads > 0 ? nthreads : (int)std::thread::hardware_concurrency();
-    if (nt > (int)nfiles * 2) nt = (int)nfiles;
+    int nt = std::min<int>(nt_req, (int)tasks.size());
     if (nt < 1) nt = 1;
     auto worker = [&]() {
       while (true) {
         size_t i = next.fetch_add(1);
-        if (i >= nfiles) break;
+        if (i >= tasks.size()) break;
         try {
-          FileData& fd = files[i];
-          fd.f = std::make_unique<ParquetFile>(paths[i]);
-          fd.chunks.resize(names.size());
-          fd.col_idx.resize(names.size());
-          size_t nrg = fd.f->num_row_groups();
-          for (size_t c = 0; c < names.size(); c++) {
-            int ci = fd.f->column_index(names[c]);
-            fd.col_idx[c] = ci;
-            if (ci < 0) continue;
-            for (size_t rg = 0; rg < nrg; rg++)
-              fd.chunks[c].push_back(fd.f->read_chunk(rg, ci));
-          }
+          const Task& t = tasks[i];
+          FileData& fd = files[t.fi];
+          fd.chunks[t.c][t.rg] = fd.f->read_chunk(t.rg, fd.col_idx[t.c]);
         } catch (std::exception& e) {
           std::lock_guard<std::mutex> lk(err_mu);
           err = e.what();
@@ -216,73 +229,85 @@ inline UnitData read_unit_raw(const std::vector<std::string>& paths,
   out.dicts.resize((size_t)dpos);
   out.soffs.resize((size_t)spos);
 
-  // phase C: fill buffers (parallel over unit-columns) + parse runs (serial
-  // for runs since they append to one vector; runs are tiny)
+  // phase C: dict columns first (serial — runs append to one shared
+  // vector and are tiny), then parallel memcpy for everything else
   for (size_t u = 0; u < out.cols.size(); u++) {
     UnitColumn& uc = out.cols[u];
-    if (!uc.present) continue;
+    if (!uc.present || !uc.is_dict) continue;
     FileData& fd = files[uc.file_idx];
-    size_t c = u % names.size();
-    auto& chs = fd.chunks[c];
-
-    if (uc.validity_off >= 0) {
-      int64_t off = uc.validity_off;
-      for (auto& ch : chs) {
-        if (!ch.validity.empty())
-          std::memcpy(out.validity.data() + off, ch.validity.data(),
-                      ch.validity.size());
-        off += ch.num_values;
-      }
-    }
-    if (uc.is_string) {
-      auto& sd = *str_cols[u];
-      std::memcpy(out.soffs.data() + uc.soff_off, sd.offs.data(),
-                  sd.offs.size() * 8);
-      if (!sd.bytes.empty())
-        std::memcpy(out.values.data() + uc.sbytes_off, sd.bytes.data(),
-                    sd.bytes.size());
-      continue;
-    }
-    if (uc.is_dict) {
-      uc.run_off = rpos / 6;
-      int64_t poff = uc.val_off;
-      int64_t doff = uc.dict_off;
-      int64_t dense_off = 0;
-      int64_t dict_elem_bias = 0;
-      int es = physical_elem_size(uc.physical);
-      for (auto& ch : chs) {
-        std::memcpy(out.values.data() + poff, ch.values.data(), ch.values.size());
-        std::memcpy(out.dicts.data() + doff, ch.dict.data(), ch.dict.size());
-        // parse runs per idx page
-        for (auto& ip : ch.idx_pages) {
-          std::vector<RleRun> rr;
-          parse_rle_runs(ch.values.data() + ip.payload_off, (size_t)ip.payload_len,
-                         ip.bit_width, ip.n, (poff + ip.payload_off) * 8, rr);
-          for (auto& r : rr) {
-            out.runs.push_back(r.out_off + dense_off);
-            out.runs.push_back(r.n);
-            out.runs.push_back(r.is_literal);
-            out.runs.push_back(r.is_literal ? r.bit_off
-                                            : (int64_t)r.value + dict_elem_bias);
-            out.runs.push_back(ip.bit_width);
-            out.runs.push_back(uc.is_dict ? dict_elem_bias : 0);
-            rpos += 6;
-          }
-          dense_off += ip.n;
-        }
-        poff += align8((int64_t)ch.values.size());
-        doff += align8((int64_t)ch.dict.size());
-        dict_elem_bias += (int64_t)(align8((int64_t)ch.dict.size()) / es);
-      }
-      uc.run_cnt = rpos / 6 - uc.run_off;
-      continue;
-    }
-    // plain fixed: contiguous copy
-    int64_t off = uc.val_off;
+    auto& chs = fd.chunks[u % names.size()];
+    uc.run_off = rpos / 6;
+    int64_t poff = uc.val_off;
+    int64_t doff = uc.dict_off;
+    int64_t dense_off = 0;
+    int64_t dict_elem_bias = 0;
+    int es = physical_elem_size(uc.physical);
     for (auto& ch : chs) {
-      std::memcpy(out.values.data() + off, ch.values.data(), ch.values.size());
-      off += (int64_t)ch.values.size();
+      std::memcpy(out.values.data() + poff, ch.values.data(), ch.values.size());
+      std::memcpy(out.dicts.data() + doff, ch.dict.data(), ch.dict.size());
+      for (auto& ip : ch.idx_pages) {
+        std::vector<RleRun> rr;
+        parse_rle_runs(ch.values.data() + ip.payload_off, (size_t)ip.payload_len,
+                       ip.bit_width, ip.n, (poff + ip.payload_off) * 8, rr);
+        for (auto& r : rr) {
+          out.runs.push_back(r.out_off + dense_off);
+          out.runs.push_back(r.n);
+          out.runs.push_back(r.is_literal);
+          out.runs.push_back(r.is_literal ? r.bit_off
+                                          : (int64_t)r.value + dict_elem_bias);
+          out.runs.push_back(ip.bit_width);
+          out.runs.push_back(dict_elem_bias);
+          rpos += 6;
+        }
+        dense_off += ip.n;
+      }
+      poff += align8((int64_t)ch.values.size());
+      doff += align8((int64_t)ch.dict.size());
+      dict_elem_bias += (int64_t)(align8((int64_t)ch.dict.size()) / es);
     }
+    uc.run_cnt = rpos / 6 - uc.run_off;
+  }
+  {
+    std::atomic<size_t> next{0};
+    int nt = std::min<int>(nt_req, (int)out.cols.size());
+    if (nt < 1) nt = 1;
+    auto worker = [&]() {
+      while (true) {
+        size_t u = next.fetch_add(1);
+        if (u >= out.cols.size()) break;
+        UnitColumn& uc = out.cols[u];
+        if (!uc.present) continue;
+        FileData& fd = files[uc.file_idx];
+        auto& chs = fd.chunks[u % names.size()];
+        if (uc.validity_off >= 0) {
+          int64_t off = uc.validity_off;
+          for (auto& ch : chs) {
+            if (!ch.validity.empty())
+              std::memcpy(out.validity.data() + off, ch.validity.data(),
+                          ch.validity.size());
+            off += ch.num_values;
+          }
+        }
+        if (uc.is_dict) continue;  // handled above
+        if (uc.is_string) {
+          auto& sd = *str_cols[u];
+          std::memcpy(out.soffs.data() + uc.soff_off, sd.offs.data(),
+                      sd.offs.size() * 8);
+          if (!sd.bytes.empty())
+            std::memcpy(out.values.data() + uc.sbytes_off, sd.bytes.data(),
+                        sd.bytes.size());
+          continue;
+        }
+        int64_t off = uc.val_off;
+        for (auto& ch : chs) {
+          std::memcpy(out.values.data() + off, ch.values.data(), ch.values.size());
+          off += (int64_t)ch.values.size();
+        }
+      }
+    };
+    std::vector<std::thread> ts;
+    for (int t = 0; t < nt; t++) ts.emplace_back(worker);
+    for (auto& t : ts) t.join();
   }
   return out;
 }

@@ -178,8 +178,12 @@ class LakeSoulScan:
     # ------------------------------------------------------------------ #
 
     def iter_batches(self) -> Iterator[Batch]:
-        for unit in self.plan():
-            batch = self._read_unit(unit)
+        units = self.plan()
+        if self.device == "cuda" and len(units) > 1:
+            gen = self._iter_units_pipelined(units)
+        else:
+            gen = (self._read_unit(u) for u in units)
+        for batch in gen:
             if batch is None:
                 continue
             batch = self._apply_filters(batch)
@@ -192,6 +196,27 @@ class LakeSoulScan:
                     yield batch.take(idx)
             else:
                 yield batch
+
+    def _iter_units_pipelined(self, units: List[ScanUnit], depth: int = 2):
+        """GPU path: prefetch the host IO/decompress of the next units on
+        background threads while the GPU decodes+merges the current one
+        (overlap engineering)."""
+        from concurrent.futures import ThreadPoolExecutor
+
+        from .reader_gpu import fetch_raw, read_unit_gpu
+
+        with ThreadPoolExecutor(max_workers=depth) as ex:
+            futs = [
+                ex.submit(fetch_raw, u.files, self.read_cols)
+                for u in units[: depth]
+            ]
+            for i, unit in enumerate(units):
+                raw = futs[i].result()
+                if i + depth < len(units):
+                    futs.append(
+                        ex.submit(fetch_raw, units[i + depth].files, self.read_cols)
+                    )
+                yield read_unit_gpu(self, unit, raw)
 
     def __iter__(self):
         return self.iter_batches()

@@ -43,12 +43,19 @@ _TARGET = {"int8": torch.int8, "int16": torch.int16}
 _ESIZE = {torch.uint8: 1, torch.int32: 4, torch.int64: 8, torch.float32: 4, torch.float64: 8}
 
 
-def read_unit_gpu(scan, unit) -> Optional[Batch]:
+def fetch_raw(files: List[str], names: List[str]) -> dict:
+    """Host phase of a unit read (releases the GIL in C++) — safe to run
+    on a prefetch thread while the GPU processes the previous unit."""
+    return cpp().read_unit_raw(files, names, 0, True)
+
+
+def read_unit_gpu(scan, unit, raw: Optional[dict] = None) -> Optional[Batch]:
     from .merge_gpu import merge_sorted_files_gpu
 
     device = torch.device("cuda")
     names = scan.read_cols
-    raw = cpp().read_unit_raw(unit.files, names, 0, True)
+    if raw is None:
+        raw = fetch_raw(unit.files, names)
 
     vals = raw["values"].to(device, non_blocking=True)
     validity_buf = (
