@@ -127,6 +127,22 @@ class Batch:
     def take(self, idx: torch.Tensor) -> "Batch":
         return Batch(self.schema, {k: v.take(idx) for k, v in self.columns.items()})
 
+    def slice(self, a: int, b: int) -> "Batch":
+        """Zero-copy row slice [a, b) (views; strings rebase offsets)."""
+        cols = {}
+        for k, c in self.columns.items():
+            v = None if c.validity is None else c.validity[a:b]
+            if c.is_string:
+                offs = c.offsets[a : b + 1]
+                base = offs[0]
+                lo = int(base)
+                hi = int(c.offsets[b])
+                cols[k] = Column(c.dtype, None, (offs - base).to(c.offsets.dtype),
+                                 c.bytes_[lo:hi], v)
+            else:
+                cols[k] = Column(c.dtype, c.data[a:b], None, None, v)
+        return Batch(self.schema, cols)
+
     # ------------------------------------------------------------------ #
 
     @classmethod

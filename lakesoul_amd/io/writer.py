@@ -256,24 +256,43 @@ def write_table_data(table, data, device: Optional[str] = None,
         os.makedirs(out_dir, exist_ok=True)
 
         if pk:
+            # one global stable sort by (bucket, pk): buckets become
+            # contiguous PK-sorted slices — a single gather for the whole
+            # batch instead of one per bucket
             buckets = _hash_bucket_ids(file_batch, pk, num_buckets)
+            order = _sort_indices(file_batch, pk)
+            border = torch.argsort(buckets.to(torch.int64)[order], stable=True)
+            perm = order[border]
+            sorted_batch = file_batch.take(perm)
+            bucket_sorted = buckets[perm].to(torch.int64)
+            counts = torch.bincount(bucket_sorted, minlength=num_buckets)
+            bounds = torch.zeros(num_buckets + 1, dtype=torch.int64)
+            torch.cumsum(counts.cpu(), 0, out=bounds[1:].view(-1))
+            bounds = bounds.tolist()
+
+            jobs = []
             for b in range(num_buckets):
-                sel = torch.nonzero(buckets == b, as_tuple=True)[0]
-                if sel.numel() == 0:
+                a, e = bounds[b], bounds[b + 1]
+                if a == e:
                     continue
-                bucket_batch = file_batch.take(sel)
-                order = _sort_indices(bucket_batch, pk)
-                bucket_batch = bucket_batch.take(order)
+                bucket_batch = sorted_batch.slice(a, e)
                 fname = f"part-{random_str(16)}_{b:04d}.parquet"
                 fpath = os.path.join(out_dir, fname)
+                jobs.append((fpath, bucket_batch))
+
+            from concurrent.futures import ThreadPoolExecutor
+
+            def _encode(job):
+                fpath, bb = job
                 size = _write_batch_to_file(
-                    fpath, bucket_batch, cfg.compression, cfg.compression_level,
+                    fpath, bb, cfg.compression, cfg.compression_level,
                     cfg.max_row_group_size,
                 )
-                results.append(
-                    FlushResult(fpath, size, bucket_batch.num_rows, desc,
-                                ",".join(data_schema.names()))
-                )
+                return FlushResult(fpath, size, bb.num_rows, desc,
+                                   ",".join(data_schema.names()))
+
+            with ThreadPoolExecutor(max_workers=min(8, max(1, len(jobs)))) as ex:
+                results.extend(ex.map(_encode, jobs))
         else:
             fname = f"part-{random_str(16)}_{0:04d}.parquet"
             fpath = os.path.join(out_dir, fname)
