@@ -283,27 +283,17 @@ static py::list read_chunks_cpu_batch(int64_t h,
   std::vector<DecodedColumn> out(rc.size());
   {
     py::gil_scoped_release rel;
-    std::atomic<size_t> next{0};
-    int nt = (int)std::min<int64_t>(nthreads > 0 ? nthreads : 8, (int64_t)rc.size());
-    std::vector<std::thread> threads;
     std::string err;
     std::mutex err_mu;
-    for (int t = 0; t < nt; t++) {
-      threads.emplace_back([&]() {
-        while (true) {
-          size_t i = next.fetch_add(1);
-          if (i >= rc.size()) break;
-          try {
-            auto ch = f->read_chunk((size_t)rc[i].first, (size_t)rc[i].second);
-            out[i] = decode_chunk_cpu(ch);
-          } catch (std::exception& e) {
-            std::lock_guard<std::mutex> lk(err_mu);
-            err = e.what();
-          }
-        }
-      });
-    }
-    for (auto& th : threads) th.join();
+    ThreadPool::instance().parallel_for((int64_t)rc.size(), [&](int64_t i) {
+      try {
+        auto ch = f->read_chunk((size_t)rc[i].first, (size_t)rc[i].second);
+        out[i] = decode_chunk_cpu(ch);
+      } catch (std::exception& e) {
+        std::lock_guard<std::mutex> lk(err_mu);
+        err = e.what();
+      }
+    });
     if (!err.empty()) throw std::runtime_error(err);
   }
   py::list result;
@@ -318,26 +308,16 @@ static py::list read_chunks_raw_batch(int64_t h,
   std::vector<ParquetFile::ChunkData> out(rc.size());
   {
     py::gil_scoped_release rel;
-    std::atomic<size_t> next{0};
-    int nt = (int)std::min<int64_t>(nthreads > 0 ? nthreads : 8, (int64_t)rc.size());
-    std::vector<std::thread> threads;
     std::string err;
     std::mutex err_mu;
-    for (int t = 0; t < nt; t++) {
-      threads.emplace_back([&]() {
-        while (true) {
-          size_t i = next.fetch_add(1);
-          if (i >= rc.size()) break;
-          try {
-            out[i] = f->read_chunk((size_t)rc[i].first, (size_t)rc[i].second);
-          } catch (std::exception& e) {
-            std::lock_guard<std::mutex> lk(err_mu);
-            err = e.what();
-          }
-        }
-      });
-    }
-    for (auto& th : threads) th.join();
+    ThreadPool::instance().parallel_for((int64_t)rc.size(), [&](int64_t i) {
+      try {
+        out[i] = f->read_chunk((size_t)rc[i].first, (size_t)rc[i].second);
+      } catch (std::exception& e) {
+        std::lock_guard<std::mutex> lk(err_mu);
+        err = e.what();
+      }
+    });
     if (!err.empty()) throw std::runtime_error(err);
   }
   py::list result;

@@ -27,6 +27,7 @@
 
 #include "parquet_file.h"
 #include "rle.h"
+#include "thread_pool.h"
 
 namespace lakesoul {
 
@@ -99,28 +100,18 @@ inline UnitData read_unit_raw(const std::vector<std::string>& paths,
         for (size_t rg = 0; rg < nrg; rg++) tasks.push_back({i, c, rg});
       }
     }
-    std::atomic<size_t> next{0};
     std::string err;
     std::mutex err_mu;
-    int nt = std::min<int>(nt_req, (int)tasks.size());
-    if (nt < 1) nt = 1;
-    auto worker = [&]() {
-      while (true) {
-        size_t i = next.fetch_add(1);
-        if (i >= tasks.size()) break;
-        try {
-          const Task& t = tasks[i];
-          FileData& fd = files[t.fi];
-          fd.chunks[t.c][t.rg] = fd.f->read_chunk(t.rg, fd.col_idx[t.c]);
-        } catch (std::exception& e) {
-          std::lock_guard<std::mutex> lk(err_mu);
-          err = e.what();
-        }
+    ThreadPool::instance().parallel_for((int64_t)tasks.size(), [&](int64_t i) {
+      try {
+        const Task& t = tasks[i];
+        FileData& fd = files[t.fi];
+        fd.chunks[t.c][t.rg] = fd.f->read_chunk(t.rg, fd.col_idx[t.c]);
+      } catch (std::exception& e) {
+        std::lock_guard<std::mutex> lk(err_mu);
+        err = e.what();
       }
-    };
-    std::vector<std::thread> ts;
-    for (int t = 0; t < nt; t++) ts.emplace_back(worker);
-    for (auto& t : ts) t.join();
+    });
     if (!err.empty()) throw std::runtime_error(err);
   }
 
@@ -268,15 +259,9 @@ inline UnitData read_unit_raw(const std::vector<std::string>& paths,
     uc.run_cnt = rpos / 6 - uc.run_off;
   }
   {
-    std::atomic<size_t> next{0};
-    int nt = std::min<int>(nt_req, (int)out.cols.size());
-    if (nt < 1) nt = 1;
-    auto worker = [&]() {
-      while (true) {
-        size_t u = next.fetch_add(1);
-        if (u >= out.cols.size()) break;
+    ThreadPool::instance().parallel_for((int64_t)out.cols.size(), [&](int64_t u) {
         UnitColumn& uc = out.cols[u];
-        if (!uc.present) continue;
+        if (!uc.present) return;
         FileData& fd = files[uc.file_idx];
         auto& chs = fd.chunks[u % names.size()];
         if (uc.validity_off >= 0) {
@@ -288,7 +273,7 @@ inline UnitData read_unit_raw(const std::vector<std::string>& paths,
             off += ch.num_values;
           }
         }
-        if (uc.is_dict) continue;  // handled above
+        if (uc.is_dict) return;  // handled above
         if (uc.is_string) {
           auto& sd = *str_cols[u];
           std::memcpy(out.soffs.data() + uc.soff_off, sd.offs.data(),
@@ -296,19 +281,16 @@ inline UnitData read_unit_raw(const std::vector<std::string>& paths,
           if (!sd.bytes.empty())
             std::memcpy(out.values.data() + uc.sbytes_off, sd.bytes.data(),
                         sd.bytes.size());
-          continue;
+          return;
         }
         int64_t off = uc.val_off;
         for (auto& ch : chs) {
           std::memcpy(out.values.data() + off, ch.values.data(), ch.values.size());
           off += (int64_t)ch.values.size();
         }
-      }
-    };
-    std::vector<std::thread> ts;
-    for (int t = 0; t < nt; t++) ts.emplace_back(worker);
-    for (auto& t : ts) t.join();
+    });
   }
+  (void)nt_req;
   return out;
 }
 
