@@ -35,6 +35,7 @@ void launch_bytes_ne_mask(const int64_t*, const uint8_t*, const uint8_t*, int, u
 template <typename T, typename ACC>
 void launch_segmented_sum(const T*, const int64_t*, const uint8_t*, const uint8_t*, ACC*, int32_t*, int64_t, hipStream_t);
 void launch_segmented_last(const int64_t*, const uint8_t*, const uint8_t*, int64_t*, int64_t, hipStream_t);
+void launch_ann_scores(const short*, const short*, float*, int64_t, int32_t, int32_t, hipStream_t);
 
 }  // namespace lakesoul
 
@@ -287,7 +288,28 @@ static torch::Tensor segmented_last(torch::Tensor grp, torch::Tensor contrib,
   return out - 1;  // -1 = no contributing row
 }
 
+// ---- ANN scoring (MFMA) ------------------------------------------------ //
+
+static torch::Tensor ann_scores(torch::Tensor X, torch::Tensor Q) {
+  CHECK_GPU(X);
+  CHECK_GPU(Q);
+  TORCH_CHECK(X.scalar_type() == torch::kBFloat16 && Q.scalar_type() == torch::kBFloat16,
+              "ann_scores expects bf16");
+  int64_t n = X.size(0);
+  int64_t nq = Q.size(0);
+  int64_t K = X.size(1);
+  TORCH_CHECK(Q.size(1) == K, "dim mismatch");
+  TORCH_CHECK(K % 32 == 0, "K must be a multiple of 32");
+  TORCH_CHECK(nq % 16 == 0, "nq must be a multiple of 16 (pad queries)");
+  auto out = torch::empty({n, nq}, X.options().dtype(torch::kFloat32));
+  launch_ann_scores((const short*)X.data_ptr(), (const short*)Q.data_ptr(),
+                    out.data_ptr<float>(), n, (int32_t)nq, (int32_t)K,
+                    cur_stream());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("ann_scores", &ann_scores);
   m.doc() = "lakesoul_amd gfx950 HIP kernels";
   m.def("hash_fixed_column", &hash_fixed_column);
   m.def("hash_string_column", &hash_string_column);

