@@ -28,6 +28,7 @@ import torch
 from .. import constants
 from ..ops import cpp
 from .batch import Batch, Column
+from .filters import decode_stat, resolve_filters
 from .merge_cpu import NpColumn, merge_sorted_files
 from .schema import FIXED_WIDTH_BYTES, Schema
 
@@ -106,7 +107,7 @@ class LakeSoulScan:
         self.partitions = list(partitions) if partitions else None
         self.version = version
         self.timestamp_ms = timestamp_ms
-        self.filters = filters or []
+        self.filter_expr = resolve_filters(filters, self.schema)
         self.device = device or table.io_config().resolve_device()
         self.batch_size = batch_size
         self.incremental = incremental
@@ -130,6 +131,19 @@ class LakeSoulScan:
         client = self.table.client
         tid = self.table.table_id
         descs = self.partitions or client.all_partition_descs(tid)
+        # range-partition pruning from the filter expression (the
+        # reference's metadata partition filtering, helpers/mod.rs)
+        if self.filter_expr is not None:
+            kept = []
+            for desc in descs:
+                pv = {}
+                for kv in desc.split(","):
+                    if "=" in kv:
+                        k, v = kv.split("=", 1)
+                        pv[k] = v
+                if self.filter_expr.partition_prune(pv):
+                    kept.append(desc)
+            descs = kept
         units: List[ScanUnit] = []
         for desc in descs:
             if self.incremental is not None:
@@ -143,6 +157,10 @@ class LakeSoulScan:
                 b = extract_hash_bucket_id(op.path)
                 by_bucket.setdefault(b if b is not None else 0, []).append(op.path)
             for b, files in sorted(by_bucket.items()):
+                if self.filter_expr is not None:
+                    files = self.prune_files_by_stats(files)
+                    if not files:
+                        continue
                 is_comp = constants.COMPACT_DIR in files[0].split(os.sep) if files else False
                 units.append(ScanUnit(desc, b, files, is_comp))
         # PK point-filter bucket pruning (reader.rs:164-225)
@@ -156,14 +174,14 @@ class LakeSoulScan:
 
     def _bucket_filter(self) -> Optional[set]:
         """If filters pin every PK column to constants, only matching
-        buckets need scanning."""
-        if not self.pk:
+        buckets need scanning (reference: reader.rs:164-225)."""
+        if not self.pk or self.filter_expr is None:
             return None
-        eq: Dict[str, object] = {}
-        for f in self.filters:
-            col, op, val = f
-            if op in ("==", "=") and col in self.pk:
-                eq[col] = val
+        eq = {
+            k: v
+            for k, v in self.filter_expr.pk_eq_values().items()
+            if k in self.pk
+        }
         if set(eq.keys()) != set(self.pk):
             return None
         from ..utils import murmur3 as m3
@@ -235,7 +253,7 @@ class LakeSoulScan:
         """Count-only fast path (EmptyScanCountExec analog,
         physical_plan/empty_schema.rs:192): row counts come from parquet
         footers; PK tables still need the merge for dedup."""
-        if not self.pk and not self.filters:
+        if not self.pk and self.filter_expr is None:
             total = 0
             for unit in self.plan():
                 for f in unit.files:
@@ -402,65 +420,54 @@ class LakeSoulScan:
     # ------------------------------------------------------------------ #
 
     def _apply_filters(self, batch: Batch) -> Batch:
-        if not self.filters:
+        if self.filter_expr is None:
             return batch
-        n = batch.num_rows
-        dev = "cpu"
-        for c in batch.columns.values():
-            t = c.data if not c.is_string else c.bytes_
-            if t is not None:
-                dev = t.device
-                break
-        keep = torch.ones(n, dtype=torch.bool, device=dev)
-        for col, op, val in self.filters:
-            c = batch.columns[col]
-            if c.is_string:
-                # string compare on CPU
-                offs = c.offsets.cpu().numpy()
-                bys = c.bytes_.cpu().numpy().tobytes()
-                enc = val.encode() if isinstance(val, str) else val
-                m = np.array(
-                    [_cmp_bytes(bys[offs[i]:offs[i + 1]], op, enc) for i in range(n)]
-                )
-                keep &= torch.from_numpy(m).to(dev)
-            else:
-                t = c.data
-                if op in ("==", "="):
-                    m = t == val
-                elif op == "!=":
-                    m = t != val
-                elif op == "<":
-                    m = t < val
-                elif op == "<=":
-                    m = t <= val
-                elif op == ">":
-                    m = t > val
-                elif op == ">=":
-                    m = t >= val
-                elif op == "in":
-                    m = torch.isin(t, torch.tensor(list(val), device=t.device))
-                else:
-                    raise ValueError(f"unsupported filter op {op}")
-                if c.validity is not None:
-                    m &= c.validity.to(torch.bool)
-                keep &= m
+        keep = self.filter_expr.evaluate(batch)
         idx = torch.nonzero(keep, as_tuple=True)[0]
-        if idx.numel() == n:
+        if idx.numel() == batch.num_rows:
             return batch
         return batch.take(idx)
 
+    def prune_files_by_stats(self, files: List[str]) -> List[str]:
+        """File-level min/max pruning (the reference's Parquet row-group
+        statistics pushdown). For PK tables only PK-column predicates are
+        safe to prune on — dropping a delta file on a non-PK predicate
+        would resurrect stale rows in the merge."""
+        if self.filter_expr is None:
+            return files
+        from ..ops import cpp
 
-def _cmp_bytes(a: bytes, op: str, b: bytes) -> bool:
-    if op in ("==", "="):
-        return a == b
-    if op == "!=":
-        return a != b
-    if op == "<":
-        return a < b
-    if op == "<=":
-        return a <= b
-    if op == ">":
-        return a > b
-    if op == ">=":
-        return a >= b
-    raise ValueError(op)
+        out = []
+        for path in files:
+            try:
+                h = cpp().open_parquet(path)
+            except Exception:
+                out.append(path)
+                continue
+            try:
+                meta = cpp().parquet_meta(h)
+                cols = meta["columns"]
+                stats: Dict[str, tuple] = {}
+                for ci, cinfo in enumerate(cols):
+                    name = cinfo["name"]
+                    if self.pk and name not in self.pk:
+                        continue
+                    dtype = cinfo["dtype"]
+                    mn = mx = None
+                    ok = True
+                    for rg in meta["row_groups"]:
+                        sd = rg["columns"][ci]
+                        if "min" not in sd:
+                            ok = False
+                            break
+                        lo = decode_stat(sd["min"], dtype)
+                        hi = decode_stat(sd["max"], dtype)
+                        mn = lo if mn is None else min(mn, lo)
+                        mx = hi if mx is None else max(mx, hi)
+                    if ok and mn is not None:
+                        stats[name] = (mn, mx)
+                if self.filter_expr.prune_stats(stats):
+                    out.append(path)
+            finally:
+                cpp().close_parquet(h)
+        return out
