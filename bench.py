@@ -218,6 +218,11 @@ def main():
     ms_per_step = elapsed / args.steps * 1000.0
     value = total_rows_per_step / (elapsed / args.steps)
 
+    from lakesoul_amd.utils import timing as _timing
+
+    if _timing.ENABLED and rank == 0:
+        print("[timing]\n" + _timing.report(), file=sys.stderr, flush=True)
+
     if rank == 0:
         result = {
             "metric": "mor_scan_rows_per_sec",

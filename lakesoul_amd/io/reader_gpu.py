@@ -23,6 +23,7 @@ from typing import Dict, List, Optional, Tuple
 import torch
 
 from ..ops import cpp, hip
+from ..utils import timing
 from .batch import Batch, Column, torch_dtype_for
 from .schema import Schema
 
@@ -46,7 +47,8 @@ _ESIZE = {torch.uint8: 1, torch.int32: 4, torch.int64: 8, torch.float32: 4, torc
 def fetch_raw(files: List[str], names: List[str]) -> dict:
     """Host phase of a unit read (releases the GIL in C++) — safe to run
     on a prefetch thread while the GPU processes the previous unit."""
-    return cpp().read_unit_raw(files, names, 0, True)
+    with timing.phase("host_fetch"):
+        return cpp().read_unit_raw(files, names, 0, True)
 
 
 def read_unit_gpu(scan, unit, raw: Optional[dict] = None) -> Optional[Batch]:
@@ -57,7 +59,8 @@ def read_unit_gpu(scan, unit, raw: Optional[dict] = None) -> Optional[Batch]:
     if raw is None:
         raw = fetch_raw(unit.files, names)
 
-    vals = raw["values"].to(device, non_blocking=True)
+    with timing.phase("h2d", sync_gpu=True):
+        vals = raw["values"].to(device, non_blocking=True)
     validity_buf = (
         raw["validity"].to(device, non_blocking=True) if raw["validity"].numel() else None
     )
@@ -74,6 +77,8 @@ def read_unit_gpu(scan, unit, raw: Optional[dict] = None) -> Optional[Batch]:
 
     file_batches: List[Batch] = []
     present: List[set] = []
+    _dec = timing.phase("gpu_decode", sync_gpu=True)
+    _dec.__enter__()
     for fi, nrows in enumerate(raw["file_rows"]):
         cols: Dict[str, Column] = {}
         pres = set()
@@ -134,15 +139,17 @@ def read_unit_gpu(scan, unit, raw: Optional[dict] = None) -> Optional[Batch]:
             cols[name] = Column(f.dtype, data=data, validity=vmask)
         file_batches.append(Batch(read_schema, cols))
         present.append(pres)
+    _dec.__exit__(None, None, None)
 
     needs_merge = bool(scan.pk) and (
         len(file_batches) > 1 or scan.cdc_column is not None or bool(scan.merge_ops)
     )
     if needs_merge:
-        merged = merge_sorted_files_gpu(
-            file_batches, scan.pk, scan.merge_ops, scan.cdc_column, present
-        )
-    else:
+        with timing.phase("gpu_merge", sync_gpu=True):
+            merged = merge_sorted_files_gpu(
+                file_batches, scan.pk, scan.merge_ops, scan.cdc_column, present
+            )
+    elif True:
         from .batch import concat_batches
 
         merged = concat_batches(file_batches)
