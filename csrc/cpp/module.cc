@@ -371,37 +371,41 @@ static py::tuple prep_rle_runs(torch::Tensor values, torch::Tensor idx_pages) {
 static py::dict read_unit_raw_py(const std::vector<std::string>& paths,
                                  const std::vector<std::string>& names,
                                  int64_t nthreads, bool pin) {
-  UnitData ud;
+  (void)nthreads;
+  std::unique_ptr<UnitStage> st;
   {
     py::gil_scoped_release rel;
-    ud = read_unit_raw(paths, names, (int)nthreads);
+    st = read_unit_stage1(paths, names);
   }
-  auto mk_u8 = [&](std::vector<uint8_t>& v) {
+  UnitStage& ud = *st;
+  auto alloc_u8 = [&](int64_t n) {
     auto opts = torch::TensorOptions().dtype(torch::kUInt8);
-    torch::Tensor t;
     if (pin) {
       try {
-        t = torch::empty({(int64_t)v.size()}, opts.pinned_memory(true));
+        return torch::empty({n}, opts.pinned_memory(true));
       } catch (...) {
-        t = torch::empty({(int64_t)v.size()}, opts);
       }
-    } else {
-      t = torch::empty({(int64_t)v.size()}, opts);
     }
-    if (!v.empty()) std::memcpy(t.data_ptr(), v.data(), v.size());
-    return t;
+    return torch::empty({n}, opts);
   };
-  auto mk_i64 = [&](std::vector<int64_t>& v) {
-    auto t = torch::empty({(int64_t)v.size()}, torch::kInt64);
-    if (!v.empty()) std::memcpy(t.data_ptr(), v.data(), v.size() * 8);
-    return t;
-  };
+  torch::Tensor values = alloc_u8(ud.values_size);
+  torch::Tensor validity = alloc_u8(ud.validity_size);
+  torch::Tensor dicts = alloc_u8(ud.dicts_size);
+  torch::Tensor soffs = torch::empty({ud.soffs_size}, torch::kInt64);
+  torch::Tensor runs = torch::empty({(int64_t)ud.runs.size()}, torch::kInt64);
+  if (!ud.runs.empty())
+    std::memcpy(runs.data_ptr(), ud.runs.data(), ud.runs.size() * 8);
+  {
+    py::gil_scoped_release rel;
+    read_unit_fill(ud, (uint8_t*)values.data_ptr(), (uint8_t*)validity.data_ptr(),
+                   (uint8_t*)dicts.data_ptr(), soffs.data_ptr<int64_t>());
+  }
   py::dict d;
-  d["values"] = mk_u8(ud.values);
-  d["validity"] = mk_u8(ud.validity);
-  d["dicts"] = mk_u8(ud.dicts);
-  d["runs"] = mk_i64(ud.runs);
-  d["soffs"] = mk_i64(ud.soffs);
+  d["values"] = values;
+  d["validity"] = validity;
+  d["dicts"] = dicts;
+  d["runs"] = runs;
+  d["soffs"] = soffs;
   py::list frows;
   for (auto r : ud.file_rows) frows.append(r);
   d["file_rows"] = frows;

@@ -1,28 +1,34 @@
 // Unit reader: read ALL files of one scan unit (bucket) in one call.
 //
-// Parallelizes footer/page/zstd work across files and chunks, then packs
-// every chunk's payload into a handful of contiguous buffers so the GPU
-// path does ONE H2D per buffer instead of hundreds of small copies:
+// Two-stage design so the python binding can allocate pinned torch
+// tensors and have chunk payloads land in them directly (no second
+// copy):
+//   stage1: open files (via a footer/mmap cache — the analog of the
+//           reference's file-meta cache, session.rs:86-105), decompress
+//           every (file, column, row-group) chunk across the persistent
+//           thread pool, compute the buffer layout and RLE run table.
+//   fill:   parallel memcpy of chunk payloads straight into the caller's
+//           destination buffers.
 //
-//   values  : per-(file,col) dense PLAIN payloads, row-group chunks laid
-//             out contiguously (a no-null column is a single device view);
-//             dict-index payloads for dict-encoded chunks (+8 pad)
-//   validity: per-(file,col) validity bytes, contiguous across row groups
-//   dicts   : per-(file,col) concatenated dictionary values; per-row-group
-//             dictionaries get an element bias applied during expansion
-//   runs    : RLE run table for the GPU expansion kernel, int64 [m,6]:
-//             {dense_out_off, n, is_literal, value_or_abs_bitoff(values buf),
-//              bit_width, dict_elem_bias}
+// Buffer layout (8-byte aligned so torch dtype views work):
+//   values  : per-(file,col) dense PLAIN payloads, row-group chunks
+//             contiguous (a no-null column is ONE device view);
+//             dict-index payloads (+8 pad); string bytes
+//   validity: per-(file,col) validity bytes, contiguous across RGs
+//   dicts   : concatenated per-RG dictionaries (element bias applied in
+//             the GPU expansion kernel)
+//   runs    : int64 [m][6] {dense_out_off, n, is_literal,
+//             value_or_abs_bitoff, bit_width, dict_elem_bias}
 //   soffs   : int64 string offsets per (file,col), rebased across chunks
-//
-// Layout is 8-byte aligned so torch views work for any element size.
 #pragma once
 
 #include <atomic>
 #include <cstring>
+#include <list>
 #include <memory>
 #include <mutex>
 #include <thread>
+#include <unordered_map>
 #include <vector>
 
 #include "parquet_file.h"
@@ -31,6 +37,55 @@
 
 namespace lakesoul {
 
+// ---------------------------------------------------------------------- //
+// file handle cache (footer + mmap reuse across scans/steps)
+// ---------------------------------------------------------------------- //
+
+class FileCache {
+ public:
+  static FileCache& instance() {
+    static FileCache* c = new FileCache();
+    return *c;
+  }
+
+  std::shared_ptr<ParquetFile> open(const std::string& path) {
+    std::lock_guard<std::mutex> lk(mu_);
+    auto it = map_.find(path);
+    if (it != map_.end()) {
+      lru_.splice(lru_.begin(), lru_, it->second.second);
+      return it->second.first;
+    }
+    auto f = std::make_shared<ParquetFile>(path);
+    lru_.push_front(path);
+    map_[path] = {f, lru_.begin()};
+    while (map_.size() > capacity_) {
+      map_.erase(lru_.back());
+      lru_.pop_back();
+    }
+    return f;
+  }
+
+  void invalidate(const std::string& path) {
+    std::lock_guard<std::mutex> lk(mu_);
+    auto it = map_.find(path);
+    if (it != map_.end()) {
+      lru_.erase(it->second.second);
+      map_.erase(it);
+    }
+  }
+
+ private:
+  std::mutex mu_;
+  size_t capacity_ = 512;
+  std::list<std::string> lru_;
+  std::unordered_map<
+      std::string,
+      std::pair<std::shared_ptr<ParquetFile>, std::list<std::string>::iterator>>
+      map_;
+};
+
+// ---------------------------------------------------------------------- //
+
 struct UnitColumn {
   int file_idx;
   std::string name;
@@ -38,74 +93,75 @@ struct UnitColumn {
   bool is_string = false;
   bool is_dict = false;
   int physical = 0;
-  int64_t num_values = 0;   // rows in the file
+  int64_t num_values = 0;
   int64_t null_count = 0;
-  int64_t val_off = 0, val_len = 0;     // bytes in values buffer
-  int64_t validity_off = 0;             // bytes in validity buffer (num_values) or -1
-  int64_t dict_off = 0, dict_len = 0;   // bytes in dict buffer (fixed dicts)
-  int64_t run_off = 0, run_cnt = 0;     // rows in runs table
-  int64_t dense_n = 0;                  // non-null count (dict expansion size)
-  int64_t soff_off = 0;                 // entries in soffs buffer (num_values+1) or -1
-  int64_t sbytes_off = 0, sbytes_len = 0;  // string bytes in values buffer
+  int64_t val_off = 0, val_len = 0;
+  int64_t validity_off = -1;
+  int64_t dict_off = 0, dict_len = 0;
+  int64_t run_off = 0, run_cnt = 0;
+  int64_t dense_n = 0;
+  int64_t soff_off = -1;
+  int64_t sbytes_off = 0, sbytes_len = 0;
 };
 
-struct UnitData {
-  std::vector<UnitColumn> cols;        // file-major, then requested order
-  std::vector<int64_t> file_rows;
-  std::vector<uint8_t> values;
-  std::vector<uint8_t> validity;
-  std::vector<uint8_t> dicts;
-  std::vector<int64_t> runs;           // [m][6]
-  std::vector<int64_t> soffs;
-};
+inline int64_t ru_align8(int64_t x) { return (x + 7) & ~7LL; }
 
-inline int64_t align8(int64_t x) { return (x + 7) & ~7LL; }
-
-inline UnitData read_unit_raw(const std::vector<std::string>& paths,
-                              const std::vector<std::string>& names,
-                              int nthreads) {
-  size_t nfiles = paths.size();
-  UnitData out;
-  out.file_rows.resize(nfiles);
-
-  // phase A: read+decompress all chunks (parallel over files)
+struct UnitStage {
   struct FileData {
-    std::unique_ptr<ParquetFile> f;
-    // per requested name: decoded per-rg chunks (empty if absent)
+    std::shared_ptr<ParquetFile> f;
     std::vector<std::vector<ParquetFile::ChunkData>> chunks;
-    std::vector<int> col_idx;  // -1 if absent
+    std::vector<int> col_idx;
   };
-  std::vector<FileData> files(nfiles);
-  int nt_req = nthreads > 0 ? nthreads : (int)std::thread::hardware_concurrency();
-  if (nt_req < 1) nt_req = 1;
-  {
-    // open + footer parse (serial — cheap), then one task per
-    // (file, column, row-group) chunk so a single large base file still
-    // decompresses across every core
-    struct Task {
-      size_t fi, c, rg;
-    };
-    std::vector<Task> tasks;
-    for (size_t i = 0; i < nfiles; i++) {
-      FileData& fd = files[i];
-      fd.f = std::make_unique<ParquetFile>(paths[i]);
-      fd.chunks.resize(names.size());
-      fd.col_idx.resize(names.size());
-      size_t nrg = fd.f->num_row_groups();
-      for (size_t c = 0; c < names.size(); c++) {
-        int ci = fd.f->column_index(names[c]);
-        fd.col_idx[c] = ci;
-        if (ci < 0) continue;
-        fd.chunks[c].resize(nrg);
-        for (size_t rg = 0; rg < nrg; rg++) tasks.push_back({i, c, rg});
-      }
+  struct StrDecoded {
+    std::vector<int64_t> offs;
+    std::vector<uint8_t> bytes;
+  };
+
+  size_t ncols = 0;
+  std::vector<FileData> files;
+  std::vector<UnitColumn> cols;
+  std::vector<int64_t> file_rows;
+  std::vector<int64_t> runs;  // [m][6]
+  std::vector<std::unique_ptr<StrDecoded>> str_cols;
+  int64_t values_size = 0, validity_size = 0, dicts_size = 0, soffs_size = 0;
+};
+
+inline std::unique_ptr<UnitStage> read_unit_stage1(
+    const std::vector<std::string>& paths, const std::vector<std::string>& names) {
+  auto st = std::make_unique<UnitStage>();
+  UnitStage& S = *st;
+  size_t nfiles = paths.size();
+  S.ncols = names.size();
+  S.files.resize(nfiles);
+  S.file_rows.resize(nfiles);
+
+  // open via cache + build the (file, col, rg) task list
+  struct Task {
+    size_t fi, c, rg;
+  };
+  std::vector<Task> tasks;
+  for (size_t i = 0; i < nfiles; i++) {
+    auto& fd = S.files[i];
+    fd.f = FileCache::instance().open(paths[i]);
+    S.file_rows[i] = fd.f->num_rows();
+    fd.chunks.resize(names.size());
+    fd.col_idx.resize(names.size());
+    size_t nrg = fd.f->num_row_groups();
+    for (size_t c = 0; c < names.size(); c++) {
+      int ci = fd.f->column_index(names[c]);
+      fd.col_idx[c] = ci;
+      if (ci < 0) continue;
+      fd.chunks[c].resize(nrg);
+      for (size_t rg = 0; rg < nrg; rg++) tasks.push_back({i, c, rg});
     }
+  }
+  {
     std::string err;
     std::mutex err_mu;
     ThreadPool::instance().parallel_for((int64_t)tasks.size(), [&](int64_t i) {
       try {
         const Task& t = tasks[i];
-        FileData& fd = files[t.fi];
+        auto& fd = S.files[t.fi];
         fd.chunks[t.c][t.rg] = fd.f->read_chunk(t.rg, fd.col_idx[t.c]);
       } catch (std::exception& e) {
         std::lock_guard<std::mutex> lk(err_mu);
@@ -115,19 +171,16 @@ inline UnitData read_unit_raw(const std::vector<std::string>& paths,
     if (!err.empty()) throw std::runtime_error(err);
   }
 
-  // phase B: layout
-  int64_t vpos = 0, vapos = 0, dpos = 0, rpos = 0, spos = 0;
+  // layout
+  int64_t vpos = 0, vapos = 0, dpos = 0, spos = 0;
   for (size_t fi = 0; fi < nfiles; fi++) {
-    FileData& fd = files[fi];
-    out.file_rows[fi] = fd.f->num_rows();
+    auto& fd = S.files[fi];
     for (size_t c = 0; c < names.size(); c++) {
       UnitColumn uc;
       uc.file_idx = (int)fi;
       uc.name = names[c];
-      uc.validity_off = -1;
-      uc.soff_off = -1;
       if (fd.col_idx[c] < 0) {
-        out.cols.push_back(uc);
+        S.cols.push_back(uc);
         continue;
       }
       uc.present = true;
@@ -153,145 +206,142 @@ inline UnitData read_unit_raw(const std::vector<std::string>& paths,
       if (uc.is_string) {
         uc.soff_off = spos;
         spos += nv + 1;
-        uc.sbytes_off = vpos;  // filled in phase C
-        // compute string bytes total
-        int64_t total_bytes = 0;
-        for (auto& ch : chs) {
-          // decode lazily in phase C; conservatively bound by payload size
-          (void)ch;
-        }
-        // defer length; use decode pass below
       } else if (uc.is_dict) {
-        uc.val_off = vpos;  // index payloads
-        int64_t plen = 0, dlen = 0, runs = 0, dense = 0;
+        uc.val_off = vpos;
+        int64_t plen = 0, dlen = 0, dense = 0;
         for (auto& ch : chs) {
-          plen += align8((int64_t)ch.values.size());
-          dlen += align8((int64_t)ch.dict.size());
+          plen += ru_align8((int64_t)ch.values.size());
+          dlen += ru_align8((int64_t)ch.dict.size());
           dense += ch.num_values - ch.null_count;
-          // run count determined in phase C (parse); reserve later
-          (void)runs;
         }
-        uc.val_len = plen + 8;  // +8 pad for the bit reader
+        uc.val_len = plen + 8;
         vpos += uc.val_len;
         uc.dict_off = dpos;
         uc.dict_len = dlen;
         dpos += dlen;
         uc.dense_n = dense;
-        uc.run_off = -1;  // filled in phase C
+        // runs (bit offsets are relative to the values buffer)
+        uc.run_off = (int64_t)S.runs.size() / 6;
+        int64_t poff = uc.val_off;
+        int64_t dense_off = 0;
+        int64_t bias = 0;
+        int es = physical_elem_size(uc.physical);
+        for (auto& ch : chs) {
+          for (auto& ip : ch.idx_pages) {
+            std::vector<RleRun> rr;
+            parse_rle_runs(ch.values.data() + ip.payload_off,
+                           (size_t)ip.payload_len, ip.bit_width, ip.n,
+                           (poff + ip.payload_off) * 8, rr);
+            for (auto& r : rr) {
+              S.runs.push_back(r.out_off + dense_off);
+              S.runs.push_back(r.n);
+              S.runs.push_back(r.is_literal);
+              S.runs.push_back(r.is_literal ? r.bit_off : (int64_t)r.value + bias);
+              S.runs.push_back(ip.bit_width);
+              S.runs.push_back(bias);
+            }
+            dense_off += ip.n;
+          }
+          poff += ru_align8((int64_t)ch.values.size());
+          bias += (int64_t)(ru_align8((int64_t)ch.dict.size()) / es);
+        }
+        uc.run_cnt = (int64_t)S.runs.size() / 6 - uc.run_off;
       } else {
         uc.val_off = vpos;
         int64_t plen = 0;
         for (auto& ch : chs) plen += (int64_t)ch.values.size();
         uc.val_len = plen;
-        vpos += align8(plen);
+        vpos += ru_align8(plen);
       }
-      out.cols.push_back(uc);
+      S.cols.push_back(uc);
     }
   }
 
-  // strings need decode to know byte totals: do a pre-pass
-  struct StrDecoded {
-    std::vector<int64_t> offs;
-    std::vector<uint8_t> bytes;
-  };
-  std::vector<std::unique_ptr<StrDecoded>> str_cols(out.cols.size());
-  for (size_t u = 0; u < out.cols.size(); u++) {
-    UnitColumn& uc = out.cols[u];
-    if (!uc.present || !uc.is_string) continue;
-    FileData& fd = files[uc.file_idx];
-    size_t c = u % names.size();
-    auto sd = std::make_unique<StrDecoded>();
-    sd->offs.push_back(0);
-    for (auto& ch : fd.chunks[c]) {
-      DecodedColumn dc = decode_chunk_cpu(ch);
-      int64_t base = (int64_t)sd->bytes.size();
-      sd->bytes.insert(sd->bytes.end(), dc.bytes.begin(), dc.bytes.end());
-      for (size_t i = 1; i < dc.offsets.size(); i++)
-        sd->offs.push_back(base + dc.offsets[i]);
-    }
-    uc.sbytes_off = vpos;
-    uc.sbytes_len = (int64_t)sd->bytes.size();
-    vpos += align8(uc.sbytes_len);
-    str_cols[u] = std::move(sd);
-  }
-
-  out.values.resize((size_t)vpos);
-  out.validity.assign((size_t)vapos, 1);
-  out.dicts.resize((size_t)dpos);
-  out.soffs.resize((size_t)spos);
-
-  // phase C: dict columns first (serial — runs append to one shared
-  // vector and are tiny), then parallel memcpy for everything else
-  for (size_t u = 0; u < out.cols.size(); u++) {
-    UnitColumn& uc = out.cols[u];
-    if (!uc.present || !uc.is_dict) continue;
-    FileData& fd = files[uc.file_idx];
-    auto& chs = fd.chunks[u % names.size()];
-    uc.run_off = rpos / 6;
-    int64_t poff = uc.val_off;
-    int64_t doff = uc.dict_off;
-    int64_t dense_off = 0;
-    int64_t dict_elem_bias = 0;
-    int es = physical_elem_size(uc.physical);
-    for (auto& ch : chs) {
-      std::memcpy(out.values.data() + poff, ch.values.data(), ch.values.size());
-      std::memcpy(out.dicts.data() + doff, ch.dict.data(), ch.dict.size());
-      for (auto& ip : ch.idx_pages) {
-        std::vector<RleRun> rr;
-        parse_rle_runs(ch.values.data() + ip.payload_off, (size_t)ip.payload_len,
-                       ip.bit_width, ip.n, (poff + ip.payload_off) * 8, rr);
-        for (auto& r : rr) {
-          out.runs.push_back(r.out_off + dense_off);
-          out.runs.push_back(r.n);
-          out.runs.push_back(r.is_literal);
-          out.runs.push_back(r.is_literal ? r.bit_off
-                                          : (int64_t)r.value + dict_elem_bias);
-          out.runs.push_back(ip.bit_width);
-          out.runs.push_back(dict_elem_bias);
-          rpos += 6;
-        }
-        dense_off += ip.n;
-      }
-      poff += align8((int64_t)ch.values.size());
-      doff += align8((int64_t)ch.dict.size());
-      dict_elem_bias += (int64_t)(align8((int64_t)ch.dict.size()) / es);
-    }
-    uc.run_cnt = rpos / 6 - uc.run_off;
-  }
+  // string pre-pass (decode to learn byte totals) — parallel per column
+  S.str_cols.resize(S.cols.size());
   {
-    ThreadPool::instance().parallel_for((int64_t)out.cols.size(), [&](int64_t u) {
-        UnitColumn& uc = out.cols[u];
-        if (!uc.present) return;
-        FileData& fd = files[uc.file_idx];
-        auto& chs = fd.chunks[u % names.size()];
-        if (uc.validity_off >= 0) {
-          int64_t off = uc.validity_off;
-          for (auto& ch : chs) {
-            if (!ch.validity.empty())
-              std::memcpy(out.validity.data() + off, ch.validity.data(),
-                          ch.validity.size());
-            off += ch.num_values;
-          }
+    std::vector<size_t> str_idx;
+    for (size_t u = 0; u < S.cols.size(); u++)
+      if (S.cols[u].present && S.cols[u].is_string) str_idx.push_back(u);
+    std::string err;
+    std::mutex err_mu;
+    ThreadPool::instance().parallel_for((int64_t)str_idx.size(), [&](int64_t k) {
+      try {
+        size_t u = str_idx[k];
+        UnitColumn& uc = S.cols[u];
+        auto& fd = S.files[uc.file_idx];
+        auto sd = std::make_unique<UnitStage::StrDecoded>();
+        sd->offs.push_back(0);
+        for (auto& ch : fd.chunks[u % S.ncols]) {
+          DecodedColumn dc = decode_chunk_cpu(ch);
+          int64_t base = (int64_t)sd->bytes.size();
+          sd->bytes.insert(sd->bytes.end(), dc.bytes.begin(), dc.bytes.end());
+          for (size_t i = 1; i < dc.offsets.size(); i++)
+            sd->offs.push_back(base + dc.offsets[i]);
         }
-        if (uc.is_dict) return;  // handled above
-        if (uc.is_string) {
-          auto& sd = *str_cols[u];
-          std::memcpy(out.soffs.data() + uc.soff_off, sd.offs.data(),
-                      sd.offs.size() * 8);
-          if (!sd.bytes.empty())
-            std::memcpy(out.values.data() + uc.sbytes_off, sd.bytes.data(),
-                        sd.bytes.size());
-          return;
-        }
-        int64_t off = uc.val_off;
-        for (auto& ch : chs) {
-          std::memcpy(out.values.data() + off, ch.values.data(), ch.values.size());
-          off += (int64_t)ch.values.size();
-        }
+        S.str_cols[u] = std::move(sd);
+      } catch (std::exception& e) {
+        std::lock_guard<std::mutex> lk(err_mu);
+        err = e.what();
+      }
     });
+    if (!err.empty()) throw std::runtime_error(err);
+    for (size_t u : str_idx) {
+      UnitColumn& uc = S.cols[u];
+      uc.sbytes_off = vpos;
+      uc.sbytes_len = (int64_t)S.str_cols[u]->bytes.size();
+      vpos += ru_align8(uc.sbytes_len);
+    }
   }
-  (void)nt_req;
-  return out;
+
+  S.values_size = vpos;
+  S.validity_size = vapos;
+  S.dicts_size = dpos;
+  S.soffs_size = spos;
+  return st;
+}
+
+// Fill caller-allocated buffers (sized from stage1) — pure parallel memcpy.
+inline void read_unit_fill(UnitStage& S, uint8_t* values, uint8_t* validity,
+                           uint8_t* dicts, int64_t* soffs) {
+  if (S.validity_size) std::memset(validity, 1, (size_t)S.validity_size);
+  ThreadPool::instance().parallel_for((int64_t)S.cols.size(), [&](int64_t u) {
+    UnitColumn& uc = S.cols[u];
+    if (!uc.present) return;
+    auto& fd = S.files[uc.file_idx];
+    auto& chs = fd.chunks[u % S.ncols];
+    if (uc.validity_off >= 0) {
+      int64_t off = uc.validity_off;
+      for (auto& ch : chs) {
+        if (!ch.validity.empty())
+          std::memcpy(validity + off, ch.validity.data(), ch.validity.size());
+        off += ch.num_values;
+      }
+    }
+    if (uc.is_string) {
+      auto& sd = *S.str_cols[u];
+      std::memcpy(soffs + uc.soff_off, sd.offs.data(), sd.offs.size() * 8);
+      if (!sd.bytes.empty())
+        std::memcpy(values + uc.sbytes_off, sd.bytes.data(), sd.bytes.size());
+      return;
+    }
+    if (uc.is_dict) {
+      int64_t poff = uc.val_off;
+      int64_t doff = uc.dict_off;
+      for (auto& ch : chs) {
+        std::memcpy(values + poff, ch.values.data(), ch.values.size());
+        std::memcpy(dicts + doff, ch.dict.data(), ch.dict.size());
+        poff += ru_align8((int64_t)ch.values.size());
+        doff += ru_align8((int64_t)ch.dict.size());
+      }
+      return;
+    }
+    int64_t off = uc.val_off;
+    for (auto& ch : chs) {
+      std::memcpy(values + off, ch.values.data(), ch.values.size());
+      off += (int64_t)ch.values.size();
+    }
+  });
 }
 
 }  // namespace lakesoul
