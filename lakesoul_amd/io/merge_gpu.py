@@ -46,9 +46,12 @@ def _pack_keys_one_file(cols: List[Column]) -> Optional[torch.Tensor]:
 
 def merge_key_order(
     file_pk_cols: List[List[Column]], counts: List[int], device
-) -> torch.Tensor:
-    """Return global row order (indices into the concatenated rows) sorted
-    by (pk..., file_seq, row) — i.e. the MOR merge order."""
+):
+    """Return (order, sorted_keys_or_None): global row order (indices into
+    the concatenated rows) sorted by (pk..., file_seq, row) — the MOR
+    merge order. When the PK packs into a u64 the merge-path kernel is
+    used and the merged key array comes back for free (boundary
+    detection reuses it, no re-gather)."""
     offsets = [0]
     for c in counts:
         offsets.append(offsets[-1] + c)
@@ -73,7 +76,7 @@ def merge_key_order(
             if len(streams) % 2:
                 nxt.append(streams[-1])
             streams = nxt
-        return streams[0][1]
+        return streams[0][1], streams[0][0]
 
     # generic lexsort on concatenated keys (stable; ties keep concat order)
     perm = torch.arange(total, dtype=torch.int64, device=device)
@@ -91,7 +94,7 @@ def merge_key_order(
         k = cat_cols[ci][perm]
         order = torch.argsort(k, stable=True)
         perm = perm[order]
-    return perm
+    return perm, None
 
 
 def sorted_keys_for_order(
@@ -128,16 +131,20 @@ def merge_sorted_files_gpu(
     total = sum(counts)
     file_pk_cols = [[b.columns[p] for p in pk] for b in file_batches]
 
-    order = merge_key_order(file_pk_cols, counts, device)
+    order, sorted_keys = merge_key_order(file_pk_cols, counts, device)
 
-    # group boundaries from sorted pk values
-    sorted_pks = sorted_keys_for_order(file_pk_cols, order)
+    # group boundaries: from the merged u64 keys when available (one
+    # kernel), else from gathered pk values
     n = total
-    start = torch.zeros(n, dtype=torch.bool, device=device)
-    if n:
-        start[0] = True
-        for sp in sorted_pks:
-            start[1:] |= sp[1:] != sp[:-1]
+    if sorted_keys is not None:
+        start = hip().group_start_mask(sorted_keys).to(torch.bool) if n else torch.zeros(0, dtype=torch.bool, device=device)
+    else:
+        sorted_pks = sorted_keys_for_order(file_pk_cols, order)
+        start = torch.zeros(n, dtype=torch.bool, device=device)
+        if n:
+            start[0] = True
+            for sp in sorted_pks:
+                start[1:] |= sp[1:] != sp[:-1]
     grp = torch.cumsum(start.to(torch.int64), 0) - 1
     ngroups = int(grp[-1].item()) + 1 if n else 0
     # last row index (in sorted space) per group

@@ -215,7 +215,7 @@ class LakeSoulScan:
             else:
                 yield batch
 
-    def _iter_units_pipelined(self, units: List[ScanUnit], depth: int = 2):
+    def _iter_units_pipelined(self, units: List[ScanUnit], depth: int = 3):
         """GPU path: prefetch the host IO/decompress of the next units on
         background threads while the GPU decodes+merges the current one
         (overlap engineering)."""
