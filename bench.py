@@ -165,23 +165,17 @@ def main():
     def one_scan() -> int:
         scan = table.scan(device=device).shard(rank, world)
         total = 0
-        exchange_bufs = []
         for batch in scan.iter_batches():
             n = batch.num_rows
             total += n
             if dist is not None and device == "cuda":
-                # RCCL all-to-all shard exchange over xGMI: redistribute the
-                # merged rows round-robin to consumer ranks
-                for f in batch.schema:
-                    c = batch.columns[f.name]
-                    if c.is_string:
-                        continue
-                    t = c.data
-                    cut = (n // world) * world
-                    send = t[:cut].reshape(world, -1)
-                    recv = torch.empty_like(send)
-                    dist.all_to_all_single(recv, send)
-                    exchange_bufs.append(recv.shape)
+                # RCCL all-to-all shard exchange over xGMI: redistribute
+                # the merged rows round-robin to consumer ranks (variable
+                # per-peer splits handled by the shard exchange)
+                from lakesoul_amd.parallel.shard import exchange_batch_all_to_all
+
+                dest = torch.arange(n, device="cuda") % world
+                exchange_batch_all_to_all(batch, dest)
         return total
 
     for w in range(args.warmup):

@@ -1,0 +1,168 @@
+#!/usr/bin/env python3
+"""TPC-H-style scan benchmark (BASELINE config 4): lineitem table through
+the lakehouse scan path with filter/projection on GPU.
+
+The reference runs TPC-H through lakesoul-datafusion
+(``rust/lakesoul-datafusion/src/tests/benchmarks/tpch/``); here the same
+role is played by the native scan + torch reductions on HBM-resident
+columns. No network: lineitem is generated synthetically with TPC-H-like
+value distributions (string dimensions as dictionary codes).
+
+    python benchmarks/tpch.py --sf 1 --steps 3 [--device cuda]
+
+Queries:
+  q6: SELECT sum(l_extendedprice*l_discount) WHERE l_shipdate in year
+      AND l_discount BETWEEN .05 AND .07 AND l_quantity < 24
+  q1lite: per-(returnflag,linestatus) sums/avgs over a shipdate cutoff
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+ROWS_PER_SF = 6_000_000
+
+
+def make_lineitem(catalog, sf: float, device):
+    from lakesoul_amd.io.schema import Field, Schema
+    from lakesoul_amd.io.stream_writer import StreamingWriter
+    from lakesoul_amd.meta.entities import CommitOp
+
+    n = int(ROWS_PER_SF * sf)
+    schema = Schema(
+        [
+            Field("l_orderkey", "int64", False),
+            Field("l_quantity", "float64", False),
+            Field("l_extendedprice", "float64", False),
+            Field("l_discount", "float64", False),
+            Field("l_tax", "float64", False),
+            Field("l_returnflag", "int8", False),   # A/N/R -> 0/1/2
+            Field("l_linestatus", "int8", False),   # F/O -> 0/1
+            Field("l_shipdate", "date32", False),
+        ]
+    )
+    if catalog.table_exists("lineitem"):
+        catalog.drop_table("lineitem", delete_data=True)
+    t = catalog.create_table("lineitem", schema, hash_bucket_num=16)
+    rng = np.random.default_rng(7)
+    chunk = 2_000_000
+    with StreamingWriter(t, commit_op=CommitOp.AppendCommit,
+                         max_rows_per_flush=chunk, device=device) as w:
+        done = 0
+        while done < n:
+            m = min(chunk, n - done)
+            w.write(
+                {
+                    "l_orderkey": rng.integers(0, n, m, dtype=np.int64),
+                    "l_quantity": rng.integers(1, 51, m).astype(np.float64),
+                    "l_extendedprice": rng.uniform(900, 105000, m),
+                    "l_discount": np.round(rng.uniform(0.0, 0.1, m), 2),
+                    "l_tax": np.round(rng.uniform(0.0, 0.08, m), 2),
+                    "l_returnflag": rng.integers(0, 3, m, dtype=np.int8),
+                    "l_linestatus": rng.integers(0, 2, m, dtype=np.int8),
+                    # dates over 7 years starting 1992-01-01 (day 8035)
+                    "l_shipdate": (8035 + rng.integers(0, 2557, m)).astype(np.int32),
+                }
+            )
+            done += m
+    return t
+
+
+def q6(table, device) -> float:
+    # year 1994 = days [8766, 9131)
+    scan = table.scan(
+        columns=["l_extendedprice", "l_discount"],
+        filters=[
+            ("l_shipdate", ">=", 8766),
+            ("l_shipdate", "<", 9131),
+            ("l_discount", ">=", 0.05),
+            ("l_discount", "<=", 0.07),
+            ("l_quantity", "<", 24.0),
+        ],
+        device=device,
+    )
+    total = 0.0
+    for batch in scan.iter_batches():
+        ep = batch.columns["l_extendedprice"].data
+        di = batch.columns["l_discount"].data
+        total += float((ep * di).sum())
+    return total
+
+
+def q1lite(table, device):
+    scan = table.scan(
+        columns=["l_returnflag", "l_linestatus", "l_quantity", "l_extendedprice", "l_discount", "l_tax"],
+        filters=[("l_shipdate", "<=", 10471)],
+        device=device,
+    )
+    sums = torch.zeros(6, 4, dtype=torch.float64)
+    counts = torch.zeros(6, dtype=torch.float64)
+    for batch in scan.iter_batches():
+        g = (batch.columns["l_returnflag"].data.to(torch.int64) * 2
+             + batch.columns["l_linestatus"].data.to(torch.int64))
+        qty = batch.columns["l_quantity"].data
+        ep = batch.columns["l_extendedprice"].data
+        di = batch.columns["l_discount"].data
+        tax = batch.columns["l_tax"].data
+        disc_price = ep * (1 - di)
+        charge = disc_price * (1 + tax)
+        dev_sums = torch.zeros(6, 4, dtype=torch.float64, device=g.device)
+        dev_counts = torch.zeros(6, dtype=torch.float64, device=g.device)
+        for j, col in enumerate((qty, ep, disc_price, charge)):
+            dev_sums[:, j].scatter_add_(0, g, col)
+        dev_counts.scatter_add_(0, g, torch.ones_like(qty))
+        sums += dev_sums.cpu()
+        counts += dev_counts.cpu()
+    return sums, counts
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--sf", type=float, default=1.0)
+    p.add_argument("--steps", type=int, default=3)
+    p.add_argument("--device", default=None)
+    p.add_argument("--workdir", default=None)
+    args = p.parse_args()
+    device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
+
+    workdir = args.workdir or os.path.join(os.environ.get("TMPDIR", "/tmp"), "lakesoul_tpch")
+    os.makedirs(workdir, exist_ok=True)
+    os.environ["LAKESOUL_META_DB"] = os.path.join(workdir, "meta.db")
+    from lakesoul_amd.meta.client import MetaClient
+    from lakesoul_amd.meta.store import SqliteMetaStore
+    from lakesoul_amd.tables.catalog import LakeSoulCatalog
+
+    catalog = LakeSoulCatalog(
+        MetaClient(SqliteMetaStore(os.environ["LAKESOUL_META_DB"])),
+        warehouse=os.path.join(workdir, "wh"),
+    )
+    t0 = time.time()
+    t = make_lineitem(catalog, args.sf, device)
+    print(f"generated lineitem sf={args.sf} in {time.time()-t0:.1f}s", file=sys.stderr)
+
+    results = {}
+    for name, fn in (("q6", q6), ("q1lite", q1lite)):
+        fn(t, device)  # warmup
+        if device == "cuda":
+            torch.cuda.synchronize()
+        t0 = time.time()
+        for _ in range(args.steps):
+            out = fn(t, device)
+        if device == "cuda":
+            torch.cuda.synchronize()
+        dt = (time.time() - t0) / args.steps
+        n = int(ROWS_PER_SF * args.sf)
+        results[name] = {"s_per_query": dt, "rows_per_sec": n / dt}
+        print(f"{name}: {dt*1000:.1f} ms ({n/dt/1e6:.1f}M rows/s)", file=sys.stderr)
+    print(json.dumps({"metric": "tpch_scan", "sf": args.sf, "device": device, "queries": results}))
+
+
+if __name__ == "__main__":
+    main()

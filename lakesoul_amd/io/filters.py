@@ -41,6 +41,17 @@ class Expr:
         """col -> constant for top-level AND-ed equality predicates."""
         return {}
 
+    def columns(self) -> set:
+        """All column names referenced by this expression."""
+        out = set()
+        for f in getattr(self, "__dataclass_fields__", {}):
+            v = getattr(self, f)
+            if isinstance(v, Expr):
+                out |= v.columns()
+        if hasattr(self, "col"):
+            out.add(self.col)
+        return out
+
 
 @dataclass
 class Literal(Expr):

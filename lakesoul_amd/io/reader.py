@@ -108,6 +108,20 @@ class LakeSoulScan:
         self.version = version
         self.timestamp_ms = timestamp_ms
         self.filter_expr = resolve_filters(filters, self.schema)
+        # filter columns must be read (and materialized) even when not
+        # selected; the final projection drops them (session.rs:650-730
+        # projection computation analog)
+        if self.filter_expr is not None:
+            for c in sorted(self.filter_expr.columns()):
+                if c not in self.read_cols and c not in self.range_cols:
+                    self.read_cols.append(c)
+        self.eval_fields = list(self.out_schema.fields)
+        have = {f.name for f in self.eval_fields}
+        if self.filter_expr is not None:
+            for c in sorted(self.filter_expr.columns()):
+                if c not in have:
+                    self.eval_fields.append(self.schema.field(c))
+        self.eval_schema = Schema(self.eval_fields)
         self.device = device or table.io_config().resolve_device()
         self.batch_size = batch_size
         self.incremental = incremental
@@ -205,6 +219,11 @@ class LakeSoulScan:
             if batch is None:
                 continue
             batch = self._apply_filters(batch)
+            if len(self.eval_schema) != len(self.out_schema):
+                batch = Batch(
+                    self.out_schema,
+                    {f.name: batch.columns[f.name] for f in self.out_schema},
+                )
             if self.batch_size:
                 n = batch.num_rows
                 for off in range(0, n, self.batch_size):
@@ -300,7 +319,7 @@ class LakeSoulScan:
 
     def _np_to_batch(self, merged: Dict[str, NpColumn], unit: ScanUnit) -> Batch:
         cols: Dict[str, Column] = {}
-        for f in self.out_schema:
+        for f in self.eval_schema:
             if f.name in self.range_cols:
                 cols[f.name] = self._range_value_column(f, unit, merged)
                 continue
@@ -319,7 +338,7 @@ class LakeSoulScan:
                     data=torch.from_numpy(np.ascontiguousarray(data)),
                     validity=None if npc.validity is None else torch.from_numpy(npc.validity),
                 )
-        return Batch(self.out_schema, cols)
+        return Batch(self.eval_schema, cols)
 
     def _range_value_column(self, f, unit: ScanUnit, merged) -> Column:
         """Materialize a range-partition column from the partition_desc."""

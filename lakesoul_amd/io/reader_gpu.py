@@ -154,15 +154,16 @@ def read_unit_gpu(scan, unit, raw: Optional[dict] = None) -> Optional[Batch]:
 
         merged = concat_batches(file_batches)
 
-    # project to out_schema (+ materialize range-partition columns)
+    # project to eval schema (+ materialize range-partition columns);
+    # filter-only columns are dropped after filter evaluation
     cols: Dict[str, Column] = {}
     nrows = merged.num_rows
-    for f in scan.out_schema:
+    for f in scan.eval_schema:
         if f.name in scan.range_cols:
             cols[f.name] = _range_col_gpu(scan, f, unit, nrows, device)
         else:
             cols[f.name] = merged.columns[f.name]
-    return Batch(scan.out_schema, cols)
+    return Batch(scan.eval_schema, cols)
 
 
 def _decode_fixed_chunk_gpu(d: dict, dtype: str, device) -> Column:

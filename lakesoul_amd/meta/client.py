@@ -25,7 +25,7 @@ from .entities import (
 )
 from .store import CommitConflictError, SqliteMetaStore, open_meta_store
 
-DEFAULT_MAX_RETRY = 5
+DEFAULT_MAX_RETRY = 25
 
 
 class MetaClient:
