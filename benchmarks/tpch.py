@@ -102,25 +102,26 @@ def q1lite(table, device):
         filters=[("l_shipdate", "<=", 10471)],
         device=device,
     )
-    sums = torch.zeros(6, 4, dtype=torch.float64)
-    counts = torch.zeros(6, dtype=torch.float64)
+    sums = None  # accumulate on GPU; fetch once at the end
+    counts = None
     for batch in scan.iter_batches():
         g = (batch.columns["l_returnflag"].data.to(torch.int64) * 2
-             + batch.columns["l_linestatus"].data.to(torch.int64))
+             + batch.columns["l_linestatus"].data.to(torch.int64)).contiguous()
         qty = batch.columns["l_quantity"].data
         ep = batch.columns["l_extendedprice"].data
         di = batch.columns["l_discount"].data
         tax = batch.columns["l_tax"].data
         disc_price = ep * (1 - di)
         charge = disc_price * (1 + tax)
-        dev_sums = torch.zeros(6, 4, dtype=torch.float64, device=g.device)
-        dev_counts = torch.zeros(6, dtype=torch.float64, device=g.device)
+        if sums is None:
+            sums = [torch.zeros(6, dtype=torch.float64, device=g.device) for _ in range(4)]
+            counts = torch.zeros(6, dtype=torch.float64, device=g.device)
         for j, col in enumerate((qty, ep, disc_price, charge)):
-            dev_sums[:, j].scatter_add_(0, g, col)
-        dev_counts.scatter_add_(0, g, torch.ones_like(qty))
-        sums += dev_sums.cpu()
-        counts += dev_counts.cpu()
-    return sums, counts
+            sums[j].scatter_add_(0, g, col.contiguous())
+        counts.scatter_add_(0, g, torch.ones_like(qty))
+    if sums is None:
+        return torch.zeros(6, 4, dtype=torch.float64), torch.zeros(6, dtype=torch.float64)
+    return torch.stack(sums, 1).cpu(), counts.cpu()
 
 
 def main():
