@@ -244,14 +244,18 @@ class LakeSoulScan:
 
         with ThreadPoolExecutor(max_workers=depth) as ex:
             futs = [
-                ex.submit(fetch_raw, u.files, self.read_cols)
+                ex.submit(fetch_raw, self._localize(u.files), self.read_cols)
                 for u in units[: depth]
             ]
             for i, unit in enumerate(units):
                 raw = futs[i].result()
                 if i + depth < len(units):
                     futs.append(
-                        ex.submit(fetch_raw, units[i + depth].files, self.read_cols)
+                        ex.submit(
+                            fetch_raw,
+                            self._localize(units[i + depth].files),
+                            self.read_cols,
+                        )
                     )
                 yield read_unit_gpu(self, unit, raw)
 
@@ -275,7 +279,7 @@ class LakeSoulScan:
         if not self.pk and self.filter_expr is None:
             total = 0
             for unit in self.plan():
-                for f in unit.files:
+                for f in self._localize(unit.files):
                     h = cpp().open_parquet(f)
                     total += cpp().parquet_meta(h)["num_rows"]
                     cpp().close_parquet(h)
@@ -284,9 +288,19 @@ class LakeSoulScan:
 
     # ------------------------------------------------------------------ #
 
+    def _localize(self, files: List[str]) -> List[str]:
+        from .fs import default_fs, is_remote
+
+        if not any(is_remote(p) for p in files):
+            return files
+        fs = default_fs()
+        return [fs.localize(p) if is_remote(p) else p for p in files]
+
     def _read_unit(self, unit: ScanUnit) -> Optional[Batch]:
         if not unit.files:
             return None
+        unit = ScanUnit(unit.partition_desc, unit.bucket_id,
+                        self._localize(unit.files), unit.is_compacted_first)
         if self.device == "cuda":
             from .reader_gpu import read_unit_gpu
 
@@ -459,7 +473,7 @@ class LakeSoulScan:
         out = []
         for path in files:
             try:
-                h = cpp().open_parquet(path)
+                h = cpp().open_parquet(self._localize([path])[0])
             except Exception:
                 out.append(path)
                 continue

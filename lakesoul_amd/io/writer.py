@@ -27,6 +27,7 @@ import torch
 from .. import constants
 from ..ops import cpp
 from .batch import Batch, Column
+from .fs import default_fs, is_remote
 from .schema import Schema
 
 _CODEC_ID = {"zstd": 6, "none": 0, "uncompressed": 0}
@@ -178,6 +179,25 @@ def _partition_descs(batch: Batch, range_cols: Sequence[str]):
 
 def _write_batch_to_file(path: str, batch: Batch, compression: str, level: int,
                          row_group_size: int) -> int:
+    """Write one parquet file; remote destinations write locally then
+    upload (multipart-writer analog, multipart_writer.rs:43)."""
+    if is_remote(path):
+        import tempfile
+
+        fd, tmp = tempfile.mkstemp(suffix=".parquet")
+        os.close(fd)
+        try:
+            size = _write_batch_to_file_local(tmp, batch, compression, level, row_group_size)
+            default_fs().upload(tmp, path)
+            return size
+        finally:
+            if os.path.exists(tmp):
+                os.remove(tmp)
+    return _write_batch_to_file_local(path, batch, compression, level, row_group_size)
+
+
+def _write_batch_to_file_local(path: str, batch: Batch, compression: str, level: int,
+                               row_group_size: int) -> int:
     names, dtypes, columns, offsets, validity, nullable = [], [], [], [], [], []
     for f in batch.schema:
         c = batch.columns[f.name]
@@ -253,7 +273,10 @@ def write_table_data(table, data, device: Optional[str] = None,
         out_dir = os.path.join(table.table_path, subdir) if subdir else table.table_path
         if compact:
             out_dir = os.path.join(out_dir, constants.COMPACT_DIR)
-        os.makedirs(out_dir, exist_ok=True)
+        if is_remote(out_dir):
+            default_fs().makedirs(out_dir)
+        else:
+            os.makedirs(out_dir, exist_ok=True)
 
         if pk:
             # one global stable sort by (bucket, pk): buckets become

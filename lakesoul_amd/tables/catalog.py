@@ -50,7 +50,12 @@ class LakeSoulCatalog:
         schema = normalize_schema(schema)
         if table_path is None:
             table_path = os.path.join(self.warehouse, namespace, table_name)
-        os.makedirs(table_path, exist_ok=True)
+        from ..io.fs import default_fs, is_remote
+
+        if is_remote(table_path):
+            default_fs().makedirs(table_path)
+        else:
+            os.makedirs(table_path, exist_ok=True)
         props = dict(properties or {})
         props.setdefault("hashBucketNum", str(int(hash_bucket_num)))
         partitions = ",".join(range_partitions) + ";" + ",".join(primary_keys)
