@@ -36,6 +36,7 @@ template <typename T, typename ACC>
 void launch_segmented_sum(const T*, const int64_t*, const uint8_t*, const uint8_t*, ACC*, int32_t*, int64_t, hipStream_t);
 void launch_segmented_last(const int64_t*, const uint8_t*, const uint8_t*, int64_t*, int64_t, hipStream_t);
 void launch_ann_scores(const short*, const short*, float*, int64_t, int32_t, int32_t, hipStream_t);
+void launch_snappy_decompress(const uint8_t*, const int64_t*, int64_t, uint8_t*, int32_t*, hipStream_t);
 
 }  // namespace lakesoul
 
@@ -308,8 +309,24 @@ static torch::Tensor ann_scores(torch::Tensor X, torch::Tensor Q) {
   return out;
 }
 
+// jobs int64 [n,4] = {src_off, src_len, dst_off, dst_len}; returns
+// (dst_buffer, status int32[n]) — status 0 = ok.
+static std::vector<torch::Tensor> snappy_decompress(torch::Tensor src,
+                                                    torch::Tensor jobs,
+                                                    int64_t total_dst) {
+  CHECK_GPU(src);
+  CHECK_GPU(jobs);
+  auto dst = torch::empty({total_dst}, src.options());
+  auto status = torch::empty({jobs.size(0)}, src.options().dtype(torch::kInt32));
+  launch_snappy_decompress(src.data_ptr<uint8_t>(), jobs.data_ptr<int64_t>(),
+                           jobs.size(0), dst.data_ptr<uint8_t>(),
+                           status.data_ptr<int32_t>(), cur_stream());
+  return {dst, status};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ann_scores", &ann_scores);
+  m.def("snappy_decompress", &snappy_decompress);
   m.doc() = "lakesoul_amd gfx950 HIP kernels";
   m.def("hash_fixed_column", &hash_fixed_column);
   m.def("hash_string_column", &hash_string_column);
