@@ -211,7 +211,7 @@ class LakeSoulScan:
 
     def iter_batches(self) -> Iterator[Batch]:
         units = self.plan()
-        if self.device == "cuda" and len(units) > 1:
+        if self.device == "cuda" and len(units) > 1 and self._gpu_merge_supported():
             gen = self._iter_units_pipelined(units)
         else:
             gen = (self._read_unit(u) for u in units)
@@ -296,16 +296,32 @@ class LakeSoulScan:
         fs = default_fs()
         return [fs.localize(p) if is_remote(p) else p for p in files]
 
+    def _gpu_merge_supported(self) -> bool:
+        """String PKs need byte-wise lexicographic merge order, which the
+        GPU merge does not implement yet — those units run the CPU merge
+        (decode still benefits from the host pipeline)."""
+        if not self.pk:
+            return True
+        return all(
+            self.schema.field(p).dtype not in ("string", "binary") for p in self.pk
+        )
+
     def _read_unit(self, unit: ScanUnit) -> Optional[Batch]:
         if not unit.files:
             return None
         unit = ScanUnit(unit.partition_desc, unit.bucket_id,
                         self._localize(unit.files), unit.is_compacted_first)
+        if self.device == "cuda" and not self._gpu_merge_supported() and len(unit.files) > 1:
+            # hybrid: CPU merge, then ship the merged batch to HBM
+            batch = self._read_unit_cpu(unit)
+            return batch.to_device("cuda") if batch is not None else None
         if self.device == "cuda":
             from .reader_gpu import read_unit_gpu
 
             return read_unit_gpu(self, unit)
-        # ---- CPU path ----
+        return self._read_unit_cpu(unit)
+
+    def _read_unit_cpu(self, unit: ScanUnit) -> Optional[Batch]:
         file_cols: List[Dict[str, NpColumn]] = []
         present: List[set] = []
         for path in unit.files:
