@@ -116,9 +116,13 @@ def q1lite(table, device):
         if sums is None:
             sums = [torch.zeros(6, dtype=torch.float64, device=g.device) for _ in range(4)]
             counts = torch.zeros(6, dtype=torch.float64, device=g.device)
-        for j, col in enumerate((qty, ep, disc_price, charge)):
-            sums[j].scatter_add_(0, g, col.contiguous())
-        counts.scatter_add_(0, g, torch.ones_like(qty))
+        # masked reductions (6 groups): scatter_add into 6 slots would
+        # serialize on atomic contention
+        for grp in range(6):
+            m = g == grp
+            for j, col in enumerate((qty, ep, disc_price, charge)):
+                sums[j][grp] += col[m].sum()
+            counts[grp] += m.sum()
     if sums is None:
         return torch.zeros(6, 4, dtype=torch.float64), torch.zeros(6, dtype=torch.float64)
     return torch.stack(sums, 1).cpu(), counts.cpu()

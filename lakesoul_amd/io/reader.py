@@ -311,7 +311,10 @@ class LakeSoulScan:
             return None
         unit = ScanUnit(unit.partition_desc, unit.bucket_id,
                         self._localize(unit.files), unit.is_compacted_first)
-        if self.device == "cuda" and not self._gpu_merge_supported() and len(unit.files) > 1:
+        needs_merge = bool(self.pk) and (
+            len(unit.files) > 1 or self.cdc_column is not None or bool(self.merge_ops)
+        )
+        if self.device == "cuda" and needs_merge and not self._gpu_merge_supported():
             # hybrid: CPU merge, then ship the merged batch to HBM
             batch = self._read_unit_cpu(unit)
             return batch.to_device("cuda") if batch is not None else None

@@ -123,15 +123,17 @@ def _sort_indices(batch: Batch, sort_cols: Sequence[str]) -> torch.Tensor:
     first_col = batch.columns[sort_cols[0]]
     dev_t = first_col.data if not first_col.is_string else first_col.bytes_
     if dev_t is not None and dev_t.device.type == "cuda":
-        idx = torch.arange(n, dtype=torch.int64, device=dev_t.device)
-        for name in reversed(sort_cols):
-            c = batch.columns[name]
-            if c.is_string:
-                raise NotImplementedError("string PK sort on GPU not yet supported")
-            keys = c.data[idx]
-            order = torch.argsort(keys, stable=True)
-            idx = idx[order]
-        return idx
+        if not any(batch.columns[c].is_string for c in sort_cols):
+            idx = torch.arange(n, dtype=torch.int64, device=dev_t.device)
+            for name in reversed(sort_cols):
+                keys = batch.columns[name].data[idx]
+                order = torch.argsort(keys, stable=True)
+                idx = idx[order]
+            return idx
+        # string sort keys: order computed on host (lexicographic byte
+        # compare), applied on device
+        cpu_batch = batch.to_device("cpu")
+        return _sort_indices(cpu_batch, sort_cols).to(dev_t.device)
     # CPU: numpy lexsort
     keys = [np.arange(n)]
     for name in reversed(sort_cols):
