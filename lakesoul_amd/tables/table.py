@@ -158,6 +158,180 @@ class LakeSoulTable:
     def to_pandas(self, **kwargs):
         return self.to_arrow(**kwargs).to_pandas()
 
+    # -- DML: update / delete (reference: Spark LakeSoulTable.update /
+    #    delete, commands/UpdateCommand + DeleteCommand) ----------------- #
+
+    def update(self, filters, assignments: Dict[str, object],
+               device: Optional[str] = None) -> int:
+        """Update rows matching ``filters``: set ``assignments`` column
+        values. PK tables rewrite via upsert (MergeCommit). Returns the
+        number of rows updated."""
+        if not self.primary_keys:
+            raise ValueError("update requires a primary-key table")
+        from ..io.batch import concat_batches
+
+        scan = self.scan(filters=filters, device=device)
+        batches = list(scan.iter_batches())
+        if not batches:
+            return 0
+        batch = concat_batches(batches)
+        import numpy as np
+        import torch
+
+        from ..io.batch import Batch, Column, torch_dtype_for
+
+        n = batch.num_rows
+        for col, val in assignments.items():
+            f = self.schema.field(col)
+            if col in self.primary_keys:
+                raise ValueError("cannot update a primary-key column")
+            if f.is_fixed_width:
+                dev = batch.columns[col].data.device if batch.columns[col].data is not None else "cpu"
+                batch.columns[col] = Column(
+                    f.dtype,
+                    data=torch.full((n,), val, dtype=torch_dtype_for(f.dtype), device=dev),
+                )
+            else:
+                enc = (val.encode() if isinstance(val, str) else bytes(val))
+                offs = np.arange(n + 1, dtype=np.int32) * len(enc)
+                bys = np.frombuffer(enc * n, dtype=np.uint8).copy() if n else np.empty(0, np.uint8)
+                batch.columns[col] = Column(
+                    f.dtype, offsets=torch.from_numpy(offs), bytes_=torch.from_numpy(bys)
+                )
+        self.upsert(batch, device=device)
+        return n
+
+    def delete(self, filters=None, device: Optional[str] = None) -> int:
+        """Delete matching rows.
+
+        - CDC tables: write delete-marker rows (MergeCommit) — the scan
+          drops them (reference CDC semantics).
+        - non-CDC PK tables: rewrite each touched bucket without the
+          matching rows and commit an UpdateCommit whose snapshot
+          replaces the bucket's files.
+        - no filters: delete the whole partition (DeleteCommit).
+        """
+        if filters is None:
+            for desc in self.partition_descs():
+                self.delete_partition(desc)
+            return -1
+        props = self.info.get_properties()
+        cdc_col = props.get("lakesoul_cdc_change_column")
+        from ..io.batch import concat_batches
+
+        if cdc_col:
+            scan = self.scan(filters=filters, device=device)
+            batches = list(scan.iter_batches())
+            if not batches:
+                return 0
+            batch = concat_batches(batches)
+            import numpy as np
+            import torch
+
+            from ..io.batch import Column
+
+            n = batch.num_rows
+            enc = b"delete"
+            offs = np.arange(n + 1, dtype=np.int32) * len(enc)
+            bys = np.frombuffer(enc * n, dtype=np.uint8).copy() if n else np.empty(0, np.uint8)
+            batch.columns[cdc_col] = Column(
+                "string", offsets=torch.from_numpy(offs), bytes_=torch.from_numpy(bys)
+            )
+            self.upsert(batch, device=device)
+            return n
+        # rewrite path
+        if not self.primary_keys:
+            raise ValueError("delete with filters requires a PK or CDC table")
+        from ..io.reader import LakeSoulScan
+        from ..io.writer import _write_batch_to_file, random_str
+        from ..io.filters import resolve_filters
+        import os as _os
+        import torch
+
+        expr = resolve_filters(filters, self.schema)
+        deleted = 0
+        cfg = self.io_config()
+        full = LakeSoulScan(self, device=device)  # no filters: full rows
+        for unit in full.plan():
+            batch = full._read_unit(unit)
+            if batch is None:
+                continue
+            keep = ~expr.evaluate(batch)
+            n_del = int((~keep).sum())
+            if n_del == 0:
+                continue
+            deleted += n_del
+            kept = batch.take(torch.nonzero(keep, as_tuple=True)[0])
+            subdir = (
+                "/".join(unit.partition_desc.split(","))
+                if unit.partition_desc != constants.NON_PARTITION_TABLE_PART_DESC
+                else ""
+            )
+            out_dir = _os.path.join(self.table_path, subdir, constants.COMPACT_DIR)
+            _os.makedirs(out_dir, exist_ok=True)
+            fpath = _os.path.join(out_dir, f"part-{random_str(16)}_{unit.bucket_id:04d}.parquet")
+            size = _write_batch_to_file(
+                fpath, kept, cfg.compression, cfg.compression_level, cfg.max_row_group_size
+            )
+            cur = self.client.store.get_latest_partition_info(self.table_id, unit.partition_desc)
+            dci = DataCommitInfo(
+                table_id=self.table_id,
+                partition_desc=unit.partition_desc,
+                file_ops=[DataFileOp(fpath, FileOp.add, size, ",".join(kept.schema.names()))]
+                + [DataFileOp(p, FileOp.del_) for p in unit.files],
+                commit_op=CommitOp.UpdateCommit,
+            )
+            self.client.store.insert_data_commit_info(dci)
+            # snapshot-replace for this bucket: other buckets' files must
+            # stay, so resolve current files minus this bucket's + new
+            keep_commits = [dci.commit_id]
+            cur_files = self.client.files_for_partition(self.table_id, unit.partition_desc)
+            other = [f for f in cur_files if f.path not in set(unit.files)]
+            if other:
+                keep_dci = DataCommitInfo(
+                    table_id=self.table_id,
+                    partition_desc=unit.partition_desc,
+                    file_ops=other,
+                    commit_op=CommitOp.UpdateCommit,
+                )
+                self.client.store.insert_data_commit_info(keep_dci)
+                keep_commits = [keep_dci.commit_id, dci.commit_id]
+            self.client.commit_data(
+                MetaInfo(
+                    table_info=self.info,
+                    list_partition=[
+                        PartitionInfo(
+                            table_id=self.table_id,
+                            partition_desc=unit.partition_desc,
+                            snapshot=keep_commits,
+                            commit_op=CommitOp.UpdateCommit,
+                        )
+                    ],
+                    read_partition_info=[cur] if cur else [],
+                ),
+                CommitOp.UpdateCommit,
+            )
+        return deleted
+
+    # -- schema evolution (reference: alterTableCommands add columns) --- #
+
+    def add_columns(self, fields) -> None:
+        """Append nullable columns to the table schema; existing files
+        simply lack them (reads null-fill via exist_cols semantics)."""
+        from ..io.schema import Field, Schema, schema_to_json
+
+        schema = self.schema
+        for f in fields:
+            if not isinstance(f, Field):
+                f = Field(*f)
+            if not f.nullable:
+                raise ValueError("added columns must be nullable")
+            if f.name in schema.names():
+                raise ValueError(f"column {f.name} already exists")
+            schema.fields.append(f)
+        self.info.table_schema = schema_to_json(schema)
+        self.client.store.update_table_schema(self.table_id, self.info.table_schema)
+
     # -- maintenance ----------------------------------------------------- #
 
     def compaction(self, partition_desc: Optional[str] = None, device: Optional[str] = None) -> None:
