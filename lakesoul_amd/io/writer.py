@@ -26,6 +26,7 @@ import torch
 
 from .. import constants
 from ..ops import cpp
+from ..utils import timing
 from .batch import Batch, Column
 from .fs import default_fs, is_remote
 from .schema import Schema
@@ -252,12 +253,14 @@ def write_table_data(table, data, device: Optional[str] = None,
                     f"partial write must include pk+range columns, missing {missing_pk + missing_range}"
                 )
             schema = schema.select([n for n in schema.names() if n in provided])
-        batch = Batch.from_any(data, schema)
+        with timing.phase("w_from_any"):
+            batch = Batch.from_any(data, schema)
     else:
         batch = data
         schema = batch.schema
     if device == "cuda":
-        batch = batch.to_device("cuda")
+        with timing.phase("w_h2d", sync_gpu=True):
+            batch = batch.to_device("cuda")
 
     pk = table.primary_keys
     range_cols = table.range_keys
@@ -284,11 +287,12 @@ def write_table_data(table, data, device: Optional[str] = None,
             # one global stable sort by (bucket, pk): buckets become
             # contiguous PK-sorted slices — a single gather for the whole
             # batch instead of one per bucket
-            buckets = _hash_bucket_ids(file_batch, pk, num_buckets)
-            order = _sort_indices(file_batch, pk)
-            border = torch.argsort(buckets.to(torch.int64)[order], stable=True)
-            perm = order[border]
-            sorted_batch = file_batch.take(perm)
+            with timing.phase("w_hash_sort", sync_gpu=True):
+                buckets = _hash_bucket_ids(file_batch, pk, num_buckets)
+                order = _sort_indices(file_batch, pk)
+                border = torch.argsort(buckets.to(torch.int64)[order], stable=True)
+                perm = order[border]
+                sorted_batch = file_batch.take(perm)
             bucket_sorted = buckets[perm].to(torch.int64)
             counts = torch.bincount(bucket_sorted, minlength=num_buckets)
             bounds = torch.zeros(num_buckets + 1, dtype=torch.int64)
@@ -316,8 +320,9 @@ def write_table_data(table, data, device: Optional[str] = None,
                 return FlushResult(fpath, size, bb.num_rows, desc,
                                    ",".join(data_schema.names()))
 
-            with ThreadPoolExecutor(max_workers=min(8, max(1, len(jobs)))) as ex:
-                results.extend(ex.map(_encode, jobs))
+            with timing.phase("w_encode"):
+                with ThreadPoolExecutor(max_workers=min(16, max(1, len(jobs)))) as ex:
+                    results.extend(ex.map(_encode, jobs))
         else:
             fname = f"part-{random_str(16)}_{0:04d}.parquet"
             fpath = os.path.join(out_dir, fname)
