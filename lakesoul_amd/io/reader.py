@@ -240,15 +240,18 @@ class LakeSoulScan:
         (overlap engineering)."""
         from concurrent.futures import ThreadPoolExecutor
 
-        from .reader_gpu import fetch_raw, read_unit_gpu
+        import torch as _torch
 
+        from .reader_gpu import UnitTransfer, fetch_raw, read_unit_gpu
+
+        device = _torch.device("cuda")
         with ThreadPoolExecutor(max_workers=depth) as ex:
             futs = [
                 ex.submit(fetch_raw, self._localize(u.files), self.read_cols)
                 for u in units[: depth]
             ]
+            transfers: dict = {}
             for i, unit in enumerate(units):
-                raw = futs[i].result()
                 if i + depth < len(units):
                     futs.append(
                         ex.submit(
@@ -257,7 +260,14 @@ class LakeSoulScan:
                             self.read_cols,
                         )
                     )
-                yield read_unit_gpu(self, unit, raw)
+                if i not in transfers:
+                    transfers[i] = UnitTransfer(futs[i].result(), device)
+                # issue the NEXT unit's H2D on the copy stream before
+                # processing this one, so the transfer overlaps compute
+                j = i + 1
+                if j < len(units) and j not in transfers and futs[j].done():
+                    transfers[j] = UnitTransfer(futs[j].result(), device)
+                yield read_unit_gpu(self, unit, transfer=transfers.pop(i))
 
     def __iter__(self):
         return self.iter_batches()

@@ -51,24 +51,58 @@ def fetch_raw(files: List[str], names: List[str]) -> dict:
         return cpp().read_unit_raw(files, names, 0, True)
 
 
-def read_unit_gpu(scan, unit, raw: Optional[dict] = None) -> Optional[Batch]:
+_copy_stream: Optional["torch.cuda.Stream"] = None
+
+
+def _get_copy_stream():
+    global _copy_stream
+    if _copy_stream is None:
+        _copy_stream = torch.cuda.Stream()
+    return _copy_stream
+
+
+class UnitTransfer:
+    """H2D of one unit's buffers on the side copy stream — lets unit
+    k+1's transfer overlap unit k's decode/merge kernels (overlap
+    engineering)."""
+
+    def __init__(self, raw: dict, device):
+        self.raw = raw  # hold pinned host buffers until consumed
+        s = _get_copy_stream()
+        with torch.cuda.stream(s):
+            self.vals = raw["values"].to(device, non_blocking=True)
+            self.validity = (
+                raw["validity"].to(device, non_blocking=True)
+                if raw["validity"].numel() else None
+            )
+            self.dicts = raw["dicts"].to(device, non_blocking=True) if raw["dicts"].numel() else None
+            self.runs = (
+                raw["runs"].view(-1, 6).to(device, non_blocking=True)
+                if raw["runs"].numel() else None
+            )
+            self.soffs = raw["soffs"].to(device, non_blocking=True) if raw["soffs"].numel() else None
+        self.event = torch.cuda.Event()
+        self.event.record(s)
+
+
+def read_unit_gpu(scan, unit, raw: Optional[dict] = None,
+                  transfer: Optional[UnitTransfer] = None) -> Optional[Batch]:
     from .merge_gpu import merge_sorted_files_gpu
 
     device = torch.device("cuda")
     names = scan.read_cols
-    if raw is None:
-        raw = fetch_raw(unit.files, names)
-
-    with timing.phase("h2d", sync_gpu=True):
-        vals = raw["values"].to(device, non_blocking=True)
-    validity_buf = (
-        raw["validity"].to(device, non_blocking=True) if raw["validity"].numel() else None
-    )
-    dicts_buf = raw["dicts"].to(device, non_blocking=True) if raw["dicts"].numel() else None
-    runs_buf = (
-        raw["runs"].view(-1, 6).to(device, non_blocking=True) if raw["runs"].numel() else None
-    )
-    soffs_buf = raw["soffs"].to(device, non_blocking=True) if raw["soffs"].numel() else None
+    if transfer is None:
+        if raw is None:
+            raw = fetch_raw(unit.files, names)
+        with timing.phase("h2d"):
+            transfer = UnitTransfer(raw, device)
+    raw = transfer.raw
+    torch.cuda.current_stream().wait_event(transfer.event)
+    vals = transfer.vals
+    validity_buf = transfer.validity
+    dicts_buf = transfer.dicts
+    runs_buf = transfer.runs
+    soffs_buf = transfer.soffs
 
     read_schema = scan.schema.select(names)
     ncols = len(names)
