@@ -90,6 +90,7 @@ class LakeSoulScan:
         device: Optional[str] = None,
         batch_size: Optional[int] = None,
         incremental: Optional[Tuple[int, int]] = None,
+        options: Optional[dict] = None,
     ):
         self.table = table
         self.schema: Schema = table.schema
@@ -123,6 +124,9 @@ class LakeSoulScan:
                     self.eval_fields.append(self.schema.field(c))
         self.eval_schema = Schema(self.eval_fields)
         self.device = device or table.io_config().resolve_device()
+        opts = dict(options or {})
+        cfg_opt = opts.get("scan_cache", table.io_config().option("scan_cache", "0"))
+        self.use_cache = str(cfg_opt) == "1"
         self.batch_size = batch_size
         self.incremental = incremental
         self._shard: Optional[Tuple[int, int]] = None
@@ -211,7 +215,8 @@ class LakeSoulScan:
 
     def iter_batches(self) -> Iterator[Batch]:
         units = self.plan()
-        if self.device == "cuda" and len(units) > 1 and self._gpu_merge_supported():
+        if (self.device == "cuda" and len(units) > 1
+                and self._gpu_merge_supported() and not self.use_cache):
             gen = self._iter_units_pipelined(units)
         else:
             gen = (self._read_unit(u) for u in units)
@@ -319,6 +324,14 @@ class LakeSoulScan:
     def _read_unit(self, unit: ScanUnit) -> Optional[Batch]:
         if not unit.files:
             return None
+        cache_key = None
+        if self.use_cache:
+            from .hbm_cache import scan_cache
+
+            cache_key = scan_cache().key(unit, self.read_cols)
+            hit = scan_cache().get(cache_key)
+            if hit is not None:
+                return hit
         unit = ScanUnit(unit.partition_desc, unit.bucket_id,
                         self._localize(unit.files), unit.is_compacted_first)
         needs_merge = bool(self.pk) and (
@@ -331,8 +344,14 @@ class LakeSoulScan:
         if self.device == "cuda":
             from .reader_gpu import read_unit_gpu
 
-            return read_unit_gpu(self, unit)
-        return self._read_unit_cpu(unit)
+            batch = read_unit_gpu(self, unit)
+        else:
+            batch = self._read_unit_cpu(unit)
+        if cache_key is not None and batch is not None:
+            from .hbm_cache import scan_cache
+
+            scan_cache().put(cache_key, batch)
+        return batch
 
     def _read_unit_cpu(self, unit: ScanUnit) -> Optional[Batch]:
         file_cols: List[Dict[str, NpColumn]] = []

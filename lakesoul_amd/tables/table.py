@@ -137,6 +137,7 @@ class LakeSoulTable:
         filters: Optional[list] = None,
         device: Optional[str] = None,
         batch_size: Optional[int] = None,
+        options: Optional[Dict[str, str]] = None,
     ):
         """Build a LakeSoulScan (reference: catalog.py:740 LakeSoulScan)."""
         from ..io.reader import LakeSoulScan
@@ -150,6 +151,7 @@ class LakeSoulTable:
             filters=filters,
             device=device,
             batch_size=batch_size,
+            options=options,
         )
 
     def to_arrow(self, **kwargs):

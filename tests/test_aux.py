@@ -112,3 +112,35 @@ def test_table_stats(catalog):
     assert s.file_count == 4  # 2 buckets x 2 commits
     assert s.total_bytes > 0
     assert s.partitions[0].version == 1
+
+
+def test_hbm_scan_cache(catalog, monkeypatch):
+    from lakesoul_amd.io.hbm_cache import scan_cache
+
+    scan_cache().clear()
+    t = catalog.create_table(
+        "cached",
+        Schema([Field("id", "int64", False), Field("v", "float64")]),
+        primary_keys=["id"],
+        hash_bucket_num=2,
+    )
+    n = 1000
+    t.upsert({"id": np.arange(n, dtype=np.int64), "v": np.zeros(n)})
+    t.upsert({"id": np.arange(100, dtype=np.int64), "v": np.ones(100)})
+
+    import pandas as pd
+
+    df1 = t.scan(device="cpu", options={"scan_cache": "1"}).to_arrow().to_pandas()
+    s0 = scan_cache().stats()
+    assert s0["entries"] == 2 and s0["misses"] >= 2
+    df2 = t.scan(device="cpu", options={"scan_cache": "1"}).to_arrow().to_pandas()
+    s1 = scan_cache().stats()
+    assert s1["hits"] >= 2
+    pd.testing.assert_frame_equal(
+        df1.sort_values("id").reset_index(drop=True),
+        df2.sort_values("id").reset_index(drop=True),
+    )
+    # a new commit changes the key -> fresh read, correct data
+    t.upsert({"id": np.array([5], dtype=np.int64), "v": np.array([42.0])})
+    df3 = t.scan(device="cpu", options={"scan_cache": "1"}).to_arrow().to_pandas()
+    assert df3.loc[df3["id"] == 5, "v"].iloc[0] == 42.0
