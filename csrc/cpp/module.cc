@@ -370,12 +370,12 @@ static py::tuple prep_rle_runs(torch::Tensor values, torch::Tensor idx_pages) {
 // memory for the big buffers (GPU boxes).
 static py::dict read_unit_raw_py(const std::vector<std::string>& paths,
                                  const std::vector<std::string>& names,
-                                 int64_t nthreads, bool pin) {
+                                 int64_t nthreads, bool pin, bool gpu_snappy) {
   (void)nthreads;
   std::unique_ptr<UnitStage> st;
   {
     py::gil_scoped_release rel;
-    st = read_unit_stage1(paths, names);
+    st = read_unit_stage1(paths, names, gpu_snappy);
   }
   UnitStage& ud = *st;
   auto alloc_u8 = [&](int64_t n) {
@@ -392,13 +392,18 @@ static py::dict read_unit_raw_py(const std::vector<std::string>& paths,
   torch::Tensor validity = alloc_u8(ud.validity_size);
   torch::Tensor dicts = alloc_u8(ud.dicts_size);
   torch::Tensor soffs = torch::empty({ud.soffs_size}, torch::kInt64);
+  torch::Tensor comp = alloc_u8(ud.comp_size);
   torch::Tensor runs = torch::empty({(int64_t)ud.runs.size()}, torch::kInt64);
   if (!ud.runs.empty())
     std::memcpy(runs.data_ptr(), ud.runs.data(), ud.runs.size() * 8);
+  torch::Tensor sjobs = torch::empty({(int64_t)ud.snappy_jobs.size()}, torch::kInt64);
+  if (!ud.snappy_jobs.empty())
+    std::memcpy(sjobs.data_ptr(), ud.snappy_jobs.data(), ud.snappy_jobs.size() * 8);
   {
     py::gil_scoped_release rel;
     read_unit_fill(ud, (uint8_t*)values.data_ptr(), (uint8_t*)validity.data_ptr(),
-                   (uint8_t*)dicts.data_ptr(), soffs.data_ptr<int64_t>());
+                   (uint8_t*)dicts.data_ptr(), soffs.data_ptr<int64_t>(),
+                   ud.comp_size ? (uint8_t*)comp.data_ptr() : nullptr);
   }
   py::dict d;
   d["values"] = values;
@@ -406,6 +411,8 @@ static py::dict read_unit_raw_py(const std::vector<std::string>& paths,
   d["dicts"] = dicts;
   d["runs"] = runs;
   d["soffs"] = soffs;
+  d["comp"] = comp;
+  d["snappy_jobs"] = sjobs;
   py::list frows;
   for (auto r : ud.file_rows) frows.append(r);
   d["file_rows"] = frows;
@@ -559,7 +566,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("read_chunk_cpu", &read_chunk_cpu);
   m.def("read_chunks_cpu_batch", &read_chunks_cpu_batch);
   m.def("prep_rle_runs", &prep_rle_runs);
-  m.def("read_unit_raw", &read_unit_raw_py);
+  m.def("read_unit_raw", &read_unit_raw_py, py::arg("paths"), py::arg("names"),
+        py::arg("nthreads") = 0, py::arg("pin") = true, py::arg("gpu_snappy") = false);
   m.def("read_chunks_raw_batch", &read_chunks_raw_batch);
   m.def("hash_columns_cpu", &hash_columns_cpu);
   m.def("hash_string_column_cpu", &hash_string_column_cpu);

@@ -361,6 +361,13 @@ class ParquetFile {
     int32_t bit_width;
   };
 
+  struct CompPage {  // GPU-decompress job (snappy, levels-free pages)
+    int64_t comp_off;   // into comp buffer
+    int64_t comp_len;
+    int64_t out_off;    // into the chunk's values stream
+    int64_t out_len;
+  };
+
   struct ChunkData {
     int32_t physical = 0;
     int64_t num_values = 0;  // rows in chunk
@@ -371,9 +378,19 @@ class ParquetFile {
     std::vector<uint8_t> dict;      // PLAIN dictionary payload
     int64_t dict_num_values = 0;
     std::vector<IdxPage> idx_pages;
+    // gpu_compressed mode: values stays EMPTY; `comp` holds the raw
+    // snappy page bodies and `comp_pages` the decompress jobs; values
+    // stream length is values_len (sum of uncompressed page payloads)
+    bool gpu_compressed = false;
+    int64_t values_len = 0;
+    std::vector<uint8_t> comp;
+    std::vector<CompPage> comp_pages;
   };
 
-  ChunkData read_chunk(size_t rg, size_t col) const {
+  // gpu_snappy: defer SNAPPY page decompression to the GPU kernel when
+  // the column is REQUIRED (no def-level section inside the compressed
+  // blob), PLAIN-encoded, non-boolean v1 pages.
+  ChunkData read_chunk(size_t rg, size_t col, bool gpu_snappy = false) const {
     const RowGroup& g = meta_.row_groups.at(rg);
     const ColumnMeta& cm = g.columns.at(col);
     const ColumnDesc& cd = cols_.at(col);
@@ -410,6 +427,29 @@ class ParquetFile {
         continue;
       }
       if (ph.type != PAGE_DATA && ph.type != PAGE_DATA_V2) continue;
+
+      if (out.gpu_compressed &&
+          !(cm.codec == CODEC_SNAPPY && ph.type == PAGE_DATA &&
+            ph.encoding == ENC_PLAIN)) {
+        // mixed chunk (e.g. dict fallback) — redo fully on host
+        return read_chunk(rg, col, false);
+      }
+      if (gpu_snappy && cm.codec == CODEC_SNAPPY && ph.type == PAGE_DATA &&
+          ph.encoding == ENC_PLAIN && !cd.nullable &&
+          cd.physical != PT_BOOLEAN && cd.physical != PT_BYTE_ARRAY &&
+          out.dict.empty() && out.values.empty()) {
+        out.gpu_compressed = true;
+        CompPage cp;
+        cp.comp_off = (int64_t)out.comp.size();
+        cp.comp_len = ph.compressed_size;
+        cp.out_off = out.values_len;
+        cp.out_len = ph.uncompressed_size;
+        out.comp.insert(out.comp.end(), body, body + ph.compressed_size);
+        out.comp_pages.push_back(cp);
+        out.values_len += ph.uncompressed_size;
+        values_seen += ph.num_values;
+        continue;
+      }
 
       std::vector<uint8_t> page;
       const uint8_t* vals;

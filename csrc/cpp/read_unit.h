@@ -122,12 +122,15 @@ struct UnitStage {
   std::vector<UnitColumn> cols;
   std::vector<int64_t> file_rows;
   std::vector<int64_t> runs;  // [m][6]
+  std::vector<int64_t> snappy_jobs;  // [m][4]: comp_off, comp_len, dst_off(values), dst_len
   std::vector<std::unique_ptr<StrDecoded>> str_cols;
-  int64_t values_size = 0, validity_size = 0, dicts_size = 0, soffs_size = 0;
+  int64_t values_size = 0, validity_size = 0, dicts_size = 0, soffs_size = 0,
+          comp_size = 0;
 };
 
 inline std::unique_ptr<UnitStage> read_unit_stage1(
-    const std::vector<std::string>& paths, const std::vector<std::string>& names) {
+    const std::vector<std::string>& paths, const std::vector<std::string>& names,
+    bool gpu_snappy = false) {
   auto st = std::make_unique<UnitStage>();
   UnitStage& S = *st;
   size_t nfiles = paths.size();
@@ -162,7 +165,7 @@ inline std::unique_ptr<UnitStage> read_unit_stage1(
       try {
         const Task& t = tasks[i];
         auto& fd = S.files[t.fi];
-        fd.chunks[t.c][t.rg] = fd.f->read_chunk(t.rg, fd.col_idx[t.c]);
+        fd.chunks[t.c][t.rg] = fd.f->read_chunk(t.rg, fd.col_idx[t.c], gpu_snappy);
       } catch (std::exception& e) {
         std::lock_guard<std::mutex> lk(err_mu);
         err = e.what();
@@ -249,7 +252,20 @@ inline std::unique_ptr<UnitStage> read_unit_stage1(
       } else {
         uc.val_off = vpos;
         int64_t plen = 0;
-        for (auto& ch : chs) plen += (int64_t)ch.values.size();
+        for (auto& ch : chs) {
+          if (ch.gpu_compressed) {
+            for (auto& cp : ch.comp_pages) {
+              S.snappy_jobs.push_back(S.comp_size + cp.comp_off);
+              S.snappy_jobs.push_back(cp.comp_len);
+              S.snappy_jobs.push_back(vpos + plen + cp.out_off);
+              S.snappy_jobs.push_back(cp.out_len);
+            }
+            S.comp_size += (int64_t)ch.comp.size();
+            plen += ch.values_len;
+          } else {
+            plen += (int64_t)ch.values.size();
+          }
+        }
         uc.val_len = plen;
         vpos += ru_align8(plen);
       }
@@ -303,7 +319,23 @@ inline std::unique_ptr<UnitStage> read_unit_stage1(
 
 // Fill caller-allocated buffers (sized from stage1) — pure parallel memcpy.
 inline void read_unit_fill(UnitStage& S, uint8_t* values, uint8_t* validity,
-                           uint8_t* dicts, int64_t* soffs) {
+                           uint8_t* dicts, int64_t* soffs, uint8_t* comp) {
+  // compressed snappy page bodies, in the same chunk order the layout
+  // assigned (serial: tiny relative to values)
+  if (comp) {
+    int64_t cpos = 0;
+    for (size_t u = 0; u < S.cols.size(); u++) {
+      UnitColumn& uc = S.cols[u];
+      if (!uc.present || uc.is_string || uc.is_dict) continue;
+      auto& chs = S.files[uc.file_idx].chunks[u % S.ncols];
+      for (auto& ch : chs) {
+        if (ch.gpu_compressed && !ch.comp.empty()) {
+          std::memcpy(comp + cpos, ch.comp.data(), ch.comp.size());
+          cpos += (int64_t)ch.comp.size();
+        }
+      }
+    }
+  }
   if (S.validity_size) std::memset(validity, 1, (size_t)S.validity_size);
   ThreadPool::instance().parallel_for((int64_t)S.cols.size(), [&](int64_t u) {
     UnitColumn& uc = S.cols[u];
@@ -338,6 +370,10 @@ inline void read_unit_fill(UnitStage& S, uint8_t* values, uint8_t* validity,
     }
     int64_t off = uc.val_off;
     for (auto& ch : chs) {
+      if (ch.gpu_compressed) {
+        off += ch.values_len;  // filled by the GPU snappy kernel
+        continue;
+      }
       std::memcpy(values + off, ch.values.data(), ch.values.size());
       off += (int64_t)ch.values.size();
     }

@@ -324,9 +324,23 @@ static std::vector<torch::Tensor> snappy_decompress(torch::Tensor src,
   return {dst, status};
 }
 
+// decompress into an existing device buffer (jobs dst offsets index it)
+static torch::Tensor snappy_decompress_into(torch::Tensor src, torch::Tensor jobs,
+                                            torch::Tensor dst) {
+  CHECK_GPU(src);
+  CHECK_GPU(jobs);
+  CHECK_GPU(dst);
+  auto status = torch::empty({jobs.size(0)}, src.options().dtype(torch::kInt32));
+  launch_snappy_decompress(src.data_ptr<uint8_t>(), jobs.data_ptr<int64_t>(),
+                           jobs.size(0), dst.data_ptr<uint8_t>(),
+                           status.data_ptr<int32_t>(), cur_stream());
+  return status;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ann_scores", &ann_scores);
   m.def("snappy_decompress", &snappy_decompress);
+  m.def("snappy_decompress_into", &snappy_decompress_into);
   m.doc() = "lakesoul_amd gfx950 HIP kernels";
   m.def("hash_fixed_column", &hash_fixed_column);
   m.def("hash_string_column", &hash_string_column);

@@ -48,7 +48,7 @@ def fetch_raw(files: List[str], names: List[str]) -> dict:
     """Host phase of a unit read (releases the GIL in C++) — safe to run
     on a prefetch thread while the GPU processes the previous unit."""
     with timing.phase("host_fetch"):
-        return cpp().read_unit_raw(files, names, 0, True)
+        return cpp().read_unit_raw(files, names, 0, True, True)
 
 
 _copy_stream: Optional["torch.cuda.Stream"] = None
@@ -80,6 +80,12 @@ class UnitTransfer:
             if raw["runs"].numel() else None
         )
         self.soffs = raw["soffs"].to(device, non_blocking=True) if raw["soffs"].numel() else None
+        self.comp = raw["comp"].to(device, non_blocking=True) if raw.get("comp") is not None and raw["comp"].numel() else None
+        self.snappy_jobs = (
+            raw["snappy_jobs"].view(-1, 4).to(device, non_blocking=True)
+            if raw.get("snappy_jobs") is not None and raw["snappy_jobs"].numel()
+            else None
+        )
         self.event = None
 
 
@@ -102,6 +108,13 @@ def read_unit_gpu(scan, unit, raw: Optional[dict] = None,
     dicts_buf = transfer.dicts
     runs_buf = transfer.runs
     soffs_buf = transfer.soffs
+    if transfer.snappy_jobs is not None:
+        # GPU snappy: decompress page bodies straight into the values
+        # buffer (wave-per-page kernel) — no host decompression happened
+        # for these chunks
+        status = hip().snappy_decompress_into(transfer.comp, transfer.snappy_jobs, vals)
+        if bool((status != 0).any()):
+            raise RuntimeError(f"GPU snappy decompression failed: {status.cpu().tolist()}")
 
     read_schema = scan.schema.select(names)
     ncols = len(names)
