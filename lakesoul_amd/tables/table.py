@@ -372,6 +372,57 @@ class LakeSoulTable:
     def rollback(self, partition_desc: str, version: int) -> None:
         self.client.rollback_partition(self.table_id, partition_desc, version)
 
+    def cleanup_old_versions(self, keep_latest: int = 1, delete_files: bool = True) -> int:
+        """Vacuum: drop partition versions older than the newest
+        ``keep_latest`` and delete data files no longer referenced by any
+        kept snapshot (reference: CleanExpiredData.scala /
+        LakeSoulTable.cleanUpPartitionData). Returns files removed."""
+        import os as _os
+
+        removed = 0
+        store = self.client.store
+        for desc in self.partition_descs():
+            cur = store.get_latest_partition_info(self.table_id, desc)
+            if cur is None:
+                continue
+            cutoff = cur.version - keep_latest + 1
+            if cutoff <= 0:
+                continue
+            old = store.get_partition_versions_in_range(self.table_id, desc, 0, cutoff - 1)
+            kept = store.get_partition_versions_in_range(
+                self.table_id, desc, cutoff, cur.version
+            )
+            kept_files = set()
+            kept_commits = set()
+            for p in kept:
+                kept_commits.update(p.snapshot)
+                for f in self.client._resolve_snapshot_files(self.table_id, desc, p.snapshot):
+                    kept_files.add(f.path)
+            old_files = set()
+            for p in old:
+                for f in self.client._resolve_snapshot_files(self.table_id, desc, p.snapshot):
+                    old_files.add(f.path)
+                for cid in p.snapshot:
+                    if cid not in kept_commits:
+                        store.delete_data_commit_info(self.table_id, desc, cid)
+            store.delete_partition_versions_since(self.table_id, desc, 0)
+            # reinsert kept versions (delete_partition_versions_since drops all)
+            from ..meta.store import CommitConflictError
+
+            try:
+                store.transaction_insert_partition_info(kept)
+            except CommitConflictError:
+                pass
+            if delete_files:
+                for path in old_files - kept_files:
+                    try:
+                        if _os.path.exists(path):
+                            _os.remove(path)
+                        removed += 1
+                    except OSError:
+                        pass
+        return removed
+
     # -- introspection ---------------------------------------------------- #
 
     def partition_descs(self) -> List[str]:

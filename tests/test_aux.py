@@ -144,3 +144,26 @@ def test_hbm_scan_cache(catalog, monkeypatch):
     t.upsert({"id": np.array([5], dtype=np.int64), "v": np.array([42.0])})
     df3 = t.scan(device="cpu", options={"scan_cache": "1"}).to_arrow().to_pandas()
     assert df3.loc[df3["id"] == 5, "v"].iloc[0] == 42.0
+
+
+def test_cleanup_old_versions(catalog):
+    t = catalog.create_table(
+        "vac",
+        Schema([Field("id", "int64", False), Field("v", "float64")]),
+        primary_keys=["id"],
+        hash_bucket_num=1,
+    )
+    for i in range(5):
+        t.upsert({"id": np.arange(10, dtype=np.int64), "v": np.full(10, float(i))})
+    t.compaction()  # compacted snapshot replaces the deltas
+    import os
+
+    all_files_before = [f.path for f in t.files()]
+    removed = t.cleanup_old_versions(keep_latest=1)
+    assert removed >= 5  # old delta files physically gone
+    df = t.to_pandas()
+    assert df["v"].tolist() == [4.0] * 10
+    # old version no longer resolvable
+    assert t.client.files_for_partition(t.table_id, "-5", version=0) == []
+    for f in t.files():
+        assert os.path.exists(f.path)
