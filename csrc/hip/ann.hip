@@ -37,20 +37,36 @@ __global__ __launch_bounds__(256) void ann_scores_kernel(
   bool row_ok = xrow < n;
   const short* xp = X + (row_ok ? xrow * K : 0);
 
-  for (int q0 = 0; q0 < nq; q0 += 16) {
-    const short* qp = Q + (int64_t)(q0 + r) * K;
-    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  // process up to 4 query tiles (64 queries) per X pass: X bytes are the
+  // memory-bound term, so amortizing them 4x quadruples arithmetic
+  // intensity (each X fragment feeds 4 MFMAs)
+  for (int q0 = 0; q0 < nq; q0 += 64) {
+    int ntiles = (nq - q0) / 16;
+    if (ntiles > 4) ntiles = 4;
+    f32x4 acc[4] = {{0.f, 0.f, 0.f, 0.f},
+                    {0.f, 0.f, 0.f, 0.f},
+                    {0.f, 0.f, 0.f, 0.f},
+                    {0.f, 0.f, 0.f, 0.f}};
     for (int k0 = 0; k0 < K; k0 += 32) {
       bf16x8 a = row_ok ? *(const bf16x8*)(xp + k0 + khalf * 8)
                         : (bf16x8)(short)0;
-      bf16x8 b = *(const bf16x8*)(qp + k0 + khalf * 8);
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+#pragma unroll
+      for (int t = 0; t < 4; t++) {
+        if (t >= ntiles) break;
+        const short* qp = Q + (int64_t)(q0 + t * 16 + r) * K;
+        bf16x8 b = *(const bf16x8*)(qp + k0 + khalf * 8);
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t], 0, 0, 0);
+      }
     }
     // D mapping: col=lane&15 (query), row=(lane>>4)*4+reg (vector)
 #pragma unroll
-    for (int reg = 0; reg < 4; reg++) {
-      int64_t orow = row0 + khalf * 4 + reg;
-      if (orow < n) out[orow * nq + q0 + r] = acc[reg];
+    for (int t = 0; t < 4; t++) {
+      if (t >= ntiles) break;
+#pragma unroll
+      for (int reg = 0; reg < 4; reg++) {
+        int64_t orow = row0 + khalf * 4 + reg;
+        if (orow < n) out[orow * nq + q0 + t * 16 + r] = acc[t][reg];
+      }
     }
   }
 }
