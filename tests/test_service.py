@@ -106,3 +106,64 @@ def test_cli_basics(catalog, tmp_path, monkeypatch):
     assert r.exit_code == 0 and "10 rows" in r.output
     r = runner.invoke(cli_mod.cli, ["history", "clit"])
     assert "MergeCommit" in r.output
+
+
+def test_s3_proxy_rbac(tmp_path, monkeypatch):
+    fastapi = pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+
+    from lakesoul_amd.meta.client import MetaClient
+    from lakesoul_amd.meta.store import SqliteMetaStore
+    from lakesoul_amd.tables.catalog import LakeSoulCatalog
+    from lakesoul_amd.service.s3_proxy import create_s3_proxy
+
+    monkeypatch.setenv("LAKESOUL_MOCK_FS_ROOT", str(tmp_path / "remote"))
+    monkeypatch.setenv("LAKESOUL_CACHE_DIR", str(tmp_path / "cache"))
+    import lakesoul_amd.io.fs as fsmod
+
+    fsmod._default_fs = None
+    (tmp_path / "remote").mkdir()
+
+    catalog = LakeSoulCatalog(
+        MetaClient(SqliteMetaStore(str(tmp_path / "meta.db"))),
+        warehouse=str(tmp_path / "wh"),
+    )
+    # a table in a restricted domain at a mock:// path
+    t = catalog.create_table(
+        "secret_t",
+        Schema([Field("id", "int64", False)]),
+        table_path="mock://lake/secret_t",
+    )
+    t.info.domain = "teamA"
+    catalog.client.store._conn().execute(
+        "UPDATE table_info SET domain='teamA' WHERE table_id=?", (t.table_id,)
+    )
+    catalog.client.store._conn().commit()
+
+    client = TestClient(create_s3_proxy(catalog, backend_scheme="mock", secret="s"))
+
+    def tok(domain):
+        from lakesoul_amd.service.server import TokenService
+
+        return {"Authorization": f"Bearer {TokenService('s').issue('u', domain)}"}
+
+    # write an object into the table path as teamA
+    r = client.put("/lake/secret_t/part-abc_0000.parquet", content=b"DATA",
+                   headers=tok("teamA"))
+    assert r.status_code == 200
+    # teamA can read it back
+    r = client.get("/lake/secret_t/part-abc_0000.parquet", headers=tok("teamA"))
+    assert r.status_code == 200 and r.content == b"DATA"
+    # teamB is denied
+    r = client.get("/lake/secret_t/part-abc_0000.parquet", headers=tok("teamB"))
+    assert r.status_code == 403
+    # unauthenticated is rejected
+    r = client.get("/lake/secret_t/part-abc_0000.parquet")
+    assert r.status_code == 401
+    # objects outside any table pass through for any authenticated domain
+    r = client.put("/lake/free/obj.bin", content=b"x", headers=tok("teamB"))
+    assert r.status_code == 200
+    r = client.get("/lake/free/obj.bin", headers=tok("teamB"))
+    assert r.status_code == 200
+    m = client.get("/__metrics").json()
+    assert m["denied"] == 1 and m["allowed"] >= 4
