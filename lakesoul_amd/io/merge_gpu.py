@@ -145,28 +145,31 @@ def merge_sorted_files_gpu(
             start[0] = True
             for sp in sorted_pks:
                 start[1:] |= sp[1:] != sp[:-1]
-    grp = torch.cumsum(start.to(torch.int64), 0) - 1
-    ngroups = int(grp[-1].item()) + 1 if n else 0
     # last row index (in sorted space) per group
     end_mask = torch.zeros(n, dtype=torch.bool, device=device)
     if n:
         end_mask[-1] = True
         end_mask[:-1] = start[1:]
 
-    # seq (file index) per sorted row — needed for *Last-per-stream ops
-    # and partial-column contributions
-    seq = torch.cat(
-        [
-            torch.full((c,), i, dtype=torch.int32, device=device)
-            for i, c in enumerate(counts)
-        ]
-    ) if n else torch.empty(0, dtype=torch.int32, device=device)
-    seq_sorted = seq[order]
-
     simple = all(merge_ops.get(nm, "UseLast") == "UseLast" for nm in names)
     uniform_cols = present is None or all(
         set(names) <= p for p in present
     )
+
+    grp = seq_sorted = None
+    ngroups = 0
+    if not (simple and uniform_cols):
+        # group ids + file seq per sorted row: only the merge-operator /
+        # partial-column paths need them (one extra device sync)
+        grp = torch.cumsum(start.to(torch.int64), 0) - 1
+        ngroups = int(grp[-1].item()) + 1 if n else 0
+        seq = torch.cat(
+            [
+                torch.full((c,), i, dtype=torch.int32, device=device)
+                for i, c in enumerate(counts)
+            ]
+        ) if n else torch.empty(0, dtype=torch.int32, device=device)
+        seq_sorted = seq[order]
 
     cat_cols: Dict[str, Column] = {}
 
