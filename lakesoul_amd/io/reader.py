@@ -240,39 +240,55 @@ class LakeSoulScan:
                 yield batch
 
     def _iter_units_pipelined(self, units: List[ScanUnit], depth: int = 3):
-        """GPU path: prefetch the host IO/decompress of the next units on
-        background threads while the GPU decodes+merges the current one
-        (overlap engineering)."""
+        """GPU path, two-level pipeline (overlap engineering, SURVEY.md
+        §7.2 item 5):
+        - host stage: prefetch IO/decompress of upcoming units on
+          ``depth`` threads (C++ releases the GIL);
+        - GPU stage: two worker threads, each with its own HIP stream,
+          run H2D + decode + merge for alternating units — unit k+1's
+          transfers and kernels overlap unit k's (each unit ends with an
+          unavoidable sync when the merged row count materializes).
+        Results are yielded in plan order."""
         from concurrent.futures import ThreadPoolExecutor
 
         import torch as _torch
 
-        from .reader_gpu import UnitTransfer, fetch_raw, read_unit_gpu
+        from .reader_gpu import fetch_raw, read_unit_gpu
 
-        device = _torch.device("cuda")
-        with ThreadPoolExecutor(max_workers=depth) as ex:
-            futs = [
-                ex.submit(fetch_raw, self._localize(u.files), self.read_cols)
+        nstreams = int(os.environ.get("LAKESOUL_SCAN_STREAMS", "2"))
+        streams = [_torch.cuda.Stream() for _ in range(max(1, nstreams))]
+
+        def proc(i, unit, fetch_fut):
+            raw = fetch_fut.result()  # resolve on the GPU worker thread
+            s = streams[i % len(streams)]
+            with _torch.cuda.stream(s):
+                batch = read_unit_gpu(self, unit, raw)
+            s.synchronize()
+            return batch
+
+        with ThreadPoolExecutor(max_workers=depth) as fex, ThreadPoolExecutor(
+            max_workers=len(streams)
+        ) as gex:
+            fetches = [
+                fex.submit(fetch_raw, self._localize(u.files), self.read_cols)
                 for u in units[: depth]
             ]
-            transfers: dict = {}
+            procs: dict = {}
+            for i in range(min(len(streams), len(units))):
+                procs[i] = gex.submit(proc, i, units[i], fetches[i])
             for i, unit in enumerate(units):
                 if i + depth < len(units):
-                    futs.append(
-                        ex.submit(
+                    fetches.append(
+                        fex.submit(
                             fetch_raw,
                             self._localize(units[i + depth].files),
                             self.read_cols,
                         )
                     )
-                if i not in transfers:
-                    transfers[i] = UnitTransfer(futs[i].result(), device)
-                # issue the NEXT unit's H2D on the copy stream before
-                # processing this one, so the transfer overlaps compute
-                j = i + 1
-                if j < len(units) and j not in transfers and futs[j].done():
-                    transfers[j] = UnitTransfer(futs[j].result(), device)
-                yield read_unit_gpu(self, unit, transfer=transfers.pop(i))
+                j = i + len(streams)
+                if j < len(units):
+                    procs[j] = gex.submit(proc, j, units[j], fetches[j])
+                yield procs.pop(i).result()
 
     def __iter__(self):
         return self.iter_batches()
