@@ -189,6 +189,16 @@ def main():
     for w in range(args.warmup):
         one_scan()
     barrier()
+    if os.environ.get("LAKESOUL_PROFILE") == "1" and rank == 0:
+        import cProfile, pstats, io as _io
+
+        pr = cProfile.Profile()
+        pr.enable()
+        one_scan()
+        pr.disable()
+        s = _io.StringIO()
+        pstats.Stats(pr, stream=s).sort_stats("cumulative").print_stats(35)
+        print(s.getvalue(), file=sys.stderr, flush=True)
     t_start = time.time()
     rows_per_step = 0
     for s in range(args.steps):
