@@ -216,7 +216,8 @@ class LakeSoulScan:
     def iter_batches(self) -> Iterator[Batch]:
         units = self.plan()
         if (self.device == "cuda" and len(units) > 1
-                and self._gpu_merge_supported() and not self.use_cache):
+                and self._gpu_merge_supported() and not self.use_cache
+                and os.environ.get("LAKESOUL_SCAN_PIPELINE", "1") != "0"):
             gen = self._iter_units_pipelined(units)
         else:
             gen = (self._read_unit(u) for u in units)
