@@ -8,6 +8,7 @@
 #include <mutex>
 #include <thread>
 #include <unordered_map>
+#include <chrono>
 
 #include "murmur3.h"
 #include "parquet_file.h"
@@ -373,10 +374,12 @@ static py::dict read_unit_raw_py(const std::vector<std::string>& paths,
                                  int64_t nthreads, bool pin, bool gpu_snappy) {
   (void)nthreads;
   std::unique_ptr<UnitStage> st;
+  auto t0 = std::chrono::steady_clock::now();
   {
     py::gil_scoped_release rel;
     st = read_unit_stage1(paths, names, gpu_snappy);
   }
+  auto t1 = std::chrono::steady_clock::now();
   UnitStage& ud = *st;
   auto alloc_u8 = [&](int64_t n) {
     auto opts = torch::TensorOptions().dtype(torch::kUInt8);
@@ -399,13 +402,21 @@ static py::dict read_unit_raw_py(const std::vector<std::string>& paths,
   torch::Tensor sjobs = torch::empty({(int64_t)ud.snappy_jobs.size()}, torch::kInt64);
   if (!ud.snappy_jobs.empty())
     std::memcpy(sjobs.data_ptr(), ud.snappy_jobs.data(), ud.snappy_jobs.size() * 8);
+  auto t2 = std::chrono::steady_clock::now();
   {
     py::gil_scoped_release rel;
     read_unit_fill(ud, (uint8_t*)values.data_ptr(), (uint8_t*)validity.data_ptr(),
                    (uint8_t*)dicts.data_ptr(), soffs.data_ptr<int64_t>(),
                    ud.comp_size ? (uint8_t*)comp.data_ptr() : nullptr);
   }
+  auto t3 = std::chrono::steady_clock::now();
   py::dict d;
+  auto us = [](auto a, auto b) {
+    return std::chrono::duration_cast<std::chrono::microseconds>(b - a).count();
+  };
+  d["t_stage1_us"] = us(t0, t1);
+  d["t_alloc_us"] = us(t1, t2);
+  d["t_fill_us"] = us(t2, t3);
   d["values"] = values;
   d["validity"] = validity;
   d["dicts"] = dicts;

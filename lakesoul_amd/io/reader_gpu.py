@@ -48,7 +48,13 @@ def fetch_raw(files: List[str], names: List[str]) -> dict:
     """Host phase of a unit read (releases the GIL in C++) — safe to run
     on a prefetch thread while the GPU processes the previous unit."""
     with timing.phase("host_fetch"):
-        return cpp().read_unit_raw(files, names, 0, True, True)
+        raw = cpp().read_unit_raw(files, names, 0, True, True)
+    if timing.ENABLED:
+        timing._acc["fetch.stage1"] += raw["t_stage1_us"] / 1e6
+        timing._acc["fetch.alloc"] += raw["t_alloc_us"] / 1e6
+        timing._acc["fetch.fill"] += raw["t_fill_us"] / 1e6
+        timing._cnt["fetch.stage1"] += 1
+    return raw
 
 
 _copy_stream: Optional["torch.cuda.Stream"] = None
