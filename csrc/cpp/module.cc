@@ -451,6 +451,9 @@ static py::dict read_unit_raw_py(const std::vector<std::string>& paths,
     cols.append(cd);
   }
   d["cols"] = cols;
+  // free the staging structures (dozens of MB of chunk vectors) off the
+  // critical path — destructor cost was visible in the fetch time
+  std::thread([p = st.release()]() { delete p; }).detach();
   return d;
 }
 

@@ -408,8 +408,13 @@ class ParquetFile {
       off = cm.data_page_offset;
     int64_t end = off + cm.total_compressed_size;
     int64_t values_seen = 0;
+    out.values.reserve((size_t)cm.total_uncompressed_size);
     bool any_null_page = false;
-    out.validity.assign((size_t)cm.num_values, 1);
+    // validity allocated lazily on the first null (REQUIRED columns and
+    // fully-valid chunks never pay the memset)
+    auto ensure_validity = [&]() {
+      if (out.validity.empty()) out.validity.assign((size_t)cm.num_values, 1);
+    };
 
     while (off < end && values_seen < cm.num_values) {
       ThriftReader r(map_ + off, (size_t)(end - off));
@@ -472,6 +477,7 @@ class ParquetFile {
           rle_decode<uint8_t>(p + 4, lv_len, 1, nv, levels.data());
           for (int64_t i = 0; i < nv; i++) {
             if (!levels[i]) {
+              ensure_validity();
               out.validity[values_seen + i] = 0;
               page_nulls++;
             }
@@ -489,6 +495,7 @@ class ParquetFile {
                               ph.def_levels_byte_length, 1, nv, levels.data());
           for (int64_t i = 0; i < nv; i++) {
             if (!levels[i]) {
+              ensure_validity();
               out.validity[values_seen + i] = 0;
               page_nulls++;
             }
@@ -538,7 +545,7 @@ class ParquetFile {
       }
       values_seen += nv;
     }
-    if (!any_null_page && out.null_count == 0) out.validity.clear();
+    (void)any_null_page;
     return out;
   }
 
