@@ -415,6 +415,9 @@ static py::dict read_unit_raw_py(const std::vector<std::string>& paths,
     return std::chrono::duration_cast<std::chrono::microseconds>(b - a).count();
   };
   d["t_stage1_us"] = us(t0, t1);
+  d["t_open_us"] = ud.t_open_us;
+  d["t_chunks_us"] = ud.t_chunks_us;
+  d["t_layout_us"] = ud.t_layout_us;
   d["t_alloc_us"] = us(t1, t2);
   d["t_fill_us"] = us(t2, t3);
   d["values"] = values;

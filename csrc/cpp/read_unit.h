@@ -31,6 +31,8 @@
 #include <unordered_map>
 #include <vector>
 
+#include <chrono>
+
 #include "parquet_file.h"
 #include "rle.h"
 #include "thread_pool.h"
@@ -117,6 +119,7 @@ struct UnitStage {
     std::vector<uint8_t> bytes;
   };
 
+  int64_t t_open_us = 0, t_chunks_us = 0, t_layout_us = 0;
   size_t ncols = 0;
   std::vector<FileData> files;
   std::vector<UnitColumn> cols;
@@ -133,6 +136,11 @@ inline std::unique_ptr<UnitStage> read_unit_stage1(
     bool gpu_snappy = false) {
   auto st = std::make_unique<UnitStage>();
   UnitStage& S = *st;
+  auto now = [] { return std::chrono::steady_clock::now(); };
+  auto us = [](auto a, auto b) {
+    return std::chrono::duration_cast<std::chrono::microseconds>(b - a).count();
+  };
+  auto tp0 = now();
   size_t nfiles = paths.size();
   S.ncols = names.size();
   S.files.resize(nfiles);
@@ -158,6 +166,7 @@ inline std::unique_ptr<UnitStage> read_unit_stage1(
       for (size_t rg = 0; rg < nrg; rg++) tasks.push_back({i, c, rg});
     }
   }
+  auto tp1 = now();
   {
     std::string err;
     std::mutex err_mu;
@@ -173,6 +182,9 @@ inline std::unique_ptr<UnitStage> read_unit_stage1(
     });
     if (!err.empty()) throw std::runtime_error(err);
   }
+  auto tp2 = now();
+  S.t_open_us = us(tp0, tp1);
+  S.t_chunks_us = us(tp1, tp2);
 
   // layout
   int64_t vpos = 0, vapos = 0, dpos = 0, spos = 0;
@@ -314,6 +326,7 @@ inline std::unique_ptr<UnitStage> read_unit_stage1(
   S.validity_size = vapos;
   S.dicts_size = dpos;
   S.soffs_size = spos;
+  S.t_layout_us = us(tp2, now());
   return st;
 }
 

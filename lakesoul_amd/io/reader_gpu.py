@@ -51,6 +51,9 @@ def fetch_raw(files: List[str], names: List[str]) -> dict:
         raw = cpp().read_unit_raw(files, names, 0, True, True)
     if timing.ENABLED:
         timing._acc["fetch.stage1"] += raw["t_stage1_us"] / 1e6
+        timing._acc["fetch.s1_open"] += raw["t_open_us"] / 1e6
+        timing._acc["fetch.s1_chunks"] += raw["t_chunks_us"] / 1e6
+        timing._acc["fetch.s1_layout"] += raw["t_layout_us"] / 1e6
         timing._acc["fetch.alloc"] += raw["t_alloc_us"] / 1e6
         timing._acc["fetch.fill"] += raw["t_fill_us"] / 1e6
         timing._cnt["fetch.stage1"] += 1
