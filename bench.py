@@ -189,6 +189,10 @@ def main():
     for w in range(args.warmup):
         one_scan()
     barrier()
+    from lakesoul_amd.utils import timing as _tm
+
+    if _tm.ENABLED:
+        _tm.reset()  # report warm steps only
     if os.environ.get("LAKESOUL_PROFILE") == "1" and rank == 0:
         import cProfile, pstats, io as _io
 
