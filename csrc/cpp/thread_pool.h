@@ -1,11 +1,15 @@
 // Persistent host thread pool — avoids spawning threads per read call
 // (the GPU boxes have 256 cores; spawn cost per call would dominate).
-// Safe under concurrent parallel_for callers: jobs are reference-counted
-// and a task index is only dispatched while its own job is live, so a
-// straggler worker can never touch a completed job's (stack-allocated)
-// function object.
+//
+// Supports CONCURRENT parallel_for callers: the scan pipeline prefetches
+// several units at once and each fetch issues its own job; workers drain
+// every active job, so overlapping fetches genuinely overlap instead of
+// serializing on a single-job mutex. Jobs are reference-counted; a task
+// index is only dispatched while its job is live, so a straggler worker
+// can never touch a completed job's (stack-allocated) function object.
 #pragma once
 
+#include <algorithm>
 #include <atomic>
 #include <condition_variable>
 #include <functional>
@@ -24,27 +28,27 @@ class ThreadPool {
   }
 
   // Run fn(i) for i in [0, n) across the pool; blocks until done.
-  // Concurrent callers serialize (each still uses the whole pool).
   void parallel_for(int64_t n, const std::function<void(int64_t)>& fn) {
     if (n <= 0) return;
     if (n == 1) {
       fn(0);
       return;
     }
-    std::lock_guard<std::mutex> job_lock(job_mu_);
     auto job = std::make_shared<Job>();
     job->fn = &fn;
     job->n = n;
     {
       std::lock_guard<std::mutex> lk(mu_);
-      cur_ = job;
+      active_.push_back(job);
+      gen_++;
       cv_.notify_all();
     }
-    run_job(*job);  // calling thread participates
+    run_job(*job);  // calling thread participates in its own job
     {
       std::unique_lock<std::mutex> lk(mu_);
       done_cv_.wait(lk, [&] { return job->done.load() == job->n; });
-      cur_.reset();
+      active_.erase(std::remove(active_.begin(), active_.end(), job),
+                    active_.end());
     }
   }
 
@@ -64,16 +68,16 @@ class ThreadPool {
   }
 
   void loop() {
-    std::shared_ptr<Job> last;
+    uint64_t seen = 0;
     while (true) {
-      std::shared_ptr<Job> j;
+      std::vector<std::shared_ptr<Job>> jobs;
       {
         std::unique_lock<std::mutex> lk(mu_);
-        cv_.wait(lk, [&] { return cur_ && cur_ != last; });
-        j = cur_;
+        cv_.wait(lk, [&] { return gen_ != seen; });
+        seen = gen_;
+        jobs = active_;
       }
-      last = j;
-      run_job(*j);
+      for (auto& j : jobs) run_job(*j);
     }
   }
 
@@ -90,10 +94,10 @@ class ThreadPool {
   }
 
   std::vector<std::thread> threads_;
-  std::mutex job_mu_;
   std::mutex mu_;
   std::condition_variable cv_, done_cv_;
-  std::shared_ptr<Job> cur_;
+  std::vector<std::shared_ptr<Job>> active_;
+  uint64_t gen_ = 0;
 };
 
 }  // namespace lakesoul

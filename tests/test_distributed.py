@@ -21,14 +21,15 @@ def _free_port():
     return p
 
 
-def _init(rank, world, port):
-    os.environ["MASTER_ADDR"] = "127.0.0.1"
-    os.environ["MASTER_PORT"] = str(port)
+def _init(rank, world, rdv_file):
     os.environ["RANK"] = str(rank)
     os.environ["WORLD_SIZE"] = str(world)
     import torch.distributed as dist
 
-    dist.init_process_group("gloo", rank=rank, world_size=world)
+    # file rendezvous: immune to TCP port reuse races across test runs
+    dist.init_process_group(
+        "gloo", rank=rank, world_size=world, init_method=f"file://{rdv_file}"
+    )
     return dist
 
 
@@ -111,7 +112,7 @@ def _run_table_shard(rank, world, port, tmpdir, results):
 @pytest.mark.parametrize("fn", [_run_exchange, _run_table_shard])
 def test_multiprocess_gloo(fn, tmp_path):
     world = 2
-    port = _free_port()
+    port = str(tmp_path / "rdv")
     ctx = mp.get_context("spawn")
     with ctx.Manager() as mgr:
         results = mgr.dict()
