@@ -430,6 +430,33 @@ static py::dict read_unit_raw_py(const std::vector<std::string>& paths,
   py::list frows;
   for (auto r : ud.file_rows) frows.append(r);
   d["file_rows"] = frows;
+  // descriptor tensor for the C++ scan driver: [nuc, 17]
+  {
+    auto dt = torch::empty({(int64_t)ud.cols.size(), 17}, torch::kInt64);
+    auto da = dt.accessor<int64_t, 2>();
+    for (size_t i = 0; i < ud.cols.size(); i++) {
+      const UnitColumn& c = ud.cols[i];
+      int es = c.present && !c.is_string ? physical_elem_size(c.physical) : 0;
+      da[i][0] = c.present;
+      da[i][1] = c.is_string;
+      da[i][2] = c.is_dict;
+      da[i][3] = es;
+      da[i][4] = c.num_values;
+      da[i][5] = c.null_count;
+      da[i][6] = c.val_off;
+      da[i][7] = c.val_len;
+      da[i][8] = c.validity_off;
+      da[i][9] = c.dict_off;
+      da[i][10] = c.dict_len;
+      da[i][11] = c.run_off;
+      da[i][12] = c.run_cnt;
+      da[i][13] = c.dense_n;
+      da[i][14] = c.soff_off;
+      da[i][15] = c.sbytes_off;
+      da[i][16] = c.sbytes_len;
+    }
+    d["desc"] = dt;
+  }
   py::list cols;
   for (auto& c : ud.cols) {
     py::dict cd;

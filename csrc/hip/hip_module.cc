@@ -324,6 +324,258 @@ static std::vector<torch::Tensor> snappy_decompress(torch::Tensor src,
   return {dst, status};
 }
 
+// ---------------------------------------------------------------------- //
+// C++ scan-unit driver (simple path): decode + UseLast merge + gather for
+// one unit entirely from C++ — the python per-unit op storm (GIL-bound)
+// was the scan's critical path. Covers: integer single-PK (int64/int32),
+// every-column UseLast, no CDC. Python falls back otherwise.
+//
+// desc: int64 [nuc,17] from _cpp.read_unit_raw (see module.cc):
+// {present,is_string,is_dict,esize,num_values,null_count,val_off,val_len,
+//  validity_off,dict_off,dict_len,run_off,run_cnt,dense_n,soff_off,
+//  sbytes_off,sbytes_len}
+// Returns: [n_out scalar?] -> list: for each read col:
+//   fixed: [data_u8_gathered, validity_or_empty]
+//   string: [offsets_i64, bytes_u8, validity_or_empty]
+// plus the merged row count implicit in tensor sizes.
+static py::list scan_unit_uselast(torch::Tensor vals, torch::Tensor validity_buf,
+                                  torch::Tensor dicts, torch::Tensor runs,
+                                  torch::Tensor soffs, torch::Tensor desc,
+                                  int64_t nfiles, int64_t ncols, int64_t pk_ci,
+                                  int64_t pk_es) {
+  CHECK_GPU(vals);
+  auto device = vals.device();
+  auto u8 = torch::TensorOptions().dtype(torch::kUInt8).device(device);
+  auto i64 = torch::TensorOptions().dtype(torch::kInt64).device(device);
+  auto da = desc.accessor<int64_t, 2>();
+  auto stream = cur_stream();
+
+  torch::Tensor empty_u8 = torch::empty({0}, u8);
+  torch::Tensor runs2 = runs.numel() ? runs.view({-1, 6}) : runs;
+
+  // ---- decode every (file, col) into dense device columns ----
+  // decoded[f][c] = {data_u8 (es per elem), validity_u8 (or undef),
+  //                  offsets_i64/bytes_u8 for strings}
+  struct Col {
+    torch::Tensor data, validity, offsets, bytes;
+    bool is_string = false;
+  };
+  std::vector<std::vector<Col>> cols((size_t)nfiles,
+                                     std::vector<Col>((size_t)ncols));
+  for (int64_t f = 0; f < nfiles; f++) {
+    for (int64_t c = 0; c < ncols; c++) {
+      int64_t u = f * ncols + c;
+      Col& out = cols[f][c];
+      TORCH_CHECK(da[u][0], "scan_unit_uselast: absent column (fallback)");
+      int64_t nv = da[u][4];
+      bool has_null = da[u][8] >= 0 && da[u][5] > 0;
+      torch::Tensor vmask;
+      if (has_null) vmask = validity_buf.narrow(0, da[u][8], nv);
+      if (da[u][1]) {  // string
+        out.is_string = true;
+        out.offsets = soffs.narrow(0, da[u][14], nv + 1);
+        out.bytes = vals.narrow(0, da[u][15], da[u][16]);
+        out.validity = vmask;
+        continue;
+      }
+      int64_t es = da[u][3];
+      if (da[u][2]) {  // dict
+        auto rr = runs2.narrow(0, da[u][11], da[u][12]);
+        auto payload = vals.narrow(0, da[u][6], da[u][7]);
+        auto idx = torch::empty({da[u][13]}, torch::TensorOptions().dtype(torch::kInt32).device(device));
+        launch_rle_expand(payload.data_ptr<uint8_t>(), rr.data_ptr<int64_t>(),
+                          rr.size(0), idx.data_ptr<int32_t>(), da[u][13], stream);
+        auto dv = dicts.narrow(0, da[u][9], da[u][10]);
+        auto out_d = torch::empty({nv * es}, u8);
+        const uint8_t* vm = has_null ? vmask.data_ptr<uint8_t>() : nullptr;
+        torch::Tensor pos;
+        const int64_t* posp = nullptr;
+        if (has_null) {
+          pos = at::cumsum(vmask, 0, torch::kInt64) - 1;
+          posp = pos.data_ptr<int64_t>();
+        }
+        if (es == 4)
+          launch_dict_gather_scatter<uint32_t>((const uint32_t*)dv.data_ptr(), idx.data_ptr<int32_t>(), vm, posp, (uint32_t*)out_d.data_ptr(), nv, stream);
+        else if (es == 8)
+          launch_dict_gather_scatter<uint64_t>((const uint64_t*)dv.data_ptr(), idx.data_ptr<int32_t>(), vm, posp, (uint64_t*)out_d.data_ptr(), nv, stream);
+        else
+          launch_dict_gather_scatter<uint8_t>(dv.data_ptr<uint8_t>(), idx.data_ptr<int32_t>(), vm, posp, out_d.data_ptr<uint8_t>(), nv, stream);
+        out.data = out_d;
+      } else {
+        auto dense = vals.narrow(0, da[u][6], da[u][7]);
+        if (has_null) {
+          auto pos = at::cumsum(vmask, 0, torch::kInt64) - 1;
+          auto out_d = torch::zeros({nv * es}, u8);
+          if (es == 1)
+            launch_scatter_valid<uint8_t>(dense.data_ptr<uint8_t>(), vmask.data_ptr<uint8_t>(), pos.data_ptr<int64_t>(), out_d.data_ptr<uint8_t>(), nv, stream);
+          else if (es == 2)
+            launch_scatter_valid<uint16_t>((const uint16_t*)dense.data_ptr(), vmask.data_ptr<uint8_t>(), pos.data_ptr<int64_t>(), (uint16_t*)out_d.data_ptr(), nv, stream);
+          else if (es == 4)
+            launch_scatter_valid<uint32_t>((const uint32_t*)dense.data_ptr(), vmask.data_ptr<uint8_t>(), pos.data_ptr<int64_t>(), (uint32_t*)out_d.data_ptr(), nv, stream);
+          else
+            launch_scatter_valid<uint64_t>((const uint64_t*)dense.data_ptr(), vmask.data_ptr<uint8_t>(), pos.data_ptr<int64_t>(), (uint64_t*)out_d.data_ptr(), nv, stream);
+          out.data = out_d;
+        } else {
+          out.data = dense;
+        }
+      }
+      out.validity = vmask;
+    }
+  }
+
+  // ---- pack keys per file + pairwise merge-path merges ----
+  std::vector<std::pair<torch::Tensor, torch::Tensor>> streams;
+  int64_t base = 0;
+  for (int64_t f = 0; f < nfiles; f++) {
+    int64_t nv = da[f * ncols + pk_ci][4];
+    auto& pkc = cols[f][pk_ci];
+    torch::Tensor key64;
+    if (pk_es == 8) {
+      key64 = pkc.data.view(torch::kInt64);
+    } else {
+      key64 = pkc.data.view(torch::kInt32).to(torch::kInt64);
+    }
+    auto keys = torch::empty({nv}, i64);
+    launch_pack_key_i64(key64.data_ptr<int64_t>(), (uint64_t*)keys.data_ptr(), nv, stream);
+    auto vals_idx = torch::arange(base, base + nv, i64);
+    streams.emplace_back(keys, vals_idx);
+    base += nv;
+  }
+  while (streams.size() > 1) {
+    std::vector<std::pair<torch::Tensor, torch::Tensor>> nxt;
+    for (size_t j = 0; j + 1 < streams.size(); j += 2) {
+      auto& A = streams[j];
+      auto& B = streams[j + 1];
+      int64_t nA = A.first.numel(), nB = B.first.numel();
+      auto kO = torch::empty({nA + nB}, i64);
+      auto vO = torch::empty({nA + nB}, i64);
+      launch_merge_pairs((const uint64_t*)A.first.data_ptr(), (const uint64_t*)A.second.data_ptr(), nA,
+                         (const uint64_t*)B.first.data_ptr(), (const uint64_t*)B.second.data_ptr(), nB,
+                         (uint64_t*)kO.data_ptr(), (uint64_t*)vO.data_ptr(), stream);
+      nxt.emplace_back(kO, vO);
+    }
+    if (streams.size() % 2) nxt.push_back(streams.back());
+    streams = std::move(nxt);
+  }
+  torch::Tensor keys = streams[0].first;
+  torch::Tensor order = streams[0].second;
+  int64_t n = keys.numel();
+
+  // ---- dedup keep-last + survivor source indices ----
+  auto keep = torch::empty({n}, u8);
+  launch_keep_last((const uint64_t*)keys.data_ptr(), keep.data_ptr<uint8_t>(), n, stream);
+  auto surv = at::nonzero(keep).view(-1);  // the unit's single device sync
+  auto src_idx = order.index({surv});
+
+  // ---- gather every column (concat across files first) ----
+  py::list out_list;
+  std::vector<torch::Tensor> g_in;
+  std::vector<int64_t> g_col;  // read col index per g_in entry (fixed/validity)
+  std::vector<int> g_kind;     // 0=data, 1=validity
+  for (int64_t c = 0; c < ncols; c++) {
+    bool is_str = cols[0][c].is_string;
+    bool any_valid = false;
+    for (int64_t f = 0; f < nfiles; f++)
+      if (cols[f][c].validity.defined()) any_valid = true;
+    if (!is_str) {
+      std::vector<torch::Tensor> parts;
+      int64_t es = da[c][3];
+      for (int64_t f = 0; f < nfiles; f++)
+        parts.push_back(cols[f][c].data.view({-1, es}));
+      auto cat = nfiles == 1 ? parts[0] : at::cat(parts, 0);
+      // gather as typed elements (view to es-wide rows then index)
+      g_in.push_back(cat);
+      g_col.push_back(c);
+      g_kind.push_back(0);
+    }
+    if (any_valid || is_str) {
+      std::vector<torch::Tensor> parts;
+      for (int64_t f = 0; f < nfiles; f++) {
+        auto v = cols[f][c].validity;
+        if (v.defined()) parts.push_back(v);
+        else parts.push_back(torch::ones({da[f * ncols + c][4]}, u8));
+      }
+      if (any_valid) {
+        g_in.push_back(nfiles == 1 ? parts[0] : at::cat(parts, 0));
+        g_col.push_back(c);
+        g_kind.push_back(1);
+      }
+    }
+  }
+  // fused fixed-width gathers (16 cols per launch)
+  std::vector<torch::Tensor> g_out(g_in.size());
+  {
+    size_t i = 0;
+    int64_t nsurv = src_idx.numel();
+    while (i < g_in.size()) {
+      GatherTable tbl;
+      tbl.ncols = 0;
+      size_t start = i;
+      for (; i < g_in.size() && tbl.ncols < 16; i++) {
+        auto& t = g_in[i];
+        int64_t es = t.dim() == 2 ? t.size(1) * t.element_size() : t.element_size();
+        auto o = torch::empty({nsurv, t.dim() == 2 ? t.size(1) : 1},
+                              t.options());
+        tbl.src[tbl.ncols] = t.data_ptr();
+        tbl.dst[tbl.ncols] = o.data_ptr();
+        tbl.esize[tbl.ncols] = (int)es;
+        tbl.ncols++;
+        g_out[i] = o;
+      }
+      launch_gather_fixed_multi(tbl, src_idx.data_ptr<int64_t>(), nsurv, stream);
+      (void)start;
+    }
+  }
+  // build outputs per column
+  std::vector<torch::Tensor> col_data((size_t)ncols), col_valid((size_t)ncols);
+  for (size_t i = 0; i < g_in.size(); i++) {
+    if (g_kind[i] == 0) col_data[(size_t)g_col[i]] = g_out[i].view(-1);
+    else col_valid[(size_t)g_col[i]] = g_out[i].view(-1);
+  }
+  for (int64_t c = 0; c < ncols; c++) {
+    py::list entry;
+    if (!cols[0][c].is_string) {
+      entry.append(col_data[(size_t)c]);
+      if (col_valid[(size_t)c].defined()) entry.append(col_valid[(size_t)c]);
+      else entry.append(py::none());
+    } else {
+      // string gather: offsets via cumsum of gathered lens, bytes kernel
+      std::vector<torch::Tensor> offs_parts, bytes_parts;
+      int64_t byte_base = 0;
+      std::vector<torch::Tensor> adj_offs;
+      for (int64_t f = 0; f < nfiles; f++) {
+        auto o = cols[f][c].offsets;
+        adj_offs.push_back((f == 0 ? o : o + byte_base));
+        bytes_parts.push_back(cols[f][c].bytes);
+        byte_base += cols[f][c].bytes.numel();
+      }
+      // concatenated offsets: drop leading 0 of subsequent files
+      std::vector<torch::Tensor> oparts;
+      for (int64_t f = 0; f < nfiles; f++)
+        oparts.push_back(f == 0 ? adj_offs[f] : adj_offs[f].narrow(0, 1, adj_offs[f].numel() - 1));
+      auto cat_offs = nfiles == 1 ? oparts[0] : at::cat(oparts, 0);
+      auto cat_bytes = nfiles == 1 ? bytes_parts[0] : at::cat(bytes_parts, 0);
+      auto lens = cat_offs.narrow(0, 1, cat_offs.numel() - 1) -
+                  cat_offs.narrow(0, 0, cat_offs.numel() - 1);
+      auto sel_lens = lens.index({src_idx});
+      int64_t nsurv = src_idx.numel();
+      auto new_offs = torch::zeros({nsurv + 1}, i64);
+      new_offs.narrow(0, 1, nsurv).copy_(at::cumsum(sel_lens, 0));
+      int64_t total = nsurv ? new_offs[nsurv].item<int64_t>() : 0;
+      auto new_bytes = torch::empty({total}, u8);
+      launch_gather_strings(cat_bytes.data_ptr<uint8_t>(), cat_offs.data_ptr<int64_t>(),
+                            src_idx.data_ptr<int64_t>(), new_offs.data_ptr<int64_t>(),
+                            new_bytes.data_ptr<uint8_t>(), nsurv, stream);
+      entry.append(new_offs);
+      entry.append(new_bytes);
+      if (col_valid[(size_t)c].defined()) entry.append(col_valid[(size_t)c]);
+      else entry.append(py::none());
+    }
+    out_list.append(entry);
+  }
+  return out_list;
+}
+
 // decompress into an existing device buffer (jobs dst offsets index it)
 static torch::Tensor snappy_decompress_into(torch::Tensor src, torch::Tensor jobs,
                                             torch::Tensor dst) {
@@ -341,6 +593,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ann_scores", &ann_scores);
   m.def("snappy_decompress", &snappy_decompress);
   m.def("snappy_decompress_into", &snappy_decompress_into);
+  m.def("scan_unit_uselast", &scan_unit_uselast);
   m.doc() = "lakesoul_amd gfx950 HIP kernels";
   m.def("hash_fixed_column", &hash_fixed_column);
   m.def("hash_string_column", &hash_string_column);

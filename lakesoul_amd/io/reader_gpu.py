@@ -130,6 +130,57 @@ def read_unit_gpu(scan, unit, raw: Optional[dict] = None,
     empty_u8 = torch.empty(0, dtype=torch.uint8, device=device)
     empty_i64 = torch.empty(0, dtype=torch.int64, device=device)
 
+    # ---- C++ fast path: decode + UseLast merge + gather in ONE call ----
+    # (the per-unit python op storm was the scan's critical path)
+    nfiles = len(raw["file_rows"])
+    desc = raw.get("desc")
+    if (
+        desc is not None
+        and nfiles > 1
+        and len(scan.pk) == 1
+        and not scan.merge_ops
+        and scan.cdc_column is None
+        and scan.pk[0] in names
+        and scan.schema.field(scan.pk[0]).dtype in ("int64", "int32")
+        and bool(desc[:, 0].all())
+    ):
+        with timing.phase("gpu_unit_cpp", sync_gpu=False):
+            pk_ci = names.index(scan.pk[0])
+            pk_es = 8 if scan.schema.field(scan.pk[0]).dtype == "int64" else 4
+            outs = hip().scan_unit_uselast(
+                vals,
+                validity_buf if validity_buf is not None else empty_u8,
+                dicts_buf if dicts_buf is not None else empty_u8,
+                runs_buf if runs_buf is not None else empty_i64,
+                soffs_buf if soffs_buf is not None else empty_i64,
+                desc,
+                nfiles,
+                ncols,
+                pk_ci,
+                pk_es,
+            )
+        cols: Dict[str, Column] = {}
+        for ci, name in enumerate(names):
+            f = scan.schema.field(name)
+            entry = outs[ci]
+            if f.dtype in ("string", "binary"):
+                offs, by, vmask = entry[0], entry[1], entry[2]
+                cols[name] = Column(f.dtype, offsets=offs, bytes_=by, validity=vmask)
+            else:
+                data, vmask = entry[0], entry[1]
+                data = data.view(_TORCH_VIEW[f.dtype])
+                if f.dtype in _TARGET:
+                    data = data.to(_TARGET[f.dtype])
+                cols[name] = Column(f.dtype, data=data, validity=vmask)
+        merged = Batch(read_schema, cols)
+        out_cols: Dict[str, Column] = {}
+        for f in scan.eval_schema:
+            if f.name in scan.range_cols:
+                out_cols[f.name] = _range_col_gpu(scan, f, unit, merged.num_rows, device)
+            else:
+                out_cols[f.name] = merged.columns[f.name]
+        return Batch(scan.eval_schema, out_cols)
+
     file_batches: List[Batch] = []
     present: List[set] = []
     _dec = timing.phase("gpu_decode", sync_gpu=True)
