@@ -70,11 +70,24 @@ class ThreadPool {
   void loop() {
     uint64_t seen = 0;
     while (true) {
+      // spin briefly before sleeping: scan tasks are ~0.5 ms, and the
+      // cv wake latency of 60+ sleeping threads would otherwise leave
+      // most of the pool asleep while a few threads drain the job
+      bool have = false;
+      for (int spin = 0; spin < 20000; spin++) {
+        if (gen_.load(std::memory_order_acquire) != seen) {
+          have = true;
+          break;
+        }
+#if defined(__x86_64__)
+        __builtin_ia32_pause();
+#endif
+      }
       std::vector<std::shared_ptr<Job>> jobs;
       {
         std::unique_lock<std::mutex> lk(mu_);
-        cv_.wait(lk, [&] { return gen_ != seen; });
-        seen = gen_;
+        if (!have) cv_.wait(lk, [&] { return gen_.load() != seen; });
+        seen = gen_.load();
         jobs = active_;
       }
       for (auto& j : jobs) run_job(*j);
@@ -97,7 +110,7 @@ class ThreadPool {
   std::mutex mu_;
   std::condition_variable cv_, done_cv_;
   std::vector<std::shared_ptr<Job>> active_;
-  uint64_t gen_ = 0;
+  std::atomic<uint64_t> gen_{0};
 };
 
 }  // namespace lakesoul
