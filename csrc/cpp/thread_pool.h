@@ -11,6 +11,8 @@
 
 #include <algorithm>
 #include <atomic>
+#include <cstdio>
+#include <cstdlib>
 #include <condition_variable>
 #include <functional>
 #include <memory>
@@ -60,8 +62,33 @@ class ThreadPool {
     int64_t n = 0;
   };
 
+  static int effective_cpus() {
+    // the GPU boxes report 256 cores but the job cgroup may cap the
+    // quota far lower — oversubscribing the pool thrashes. cgroup v2:
+    // /sys/fs/cgroup/cpu.max = "<quota> <period>" or "max <period>".
+    int hw = (int)std::thread::hardware_concurrency();
+    if (const char* env = std::getenv("LAKESOUL_POOL_THREADS")) {
+      int v = atoi(env);
+      if (v > 0) return v;
+    }
+    FILE* f = std::fopen("/sys/fs/cgroup/cpu.max", "r");
+    if (f) {
+      char buf[64] = {0};
+      if (std::fgets(buf, sizeof(buf), f)) {
+        long quota, period;
+        if (std::sscanf(buf, "%ld %ld", &quota, &period) == 2 && quota > 0 &&
+            period > 0) {
+          int q = (int)(quota / period);
+          if (q > 0 && q < hw) hw = q;
+        }
+      }
+      std::fclose(f);
+    }
+    return hw;
+  }
+
   ThreadPool() {
-    int n = (int)std::thread::hardware_concurrency();
+    int n = effective_cpus();
     if (n > 64) n = 64;
     if (n < 2) n = 2;
     for (int i = 0; i < n - 1; i++) threads_.emplace_back([this] { loop(); });
@@ -70,23 +97,10 @@ class ThreadPool {
   void loop() {
     uint64_t seen = 0;
     while (true) {
-      // spin briefly before sleeping: scan tasks are ~0.5 ms, and the
-      // cv wake latency of 60+ sleeping threads would otherwise leave
-      // most of the pool asleep while a few threads drain the job
-      bool have = false;
-      for (int spin = 0; spin < 20000; spin++) {
-        if (gen_.load(std::memory_order_acquire) != seen) {
-          have = true;
-          break;
-        }
-#if defined(__x86_64__)
-        __builtin_ia32_pause();
-#endif
-      }
       std::vector<std::shared_ptr<Job>> jobs;
       {
         std::unique_lock<std::mutex> lk(mu_);
-        if (!have) cv_.wait(lk, [&] { return gen_.load() != seen; });
+        cv_.wait(lk, [&] { return gen_.load() != seen; });
         seen = gen_.load();
         jobs = active_;
       }
