@@ -368,6 +368,14 @@ class ParquetFile {
     int64_t out_len;
   };
 
+  struct DeferPage {  // host direct-decompress job (zstd/uncompressed,
+    int64_t file_off;  // levels-free PLAIN pages): source bytes stay in
+    int64_t comp_len;  // the mmap until fill() decompresses straight into
+    int64_t out_off;   // the caller's pinned buffer
+    int64_t out_len;
+    int32_t codec;
+  };
+
   struct ChunkData {
     int32_t physical = 0;
     int64_t num_values = 0;  // rows in chunk
@@ -385,12 +393,19 @@ class ParquetFile {
     int64_t values_len = 0;
     std::vector<uint8_t> comp;
     std::vector<CompPage> comp_pages;
+    // host_deferred mode: values stays EMPTY; defer_pages decompress
+    // later directly into the final buffer (no staging copy)
+    bool host_deferred = false;
+    std::vector<DeferPage> defer_pages;
   };
+
+  const uint8_t* data_at(int64_t off) const { return map_ + off; }
 
   // gpu_snappy: defer SNAPPY page decompression to the GPU kernel when
   // the column is REQUIRED (no def-level section inside the compressed
   // blob), PLAIN-encoded, non-boolean v1 pages.
-  ChunkData read_chunk(size_t rg, size_t col, bool gpu_snappy = false) const {
+  ChunkData read_chunk(size_t rg, size_t col, bool gpu_snappy = false,
+                       bool defer_host = false) const {
     const RowGroup& g = meta_.row_groups.at(rg);
     const ColumnMeta& cm = g.columns.at(col);
     const ColumnDesc& cd = cols_.at(col);
@@ -438,6 +453,30 @@ class ParquetFile {
             ph.encoding == ENC_PLAIN)) {
         // mixed chunk (e.g. dict fallback) — redo fully on host
         return read_chunk(rg, col, false);
+      }
+      if (defer_host &&
+          (cm.codec == CODEC_ZSTD || cm.codec == CODEC_UNCOMPRESSED) &&
+          ph.type == PAGE_DATA && ph.encoding == ENC_PLAIN && !cd.nullable &&
+          cd.physical != PT_BOOLEAN && cd.physical != PT_BYTE_ARRAY &&
+          out.dict.empty() && out.values.empty() && !out.gpu_compressed) {
+        // levels-free PLAIN page: defer decompression — fill() writes it
+        // straight into the destination buffer
+        out.host_deferred = true;
+        DeferPage dp;
+        dp.file_off = (int64_t)(body - map_);
+        dp.comp_len = ph.compressed_size;
+        dp.out_off = out.values_len;
+        dp.out_len = ph.uncompressed_size;
+        dp.codec = cm.codec;
+        out.defer_pages.push_back(dp);
+        out.values_len += ph.uncompressed_size;
+        values_seen += ph.num_values;
+        continue;
+      }
+      if (out.host_deferred &&
+          !((cm.codec == CODEC_ZSTD || cm.codec == CODEC_UNCOMPRESSED) &&
+            ph.type == PAGE_DATA && ph.encoding == ENC_PLAIN)) {
+        return read_chunk(rg, col, false, false);  // mixed: redo staged
       }
       if (gpu_snappy && cm.codec == CODEC_SNAPPY && ph.type == PAGE_DATA &&
           ph.encoding == ENC_PLAIN && !cd.nullable &&

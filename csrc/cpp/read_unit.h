@@ -126,6 +126,12 @@ struct UnitStage {
   std::vector<int64_t> file_rows;
   std::vector<int64_t> runs;  // [m][6]
   std::vector<int64_t> snappy_jobs;  // [m][4]: comp_off, comp_len, dst_off(values), dst_len
+  struct HostJob {
+    int file_idx;
+    int64_t file_off, comp_len, dst_off, dst_len;
+    int32_t codec;
+  };
+  std::vector<HostJob> host_jobs;
   std::vector<std::unique_ptr<StrDecoded>> str_cols;
   int64_t values_size = 0, validity_size = 0, dicts_size = 0, soffs_size = 0,
           comp_size = 0;
@@ -174,7 +180,8 @@ inline std::unique_ptr<UnitStage> read_unit_stage1(
       try {
         const Task& t = tasks[i];
         auto& fd = S.files[t.fi];
-        fd.chunks[t.c][t.rg] = fd.f->read_chunk(t.rg, fd.col_idx[t.c], gpu_snappy);
+        fd.chunks[t.c][t.rg] =
+            fd.f->read_chunk(t.rg, fd.col_idx[t.c], gpu_snappy, true);
       } catch (std::exception& e) {
         std::lock_guard<std::mutex> lk(err_mu);
         err = e.what();
@@ -273,6 +280,13 @@ inline std::unique_ptr<UnitStage> read_unit_stage1(
               S.snappy_jobs.push_back(cp.out_len);
             }
             S.comp_size += (int64_t)ch.comp.size();
+            plen += ch.values_len;
+          } else if (ch.host_deferred) {
+            for (auto& dp : ch.defer_pages) {
+              S.host_jobs.push_back(
+                  {(int)fi, dp.file_off, dp.comp_len, vpos + plen + dp.out_off,
+                   dp.out_len, dp.codec});
+            }
             plen += ch.values_len;
           } else {
             plen += (int64_t)ch.values.size();
@@ -383,13 +397,21 @@ inline void read_unit_fill(UnitStage& S, uint8_t* values, uint8_t* validity,
     }
     int64_t off = uc.val_off;
     for (auto& ch : chs) {
-      if (ch.gpu_compressed) {
-        off += ch.values_len;  // filled by the GPU snappy kernel
+      if (ch.gpu_compressed || ch.host_deferred) {
+        off += ch.values_len;  // GPU snappy kernel / host job fills it
         continue;
       }
       std::memcpy(values + off, ch.values.data(), ch.values.size());
       off += (int64_t)ch.values.size();
     }
+  });
+  // deferred pages: decompress straight from the file mmap into the
+  // destination buffer (one pass, zero staging)
+  ThreadPool::instance().parallel_for((int64_t)S.host_jobs.size(), [&](int64_t i) {
+    const UnitStage::HostJob& hj = S.host_jobs[i];
+    const uint8_t* src = S.files[hj.file_idx].f->data_at(hj.file_off);
+    decompress_into(hj.codec, values + hj.dst_off, (size_t)hj.dst_len, src,
+                    (size_t)hj.comp_len);
   });
 }
 
