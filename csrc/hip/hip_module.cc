@@ -344,6 +344,7 @@ static py::list scan_unit_uselast(torch::Tensor vals, torch::Tensor validity_buf
                                   int64_t nfiles, int64_t ncols, int64_t pk_ci,
                                   int64_t pk_es) {
   CHECK_GPU(vals);
+  py::gil_scoped_release rel;  // long device-side section (incl. one sync)
   auto device = vals.device();
   auto u8 = torch::TensorOptions().dtype(torch::kUInt8).device(device);
   auto i64 = torch::TensorOptions().dtype(torch::kInt64).device(device);
@@ -468,7 +469,6 @@ static py::list scan_unit_uselast(torch::Tensor vals, torch::Tensor validity_buf
   auto src_idx = order.index({surv});
 
   // ---- gather every column (concat across files first) ----
-  py::list out_list;
   std::vector<torch::Tensor> g_in;
   std::vector<int64_t> g_col;  // read col index per g_in entry (fixed/validity)
   std::vector<int> g_kind;     // 0=data, 1=validity
@@ -528,16 +528,16 @@ static py::list scan_unit_uselast(torch::Tensor vals, torch::Tensor validity_buf
   }
   // build outputs per column
   std::vector<torch::Tensor> col_data((size_t)ncols), col_valid((size_t)ncols);
+  std::vector<std::vector<torch::Tensor>> out_cols;
   for (size_t i = 0; i < g_in.size(); i++) {
     if (g_kind[i] == 0) col_data[(size_t)g_col[i]] = g_out[i].view(-1);
     else col_valid[(size_t)g_col[i]] = g_out[i].view(-1);
   }
   for (int64_t c = 0; c < ncols; c++) {
-    py::list entry;
+    std::vector<torch::Tensor> entry;
     if (!cols[0][c].is_string) {
-      entry.append(col_data[(size_t)c]);
-      if (col_valid[(size_t)c].defined()) entry.append(col_valid[(size_t)c]);
-      else entry.append(py::none());
+      entry.push_back(col_data[(size_t)c]);
+      entry.push_back(col_valid[(size_t)c]);  // may be undefined
     } else {
       // string gather: offsets via cumsum of gathered lens, bytes kernel
       std::vector<torch::Tensor> offs_parts, bytes_parts;
@@ -566,12 +566,21 @@ static py::list scan_unit_uselast(torch::Tensor vals, torch::Tensor validity_buf
       launch_gather_strings(cat_bytes.data_ptr<uint8_t>(), cat_offs.data_ptr<int64_t>(),
                             src_idx.data_ptr<int64_t>(), new_offs.data_ptr<int64_t>(),
                             new_bytes.data_ptr<uint8_t>(), nsurv, stream);
-      entry.append(new_offs);
-      entry.append(new_bytes);
-      if (col_valid[(size_t)c].defined()) entry.append(col_valid[(size_t)c]);
-      else entry.append(py::none());
+      entry.push_back(new_offs);
+      entry.push_back(new_bytes);
+      entry.push_back(col_valid[(size_t)c]);  // may be undefined
     }
-    out_list.append(entry);
+    out_cols.push_back(std::move(entry));
+  }
+  py::gil_scoped_acquire acq;
+  py::list out_list;
+  for (auto& entry : out_cols) {
+    py::list e;
+    for (auto& t : entry) {
+      if (t.defined()) e.append(t);
+      else e.append(py::none());
+    }
+    out_list.append(e);
   }
   return out_list;
 }
