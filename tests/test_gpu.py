@@ -320,3 +320,44 @@ def test_gpu_string_pk_hybrid(dev, tmp_path):
     assert len(df) == 2001
     got = dict(zip(df["key"], df["v"]))
     assert got["customer_00005"] == -5 and got["zzz"] == 1
+
+
+def test_gpu_two_int32_pk_merge(dev, tmp_path):
+    """Two-int32 composite PK uses the pack_key_2xi32 merge path."""
+    catalog = _mk_catalog(tmp_path)
+    from lakesoul_amd.io.schema import Field, Schema
+
+    t = catalog.create_table(
+        "g2pk",
+        Schema([Field("a", "int32", False), Field("b", "int32", False), Field("v", "float64")]),
+        primary_keys=["a", "b"],
+        hash_bucket_num=2,
+    )
+    rng = np.random.default_rng(7)
+    a = rng.integers(-1000, 1000, 5000, dtype=np.int32)
+    b = rng.integers(-1000, 1000, 5000, dtype=np.int32)
+    t.upsert({"a": a, "b": b, "v": np.zeros(5000)})
+    t.upsert({"a": a[:500], "b": b[:500], "v": np.ones(500)})
+    import pandas as pd
+
+    cpu = t.scan(device="cpu").to_arrow().to_pandas().sort_values(["a", "b"]).reset_index(drop=True)
+    gpu = t.scan(device="cuda").to_arrow().to_pandas().sort_values(["a", "b"]).reset_index(drop=True)
+    pd.testing.assert_frame_equal(cpu, gpu)
+
+
+def test_gpu_int32_single_pk(dev, tmp_path):
+    catalog = _mk_catalog(tmp_path)
+    from lakesoul_amd.io.schema import Field, Schema
+
+    t = catalog.create_table(
+        "gi32",
+        Schema([Field("id", "int32", False), Field("v", "float64")]),
+        primary_keys=["id"],
+        hash_bucket_num=2,
+    )
+    n = 30000
+    t.upsert({"id": np.arange(n, dtype=np.int32), "v": np.zeros(n)})
+    t.upsert({"id": np.arange(0, n, 7, dtype=np.int32), "v": np.ones(len(range(0, n, 7)))})
+    df = t.scan(device="cuda").to_arrow().to_pandas().sort_values("id").reset_index(drop=True)
+    assert len(df) == n
+    assert (df["v"][::7] == 1.0).all() and (df["v"][1::7] == 0.0).all()
