@@ -262,17 +262,24 @@ class Batch:
                     mask = ~(c.validity.cpu().numpy().astype(bool))
                 arrays.append(pa.array(np_arr, type=dtype_to_arrow(f.dtype), from_pandas=False, mask=mask))
             else:
-                offs = c.offsets.cpu().numpy()
-                bys = c.bytes_.cpu().numpy().tobytes()
-                items = []
-                val = c.validity.cpu().numpy() if c.validity is not None else None
-                for i in range(len(c)):
-                    if val is not None and not val[i]:
-                        items.append(None)
-                    else:
-                        raw = bys[offs[i]:offs[i + 1]]
-                        items.append(raw.decode() if f.dtype == "string" else raw)
-                arrays.append(pa.array(items, type=dtype_to_arrow(f.dtype)))
+                # zero-copy-ish buffer construction (offsets may be i64
+                # from the GPU path; arrow string/binary wants i32)
+                offs_np = c.offsets.cpu().numpy().astype(np.int32, copy=False)
+                bys_np = c.bytes_.cpu().numpy()
+                n = len(c)
+                validity_buf = None
+                null_count = 0
+                if c.validity is not None:
+                    v = c.validity.cpu().numpy().astype(bool)
+                    null_count = int(n - v.sum())
+                    validity_buf = pa.py_buffer(np.packbits(v, bitorder="little").tobytes())
+                arr = pa.Array.from_buffers(
+                    dtype_to_arrow(f.dtype),
+                    n,
+                    [validity_buf, pa.py_buffer(offs_np.tobytes()), pa.py_buffer(bys_np.tobytes())],
+                    null_count=null_count,
+                )
+                arrays.append(arr)
         return pa.table(dict(zip(self.schema.names(), arrays)))
 
 
