@@ -38,9 +38,28 @@ class ShardInfo:
     path: str
 
 
+def kmeans(x: torch.Tensor, k: int, iters: int = 10, seed: int = 0) -> torch.Tensor:
+    """Plain Lloyd k-means on the active device (the reference trains its
+    IVF coarse quantizer the same way, rabitq/kmeans.rs). Returns
+    L2-normalized centroids (cosine geometry)."""
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    init = torch.randperm(x.shape[0], generator=g)[:k]
+    c = x[init.to(x.device)].clone()
+    for _ in range(iters):
+        scores = x @ c.T  # cosine (inputs normalized)
+        assign = scores.argmax(dim=1)
+        for j in range(k):
+            m = assign == j
+            if bool(m.any()):
+                c[j] = x[m].mean(dim=0)
+        c = c / c.norm(dim=1, keepdim=True).clamp_min(1e-30)
+    return c
+
+
 class VectorIndex:
     def __init__(self, root: str, column: str, dim: int, metric: str,
-                 shards: List[ShardInfo], pk_dtype: str, version: int):
+                 shards: List[ShardInfo], pk_dtype: str, version: int,
+                 ivf_clusters: int = 0):
         self.root = root
         self.column = column
         self.dim = dim
@@ -48,6 +67,7 @@ class VectorIndex:
         self.shards = shards
         self.pk_dtype = pk_dtype
         self.version = version
+        self.ivf_clusters = ivf_clusters
         self._gpu_cache: dict = {}
 
     # -- persistence ---------------------------------------------------- #
@@ -63,7 +83,8 @@ class VectorIndex:
             "pk_dtype": self.pk_dtype,
             "version": self.version,
             "created_ms": int(time.time() * 1000),
-            "engine": "mfma-exact-bf16",
+            "engine": "mfma-exact-bf16" if not self.ivf_clusters else "ivf+mfma-bf16",
+            "ivf_clusters": self.ivf_clusters,
             "shards": [
                 {"bucket_id": s.bucket_id, "num_rows": s.num_rows, "path": s.path}
                 for s in self.shards
@@ -78,26 +99,46 @@ class VectorIndex:
         with open(os.path.join(root, "manifest.json")) as f:
             m = json.load(f)
         shards = [ShardInfo(s["bucket_id"], s["num_rows"], s["path"]) for s in m["shards"]]
-        return cls(root, m["column"], m["dim"], m["metric"], shards, m["pk_dtype"], m["version"])
+        return cls(root, m["column"], m["dim"], m["metric"], shards,
+                   m["pk_dtype"], m["version"], m.get("ivf_clusters", 0))
 
     # -- shard data ----------------------------------------------------- #
 
-    def _load_shard(self, s: ShardInfo, device) -> Tuple[torch.Tensor, torch.Tensor]:
+    def _load_shard(self, s: ShardInfo, device):
         key = (s.path, str(device))
         if key in self._gpu_cache:
             return self._gpu_cache[key]
         raw = np.fromfile(s.path + ".vec", dtype=np.uint16).reshape(s.num_rows, self.dim)
         vecs = torch.from_numpy(raw.view(np.int16)).view(torch.bfloat16).to(device)
         ids = torch.from_numpy(np.fromfile(s.path + ".ids", dtype=np.int64)).to(device)
-        self._gpu_cache[key] = (vecs, ids)
-        return vecs, ids
+        clu = None
+        if self.ivf_clusters and os.path.exists(s.path + ".clu"):
+            clu = torch.from_numpy(np.fromfile(s.path + ".clu", dtype=np.int64))
+        entry = (vecs, ids, clu)
+        self._gpu_cache[key] = entry
+        return entry
+
+    def _centroids(self, device) -> Optional[torch.Tensor]:
+        if not self.ivf_clusters:
+            return None
+        key = ("__centroids__", str(device))
+        if key not in self._gpu_cache:
+            raw = np.fromfile(os.path.join(self.root, "centroids.vec"),
+                              dtype=np.float32).reshape(self.ivf_clusters, self.dim)
+            self._gpu_cache[key] = torch.from_numpy(raw).to(device)
+        return self._gpu_cache[key]
 
     # -- search --------------------------------------------------------- #
 
-    def search(self, queries, k: int = 10, device: Optional[str] = None):
+    def search(self, queries, k: int = 10, device: Optional[str] = None,
+               nprobe: Optional[int] = None):
         """Top-k over all shards. Returns (ids, scores) arrays of shape
         (nq, k). Cosine: inputs are normalized; score = cosine similarity.
-        L2: score = -||x-q||^2 (larger is better)."""
+        L2: score = -||x-q||^2 (larger is better).
+
+        With an IVF index, only the members of the ``nprobe`` nearest
+        coarse clusters per query are scored (default: clusters/8, min 4
+        — the reference's IVF SearchParams analog)."""
         if device is None:
             device = "cuda" if torch.cuda.is_available() else "cpu"
         q = torch.as_tensor(np.asarray(queries, dtype=np.float32))
@@ -110,8 +151,26 @@ class VectorIndex:
 
         best_scores = torch.full((nq, k), -float("inf"), device=device)
         best_ids = torch.full((nq, k), -1, dtype=torch.int64, device=device)
+        cents = self._centroids(device)
+        if cents is not None and nprobe is None:
+            nprobe = max(4, self.ivf_clusters // 8)
         for s in self.shards:
-            vecs, ids = self._load_shard(s, device)
+            vecs, ids, clu = self._load_shard(s, device)
+            if cents is not None and clu is not None:
+                # union of the nprobe nearest clusters over the query batch
+                cscores = q_dev.to(cents.dtype) @ cents.T  # (nq, k_c)
+                probe = torch.topk(cscores, min(nprobe, self.ivf_clusters), dim=1).indices
+                wanted = torch.unique(probe.flatten()).cpu()
+                # vectors are stored cluster-sorted: gather member ranges
+                segs = []
+                for cid in wanted.tolist():
+                    a, b = int(clu[cid]), int(clu[cid + 1])
+                    if b > a:
+                        segs.append((a, b))
+                if not segs:
+                    continue
+                vecs = torch.cat([vecs[a:b] for a, b in segs])
+                ids = torch.cat([ids[a:b] for a, b in segs])
             scores = self._scores(vecs, q_dev, device)  # (n, nq) f32
             kk = min(k, scores.shape[0])
             top = torch.topk(scores, kk, dim=0)  # (kk, nq)
@@ -155,6 +214,7 @@ def build_vector_index(
     pk: Optional[str] = None,
     metric: str = "cosine",
     device: Optional[str] = None,
+    ivf_clusters: int = 0,
 ) -> VectorIndex:
     """Build per-bucket exact-search shards for a fixed-size-list float
     column stored as ``dim`` float32/float64 scalar columns or via numpy
@@ -172,6 +232,7 @@ def build_vector_index(
     root = os.path.join(table.table_path, "_vector_index", column)
     os.makedirs(root, exist_ok=True)
     shards: List[ShardInfo] = []
+    _shard_payloads: list = []
     dim = None
     scan = table.scan(columns=[pk, column], device=device or "cpu")
     for unit in scan.plan():
@@ -190,15 +251,41 @@ def build_vector_index(
             norms = np.linalg.norm(vecs, axis=1, keepdims=True)
             norms[norms == 0] = 1
             vecs = vecs / norms
-        vbf = torch.from_numpy(vecs).to(torch.bfloat16).view(torch.int16).numpy().view(np.uint16)
         spath = os.path.join(root, f"shard_{unit.bucket_id:04d}")
-        vbf.tofile(spath + ".vec")
-        ids_t.cpu().numpy().astype(np.int64).tofile(spath + ".ids")
-        shards.append(ShardInfo(unit.bucket_id, n, spath))
+        _shard_payloads.append((spath, unit.bucket_id, vecs, ids_t.cpu().numpy().astype(np.int64)))
     if dim is None:
         raise ValueError("no data to index")
+
+    dev = device or ("cuda" if torch.cuda.is_available() else "cpu")
+    centroids = None
+    if ivf_clusters:
+        # coarse quantizer on a sample (GPU k-means; rabitq/kmeans.rs analog)
+        sample = np.concatenate([p[2] for p in _shard_payloads])
+        if len(sample) > 200_000:
+            sel = np.random.default_rng(0).choice(len(sample), 200_000, replace=False)
+            sample = sample[sel]
+        centroids = kmeans(torch.from_numpy(sample).to(dev), ivf_clusters).cpu()
+        centroids.numpy().astype(np.float32).tofile(os.path.join(root, "centroids.vec"))
+
+    for spath, bucket_id, vecs, ids_np in _shard_payloads:
+        n = len(vecs)
+        order = np.arange(n)
+        if centroids is not None:
+            assign = (torch.from_numpy(vecs).to(dev) @ centroids.to(dev).T).argmax(dim=1).cpu().numpy()
+            order = np.argsort(assign, kind="stable")
+            counts = np.bincount(assign, minlength=ivf_clusters)
+            clu = np.zeros(ivf_clusters + 1, dtype=np.int64)
+            np.cumsum(counts, out=clu[1:])
+            clu.tofile(spath + ".clu")
+        v_sorted = vecs[order]
+        vbf = torch.from_numpy(v_sorted).to(torch.bfloat16).view(torch.int16).numpy().view(np.uint16)
+        vbf.tofile(spath + ".vec")
+        ids_np[order].tofile(spath + ".ids")
+        shards.append(ShardInfo(bucket_id, n, spath))
+
     version = table.latest_version() or 0
-    idx = VectorIndex(root, column, int(dim), metric, shards, "int64", version)
+    idx = VectorIndex(root, column, int(dim), metric, shards, "int64", version,
+                      ivf_clusters=ivf_clusters)
     idx.save_manifest()
     return idx
 
