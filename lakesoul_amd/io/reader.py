@@ -338,9 +338,30 @@ class LakeSoulScan:
             self.schema.field(p).dtype not in ("string", "binary") for p in self.pk
         )
 
+    def _check_unit_size(self, unit: ScanUnit) -> None:
+        """Guard against buckets whose decoded size would not fit the
+        configured HBM budget (chunked spill merge is a round-2 item,
+        ROADMAP.md): fail with actionable advice instead of OOM-ing."""
+        limit = int(os.environ.get("LAKESOUL_MAX_UNIT_BYTES", str(64 * 1024**3)))
+        total = 0
+        for path in unit.files:
+            try:
+                total += os.path.getsize(path)
+            except OSError:
+                pass
+        # decompressed estimate: zstd(1) on typical columns ~2x
+        if total * 2 > limit:
+            raise MemoryError(
+                f"scan unit bucket={unit.bucket_id} estimated decoded size "
+                f"~{total * 2 / 1e9:.0f} GB exceeds LAKESOUL_MAX_UNIT_BYTES "
+                f"({limit / 1e9:.0f} GB). Recreate the table with more hash "
+                f"buckets, compact the partition, or raise the limit."
+            )
+
     def _read_unit(self, unit: ScanUnit) -> Optional[Batch]:
         if not unit.files:
             return None
+        self._check_unit_size(unit)
         cache_key = None
         if self.use_cache:
             from .hbm_cache import scan_cache
