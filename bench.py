@@ -48,6 +48,8 @@ def build_args():
     p.add_argument("--rows-upsert", type=int, default=2_000_000)
     p.add_argument("--upserts", type=int, default=10)
     p.add_argument("--buckets-per-gpu", type=int, default=16)
+    p.add_argument("--setup-chunk-rows", type=int, default=50_000_000,
+                   help="generate+write setup data in chunks to bound host RAM")
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--workdir", type=str, default=None)
     p.add_argument("--keep", action="store_true")
@@ -147,12 +149,18 @@ def main():
 
     t0 = time.time()
     upsert_s = 0.0
-    base_ids = filter_my(np.arange(rows_base * world, dtype=np.int64))
-    data = gen(base_ids)
-    tu = time.time()
-    table.upsert(data, device=device)
-    upsert_s += time.time() - tu
-    upsert_rows = len(base_ids)
+    upsert_rows = 0
+    chunk = max(1, args.setup_chunk_rows)
+    for lo in range(0, rows_base * world, chunk):
+        hi = min(lo + chunk, rows_base * world)
+        base_ids = filter_my(np.arange(lo, hi, dtype=np.int64))
+        if len(base_ids) == 0:
+            continue
+        data = gen(base_ids)
+        tu = time.time()
+        table.upsert(data, device=device)
+        upsert_s += time.time() - tu
+        upsert_rows += len(base_ids)
     for u in range(args.upserts):
         up_ids = filter_my(
             rng.choice(rows_base * world, args.rows_upsert * world, replace=False).astype(np.int64)

@@ -172,6 +172,22 @@ class LakeSoulScan:
                 )
             by_bucket: Dict[int, List[str]] = {}
             for op in file_ops:
+                # invalid-file tolerance (reference session.rs:440-450:
+                # listing drops files < 8 bytes with an error log)
+                from .fs import is_remote as _is_remote
+
+                if not _is_remote(op.path):
+                    try:
+                        if os.path.getsize(op.path) < 8:
+                            import warnings
+
+                            warnings.warn(f"skipping invalid file {op.path}")
+                            continue
+                    except OSError:
+                        import warnings
+
+                        warnings.warn(f"skipping missing file {op.path}")
+                        continue
                 b = extract_hash_bucket_id(op.path)
                 by_bucket.setdefault(b if b is not None else 0, []).append(op.path)
             for b, files in sorted(by_bucket.items()):
