@@ -269,3 +269,40 @@ def test_delete_partition(catalog):
     t.upsert({"id": np.array([1], dtype=np.int64), "v": np.array([1.0]), "s": ["a"]})
     t.delete_partition("-5")
     assert t.scan().count() == 0
+
+
+def test_invalid_file_tolerance(catalog):
+    """Files <8 bytes or missing are skipped with a warning
+    (reference session.rs:440-450)."""
+    import os
+    import warnings
+
+    t = _mk_pk_table(catalog, "tol", buckets=1)
+    t.upsert({"id": np.arange(10, dtype=np.int64), "v": np.zeros(10), "s": ["a"] * 10})
+    # register a bogus tiny file + a missing file into the snapshot
+    from lakesoul_amd.meta.entities import CommitOp, DataCommitInfo, DataFileOp, FileOp
+
+    bad = os.path.join(t.table_path, "part-bogus000000000_0000.parquet")
+    with open(bad, "wb") as f:
+        f.write(b"xx")
+    missing = os.path.join(t.table_path, "part-gone0000000000_0000.parquet")
+    t.client.commit_data_commit_info(
+        DataCommitInfo(
+            table_id=t.table_id, partition_desc="-5",
+            file_ops=[DataFileOp(bad, FileOp.add, 2), DataFileOp(missing, FileOp.add, 2)],
+            commit_op=CommitOp.MergeCommit,
+        )
+    )
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        df = t.to_pandas()
+    assert len(df) == 10
+    assert any("invalid" in str(x.message) or "missing" in str(x.message) for x in w)
+
+
+def test_oversized_unit_guard(catalog, monkeypatch):
+    t = _mk_pk_table(catalog, "huge", buckets=1)
+    t.upsert({"id": np.arange(1000, dtype=np.int64), "v": np.zeros(1000), "s": ["a"] * 1000})
+    monkeypatch.setenv("LAKESOUL_MAX_UNIT_BYTES", "1000")
+    with pytest.raises(MemoryError, match="hash\\s*buckets|buckets"):
+        t.to_pandas()
