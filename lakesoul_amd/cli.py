@@ -27,6 +27,27 @@ def cli():
     """lakesoul_amd — MI355X-native lakehouse engine."""
 
 
+@cli.command("sql")
+@click.argument("query", required=False)
+@click.option("--device", default=None, help="scan device (e.g. cuda:0)")
+def sql(query, device):
+    """Run one SQL statement, or start the interactive console
+    (reference: rust/lakesoul-console)."""
+    from .sql import execute_sql, repl
+
+    cat = _catalog()
+    if query:
+        df = execute_sql(cat, query, device=device)
+        try:
+            from tabulate import tabulate
+
+            click.echo(tabulate(df, headers="keys", tablefmt="psql", showindex=False))
+        except ImportError:
+            click.echo(df.to_string(index=False))
+    else:
+        repl(cat, device=device)
+
+
 @cli.command("list-tables")
 @click.option("--namespace", default="default")
 def list_tables(namespace):

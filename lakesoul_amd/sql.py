@@ -1,0 +1,437 @@
+"""Minimal SQL layer over the catalog — the MI355X-native stand-in for
+the reference's ``lakesoul-console`` interactive SQL REPL
+(``rust/lakesoul-console/src/exec.rs``) and the datafusion CLI
+(``rust/lakesoul-datafusion/src/cli.rs``).
+
+The reference delegates SQL to DataFusion; here a hand-rolled
+recursive-descent parser covers the console's practical surface —
+``SELECT`` with projection, aggregates, ``WHERE`` (pushed down into the
+scan's filter/stats/bucket pruning), ``GROUP BY``, ``ORDER BY``,
+``LIMIT``, plus ``SHOW TABLES / NAMESPACES`` and ``DESCRIBE`` — and
+executes against :class:`~lakesoul_amd.tables.catalog.Catalog` scans, so
+queries get the same MOR + pruning machinery as the programmatic API.
+"""
+
+from __future__ import annotations
+
+import re
+from dataclasses import dataclass, field
+from typing import List, Optional, Tuple
+
+from .io.filters import And, Cmp, Expr, IsNull, Not, Or
+
+_TOKEN_RE = re.compile(
+    r"""\s*(?:
+        (?P<num>-?\d+\.\d+(?:[eE][+-]?\d+)?|-?\.\d+|-?\d+)
+      | (?P<str>'(?:[^']|'')*')
+      | (?P<op><=|>=|<>|!=|=|<|>|\(|\)|,|\*|\.)
+      | (?P<id>[A-Za-z_][A-Za-z_0-9]*)
+    )""",
+    re.VERBOSE,
+)
+
+_KEYWORDS = {
+    "select", "from", "where", "group", "order", "by", "limit", "and",
+    "or", "not", "in", "is", "null", "between", "as", "asc", "desc",
+    "show", "tables", "namespaces", "describe", "distinct", "version",
+}
+
+_AGGS = {"count", "sum", "min", "max", "avg"}
+
+
+class SqlError(ValueError):
+    pass
+
+
+def tokenize(sql: str) -> List[Tuple[str, str]]:
+    out, pos = [], 0
+    sql = sql.strip().rstrip(";")
+    while pos < len(sql):
+        m = _TOKEN_RE.match(sql, pos)
+        if not m:
+            raise SqlError(f"bad token at: {sql[pos:pos+20]!r}")
+        pos = m.end()
+        if m.lastgroup == "num":
+            out.append(("num", m.group("num")))
+        elif m.lastgroup == "str":
+            out.append(("str", m.group("str")[1:-1].replace("''", "'")))
+        elif m.lastgroup == "op":
+            out.append(("op", m.group("op")))
+        else:
+            word = m.group("id")
+            kind = "kw" if word.lower() in _KEYWORDS else "id"
+            out.append((kind, word.lower() if kind == "kw" else word))
+    return out
+
+
+# --------------------------------------------------------------------- #
+# AST
+
+
+@dataclass
+class SelectItem:
+    # one of: column name | ("agg", fn, col_or_star) | ("lit", value)
+    kind: str                      # "col" | "agg" | "star"
+    name: str = ""                 # column name (col/agg arg)
+    fn: str = ""                   # aggregate fn
+    alias: str = ""
+
+    @property
+    def out_name(self) -> str:
+        if self.alias:
+            return self.alias
+        if self.kind == "agg":
+            return f"{self.fn}({self.name or '*'})"
+        return self.name
+
+
+@dataclass
+class Query:
+    table: str
+    namespace: str = "default"
+    items: List[SelectItem] = field(default_factory=list)
+    where: Optional[Expr] = None
+    group_by: List[str] = field(default_factory=list)
+    order_by: List[Tuple[str, bool]] = field(default_factory=list)  # (name, desc)
+    limit: Optional[int] = None
+    distinct: bool = False
+    version: Optional[int] = None
+
+
+class _Parser:
+    def __init__(self, tokens):
+        self.t = tokens
+        self.i = 0
+
+    def peek(self, k=0):
+        j = self.i + k
+        return self.t[j] if j < len(self.t) else ("eof", "")
+
+    def next(self):
+        tok = self.peek()
+        self.i += 1
+        return tok
+
+    def expect(self, kind, val=None):
+        k, v = self.next()
+        if k != kind or (val is not None and v != val):
+            raise SqlError(f"expected {val or kind}, got {v!r}")
+        return v
+
+    def accept(self, kind, val=None):
+        k, v = self.peek()
+        if k == kind and (val is None or v == val):
+            self.i += 1
+            return True
+        return False
+
+    # -- statements ----------------------------------------------------- #
+
+    def statement(self):
+        k, v = self.peek()
+        if (k, v) == ("kw", "show"):
+            self.next()
+            kind = self.expect("kw")
+            if kind not in ("tables", "namespaces"):
+                raise SqlError(f"SHOW {kind}?")
+            return ("show", kind)
+        if (k, v) == ("kw", "describe"):
+            self.next()
+            ns, name = self.table_name()
+            return ("describe", (ns, name))
+        if (k, v) == ("kw", "select"):
+            return ("select", self.select())
+        raise SqlError(f"unsupported statement start: {v!r}")
+
+    def table_name(self):
+        name = self.expect("id")
+        if self.accept("op", "."):
+            return name, self.expect("id")
+        return "default", name
+
+    # -- SELECT ---------------------------------------------------------- #
+
+    def select(self) -> Query:
+        self.expect("kw", "select")
+        q = Query(table="")
+        q.distinct = self.accept("kw", "distinct")
+        q.items = [self.select_item()]
+        while self.accept("op", ","):
+            q.items.append(self.select_item())
+        self.expect("kw", "from")
+        q.namespace, q.table = self.table_name()
+        if self.accept("kw", "version"):   # time travel: FROM t VERSION 3
+            q.version = int(self.expect("num"))
+        if self.accept("kw", "where"):
+            q.where = self.or_expr()
+        if self.accept("kw", "group"):
+            self.expect("kw", "by")
+            q.group_by = [self.expect("id")]
+            while self.accept("op", ","):
+                q.group_by.append(self.expect("id"))
+        if self.accept("kw", "order"):
+            self.expect("kw", "by")
+            q.order_by = [self.order_item(q)]
+            while self.accept("op", ","):
+                q.order_by.append(self.order_item(q))
+        if self.accept("kw", "limit"):
+            q.limit = int(self.expect("num"))
+        if self.peek()[0] != "eof":
+            raise SqlError(f"trailing tokens: {self.peek()[1]!r}")
+        return q
+
+    def select_item(self) -> SelectItem:
+        k, v = self.peek()
+        if (k, v) == ("op", "*"):
+            self.next()
+            return SelectItem("star")
+        if k == "id" and v.lower() in _AGGS and self.peek(1) == ("op", "("):
+            fn = self.next()[1].lower()
+            self.expect("op", "(")
+            if self.accept("op", "*"):
+                if fn != "count":
+                    raise SqlError(f"{fn}(*) not supported")
+                arg = ""
+            else:
+                arg = self.expect("id")
+            self.expect("op", ")")
+            item = SelectItem("agg", name=arg, fn=fn)
+        elif k == "id":
+            item = SelectItem("col", name=self.next()[1])
+        else:
+            raise SqlError(f"bad select item near {v!r}")
+        if self.accept("kw", "as"):
+            item.alias = self.expect("id")
+        elif self.peek()[0] == "id":   # bare alias
+            item.alias = self.next()[1]
+        return item
+
+    def order_item(self, q: Query):
+        name = self.expect("id") if self.peek()[0] == "id" else str(self.expect("num"))
+        if name.isdigit():
+            name = q.items[int(name) - 1].out_name
+        desc = False
+        if self.accept("kw", "desc"):
+            desc = True
+        else:
+            self.accept("kw", "asc")
+        return name, desc
+
+    # -- WHERE ----------------------------------------------------------- #
+
+    def or_expr(self) -> Expr:
+        e = self.and_expr()
+        while self.accept("kw", "or"):
+            e = Or(e, self.and_expr())
+        return e
+
+    def and_expr(self) -> Expr:
+        e = self.not_expr()
+        while self.accept("kw", "and"):
+            e = And(e, self.not_expr())
+        return e
+
+    def not_expr(self) -> Expr:
+        if self.accept("kw", "not"):
+            return Not(self.not_expr())
+        return self.predicate()
+
+    def literal(self):
+        k, v = self.next()
+        if k == "num":
+            return float(v) if ("." in v or "e" in v.lower()) else int(v)
+        if k == "str":
+            return v
+        if k == "id" and v.lower() in ("true", "false"):
+            return v.lower() == "true"
+        raise SqlError(f"expected literal, got {v!r}")
+
+    def predicate(self) -> Expr:
+        if self.accept("op", "("):
+            e = self.or_expr()
+            self.expect("op", ")")
+            return e
+        col = self.expect("id")
+        k, v = self.peek()
+        if (k, v) == ("kw", "is"):
+            self.next()
+            neg = self.accept("kw", "not")
+            self.expect("kw", "null")
+            return IsNull(col, negate=neg)
+        if (k, v) == ("kw", "in") or ((k, v) == ("kw", "not") and self.peek(1) == ("kw", "in")):
+            neg = self.accept("kw", "not")
+            self.expect("kw", "in")
+            self.expect("op", "(")
+            vals = [self.literal()]
+            while self.accept("op", ","):
+                vals.append(self.literal())
+            self.expect("op", ")")
+            e: Expr = Cmp(col, "eq", vals[0])
+            for x in vals[1:]:
+                e = Or(e, Cmp(col, "eq", x))
+            return Not(e) if neg else e
+        if (k, v) == ("kw", "between") or ((k, v) == ("kw", "not") and self.peek(1) == ("kw", "between")):
+            neg = self.accept("kw", "not")
+            self.expect("kw", "between")
+            lo = self.literal()
+            self.expect("kw", "and")
+            hi = self.literal()
+            e = And(Cmp(col, "gteq", lo), Cmp(col, "lteq", hi))
+            return Not(e) if neg else e
+        op = self.expect("op")
+        ops = {"=": "eq", "!=": "noteq", "<>": "noteq", "<": "lt",
+               "<=": "lteq", ">": "gt", ">=": "gteq"}
+        if op not in ops:
+            raise SqlError(f"bad comparison operator {op!r}")
+        return Cmp(col, ops[op], self.literal())
+
+
+def parse_sql(sql: str):
+    return _Parser(tokenize(sql)).statement()
+
+
+# --------------------------------------------------------------------- #
+# Execution
+
+
+def execute_sql(catalog, sql: str, device: Optional[str] = None):
+    """Run a SQL statement and return a pandas DataFrame."""
+    import pandas as pd
+
+    kind, payload = parse_sql(sql)
+    if kind == "show":
+        if payload == "namespaces":
+            return pd.DataFrame({"namespace": catalog.list_namespaces()})
+        return pd.DataFrame({"table": catalog.list_tables()})
+    if kind == "describe":
+        ns, name = payload
+        t = catalog.table(name, ns)
+        sch = t.schema
+        return pd.DataFrame({
+            "column": [f.name for f in sch],
+            "type": [f.dtype for f in sch],
+            "nullable": [f.nullable for f in sch],
+            "primary_key": [f.name in (t.primary_keys or []) for f in sch],
+        })
+    return _execute_select(catalog, payload, device=device)
+
+
+def _execute_select(catalog, q: Query, device=None):
+    import numpy as np
+    import pandas as pd
+
+    t = catalog.table(q.table, q.namespace)
+    schema_cols = t.schema.names()
+
+    # columns actually needed from storage
+    need = set(q.group_by)
+    star = any(it.kind == "star" for it in q.items)
+    has_agg = any(it.kind == "agg" for it in q.items)
+    for it in q.items:
+        if it.kind in ("col", "agg") and it.name:
+            need.add(it.name)
+    for name, _ in q.order_by:
+        if name in schema_cols:
+            need.add(name)
+    if star:
+        need = set(schema_cols)
+    for c in need:
+        if c not in schema_cols:
+            raise SqlError(f"unknown column {c!r} in {q.namespace}.{q.table}")
+
+    # count(*) with no WHERE/grouping: metadata-free count-only scan
+    # (reference: EmptyScanCountExec, physical_plan/empty_schema.rs:192)
+    if (len(q.items) == 1 and q.items[0].kind == "agg"
+            and q.items[0].fn == "count" and not q.items[0].name
+            and q.where is None and not q.group_by and not q.distinct):
+        n = t.scan(version=q.version, device=device).count()
+        return pd.DataFrame({q.items[0].out_name: [n]})
+
+    scan = t.scan(columns=sorted(need) or None, filters=q.where,
+                  version=q.version, device=device)
+    df = scan.to_arrow().to_pandas()
+
+    # aggregate / project
+    if has_agg or q.group_by:
+        def agg_series(sub: pd.DataFrame):
+            row = {}
+            for it in q.items:
+                if it.kind == "col":
+                    if it.name not in q.group_by:
+                        raise SqlError(
+                            f"column {it.name!r} must appear in GROUP BY")
+                    continue
+                if it.kind == "star":
+                    raise SqlError("SELECT * with aggregates is not valid")
+                s = sub[it.name] if it.name else None
+                if it.fn == "count":
+                    row[it.out_name] = len(sub) if s is None else int(s.notna().sum())
+                elif it.fn == "sum":
+                    row[it.out_name] = s.sum()
+                elif it.fn == "min":
+                    row[it.out_name] = s.min()
+                elif it.fn == "max":
+                    row[it.out_name] = s.max()
+                elif it.fn == "avg":
+                    row[it.out_name] = float(s.mean())
+            return pd.Series(row)
+
+        if q.group_by:
+            out = (df.groupby(q.group_by, as_index=False, sort=False)
+                     .apply(agg_series, include_groups=False)
+                   if pd.__version__ >= "2.2"
+                   else df.groupby(q.group_by, as_index=False).apply(agg_series))
+            # keep declared item order, renaming group cols per alias
+            cols = []
+            for it in q.items:
+                cols.append(it.name if it.kind == "col" and not it.alias else it.out_name)
+                if it.kind == "col" and it.alias:
+                    out = out.rename(columns={it.name: it.alias})
+            out = out[cols]
+        else:
+            out = pd.DataFrame([agg_series(df)])
+    else:
+        cols, ren = [], {}
+        for it in q.items:
+            if it.kind == "star":
+                cols.extend([c for c in schema_cols if c not in cols])
+            else:
+                cols.append(it.name)
+                if it.alias:
+                    ren[it.name] = it.alias
+        out = df[cols].rename(columns=ren)
+        if q.distinct:
+            out = out.drop_duplicates().reset_index(drop=True)
+
+    if q.order_by:
+        names = [n for n, _ in q.order_by]
+        asc = [not d for _, d in q.order_by]
+        out = out.sort_values(names, ascending=asc).reset_index(drop=True)
+    if q.limit is not None:
+        out = out.head(q.limit).reset_index(drop=True)
+    return out.reset_index(drop=True)
+
+
+def repl(catalog, device=None, input_fn=input, print_fn=print):
+    """Interactive console loop (reference: lakesoul-console/src/main.rs)."""
+    print_fn("lakesoul_amd SQL console — \\q to quit")
+    while True:
+        try:
+            line = input_fn("lakesoul> ")
+        except (EOFError, KeyboardInterrupt):
+            break
+        line = line.strip()
+        if not line:
+            continue
+        if line in ("\\q", "quit", "exit"):
+            break
+        try:
+            df = execute_sql(catalog, line, device=device)
+            try:
+                from tabulate import tabulate
+                print_fn(tabulate(df, headers="keys", tablefmt="psql", showindex=False))
+            except ImportError:
+                print_fn(df.to_string(index=False))
+            print_fn(f"({len(df)} rows)")
+        except Exception as e:  # console: report, keep looping
+            print_fn(f"error: {e}")
