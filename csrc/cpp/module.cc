@@ -26,6 +26,8 @@ struct DtypeInfo {
   int32_t converted;
   LogicalTag logical;
   int32_t bit_width;
+  int32_t dec_precision = 0;
+  int32_t dec_scale = 0;
 };
 
 static DtypeInfo dtype_info(const std::string& dt) {
@@ -45,6 +47,18 @@ static DtypeInfo dtype_info(const std::string& dt) {
     return {PT_INT64, CV_TIMESTAMP_MICROS, LogicalTag::TIMESTAMP_MICROS, 0};
   if (dt == "timestamp[ns]")
     return {PT_INT64, CV_NONE, LogicalTag::TIMESTAMP_NANOS, 0};
+  if (dt.rfind("decimal(", 0) == 0 && dt.back() == ')') {
+    // decimal(p,s), p<=18: stored as INT64 unscaled (parquet spec allows
+    // INT32/INT64 physical for small precisions; arrow reads it with
+    // store_decimal_as_integer parity)
+    int p = 0, sc = 0;
+    if (sscanf(dt.c_str(), "decimal(%d,%d)", &p, &sc) != 2 || p <= 0 || p > 18)
+      throw std::runtime_error("unsupported decimal dtype: " + dt);
+    DtypeInfo di{PT_INT64, CV_DECIMAL, LogicalTag::DECIMAL, 0};
+    di.dec_precision = p;
+    di.dec_scale = sc;
+    return di;
+  }
   throw std::runtime_error("unsupported dtype: " + dt);
 }
 
@@ -60,6 +74,9 @@ static std::string dtype_name(const ColumnDesc& c) {
       if (c.converted == CV_DATE || c.logical == LogicalTag::DATE) return "date32";
       return "int32";
     case PT_INT64:
+      if (c.converted == CV_DECIMAL || c.logical == LogicalTag::DECIMAL)
+        return "decimal(" + std::to_string(c.dec_precision) + "," +
+               std::to_string(c.dec_scale) + ")";
       if (c.converted == CV_TIMESTAMP_MILLIS || c.logical == LogicalTag::TIMESTAMP_MILLIS)
         return "timestamp[ms]";
       if (c.converted == CV_TIMESTAMP_MICROS || c.logical == LogicalTag::TIMESTAMP_MICROS)
@@ -103,6 +120,8 @@ static int64_t write_parquet(
     descs[i].converted = di.converted;
     descs[i].logical = di.logical;
     descs[i].int_bit_width = di.bit_width;
+    descs[i].dec_precision = di.dec_precision;
+    descs[i].dec_scale = di.dec_scale;
     descs[i].nullable = nullable[i];
 
     torch::Tensor col = columns[i].contiguous().cpu();

@@ -45,6 +45,8 @@ struct ColumnDesc {
   LogicalTag logical = LogicalTag::NONE;
   int32_t int_bit_width = 0;
   bool int_signed = true;
+  int32_t dec_precision = 0;
+  int32_t dec_scale = 0;
 };
 
 inline int physical_elem_size(int32_t pt) {
@@ -123,6 +125,8 @@ class ParquetWriter {
       e.logical = c.logical;
       e.int_bit_width = c.int_bit_width;
       e.int_signed = c.int_signed;
+      e.dec_precision = c.dec_precision;
+      e.dec_scale = c.dec_scale;
       fm.schema.push_back(e);
     }
     fm.num_rows = total_rows_;
@@ -604,6 +608,8 @@ class ParquetFile {
       c.logical = e.logical;
       c.int_bit_width = e.int_bit_width;
       c.int_signed = e.int_signed;
+      c.dec_precision = e.dec_precision;
+      c.dec_scale = e.dec_scale;
       cols_.push_back(c);
     }
   }

@@ -80,6 +80,7 @@ enum class LogicalTag {
   TIMESTAMP_NANOS,
   INT,       // with bit_width / signed
   FLOAT16,
+  DECIMAL,   // with dec_precision / dec_scale
 };
 
 struct SchemaElement {
@@ -93,6 +94,8 @@ struct SchemaElement {
   int32_t int_bit_width = 0;
   bool int_signed = true;
   bool ts_utc = true;
+  int32_t dec_precision = 0;
+  int32_t dec_scale = 0;
 };
 
 struct Statistics {
@@ -191,12 +194,25 @@ inline SchemaElement parse_schema_element(ThriftReader& r) {
       case 4: e.name = r.read_binary(); break;
       case 5: e.num_children = (int32_t)r.read_zigzag(); break;
       case 6: e.converted = (int32_t)r.read_zigzag(); break;
+      case 7: e.dec_scale = (int32_t)r.read_zigzag(); break;
+      case 8: e.dec_precision = (int32_t)r.read_zigzag(); break;
       case 10: {  // LogicalType union
         int16_t f2 = 0;
         CType t2;
         while (r.read_field_header(f2, t2)) {
           switch (f2) {
             case 1: e.logical = LogicalTag::STRING; r.skip(t2); break;
+            case 5: {  // DecimalType { 1: i32 scale, 2: i32 precision }
+              e.logical = LogicalTag::DECIMAL;
+              int16_t f3 = 0;
+              CType t3;
+              while (r.read_field_header(f3, t3)) {
+                if (f3 == 1) e.dec_scale = (int32_t)r.read_zigzag();
+                else if (f3 == 2) e.dec_precision = (int32_t)r.read_zigzag();
+                else r.skip(t3);
+              }
+              break;
+            }
             case 6: e.logical = LogicalTag::DATE; r.skip(t2); break;
             case 8: {  // TIMESTAMP
               int16_t f3 = 0;
@@ -421,6 +437,10 @@ inline void write_schema_element(ThriftWriter& w, const SchemaElement& e) {
   w.field_binary(last, 4, e.name);
   if (e.num_children > 0) w.field_i32(last, 5, e.num_children);
   if (e.converted != CV_NONE) w.field_i32(last, 6, e.converted);
+  if (e.converted == CV_DECIMAL || e.logical == LogicalTag::DECIMAL) {
+    w.field_i32(last, 7, e.dec_scale);
+    w.field_i32(last, 8, e.dec_precision);
+  }
   if (e.logical != LogicalTag::NONE) {
     w.field(last, 10, CType::STRUCT);
     int16_t l2 = 0;
@@ -463,6 +483,14 @@ inline void write_schema_element(ThriftWriter& w, const SchemaElement& e) {
         w.field(l2, 15, CType::STRUCT);
         w.stop();
         break;
+      case LogicalTag::DECIMAL: {
+        w.field(l2, 5, CType::STRUCT);
+        int16_t l3 = 0;
+        w.field_i32(l3, 1, e.dec_scale);
+        w.field_i32(l3, 2, e.dec_precision);
+        w.stop();
+        break;
+      }
       default:
         break;
     }

@@ -48,7 +48,15 @@ _NP_DTYPE = {
 
 
 def torch_dtype_for(dtype: str):
+    if dtype.startswith("decimal"):
+        return torch.int64   # unscaled int64 backing
     return _TORCH_DTYPE[dtype]
+
+
+def np_dtype_for(dtype: str):
+    if dtype.startswith("decimal"):
+        return np.int64
+    return _NP_DTYPE[dtype]
 
 
 @dataclass
@@ -173,8 +181,8 @@ class Batch:
                         v = v.astype(np.uint8)
                     t = torch.from_numpy(np.ascontiguousarray(v))
                 else:
-                    t = torch.from_numpy(np.asarray(v, dtype=_NP_DTYPE[f.dtype]))
-                t = t.to(_TORCH_DTYPE[f.dtype])
+                    t = torch.from_numpy(np.asarray(v, dtype=np_dtype_for(f.dtype)))
+                t = t.to(torch_dtype_for(f.dtype))
                 cols[f.name] = Column(f.dtype, data=t)
             else:
                 if isinstance(v, tuple) and len(v) in (2, 3):
@@ -220,6 +228,25 @@ class Batch:
         for f in schema:
             col = t.column(f.name)
             arr = col.combine_chunks() if hasattr(col, "combine_chunks") else col
+            if f.dtype.startswith("decimal"):
+                from .schema import decimal_params
+
+                _, sc = decimal_params(f.dtype)
+                import decimal as _dec
+
+                vals = arr.to_pylist()
+                validity = None
+                if arr.null_count:
+                    validity = torch.tensor(
+                        [0 if x is None else 1 for x in vals], dtype=torch.uint8)
+                q = _dec.Decimal(1).scaleb(-sc)
+                unscaled = np.asarray(
+                    [0 if x is None else int(x.quantize(q).scaleb(sc)) for x in vals],
+                    dtype=np.int64,
+                )
+                c = Column(f.dtype, data=torch.from_numpy(unscaled), validity=validity)
+                d[f.name] = c
+                continue
             if f.is_fixed_width:
                 np_arr = arr.to_numpy(zero_copy_only=False)
                 validity = None
@@ -227,13 +254,13 @@ class Batch:
                     validity = (~np.asarray(arr.is_null())).astype(np.uint8)
                     np_arr = np.nan_to_num(np_arr) if np_arr.dtype.kind == "f" else np_arr
                     if np_arr.dtype == object:
-                        np_arr = np.where(validity, np_arr, 0).astype(_NP_DTYPE[f.dtype])
+                        np_arr = np.where(validity, np_arr, 0).astype(np_dtype_for(f.dtype))
                 if np_arr.dtype == object or (np_arr.dtype.kind == "f" and f.dtype.startswith("int")):
                     np_arr = np.asarray(
                         [0 if x is None else x for x in arr.to_pylist()],
-                        dtype=_NP_DTYPE[f.dtype],
+                        dtype=np_dtype_for(f.dtype),
                     )
-                b = cls.from_dict({f.name: np_arr.astype(_NP_DTYPE[f.dtype])}, Schema([f]))
+                b = cls.from_dict({f.name: np_arr.astype(np_dtype_for(f.dtype))}, Schema([f]))
                 c = b.columns[f.name]
                 if validity is not None:
                     c.validity = torch.from_numpy(validity)
@@ -253,6 +280,24 @@ class Batch:
         arrays = []
         for f in self.schema:
             c = self.columns[f.name]
+            if f.dtype.startswith("decimal"):
+                v = c.data.cpu().numpy().astype(np.int64)
+                n = v.size
+                wide = np.empty((n, 2), dtype=np.int64)   # little-endian lo, hi
+                wide[:, 0] = v
+                wide[:, 1] = v >> 63                      # sign extension
+                validity_buf = None
+                null_count = 0
+                if c.validity is not None:
+                    vv = c.validity.cpu().numpy().astype(bool)
+                    null_count = int(n - vv.sum())
+                    validity_buf = pa.py_buffer(np.packbits(vv, bitorder="little").tobytes())
+                arrays.append(pa.Array.from_buffers(
+                    dtype_to_arrow(f.dtype), n,
+                    [validity_buf, pa.py_buffer(wide.tobytes())],
+                    null_count=null_count,
+                ))
+                continue
             if f.is_fixed_width:
                 np_arr = c.data.cpu().numpy()
                 if f.dtype == "bool":

@@ -147,6 +147,15 @@ class Cmp(Expr):
         else:
             t = c.data
             v = self.value
+            if c.dtype.startswith("decimal") and not isinstance(v, (list, tuple, set)):
+                # logical literal -> unscaled int64 (decimal columns store
+                # unscaled values; exact scaling via decimal module)
+                import decimal as _dec
+
+                from .schema import decimal_params
+
+                _, sc = decimal_params(c.dtype)
+                v = int(_dec.Decimal(str(v)).scaleb(sc).to_integral_value())
             if self.op == "eq":
                 out = t == v
             elif self.op == "noteq":
@@ -288,6 +297,8 @@ def parse_filter_dsl(s: str, schema: Schema) -> Expr:
 
 
 def _parse_literal(s: str, dtype: str):
+    if dtype.startswith("decimal"):
+        return float(s)   # scaled at evaluate() against the unscaled column
     if dtype in ("string", "binary"):
         if len(s) >= 2 and s[0] == s[-1] and s[0] in "'\"":
             s = s[1:-1]

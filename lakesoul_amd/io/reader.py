@@ -48,6 +48,14 @@ _NP_FROM_PHYS = {
     "timestamp[ns]": np.int64,
 }
 
+def _np_phys(dtype):
+    return np.int64 if dtype.startswith("decimal") else _NP_FROM_PHYS[dtype]
+
+
+def _np_target(dtype):
+    return np.int64 if dtype.startswith("decimal") else _NP_TARGET[dtype]
+
+
 _NP_TARGET = {
     "bool": np.uint8,
     "int8": np.int8,
@@ -448,7 +456,7 @@ class LakeSoulScan:
                     validity=None if npc.validity is None else torch.from_numpy(npc.validity),
                 )
             else:
-                data = npc.data.astype(_NP_TARGET[f.dtype], copy=False)
+                data = npc.data.astype(_np_target(f.dtype), copy=False)
                 cols[f.name] = Column(
                     f.dtype,
                     data=torch.from_numpy(np.ascontiguousarray(data)),
@@ -464,7 +472,7 @@ class LakeSoulScan:
             if "=" in kv and kv.split("=", 1)[0] == f.name:
                 val = kv.split("=", 1)[1]
         if f.is_fixed_width:
-            npdt = _NP_TARGET[f.dtype]
+            npdt = _np_target(f.dtype)
             arr = np.full(n, npdt(val) if val is not None else 0, dtype=npdt)
             return Column(f.dtype, data=torch.from_numpy(arr))
         enc = (val or "").encode()
@@ -501,7 +509,7 @@ class LakeSoulScan:
                         offs_parts.append(d["offsets"].numpy())
                         bytes_parts.append(d["bytes"].numpy())
                     else:
-                        parts.append(d["data"].numpy().view(_NP_FROM_PHYS[f.dtype]))
+                        parts.append(d["data"].numpy().view(_np_phys(f.dtype)))
                     v = d["validity"].numpy()
                     if len(v):
                         any_null = True
@@ -525,7 +533,7 @@ class LakeSoulScan:
                     )
                     out[name] = NpColumn(f.dtype, None, offs, bys, validity)
                 else:
-                    data = np.concatenate(parts) if parts else np.empty(0, _NP_FROM_PHYS[f.dtype])
+                    data = np.concatenate(parts) if parts else np.empty(0, _np_phys(f.dtype))
                     out[name] = NpColumn(f.dtype, data, None, None, validity)
             # schema evolution: missing columns become nulls
             for name in names:
@@ -543,7 +551,7 @@ class LakeSoulScan:
                 else:
                     out[name] = NpColumn(
                         f.dtype,
-                        np.zeros(total, dtype=_NP_FROM_PHYS[f.dtype]),
+                        np.zeros(total, dtype=_np_phys(f.dtype)),
                         None,
                         None,
                         np.zeros(total, dtype=np.uint8),

@@ -41,6 +41,10 @@ _TORCH_VIEW = {
     "timestamp[ns]": torch.int64,
 }
 _TARGET = {"int8": torch.int8, "int16": torch.int16}
+
+
+def _torch_view(dtype):
+    return torch.int64 if dtype.startswith("decimal") else _torch_view(dtype)
 _ESIZE = {torch.uint8: 1, torch.int32: 4, torch.int64: 8, torch.float32: 4, torch.float64: 8}
 
 
@@ -168,7 +172,7 @@ def read_unit_gpu(scan, unit, raw: Optional[dict] = None,
                 cols[name] = Column(f.dtype, offsets=offs, bytes_=by, validity=vmask)
             else:
                 data, vmask = entry[0], entry[1]
-                data = data.view(_TORCH_VIEW[f.dtype])
+                data = data.view(_torch_view(f.dtype))
                 if f.dtype in _TARGET:
                     data = data.to(_TARGET[f.dtype])
                 cols[name] = Column(f.dtype, data=data, validity=vmask)
@@ -218,7 +222,7 @@ def read_unit_gpu(scan, unit, raw: Optional[dict] = None,
                 cols[name] = Column(f.dtype, offsets=offs, bytes_=by, validity=vmask)
                 continue
 
-            tdt = _TORCH_VIEW[f.dtype]
+            tdt = _torch_view(f.dtype)
             esize = _ESIZE[tdt]
             if cd["is_dict"]:
                 runs = runs_buf.narrow(0, cd["run_off"], cd["run_cnt"])
@@ -275,7 +279,7 @@ def read_unit_gpu(scan, unit, raw: Optional[dict] = None,
 def _decode_fixed_chunk_gpu(d: dict, dtype: str, device) -> Column:
     """Decode one raw column chunk (fixed width) — standalone helper used
     by kernel unit tests; the production path is read_unit_gpu."""
-    tdt = _TORCH_VIEW[dtype]
+    tdt = _torch_view(dtype)
     esize = _ESIZE[tdt]
     nv = d["num_values"]
     has_nulls = d["validity"].numel() > 0 and d["null_count"] > 0

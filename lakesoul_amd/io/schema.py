@@ -78,7 +78,27 @@ FIXED_WIDTH_BYTES = {
 }
 
 
+_DECIMAL_RE = None
+
+
+def decimal_params(dtype: str):
+    """Return (precision, scale) for a decimal dtype string, else None."""
+    global _DECIMAL_RE
+    if _DECIMAL_RE is None:
+        import re
+        _DECIMAL_RE = re.compile(r"decimal\s*\(\s*(\d+)\s*,\s*(\d+)\s*\)$")
+    m = _DECIMAL_RE.match(dtype.strip().lower())
+    return (int(m.group(1)), int(m.group(2))) if m else None
+
+
 def canonical_dtype(dt: str) -> str:
+    dp = decimal_params(dt)
+    if dp is not None:
+        p, sc = dp
+        if not (0 < p <= 18) or not (0 <= sc <= p):
+            raise TypeError(
+                f"unsupported decimal precision/scale {dt} (int64-backed: p<=18)")
+        return f"decimal({p},{sc})"
     try:
         return _CANONICAL[dt.lower()]
     except KeyError:
@@ -96,7 +116,7 @@ class Field:
 
     @property
     def is_fixed_width(self) -> bool:
-        return self.dtype in FIXED_WIDTH_BYTES
+        return self.dtype in FIXED_WIDTH_BYTES or self.dtype.startswith("decimal")
 
 
 @dataclass
@@ -177,6 +197,8 @@ def _arrow_type_to_dtype(t) -> str:
 
     if pt.is_timestamp(t):
         return f"timestamp[{t.unit}]"
+    if pt.is_decimal(t):
+        return canonical_dtype(f"decimal({t.precision},{t.scale})")
     raise TypeError(f"unsupported arrow type {t}")
 
 
@@ -199,6 +221,9 @@ def dtype_to_arrow(dtype: str):
         "timestamp[ms]": pa.timestamp("ms"),
         "timestamp[ns]": pa.timestamp("ns"),
     }
+    dp = decimal_params(dtype)
+    if dp is not None:
+        return pa.decimal128(*dp)
     return mapping[dtype]
 
 
@@ -217,7 +242,7 @@ def schema_to_json(schema: Schema) -> str:
             "fields": [
                 {
                     "name": f.name,
-                    "type": _TO_SPARK[f.dtype],
+                    "type": _TO_SPARK.get(f.dtype, f.dtype),
                     "nullable": f.nullable,
                     "metadata": {},
                 }

@@ -361,3 +361,29 @@ def test_gpu_int32_single_pk(dev, tmp_path):
     df = t.scan(device="cuda").to_arrow().to_pandas().sort_values("id").reset_index(drop=True)
     assert len(df) == n
     assert (df["v"][::7] == 1.0).all() and (df["v"][1::7] == 0.0).all()
+
+
+@pytest.mark.gpu
+def test_gpu_decimal_mor(gpu_catalog):
+    """decimal column through the GPU MOR path (int64 unscaled on device)."""
+    import decimal
+
+    from lakesoul_amd.io.schema import Field, Schema
+
+    t = gpu_catalog.create_table(
+        "gdec",
+        Schema([Field("id", "int64", False), Field("amt", "decimal(12,2)")]),
+        primary_keys=["id"],
+        hash_bucket_num=2,
+    )
+    n = 50000
+    t.upsert({"id": np.arange(n, dtype=np.int64),
+              "amt": (np.arange(n, dtype=np.int64) * 7) % 100000})
+    t.upsert({"id": np.arange(0, n, 3, dtype=np.int64),
+              "amt": np.full((n + 2) // 3, 123456, dtype=np.int64)})
+    df = t.scan(device="cuda:0").to_arrow().to_pandas()
+    df = df.sort_values("id").reset_index(drop=True)
+    expect = (np.arange(n, dtype=np.int64) * 7) % 100000
+    expect[::3] = 123456
+    got = np.array([int(x.scaleb(2)) for x in df["amt"]])
+    np.testing.assert_array_equal(got, expect)

@@ -214,3 +214,61 @@ def test_multiple_row_groups_and_stats(tmp_path):
     pf = pq.ParquetFile(str(path))
     assert pf.metadata.num_row_groups == 3
     assert pf.metadata.row_group(0).column(0).statistics.min == 0
+
+def test_decimal_roundtrip_ours_to_pyarrow(tmp_path):
+    """decimal(p,s) written as INT64 unscaled with DECIMAL logical type;
+    pyarrow must read it back as decimal128 with the right values."""
+    import decimal
+
+    unscaled = np.array([12345, -995, 0, 10**10], dtype=np.int64)
+    path = tmp_path / "dec.parquet"
+    _write_ours(path, ["id", "amt"], ["int64", "decimal(12,2)"],
+                [torch.arange(4, dtype=torch.int64), torch.from_numpy(unscaled)])
+    t = pq.read_table(path)
+    assert t.schema.field("amt").type == pa.decimal128(12, 2)
+    assert t.column("amt").to_pylist() == [
+        decimal.Decimal(int(x)).scaleb(-2) for x in unscaled
+    ]
+
+
+def test_decimal_roundtrip_pyarrow_to_ours(tmp_path):
+    """pyarrow writing decimal as integer (store_decimal_as_integer) must be
+    readable by our reader with matching unscaled values + validity."""
+    import decimal
+
+    vals = [decimal.Decimal("123.45"), decimal.Decimal("-9.99"), None,
+            decimal.Decimal("0.01")]
+    t = pa.table({"id": pa.array([1, 2, 3, 4], pa.int64()),
+                  "amt": pa.array(vals, pa.decimal128(10, 2))})
+    path = tmp_path / "pad.parquet"
+    pq.write_table(t, path, store_decimal_as_integer=True,
+                   use_dictionary=False, compression="zstd")
+
+    h = cpp().open_parquet(str(path))
+    try:
+        meta = cpp().parquet_meta(h)
+        amt_ci = [i for i, c in enumerate(meta["columns"]) if c["name"] == "amt"][0]
+        assert meta["columns"][amt_ci]["dtype"] == "decimal(10,2)"
+        d = cpp().read_chunk_cpu(h, 0, amt_ci)
+        got = d["data"].numpy().view(np.int64)
+        mask = d["validity"].numpy()
+    finally:
+        cpp().close_parquet(h)
+    assert mask.tolist() == [1, 1, 0, 1]
+    np.testing.assert_array_equal(got[[0, 1, 3]], [12345, -999, 1])
+
+
+def test_decimal_batch_arrow_bridge():
+    """Batch.from_arrow / to_arrow map decimal128 <-> unscaled int64."""
+    import decimal
+
+    from lakesoul_amd.io.batch import Batch
+    from lakesoul_amd.io.schema import Field, Schema
+
+    vals = [decimal.Decimal("1.50"), None, decimal.Decimal("-2.25")]
+    t = pa.table({"amt": pa.array(vals, pa.decimal128(8, 2))})
+    sch = Schema([Field("amt", "decimal(8,2)")])
+    b = Batch.from_arrow(t, sch)
+    np.testing.assert_array_equal(b.columns["amt"].data.numpy()[[0, 2]], [150, -225])
+    assert b.columns["amt"].validity.numpy().tolist() == [1, 0, 1]
+    assert b.to_arrow().column("amt").to_pylist() == vals

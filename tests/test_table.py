@@ -306,3 +306,60 @@ def test_oversized_unit_guard(catalog, monkeypatch):
     monkeypatch.setenv("LAKESOUL_MAX_UNIT_BYTES", "1000")
     with pytest.raises(MemoryError, match="hash\\s*buckets|buckets"):
         t.to_pandas()
+
+
+def test_decimal_column_table(catalog):
+    """decimal(p,s) end-to-end: upsert, MOR merge, filters, SQL, SumAll."""
+    import decimal
+
+    import pyarrow as pa
+
+    from lakesoul_amd.io.schema import Field, Schema
+
+    t = catalog.create_table(
+        "dec",
+        Schema([Field("id", "int64", False), Field("amt", "decimal(12,2)")]),
+        primary_keys=["id"],
+        hash_bucket_num=2,
+    )
+    # write via arrow decimals
+    t.upsert(pa.table({
+        "id": pa.array([1, 2, 3], pa.int64()),
+        "amt": pa.array([decimal.Decimal("10.50"), decimal.Decimal("-1.25"),
+                         decimal.Decimal("3.00")], pa.decimal128(12, 2)),
+    }))
+    # upsert overwrite via raw unscaled ints
+    t.upsert({"id": np.array([2], dtype=np.int64),
+              "amt": np.array([9900], dtype=np.int64)})
+    df = _df(t)
+    assert df["amt"].tolist() == [decimal.Decimal("10.50"),
+                                  decimal.Decimal("99.00"),
+                                  decimal.Decimal("3.00")]
+    # filter with a logical literal
+    got = t.to_pandas(filters=[("amt", ">", 5.0)])
+    assert sorted(got["id"]) == [1, 2]
+    # SQL over decimal
+    from lakesoul_amd.sql import execute_sql
+
+    q = execute_sql(catalog, "SELECT id FROM dec WHERE amt >= 10.5 ORDER BY id")
+    assert q["id"].tolist() == [1, 2]
+
+
+def test_decimal_sum_all_merge(catalog):
+    from lakesoul_amd.io.schema import Field, Schema
+
+    t = catalog.create_table(
+        "decsum",
+        Schema([Field("id", "int64", False), Field("amt", "decimal(10,2)")]),
+        primary_keys=["id"],
+        properties={"merge_op.amt": "SumAll"},
+    )
+    t.upsert({"id": np.array([1, 2], dtype=np.int64),
+              "amt": np.array([100, 250], dtype=np.int64)})
+    t.upsert({"id": np.array([2, 3], dtype=np.int64),
+              "amt": np.array([50, 75], dtype=np.int64)})
+    import decimal
+
+    df = _df(t)
+    assert df["amt"].tolist() == [decimal.Decimal("1.00"), decimal.Decimal("3.00"),
+                                  decimal.Decimal("0.75")]
