@@ -241,6 +241,7 @@ class LakeSoulScan:
         units = self.plan()
         if (self.device == "cuda" and len(units) > 1
                 and self._gpu_merge_supported() and not self.use_cache
+                and all(self._unit_fits(u) for u in units)
                 and os.environ.get("LAKESOUL_SCAN_PIPELINE", "1") != "0"):
             gen = self._iter_units_pipelined(units)
         else:
@@ -362,10 +363,10 @@ class LakeSoulScan:
             self.schema.field(p).dtype not in ("string", "binary") for p in self.pk
         )
 
-    def _check_unit_size(self, unit: ScanUnit) -> None:
-        """Guard against buckets whose decoded size would not fit the
-        configured HBM budget (chunked spill merge is a round-2 item,
-        ROADMAP.md): fail with actionable advice instead of OOM-ing."""
+    def _unit_fits(self, unit: ScanUnit) -> bool:
+        """Estimate whether a bucket's decoded size fits the configured
+        device-memory budget (LAKESOUL_MAX_UNIT_BYTES, default 64 GB —
+        leaves headroom in 288 GB HBM for merge intermediates)."""
         limit = int(os.environ.get("LAKESOUL_MAX_UNIT_BYTES", str(64 * 1024**3)))
         total = 0
         for path in unit.files:
@@ -374,18 +375,18 @@ class LakeSoulScan:
             except OSError:
                 pass
         # decompressed estimate: zstd(1) on typical columns ~2x
-        if total * 2 > limit:
-            raise MemoryError(
-                f"scan unit bucket={unit.bucket_id} estimated decoded size "
-                f"~{total * 2 / 1e9:.0f} GB exceeds LAKESOUL_MAX_UNIT_BYTES "
-                f"({limit / 1e9:.0f} GB). Recreate the table with more hash "
-                f"buckets, compact the partition, or raise the limit."
-            )
+        return total * 2 <= limit
 
     def _read_unit(self, unit: ScanUnit) -> Optional[Batch]:
         if not unit.files:
             return None
-        self._check_unit_size(unit)
+        oversized = not self._unit_fits(unit)
+        if oversized and self.device != "cuda":
+            raise MemoryError(
+                f"scan unit bucket={unit.bucket_id} estimated decoded size "
+                f"exceeds LAKESOUL_MAX_UNIT_BYTES. Recreate the table with "
+                f"more hash buckets, compact the partition, or raise the limit."
+            )
         cache_key = None
         if self.use_cache:
             from .hbm_cache import scan_cache
@@ -399,8 +400,22 @@ class LakeSoulScan:
         needs_merge = bool(self.pk) and (
             len(unit.files) > 1 or self.cdc_column is not None or bool(self.merge_ops)
         )
-        if self.device == "cuda" and needs_merge and not self._gpu_merge_supported():
-            # hybrid: CPU merge, then ship the merged batch to HBM
+        if self.device == "cuda" and (
+            oversized or (needs_merge and not self._gpu_merge_supported())
+        ):
+            # hybrid: CPU decode+merge (host RAM), then ship the merged
+            # batch to HBM — for string-PK merges and for buckets whose
+            # working set would not fit the HBM budget (the merged output
+            # is far smaller than decode intermediates)
+            if oversized:
+                import warnings
+
+                warnings.warn(
+                    f"scan unit bucket={unit.bucket_id} exceeds "
+                    "LAKESOUL_MAX_UNIT_BYTES; falling back to host-side "
+                    "decode+merge for this bucket (slower). Compact or "
+                    "re-bucket the table to restore the GPU path."
+                )
             batch = self._read_unit_cpu(unit)
             return batch.to_device("cuda") if batch is not None else None
         if self.device == "cuda":

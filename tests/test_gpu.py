@@ -387,3 +387,32 @@ def test_gpu_decimal_mor(gpu_catalog):
     expect[::3] = 123456
     got = np.array([int(x.scaleb(2)) for x in df["amt"]])
     np.testing.assert_array_equal(got, expect)
+
+
+@pytest.mark.gpu
+def test_gpu_oversized_unit_fallback(gpu_catalog, monkeypatch):
+    """A bucket estimated over LAKESOUL_MAX_UNIT_BYTES must fall back to
+    host decode+merge and still return correct HBM-resident results."""
+    import warnings
+
+    from lakesoul_amd.io.schema import Field, Schema
+
+    t = gpu_catalog.create_table(
+        "big1",
+        Schema([Field("id", "int64", False), Field("v", "float64")]),
+        primary_keys=["id"],
+        hash_bucket_num=1,
+    )
+    n = 100000
+    t.upsert({"id": np.arange(n, dtype=np.int64), "v": np.zeros(n)})
+    t.upsert({"id": np.arange(0, n, 2, dtype=np.int64), "v": np.ones(n // 2)})
+    monkeypatch.setenv("LAKESOUL_MAX_UNIT_BYTES", "1000")
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        df = t.scan(device="cuda:0").to_arrow().to_pandas()
+    assert any("falling back" in str(x.message) for x in w)
+    df = df.sort_values("id").reset_index(drop=True)
+    assert len(df) == n
+    expect = np.zeros(n)
+    expect[::2] = 1.0
+    np.testing.assert_allclose(df["v"].to_numpy(), expect)
