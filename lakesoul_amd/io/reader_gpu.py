@@ -44,7 +44,7 @@ _TARGET = {"int8": torch.int8, "int16": torch.int16}
 
 
 def _torch_view(dtype):
-    return torch.int64 if dtype.startswith("decimal") else _torch_view(dtype)
+    return torch.int64 if dtype.startswith("decimal") else _TORCH_VIEW[dtype]
 _ESIZE = {torch.uint8: 1, torch.int32: 4, torch.int64: 8, torch.float32: 4, torch.float64: 8}
 
 
