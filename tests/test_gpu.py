@@ -364,12 +364,13 @@ def test_gpu_int32_single_pk(dev, tmp_path):
 
 
 @pytest.mark.gpu
-def test_gpu_decimal_mor(gpu_catalog):
+def test_gpu_decimal_mor(dev, tmp_path):
     """decimal column through the GPU MOR path (int64 unscaled on device)."""
     import decimal
 
     from lakesoul_amd.io.schema import Field, Schema
 
+    gpu_catalog = _mk_catalog(tmp_path)
     t = gpu_catalog.create_table(
         "gdec",
         Schema([Field("id", "int64", False), Field("amt", "decimal(12,2)")]),
@@ -390,13 +391,14 @@ def test_gpu_decimal_mor(gpu_catalog):
 
 
 @pytest.mark.gpu
-def test_gpu_oversized_unit_fallback(gpu_catalog, monkeypatch):
+def test_gpu_oversized_unit_fallback(dev, tmp_path, monkeypatch):
     """A bucket estimated over LAKESOUL_MAX_UNIT_BYTES must fall back to
     host decode+merge and still return correct HBM-resident results."""
     import warnings
 
     from lakesoul_amd.io.schema import Field, Schema
 
+    gpu_catalog = _mk_catalog(tmp_path)
     t = gpu_catalog.create_table(
         "big1",
         Schema([Field("id", "int64", False), Field("v", "float64")]),
