@@ -67,6 +67,7 @@ class LakeSoulCatalog:
             table_schema=schema_to_json(schema),
             properties=json.dumps(props),
             partitions=partitions,
+            domain=props.get("domain", "public"),
         )
         self.client.create_table(info)
         return LakeSoulTable(self.client, info)
