@@ -149,9 +149,19 @@ class SqliteMetaStore:
         conn = getattr(self._local, "conn", None)
         if conn is None:
             conn = sqlite3.connect(self.path, timeout=60.0)
-            conn.execute("PRAGMA journal_mode=WAL")
-            conn.execute("PRAGMA synchronous=NORMAL")
+            # busy_timeout FIRST: switching journal_mode takes a write lock
+            # and races with other ranks opening the same store
             conn.execute("PRAGMA busy_timeout=60000")
+            for attempt in range(20):
+                try:
+                    conn.execute("PRAGMA journal_mode=WAL")
+                    break
+                except sqlite3.OperationalError:
+                    import random
+                    import time as _time
+
+                    _time.sleep(random.uniform(0.01, 0.1))
+            conn.execute("PRAGMA synchronous=NORMAL")
             self._local.conn = conn
         return conn
 
