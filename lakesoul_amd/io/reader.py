@@ -131,7 +131,14 @@ class LakeSoulScan:
                 if c not in have:
                     self.eval_fields.append(self.schema.field(c))
         self.eval_schema = Schema(self.eval_fields)
-        self.device = device or table.io_config().resolve_device()
+        self.device = str(device or table.io_config().resolve_device())
+        if self.device.startswith("cuda"):
+            # normalize "cuda:N" -> select device N, compare as "cuda"
+            import torch as _torch
+
+            if ":" in self.device and _torch.cuda.is_available():
+                _torch.cuda.set_device(self.device)
+            self.device = "cuda"
         opts = dict(options or {})
         cfg_opt = opts.get("scan_cache", table.io_config().option("scan_cache", "0"))
         self.use_cache = str(cfg_opt) == "1"

@@ -258,9 +258,9 @@ def write_table_data(table, data, device: Optional[str] = None,
     else:
         batch = data
         schema = batch.schema
-    if device == "cuda":
+    if device is not None and str(device).startswith("cuda"):
         with timing.phase("w_h2d", sync_gpu=True):
-            batch = batch.to_device("cuda")
+            batch = batch.to_device(device)
 
     pk = table.primary_keys
     range_cols = table.range_keys

@@ -49,5 +49,5 @@ def barrier(device: Optional[str] = None):
 
     if dist.is_available() and dist.is_initialized():
         dist.barrier()
-    if (device == "cuda") or (device is None and torch.cuda.is_available()):
+    if (device is not None and str(device).startswith("cuda")) or (device is None and torch.cuda.is_available()):
         torch.cuda.synchronize()
