@@ -164,8 +164,28 @@ int64_t int_value(const DecodedColumn& dc, const ColumnDesc& cd, int64_t row) {
       return v;
     }
     default:
-      throw std::runtime_error("C reader merge supports integer primary keys");
+      throw std::runtime_error(
+          "C reader merge supports integer and string primary keys");
   }
+}
+
+// three-way PK value compare: integers numerically, BYTE_ARRAY (string)
+// byte-wise lexicographic — same order the writer sorted by
+int cmp_value(const DecodedColumn& da, const ColumnDesc& ca, int64_t ra,
+              const DecodedColumn& db, const ColumnDesc& cb, int64_t rb) {
+  if (ca.physical == PT_BYTE_ARRAY && cb.physical == PT_BYTE_ARRAY) {
+    int32_t a0 = da.offsets[(size_t)ra], a1 = da.offsets[(size_t)ra + 1];
+    int32_t b0 = db.offsets[(size_t)rb], b1 = db.offsets[(size_t)rb + 1];
+    size_t la = (size_t)(a1 - a0), lb = (size_t)(b1 - b0);
+    int c = std::memcmp(da.bytes.data() + a0, db.bytes.data() + b0,
+                        la < lb ? la : lb);
+    if (c != 0) return c < 0 ? -1 : 1;
+    if (la != lb) return la < lb ? -1 : 1;
+    return 0;
+  }
+  int64_t va = int_value(da, ca, ra);
+  int64_t vb = int_value(db, cb, rb);
+  return va < vb ? -1 : (va > vb ? 1 : 0);
 }
 
 }  // namespace
@@ -237,9 +257,9 @@ extern "C" int lakesoul_c_reader_start(LakesoulCReader* r) {
     if (!r->pks.empty()) {
       std::stable_sort(refs.begin(), refs.end(), [&](const RowRef& a, const RowRef& b) {
         for (size_t k : pk_idx) {
-          int64_t va = int_value(fcs[a.seq].cols[k], fcs[a.seq].descs[k], a.row);
-          int64_t vb = int_value(fcs[b.seq].cols[k], fcs[b.seq].descs[k], b.row);
-          if (va != vb) return va < vb;
+          int c = cmp_value(fcs[a.seq].cols[k], fcs[a.seq].descs[k], a.row,
+                            fcs[b.seq].cols[k], fcs[b.seq].descs[k], b.row);
+          if (c != 0) return c < 0;
         }
         if (a.seq != b.seq) return a.seq < b.seq;
         return a.row < b.row;
@@ -251,8 +271,10 @@ extern "C" int lakesoul_c_reader_start(LakesoulCReader* r) {
         if (!last) {
           bool same = true;
           for (size_t k : pk_idx) {
-            if (int_value(fcs[refs[i].seq].cols[k], fcs[refs[i].seq].descs[k], refs[i].row) !=
-                int_value(fcs[refs[i + 1].seq].cols[k], fcs[refs[i + 1].seq].descs[k], refs[i + 1].row)) {
+            if (cmp_value(fcs[refs[i].seq].cols[k], fcs[refs[i].seq].descs[k],
+                          refs[i].row,
+                          fcs[refs[i + 1].seq].cols[k],
+                          fcs[refs[i + 1].seq].descs[k], refs[i + 1].row) != 0) {
               same = false;
               break;
             }

@@ -130,3 +130,46 @@ def test_c_reader_merge_on_read(lib, tmp_path, catalog):
     assert len(got) == n
     assert (got["v"][::4] == 2.5).all()
     assert (got["v"][1::4] == 0).all()
+
+
+def test_capi_reader_string_pk(tmp_path, lib):
+    """C ABI MOR merge with a string primary key (byte-lexicographic
+    ordering, UseLast dedup across files)."""
+    import pyarrow.parquet as pq
+
+    f1 = str(tmp_path / "a.parquet")
+    f2 = str(tmp_path / "b.parquet")
+    # each file writer-sorted by pk, file order = commit order
+    pq.write_table(pa.table({
+        "k": pa.array(["apple", "kiwi", "pear"], pa.string()),
+        "v": pa.array([1.0, 2.0, 3.0], pa.float64()),
+    }), f1, use_dictionary=False, compression="zstd")
+    pq.write_table(pa.table({
+        "k": pa.array(["banana", "kiwi"], pa.string()),
+        "v": pa.array([10.0, 20.0], pa.float64()),
+    }), f2, use_dictionary=False, compression="zstd")
+
+    r = lib.lakesoul_c_reader_create()
+    for f in (f1, f2):
+        lib.lakesoul_c_reader_add_file(ctypes.c_void_p(r), f.encode())
+    lib.lakesoul_c_reader_add_primary_key(ctypes.c_void_p(r), b"k")
+    assert lib.lakesoul_c_reader_start(ctypes.c_void_p(r)) == 0, \
+        lib.lakesoul_c_last_error()
+
+    schema_holder = ctypes.create_string_buffer(72)
+    assert lib.lakesoul_c_reader_schema(ctypes.c_void_p(r), ctypes.addressof(schema_holder)) == 0
+    schema = pa.Schema._import_from_c(ctypes.addressof(schema_holder))
+    rows = []
+    while True:
+        arr_holder = ctypes.create_string_buffer(80)
+        rc = lib.lakesoul_c_reader_next(ctypes.c_void_p(r), ctypes.addressof(arr_holder))
+        assert rc >= 0, lib.lakesoul_c_last_error()
+        if rc == 0:
+            break
+        sa = pa.Array._import_from_c(ctypes.addressof(arr_holder),
+                                     pa.struct(list(schema)))
+        rows.append(pa.RecordBatch.from_struct_array(sa))
+    lib.lakesoul_c_reader_close(ctypes.c_void_p(r))
+    got = pa.Table.from_batches(rows).to_pandas()
+    assert got["k"].tolist() == ["apple", "banana", "kiwi", "pear"]
+    assert got["v"].tolist() == [1.0, 10.0, 20.0, 3.0]
