@@ -84,6 +84,12 @@ class ThreadPool {
       }
       std::fclose(f);
     }
+    // one process per GPU shares the node's quota: divide by the rank
+    // count so 8 ranks don't each spin up a full-quota pool and thrash
+    if (const char* ws = std::getenv("WORLD_SIZE")) {
+      int w = atoi(ws);
+      if (w > 1) hw = hw / w;
+    }
     return hw;
   }
 
