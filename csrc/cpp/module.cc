@@ -13,6 +13,7 @@
 #include "murmur3.h"
 #include "parquet_file.h"
 #include "read_unit.h"
+#include "zstd_dec.h"
 
 namespace py = pybind11;
 using namespace lakesoul;
@@ -632,6 +633,29 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("read_unit_raw", &read_unit_raw_py, py::arg("paths"), py::arg("names"),
         py::arg("nthreads") = 0, py::arg("pin") = true, py::arg("gpu_snappy") = false);
   m.def("read_chunks_raw_batch", &read_chunks_raw_batch);
+  m.def("zstd_compress_ref", [](py::bytes src, int64_t level) {
+    std::string b = src;
+    auto out = zstd_compress((const uint8_t*)b.data(), b.size(), (int)level);
+    return py::bytes((const char*)out.data(), out.size());
+  });
+  m.def("zstd_decompress_ref", [](py::bytes src, int64_t cap) {
+    std::string b = src;
+    std::vector<uint8_t> out((size_t)cap);
+    zstd_decompress_into(out.data(), (size_t)cap, (const uint8_t*)b.data(),
+                         b.size());
+    return py::bytes((const char*)out.data(), (size_t)cap);
+  });
+  m.def("zstd_decode_ref", [](py::bytes src, int64_t cap) {
+    // differential-test hook for the from-scratch zstd decoder
+    std::string b = src;
+    std::vector<uint8_t> out((size_t)cap);
+    static thread_local std::unique_ptr<lszstd::Ctx> ctx;
+    if (!ctx) ctx.reset(new lszstd::Ctx());
+    int64_t n = lszstd::decode((const uint8_t*)b.data(), (int64_t)b.size(),
+                               out.data(), cap, ctx.get());
+    if (n < 0) throw std::runtime_error("lszstd decode failed");
+    return py::bytes((const char*)out.data(), (size_t)n);
+  });
   m.def("hash_columns_cpu", &hash_columns_cpu);
   m.def("hash_string_column_cpu", &hash_string_column_cpu);
   m.def("bucket_ids_from_hashes", &bucket_ids_from_hashes);
