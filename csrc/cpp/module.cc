@@ -391,13 +391,14 @@ static py::tuple prep_rle_runs(torch::Tensor values, torch::Tensor idx_pages) {
 // memory for the big buffers (GPU boxes).
 static py::dict read_unit_raw_py(const std::vector<std::string>& paths,
                                  const std::vector<std::string>& names,
-                                 int64_t nthreads, bool pin, bool gpu_snappy) {
+                                 int64_t nthreads, bool pin, bool gpu_snappy,
+                                 bool gpu_zstd) {
   (void)nthreads;
   std::unique_ptr<UnitStage> st;
   auto t0 = std::chrono::steady_clock::now();
   {
     py::gil_scoped_release rel;
-    st = read_unit_stage1(paths, names, gpu_snappy);
+    st = read_unit_stage1(paths, names, gpu_snappy, gpu_zstd);
   }
   auto t1 = std::chrono::steady_clock::now();
   UnitStage& ud = *st;
@@ -422,6 +423,9 @@ static py::dict read_unit_raw_py(const std::vector<std::string>& paths,
   torch::Tensor sjobs = torch::empty({(int64_t)ud.snappy_jobs.size()}, torch::kInt64);
   if (!ud.snappy_jobs.empty())
     std::memcpy(sjobs.data_ptr(), ud.snappy_jobs.data(), ud.snappy_jobs.size() * 8);
+  torch::Tensor zjobs = torch::empty({(int64_t)ud.zstd_jobs.size()}, torch::kInt64);
+  if (!ud.zstd_jobs.empty())
+    std::memcpy(zjobs.data_ptr(), ud.zstd_jobs.data(), ud.zstd_jobs.size() * 8);
   auto t2 = std::chrono::steady_clock::now();
   {
     py::gil_scoped_release rel;
@@ -447,6 +451,7 @@ static py::dict read_unit_raw_py(const std::vector<std::string>& paths,
   d["soffs"] = soffs;
   d["comp"] = comp;
   d["snappy_jobs"] = sjobs;
+  d["zstd_jobs"] = zjobs;
   py::list frows;
   for (auto r : ud.file_rows) frows.append(r);
   d["file_rows"] = frows;
@@ -631,7 +636,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("read_chunks_cpu_batch", &read_chunks_cpu_batch);
   m.def("prep_rle_runs", &prep_rle_runs);
   m.def("read_unit_raw", &read_unit_raw_py, py::arg("paths"), py::arg("names"),
-        py::arg("nthreads") = 0, py::arg("pin") = true, py::arg("gpu_snappy") = false);
+        py::arg("nthreads") = 0, py::arg("pin") = true,
+        py::arg("gpu_snappy") = false, py::arg("gpu_zstd") = false);
   m.def("read_chunks_raw_batch", &read_chunks_raw_batch);
   m.def("zstd_compress_ref", [](py::bytes src, int64_t level) {
     std::string b = src;

@@ -365,11 +365,12 @@ class ParquetFile {
     int32_t bit_width;
   };
 
-  struct CompPage {  // GPU-decompress job (snappy, levels-free pages)
+  struct CompPage {  // GPU-decompress job (snappy/zstd, levels-free pages)
     int64_t comp_off;   // into comp buffer
     int64_t comp_len;
     int64_t out_off;    // into the chunk's values stream
     int64_t out_len;
+    int32_t codec = CODEC_SNAPPY;
   };
 
   struct DeferPage {  // host direct-decompress job (zstd/uncompressed,
@@ -405,11 +406,11 @@ class ParquetFile {
 
   const uint8_t* data_at(int64_t off) const { return map_ + off; }
 
-  // gpu_snappy: defer SNAPPY page decompression to the GPU kernel when
-  // the column is REQUIRED (no def-level section inside the compressed
-  // blob), PLAIN-encoded, non-boolean v1 pages.
+  // gpu_snappy / gpu_zstd: defer page decompression to the GPU kernels
+  // when the column is REQUIRED (no def-level section inside the
+  // compressed blob), PLAIN-encoded, non-boolean v1 pages.
   ChunkData read_chunk(size_t rg, size_t col, bool gpu_snappy = false,
-                       bool defer_host = false) const {
+                       bool defer_host = false, bool gpu_zstd = false) const {
     const RowGroup& g = meta_.row_groups.at(rg);
     const ColumnMeta& cm = g.columns.at(col);
     const ColumnDesc& cd = cols_.at(col);
@@ -453,10 +454,28 @@ class ParquetFile {
       if (ph.type != PAGE_DATA && ph.type != PAGE_DATA_V2) continue;
 
       if (out.gpu_compressed &&
-          !(cm.codec == CODEC_SNAPPY && ph.type == PAGE_DATA &&
-            ph.encoding == ENC_PLAIN)) {
+          !((cm.codec == CODEC_SNAPPY || cm.codec == CODEC_ZSTD) &&
+            ph.type == PAGE_DATA && ph.encoding == ENC_PLAIN)) {
         // mixed chunk (e.g. dict fallback) — redo fully on host
         return read_chunk(rg, col, false);
+      }
+      if (((gpu_snappy && cm.codec == CODEC_SNAPPY) ||
+           (gpu_zstd && cm.codec == CODEC_ZSTD)) &&
+          ph.type == PAGE_DATA && ph.encoding == ENC_PLAIN && !cd.nullable &&
+          cd.physical != PT_BOOLEAN && cd.physical != PT_BYTE_ARRAY &&
+          out.dict.empty() && out.values.empty() && !out.host_deferred) {
+        out.gpu_compressed = true;
+        CompPage cp;
+        cp.comp_off = (int64_t)out.comp.size();
+        cp.comp_len = ph.compressed_size;
+        cp.out_off = out.values_len;
+        cp.out_len = ph.uncompressed_size;
+        cp.codec = cm.codec;
+        out.comp.insert(out.comp.end(), body, body + ph.compressed_size);
+        out.comp_pages.push_back(cp);
+        out.values_len += ph.uncompressed_size;
+        values_seen += ph.num_values;
+        continue;
       }
       if (defer_host &&
           (cm.codec == CODEC_ZSTD || cm.codec == CODEC_UNCOMPRESSED) &&
@@ -481,22 +500,6 @@ class ParquetFile {
           !((cm.codec == CODEC_ZSTD || cm.codec == CODEC_UNCOMPRESSED) &&
             ph.type == PAGE_DATA && ph.encoding == ENC_PLAIN)) {
         return read_chunk(rg, col, false, false);  // mixed: redo staged
-      }
-      if (gpu_snappy && cm.codec == CODEC_SNAPPY && ph.type == PAGE_DATA &&
-          ph.encoding == ENC_PLAIN && !cd.nullable &&
-          cd.physical != PT_BOOLEAN && cd.physical != PT_BYTE_ARRAY &&
-          out.dict.empty() && out.values.empty()) {
-        out.gpu_compressed = true;
-        CompPage cp;
-        cp.comp_off = (int64_t)out.comp.size();
-        cp.comp_len = ph.compressed_size;
-        cp.out_off = out.values_len;
-        cp.out_len = ph.uncompressed_size;
-        out.comp.insert(out.comp.end(), body, body + ph.compressed_size);
-        out.comp_pages.push_back(cp);
-        out.values_len += ph.uncompressed_size;
-        values_seen += ph.num_values;
-        continue;
       }
 
       std::vector<uint8_t> page;

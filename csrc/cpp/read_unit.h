@@ -126,6 +126,7 @@ struct UnitStage {
   std::vector<int64_t> file_rows;
   std::vector<int64_t> runs;  // [m][6]
   std::vector<int64_t> snappy_jobs;  // [m][4]: comp_off, comp_len, dst_off(values), dst_len
+  std::vector<int64_t> zstd_jobs;    // [m][4]: same layout, zstd frames
   struct HostJob {
     int file_idx;
     int64_t file_off, comp_len, dst_off, dst_len;
@@ -139,7 +140,7 @@ struct UnitStage {
 
 inline std::unique_ptr<UnitStage> read_unit_stage1(
     const std::vector<std::string>& paths, const std::vector<std::string>& names,
-    bool gpu_snappy = false) {
+    bool gpu_snappy = false, bool gpu_zstd = false) {
   auto st = std::make_unique<UnitStage>();
   UnitStage& S = *st;
   auto now = [] { return std::chrono::steady_clock::now(); };
@@ -181,7 +182,7 @@ inline std::unique_ptr<UnitStage> read_unit_stage1(
         const Task& t = tasks[i];
         auto& fd = S.files[t.fi];
         fd.chunks[t.c][t.rg] =
-            fd.f->read_chunk(t.rg, fd.col_idx[t.c], gpu_snappy, true);
+            fd.f->read_chunk(t.rg, fd.col_idx[t.c], gpu_snappy, true, gpu_zstd);
       } catch (std::exception& e) {
         std::lock_guard<std::mutex> lk(err_mu);
         err = e.what();
@@ -274,10 +275,11 @@ inline std::unique_ptr<UnitStage> read_unit_stage1(
         for (auto& ch : chs) {
           if (ch.gpu_compressed) {
             for (auto& cp : ch.comp_pages) {
-              S.snappy_jobs.push_back(S.comp_size + cp.comp_off);
-              S.snappy_jobs.push_back(cp.comp_len);
-              S.snappy_jobs.push_back(vpos + plen + cp.out_off);
-              S.snappy_jobs.push_back(cp.out_len);
+              auto& jobs = (cp.codec == CODEC_ZSTD) ? S.zstd_jobs : S.snappy_jobs;
+              jobs.push_back(S.comp_size + cp.comp_off);
+              jobs.push_back(cp.comp_len);
+              jobs.push_back(vpos + plen + cp.out_off);
+              jobs.push_back(cp.out_len);
             }
             S.comp_size += (int64_t)ch.comp.size();
             plen += ch.values_len;

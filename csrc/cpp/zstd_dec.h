@@ -69,13 +69,25 @@ struct BitBwd {
     if (n == 0) return 0;
     bitpos -= n;
     int64_t lo = bitpos;
+    if (bitpos < 0) overflow = true;
+    if (lo >= 0) {
+      // fast path: unaligned LE u64 window ENDING at the cursor byte so
+      // the load never overruns the buffer end (bits exist => bytes do)
+      int64_t hi_byte = (lo + n + 7) >> 3;
+      if (hi_byte >= 8) {
+        uint64_t w;
+        memcpy(&w, buf + hi_byte - 8, 8);
+        int shift = (int)(lo - (hi_byte - 8) * 8);
+        return (uint32_t)((w >> shift) &
+                          ((n == 32) ? 0xFFFFFFFFu : ((1u << n) - 1)));
+      }
+    }
     if (lo < -63) lo = -63;
     uint32_t out = 0;
     for (int i = 0; i < n; i++) {
       int64_t b = lo + i;
       if (b >= 0) out |= (uint32_t)((buf[b >> 3] >> (b & 7)) & 1) << i;
     }
-    if (bitpos < 0) overflow = true;
     return out;
   }
 
@@ -121,11 +133,11 @@ struct FseEntry {
   uint16_t newState;  // baseline; nextState = newState + read(nbBits)
 };
 
-struct FseTable {
+struct FseTable {  // no default initializers: instances live in LDS
   FseEntry e[1 << kMaxTableLog];
-  int tableLog = 0;
-  bool rle = false;     // degenerate: single symbol, no bits
-  uint8_t rleSym = 0;
+  int tableLog;
+  bool rle;          // degenerate: single symbol, no bits
+  uint8_t rleSym;
 };
 
 // read normalized counts (RFC 8878 4.1.1) from a forward bitstream.
@@ -293,9 +305,9 @@ struct HufEntry {
   uint8_t nbBits;
 };
 
-struct HufTable {
+struct HufTable {  // no default initializers: instances live in LDS
   HufEntry e[1 << kHufMaxBits];
-  int maxBits = 0;
+  int maxBits;
 };
 
 // build decode table from weights[0..nsym-1] (HUF_readDTableX1 layout)

@@ -37,6 +37,9 @@ void launch_segmented_sum(const T*, const int64_t*, const uint8_t*, const uint8_
 void launch_segmented_last(const int64_t*, const uint8_t*, const uint8_t*, int64_t*, int64_t, hipStream_t);
 void launch_ann_scores(const short*, const short*, float*, int64_t, int32_t, int32_t, hipStream_t);
 void launch_snappy_decompress(const uint8_t*, const int64_t*, int64_t, uint8_t*, int32_t*, hipStream_t);
+void launch_zstd_decompress(const uint8_t*, const int64_t*, int64_t, uint8_t*,
+                            uint8_t*, int64_t, int32_t*, hipStream_t);
+int64_t lsz_gpu_litbuf_bytes();
 
 }  // namespace lakesoul
 
@@ -598,8 +601,27 @@ static torch::Tensor snappy_decompress_into(torch::Tensor src, torch::Tensor job
   return status;
 }
 
+// GPU zstd: one wave per page, scratch literals buffer per block slot
+static torch::Tensor zstd_decompress_into(torch::Tensor src, torch::Tensor jobs,
+                                          torch::Tensor dst) {
+  CHECK_GPU(src);
+  CHECK_GPU(jobs);
+  CHECK_GPU(dst);
+  int64_t njobs = jobs.size(0);
+  int64_t nblocks = njobs < 2048 ? njobs : 2048;
+  auto scratch = torch::empty({nblocks * lsz_gpu_litbuf_bytes()},
+                              src.options().dtype(torch::kUInt8));
+  auto status = torch::empty({njobs}, src.options().dtype(torch::kInt32));
+  launch_zstd_decompress(src.data_ptr<uint8_t>(), jobs.data_ptr<int64_t>(),
+                         njobs, dst.data_ptr<uint8_t>(),
+                         scratch.data_ptr<uint8_t>(), nblocks,
+                         status.data_ptr<int32_t>(), cur_stream());
+  return status;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ann_scores", &ann_scores);
+  m.def("zstd_decompress_into", &zstd_decompress_into);
   m.def("snappy_decompress", &snappy_decompress);
   m.def("snappy_decompress_into", &snappy_decompress_into);
   m.def("scan_unit_uselast", &scan_unit_uselast);

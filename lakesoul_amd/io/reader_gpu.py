@@ -48,11 +48,16 @@ def _torch_view(dtype):
 _ESIZE = {torch.uint8: 1, torch.int32: 4, torch.int64: 8, torch.float32: 4, torch.float64: 8}
 
 
+import os as _os
+
+_GPU_ZSTD = _os.environ.get("LAKESOUL_GPU_ZSTD", "1") != "0"
+
+
 def fetch_raw(files: List[str], names: List[str]) -> dict:
     """Host phase of a unit read (releases the GIL in C++) — safe to run
     on a prefetch thread while the GPU processes the previous unit."""
     with timing.phase("host_fetch"):
-        raw = cpp().read_unit_raw(files, names, 0, True, True)
+        raw = cpp().read_unit_raw(files, names, 0, True, True, _GPU_ZSTD)
     if timing.ENABLED:
         timing._acc["fetch.stage1"] += raw["t_stage1_us"] / 1e6
         timing._acc["fetch.s1_open"] += raw["t_open_us"] / 1e6
@@ -82,7 +87,33 @@ class UnitTransfer:
 
     def __init__(self, raw: dict, device):
         self.raw = raw  # hold pinned host buffers until consumed
-        self.vals = raw["values"].to(device, non_blocking=True)
+        # values regions covered by GPU decompress jobs carry no host
+        # data — ship only the host-filled gaps (for all-zstd units the
+        # H2D volume drops to the compressed bytes)
+        jobs_host = []
+        for k in ("snappy_jobs", "zstd_jobs"):
+            t = raw.get(k)
+            if t is not None and t.numel():
+                jobs_host.append(t.view(-1, 4))
+        nvals = raw["values"].numel()
+        if jobs_host and nvals:
+            import torch as _t
+
+            j = _t.cat(jobs_host) if len(jobs_host) > 1 else jobs_host[0]
+            dst = j[:, 2].numpy()
+            ln = j[:, 3].numpy()
+            order = dst.argsort()
+            self.vals = _t.empty(nvals, dtype=_t.uint8, device=device)
+            hv = raw["values"]
+            pos = 0
+            for off, l in zip(dst[order], ln[order]):
+                if off > pos:
+                    self.vals[pos:off].copy_(hv[pos:off], non_blocking=True)
+                pos = max(pos, int(off + l))
+            if pos < nvals:
+                self.vals[pos:nvals].copy_(hv[pos:nvals], non_blocking=True)
+        else:
+            self.vals = raw["values"].to(device, non_blocking=True)
         self.validity = (
             raw["validity"].to(device, non_blocking=True)
             if raw["validity"].numel() else None
@@ -97,6 +128,11 @@ class UnitTransfer:
         self.snappy_jobs = (
             raw["snappy_jobs"].view(-1, 4).to(device, non_blocking=True)
             if raw.get("snappy_jobs") is not None and raw["snappy_jobs"].numel()
+            else None
+        )
+        self.zstd_jobs = (
+            raw["zstd_jobs"].view(-1, 4).to(device, non_blocking=True)
+            if raw.get("zstd_jobs") is not None and raw["zstd_jobs"].numel()
             else None
         )
         self.event = None
@@ -128,6 +164,14 @@ def read_unit_gpu(scan, unit, raw: Optional[dict] = None,
         status = hip().snappy_decompress_into(transfer.comp, transfer.snappy_jobs, vals)
         if bool((status != 0).any()):
             raise RuntimeError(f"GPU snappy decompression failed: {status.cpu().tolist()}")
+    if transfer.zstd_jobs is not None:
+        # GPU zstd: the from-scratch RFC 8878 decoder (csrc/hip/zstd.hip)
+        # decompresses zstd page frames straight into the values buffer —
+        # compressed bytes crossed the bus, the cgroup-capped host CPUs
+        # never touched them
+        status = hip().zstd_decompress_into(transfer.comp, transfer.zstd_jobs, vals)
+        if bool((status != 0).any()):
+            raise RuntimeError(f"GPU zstd decompression failed: status={int((status != 0).sum())} pages")
 
     read_schema = scan.schema.select(names)
     ncols = len(names)

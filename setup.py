@@ -44,6 +44,7 @@ if _WITH_HIP:
                 os.path.join("csrc", "hip", "kernels.hip"),
                 os.path.join("csrc", "hip", "ann.hip"),
                 os.path.join("csrc", "hip", "snappy.hip"),
+                os.path.join("csrc", "hip", "zstd.hip"),
             ],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
