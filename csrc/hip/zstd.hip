@@ -436,6 +436,8 @@ __global__ void __launch_bounds__(LANES) zstd_pages_kernel(
 
 }  // namespace lsz_gpu
 
+namespace lakesoul {
+
 void launch_zstd_decompress(const uint8_t* src, const int64_t* jobs,
                             int64_t njobs, uint8_t* dst, uint8_t* scratch,
                             int64_t nblocks, int32_t* status,
@@ -447,3 +449,5 @@ void launch_zstd_decompress(const uint8_t* src, const int64_t* jobs,
 }
 
 int64_t lsz_gpu_litbuf_bytes() { return lsz_gpu::kLitBufCap; }
+
+}  // namespace lakesoul
