@@ -157,47 +157,25 @@ class ParquetWriter {
       const ColumnDesc& cd = cols_[ci];
       const ColumnData& col = data[ci];
 
-      // ---- assemble page payload (def levels + PLAIN values) ----
-      std::vector<uint8_t> payload;
-      const uint8_t* validity =
-          cd.nullable && col.validity ? col.validity + row_off : nullptr;
-      int64_t null_count = 0;
-      if (cd.nullable) {
-        std::vector<uint8_t> levels = encode_def_levels(validity, n);
-        uint32_t lv_len = (uint32_t)levels.size();
-        payload.insert(payload.end(), (uint8_t*)&lv_len, (uint8_t*)&lv_len + 4);
-        payload.insert(payload.end(), levels.begin(), levels.end());
-        if (validity)
-          for (int64_t i = 0; i < n; i++) null_count += validity[i] ? 0 : 1;
+      // page split: target ~LAKESOUL_PAGE_BYTES decompressed bytes per
+      // page (default 128 KB) so the GPU zstd kernel gets thousands of
+      // pages in flight per scan unit instead of a few multi-MB frames.
+      // Strings keep one page per chunk (host-assembled path).
+      static const int64_t kPageBytes = []() {
+        const char* e = std::getenv("LAKESOUL_PAGE_BYTES");
+        int64_t v = e ? atoll(e) : 0;
+        return v > 0 ? v : (int64_t)(128 << 10);
+      }();
+      int64_t page_rows = n;
+      if (cd.physical != PT_BYTE_ARRAY) {
+        int es = physical_elem_size(cd.physical);
+        if (es < 1) es = 1;
+        page_rows = kPageBytes / es;
+        if (page_rows < 1) page_rows = 1;
       }
 
       Statistics stats;
-      stats.null_count = cd.nullable ? null_count : 0;
-      append_plain_values(cd, col, row_off, n, validity, payload, stats);
-
-      // ---- compress ----
-      std::vector<uint8_t> compressed;
-      const uint8_t* body = payload.data();
-      size_t body_n = payload.size();
-      if (codec_ == CODEC_ZSTD) {
-        compressed = zstd_compress(payload.data(), payload.size(), level_);
-        body = compressed.data();
-        body_n = compressed.size();
-      } else if (codec_ != CODEC_UNCOMPRESSED) {
-        throw std::runtime_error("writer supports zstd/uncompressed only");
-      }
-
-      // ---- page header + emit ----
-      PageHeader ph;
-      ph.type = PAGE_DATA;
-      ph.uncompressed_size = (int32_t)payload.size();
-      ph.compressed_size = (int32_t)body_n;
-      ph.num_values = (int32_t)n;
-      ph.encoding = ENC_PLAIN;
-      ph.def_encoding = ENC_RLE;
-      ph.rep_encoding = ENC_RLE;
-      auto ph_bytes = serialize_page_header(ph);
-
+      stats.null_count = 0;
       ColumnMeta cm;
       cm.type = cd.physical;
       cm.encodings = {ENC_PLAIN, ENC_RLE};
@@ -205,14 +183,54 @@ class ParquetWriter {
       cm.codec = codec_;
       cm.num_values = n;
       cm.data_page_offset = pos_;
-      cm.total_uncompressed_size = (int64_t)(ph_bytes.size() + payload.size());
-      cm.total_compressed_size = (int64_t)(ph_bytes.size() + body_n);
+
+      for (int64_t poff = 0; poff < n || n == 0; poff += page_rows) {
+        int64_t pn = n - poff < page_rows ? n - poff : page_rows;
+        // ---- assemble page payload (def levels + PLAIN values) ----
+        std::vector<uint8_t> payload;
+        const uint8_t* validity =
+            cd.nullable && col.validity ? col.validity + row_off + poff : nullptr;
+        if (cd.nullable) {
+          std::vector<uint8_t> levels = encode_def_levels(validity, pn);
+          uint32_t lv_len = (uint32_t)levels.size();
+          payload.insert(payload.end(), (uint8_t*)&lv_len, (uint8_t*)&lv_len + 4);
+          payload.insert(payload.end(), levels.begin(), levels.end());
+          if (validity)
+            for (int64_t i = 0; i < pn; i++) stats.null_count += validity[i] ? 0 : 1;
+        }
+        append_plain_values(cd, col, row_off + poff, pn, validity, payload, stats);
+
+        // ---- compress ----
+        std::vector<uint8_t> compressed;
+        const uint8_t* body = payload.data();
+        size_t body_n = payload.size();
+        if (codec_ == CODEC_ZSTD) {
+          compressed = zstd_compress(payload.data(), payload.size(), level_);
+          body = compressed.data();
+          body_n = compressed.size();
+        } else if (codec_ != CODEC_UNCOMPRESSED) {
+          throw std::runtime_error("writer supports zstd/uncompressed only");
+        }
+
+        // ---- page header + emit ----
+        PageHeader ph;
+        ph.type = PAGE_DATA;
+        ph.uncompressed_size = (int32_t)payload.size();
+        ph.compressed_size = (int32_t)body_n;
+        ph.num_values = (int32_t)pn;
+        ph.encoding = ENC_PLAIN;
+        ph.def_encoding = ENC_RLE;
+        ph.rep_encoding = ENC_RLE;
+        auto ph_bytes = serialize_page_header(ph);
+        cm.total_uncompressed_size += (int64_t)(ph_bytes.size() + payload.size());
+        cm.total_compressed_size += (int64_t)(ph_bytes.size() + body_n);
+        fwrite_all(ph_bytes.data(), ph_bytes.size());
+        fwrite_all(body, body_n);
+        if (n == 0) break;
+      }
       cm.stats = stats;
       rg.columns.push_back(cm);
       rg.total_byte_size += cm.total_uncompressed_size;
-
-      fwrite_all(ph_bytes.data(), ph_bytes.size());
-      fwrite_all(body, body_n);
     }
     row_groups_.push_back(std::move(rg));
     total_rows_ += n;
@@ -454,13 +472,16 @@ class ParquetFile {
       if (ph.type != PAGE_DATA && ph.type != PAGE_DATA_V2) continue;
 
       if (out.gpu_compressed &&
-          !((cm.codec == CODEC_SNAPPY || cm.codec == CODEC_ZSTD) &&
+          !((cm.codec == CODEC_SNAPPY ||
+             (cm.codec == CODEC_ZSTD &&
+              ph.uncompressed_size <= (512 << 10))) &&
             ph.type == PAGE_DATA && ph.encoding == ENC_PLAIN)) {
-        // mixed chunk (e.g. dict fallback) — redo fully on host
+        // mixed chunk (dict fallback / oversized page) — redo on host
         return read_chunk(rg, col, false);
       }
       if (((gpu_snappy && cm.codec == CODEC_SNAPPY) ||
-           (gpu_zstd && cm.codec == CODEC_ZSTD)) &&
+           (gpu_zstd && cm.codec == CODEC_ZSTD &&
+            ph.uncompressed_size <= (512 << 10))) &&
           ph.type == PAGE_DATA && ph.encoding == ENC_PLAIN && !cd.nullable &&
           cd.physical != PT_BOOLEAN && cd.physical != PT_BYTE_ARRAY &&
           out.dict.empty() && out.values.empty() && !out.host_deferred) {
