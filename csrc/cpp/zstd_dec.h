@@ -51,44 +51,51 @@ LSZ_HD inline int highbit32(uint32_t v) {
 
 struct BitBwd {
   const uint8_t* buf;
-  int64_t bitpos;   // bits remaining below the cursor
+  int64_t bitpos;    // bits remaining below the cursor
+  uint64_t cont;     // register window: stream bits [contEnd-64, contEnd)
+  int64_t contEnd;   // byte-aligned top of the window (bit index)
   bool overflow;
+
+  LSZ_HD void refill() {
+    // place an 8-byte window ending at the cursor byte; never reads past
+    // the buffer end (bits exist => bytes do) and zero-pads small bufs
+    int64_t byteHi = (bitpos + 7) >> 3;
+    if (byteHi >= 8) {
+      memcpy(&cont, buf + byteHi - 8, 8);
+      contEnd = byteHi * 8;
+    } else {
+      uint64_t w = 0;
+      for (int64_t i = 0; i < byteHi; i++) w |= (uint64_t)buf[i] << (8 * i);
+      contEnd = 64;          // treat bits [0,64); bits >= byteHi*8 unused
+      cont = w;
+    }
+  }
 
   LSZ_HD bool init(const uint8_t* p, int64_t n) {
     buf = p;
     overflow = false;
+    cont = 0;
+    contEnd = 0;
     if (n <= 0) { bitpos = 0; overflow = true; return false; }
     uint8_t last = p[n - 1];
     if (last == 0) { bitpos = 0; overflow = true; return false; }
     bitpos = (n - 1) * 8 + highbit32(last);  // sentinel bit excluded
+    refill();
     return true;
   }
 
   // read n bits [bitpos-n, bitpos); zero-padded when under-running
   LSZ_HD uint32_t read(int n) {
     if (n == 0) return 0;
+    if (bitpos - n < contEnd - 64) refill();
     bitpos -= n;
-    int64_t lo = bitpos;
+    uint64_t mask = (n >= 32) ? 0xFFFFFFFFull : ((1ull << n) - 1);
+    int64_t shift = bitpos - (contEnd - 64);
     if (bitpos < 0) overflow = true;
-    if (lo >= 0) {
-      // fast path: unaligned LE u64 window ENDING at the cursor byte so
-      // the load never overruns the buffer end (bits exist => bytes do)
-      int64_t hi_byte = (lo + n + 7) >> 3;
-      if (hi_byte >= 8) {
-        uint64_t w;
-        memcpy(&w, buf + hi_byte - 8, 8);
-        int shift = (int)(lo - (hi_byte - 8) * 8);
-        return (uint32_t)((w >> shift) &
-                          ((n == 32) ? 0xFFFFFFFFu : ((1u << n) - 1)));
-      }
-    }
-    if (lo < -63) lo = -63;
-    uint32_t out = 0;
-    for (int i = 0; i < n; i++) {
-      int64_t b = lo + i;
-      if (b >= 0) out |= (uint32_t)((buf[b >> 3] >> (b & 7)) & 1) << i;
-    }
-    return out;
+    if (shift >= 0) return (uint32_t)((cont >> shift) & mask);
+    // under-run: stream bits below 0 read as zeros
+    if (shift <= -64) return 0;
+    return (uint32_t)((cont << (-shift)) & mask);
   }
 
   LSZ_HD bool done() const { return bitpos <= 0; }
