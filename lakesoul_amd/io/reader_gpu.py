@@ -99,19 +99,21 @@ class UnitTransfer:
         if jobs_host and nvals:
             import torch as _t
 
+            import numpy as _np
+
             j = _t.cat(jobs_host) if len(jobs_host) > 1 else jobs_host[0]
             dst = j[:, 2].numpy()
             ln = j[:, 3].numpy()
             order = dst.argsort()
+            dst, ln = dst[order], ln[order]
+            ends = _np.maximum.accumulate(dst + ln)
+            gap_starts = _np.concatenate(([0], ends))
+            gap_ends = _np.concatenate((dst, [nvals]))
+            keep = gap_ends > gap_starts
             self.vals = _t.empty(nvals, dtype=_t.uint8, device=device)
             hv = raw["values"]
-            pos = 0
-            for off, l in zip(dst[order], ln[order]):
-                if off > pos:
-                    self.vals[pos:off].copy_(hv[pos:off], non_blocking=True)
-                pos = max(pos, int(off + l))
-            if pos < nvals:
-                self.vals[pos:nvals].copy_(hv[pos:nvals], non_blocking=True)
+            for a, b in zip(gap_starts[keep], gap_ends[keep]):
+                self.vals[a:b].copy_(hv[a:b], non_blocking=True)
         else:
             self.vals = raw["values"].to(device, non_blocking=True)
         self.validity = (
