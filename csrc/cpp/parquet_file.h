@@ -158,13 +158,13 @@ class ParquetWriter {
       const ColumnData& col = data[ci];
 
       // page split: target ~LAKESOUL_PAGE_BYTES decompressed bytes per
-      // page (default 128 KB) so the GPU zstd kernel gets thousands of
+      // page (default 32 KB) so the GPU zstd kernel gets thousands of
       // pages in flight per scan unit instead of a few multi-MB frames.
       // Strings keep one page per chunk (host-assembled path).
       static const int64_t kPageBytes = []() {
         const char* e = std::getenv("LAKESOUL_PAGE_BYTES");
         int64_t v = e ? atoll(e) : 0;
-        return v > 0 ? v : (int64_t)(128 << 10);
+        return v > 0 ? v : (int64_t)(32 << 10);
       }();
       int64_t page_rows = n;
       if (cd.physical != PT_BYTE_ARRAY) {

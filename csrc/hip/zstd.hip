@@ -67,6 +67,10 @@ struct LdsCtx {
   FseTable ll, of, ml;
   HufTable huf;
   int haveLl, haveOf, haveMl, haveHuf;
+  int64_t scratch_i64;     // lane0 -> wave value handoff
+  int16_t counts[64];      // ncount output (seq tables)
+  uint8_t weights[256];    // huffman weights
+  int scratch_i32;
 };
 
 // decode the literals section. All lanes in lockstep; table builds by
@@ -139,32 +143,36 @@ __device__ inline int64_t dev_literals(LdsCtx& c, uint8_t* lit,
   if (type == 2) {
     if (rem < 1) return -1;
     uint8_t hb = p[0];
-    uint8_t weights[256];
-    int nw;
-    if (hb >= 128) {
-      nw = hb - 127;
-      int64_t wb = (nw + 1) / 2;
-      if (1 + wb > rem) return -1;
-      for (int i = 0; i < nw; i++) {
-        uint8_t b = p[1 + i / 2];
-        weights[i] = (i & 1) ? (b & 0xF) : (b >> 4);
+    // weights parse + table build on lane 0 only (keeps the 256-byte
+    // weights array and FSE decode state out of per-lane scratch)
+    if (lane == 0) {
+      int nw = -1;
+      int64_t consumed = -1;
+      if (hb >= 128) {
+        nw = hb - 127;
+        int64_t wb = (nw + 1) / 2;
+        if (1 + wb <= rem) {
+          for (int i = 0; i < nw; i++) {
+            uint8_t b = p[1 + i / 2];
+            c.weights[i] = (i & 1) ? (b & 0xF) : (b >> 4);
+          }
+          consumed = 1 + wb;
+        }
+      } else if (1 + hb <= rem) {
+        uint8_t wtmp[256];
+        nw = fse_decompress(p + 1, hb, wtmp, 255);
+        for (int i = 0; i < nw; i++) c.weights[i] = wtmp[i];
+        consumed = 1 + hb;
       }
-      p += 1 + wb;
-      rem -= 1 + wb;
-    } else {
-      if (1 + hb > rem) return -1;
-      nw = fse_decompress(p + 1, hb, weights, 255);  // redundant per lane
-      if (nw < 0) return -1;
-      p += 1 + hb;
-      rem -= 1 + hb;
+      bool ok = consumed >= 0 && nw >= 0 && huf_build(c.huf, c.weights, nw);
+      c.haveHuf = ok ? 1 : -1;
+      c.scratch_i64 = consumed;
     }
-    bool ok = true;
-    if (lane == 0) ok = huf_build(c.huf, weights, nw);
-    __syncthreads();
-    // broadcast lane 0's verdict via LDS flag
-    if (lane == 0) c.haveHuf = ok ? 1 : -1;
     __syncthreads();
     if (c.haveHuf < 0) return -1;
+    p += c.scratch_i64;
+    rem -= c.scratch_i64;
+    __syncthreads();
   } else if (c.haveHuf != 1) {
     return -1;
   }
@@ -198,15 +206,14 @@ __device__ inline int64_t dev_read_seq_table(
     LdsCtx& c, FseTable& t, int& have, int mode, const uint8_t* src, int64_t n,
     int which, int maxSymLimit, int lane) {  // which: 0=LL 1=OF 2=ML
   if (mode == 0) {
-    int16_t cnt[64];
-    int maxSym, tlog;
-    if (which == 0) predef_ll(cnt, maxSym, tlog);
-    else if (which == 1) predef_of(cnt, maxSym, tlog);
-    else predef_ml(cnt, maxSym, tlog);
-    bool ok = true;
-    if (lane == 0) ok = fse_build(t, cnt, maxSym, tlog);
-    __syncthreads();
-    if (lane == 0) have = ok ? 1 : -1;
+    if (lane == 0) {
+      int16_t cnt[64];
+      int maxSym, tlog;
+      if (which == 0) predef_ll(cnt, maxSym, tlog);
+      else if (which == 1) predef_of(cnt, maxSym, tlog);
+      else predef_ml(cnt, maxSym, tlog);
+      have = fse_build(t, cnt, maxSym, tlog) ? 1 : -1;
+    }
     __syncthreads();
     return have < 0 ? -1 : 0;
   }
@@ -220,17 +227,17 @@ __device__ inline int64_t dev_read_seq_table(
     return 1;
   }
   if (mode == 2) {
-    BitFwd br{src, n};
-    int16_t cnt[64];
-    int tlog;
-    int maxSym = fse_read_ncount(br, cnt, maxSymLimit, tlog);  // redundant
-    if (maxSym < 0) return -1;
-    bool ok = true;
-    if (lane == 0) ok = fse_build(t, cnt, maxSym, tlog);
+    if (lane == 0) {
+      BitFwd br{src, n};
+      int16_t cnt[64];
+      int tlog;
+      int maxSym = fse_read_ncount(br, cnt, maxSymLimit, tlog);
+      bool ok = maxSym >= 0 && fse_build(t, cnt, maxSym, tlog);
+      have = ok ? 1 : -1;
+      c.scratch_i64 = br.bytes_consumed();
+    }
     __syncthreads();
-    if (lane == 0) have = ok ? 1 : -1;
-    __syncthreads();
-    return have < 0 ? -1 : br.bytes_consumed();
+    return have < 0 ? -1 : c.scratch_i64;
   }
   return have == 1 ? 0 : -1;
 }
