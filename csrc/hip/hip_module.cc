@@ -608,7 +608,12 @@ static torch::Tensor zstd_decompress_into(torch::Tensor src, torch::Tensor jobs,
   CHECK_GPU(jobs);
   CHECK_GPU(dst);
   int64_t njobs = jobs.size(0);
-  int64_t nblocks = njobs < 2048 ? njobs : 2048;
+  int64_t cap = 2048;
+  if (const char* e = std::getenv("LAKESOUL_ZSTD_BLOCKS")) {
+    int64_t v = atoll(e);
+    if (v > 0) cap = v;
+  }
+  int64_t nblocks = njobs < cap ? njobs : cap;
   auto scratch = torch::empty({nblocks * lsz_gpu_litbuf_bytes()},
                               src.options().dtype(torch::kUInt8));
   auto status = torch::empty({njobs}, src.options().dtype(torch::kInt32));
