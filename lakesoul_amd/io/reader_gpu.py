@@ -50,7 +50,12 @@ _ESIZE = {torch.uint8: 1, torch.int32: 4, torch.int64: 8, torch.float32: 4, torc
 
 import os as _os
 
-_GPU_ZSTD = _os.environ.get("LAKESOUL_GPU_ZSTD", "1") != "0"
+# GPU zstd decode is OFF by default: the wave-per-page kernel is correct
+# (tests/test_gpu.py) but measured 2.3-5.6 GB/s aggregate vs ~12 GB/s for
+# the 16-thread host zstd path (profiles/r01_gpu_zstd.md) — sequence
+# execution is serial-latency-bound. Round-2 redesign: wide copies,
+# lane-per-stream huffman, two-phase sequence metadata.
+_GPU_ZSTD = _os.environ.get("LAKESOUL_GPU_ZSTD", "0") == "1"
 
 
 def fetch_raw(files: List[str], names: List[str]) -> dict:

@@ -460,9 +460,12 @@ def test_gpu_zstd_kernel_matches_reference(dev):
 
 
 @pytest.mark.gpu
-def test_gpu_zstd_scan_path_active(dev, tmp_path):
-    """End-to-end MOR scan with the GPU zstd path on: results equal the
-    CPU scan, and the unit fetch reports zstd jobs (not host decode)."""
+def test_gpu_zstd_scan_path_active(dev, tmp_path, monkeypatch):
+    """End-to-end MOR scan with the GPU zstd path opted in: results equal
+    the CPU scan, and the unit fetch reports zstd jobs (not host decode)."""
+    import lakesoul_amd.io.reader_gpu as rgpu
+
+    monkeypatch.setattr(rgpu, "_GPU_ZSTD", True)
     from lakesoul_amd.io.reader_gpu import fetch_raw
 
     catalog = _mk_catalog(tmp_path)
