@@ -71,7 +71,31 @@ struct LdsCtx {
   int16_t counts[64];      // ncount output (seq tables)
   uint8_t weights[256];    // huffman weights
   int scratch_i32;
+  // rolling window of the last 32 output bytes (win[31] most recent).
+  // Matches with offset <= 32 read ONLY this window: no global reads,
+  // no fences — the serial cost per sequence collapses to LDS ops.
+  uint8_t win[32];
 };
+
+// slide `src[0..n)` into the 32-byte window (volatile read: src may be
+// wave-written global memory that has been fenced)
+__device__ inline void win_append_from(LdsCtx& c, const uint8_t* src,
+                                       int64_t n, int lane) {
+  if (n <= 0) return;
+  uint8_t nb = 0;
+  if (lane < 32) {
+    int64_t k = (int64_t)lane + n;
+    nb = (k < 32) ? c.win[k] : ((volatile const uint8_t*)src)[k - 32];
+  }
+  __syncthreads();
+  if (lane < 32) c.win[lane] = nb;
+  __syncthreads();
+}
+
+__device__ inline void win_fill(LdsCtx& c, uint8_t v, int lane) {
+  if (lane < 32) c.win[lane] = v;
+  __syncthreads();
+}
 
 // decode the literals section. All lanes in lockstep; table builds by
 // lane 0 into LDS; stream decode on up to 4 lanes. Returns bytes
@@ -271,6 +295,7 @@ __device__ inline int64_t dev_block(LdsCtx& c, uint8_t* lit, uint32_t* rep,
   if (nSeq == 0) {
     if (pos + litLen > dstCap) return -1;
     wcopy_from_lit(dstBase + pos, lit, (uint32_t)litLen, lane);
+    win_append_from(c, lit, litLen, lane);
     return litLen;
   }
 
@@ -298,6 +323,10 @@ __device__ inline int64_t dev_block(LdsCtx& c, uint8_t* lit, uint32_t* rep,
   sOf.init(c.of, br);
   sMl.init(c.ml, br);
 
+  // W = everything below this output position is store-fence'd (visible
+  // to plain/GLC reads); advance lazily, fencing only when a large-offset
+  // match must read not-yet-fenced bytes
+  int64_t W = pos;
   int64_t litPos = 0;
   int64_t out = pos;
   for (int64_t s = 0; s < nSeq; s++) {
@@ -331,12 +360,64 @@ __device__ inline int64_t dev_block(LdsCtx& c, uint8_t* lit, uint32_t* rep,
       }
     }
 
+    // ---- literals: litBuf reads are hazard-free (fenced after the
+    // literals phase); dst writes are fire-and-forget ----
     if (litPos + litLenSeq > litLen || out + litLenSeq > dstCap) return -1;
-    wcopy_from_lit(dstBase + out, lit + litPos, litLenSeq, lane);
-    litPos += litLenSeq;
-    out += litLenSeq;
+    if (litLenSeq) {
+      for (uint32_t j = lane; j < litLenSeq; j += LANES)
+        dstBase[out + j] = ((volatile const uint8_t*)lit)[litPos + j];
+      // window <- last 32 bytes of (window ++ literals)
+      uint8_t nb = 0;
+      if (lane < 32) {
+        int64_t k = (int64_t)lane + litLenSeq;   // index into win ++ lits
+        nb = (k < 32) ? c.win[k]
+                      : ((volatile const uint8_t*)lit)[litPos + (k - 32)];
+      }
+      __syncthreads();
+      if (lane < 32) c.win[lane] = nb;
+      __syncthreads();
+      litPos += litLenSeq;
+      out += litLenSeq;
+    }
+
+    // ---- match ----
     if ((int64_t)offset > out || out + matchLen > dstCap) return -1;
-    wcopy_match(dstBase, out, offset, matchLen, lane);
+    if (offset <= 32) {
+      // source is entirely inside the register window: no global reads
+      for (uint32_t j = lane; j < matchLen; j += LANES)
+        dstBase[out + j] = c.win[(32 - offset) + (j % offset)];
+      uint8_t nb = 0;
+      if (lane < 32) {
+        int64_t k = (int64_t)lane + matchLen;
+        nb = (k < 32) ? c.win[k]
+                      : c.win[(32 - offset) + (uint32_t)((k - 32) % offset)];
+      }
+      __syncthreads();
+      if (lane < 32) c.win[lane] = nb;
+      __syncthreads();
+    } else {
+      // large offset: source bytes live below `out`; fence once if they
+      // reach into the unfenced region
+      uint32_t span = offset < matchLen ? offset : matchLen;
+      if (out - (int64_t)offset + (int64_t)span > W) {
+        waitcnt0();
+        W = out;
+      }
+      for (uint32_t j = lane; j < matchLen; j += LANES)
+        dstBase[out + j] =
+            ((volatile const uint8_t*)dstBase)[out - offset + (j % offset)];
+      uint8_t nb = 0;
+      if (lane < 32) {
+        int64_t k = (int64_t)lane + matchLen;
+        nb = (k < 32)
+                 ? c.win[k]
+                 : ((volatile const uint8_t*)
+                        dstBase)[out - offset + (uint32_t)((k - 32) % offset)];
+      }
+      __syncthreads();
+      if (lane < 32) c.win[lane] = nb;
+      __syncthreads();
+    }
     out += matchLen;
 
     if (s + 1 < nSeq) {
@@ -349,6 +430,7 @@ __device__ inline int64_t dev_block(LdsCtx& c, uint8_t* lit, uint32_t* rep,
   int64_t tail = litLen - litPos;
   if (tail < 0 || out + tail > dstCap) return -1;
   wcopy_from_lit(dstBase + out, lit + litPos, (uint32_t)tail, lane);
+  win_append_from(c, lit + litPos, tail, lane);
   out += tail;
   return out - pos;
 }
@@ -388,6 +470,7 @@ __device__ inline int64_t dev_frame(LdsCtx& c, uint8_t* lit, const uint8_t* src,
     if (ip > n) return -1;
 
     if (lane == 0) { c.haveLl = c.haveOf = c.haveMl = 0; c.haveHuf = 0; }
+    if (lane < 32) c.win[lane] = 0;
     __syncthreads();
     rep[0] = 1; rep[1] = 4; rep[2] = 8;
 
@@ -403,11 +486,25 @@ __device__ inline int64_t dev_frame(LdsCtx& c, uint8_t* lit, const uint8_t* src,
       if (btype == 0) {
         if (ip + bsize > n || pos + bsize > dstCap) return -1;
         wcopy_src(dst + pos, src + ip, bsize, lane);
+        win_append_from(c, src + ip, bsize, lane);
         ip += bsize;
         pos += bsize;
       } else if (btype == 1) {
         if (ip + 1 > n || pos + bsize > dstCap) return -1;
         wfill(dst + pos, src[ip], bsize, lane);
+        if (bsize >= 32) {
+          win_fill(c, src[ip], lane);
+        } else {
+          uint8_t v = src[ip];
+          uint8_t nb = 0;
+          if (lane < 32) {
+            int64_t k = (int64_t)lane + bsize;
+            nb = (k < 32) ? c.win[k] : v;
+          }
+          __syncthreads();
+          if (lane < 32) c.win[lane] = nb;
+          __syncthreads();
+        }
         ip += 1;
         pos += bsize;
       } else if (btype == 2) {
