@@ -22,8 +22,10 @@
 #ifndef LSZ_HD
 #if defined(__HIPCC__) || defined(__HIP_DEVICE_COMPILE__)
 #define LSZ_HD __host__ __device__
+#define LSZ_COLD __host__ __device__ __attribute__((noinline))
 #else
 #define LSZ_HD
+#define LSZ_COLD
 #endif
 #endif
 
@@ -149,7 +151,7 @@ struct FseTable {  // no default initializers: instances live in LDS
 
 // read normalized counts (RFC 8878 4.1.1) from a forward bitstream.
 // returns max symbol (inclusive) or -1 on error.
-LSZ_HD inline int fse_read_ncount(BitFwd& br, int16_t* counts, int maxSymLimit,
+LSZ_COLD inline int fse_read_ncount(BitFwd& br, int16_t* counts, int maxSymLimit,
                                   int& tableLog) {
   tableLog = (int)br.read(4) + 5;
   if (tableLog > kMaxTableLog) return -1;
@@ -197,15 +199,16 @@ LSZ_HD inline int fse_read_ncount(BitFwd& br, int16_t* counts, int maxSymLimit,
   return sym - 1;
 }
 
-// build a decode table from normalized counts (FSE_buildDTable)
-LSZ_HD inline bool fse_build(FseTable& t, const int16_t* counts, int maxSym,
-                             int tableLog) {
+// build a decode table from normalized counts (FSE_buildDTable).
+// symTab/symbolNext are caller scratch (>= 512 / 256 entries) so the
+// device build can keep them in LDS instead of per-lane scratch memory.
+LSZ_COLD inline bool fse_build(FseTable& t, const int16_t* counts, int maxSym,
+                               int tableLog, uint8_t* symTab,
+                               uint16_t* symbolNext) {
   t.tableLog = tableLog;
   t.rle = false;
   int tableSize = 1 << tableLog;
   int highThreshold = tableSize - 1;
-  uint16_t symbolNext[kMaxSymbols];
-  uint8_t symTab[1 << kMaxTableLog];
 
   for (int s = 0; s <= maxSym; s++) {
     if (counts[s] == -1) {
@@ -239,6 +242,13 @@ LSZ_HD inline bool fse_build(FseTable& t, const int16_t* counts, int maxSym,
   return true;
 }
 
+LSZ_HD inline bool fse_build(FseTable& t, const int16_t* counts, int maxSym,
+                             int tableLog) {
+  uint8_t symTab[1 << kMaxTableLog];
+  uint16_t symbolNext[kMaxSymbols];
+  return fse_build(t, counts, maxSym, tableLog, symTab, symbolNext);
+}
+
 LSZ_HD inline void fse_build_rle(FseTable& t, uint8_t sym) {
   t.rle = true;
   t.rleSym = sym;
@@ -264,15 +274,16 @@ struct FseState {
 // generic FSE decompression (used for huffman weights): strict s1/s2
 // alternation, stop after overflow emits the final symbol (libzstd
 // FSE_decompress_usingDTable tail semantics)
-LSZ_HD inline int fse_decompress(const uint8_t* src, int64_t n, uint8_t* dst,
-                                 int dstCap) {
+LSZ_COLD inline int fse_decompress_ws(const uint8_t* src, int64_t n, uint8_t* dst,
+                                    int dstCap, FseTable& table,
+                                    int16_t* counts, uint8_t* symTab,
+                                    uint16_t* symbolNext) {
   BitFwd hdr{src, n};
-  int16_t counts[kMaxSymbols];
   int tableLog;
   int maxSym = fse_read_ncount(hdr, counts, 255, tableLog);
   if (maxSym < 0) return -1;
-  FseTable table;
-  if (!fse_build(table, counts, maxSym, tableLog)) return -1;
+  if (!fse_build(table, counts, maxSym, tableLog, symTab, symbolNext))
+    return -1;
   int64_t consumed = hdr.bytes_consumed();
   BitBwd br;
   if (!br.init(src + consumed, n - consumed)) return -1;
@@ -301,6 +312,16 @@ LSZ_HD inline int fse_decompress(const uint8_t* src, int64_t n, uint8_t* dst,
   return out;
 }
 
+LSZ_HD inline int fse_decompress(const uint8_t* src, int64_t n, uint8_t* dst,
+                                 int dstCap) {
+  FseTable table;
+  int16_t counts[kMaxSymbols];
+  uint8_t symTab[1 << kMaxTableLog];
+  uint16_t symbolNext[kMaxSymbols];
+  return fse_decompress_ws(src, n, dst, dstCap, table, counts, symTab,
+                           symbolNext);
+}
+
 // ---------------------------------------------------------------------- //
 // Huffman literals
 // ---------------------------------------------------------------------- //
@@ -318,7 +339,7 @@ struct HufTable {  // no default initializers: instances live in LDS
 };
 
 // build decode table from weights[0..nsym-1] (HUF_readDTableX1 layout)
-LSZ_HD inline bool huf_build(HufTable& t, const uint8_t* weights, int nsym) {
+LSZ_COLD inline bool huf_build(HufTable& t, const uint8_t* weights, int nsym) {
   uint32_t rankCount[kHufMaxBits + 2] = {0};
   uint32_t total = 0;
   for (int s = 0; s < nsym; s++) {

@@ -71,6 +71,13 @@ struct LdsCtx {
   int16_t counts[64];      // ncount output (seq tables)
   uint8_t weights[256];    // huffman weights
   int scratch_i32;
+  // lane0 table-build workspaces (LDS, not per-lane scratch memory —
+  // dynamically-indexed locals would otherwise force a multi-KB private
+  // segment that craters wave residency)
+  FseTable wt;             // huffman-weight FSE table
+  int16_t wcounts[256];
+  uint8_t symTab[1 << kMaxTableLog];
+  uint16_t symNext[256];
   // rolling window of the last 32 output bytes (win[31] most recent).
   // Matches with offset <= 32 read ONLY this window: no global reads,
   // no fences — the serial cost per sequence collapses to LDS ops.
@@ -183,9 +190,8 @@ __device__ inline int64_t dev_literals(LdsCtx& c, uint8_t* lit,
           consumed = 1 + wb;
         }
       } else if (1 + hb <= rem) {
-        uint8_t wtmp[256];
-        nw = fse_decompress(p + 1, hb, wtmp, 255);
-        for (int i = 0; i < nw; i++) c.weights[i] = wtmp[i];
+        nw = fse_decompress_ws(p + 1, hb, c.weights, 255, c.wt, c.wcounts,
+                               c.symTab, c.symNext);
         consumed = 1 + hb;
       }
       bool ok = consumed >= 0 && nw >= 0 && huf_build(c.huf, c.weights, nw);
@@ -215,10 +221,13 @@ __device__ inline int64_t dev_literals(LdsCtx& c, uint8_t* lit,
     int64_t o4 = rs - 3 * o123;
     if (o4 < 0) return -1;
     const uint8_t* q = p + 6;
-    if (lane == 0) ok = huf_stream(c.huf, q, s1, lit, o123);
-    if (lane == 1) ok = huf_stream(c.huf, q + s1, s2, lit + o123, o123);
-    if (lane == 2) ok = huf_stream(c.huf, q + s1 + s2, s3, lit + 2 * o123, o123);
-    if (lane == 3) ok = huf_stream(c.huf, q + s1 + s2 + s3, s4, lit + 3 * o123, o4);
+    if (lane < 4) {
+      // one shared decode instance: lanes 0-3 each take a stream
+      int64_t starts[4] = {0, s1, s1 + s2, s1 + s2 + s3};
+      int64_t lens[4] = {s1, s2, s3, s4};
+      ok = huf_stream(c.huf, q + starts[lane], lens[lane], lit + lane * o123,
+                      lane == 3 ? o4 : o123);
+    }
   }
   waitcnt0();
   __syncthreads();
@@ -231,12 +240,11 @@ __device__ inline int64_t dev_read_seq_table(
     int which, int maxSymLimit, int lane) {  // which: 0=LL 1=OF 2=ML
   if (mode == 0) {
     if (lane == 0) {
-      int16_t cnt[64];
       int maxSym, tlog;
-      if (which == 0) predef_ll(cnt, maxSym, tlog);
-      else if (which == 1) predef_of(cnt, maxSym, tlog);
-      else predef_ml(cnt, maxSym, tlog);
-      have = fse_build(t, cnt, maxSym, tlog) ? 1 : -1;
+      if (which == 0) predef_ll(c.counts, maxSym, tlog);
+      else if (which == 1) predef_of(c.counts, maxSym, tlog);
+      else predef_ml(c.counts, maxSym, tlog);
+      have = fse_build(t, c.counts, maxSym, tlog, c.symTab, c.symNext) ? 1 : -1;
     }
     __syncthreads();
     return have < 0 ? -1 : 0;
@@ -253,10 +261,10 @@ __device__ inline int64_t dev_read_seq_table(
   if (mode == 2) {
     if (lane == 0) {
       BitFwd br{src, n};
-      int16_t cnt[64];
       int tlog;
-      int maxSym = fse_read_ncount(br, cnt, maxSymLimit, tlog);
-      bool ok = maxSym >= 0 && fse_build(t, cnt, maxSym, tlog);
+      int maxSym = fse_read_ncount(br, c.counts, maxSymLimit, tlog);
+      bool ok = maxSym >= 0 &&
+                fse_build(t, c.counts, maxSym, tlog, c.symTab, c.symNext);
       have = ok ? 1 : -1;
       c.scratch_i64 = br.bytes_consumed();
     }
@@ -523,7 +531,7 @@ __device__ inline int64_t dev_frame(LdsCtx& c, uint8_t* lit, const uint8_t* src,
   return pos;
 }
 
-__global__ void __launch_bounds__(LANES) zstd_pages_kernel(
+__global__ void __launch_bounds__(LANES, 4) zstd_pages_kernel(
     const uint8_t* __restrict__ src, const int64_t* __restrict__ jobs,
     int64_t njobs, uint8_t* __restrict__ dst, uint8_t* __restrict__ scratch,
     int32_t* __restrict__ status) {
