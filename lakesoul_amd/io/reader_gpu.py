@@ -50,12 +50,28 @@ _ESIZE = {torch.uint8: 1, torch.int32: 4, torch.int64: 8, torch.float32: 4, torc
 
 import os as _os
 
-# GPU zstd decode is OFF by default: the wave-per-page kernel is correct
-# (tests/test_gpu.py) but measured 2.3-5.6 GB/s aggregate vs ~12 GB/s for
-# the 16-thread host zstd path (profiles/r01_gpu_zstd.md) — sequence
-# execution is serial-latency-bound. Round-2 redesign: wide copies,
-# lane-per-stream huffman, two-phase sequence metadata.
-_GPU_ZSTD = _os.environ.get("LAKESOUL_GPU_ZSTD", "0") == "1"
+def _default_gpu_zstd() -> bool:
+    """GPU zstd decode (wave-per-page kernel, ~6-13 GB/s measured) beats
+    the host pool only when host CPUs are scarce: the cgroup CPU quota is
+    shared by every rank on the node, so at 8 ranks each rank gets ~2
+    threads (~1.5 GB/s zstd) while the GPU path holds its rate. Default:
+    ON when fewer than 6 host threads are available per rank, OFF
+    otherwise (measured N=1 crossover, profiles/r01_gpu_zstd.md)."""
+    env = _os.environ.get("LAKESOUL_GPU_ZSTD")
+    if env is not None:
+        return env == "1"
+    try:
+        quota_s = open("/sys/fs/cgroup/cpu.max").read().split()
+        ncpu = _os.cpu_count() or 16
+        quota = ncpu if quota_s[0] == "max" else max(
+            1, int(quota_s[0]) // int(quota_s[1]))
+    except OSError:
+        quota = _os.cpu_count() or 16
+    world = int(_os.environ.get("WORLD_SIZE", "1"))
+    return quota // max(1, world) < 6
+
+
+_GPU_ZSTD = _default_gpu_zstd()
 
 
 def fetch_raw(files: List[str], names: List[str]) -> dict:
