@@ -32,6 +32,12 @@ static const int kLitBufCap = 1 << 17;  // 128 KB literals per block
 
 __device__ inline void waitcnt0() { __builtin_amdgcn_s_waitcnt(0); }
 
+// order LDS ops only (lgkmcnt(0); vmcnt=63, expcnt=7 untouched): the
+// window read/write shuffle must not drain in-flight global stores —
+// a full __syncthreads here costs ~1000 cycles per sequence (measured,
+// profiles/r01_gpu_zstd.md)
+__device__ inline void lds_sync() { __builtin_amdgcn_s_waitcnt(0xC07F); }
+
 // unaligned-capable wide element copy: 8 bytes per lane per iteration
 // (gfx950 flat loads support byte-aligned addresses), byte tail
 template <bool VOL>
@@ -400,9 +406,9 @@ __device__ inline int64_t dev_block(LdsCtx& c, uint8_t* lit, uint32_t* rep,
         nb = (k < 32) ? c.win[k]
                       : ((volatile const uint8_t*)lit)[litPos + (k - 32)];
       }
-      __syncthreads();
+      lds_sync();
       if (lane < 32) c.win[lane] = nb;
-      __syncthreads();
+      lds_sync();
       litPos += litLenSeq;
       out += litLenSeq;
     }
@@ -419,9 +425,9 @@ __device__ inline int64_t dev_block(LdsCtx& c, uint8_t* lit, uint32_t* rep,
         nb = (k < 32) ? c.win[k]
                       : c.win[(32 - offset) + (uint32_t)((k - 32) % offset)];
       }
-      __syncthreads();
+      lds_sync();
       if (lane < 32) c.win[lane] = nb;
-      __syncthreads();
+      lds_sync();
     } else {
       // large offset: source bytes live below `out`; fence once if they
       // reach into the unfenced region
@@ -441,9 +447,9 @@ __device__ inline int64_t dev_block(LdsCtx& c, uint8_t* lit, uint32_t* rep,
                  : ((volatile const uint8_t*)
                         dstBase)[out - offset + (uint32_t)((k - 32) % offset)];
       }
-      __syncthreads();
+      lds_sync();
       if (lane < 32) c.win[lane] = nb;
-      __syncthreads();
+      lds_sync();
     }
     out += matchLen;
 
