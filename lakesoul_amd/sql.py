@@ -34,6 +34,7 @@ _KEYWORDS = {
     "select", "from", "where", "group", "order", "by", "limit", "and",
     "or", "not", "in", "is", "null", "between", "as", "asc", "desc",
     "show", "tables", "namespaces", "describe", "distinct", "version",
+    "join", "inner", "left", "on", "insert", "into", "values",
 }
 
 _AGGS = {"count", "sum", "min", "max", "avg"}
@@ -86,9 +87,20 @@ class SelectItem:
 
 
 @dataclass
+class JoinSpec:
+    table: str
+    namespace: str = "default"
+    alias: str = ""
+    kind: str = "inner"               # inner | left
+    on: List[Tuple[str, str]] = field(default_factory=list)  # (left col, right col)
+
+
+@dataclass
 class Query:
     table: str
     namespace: str = "default"
+    alias: str = ""
+    join: Optional["JoinSpec"] = None
     items: List[SelectItem] = field(default_factory=list)
     where: Optional[Expr] = None
     group_by: List[str] = field(default_factory=list)
@@ -141,13 +153,50 @@ class _Parser:
             return ("describe", (ns, name))
         if (k, v) == ("kw", "select"):
             return ("select", self.select())
+        if (k, v) == ("kw", "insert"):
+            return ("insert", self.insert())
         raise SqlError(f"unsupported statement start: {v!r}")
+
+    def insert(self):
+        self.expect("kw", "insert")
+        self.expect("kw", "into")
+        ns, name = self.table_name()
+        cols = []
+        if self.accept("op", "("):
+            cols.append(self.expect("id"))
+            while self.accept("op", ","):
+                cols.append(self.expect("id"))
+            self.expect("op", ")")
+        k, v = self.peek()
+        if (k, v) == ("kw", "values"):
+            self.next()
+            rows = []
+            while True:
+                self.expect("op", "(")
+                row = [self.literal()]
+                while self.accept("op", ","):
+                    row.append(self.literal())
+                self.expect("op", ")")
+                rows.append(row)
+                if not self.accept("op", ","):
+                    break
+            return {"namespace": ns, "table": name, "columns": cols, "rows": rows}
+        if (k, v) == ("kw", "select"):
+            return {"namespace": ns, "table": name, "columns": cols,
+                    "select": self.select()}
+        raise SqlError("INSERT expects VALUES or SELECT")
 
     def table_name(self):
         name = self.expect("id")
         if self.accept("op", "."):
             return name, self.expect("id")
         return "default", name
+
+    def qualified_id(self) -> str:
+        name = self.expect("id")
+        if self.accept("op", "."):
+            return f"{name}.{self.expect('id')}"
+        return name
 
     # -- SELECT ---------------------------------------------------------- #
 
@@ -160,15 +209,38 @@ class _Parser:
             q.items.append(self.select_item())
         self.expect("kw", "from")
         q.namespace, q.table = self.table_name()
+        if self.peek()[0] == "id":
+            q.alias = self.next()[1]
         if self.accept("kw", "version"):   # time travel: FROM t VERSION 3
             q.version = int(self.expect("num"))
+        k, v = self.peek()
+        if (k == "kw" and v in ("join", "inner", "left")):
+            kind = "inner"
+            if self.accept("kw", "left"):
+                kind = "left"
+            else:
+                self.accept("kw", "inner")
+            self.expect("kw", "join")
+            jns, jname = self.table_name()
+            j = JoinSpec(jname, jns, kind=kind)
+            if self.peek()[0] == "id":
+                j.alias = self.next()[1]
+            self.expect("kw", "on")
+            while True:
+                lcol = self.qualified_id()
+                self.expect("op", "=")
+                rcol = self.qualified_id()
+                j.on.append((lcol, rcol))
+                if not self.accept("kw", "and"):
+                    break
+            q.join = j
         if self.accept("kw", "where"):
             q.where = self.or_expr()
         if self.accept("kw", "group"):
             self.expect("kw", "by")
-            q.group_by = [self.expect("id")]
+            q.group_by = [self.qualified_id()]
             while self.accept("op", ","):
-                q.group_by.append(self.expect("id"))
+                q.group_by.append(self.qualified_id())
         if self.accept("kw", "order"):
             self.expect("kw", "by")
             q.order_by = [self.order_item(q)]
@@ -193,11 +265,11 @@ class _Parser:
                     raise SqlError(f"{fn}(*) not supported")
                 arg = ""
             else:
-                arg = self.expect("id")
+                arg = self.qualified_id()
             self.expect("op", ")")
             item = SelectItem("agg", name=arg, fn=fn)
         elif k == "id":
-            item = SelectItem("col", name=self.next()[1])
+            item = SelectItem("col", name=self.qualified_id())
         else:
             raise SqlError(f"bad select item near {v!r}")
         if self.accept("kw", "as"):
@@ -207,7 +279,7 @@ class _Parser:
         return item
 
     def order_item(self, q: Query):
-        name = self.expect("id") if self.peek()[0] == "id" else str(self.expect("num"))
+        name = self.qualified_id() if self.peek()[0] == "id" else str(self.expect("num"))
         if name.isdigit():
             name = q.items[int(name) - 1].out_name
         desc = False
@@ -251,7 +323,7 @@ class _Parser:
             e = self.or_expr()
             self.expect("op", ")")
             return e
-        col = self.expect("id")
+        col = self.qualified_id()
         k, v = self.peek()
         if (k, v) == ("kw", "is"):
             self.next()
@@ -299,6 +371,8 @@ def execute_sql(catalog, sql: str, device: Optional[str] = None):
     import pandas as pd
 
     kind, payload = parse_sql(sql)
+    if kind == "insert":
+        return _execute_insert(catalog, payload, device=device)
     if kind == "show":
         if payload == "namespaces":
             return pd.DataFrame({"namespace": catalog.list_namespaces()})
@@ -316,11 +390,268 @@ def execute_sql(catalog, sql: str, device: Optional[str] = None):
     return _execute_select(catalog, payload, device=device)
 
 
-def _execute_select(catalog, q: Query, device=None):
-    import numpy as np
+def _execute_insert(catalog, ins: dict, device=None):
     import pandas as pd
 
+    t = catalog.table(ins["table"], ins["namespace"])
+    names = ins["columns"] or t.schema.names()
+    if "rows" in ins:
+        rows = ins["rows"]
+        for r in rows:
+            if len(r) != len(names):
+                raise SqlError(f"INSERT row has {len(r)} values, expected {len(names)}")
+        df = pd.DataFrame(rows, columns=names)
+    else:
+        df = _execute_select(catalog, ins["select"], device=device)
+        if len(df.columns) != len(names):
+            raise SqlError("INSERT SELECT column count mismatch")
+        df.columns = names
+    # cast to schema dtypes
+    for f in t.schema:
+        if f.name in df.columns and f.is_fixed_width and not f.dtype.startswith("decimal"):
+            import numpy as _np
+
+            from .io.batch import np_dtype_for
+
+            df[f.name] = df[f.name].astype(np_dtype_for(f.dtype))
+    t.write(df, device=device)
+    return pd.DataFrame({"rows_inserted": [len(df)]})
+
+
+def _pd_eval(expr: Expr, df, col):
+    """Evaluate a filter Expr against a pandas DataFrame (join residuals
+    — single-table queries push the same Expr into the scan instead)."""
+    import pandas as pd
+
+    if isinstance(expr, And):
+        return _pd_eval(expr.left, df, col) & _pd_eval(expr.right, df, col)
+    if isinstance(expr, Or):
+        return _pd_eval(expr.left, df, col) | _pd_eval(expr.right, df, col)
+    if isinstance(expr, Not):
+        return ~_pd_eval(expr.inner, df, col)
+    if isinstance(expr, IsNull):
+        m = df[col(expr.col)].isna()
+        return ~m if expr.negate else m
+    if isinstance(expr, Cmp):
+        series = df[col(expr.col)]
+        v = expr.value
+        op = expr.op
+        if op == "eq":
+            return series == v
+        if op == "noteq":
+            return series != v
+        if op == "gt":
+            return series > v
+        if op == "gteq":
+            return series >= v
+        if op == "lt":
+            return series < v
+        if op == "lteq":
+            return series <= v
+        raise SqlError(f"unsupported join filter op {op}")
+    raise SqlError(f"unsupported filter node {type(expr).__name__} in join")
+
+
+def _execute_join_select(catalog, q: Query, device=None):
+    """Two-table equi-join (reference: lakesoul-datafusion delegates joins
+    to DataFusion; here a pandas hash join over two MOR scans)."""
+    import pandas as pd
+
+    lt = catalog.table(q.table, q.namespace)
+    j = q.join
+    rt = catalog.table(j.table, j.namespace)
+    lnames = set(lt.schema.names())
+    rnames = set(rt.schema.names())
+    lquals = {q.alias or q.table, q.table}
+    rquals = {j.alias or j.table, j.table}
+
+    def side_of(name: str):
+        if "." in name:
+            qual, col = name.split(".", 1)
+            if qual in lquals and col in lnames:
+                return "L", col
+            if qual in rquals and col in rnames:
+                return "R", col
+            raise SqlError(f"cannot resolve {name!r}")
+        amb = (name in lnames) + (name in rnames)
+        if amb == 0:
+            raise SqlError(f"unknown column {name!r}")
+        if amb == 2:
+            raise SqlError(f"ambiguous column {name!r}: qualify it")
+        return ("L", name) if name in lnames else ("R", name)
+
+    # referenced columns per side
+    refs = []
+    star = any(it.kind == "star" for it in q.items)
+    for it in q.items:
+        if it.kind in ("col", "agg") and it.name:
+            refs.append(it.name)
+    refs += q.group_by + [n for n, _ in q.order_by]
+    if q.where is not None:
+        refs += list(q.where.columns())
+    for lcol, rcol in j.on:
+        refs += [lcol, rcol]
+    need_l, need_r = set(), set()
+    for r in refs:
+        side, col = side_of(r)
+        (need_l if side == "L" else need_r).add(col)
+    if star:
+        need_l, need_r = set(lnames), set(rnames)
+
+    ldf = lt.scan(columns=sorted(need_l) or None, version=q.version,
+                  device=device).to_arrow().to_pandas()
+    rdf = rt.scan(columns=sorted(need_r) or None,
+                  device=device).to_arrow().to_pandas()
+    # disambiguate overlapping names: right-side dupes get "<qual>." prefix
+    rqual = j.alias or j.table
+    overlap = set(ldf.columns) & set(rdf.columns)
+    rdf = rdf.rename(columns={c: f"{rqual}.{c}" for c in overlap})
+
+    def df_col(name: str) -> str:
+        side, col = side_of(name)
+        if side == "R" and col in overlap:
+            return f"{rqual}.{col}"
+        return col
+
+    left_on = []
+    right_on = []
+    for lcol, rcol in j.on:
+        sl, cl = side_of(lcol)
+        sr, cr = side_of(rcol)
+        if sl == sr:
+            raise SqlError("JOIN ON must reference both tables")
+        if sl == "R":
+            lcol, rcol = rcol, lcol
+            cl, cr = cr, cl
+        left_on.append(cl)
+        right_on.append(f"{rqual}.{cr}" if cr in overlap else cr)
+    out = ldf.merge(rdf, how=j.kind, left_on=left_on, right_on=right_on)
+
+    if q.where is not None:
+        out = out[_pd_eval(q.where, out, df_col)].reset_index(drop=True)
+
+    # project / aggregate (shared tail with the single-table path)
+    return _project_and_finish(q, out, df_col,
+                               all_cols=list(ldf.columns) + list(rdf.columns))
+
+
+def _project_and_finish(q: Query, df, col, all_cols):
+    """Shared SELECT tail: aggregation / projection / DISTINCT / ORDER BY
+    / LIMIT over a materialized DataFrame. ``col`` maps a referenced name
+    to the DataFrame column holding it."""
+    import pandas as pd
+
+    has_agg = any(it.kind == "agg" for it in q.items)
+    if has_agg or q.group_by:
+        def agg_series(sub: pd.DataFrame):
+            row = {}
+            for it in q.items:
+                if it.kind == "col":
+                    if it.name not in q.group_by:
+                        raise SqlError(
+                            f"column {it.name!r} must appear in GROUP BY")
+                    continue
+                if it.kind == "star":
+                    raise SqlError("SELECT * with aggregates is not valid")
+                s = sub[col(it.name)] if it.name else None
+                if it.fn == "count":
+                    row[it.out_name] = len(sub) if s is None else int(s.notna().sum())
+                elif it.fn == "sum":
+                    row[it.out_name] = s.sum()
+                elif it.fn == "min":
+                    row[it.out_name] = s.min()
+                elif it.fn == "max":
+                    row[it.out_name] = s.max()
+                elif it.fn == "avg":
+                    row[it.out_name] = float(s.mean())
+            return pd.Series(row)
+
+        if q.group_by:
+            gcols = [col(g) for g in q.group_by]
+            out = (df.groupby(gcols, as_index=False, sort=False)
+                     .apply(agg_series, include_groups=False)
+                   if pd.__version__ >= "2.2"
+                   else df.groupby(gcols, as_index=False).apply(agg_series))
+            cols = []
+            for it in q.items:
+                if it.kind == "col":
+                    dfc = col(it.name)
+                    outn = it.alias or it.name.split(".")[-1]
+                    if dfc != outn:
+                        out = out.rename(columns={dfc: outn})
+                    cols.append(outn)
+                else:
+                    cols.append(it.out_name)
+            out = out[cols]
+        else:
+            out = pd.DataFrame([agg_series(df)])
+    else:
+        cols, ren = [], {}
+        for it in q.items:
+            if it.kind == "star":
+                for c in all_cols:
+                    if c not in cols:
+                        cols.append(c)
+            else:
+                dfc = col(it.name)
+                cols.append(dfc)
+                outn = it.alias or it.name.split(".")[-1]
+                if dfc != outn:
+                    ren[dfc] = outn
+        out = df[cols].rename(columns=ren)
+        if q.distinct:
+            out = out.drop_duplicates().reset_index(drop=True)
+
+    if q.order_by:
+        names = []
+        for n, _ in q.order_by:
+            # ORDER BY may reference an output alias or a source column
+            names.append(n if n in out.columns
+                         else (col(n) if col(n) in out.columns else n))
+        asc = [not d for _, d in q.order_by]
+        out = out.sort_values(names, ascending=asc).reset_index(drop=True)
+    if q.limit is not None:
+        out = out.head(q.limit).reset_index(drop=True)
+    return out.reset_index(drop=True)
+
+
+def _strip_quals(q: Query, valid_quals) -> None:
+    """Single-table queries: rewrite `t.col`/`alias.col` to `col`."""
+    def strip(name: str) -> str:
+        if "." in name:
+            qual, col = name.split(".", 1)
+            if qual in valid_quals:
+                return col
+            raise SqlError(f"unknown table qualifier {qual!r}")
+        return name
+
+    for it in q.items:
+        if it.kind in ("col", "agg") and it.name:
+            it.name = strip(it.name)
+    q.group_by = [strip(g) for g in q.group_by]
+    q.order_by = [(strip(n) if "." in n else n, d) for n, d in q.order_by]
+
+    def walk(e):
+        if e is None:
+            return
+        if isinstance(e, (And, Or)):
+            walk(e.left)
+            walk(e.right)
+        elif isinstance(e, Not):
+            walk(e.inner)
+        elif isinstance(e, (Cmp, IsNull)):
+            e.col = strip(e.col)
+
+    walk(q.where)
+
+
+def _execute_select(catalog, q: Query, device=None):
+    import pandas as pd
+
+    if q.join is not None:
+        return _execute_join_select(catalog, q, device=device)
     t = catalog.table(q.table, q.namespace)
+    _strip_quals(q, {q.table, q.alias} - {""})
     schema_cols = t.schema.names()
 
     # columns actually needed from storage
@@ -350,66 +681,7 @@ def _execute_select(catalog, q: Query, device=None):
     scan = t.scan(columns=sorted(need) or None, filters=q.where,
                   version=q.version, device=device)
     df = scan.to_arrow().to_pandas()
-
-    # aggregate / project
-    if has_agg or q.group_by:
-        def agg_series(sub: pd.DataFrame):
-            row = {}
-            for it in q.items:
-                if it.kind == "col":
-                    if it.name not in q.group_by:
-                        raise SqlError(
-                            f"column {it.name!r} must appear in GROUP BY")
-                    continue
-                if it.kind == "star":
-                    raise SqlError("SELECT * with aggregates is not valid")
-                s = sub[it.name] if it.name else None
-                if it.fn == "count":
-                    row[it.out_name] = len(sub) if s is None else int(s.notna().sum())
-                elif it.fn == "sum":
-                    row[it.out_name] = s.sum()
-                elif it.fn == "min":
-                    row[it.out_name] = s.min()
-                elif it.fn == "max":
-                    row[it.out_name] = s.max()
-                elif it.fn == "avg":
-                    row[it.out_name] = float(s.mean())
-            return pd.Series(row)
-
-        if q.group_by:
-            out = (df.groupby(q.group_by, as_index=False, sort=False)
-                     .apply(agg_series, include_groups=False)
-                   if pd.__version__ >= "2.2"
-                   else df.groupby(q.group_by, as_index=False).apply(agg_series))
-            # keep declared item order, renaming group cols per alias
-            cols = []
-            for it in q.items:
-                cols.append(it.name if it.kind == "col" and not it.alias else it.out_name)
-                if it.kind == "col" and it.alias:
-                    out = out.rename(columns={it.name: it.alias})
-            out = out[cols]
-        else:
-            out = pd.DataFrame([agg_series(df)])
-    else:
-        cols, ren = [], {}
-        for it in q.items:
-            if it.kind == "star":
-                cols.extend([c for c in schema_cols if c not in cols])
-            else:
-                cols.append(it.name)
-                if it.alias:
-                    ren[it.name] = it.alias
-        out = df[cols].rename(columns=ren)
-        if q.distinct:
-            out = out.drop_duplicates().reset_index(drop=True)
-
-    if q.order_by:
-        names = [n for n, _ in q.order_by]
-        asc = [not d for _, d in q.order_by]
-        out = out.sort_values(names, ascending=asc).reset_index(drop=True)
-    if q.limit is not None:
-        out = out.head(q.limit).reset_index(drop=True)
-    return out.reset_index(drop=True)
+    return _project_and_finish(q, df, lambda n: n, all_cols=schema_cols)
 
 
 def repl(catalog, device=None, input_fn=input, print_fn=print):

@@ -176,3 +176,87 @@ def test_repl_loop(sql_table, capsys):
     joined = "\n".join(str(o) for o in outs)
     assert "1000" in joined
     assert "error:" in joined
+
+
+@pytest.fixture
+def join_tables(catalog):
+    from lakesoul_amd.io.schema import Field, Schema
+
+    o = catalog.create_table(
+        "jorders",
+        Schema([Field("oid", "int64", False), Field("cust", "int64"),
+                Field("amount", "float64")]),
+        primary_keys=["oid"], hash_bucket_num=2,
+    )
+    o.upsert({"oid": np.arange(20, dtype=np.int64),
+              "cust": np.arange(20, dtype=np.int64) % 5,
+              "amount": np.arange(20, dtype=np.float64) * 10})
+    c = catalog.create_table(
+        "jcust",
+        Schema([Field("cid", "int64", False), Field("name", "string")]),
+        primary_keys=["cid"], hash_bucket_num=1,
+    )
+    c.upsert({"cid": np.arange(4, dtype=np.int64),
+              "name": [f"cust{i}" for i in range(4)]})  # cust 4 has no row
+    return catalog
+
+
+def test_inner_join(join_tables):
+    df = execute_sql(join_tables,
+        "SELECT o.oid, c.name, o.amount FROM jorders o "
+        "JOIN jcust c ON o.cust = c.cid ORDER BY o.oid")
+    assert len(df) == 16  # cust 4 rows dropped
+    assert list(df.columns) == ["oid", "name", "amount"]
+    assert df[df.oid == 0]["name"].iloc[0] == "cust0"
+
+
+def test_left_join_and_where(join_tables):
+    df = execute_sql(join_tables,
+        "SELECT o.oid, c.name FROM jorders o LEFT JOIN jcust c "
+        "ON o.cust = c.cid WHERE o.amount > 50 ORDER BY o.oid")
+    assert len(df) == 14  # oids 6..19
+    assert df[df.oid == 9]["name"].isna().iloc[0]  # cust 4 unmatched
+
+
+def test_join_group_by(join_tables):
+    df = execute_sql(join_tables,
+        "SELECT c.name, sum(o.amount) total FROM jorders o "
+        "JOIN jcust c ON o.cust = c.cid GROUP BY c.name ORDER BY c.name")
+    # cust0: oids 0,5,10,15 -> 0+50+100+150 = 300
+    assert df[df.name == "cust0"]["total"].iloc[0] == 300.0
+
+
+def test_insert_values_and_select(catalog):
+    from lakesoul_amd.io.schema import Field, Schema
+
+    catalog.create_table(
+        "ins", Schema([Field("id", "int64", False), Field("v", "float64"),
+                       Field("s", "string")]),
+        primary_keys=["id"], hash_bucket_num=1,
+    )
+    r = execute_sql(catalog, "INSERT INTO ins VALUES (1, 1.5, 'a'), (2, 2.5, 'b')")
+    assert r["rows_inserted"].iloc[0] == 2
+    df = execute_sql(catalog, "SELECT * FROM ins ORDER BY id")
+    assert df["s"].tolist() == ["a", "b"]
+    # upsert semantics via INSERT on PK table
+    execute_sql(catalog, "INSERT INTO ins VALUES (2, 9.0, 'z')")
+    df = execute_sql(catalog, "SELECT v FROM ins WHERE id = 2")
+    assert df["v"].iloc[0] == 9.0
+    # INSERT ... SELECT into a second table
+    catalog.create_table(
+        "ins2", Schema([Field("id", "int64", False), Field("v", "float64")]),
+        primary_keys=["id"], hash_bucket_num=1,
+    )
+    r = execute_sql(catalog, "INSERT INTO ins2 (id, v) SELECT id, v FROM ins")
+    assert r["rows_inserted"].iloc[0] == 2
+    assert execute_sql(catalog, "SELECT count(*) n FROM ins2")["n"].iloc[0] == 2
+
+
+def test_single_table_qualifiers(sql_table):
+    cat, t = sql_table
+    df = execute_sql(cat, "SELECT orders.id FROM orders WHERE orders.id < 3 ORDER BY orders.id")
+    assert df["id"].tolist() == [0, 1, 2]
+    df = execute_sql(cat, "SELECT o.id FROM orders o WHERE o.id = 5")
+    assert df["id"].tolist() == [5]
+    with pytest.raises(SqlError):
+        execute_sql(cat, "SELECT x.id FROM orders o")
