@@ -162,6 +162,7 @@ class LakeSoulFlightServer(fl.FlightServerBase if _HAVE_FLIGHT else object):
             version=d.get("version"),
             partitions=d.get("partitions"),
             batch_size=d.get("batch_size"),
+            incremental=tuple(d["incremental"]) if d.get("incremental") else None,
         )
         if d.get("world_size"):
             scan = scan.shard(int(d.get("rank", 0)), int(d["world_size"]))

@@ -138,6 +138,7 @@ class LakeSoulTable:
         device: Optional[str] = None,
         batch_size: Optional[int] = None,
         options: Optional[Dict[str, str]] = None,
+        incremental: Optional[tuple] = None,
     ):
         """Build a LakeSoulScan (reference: catalog.py:740 LakeSoulScan)."""
         from ..io.reader import LakeSoulScan
@@ -152,6 +153,7 @@ class LakeSoulTable:
             device=device,
             batch_size=batch_size,
             options=options,
+            incremental=incremental,
         )
 
     def to_arrow(self, **kwargs):

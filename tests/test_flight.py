@@ -140,3 +140,20 @@ def test_rbac_domain_enforced(flight, catalog):
     t2 = client2.do_get(fl.Ticket(json.dumps({"table": "priv"}).encode()),
                         options=opts2).read_all()
     assert t2.num_rows == 0
+
+
+def test_do_get_incremental(flight):
+    """Streaming readers poll incremental tickets (reference: Flink
+    LakeSoulSource dynamic splits over new commits)."""
+    srv, catalog = flight
+    t = catalog.table("ft")
+    import numpy as np
+
+    t.upsert({"id": np.array([1000, 1001], dtype=np.int64),
+              "v": np.array([1.0, 2.0])})
+    client, opts = _client(srv)
+    ticket = fl.Ticket(json.dumps({"table": "ft", "incremental": [0, 10**9]}).encode())
+    inc = client.do_get(ticket, options=opts).read_all()
+    ids = set(inc.column("id").to_pylist())
+    assert {1000, 1001} <= ids
+    assert len(ids) < 102  # not the full table
