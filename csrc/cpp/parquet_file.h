@@ -27,6 +27,9 @@
 #include <string>
 #include <vector>
 
+#include <mutex>
+#include <unordered_map>
+
 #include "compress.h"
 #include "parquet_types.h"
 #include "rle.h"
@@ -429,6 +432,33 @@ class ParquetFile {
   // compressed blob), PLAIN-encoded, non-boolean v1 pages.
   ChunkData read_chunk(size_t rg, size_t col, bool gpu_snappy = false,
                        bool defer_host = false, bool gpu_zstd = false) const {
+    // descriptor cache: lakehouse files are immutable, and deferred /
+    // gpu-staged chunks are pure page descriptors (no payload) — cache
+    // them so repeated scans skip the per-page thrift header walk
+    // (reference analog: the global metadata cache, session.rs:86-105)
+    uint64_t ckey = ((uint64_t)rg << 24) | ((uint64_t)col << 4) |
+                    (defer_host ? 1u : 0u) | (gpu_snappy ? 2u : 0u) |
+                    (gpu_zstd ? 4u : 0u);
+    if (defer_host || gpu_snappy || gpu_zstd) {
+      std::lock_guard<std::mutex> lk(chunk_mu_);
+      auto it = chunk_meta_cache_.find(ckey);
+      if (it != chunk_meta_cache_.end()) return *it->second;
+    }
+    ChunkData out_cached = read_chunk_impl(rg, col, gpu_snappy, defer_host,
+                                           gpu_zstd);
+    if ((out_cached.host_deferred || out_cached.gpu_compressed) &&
+        out_cached.values.empty() && out_cached.validity.empty() &&
+        out_cached.dict.empty() && out_cached.comp.empty()) {
+      std::lock_guard<std::mutex> lk(chunk_mu_);
+      chunk_meta_cache_.emplace(
+          ckey, std::make_shared<const ChunkData>(out_cached));
+    }
+    return out_cached;
+  }
+
+  ChunkData read_chunk_impl(size_t rg, size_t col, bool gpu_snappy = false,
+                            bool defer_host = false,
+                            bool gpu_zstd = false) const {
     const RowGroup& g = meta_.row_groups.at(rg);
     const ColumnMeta& cm = g.columns.at(col);
     const ColumnDesc& cd = cols_.at(col);
@@ -477,7 +507,7 @@ class ParquetFile {
               ph.uncompressed_size <= (512 << 10))) &&
             ph.type == PAGE_DATA && ph.encoding == ENC_PLAIN)) {
         // mixed chunk (dict fallback / oversized page) — redo on host
-        return read_chunk(rg, col, false);
+        return read_chunk_impl(rg, col, false);
       }
       if (((gpu_snappy && cm.codec == CODEC_SNAPPY) ||
            (gpu_zstd && cm.codec == CODEC_ZSTD &&
@@ -520,7 +550,7 @@ class ParquetFile {
       if (out.host_deferred &&
           !((cm.codec == CODEC_ZSTD || cm.codec == CODEC_UNCOMPRESSED) &&
             ph.type == PAGE_DATA && ph.encoding == ENC_PLAIN)) {
-        return read_chunk(rg, col, false, false);  // mixed: redo staged
+        return read_chunk_impl(rg, col, false, false);  // mixed: redo staged
       }
 
       std::vector<uint8_t> page;
@@ -637,6 +667,10 @@ class ParquetFile {
       cols_.push_back(c);
     }
   }
+
+  mutable std::mutex chunk_mu_;
+  mutable std::unordered_map<uint64_t, std::shared_ptr<const ChunkData>>
+      chunk_meta_cache_;
 
   std::string path_;
   int fd_ = -1;
