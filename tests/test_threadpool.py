@@ -35,3 +35,24 @@ def test_concurrent_read_unit_raw(catalog):
             outs = list(ex.map(fetch, units * 2))
     total = sum(sum(o["file_rows"]) for o in outs) // 2
     assert total == n + 4 * 2000
+
+
+def test_tsan_clean(tmp_path):
+    """ThreadPool under ThreadSanitizer: no data races in the job-queue
+    handoff (SURVEY.md §5.2 race-detection item)."""
+    import subprocess, shutil, os
+
+    src = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+                       "csrc", "cpp", "tests", "threadpool_tsan.cc")
+    exe = str(tmp_path / "tp_tsan")
+    cc = shutil.which("g++")
+    if cc is None:
+        pytest.skip("no g++")
+    b = subprocess.run([cc, "-O1", "-g", "-std=c++17", "-fsanitize=thread",
+                        "-pthread", src, "-o", exe], capture_output=True, text=True)
+    if b.returncode != 0:
+        pytest.skip(f"tsan build unavailable: {b.stderr[-200:]}")
+    env = dict(os.environ, TSAN_OPTIONS="halt_on_error=1", LAKESOUL_POOL_THREADS="8")
+    r = subprocess.run([exe], capture_output=True, text=True, timeout=300, env=env)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "WARNING: ThreadSanitizer" not in r.stderr, r.stderr[:2000]
