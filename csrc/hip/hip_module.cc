@@ -35,6 +35,8 @@ void launch_bytes_ne_mask(const int64_t*, const uint8_t*, const uint8_t*, int, u
 template <typename T, typename ACC>
 void launch_segmented_sum(const T*, const int64_t*, const uint8_t*, const uint8_t*, ACC*, int32_t*, int64_t, hipStream_t);
 void launch_segmented_last(const int64_t*, const uint8_t*, const uint8_t*, int64_t*, int64_t, hipStream_t);
+void launch_hamming_scores(const uint64_t*, const uint64_t*, int32_t*, int64_t,
+                           int, int, hipStream_t);
 void launch_ann_scores(const short*, const short*, float*, int64_t, int32_t, int32_t, hipStream_t);
 void launch_snappy_decompress(const uint8_t*, const int64_t*, int64_t, uint8_t*, int32_t*, hipStream_t);
 void launch_zstd_decompress(const uint8_t*, const int64_t*, int64_t, uint8_t*,
@@ -627,6 +629,19 @@ static torch::Tensor zstd_decompress_into(torch::Tensor src, torch::Tensor jobs,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ann_scores", &ann_scores);
   m.def("zstd_decompress_into", &zstd_decompress_into);
+  m.def("hamming_scores", [](torch::Tensor codes, torch::Tensor qcodes) {
+    CHECK_GPU(codes);
+    CHECK_GPU(qcodes);
+    int64_t n = codes.size(0);
+    int nq = (int)qcodes.size(0);
+    int words = (int)codes.size(1);
+    TORCH_CHECK(words <= 16 && qcodes.size(1) == words);
+    auto out = torch::empty({n, nq}, codes.options().dtype(torch::kInt32));
+    launch_hamming_scores((const uint64_t*)codes.data_ptr<int64_t>(),
+                          (const uint64_t*)qcodes.data_ptr<int64_t>(),
+                          out.data_ptr<int32_t>(), n, nq, words, cur_stream());
+    return out;
+  });
   m.def("snappy_decompress", &snappy_decompress);
   m.def("snappy_decompress_into", &snappy_decompress_into);
   m.def("scan_unit_uselast", &scan_unit_uselast);

@@ -543,3 +543,39 @@ void launch_segmented_last(const int64_t* grp, const uint8_t* contrib,
 }
 
 }  // namespace lakesoul
+namespace lakesoul {
+
+// ------------------------------------------------------------------ //
+// binary-code hamming scorer (vector index first pass — the RaBitQ
+// 1-bit idea, lakesoul-vector quantizer.rs): codes are sign bits of
+// rotated vectors packed into u64 words. Each thread owns one database
+// row, keeps its W<=16 words in registers, loops the query block —
+// codes read once from HBM per launch.
+// ------------------------------------------------------------------ //
+
+__global__ void hamming_scores_kernel(const uint64_t* __restrict__ codes,
+                                      const uint64_t* __restrict__ qcodes,
+                                      int32_t* __restrict__ out, int64_t n,
+                                      int nq, int words) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  uint64_t c[16];
+  for (; i < n; i += stride) {
+    for (int w = 0; w < words; w++) c[w] = codes[i * words + w];
+    for (int q = 0; q < nq; q++) {
+      int d = 0;
+      for (int w = 0; w < words; w++)
+        d += __popcll(c[w] ^ qcodes[q * words + w]);
+      out[i * nq + q] = d;
+    }
+  }
+}
+
+void launch_hamming_scores(const uint64_t* codes, const uint64_t* qcodes,
+                           int32_t* out, int64_t n, int nq, int words,
+                           hipStream_t s) {
+  hipLaunchKernelGGL(hamming_scores_kernel, dim3(ls_blocks(n)), dim3(LS_THREADS),
+                     0, s, codes, qcodes, out, n, nq, words);
+}
+
+}  // namespace lakesoul

@@ -38,6 +38,29 @@ class ShardInfo:
     path: str
 
 
+def random_rotation(dim: int, seed: int = 7) -> np.ndarray:
+    """Seeded random orthonormal rotation (QR of a gaussian) — isotropizes
+    coordinates before 1-bit sign quantization (the reference's RaBitQ
+    rotation, rabitq/rotation.rs)."""
+    rng = np.random.default_rng(seed)
+    a = rng.normal(size=(dim, dim)).astype(np.float64)
+    q_m, r = np.linalg.qr(a)
+    q_m *= np.sign(np.diag(r))  # deterministic sign convention
+    return q_m.astype(np.float32)
+
+
+def pack_sign_bits(x: np.ndarray) -> np.ndarray:
+    """(n, dim) float -> (n, ceil(dim/64)) int64 sign codes (bit i of
+    word w = sign(x[:, 64*w + i]) >= 0)."""
+    n, dim = x.shape
+    bits = (x >= 0).astype(np.uint8)
+    pad = (-dim) % 64
+    if pad:
+        bits = np.concatenate([bits, np.zeros((n, pad), np.uint8)], axis=1)
+    packed = np.packbits(bits, axis=1, bitorder="little")
+    return packed.view(np.int64).reshape(n, -1)
+
+
 def kmeans(x: torch.Tensor, k: int, iters: int = 10, seed: int = 0) -> torch.Tensor:
     """Plain Lloyd k-means on the active device (the reference trains its
     IVF coarse quantizer the same way, rabitq/kmeans.rs). Returns
@@ -59,7 +82,7 @@ def kmeans(x: torch.Tensor, k: int, iters: int = 10, seed: int = 0) -> torch.Ten
 class VectorIndex:
     def __init__(self, root: str, column: str, dim: int, metric: str,
                  shards: List[ShardInfo], pk_dtype: str, version: int,
-                 ivf_clusters: int = 0):
+                 ivf_clusters: int = 0, binary: bool = False):
         self.root = root
         self.column = column
         self.dim = dim
@@ -68,6 +91,7 @@ class VectorIndex:
         self.pk_dtype = pk_dtype
         self.version = version
         self.ivf_clusters = ivf_clusters
+        self.binary = binary
         self._gpu_cache: dict = {}
 
     # -- persistence ---------------------------------------------------- #
@@ -83,8 +107,11 @@ class VectorIndex:
             "pk_dtype": self.pk_dtype,
             "version": self.version,
             "created_ms": int(time.time() * 1000),
-            "engine": "mfma-exact-bf16" if not self.ivf_clusters else "ivf+mfma-bf16",
+            "engine": ("binary+mfma-rescore" if self.binary
+                       else "mfma-exact-bf16" if not self.ivf_clusters
+                       else "ivf+mfma-bf16"),
             "ivf_clusters": self.ivf_clusters,
+            "binary": self.binary,
             "shards": [
                 {"bucket_id": s.bucket_id, "num_rows": s.num_rows, "path": s.path}
                 for s in self.shards
@@ -100,7 +127,8 @@ class VectorIndex:
             m = json.load(f)
         shards = [ShardInfo(s["bucket_id"], s["num_rows"], s["path"]) for s in m["shards"]]
         return cls(root, m["column"], m["dim"], m["metric"], shards,
-                   m["pk_dtype"], m["version"], m.get("ivf_clusters", 0))
+                   m["pk_dtype"], m["version"], m.get("ivf_clusters", 0),
+                   m.get("binary", False))
 
     # -- shard data ----------------------------------------------------- #
 
@@ -114,9 +142,25 @@ class VectorIndex:
         clu = None
         if self.ivf_clusters and os.path.exists(s.path + ".clu"):
             clu = torch.from_numpy(np.fromfile(s.path + ".clu", dtype=np.int64))
-        entry = (vecs, ids, clu)
+        codes = None
+        if self.binary:
+            w = (self.dim + 63) // 64
+            codes = torch.from_numpy(
+                np.fromfile(s.path + ".bin", dtype=np.int64).reshape(s.num_rows, w)
+            ).to(device)
+        entry = (vecs, ids, clu, codes)
         self._gpu_cache[key] = entry
         return entry
+
+    def _rotation(self) -> Optional[np.ndarray]:
+        if not self.binary:
+            return None
+        key = "__rotation__"
+        if key not in self._gpu_cache:
+            self._gpu_cache[key] = np.fromfile(
+                os.path.join(self.root, "rotation.vec"), dtype=np.float32
+            ).reshape(self.dim, self.dim)
+        return self._gpu_cache[key]
 
     def _centroids(self, device) -> Optional[torch.Tensor]:
         if not self.ivf_clusters:
@@ -131,7 +175,7 @@ class VectorIndex:
     # -- search --------------------------------------------------------- #
 
     def search(self, queries, k: int = 10, device: Optional[str] = None,
-               nprobe: Optional[int] = None):
+               nprobe: Optional[int] = None, rescore: int = 16):
         """Top-k over all shards. Returns (ids, scores) arrays of shape
         (nq, k). Cosine: inputs are normalized; score = cosine similarity.
         L2: score = -||x-q||^2 (larger is better).
@@ -154,8 +198,13 @@ class VectorIndex:
         cents = self._centroids(device)
         if cents is not None and nprobe is None:
             nprobe = max(4, self.ivf_clusters // 8)
+        qcodes = None
+        if self.binary:
+            rot = self._rotation()
+            qr = q.numpy() @ rot
+            qcodes = torch.from_numpy(pack_sign_bits(qr)).to(device)
         for s in self.shards:
-            vecs, ids, clu = self._load_shard(s, device)
+            vecs, ids, clu, codes = self._load_shard(s, device)
             if cents is not None and clu is not None:
                 # union of the nprobe nearest clusters over the query batch
                 cscores = q_dev.to(cents.dtype) @ cents.T  # (nq, k_c)
@@ -171,7 +220,30 @@ class VectorIndex:
                     continue
                 vecs = torch.cat([vecs[a:b] for a, b in segs])
                 ids = torch.cat([ids[a:b] for a, b in segs])
-            scores = self._scores(vecs, q_dev, device)  # (n, nq) f32
+                if codes is not None:
+                    codes = torch.cat([codes[a:b] for a, b in segs])
+            if codes is not None:
+                # 1-bit first pass: hamming distance on sign codes, then
+                # exact MFMA rescore of rescore*k candidates per query
+                nc = min(max(k * rescore, k), codes.shape[0])
+                if str(device).startswith("cuda"):
+                    from ..ops import hip
+
+                    ham = hip().hamming_scores(codes, qcodes)  # (n, nq) i32
+                else:
+                    x = codes.numpy().view(np.uint64)
+                    qq = qcodes.numpy().view(np.uint64)
+                    ham = torch.from_numpy(
+                        np.bitwise_count(x[:, None, :] ^ qq[None, :, :])
+                        .sum(axis=2).astype(np.int32))
+                cand = torch.topk(-ham.to(torch.float32), nc, dim=0).indices  # (nc, nq)
+                flat = torch.unique(cand.flatten())
+                sub_scores = self._scores(vecs[flat], q_dev, device)  # (m, nq)
+                scores = torch.full((vecs.shape[0], nq), -float("inf"),
+                                    device=sub_scores.device)
+                scores[flat] = sub_scores
+            else:
+                scores = self._scores(vecs, q_dev, device)  # (n, nq) f32
             kk = min(k, scores.shape[0])
             top = torch.topk(scores, kk, dim=0)  # (kk, nq)
             cand_scores = torch.cat([best_scores, top.values.T], dim=1)
@@ -215,6 +287,7 @@ def build_vector_index(
     metric: str = "cosine",
     device: Optional[str] = None,
     ivf_clusters: int = 0,
+    binary: bool = False,
 ) -> VectorIndex:
     """Build per-bucket exact-search shards for a fixed-size-list float
     column stored as ``dim`` float32/float64 scalar columns or via numpy
@@ -257,6 +330,10 @@ def build_vector_index(
         raise ValueError("no data to index")
 
     dev = device or ("cuda" if torch.cuda.is_available() else "cpu")
+    rot = None
+    if binary:
+        rot = random_rotation(dim)
+        rot.tofile(os.path.join(root, "rotation.vec"))
     centroids = None
     if ivf_clusters:
         # coarse quantizer on a sample (GPU k-means; rabitq/kmeans.rs analog)
@@ -281,11 +358,13 @@ def build_vector_index(
         vbf = torch.from_numpy(v_sorted).to(torch.bfloat16).view(torch.int16).numpy().view(np.uint16)
         vbf.tofile(spath + ".vec")
         ids_np[order].tofile(spath + ".ids")
+        if rot is not None:
+            pack_sign_bits(v_sorted @ rot).tofile(spath + ".bin")
         shards.append(ShardInfo(bucket_id, n, spath))
 
     version = table.latest_version() or 0
     idx = VectorIndex(root, column, int(dim), metric, shards, "int64", version,
-                      ivf_clusters=ivf_clusters)
+                      ivf_clusters=ivf_clusters, binary=binary)
     idx.save_manifest()
     return idx
 
