@@ -200,6 +200,21 @@ def create_app(catalog=None, secret: Optional[str] = None):
     def get_metrics(claims: dict = Depends(auth)):
         return metrics.snapshot()
 
+    @app.get("/metrics/prometheus")
+    def get_metrics_prom(claims: dict = Depends(auth)):
+        """Prometheus exposition format (reference: the s3-proxy and
+        flight server export prometheus counters, main.rs:44-52)."""
+        from fastapi.responses import PlainTextResponse
+
+        snap = metrics.snapshot()
+        lines = []
+        for k, v in snap.items():
+            name = f"lakesoul_{k}"
+            kind = "gauge" if k == "active_streams" else "counter"
+            lines.append(f"# TYPE {name} {kind}")
+            lines.append(f"{name} {v}")
+        return PlainTextResponse("\n".join(lines) + "\n")
+
     def _check_domain(t, claims: dict, write: bool = False):
         """RBAC: table domain must match the token's domain (analog of
         rbac.rs:19-50 verify_permission_by_table_name)."""

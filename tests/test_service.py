@@ -167,3 +167,12 @@ def test_s3_proxy_rbac(tmp_path, monkeypatch):
     assert r.status_code == 200
     m = client.get("/__metrics").json()
     assert m["denied"] == 1 and m["allowed"] >= 4
+
+
+def test_metrics_prometheus_format(app_client):
+    hdr = _auth(app_client)
+    r = app_client.get("/metrics/prometheus", headers=hdr)
+    assert r.status_code == 200
+    body = r.text
+    assert "# TYPE lakesoul_total_rows counter" in body
+    assert "lakesoul_active_streams" in body
