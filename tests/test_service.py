@@ -170,7 +170,7 @@ def test_s3_proxy_rbac(tmp_path, monkeypatch):
 
 
 def test_metrics_prometheus_format(app_client):
-    hdr = _auth(app_client)
+    hdr = _token(app_client)
     r = app_client.get("/metrics/prometheus", headers=hdr)
     assert r.status_code == 200
     body = r.text
