@@ -37,6 +37,8 @@ void launch_segmented_sum(const T*, const int64_t*, const uint8_t*, const uint8_
 void launch_segmented_last(const int64_t*, const uint8_t*, const uint8_t*, int64_t*, int64_t, hipStream_t);
 void launch_hamming_scores(const uint64_t*, const uint64_t*, int32_t*, int64_t,
                            int, int, hipStream_t);
+void launch_str_chunk_keys(const int64_t*, const uint8_t*, int64_t, int64_t*,
+                           int64_t, hipStream_t);
 void launch_ann_scores(const short*, const short*, float*, int64_t, int32_t, int32_t, hipStream_t);
 void launch_snappy_decompress(const uint8_t*, const int64_t*, int64_t, uint8_t*, int32_t*, hipStream_t);
 void launch_zstd_decompress(const uint8_t*, const int64_t*, int64_t, uint8_t*,
@@ -629,6 +631,17 @@ static torch::Tensor zstd_decompress_into(torch::Tensor src, torch::Tensor jobs,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ann_scores", &ann_scores);
   m.def("zstd_decompress_into", &zstd_decompress_into);
+  m.def("str_chunk_keys", [](torch::Tensor offsets, torch::Tensor bytes,
+                             int64_t chunk) {
+    CHECK_GPU(offsets);
+    CHECK_GPU(bytes);
+    int64_t n = offsets.numel() - 1;
+    auto out = torch::empty({n}, offsets.options().dtype(torch::kInt64));
+    launch_str_chunk_keys(offsets.data_ptr<int64_t>(),
+                          bytes.data_ptr<uint8_t>(), chunk,
+                          out.data_ptr<int64_t>(), n, cur_stream());
+    return out;
+  });
   m.def("hamming_scores", [](torch::Tensor codes, torch::Tensor qcodes) {
     CHECK_GPU(codes);
     CHECK_GPU(qcodes);

@@ -571,6 +571,36 @@ __global__ void hamming_scores_kernel(const uint64_t* __restrict__ codes,
   }
 }
 
+// 8-byte big-endian chunk key of each string (zero-padded past the
+// end), sign-flipped so int64 ascending sort = unsigned byte order.
+// LSD passes over chunks + a length pass give full lexicographic order
+// on the GPU (string-PK merge; reference cursors compare byte-wise,
+// sorted/cursor.rs).
+__global__ void str_chunk_keys_kernel(const int64_t* __restrict__ offsets,
+                                      const uint8_t* __restrict__ bytes,
+                                      int64_t chunk, int64_t* __restrict__ out,
+                                      int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    int64_t a = offsets[i], b = offsets[i + 1];
+    int64_t start = a + chunk * 8;
+    uint64_t key = 0;
+    for (int k = 0; k < 8; k++) {
+      uint64_t v = (start + k < b) ? bytes[start + k] : 0;
+      key = (key << 8) | v;
+    }
+    out[i] = (int64_t)(key ^ 0x8000000000000000ull);
+  }
+}
+
+void launch_str_chunk_keys(const int64_t* offsets, const uint8_t* bytes,
+                           int64_t chunk, int64_t* out, int64_t n,
+                           hipStream_t s) {
+  hipLaunchKernelGGL(str_chunk_keys_kernel, dim3(ls_blocks(n)),
+                     dim3(LS_THREADS), 0, s, offsets, bytes, chunk, out, n);
+}
+
 void launch_hamming_scores(const uint64_t* codes, const uint64_t* qcodes,
                            int32_t* out, int64_t n, int nq, int words,
                            hipStream_t s) {

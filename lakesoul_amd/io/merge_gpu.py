@@ -44,14 +44,55 @@ def _pack_keys_one_file(cols: List[Column]) -> Optional[torch.Tensor]:
     return None
 
 
+def _string_sort_tensors(cols: List[Column], device) -> List[torch.Tensor]:
+    """Most-significant-first sort keys for a string PK column: 8-byte
+    big-endian chunk keys (GPU kernel) then length as the final
+    tiebreak. Equality across all returned tensors == string equality,
+    and stable LSD passes over them give exact lexicographic order
+    (ROADMAP round-2 item pulled into round 1)."""
+    # per-row lengths -> absolute offsets of the concatenation
+    lens = torch.cat([
+        (c.offsets[1:] - c.offsets[:-1]).to(torch.int64).to(device) for c in cols
+    ])
+    offsets = torch.zeros(lens.numel() + 1, dtype=torch.int64, device=device)
+    torch.cumsum(lens, 0, out=offsets[1:])
+    bys = torch.cat([c.bytes_.to(device) for c in cols])
+    maxlen = int(lens.max().item()) if lens.numel() else 0
+    nchunks = max(1, (maxlen + 7) // 8)
+    if str(device).startswith("cuda"):
+        out = [hip().str_chunk_keys(offsets, bys, ci) for ci in range(nchunks)]
+    else:
+        # CPU reference of the kernel (tests without a GPU)
+        import numpy as np
+
+        n = lens.numel()
+        offs_np = offsets.cpu().numpy()
+        lens_np = lens.cpu().numpy()
+        b = bys.cpu().numpy()
+        L = nchunks * 8
+        idx = offs_np[:-1, None] + np.arange(L)[None, :]
+        mask = np.arange(L)[None, :] < lens_np[:, None]
+        padded = np.where(mask, b[np.minimum(idx, max(0, len(b) - 1))], 0)
+        keys = padded.reshape(n, nchunks, 8).astype(np.uint64)
+        shifts = np.uint64(8) * np.arange(7, -1, -1, dtype=np.uint64)
+        words = (keys << shifts[None, None, :]).sum(axis=2, dtype=np.uint64)
+        words ^= np.uint64(1 << 63)
+        out = [torch.from_numpy(words[:, ci].view(np.int64).copy())
+               for ci in range(nchunks)]
+    out.append(lens)
+    return out
+
+
 def merge_key_order(
     file_pk_cols: List[List[Column]], counts: List[int], device
 ):
-    """Return (order, sorted_keys_or_None): global row order (indices into
-    the concatenated rows) sorted by (pk..., file_seq, row) — the MOR
-    merge order. When the PK packs into a u64 the merge-path kernel is
-    used and the merged key array comes back for free (boundary
-    detection reuses it, no re-gather)."""
+    """Return (order, sorted_keys_or_None, eq_tensors_or_None): global row
+    order (indices into the concatenated rows) sorted by
+    (pk..., file_seq, row) — the MOR merge order. When the PK packs into
+    a u64 the merge-path kernel is used and the merged key array comes
+    back for free (boundary detection reuses it, no re-gather). For other
+    PKs (incl. strings) eq_tensors (concat space, most-significant-first)
+    drive both the stable LSD sort and group-boundary detection."""
     offsets = [0]
     for c in counts:
         offsets.append(offsets[-1] + c)
@@ -76,25 +117,24 @@ def merge_key_order(
             if len(streams) % 2:
                 nxt.append(streams[-1])
             streams = nxt
-        return streams[0][1], streams[0][0]
+        return streams[0][1], streams[0][0], None
 
-    # generic lexsort on concatenated keys (stable; ties keep concat order)
+    # generic stable LSD sort on concatenated keys (ties keep concat
+    # order = snapshot order, the MOR tie rule)
     perm = torch.arange(total, dtype=torch.int64, device=device)
-    cat_cols = []
+    sort_tensors: List[torch.Tensor] = []   # most-significant first
     npk = len(file_pk_cols[0])
     for ci in range(npk):
         cols = [fc[ci] for fc in file_pk_cols]
         if cols[0].is_string:
-            raise NotImplementedError(
-                "string primary keys on the GPU merge path are not yet "
-                "supported — scan with device='cpu'"
-            )
-        cat_cols.append(torch.cat([c.data for c in cols]))
-    for ci in range(npk - 1, -1, -1):
-        k = cat_cols[ci][perm]
+            sort_tensors.extend(_string_sort_tensors(cols, device))
+        else:
+            sort_tensors.append(torch.cat([c.data for c in cols]))
+    for t in reversed(sort_tensors):
+        k = t[perm]
         order = torch.argsort(k, stable=True)
         perm = perm[order]
-    return perm, None
+    return perm, None, sort_tensors
 
 
 def sorted_keys_for_order(
@@ -131,7 +171,7 @@ def merge_sorted_files_gpu(
     total = sum(counts)
     file_pk_cols = [[b.columns[p] for p in pk] for b in file_batches]
 
-    order, sorted_keys = merge_key_order(file_pk_cols, counts, device)
+    order, sorted_keys, eq_tensors = merge_key_order(file_pk_cols, counts, device)
 
     # group boundaries: from the merged u64 keys when available (one
     # kernel), else from gathered pk values
@@ -139,11 +179,11 @@ def merge_sorted_files_gpu(
     if sorted_keys is not None:
         start = hip().group_start_mask(sorted_keys).to(torch.bool) if n else torch.zeros(0, dtype=torch.bool, device=device)
     else:
-        sorted_pks = sorted_keys_for_order(file_pk_cols, order)
         start = torch.zeros(n, dtype=torch.bool, device=device)
         if n:
             start[0] = True
-            for sp in sorted_pks:
+            for t in eq_tensors:
+                sp = t[order]
                 start[1:] |= sp[1:] != sp[:-1]
     # last row index (in sorted space) per group
     end_mask = torch.zeros(n, dtype=torch.bool, device=device)

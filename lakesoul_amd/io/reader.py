@@ -361,14 +361,18 @@ class LakeSoulScan:
         return [fs.localize(p) if is_remote(p) else p for p in files]
 
     def _gpu_merge_supported(self) -> bool:
-        """String PKs need byte-wise lexicographic merge order, which the
-        GPU merge does not implement yet — those units run the CPU merge
-        (decode still benefits from the host pipeline)."""
+        """All PK types merge on the GPU now: integers via packed-u64
+        merge-path, strings via stable LSD passes over 8-byte chunk keys
+        (merge_gpu._string_sort_tensors). LAKESOUL_GPU_STRING_MERGE=0
+        forces the hybrid CPU-merge path for string PKs."""
         if not self.pk:
             return True
-        return all(
-            self.schema.field(p).dtype not in ("string", "binary") for p in self.pk
-        )
+        if os.environ.get("LAKESOUL_GPU_STRING_MERGE", "1") == "0":
+            return all(
+                self.schema.field(p).dtype not in ("string", "binary")
+                for p in self.pk
+            )
+        return True
 
     def _unit_fits(self, unit: ScanUnit) -> bool:
         """Estimate whether a bucket's decoded size fits the configured

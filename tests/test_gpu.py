@@ -302,8 +302,10 @@ def test_gpu_filters_and_projection(dev, tmp_path):
     assert df["v"].tolist() == [4321.0]
 
 
-def test_gpu_string_pk_hybrid(dev, tmp_path):
-    """String-PK tables scan on GPU via the hybrid CPU-merge path."""
+def test_gpu_string_pk_hybrid(dev, tmp_path, monkeypatch):
+    """String-PK tables scan on GPU via the hybrid CPU-merge path
+    (LAKESOUL_GPU_STRING_MERGE=0 escape hatch)."""
+    monkeypatch.setenv("LAKESOUL_GPU_STRING_MERGE", "0")
     catalog = _mk_catalog(tmp_path)
     from lakesoul_amd.io.schema import Field, Schema
 
@@ -490,3 +492,34 @@ def test_gpu_zstd_scan_path_active(dev, tmp_path, monkeypatch):
     import pandas as pd
 
     pd.testing.assert_frame_equal(cpu_df, gpu_df)
+
+
+@pytest.mark.gpu
+def test_gpu_string_pk_native_merge(dev, tmp_path):
+    """String-PK MOR merge fully on GPU (LSD chunk-key sort): results
+    match the CPU merge on a multi-delta history with shared prefixes."""
+    catalog = _mk_catalog(tmp_path)
+    from lakesoul_amd.io.schema import Field, Schema
+
+    t = catalog.create_table(
+        "gsnative",
+        Schema([Field("key", "string", False), Field("v", "int64")]),
+        primary_keys=["key"],
+        hash_bucket_num=2,
+    )
+    n = 30000
+    rng = np.random.default_rng(4)
+    t.upsert({"key": [f"user_{i:08d}" for i in range(n)],
+              "v": np.zeros(n, dtype=np.int64)})
+    for it in range(4):
+        ids = rng.choice(n, 5000, replace=False)
+        t.upsert({"key": [f"user_{i:08d}" for i in ids],
+                  "v": np.full(5000, it + 1, dtype=np.int64)})
+    t.upsert({"key": ["a", "user_00000005\x00x", "zz"],
+              "v": np.array([-1, -2, -3], dtype=np.int64)})
+    cpu = t.scan(device="cpu").to_arrow().to_pandas().sort_values("key").reset_index(drop=True)
+    gpu = t.scan(device="cuda").to_arrow().to_pandas().sort_values("key").reset_index(drop=True)
+    import pandas as pd
+
+    pd.testing.assert_frame_equal(cpu, gpu)
+    assert len(gpu) == n + 3

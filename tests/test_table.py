@@ -363,3 +363,32 @@ def test_decimal_sum_all_merge(catalog):
     df = _df(t)
     assert df["amt"].tolist() == [decimal.Decimal("1.00"), decimal.Decimal("3.00"),
                                   decimal.Decimal("0.75")]
+
+
+def test_string_pk_gpu_merge_logic_cpu(catalog):
+    """merge_key_order's string LSD path (CPU reference of the chunk-key
+    kernel) must equal python sorted() byte order, including tricky
+    shared prefixes, embedded NULs and length ties."""
+    import torch
+
+    from lakesoul_amd.io.batch import Batch
+    from lakesoul_amd.io.merge_gpu import merge_key_order
+    from lakesoul_amd.io.schema import Field, Schema
+
+    sch = Schema([Field("k", "string", False)])
+    f1 = ["apple", "apple\x00", "commonprefix_aaaa", "commonprefix_aaab", "z"]
+    f2 = ["app", "apple", "commonprefix_aaaa0", "commonprefix_aa", "za"]
+    b1 = Batch.from_dict({"k": sorted(f1)}, sch)
+    b2 = Batch.from_dict({"k": sorted(f2)}, sch)
+    cols = [[b1.columns["k"]], [b2.columns["k"]]]
+    order, keys, eq = merge_key_order(cols, [5, 5], torch.device("cpu"))
+    assert keys is None and eq is not None
+    allk = sorted(f1) + sorted(f2)
+    got = [allk[i] for i in order.tolist()]
+    # expected: stable sort by byte order, ties keep concat (file) order
+    expect = [s for s in sorted(got, key=lambda x: x.encode())]
+    assert [g.encode() for g in got] == [e.encode() for e in expect]
+    # equal adjacent strings must show equal across ALL eq tensors
+    for i in range(len(got) - 1):
+        same_eq = all(bool(t[order[i]] == t[order[i + 1]]) for t in eq)
+        assert same_eq == (got[i] == got[i + 1]), (got[i], got[i + 1])
