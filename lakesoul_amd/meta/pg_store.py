@@ -157,6 +157,23 @@ class PostgresMetaStore:  # pragma: no cover — needs a live PG server
         except Exception as e:  # unique violation = MVCC CAS failure
             raise CommitConflictError(str(e)) from e
 
+    def vacuum_partition_versions(self, table_id, partition_desc, cutoff,
+                                  stale_commit_ids) -> None:
+        """Atomic vacuum (same contract as SqliteMetaStore)."""
+        with self.conn.transaction():
+            with self.conn.cursor() as cur:
+                cur.execute(
+                    "DELETE FROM partition_info WHERE table_id=%s AND"
+                    " partition_desc=%s AND version<%s",
+                    (table_id, partition_desc, cutoff),
+                )
+                for cid in stale_commit_ids:
+                    cur.execute(
+                        "DELETE FROM data_commit_info WHERE table_id=%s AND"
+                        " partition_desc=%s AND commit_id=%s",
+                        (table_id, partition_desc, cid),
+                    )
+
     def insert_data_commit_info(self, dci: DataCommitInfo) -> None:
         self._exec(
             "INSERT INTO data_commit_info(table_id, partition_desc, commit_id,"

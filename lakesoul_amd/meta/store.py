@@ -496,6 +496,27 @@ class SqliteMetaStore:
                 (table_id, partition_desc, version),
             )
 
+    def vacuum_partition_versions(
+        self, table_id: str, partition_desc: str, cutoff: int,
+        stale_commit_ids: List[str],
+    ) -> None:
+        """Vacuum: atomically drop partition versions < cutoff and their
+        no-longer-referenced commit infos in ONE transaction (a reader or
+        concurrent committer never observes an empty version history —
+        ROADMAP hygiene item)."""
+        with self._conn() as c:
+            c.execute(
+                "DELETE FROM partition_info WHERE table_id=? AND"
+                " partition_desc=? AND version<?",
+                (table_id, partition_desc, cutoff),
+            )
+            for cid in stale_commit_ids:
+                c.execute(
+                    "DELETE FROM data_commit_info WHERE table_id=? AND"
+                    " partition_desc=? AND commit_id=?",
+                    (table_id, partition_desc, cid),
+                )
+
     # -- global config -------------------------------------------------- #
 
     def set_global_config(self, key: str, value: str) -> None:

@@ -401,20 +401,16 @@ class LakeSoulTable:
                 for f in self.client._resolve_snapshot_files(self.table_id, desc, p.snapshot):
                     kept_files.add(f.path)
             old_files = set()
+            stale_cids = []
             for p in old:
                 for f in self.client._resolve_snapshot_files(self.table_id, desc, p.snapshot):
                     old_files.add(f.path)
                 for cid in p.snapshot:
                     if cid not in kept_commits:
-                        store.delete_data_commit_info(self.table_id, desc, cid)
-            store.delete_partition_versions_since(self.table_id, desc, 0)
-            # reinsert kept versions (delete_partition_versions_since drops all)
-            from ..meta.store import CommitConflictError
-
-            try:
-                store.transaction_insert_partition_info(kept)
-            except CommitConflictError:
-                pass
+                        stale_cids.append(cid)
+            # one atomic transaction: versions < cutoff + stale commits go
+            # together; kept versions are never touched
+            store.vacuum_partition_versions(self.table_id, desc, cutoff, stale_cids)
             if delete_files:
                 for path in old_files - kept_files:
                     try:
