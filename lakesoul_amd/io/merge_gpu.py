@@ -342,6 +342,67 @@ def _apply_op_gpu(f, col: Column, op: str, order, grp, ngroups, seq_sorted,
         return Column(f.dtype, data=sums.to(col.data.dtype),
                       validity=None if bool(validity.all()) else validity)
 
+    if op in ("JoinedAllByComma", "JoinedAllBySemicolon",
+              "JoinedLastByComma", "JoinedLastBySemicolon"):
+        # delimiter-join of (all | last-per-file) values per PK group.
+        # Pure tensor algebra + the string gather kernel: participation
+        # mask -> per-group ranks -> value byte layout with 1-byte gaps
+        # for delimiters -> wave-per-row byte gather + delimiter scatter.
+        delim = ord(",") if op.endswith("Comma") else ord(";")
+        use_all = "All" in op
+        participate = torch.ones(n, dtype=torch.bool, device=device)
+        if contrib is not None:
+            participate &= contrib.to(torch.bool)
+        if not use_all:
+            los = torch.zeros(n, dtype=torch.bool, device=device)
+            if n:
+                los[-1] = True
+                los[:-1] = (seq_sorted[1:] != seq_sorted[:-1]) | start[1:]
+            participate &= los
+        # group is null if ANY participating row is null
+        if col.validity is not None:
+            null_part = participate & ~validity_sorted.to(torch.bool)
+            gnull = torch.zeros(ngroups, dtype=torch.bool, device=device)
+            gnull.index_put_((grp[null_part],),
+                             torch.ones((), dtype=torch.bool, device=device))
+        else:
+            gnull = torch.zeros(ngroups, dtype=torch.bool, device=device)
+        participate &= ~gnull[grp]
+
+        offs = col.offsets.to(torch.int64).to(device)
+        lens_all = offs[1:] - offs[:-1]
+        lens_sorted = lens_all[order] * participate
+        # rank of each participating row within its group (1-based)
+        cp = torch.cumsum(participate.to(torch.int64), 0)
+        gstarts = torch.nonzero(start, as_tuple=True)[0]
+        base = torch.where(gstarts > 0, cp[gstarts - 1],
+                           torch.zeros((), dtype=torch.int64, device=device))
+        rank = cp - base[grp]
+        has_delim = participate & (rank > 1)
+        span = lens_sorted + has_delim.to(torch.int64)
+        dst_end = torch.cumsum(span, 0)
+        dst_start = dst_end - span
+        total_bytes = int(dst_end[-1].item()) if n else 0
+        # per-group output offsets from group-end cumulative bytes
+        gends = torch.nonzero(end_mask, as_tuple=True)[0]
+        out_offs = torch.zeros(ngroups + 1, dtype=torch.int64, device=device)
+        if n:
+            out_offs[1:] = dst_end[gends]
+        sel = torch.nonzero(participate, as_tuple=True)[0]
+        src_rows = order[sel]
+        val_starts = dst_start[sel] + has_delim[sel].to(torch.int64)
+        # dst_offsets arg needs n+1 entries; only starts are read (the
+        # kernel uses source lengths) — append total as sentinel
+        dstoffs = torch.cat([val_starts, torch.tensor([total_bytes], device=device)])
+        out_bytes = hip().gather_strings(col.bytes_.to(device), offs, src_rows, dstoffs)
+        dpos = dst_start[sel[has_delim[sel]]] if bool(has_delim.any()) else None
+        if dpos is not None and dpos.numel():
+            out_bytes.index_put_((dpos,), torch.tensor(
+                delim, dtype=torch.uint8, device=device))
+        validity = (~gnull).to(torch.uint8)
+        return Column(col.dtype, None, out_offs, out_bytes,
+                      None if bool(gnull.logical_not().all()) else validity)
+
     raise NotImplementedError(
         f"merge operator {op} not supported on GPU — scan with device='cpu'"
     )

@@ -523,3 +523,46 @@ def test_gpu_string_pk_native_merge(dev, tmp_path):
 
     pd.testing.assert_frame_equal(cpu, gpu)
     assert len(gpu) == n + 3
+
+
+@pytest.mark.gpu
+def test_gpu_joined_merge_operators(dev, tmp_path):
+    """JoinedAllByComma / JoinedLastBySemicolon on the GPU match the CPU
+    oracle (delimiter-join per PK group, null poisoning)."""
+    catalog = _mk_catalog(tmp_path)
+    from lakesoul_amd.io.schema import Field, Schema
+
+    t = catalog.create_table(
+        "gjoin",
+        Schema([Field("id", "int64", False), Field("tags", "string"),
+                Field("last", "string")]),
+        primary_keys=["id"],
+        hash_bucket_num=2,
+        properties={"merge_op.tags": "JoinedAllByComma",
+                    "merge_op.last": "JoinedLastBySemicolon"},
+    )
+    n = 5000
+    t.upsert({"id": np.arange(n, dtype=np.int64),
+              "tags": [f"a{i}" for i in range(n)],
+              "last": [f"x{i}" for i in range(n)]})
+    rng = np.random.default_rng(6)
+    for it in range(3):
+        ids = np.sort(rng.choice(n, 800, replace=False)).astype(np.int64)
+        t.upsert({"id": ids,
+                  "tags": [f"b{it}_{i}" for i in ids],
+                  "last": [f"y{it}_{i}" for i in ids]})
+    # a null value poisons its group
+    import pyarrow as pa
+
+    t.upsert(pa.table({
+        "id": pa.array([0, 1], pa.int64()),
+        "tags": pa.array([None, "z"], pa.string()),
+        "last": pa.array(["q", None], pa.string()),
+    }))
+    cpu = t.scan(device="cpu").to_arrow().to_pandas().sort_values("id").reset_index(drop=True)
+    gpu = t.scan(device="cuda").to_arrow().to_pandas().sort_values("id").reset_index(drop=True)
+    import pandas as pd
+
+    pd.testing.assert_frame_equal(cpu, gpu)
+    assert cpu["tags"].iloc[0] is None or pd.isna(cpu["tags"].iloc[0])
+    assert "," in cpu["tags"].iloc[2]
