@@ -28,6 +28,13 @@ from .store import CommitConflictError, SqliteMetaStore, open_meta_store
 DEFAULT_MAX_RETRY = 25
 
 
+class ConcurrentReplaceError(Exception):
+    """A Compaction/Update commit raced another snapshot REPLACE: the
+    compacted output is based on a snapshot that no longer exists. Not
+    retryable at the commit layer — the caller must redo its work from
+    the new snapshot."""
+
+
 class MetaClient:
     def __init__(self, store: Optional[SqliteMetaStore] = None, max_retry: int = DEFAULT_MAX_RETRY):
         self.store = store if store is not None else open_meta_store()
@@ -152,18 +159,28 @@ class MetaClient:
                         version=-1,
                         domain=domain,
                     )
-                read_version = read_partition_map.get(
-                    part.partition_desc, PartitionInfo("", "", 0)
-                ).version
-                # snapshot replacement; on a concurrent change between read
-                # and commit the reference leaves conflict handling TODO
-                # (metadata_client.rs:609-620) — we take the replace branch
-                # only when the read version still matches, else keep the
-                # current snapshot and append compacted commits after it.
-                if cur.version < 0 or read_version == cur.version or not meta_info.read_partition_info:
+                read_info = read_partition_map.get(part.partition_desc)
+                # snapshot replacement. The reference leaves the
+                # concurrent-change case TODO (metadata_client.rs:609-620);
+                # here the strict rule is: the replacement covers exactly
+                # the snapshot the compaction READ. Commits that landed
+                # after the read (cur.snapshot extends read.snapshot) are
+                # re-appended ON TOP of the compacted commit; if the
+                # current snapshot is NOT an extension of the read one
+                # (another replace won the race), this compaction is stale
+                # and must abort.
+                if read_info is None or cur.version < 0 or                         read_info.version == cur.version:
                     snapshot = list(part.snapshot)
                 else:
-                    snapshot = list(part.snapshot)
+                    rs = list(read_info.snapshot)
+                    cs = list(cur.snapshot)
+                    if cs[: len(rs)] != rs:
+                        raise ConcurrentReplaceError(
+                            f"partition {part.partition_desc}: snapshot was "
+                            f"replaced concurrently (read v{read_info.version},"
+                            f" now v{cur.version}) — redo the compaction"
+                        )
+                    snapshot = list(part.snapshot) + cs[len(rs):]
                 nxt = PartitionInfo(
                     table_id=table_info.table_id,
                     partition_desc=part.partition_desc,

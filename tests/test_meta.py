@@ -305,3 +305,86 @@ def test_concurrent_commit_contention(tmp_path):
     for r in range(8):
         sub = df[(df.id >= r * 1000) & (df.id < r * 1000 + 500)]
         assert (sub["v"] == r * 10 + 3).all()  # last iteration wins
+
+
+def test_compaction_replace_preserves_concurrent_merge(catalog):
+    """A MergeCommit landing between a compaction's snapshot read and its
+    CompactionCommit must survive: the commit layer re-appends post-read
+    commits on top of the compacted snapshot (strict version of the
+    reference's TODO, metadata_client.rs:609-620)."""
+    import numpy as np
+
+    from lakesoul_amd.io.schema import Field, Schema
+
+    t = catalog.create_table(
+        "race",
+        Schema([Field("id", "int64", False), Field("v", "float64")]),
+        primary_keys=["id"], hash_bucket_num=1,
+    )
+    t.upsert({"id": np.arange(10, dtype=np.int64), "v": np.zeros(10)})
+    client = t.client
+    # compaction reads the CURRENT snapshot (v0)
+    read_info = client.store.get_latest_partition_info(t.table_id, "-5")
+    # ... meanwhile a merge commit lands (v1)
+    t.upsert({"id": np.array([3], dtype=np.int64), "v": np.array([9.0])})
+    # compaction commits its replacement based on the stale read
+    import lakesoul_amd.constants as C
+    from lakesoul_amd.meta.entities import (CommitOp, DataCommitInfo,
+                                            DataFileOp, FileOp, MetaInfo,
+                                            PartitionInfo)
+
+    # re-use v0's file as the "compacted" output for the test
+    v0_files = client._resolve_snapshot_files(t.table_id, "-5", read_info.snapshot)
+    dci = DataCommitInfo(
+        table_id=t.table_id, partition_desc="-5",
+        file_ops=[DataFileOp(f.path, FileOp.add, f.size) for f in v0_files],
+        commit_op=CommitOp.CompactionCommit,
+    )
+    client.store.insert_data_commit_info(dci)
+    client.commit_data(MetaInfo(
+        table_info=client.store.get_table_info_by_id(t.table_id),
+        list_partition=[PartitionInfo(
+            table_id=t.table_id, partition_desc="-5",
+            snapshot=[dci.commit_id], commit_op=CommitOp.CompactionCommit)],
+        read_partition_info=[read_info],
+    ), CommitOp.CompactionCommit)
+    # the concurrent merge's row survives
+    df = t.to_pandas()
+    assert df[df.id == 3]["v"].iloc[0] == 9.0
+
+
+def test_compaction_replace_aborts_on_double_replace(catalog):
+    import numpy as np
+
+    from lakesoul_amd.io.schema import Field, Schema
+    from lakesoul_amd.meta.client import ConcurrentReplaceError
+    from lakesoul_amd.meta.entities import (CommitOp, DataCommitInfo,
+                                            DataFileOp, FileOp, MetaInfo,
+                                            PartitionInfo)
+
+    t = catalog.create_table(
+        "race2",
+        Schema([Field("id", "int64", False), Field("v", "float64")]),
+        primary_keys=["id"], hash_bucket_num=1,
+    )
+    t.upsert({"id": np.arange(5, dtype=np.int64), "v": np.zeros(5)})
+    for it in range(3):
+        t.upsert({"id": np.array([it], dtype=np.int64), "v": np.array([1.0 + it])})
+    client = t.client
+    stale_read = client.store.get_latest_partition_info(t.table_id, "-5")
+    t.compaction()  # a real replace lands first
+    files = t.files()
+    dci = DataCommitInfo(
+        table_id=t.table_id, partition_desc="-5",
+        file_ops=[DataFileOp(f.path, FileOp.add, f.size) for f in files],
+        commit_op=CommitOp.CompactionCommit,
+    )
+    client.store.insert_data_commit_info(dci)
+    with pytest.raises(ConcurrentReplaceError):
+        client.commit_data(MetaInfo(
+            table_info=client.store.get_table_info_by_id(t.table_id),
+            list_partition=[PartitionInfo(
+                table_id=t.table_id, partition_desc="-5",
+                snapshot=[dci.commit_id], commit_op=CommitOp.CompactionCommit)],
+            read_partition_info=[stale_read],
+        ), CommitOp.CompactionCommit)
