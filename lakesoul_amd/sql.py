@@ -35,6 +35,7 @@ _KEYWORDS = {
     "or", "not", "in", "is", "null", "between", "as", "asc", "desc",
     "show", "tables", "namespaces", "describe", "distinct", "version",
     "join", "inner", "left", "on", "insert", "into", "values",
+    "update", "set", "delete",
 }
 
 _AGGS = {"count", "sum", "min", "max", "avg"}
@@ -155,6 +156,30 @@ class _Parser:
             return ("select", self.select())
         if (k, v) == ("kw", "insert"):
             return ("insert", self.insert())
+        if (k, v) == ("kw", "update"):
+            self.next()
+            ns, name = self.table_name()
+            self.expect("kw", "set")
+            assigns = {}
+            while True:
+                col = self.expect("id")
+                self.expect("op", "=")
+                assigns[col] = self.literal()
+                if not self.accept("op", ","):
+                    break
+            where = None
+            if self.accept("kw", "where"):
+                where = self.or_expr()
+            return ("update", {"namespace": ns, "table": name,
+                               "assignments": assigns, "where": where})
+        if (k, v) == ("kw", "delete"):
+            self.next()
+            self.expect("kw", "from")
+            ns, name = self.table_name()
+            where = None
+            if self.accept("kw", "where"):
+                where = self.or_expr()
+            return ("delete", {"namespace": ns, "table": name, "where": where})
         raise SqlError(f"unsupported statement start: {v!r}")
 
     def insert(self):
@@ -373,6 +398,15 @@ def execute_sql(catalog, sql: str, device: Optional[str] = None):
     kind, payload = parse_sql(sql)
     if kind == "insert":
         return _execute_insert(catalog, payload, device=device)
+    if kind in ("update", "delete"):
+        import pandas as pd
+
+        t = catalog.table(payload["table"], payload["namespace"])
+        if kind == "update":
+            n = t.update(payload["where"], payload["assignments"], device=device)
+            return pd.DataFrame({"rows_updated": [n]})
+        n = t.delete(payload["where"], device=device)
+        return pd.DataFrame({"rows_deleted": [n]})
     if kind == "show":
         if payload == "namespaces":
             return pd.DataFrame({"namespace": catalog.list_namespaces()})

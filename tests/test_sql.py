@@ -154,7 +154,7 @@ def test_errors(sql_table):
     with pytest.raises(SqlError):
         execute_sql(cat, "SELECT nosuch FROM orders")
     with pytest.raises(SqlError):
-        execute_sql(cat, "DELETE FROM orders")
+        execute_sql(cat, "DROP TABLE orders")
     with pytest.raises(SqlError):
         execute_sql(cat, "SELECT id, sum(qty) FROM orders")  # id not grouped
     with pytest.raises(SqlError):
@@ -260,3 +260,24 @@ def test_single_table_qualifiers(sql_table):
     assert df["id"].tolist() == [5]
     with pytest.raises(SqlError):
         execute_sql(cat, "SELECT x.id FROM orders o")
+
+
+def test_update_and_delete_statements(catalog):
+    catalog_tables = catalog
+    from lakesoul_amd.io.schema import Field, Schema
+
+    catalog.create_table(
+        "dml", Schema([Field("id", "int64", False), Field("v", "float64"),
+                       Field("tag", "string")]),
+        primary_keys=["id"], hash_bucket_num=2,
+    )
+    execute_sql(catalog, "INSERT INTO dml VALUES (1, 1.0, 'a'), (2, 2.0, 'b'), (3, 3.0, 'a')")
+    r = execute_sql(catalog, "UPDATE dml SET v = 99.0, tag = 'z' WHERE id >= 2")
+    assert r["rows_updated"].iloc[0] == 2
+    df = execute_sql(catalog, "SELECT id, v, tag FROM dml ORDER BY id")
+    assert df["v"].tolist() == [1.0, 99.0, 99.0]
+    assert df["tag"].tolist() == ["a", "z", "z"]
+    r = execute_sql(catalog, "DELETE FROM dml WHERE tag = 'z'")
+    assert r["rows_deleted"].iloc[0] == 2
+    df = execute_sql(catalog, "SELECT count(*) n FROM dml")
+    assert df["n"].iloc[0] == 1
