@@ -91,6 +91,10 @@ static std::string dtype_name(const ColumnDesc& c) {
     case PT_BYTE_ARRAY:
       if (c.converted == CV_UTF8 || c.logical == LogicalTag::STRING) return "string";
       return "binary";
+    case PT_FLBA:
+      return "binary";   // fixed-len values surface as binary cells
+    case PT_INT96:
+      return "timestamp[ns]";
     default:
       return "unsupported";
   }

@@ -36,6 +36,34 @@
 
 namespace lakesoul {
 
+// foreign physical-type translation (read side): FLBA -> length-prefixed
+// byte_array stream; INT96 (legacy impala/spark timestamps: 8B
+// nanos-of-day + 4B julian day, both LE) -> int64 nanoseconds since epoch
+inline void flba_to_byte_array(const uint8_t* src, int64_t n, int32_t len,
+                               std::vector<uint8_t>& out) {
+  out.reserve(out.size() + (size_t)n * (len + 4));
+  for (int64_t i = 0; i < n; i++) {
+    uint32_t l = (uint32_t)len;
+    const uint8_t* lp = (const uint8_t*)&l;
+    out.insert(out.end(), lp, lp + 4);
+    out.insert(out.end(), src + i * len, src + (i + 1) * len);
+  }
+}
+
+inline void int96_to_ns(const uint8_t* src, int64_t n,
+                        std::vector<uint8_t>& out) {
+  size_t base = out.size();
+  out.resize(base + (size_t)n * 8);
+  for (int64_t i = 0; i < n; i++) {
+    uint64_t nanos;
+    uint32_t jd;
+    std::memcpy(&nanos, src + i * 12, 8);
+    std::memcpy(&jd, src + i * 12 + 8, 4);
+    int64_t ns = ((int64_t)jd - 2440588) * 86400000000000LL + (int64_t)nanos;
+    std::memcpy(out.data() + base + i * 8, &ns, 8);
+  }
+}
+
 // ---------------------------------------------------------------------- //
 // column description (writer input / reader output)
 // ---------------------------------------------------------------------- //
@@ -50,6 +78,7 @@ struct ColumnDesc {
   bool int_signed = true;
   int32_t dec_precision = 0;
   int32_t dec_scale = 0;
+  int32_t type_length = 0;  // FLBA
 };
 
 inline int physical_elem_size(int32_t pt) {
@@ -463,7 +492,12 @@ class ParquetFile {
     const ColumnMeta& cm = g.columns.at(col);
     const ColumnDesc& cd = cols_.at(col);
     ChunkData out;
-    out.physical = cd.physical;
+    // foreign-file physical types are translated at page-decode time:
+    // FLBA becomes a length-prefixed byte_array stream, INT96 becomes
+    // timestamp[ns] int64 — everything downstream sees standard types
+    out.physical = cd.physical == PT_FLBA ? PT_BYTE_ARRAY
+                   : cd.physical == PT_INT96 ? PT_INT64
+                                             : cd.physical;
     out.num_values = cm.num_values;
 
     int64_t off = cm.dictionary_page_offset >= 0
@@ -493,9 +527,16 @@ class ParquetFile {
       off += hdr_len + ph.compressed_size;
 
       if (ph.type == PAGE_DICTIONARY) {
-        out.dict.resize(ph.uncompressed_size);
-        decompress_into(cm.codec, out.dict.data(), out.dict.size(), body,
+        std::vector<uint8_t> draw((size_t)ph.uncompressed_size);
+        decompress_into(cm.codec, draw.data(), draw.size(), body,
                         ph.compressed_size);
+        if (cd.physical == PT_FLBA)
+          flba_to_byte_array(draw.data(), ph.dict_num_values, cd.type_length,
+                             out.dict);
+        else if (cd.physical == PT_INT96)
+          int96_to_ns(draw.data(), ph.dict_num_values, out.dict);
+        else
+          out.dict = std::move(draw);
         out.dict_num_values = ph.dict_num_values;
         continue;
       }
@@ -514,6 +555,7 @@ class ParquetFile {
             ph.uncompressed_size <= (512 << 10))) &&
           ph.type == PAGE_DATA && ph.encoding == ENC_PLAIN && !cd.nullable &&
           cd.physical != PT_BOOLEAN && cd.physical != PT_BYTE_ARRAY &&
+          cd.physical != PT_FLBA && cd.physical != PT_INT96 &&
           out.dict.empty() && out.values.empty() && !out.host_deferred) {
         out.gpu_compressed = true;
         CompPage cp;
@@ -532,6 +574,7 @@ class ParquetFile {
           (cm.codec == CODEC_ZSTD || cm.codec == CODEC_UNCOMPRESSED) &&
           ph.type == PAGE_DATA && ph.encoding == ENC_PLAIN && !cd.nullable &&
           cd.physical != PT_BOOLEAN && cd.physical != PT_BYTE_ARRAY &&
+          cd.physical != PT_FLBA && cd.physical != PT_INT96 &&
           out.dict.empty() && out.values.empty() && !out.gpu_compressed) {
         // levels-free PLAIN page: defer decompression — fill() writes it
         // straight into the destination buffer
@@ -621,6 +664,10 @@ class ParquetFile {
           out.values.resize(base + (size_t)nonnull);
           for (int64_t i = 0; i < nonnull; i++)
             out.values[base + i] = (vals[i >> 3] >> (i & 7)) & 1;
+        } else if (cd.physical == PT_FLBA) {
+          flba_to_byte_array(vals, nonnull, cd.type_length, out.values);
+        } else if (cd.physical == PT_INT96) {
+          int96_to_ns(vals, nonnull, out.values);
         } else {
           out.values.insert(out.values.end(), vals, vals + vals_len);
         }
@@ -664,6 +711,7 @@ class ParquetFile {
       c.int_signed = e.int_signed;
       c.dec_precision = e.dec_precision;
       c.dec_scale = e.dec_scale;
+      c.type_length = e.type_length;
       cols_.push_back(c);
     }
   }

@@ -272,3 +272,65 @@ def test_decimal_batch_arrow_bridge():
     np.testing.assert_array_equal(b.columns["amt"].data.numpy()[[0, 2]], [150, -225])
     assert b.columns["amt"].validity.numpy().tolist() == [1, 0, 1]
     assert b.to_arrow().column("amt").to_pylist() == vals
+
+
+def test_flba_read(tmp_path):
+    """FLBA (fixed-length byte array) columns from foreign writers read
+    as binary cells."""
+    vals = [b"0123456789abcdef", b"ffffffffffffffff", b"0000000000000000"]
+    t = pa.table({"id": pa.array([1, 2, 3], pa.int64()),
+                  "u": pa.array(vals, pa.binary(16))})
+    path = str(tmp_path / "flba.parquet")
+    pq.write_table(t, path, use_dictionary=False, compression="zstd")
+    h = cpp().open_parquet(path)
+    try:
+        meta = cpp().parquet_meta(h)
+        ci = [i for i, c in enumerate(meta["columns"]) if c["name"] == "u"][0]
+        assert meta["columns"][ci]["dtype"] == "binary"
+        d = cpp().read_chunk_cpu(h, 0, ci)
+        offs = d["offsets"].numpy()
+        bys = d["bytes"].numpy().tobytes()
+        got = [bys[offs[i]:offs[i + 1]] for i in range(3)]
+    finally:
+        cpp().close_parquet(h)
+    assert got == vals
+
+
+def test_flba_read_dict_encoded(tmp_path):
+    vals = [b"aaaaaaaa", b"bbbbbbbb"] * 50
+    t = pa.table({"u": pa.array(vals, pa.binary(8))})
+    path = str(tmp_path / "flbad.parquet")
+    pq.write_table(t, path, use_dictionary=True, compression="snappy")
+    h = cpp().open_parquet(path)
+    try:
+        d = cpp().read_chunk_cpu(h, 0, 0)
+        offs = d["offsets"].numpy()
+        bys = d["bytes"].numpy().tobytes()
+        got = [bys[offs[i]:offs[i + 1]] for i in range(100)]
+    finally:
+        cpp().close_parquet(h)
+    assert got == vals
+
+
+def test_int96_read(tmp_path):
+    """Legacy INT96 timestamps (spark/impala) read as timestamp[ns]."""
+    import datetime
+
+    ts = [datetime.datetime(2020, 1, 1, 12, 0, 0),
+          datetime.datetime(1999, 12, 31, 23, 59, 59),
+          datetime.datetime(2026, 9, 11, 1, 2, 3, 456789)]
+    t = pa.table({"t": pa.array(ts, pa.timestamp("us"))})
+    path = str(tmp_path / "i96.parquet")
+    pq.write_table(t, path, use_deprecated_int96_timestamps=True,
+                   use_dictionary=False, compression="zstd")
+    h = cpp().open_parquet(path)
+    try:
+        meta = cpp().parquet_meta(h)
+        assert meta["columns"][0]["dtype"] == "timestamp[ns]"
+        d = cpp().read_chunk_cpu(h, 0, 0)
+        got = d["data"].numpy().view(np.int64)
+    finally:
+        cpp().close_parquet(h)
+    expect = np.array([int(x.replace(tzinfo=datetime.timezone.utc).timestamp() * 1e6) * 1000
+                       for x in ts], dtype=np.int64)
+    np.testing.assert_array_equal(got, expect)
