@@ -25,6 +25,9 @@ def main():
     p.add_argument("--k", type=int, default=10)
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--device", default=None)
+    p.add_argument("--binary", action="store_true",
+                   help="1-bit sign-code first pass + MFMA rescore")
+    p.add_argument("--rescore", type=int, default=16)
     args = p.parse_args()
     device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
 
@@ -35,10 +38,27 @@ def main():
     Q = torch.from_numpy(rng.normal(size=(args.nq, args.dim)).astype(np.float32))
     Q = (Q / Q.norm(dim=1, keepdim=True)).to(torch.bfloat16).to(device)
 
-    def search():
-        if device == "cuda":
-            from lakesoul_amd.ops import hip
+    codes = qcodes = None
+    if args.binary:
+        from lakesoul_amd.vector.index import pack_sign_bits, random_rotation
 
+        rot = random_rotation(args.dim)
+        codes = torch.from_numpy(pack_sign_bits(X.numpy() @ rot)).to(device)
+        qn = Q.to(torch.float32).cpu().numpy() @ rot
+        qcodes = torch.from_numpy(pack_sign_bits(qn)).to(device)
+
+    def search():
+        from lakesoul_amd.ops import hip
+
+        if args.binary and device == "cuda":
+            ham = hip().hamming_scores(codes, qcodes)
+            nc = min(args.k * args.rescore, args.n)
+            cand = torch.topk(-ham.to(torch.float16), nc, dim=0).indices
+            flat = torch.unique(cand.flatten())
+            sub = hip().ann_scores(Xb[flat].contiguous(), Q)
+            top = torch.topk(sub, args.k, dim=0)
+            return flat[top.indices]
+        if device == "cuda":
             scores = hip().ann_scores(Xb, Q)
         else:
             scores = Xb.to(torch.float32) @ Q.to(torch.float32).T
@@ -81,6 +101,7 @@ def main():
                 "tflops": flops / dt / 1e12,
                 "x_read_gb_per_s": gbytes * ((args.nq + 15) // 16) / dt,
                 "self_recall_at_1": recall,
+                "binary_first_pass": bool(args.binary),
                 "device": device,
             }
         )
