@@ -63,7 +63,7 @@ static DtypeInfo dtype_info(const std::string& dt) {
   throw std::runtime_error("unsupported dtype: " + dt);
 }
 
-static std::string dtype_name(const ColumnDesc& c) {
+static std::string dtype_name_scalar(const ColumnDesc& c) {
   switch (c.physical) {
     case PT_BOOLEAN:
       return "bool";
@@ -103,6 +103,11 @@ static std::string dtype_name(const ColumnDesc& c) {
 // ---------------------------------------------------------------------- //
 // writer binding
 // ---------------------------------------------------------------------- //
+
+static std::string dtype_name(const ColumnDesc& c) {
+  if (c.is_list) return "list<" + dtype_name_scalar(c) + ">";
+  return dtype_name_scalar(c);
+}
 
 static int64_t write_parquet(
     const std::string& path, const std::vector<std::string>& names,
@@ -292,12 +297,29 @@ static py::dict decoded_to_dict(DecodedColumn&& dc) {
 static py::dict read_chunk_cpu(int64_t h, int64_t rg, int64_t col) {
   auto f = get_file(h);
   DecodedColumn dc;
+  torch::Tensor list_offs, list_valid;
+  bool is_list = false;
   {
     py::gil_scoped_release rel;
     auto ch = f->read_chunk((size_t)rg, (size_t)col);
+    if (ch.is_list) {
+      is_list = true;
+      list_offs = torch::empty({(int64_t)ch.list_offsets.size()}, torch::kInt64);
+      std::memcpy(list_offs.data_ptr(), ch.list_offsets.data(),
+                  ch.list_offsets.size() * 8);
+      list_valid = torch::empty({(int64_t)ch.list_validity.size()}, torch::kUInt8);
+      if (!ch.list_validity.empty())
+        std::memcpy(list_valid.data_ptr(), ch.list_validity.data(),
+                    ch.list_validity.size());
+    }
     dc = decode_chunk_cpu(ch);
   }
-  return decoded_to_dict(std::move(dc));
+  py::dict d = decoded_to_dict(std::move(dc));
+  if (is_list) {
+    d["list_offsets"] = list_offs;
+    d["list_validity"] = list_valid;
+  }
+  return d;
 }
 
 // batch: parallel host decode across (rg,col) pairs

@@ -79,6 +79,11 @@ struct ColumnDesc {
   int32_t dec_precision = 0;
   int32_t dec_scale = 0;
   int32_t type_length = 0;  // FLBA
+  // LIST columns (3-level standard layout): the desc describes the leaf
+  // ELEMENT; list structure decodes from rep/def levels
+  bool is_list = false;
+  bool elem_nullable = false;
+  int32_t max_def = 0;      // 0 for flat columns (unused)
 };
 
 inline int physical_elem_size(int32_t pt) {
@@ -452,6 +457,11 @@ class ParquetFile {
     // later directly into the final buffer (no staging copy)
     bool host_deferred = false;
     std::vector<DeferPage> defer_pages;
+    // LIST columns: rows -> element ranges (+1 sentinel) and per-row
+    // validity; values/validity above then describe the ELEMENTS
+    bool is_list = false;
+    std::vector<int64_t> list_offsets;
+    std::vector<uint8_t> list_validity;
   };
 
   const uint8_t* data_at(int64_t off) const { return map_ + off; }
@@ -555,7 +565,7 @@ class ParquetFile {
             ph.uncompressed_size <= (512 << 10))) &&
           ph.type == PAGE_DATA && ph.encoding == ENC_PLAIN && !cd.nullable &&
           cd.physical != PT_BOOLEAN && cd.physical != PT_BYTE_ARRAY &&
-          cd.physical != PT_FLBA && cd.physical != PT_INT96 &&
+          cd.physical != PT_FLBA && cd.physical != PT_INT96 && !cd.is_list &&
           out.dict.empty() && out.values.empty() && !out.host_deferred) {
         out.gpu_compressed = true;
         CompPage cp;
@@ -574,7 +584,7 @@ class ParquetFile {
           (cm.codec == CODEC_ZSTD || cm.codec == CODEC_UNCOMPRESSED) &&
           ph.type == PAGE_DATA && ph.encoding == ENC_PLAIN && !cd.nullable &&
           cd.physical != PT_BOOLEAN && cd.physical != PT_BYTE_ARRAY &&
-          cd.physical != PT_FLBA && cd.physical != PT_INT96 &&
+          cd.physical != PT_FLBA && cd.physical != PT_INT96 && !cd.is_list &&
           out.dict.empty() && out.values.empty() && !out.gpu_compressed) {
         // levels-free PLAIN page: defer decompression — fill() writes it
         // straight into the destination buffer
@@ -600,7 +610,10 @@ class ParquetFile {
       const uint8_t* vals;
       size_t vals_len;
       int64_t nv = ph.num_values;
-      int64_t page_nulls = 0;
+      int64_t page_nulls = 0;          // entries without a stored value
+      int64_t page_markers = 0;        // list row-marker entries (not null
+                                       // ELEMENTS — excluded from null_count)
+      int64_t elem_slots_page_start = (int64_t)out.validity.size();
 
       if (ph.type == PAGE_DATA) {
         page.resize(ph.uncompressed_size);
@@ -608,7 +621,43 @@ class ParquetFile {
                         ph.compressed_size);
         const uint8_t* p = page.data();
         size_t rem = page.size();
-        if (cd.nullable) {
+        if (cd.is_list) {
+          // standard 3-level LIST: [u32][rep RLE][u32][def RLE][values].
+          // rep==0 starts a row; def encodes null-list / empty-list /
+          // null-element / value (parquet.thrift nested encoding)
+          out.is_list = true;
+          uint32_t rl_len, dl_len;
+          std::memcpy(&rl_len, p, 4);
+          std::vector<uint8_t> reps((size_t)nv);
+          rle_decode<uint8_t>(p + 4, rl_len, 1, nv, reps.data());
+          p += 4 + rl_len;
+          rem -= 4 + rl_len;
+          std::memcpy(&dl_len, p, 4);
+          int def_bw = 1;
+          while ((1 << def_bw) <= cd.max_def) def_bw++;
+          std::vector<uint8_t> defs((size_t)nv);
+          rle_decode<uint8_t>(p + 4, dl_len, def_bw, nv, defs.data());
+          p += 4 + dl_len;
+          rem -= 4 + dl_len;
+          int elem_threshold = cd.max_def - (cd.elem_nullable ? 1 : 0);
+          for (int64_t i = 0; i < nv; i++) {
+            if (reps[i] == 0) {
+              out.list_offsets.push_back((int64_t)out.validity.size());
+              out.list_validity.push_back(
+                  (cd.nullable && defs[i] == 0) ? 0 : 1);
+            }
+            if (defs[i] >= elem_threshold) {
+              bool present = defs[i] == cd.max_def;
+              out.validity.push_back(present ? 1 : 0);
+              if (!present) page_nulls++;  // null ELEMENT (no stored value)
+            } else {
+              page_nulls++;      // row marker entry carries no value
+              page_markers++;
+            }
+          }
+          vals = p;
+          vals_len = rem;
+        } else if (cd.nullable) {
           if (ph.def_encoding != ENC_RLE)
             throw std::runtime_error("unsupported def-level encoding");
           uint32_t lv_len;
@@ -628,6 +677,8 @@ class ParquetFile {
         vals = p;
         vals_len = rem;
       } else {  // DATA_PAGE_V2: levels uncompressed, values possibly compressed
+        if (cd.is_list)
+          throw std::runtime_error("LIST columns: DataPageV2 not supported yet");
         int64_t lv = ph.rep_levels_byte_length + ph.def_levels_byte_length;
         if (cd.nullable && ph.def_levels_byte_length > 0) {
           std::vector<uint8_t> levels((size_t)nv);
@@ -654,8 +705,8 @@ class ParquetFile {
         vals_len = uncomp_vals;
       }
       if (page_nulls) any_null_page = true;
-      out.null_count += page_nulls;
-      int64_t nonnull = nv - page_nulls;
+      out.null_count += page_nulls - page_markers;  // null ELEMENTS only
+      int64_t nonnull = nv - page_nulls;            // stored values
 
       if (ph.encoding == ENC_PLAIN) {
         if (cd.physical == PT_BOOLEAN) {
@@ -676,7 +727,9 @@ class ParquetFile {
         out.is_dict = true;
         int bw = vals[0];
         IdxPage ip;
-        ip.out_off = values_seen;  // row offset (incl. nulls); indices are dense
+        // row offset incl. nulls (element-slot offset for LIST columns);
+        // indices themselves are dense
+        ip.out_off = cd.is_list ? elem_slots_page_start : values_seen;
         ip.n = nonnull;
         ip.payload_off = (int64_t)out.values.size();
         ip.payload_len = (int64_t)(vals_len - 1);
@@ -690,6 +743,11 @@ class ParquetFile {
       values_seen += nv;
     }
     (void)any_null_page;
+    if (cd.is_list) {
+      out.list_offsets.push_back((int64_t)out.validity.size());
+      out.num_values = (int64_t)out.validity.size();  // element slots
+      if (out.null_count == 0) out.validity.clear();  // all elements valid
+    }
     return out;
   }
 
@@ -699,8 +757,37 @@ class ParquetFile {
     // flat schema: every element after root with num_children==0
     for (size_t i = 1; i < meta_.schema.size(); i++) {
       const SchemaElement& e = meta_.schema[i];
-      if (e.num_children > 0)
+      if (e.num_children > 0) {
+        // accept exactly the standard 3-level LIST shape:
+        //   [optional] group <name> (LIST) { repeated group list {
+        //       [optional] <leaf> element; } }
+        // (CV_LIST converted value = 3)
+        bool is_list_group = (e.converted == 3);
+        if (is_list_group && i + 2 < meta_.schema.size() &&
+            meta_.schema[i + 1].repetition == REP_REPEATED &&
+            meta_.schema[i + 1].num_children == 1 &&
+            meta_.schema[i + 2].num_children == 0) {
+          const SchemaElement& leaf = meta_.schema[i + 2];
+          ColumnDesc c;
+          c.name = e.name;
+          c.physical = leaf.type;
+          c.nullable = e.repetition == REP_OPTIONAL;  // list nullability
+          c.converted = leaf.converted;
+          c.logical = leaf.logical;
+          c.int_bit_width = leaf.int_bit_width;
+          c.int_signed = leaf.int_signed;
+          c.dec_precision = leaf.dec_precision;
+          c.dec_scale = leaf.dec_scale;
+          c.type_length = leaf.type_length;
+          c.is_list = true;
+          c.elem_nullable = leaf.repetition == REP_OPTIONAL;
+          c.max_def = (c.nullable ? 1 : 0) + 1 + (c.elem_nullable ? 1 : 0);
+          cols_.push_back(c);
+          i += 2;
+          continue;
+        }
         throw std::runtime_error("nested schemas not supported yet: " + e.name);
+      }
       ColumnDesc c;
       c.name = e.name;
       c.physical = e.type;

@@ -334,3 +334,78 @@ def test_int96_read(tmp_path):
     expect = np.array([int(x.replace(tzinfo=datetime.timezone.utc).timestamp() * 1e6) * 1000
                        for x in ts], dtype=np.int64)
     np.testing.assert_array_equal(got, expect)
+
+
+def _read_list_col(path, name):
+    h = cpp().open_parquet(path)
+    try:
+        meta = cpp().parquet_meta(h)
+        ci = [i for i, c in enumerate(meta["columns"]) if c["name"] == name][0]
+        dt = meta["columns"][ci]["dtype"]
+        parts = []
+        for rg in range(meta["num_row_groups"]):
+            parts.append(cpp().read_chunk_cpu(h, rg, ci))
+        return dt, parts
+    finally:
+        cpp().close_parquet(h)
+
+
+def test_list_int64_read(tmp_path):
+    """Standard 3-level LIST<int64>: rep/def level decode vs pyarrow."""
+    data = [[1, 2, 3], [], None, [4], [5, None, 7], [8, 9]]
+    t = pa.table({"l": pa.array(data, pa.list_(pa.int64()))})
+    path = str(tmp_path / "list.parquet")
+    pq.write_table(t, path, use_dictionary=False, compression="zstd")
+    dt, parts = _read_list_col(path, "l")
+    assert dt == "list<int64>"
+    d = parts[0]
+    offs = d["list_offsets"].numpy()
+    lv = d["list_validity"].numpy()
+    vals = d["data"].numpy().view(np.int64)
+    ev = d["validity"].numpy()
+    got = []
+    for r in range(len(offs) - 1):
+        if not lv[r]:
+            got.append(None)
+            continue
+        row = []
+        for e in range(offs[r], offs[r + 1]):
+            row.append(None if (len(ev) and not ev[e]) else int(vals[e]))
+        got.append(row)
+    assert got == data
+
+
+def test_list_string_dict_read(tmp_path):
+    data = [["a", "bb"], ["a"], [], ["ccc", "a", "bb"]]
+    t = pa.table({"s": pa.array(data, pa.list_(pa.string()))})
+    path = str(tmp_path / "lists.parquet")
+    pq.write_table(t, path, use_dictionary=True, compression="snappy")
+    dt, parts = _read_list_col(path, "s")
+    assert dt == "list<string>"
+    d = parts[0]
+    offs = d["list_offsets"].numpy()
+    soffs = d["offsets"].numpy()
+    bys = d["bytes"].numpy().tobytes()
+    got = []
+    for r in range(len(offs) - 1):
+        row = [bys[soffs[e]:soffs[e + 1]].decode() for e in range(offs[r], offs[r + 1])]
+        got.append(row)
+    assert got == data
+
+
+def test_list_multirowgroup_and_longrows(tmp_path):
+    rng = np.random.default_rng(0)
+    data = [list(map(int, rng.integers(0, 100, int(rng.integers(0, 30)))))
+            for _ in range(5000)]
+    t = pa.table({"l": pa.array(data, pa.list_(pa.int64()))})
+    path = str(tmp_path / "listbig.parquet")
+    pq.write_table(t, path, use_dictionary=False, compression="zstd",
+                   row_group_size=1700)
+    dt, parts = _read_list_col(path, "l")
+    got = []
+    for d in parts:
+        offs = d["list_offsets"].numpy()
+        vals = d["data"].numpy().view(np.int64)
+        for r in range(len(offs) - 1):
+            got.append([int(vals[e]) for e in range(offs[r], offs[r + 1])])
+    assert got == data
