@@ -662,10 +662,13 @@ class ParquetFile {
             throw std::runtime_error("unsupported def-level encoding");
           uint32_t lv_len;
           std::memcpy(&lv_len, p, 4);
+          int def_bw = 1;
+          while ((1 << def_bw) <= cd.max_def) def_bw++;
           std::vector<uint8_t> levels((size_t)nv);
-          rle_decode<uint8_t>(p + 4, lv_len, 1, nv, levels.data());
+          rle_decode<uint8_t>(p + 4, lv_len, def_bw, nv, levels.data());
+          uint8_t full = (uint8_t)cd.max_def;
           for (int64_t i = 0; i < nv; i++) {
-            if (!levels[i]) {
+            if (levels[i] != full) {   // null at leaf OR a null ancestor
               ensure_validity();
               out.validity[values_seen + i] = 0;
               page_nulls++;
@@ -755,23 +758,29 @@ class ParquetFile {
   void build_columns() {
     if (meta_.schema.empty()) throw std::runtime_error("empty schema");
     // flat schema: every element after root with num_children==0
-    for (size_t i = 1; i < meta_.schema.size(); i++) {
-      const SchemaElement& e = meta_.schema[i];
-      if (e.num_children > 0) {
-        // accept exactly the standard 3-level LIST shape:
-        //   [optional] group <name> (LIST) { repeated group list {
-        //       [optional] <leaf> element; } }
-        // (CV_LIST converted value = 3)
-        bool is_list_group = (e.converted == 3);
-        if (is_list_group && i + 2 < meta_.schema.size() &&
+    size_t i = 1;
+    while (i < meta_.schema.size())
+      i = flatten_element(i, "", 0);
+  }
+
+  // recursive schema flatten: returns the index after the subtree.
+  // prefix names nested leaves "outer.inner"; def_base accumulates
+  // optional ancestors so leaf def-levels decode with the right width
+  // and null threshold. LIST groups use the dedicated 3-level path.
+  size_t flatten_element(size_t i, const std::string& prefix, int def_base) {
+    const SchemaElement& e = meta_.schema[i];
+    std::string name = prefix.empty() ? e.name : prefix + "." + e.name;
+    if (e.num_children > 0) {
+      if (e.converted == 3) {  // LIST
+        if (i + 2 < meta_.schema.size() &&
             meta_.schema[i + 1].repetition == REP_REPEATED &&
             meta_.schema[i + 1].num_children == 1 &&
-            meta_.schema[i + 2].num_children == 0) {
+            meta_.schema[i + 2].num_children == 0 && def_base == 0) {
           const SchemaElement& leaf = meta_.schema[i + 2];
           ColumnDesc c;
-          c.name = e.name;
+          c.name = name;
           c.physical = leaf.type;
-          c.nullable = e.repetition == REP_OPTIONAL;  // list nullability
+          c.nullable = e.repetition == REP_OPTIONAL;
           c.converted = leaf.converted;
           c.logical = leaf.logical;
           c.int_bit_width = leaf.int_bit_width;
@@ -783,24 +792,34 @@ class ParquetFile {
           c.elem_nullable = leaf.repetition == REP_OPTIONAL;
           c.max_def = (c.nullable ? 1 : 0) + 1 + (c.elem_nullable ? 1 : 0);
           cols_.push_back(c);
-          i += 2;
-          continue;
+          return i + 3;
         }
-        throw std::runtime_error("nested schemas not supported yet: " + e.name);
+        throw std::runtime_error("unsupported LIST shape: " + name);
       }
-      ColumnDesc c;
-      c.name = e.name;
-      c.physical = e.type;
-      c.nullable = e.repetition == REP_OPTIONAL;
-      c.converted = e.converted;
-      c.logical = e.logical;
-      c.int_bit_width = e.int_bit_width;
-      c.int_signed = e.int_signed;
-      c.dec_precision = e.dec_precision;
-      c.dec_scale = e.dec_scale;
-      c.type_length = e.type_length;
-      cols_.push_back(c);
+      if (e.repetition == REP_REPEATED)
+        throw std::runtime_error("repeated groups (MAP/legacy lists) not "
+                                 "supported yet: " + name);
+      // plain struct: flatten children as dotted leaf columns
+      int base = def_base + (e.repetition == REP_OPTIONAL ? 1 : 0);
+      size_t j = i + 1;
+      for (int k = 0; k < e.num_children; k++)
+        j = flatten_element(j, name, base);
+      return j;
     }
+    ColumnDesc c;
+    c.name = name;
+    c.physical = e.type;
+    c.max_def = def_base + (e.repetition == REP_OPTIONAL ? 1 : 0);
+    c.nullable = c.max_def > 0;  // null if ANY optional ancestor is null
+    c.converted = e.converted;
+    c.logical = e.logical;
+    c.int_bit_width = e.int_bit_width;
+    c.int_signed = e.int_signed;
+    c.dec_precision = e.dec_precision;
+    c.dec_scale = e.dec_scale;
+    c.type_length = e.type_length;
+    cols_.push_back(c);
+    return i + 1;
   }
 
   mutable std::mutex chunk_mu_;

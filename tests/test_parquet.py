@@ -409,3 +409,37 @@ def test_list_multirowgroup_and_longrows(tmp_path):
         for r in range(len(offs) - 1):
             got.append([int(vals[e]) for e in range(offs[r], offs[r + 1])])
     assert got == data
+
+
+def test_struct_read_flattened(tmp_path):
+    """Plain struct columns flatten to dotted leaf columns with correct
+    multi-level def decoding (null struct vs null field)."""
+    t = pa.table({
+        "id": pa.array([1, 2, 3, 4], pa.int64()),
+        "s": pa.array(
+            [{"a": 10, "b": "x"}, None, {"a": None, "b": "y"}, {"a": 40, "b": None}],
+            pa.struct([("a", pa.int64()), ("b", pa.string())])),
+    })
+    path = str(tmp_path / "struct.parquet")
+    pq.write_table(t, path, use_dictionary=False, compression="zstd")
+    h = cpp().open_parquet(path)
+    try:
+        meta = cpp().parquet_meta(h)
+        names = [c["name"] for c in meta["columns"]]
+        assert names == ["id", "s.a", "s.b"]
+        ai = names.index("s.a")
+        bi = names.index("s.b")
+        da = cpp().read_chunk_cpu(h, 0, ai)
+        db = cpp().read_chunk_cpu(h, 0, bi)
+    finally:
+        cpp().close_parquet(h)
+    av = da["data"].numpy().view(np.int64)
+    avm = da["validity"].numpy()
+    assert avm.tolist() == [1, 0, 0, 1]
+    assert av[0] == 10 and av[3] == 40
+    bvm = db["validity"].numpy()
+    assert bvm.tolist() == [1, 0, 1, 0]
+    offs = db["offsets"].numpy()
+    bys = db["bytes"].numpy().tobytes()
+    vals = [bys[offs[i]:offs[i + 1]].decode() for i in range(4)]
+    assert vals[0] == "x" and vals[2] == "y"
