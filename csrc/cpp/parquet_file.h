@@ -796,6 +796,37 @@ class ParquetFile {
         }
         throw std::runtime_error("unsupported LIST shape: " + name);
       }
+      if ((e.converted == 1 || e.converted == 2) && def_base == 0) {  // MAP / legacy MAP_KEY_VALUE
+        // standard shape: group (MAP) { repeated group key_value {
+        //   required key; [optional] value; } } -> surface as TWO
+        // parallel list columns name.key / name.value (same row ranges)
+        if (i + 3 < meta_.schema.size() &&
+            meta_.schema[i + 1].repetition == REP_REPEATED &&
+            meta_.schema[i + 1].num_children == 2 &&
+            meta_.schema[i + 2].num_children == 0 &&
+            meta_.schema[i + 3].num_children == 0) {
+          for (int k = 0; k < 2; k++) {
+            const SchemaElement& leaf = meta_.schema[i + 2 + k];
+            ColumnDesc c;
+            c.name = name + "." + leaf.name;
+            c.physical = leaf.type;
+            c.nullable = e.repetition == REP_OPTIONAL;
+            c.converted = leaf.converted;
+            c.logical = leaf.logical;
+            c.int_bit_width = leaf.int_bit_width;
+            c.int_signed = leaf.int_signed;
+            c.dec_precision = leaf.dec_precision;
+            c.dec_scale = leaf.dec_scale;
+            c.type_length = leaf.type_length;
+            c.is_list = true;
+            c.elem_nullable = leaf.repetition == REP_OPTIONAL;
+            c.max_def = (c.nullable ? 1 : 0) + 1 + (c.elem_nullable ? 1 : 0);
+            cols_.push_back(c);
+          }
+          return i + 4;
+        }
+        throw std::runtime_error("unsupported MAP shape: " + name);
+      }
       if (e.repetition == REP_REPEATED)
         throw std::runtime_error("repeated groups (MAP/legacy lists) not "
                                  "supported yet: " + name);

@@ -443,3 +443,35 @@ def test_struct_read_flattened(tmp_path):
     bys = db["bytes"].numpy().tobytes()
     vals = [bys[offs[i]:offs[i + 1]].decode() for i in range(4)]
     assert vals[0] == "x" and vals[2] == "y"
+
+
+def test_map_read_as_parallel_lists(tmp_path):
+    """MAP columns surface as two parallel list columns (m.key/m.value)
+    with shared row ranges."""
+    data = [{"a": 1, "b": 2}, {}, None, {"c": None, "d": 4}]
+    t = pa.table({"m": pa.array(data, pa.map_(pa.string(), pa.int64()))})
+    path = str(tmp_path / "map.parquet")
+    pq.write_table(t, path, use_dictionary=False, compression="zstd")
+    h = cpp().open_parquet(path)
+    try:
+        meta = cpp().parquet_meta(h)
+        names = [c["name"] for c in meta["columns"]]
+        assert names == ["m.key", "m.value"]
+        assert meta["columns"][0]["dtype"] == "list<string>"
+        assert meta["columns"][1]["dtype"] == "list<int64>"
+        dk = cpp().read_chunk_cpu(h, 0, 0)
+        dv = cpp().read_chunk_cpu(h, 0, 1)
+    finally:
+        cpp().close_parquet(h)
+    ko, vo = dk["list_offsets"].numpy(), dv["list_offsets"].numpy()
+    np.testing.assert_array_equal(ko, vo)
+    kv = dk["list_validity"].numpy()
+    assert kv.tolist() == [1, 1, 0, 1]
+    soffs = dk["offsets"].numpy()
+    bys = dk["bytes"].numpy().tobytes()
+    keys = [bys[soffs[i]:soffs[i + 1]].decode() for i in range(4)]
+    assert keys == ["a", "b", "c", "d"]
+    vals = dv["data"].numpy().view(np.int64)
+    vvm = dv["validity"].numpy()
+    assert vvm.tolist() == [1, 1, 0, 1]
+    assert vals[0] == 1 and vals[3] == 4
