@@ -1,0 +1,342 @@
+// Metadata C ABI — analog of the reference's rust/lakesoul-metadata-c
+// (lib.rs:116-560): open the catalog store, look up table info, resolve
+// the latest snapshot to a file list, and commit new data with the MVCC
+// version-CAS retry loop — everything a JVM/foreign-engine connector
+// needs to participate in the commit protocol (DBManager.java routes
+// the same operations through metadata-c).
+//
+// Backed by the sqlite catalog (same schema as meta/store.py, which
+// mirrors script/meta_init.sql). sqlite3 is declared extern and linked
+// against the system libsqlite3.so.0 (no dev headers in the image,
+// same technique as compress.h for libzstd). Results cross the ABI as
+// JSON strings (callers free them with lakesoul_meta_free_string).
+
+#include <cstdint>
+#include <cstdio>
+#include <cstring>
+#include <cstdlib>
+#include <stdexcept>
+#include <random>
+#include <string>
+#include <vector>
+
+// ---- minimal sqlite3 API (stable C ABI of libsqlite3.so.0) ----
+extern "C" {
+typedef struct sqlite3 sqlite3;
+typedef struct sqlite3_stmt sqlite3_stmt;
+int sqlite3_open(const char*, sqlite3**);
+int sqlite3_close(sqlite3*);
+int sqlite3_prepare_v2(sqlite3*, const char*, int, sqlite3_stmt**,
+                       const char**);
+int sqlite3_bind_text(sqlite3_stmt*, int, const char*, int, void (*)(void*));
+int sqlite3_bind_int64(sqlite3_stmt*, int, int64_t);
+int sqlite3_step(sqlite3_stmt*);
+int sqlite3_finalize(sqlite3_stmt*);
+const unsigned char* sqlite3_column_text(sqlite3_stmt*, int);
+int64_t sqlite3_column_int64(sqlite3_stmt*, int);
+int sqlite3_exec(sqlite3*, const char*, int (*)(void*, int, char**, char**),
+                 void*, char**);
+int sqlite3_busy_timeout(sqlite3*, int);
+#define SQLITE_ROW 100
+#define SQLITE_DONE 101
+#define SQLITE_OK 0
+#define SQLITE_CONSTRAINT 19
+}
+
+static thread_local std::string g_meta_err;
+
+extern "C" const char* lakesoul_meta_last_error(void) {
+  return g_meta_err.c_str();
+}
+
+namespace {
+
+struct MetaHandle {
+  sqlite3* db = nullptr;
+};
+
+std::string jesc(const std::string& s) {
+  std::string o;
+  for (char c : s) {
+    if (c == '"' || c == '\\') {
+      o += '\\';
+      o += c;
+    } else if ((unsigned char)c < 0x20) {
+      char buf[8];
+      snprintf(buf, sizeof buf, "\\u%04x", c);
+      o += buf;
+    } else {
+      o += c;
+    }
+  }
+  return o;
+}
+
+// tiny JSON scanners for the store's own serializations (lists of
+// strings / file-op dicts written by meta/store.py via json.dumps)
+std::vector<std::string> parse_str_list(const std::string& j) {
+  std::vector<std::string> out;
+  size_t i = 0;
+  while ((i = j.find('"', i)) != std::string::npos) {
+    size_t e = i + 1;
+    std::string cur;
+    while (e < j.size() && j[e] != '"') {
+      if (j[e] == '\\' && e + 1 < j.size()) e++;
+      cur += j[e++];
+    }
+    out.push_back(cur);
+    i = e + 1;
+  }
+  return out;
+}
+
+std::string q1(sqlite3* db, const std::string& sql,
+               const std::vector<std::string>& binds, bool* found) {
+  sqlite3_stmt* st = nullptr;
+  if (sqlite3_prepare_v2(db, sql.c_str(), -1, &st, nullptr) != SQLITE_OK)
+    throw std::runtime_error("sqlite prepare failed: " + sql);
+  for (size_t i = 0; i < binds.size(); i++)
+    sqlite3_bind_text(st, (int)i + 1, binds[i].c_str(), -1, nullptr);
+  std::string out;
+  *found = false;
+  if (sqlite3_step(st) == SQLITE_ROW) {
+    const unsigned char* t = sqlite3_column_text(st, 0);
+    out = t ? (const char*)t : "";
+    *found = true;
+  }
+  sqlite3_finalize(st);
+  return out;
+}
+
+std::string rand_hex(int n) {
+  static thread_local std::mt19937_64 rng{std::random_device{}()};
+  static const char* hexd = "0123456789abcdef";
+  std::string s;
+  for (int i = 0; i < n; i++) s += hexd[rng() & 15];
+  return s;
+}
+
+}  // namespace
+
+#define META_TRY(body)             \
+  try {                            \
+    body;                          \
+    return 0;                      \
+  } catch (std::exception & e) {   \
+    g_meta_err = e.what();         \
+    return -1;                     \
+  }
+
+extern "C" void* lakesoul_meta_open(const char* db_path) {
+  auto* h = new MetaHandle();
+  if (sqlite3_open(db_path, &h->db) != SQLITE_OK) {
+    g_meta_err = "cannot open metadata db";
+    delete h;
+    return nullptr;
+  }
+  sqlite3_busy_timeout(h->db, 60000);
+  return h;
+}
+
+extern "C" void lakesoul_meta_close(void* hp) {
+  auto* h = (MetaHandle*)hp;
+  if (h) {
+    sqlite3_close(h->db);
+    delete h;
+  }
+}
+
+extern "C" void lakesoul_meta_free_string(char* s) { free(s); }
+
+// table info as JSON {"table_id","table_path","table_schema",...} or NULL
+extern "C" char* lakesoul_meta_table_info(void* hp, const char* name,
+                                          const char* ns) {
+  auto* h = (MetaHandle*)hp;
+  try {
+    sqlite3_stmt* st = nullptr;
+    const char* sql =
+        "SELECT table_id, table_path, table_schema, properties, partitions"
+        " FROM table_info WHERE table_name=? AND table_namespace=?";
+    if (sqlite3_prepare_v2(h->db, sql, -1, &st, nullptr) != SQLITE_OK)
+      throw std::runtime_error("prepare failed");
+    sqlite3_bind_text(st, 1, name, -1, nullptr);
+    sqlite3_bind_text(st, 2, ns && *ns ? ns : "default", -1, nullptr);
+    char* out = nullptr;
+    if (sqlite3_step(st) == SQLITE_ROW) {
+      auto col = [&](int i) {
+        const unsigned char* t = sqlite3_column_text(st, i);
+        return std::string(t ? (const char*)t : "");
+      };
+      std::string j = "{\"table_id\":\"" + jesc(col(0)) + "\",\"table_path\":\"" +
+                      jesc(col(1)) + "\",\"table_schema\":" +
+                      (col(2).empty() ? "null" : col(2)) +
+                      ",\"properties\":" + (col(3).empty() ? "{}" : col(3)) +
+                      ",\"partitions\":\"" + jesc(col(4)) + "\"}";
+      out = strdup(j.c_str());
+    }
+    sqlite3_finalize(st);
+    if (!out) g_meta_err = "table not found";
+    return out;
+  } catch (std::exception& e) {
+    g_meta_err = e.what();
+    return nullptr;
+  }
+}
+
+// latest snapshot resolved to files: JSON [{"path":...,"size":N},...]
+extern "C" char* lakesoul_meta_files_for_latest(void* hp, const char* table_id,
+                                                const char* partition_desc) {
+  auto* h = (MetaHandle*)hp;
+  try {
+    bool found = false;
+    std::string snap = q1(
+        h->db,
+        "SELECT snapshot FROM partition_info WHERE table_id=? AND"
+        " partition_desc=? ORDER BY version DESC LIMIT 1",
+        {table_id, partition_desc}, &found);
+    std::string out = "[";
+    bool first = true;
+    if (found) {
+      // resolve add/del ops across the snapshot's commits in order
+      std::vector<std::pair<std::string, int64_t>> files;
+      for (auto& cid : parse_str_list(snap)) {
+        bool f2 = false;
+        std::string ops = q1(
+            h->db,
+            "SELECT file_ops FROM data_commit_info WHERE table_id=? AND"
+            " partition_desc=? AND commit_id=?",
+            {table_id, partition_desc, cid}, &f2);
+        if (!f2) continue;
+        // ops JSON: [{"path": "...", "file_op": "add|del", "size": N}, ...]
+        size_t i = 0;
+        while ((i = ops.find("{", i)) != std::string::npos) {
+          size_t e = ops.find("}", i);
+          std::string obj = ops.substr(i, e - i);
+          auto grab = [&](const char* key) {
+            size_t k = obj.find(std::string("\"") + key + "\"");
+            if (k == std::string::npos) return std::string();
+            k = obj.find(':', k) + 1;
+            while (k < obj.size() && (obj[k] == ' ')) k++;
+            if (obj[k] == '"') {
+              size_t q = obj.find('"', k + 1);
+              return obj.substr(k + 1, q - k - 1);
+            }
+            size_t q = obj.find_first_of(",}", k);
+            return obj.substr(k, q - k);
+          };
+          std::string path = grab("path");
+          std::string op = grab("file_op");
+          std::string sz = grab("size");
+          if (op == "del") {
+            for (auto it = files.begin(); it != files.end(); ++it)
+              if (it->first == path) {
+                files.erase(it);
+                break;
+              }
+          } else {
+            files.push_back({path, sz.empty() ? 0 : atoll(sz.c_str())});
+          }
+          i = e + 1;
+        }
+      }
+      for (auto& f : files) {
+        if (!first) out += ",";
+        first = false;
+        out += "{\"path\":\"" + jesc(f.first) +
+               "\",\"size\":" + std::to_string(f.second) + "}";
+      }
+    }
+    out += "]";
+    return strdup(out.c_str());
+  } catch (std::exception& e) {
+    g_meta_err = e.what();
+    return nullptr;
+  }
+}
+
+// commit new files (Append/Merge semantics: snapshot extend) with the
+// MVCC version-CAS retry loop. paths/sizes: parallel arrays.
+extern "C" int lakesoul_meta_commit_add_files(void* hp, const char* table_id,
+                                              const char* partition_desc,
+                                              const char** paths,
+                                              const int64_t* sizes, int nfiles,
+                                              const char* commit_op) {
+  auto* h = (MetaHandle*)hp;
+  META_TRY({
+    std::string op = commit_op && *commit_op ? commit_op : "MergeCommit";
+    std::string cid = rand_hex(32);
+    std::string ops = "[";
+    for (int i = 0; i < nfiles; i++) {
+      if (i) ops += ", ";
+      ops += "{\"path\": \"" + jesc(paths[i]) + "\", \"file_op\": \"add\", "
+             "\"size\": " + std::to_string(sizes ? sizes[i] : 0) + "}";
+    }
+    ops += "]";
+    {
+      sqlite3_stmt* st = nullptr;
+      const char* sql =
+          "INSERT INTO data_commit_info VALUES (?,?,?,?,?,0,"
+          "CAST(strftime('%s','now') AS INTEGER)*1000,'public')";
+      if (sqlite3_prepare_v2(h->db, sql, -1, &st, nullptr) != SQLITE_OK)
+        throw std::runtime_error("prepare dci failed");
+      sqlite3_bind_text(st, 1, table_id, -1, nullptr);
+      sqlite3_bind_text(st, 2, partition_desc, -1, nullptr);
+      sqlite3_bind_text(st, 3, cid.c_str(), -1, nullptr);
+      sqlite3_bind_text(st, 4, ops.c_str(), -1, nullptr);
+      sqlite3_bind_text(st, 5, op.c_str(), -1, nullptr);
+      if (sqlite3_step(st) != SQLITE_DONE) {
+        sqlite3_finalize(st);
+        throw std::runtime_error("insert data_commit_info failed");
+      }
+      sqlite3_finalize(st);
+    }
+    // CAS loop: read latest version+snapshot, insert version+1
+    for (int attempt = 0; attempt < 25; attempt++) {
+      bool found = false;
+      std::string snap = q1(
+          h->db,
+          "SELECT snapshot FROM partition_info WHERE table_id=? AND"
+          " partition_desc=? ORDER BY version DESC LIMIT 1",
+          {table_id, partition_desc}, &found);
+      int64_t ver = -1;
+      if (found) {
+        bool f2 = false;
+        std::string v = q1(
+            h->db,
+            "SELECT version FROM partition_info WHERE table_id=? AND"
+            " partition_desc=? ORDER BY version DESC LIMIT 1",
+            {table_id, partition_desc}, &f2);
+        ver = atoll(v.c_str());
+      }
+      std::string new_snap;
+      if (found && !snap.empty() && snap != "[]") {
+        new_snap = snap.substr(0, snap.size() - 1) + ", \"" + cid + "\"]";
+      } else {
+        new_snap = "[\"" + cid + "\"]";
+      }
+      sqlite3_stmt* st = nullptr;
+      const char* sql =
+          "INSERT INTO partition_info VALUES (?,?,?,?,"
+          "CAST(strftime('%s','now') AS INTEGER)*1000,?,'','public')";
+      if (sqlite3_prepare_v2(h->db, sql, -1, &st, nullptr) != SQLITE_OK)
+        throw std::runtime_error("prepare pi failed");
+      sqlite3_bind_text(st, 1, table_id, -1, nullptr);
+      sqlite3_bind_text(st, 2, partition_desc, -1, nullptr);
+      sqlite3_bind_int64(st, 3, ver + 1);
+      sqlite3_bind_text(st, 4, op.c_str(), -1, nullptr);
+      sqlite3_bind_text(st, 5, new_snap.c_str(), -1, nullptr);
+      int rc = sqlite3_step(st);
+      sqlite3_finalize(st);
+      if (rc == SQLITE_DONE) {
+        char* err = nullptr;
+        std::string upd =
+            "UPDATE data_commit_info SET committed=1 WHERE commit_id='" + cid +
+            "'";
+        sqlite3_exec(h->db, upd.c_str(), nullptr, nullptr, &err);
+        return 0;  // committed
+      }
+      // CAS conflict (PK violation): retry with fresh read
+    }
+    throw std::runtime_error("commit_add_files: CAS retries exhausted");
+  });
+}

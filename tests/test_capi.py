@@ -173,3 +173,73 @@ def test_capi_reader_string_pk(tmp_path, lib):
     got = pa.Table.from_batches(rows).to_pandas()
     assert got["k"].tolist() == ["apple", "banana", "kiwi", "pear"]
     assert got["v"].tolist() == [1.0, 10.0, 20.0, 3.0]
+
+
+def test_meta_c_abi_roundtrip(lib, catalog, tmp_path):
+    """Metadata C ABI (lakesoul-metadata-c analog): table lookup, snapshot
+    file resolution, and an MVCC commit visible to the python client."""
+    import json
+
+    from lakesoul_amd.io.schema import Field, Schema
+
+    t = catalog.create_table(
+        "cmeta",
+        Schema([Field("id", "int64", False), Field("v", "float64", False)]),
+        primary_keys=["id"], hash_bucket_num=1,
+    )
+    n = 100
+    t.upsert({"id": np.arange(n, dtype=np.int64), "v": np.zeros(n)})
+
+    L = ctypes.CDLL(LIB)
+    L.lakesoul_meta_open.restype = ctypes.c_void_p
+    L.lakesoul_meta_open.argtypes = [ctypes.c_char_p]
+    L.lakesoul_meta_table_info.restype = ctypes.c_void_p
+    L.lakesoul_meta_table_info.argtypes = [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p]
+    L.lakesoul_meta_files_for_latest.restype = ctypes.c_void_p
+    L.lakesoul_meta_files_for_latest.argtypes = [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p]
+    L.lakesoul_meta_commit_add_files.restype = ctypes.c_int
+    L.lakesoul_meta_commit_add_files.argtypes = [
+        ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p,
+        ctypes.POINTER(ctypes.c_char_p), ctypes.POINTER(ctypes.c_int64),
+        ctypes.c_int, ctypes.c_char_p]
+    L.lakesoul_meta_free_string.argtypes = [ctypes.c_void_p]
+    L.lakesoul_meta_close.argtypes = [ctypes.c_void_p]
+    L.lakesoul_meta_last_error.restype = ctypes.c_char_p
+
+    h = L.lakesoul_meta_open(t.client.store.path.encode())
+    assert h, L.lakesoul_meta_last_error()
+    try:
+        p = L.lakesoul_meta_table_info(h, b"cmeta", b"default")
+        assert p, L.lakesoul_meta_last_error()
+        info = json.loads(ctypes.string_at(p).decode())
+        L.lakesoul_meta_free_string(p)
+        assert info["table_id"] == t.table_id
+        assert info["table_schema"]["fields"][0]["name"] == "id"
+
+        p = L.lakesoul_meta_files_for_latest(h, t.table_id.encode(), b"-5")
+        files = json.loads(ctypes.string_at(p).decode())
+        L.lakesoul_meta_free_string(p)
+        py_files = {f.path for f in t.files()}
+        assert {f["path"] for f in files} == py_files
+        assert all(f["size"] > 0 for f in files)
+
+        # commit an extra file through the C ABI; python must see it
+        import pyarrow as pa
+        import pyarrow.parquet as pq
+
+        extra = str(tmp_path / "part-cabi000000000_0000.parquet")
+        pq.write_table(pa.table({
+            "id": pa.array(np.arange(100, 110, dtype=np.int64)),
+            "v": pa.array(np.full(10, 5.5)),
+        }), extra, use_dictionary=False, compression="zstd")
+        paths = (ctypes.c_char_p * 1)(extra.encode())
+        sizes = (ctypes.c_int64 * 1)(1234)
+        rc = L.lakesoul_meta_commit_add_files(
+            h, t.table_id.encode(), b"-5", paths, sizes, 1, b"MergeCommit")
+        assert rc == 0, L.lakesoul_meta_last_error()
+    finally:
+        L.lakesoul_meta_close(h)
+
+    df = t.to_pandas().sort_values("id").reset_index(drop=True)
+    assert len(df) == n + 10
+    assert (df[df.id >= 100]["v"] == 5.5).all()
