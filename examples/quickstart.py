@@ -76,6 +76,7 @@ print("compacted; vacuumed", removed, "files")
 from lakesoul_amd.tables.stream import TableStream  # noqa: E402
 
 stream = TableStream(orders, device="cpu")
+stream.poll()  # drain history (a fresh stream starts from the beginning)
 orders.upsert({"order_id": np.array([n + 1], dtype=np.int64),
                "customer": ["new"], "amount": np.array([100], dtype=np.int64),
                "qty": np.array([1], dtype=np.int64)})
