@@ -195,13 +195,16 @@ class ParquetWriter {
       const ColumnData& col = data[ci];
 
       // page split: target ~LAKESOUL_PAGE_BYTES decompressed bytes per
-      // page (default 32 KB) so the GPU zstd kernel gets thousands of
-      // pages in flight per scan unit instead of a few multi-MB frames.
+      // page. Default 128 KB: measured best for the host zstd decode
+      // path (32 KB costs ~15% on the headline scan from extra headers
+      // and worse zstd ratio). Set 32768 when running the experimental
+      // GPU zstd kernel (profiles/r01_gpu_zstd.md) — it wants thousands
+      // of pages in flight per scan unit.
       // Strings keep one page per chunk (host-assembled path).
       static const int64_t kPageBytes = []() {
         const char* e = std::getenv("LAKESOUL_PAGE_BYTES");
         int64_t v = e ? atoll(e) : 0;
-        return v > 0 ? v : (int64_t)(32 << 10);
+        return v > 0 ? v : (int64_t)(128 << 10);
       }();
       int64_t page_rows = n;
       if (cd.physical != PT_BYTE_ARRAY) {
