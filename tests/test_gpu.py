@@ -567,3 +567,38 @@ def test_gpu_joined_merge_operators(dev, tmp_path):
     assert cpu["tags"].iloc[0] is None or pd.isna(cpu["tags"].iloc[0])
     assert cpu["tags"].str.contains(",").fillna(False).any()
     assert cpu["last"].str.contains(";").fillna(False).any()
+
+
+@pytest.mark.gpu
+def test_gpu_range_partitioned_mor(dev, tmp_path):
+    """Range partitions (col=val dirs) + hash buckets + GPU MOR merge +
+    partition pruning, results equal CPU."""
+    catalog = _mk_catalog(tmp_path)
+    from lakesoul_amd.io.schema import Field, Schema
+
+    t = catalog.create_table(
+        "grange",
+        Schema([Field("dt", "string", False), Field("id", "int64", False),
+                Field("v", "float64", False)]),
+        primary_keys=["id"],
+        range_partitions=["dt"],
+        hash_bucket_num=2,
+    )
+    n = 40000
+    for day in ("2026-01-01", "2026-01-02"):
+        t.upsert({"dt": [day] * n, "id": np.arange(n, dtype=np.int64),
+                  "v": np.zeros(n)})
+        t.upsert({"dt": [day] * (n // 4),
+                  "id": np.arange(0, n, 4, dtype=np.int64),
+                  "v": np.ones(n // 4)})
+    cpu = t.scan(device="cpu", partitions=["dt=2026-01-02"]).to_arrow().to_pandas()
+    gpu = t.scan(device="cuda", partitions=["dt=2026-01-02"]).to_arrow().to_pandas()
+    cpu = cpu.sort_values("id").reset_index(drop=True)
+    gpu = gpu.sort_values("id").reset_index(drop=True)
+    import pandas as pd
+
+    pd.testing.assert_frame_equal(cpu, gpu)
+    assert len(gpu) == n and (gpu["dt"] == "2026-01-02").all()
+    expect = np.zeros(n)
+    expect[::4] = 1.0
+    np.testing.assert_allclose(gpu["v"].to_numpy(), expect)
