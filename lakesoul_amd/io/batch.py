@@ -88,7 +88,12 @@ class Column:
         return Column(self.dtype, mv(self.data), mv(self.offsets), mv(self.bytes_), mv(self.validity))
 
     def take(self, idx: torch.Tensor) -> "Column":
-        """Gather rows by index tensor (same device)."""
+        """Gather rows by index tensor (moved to this column's device —
+        a CPU index against a CUDA column would reach the string gather
+        kernel as a host pointer)."""
+        dev = self.offsets.device if self.is_string else self.data.device
+        if idx.device != dev:
+            idx = idx.to(dev)
         if self.is_string:
             offs = self.offsets
             lens = offs[1:] - offs[:-1]
