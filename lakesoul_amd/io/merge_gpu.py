@@ -137,18 +137,6 @@ def merge_key_order(
     return perm, None, sort_tensors
 
 
-def sorted_keys_for_order(
-    file_pk_cols: List[List[Column]], order: torch.Tensor
-) -> List[torch.Tensor]:
-    """PK columns in merge order (for boundary detection)."""
-    npk = len(file_pk_cols[0])
-    out = []
-    for ci in range(npk):
-        cat = torch.cat([fc[ci].data for fc in file_pk_cols])
-        out.append(cat[order])
-    return out
-
-
 def merge_sorted_files_gpu(
     file_batches: List[Batch],
     pk: List[str],
