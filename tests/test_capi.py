@@ -243,3 +243,74 @@ def test_meta_c_abi_roundtrip(lib, catalog, tmp_path):
     df = t.to_pandas().sort_values("id").reset_index(drop=True)
     assert len(df) == n + 10
     assert (df[df.id >= 100]["v"] == 5.5).all()
+
+
+def test_meta_c_abi_commit_races_python(lib, catalog, tmp_path):
+    """C-ABI commits racing python commits: both sides' CAS loops land
+    every commit (cross-language MVCC interop)."""
+    import threading
+
+    from lakesoul_amd.io.schema import Field, Schema
+
+    t = catalog.create_table(
+        "cmrace",
+        Schema([Field("id", "int64", False), Field("v", "float64", False)]),
+        primary_keys=["id"], hash_bucket_num=1,
+    )
+    t.upsert({"id": np.arange(10, dtype=np.int64), "v": np.zeros(10)})
+
+    L = ctypes.CDLL(LIB)
+    L.lakesoul_meta_open.restype = ctypes.c_void_p
+    L.lakesoul_meta_open.argtypes = [ctypes.c_char_p]
+    L.lakesoul_meta_commit_add_files.restype = ctypes.c_int
+    L.lakesoul_meta_commit_add_files.argtypes = [
+        ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p,
+        ctypes.POINTER(ctypes.c_char_p), ctypes.POINTER(ctypes.c_int64),
+        ctypes.c_int, ctypes.c_char_p]
+    L.lakesoul_meta_close.argtypes = [ctypes.c_void_p]
+
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    cfiles = []
+    for i in range(4):
+        p = str(tmp_path / f"part-crace{i:010d}_0000.parquet")
+        pq.write_table(pa.table({
+            "id": pa.array(np.arange(100 + i * 10, 110 + i * 10, dtype=np.int64)),
+            "v": pa.array(np.full(10, float(i))),
+        }), p, use_dictionary=False, compression="zstd")
+        cfiles.append(p)
+
+    errs = []
+
+    def c_committer():
+        h = L.lakesoul_meta_open(t.client.store.path.encode())
+        try:
+            for p in cfiles:
+                paths = (ctypes.c_char_p * 1)(p.encode())
+                sizes = (ctypes.c_int64 * 1)(1)
+                if L.lakesoul_meta_commit_add_files(
+                        h, t.table_id.encode(), b"-5", paths, sizes, 1,
+                        b"MergeCommit") != 0:
+                    errs.append("c commit failed")
+        finally:
+            L.lakesoul_meta_close(h)
+
+    def py_committer():
+        try:
+            for i in range(4):
+                t.upsert({"id": np.array([i], dtype=np.int64),
+                          "v": np.array([50.0 + i])})
+        except Exception as e:  # pragma: no cover
+            errs.append(repr(e))
+
+    th1 = threading.Thread(target=c_committer)
+    th2 = threading.Thread(target=py_committer)
+    th1.start(); th2.start(); th1.join(); th2.join()
+    assert not errs, errs
+    df = t.to_pandas().sort_values("id").reset_index(drop=True)
+    assert len(df) == 10 + 40       # 4 C-committed files x 10 rows
+    assert (df[df.id == 0]["v"] == 50.0).all()
+    assert (df[df.id >= 130]["v"] == 3.0).all()
+    # version advanced once per commit: 1 base + 4 py + 4 C
+    assert t.latest_version("-5") == 8
