@@ -30,12 +30,21 @@ def cli():
 @cli.command("sql")
 @click.argument("query", required=False)
 @click.option("--device", default=None, help="scan device (e.g. cuda:0)")
-def sql(query, device):
-    """Run one SQL statement, or start the interactive console
-    (reference: rust/lakesoul-console)."""
+@click.option("-f", "--file", "script", type=click.Path(exists=True),
+              help="run statements from a .sql file (';'-separated)")
+def sql(query, device, script):
+    """Run one SQL statement, a .sql script, or start the interactive
+    console (reference: rust/lakesoul-console)."""
     from .sql import execute_sql, repl
 
     cat = _catalog()
+    if script:
+        with open(script) as fh:
+            text = fh.read()
+        for stmt in [x.strip() for x in text.split(";") if x.strip()]:
+            df = execute_sql(cat, stmt, device=device)
+            click.echo(df.to_string(index=False))
+        return
     if query:
         df = execute_sql(cat, query, device=device)
         try:
