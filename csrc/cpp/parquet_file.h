@@ -224,19 +224,31 @@ class ParquetWriter {
       cm.num_values = n;
       cm.data_page_offset = pos_;
 
+      // LAKESOUL_PAGE_V2=1 emits DataPageV2 (levels outside the
+      // compressed region; reader + pyarrow both handle it)
+      static const bool kPageV2 = []() {
+        const char* e = std::getenv("LAKESOUL_PAGE_V2");
+        return e && *e == '1';
+      }();
       for (int64_t poff = 0; poff < n || n == 0; poff += page_rows) {
         int64_t pn = n - poff < page_rows ? n - poff : page_rows;
         // ---- assemble page payload (def levels + PLAIN values) ----
         std::vector<uint8_t> payload;
+        std::vector<uint8_t> levels;
+        int64_t page_nulls = 0;
         const uint8_t* validity =
             cd.nullable && col.validity ? col.validity + row_off + poff : nullptr;
         if (cd.nullable) {
-          std::vector<uint8_t> levels = encode_def_levels(validity, pn);
-          uint32_t lv_len = (uint32_t)levels.size();
-          payload.insert(payload.end(), (uint8_t*)&lv_len, (uint8_t*)&lv_len + 4);
-          payload.insert(payload.end(), levels.begin(), levels.end());
+          levels = encode_def_levels(validity, pn);
+          if (!kPageV2) {
+            uint32_t lv_len = (uint32_t)levels.size();
+            payload.insert(payload.end(), (uint8_t*)&lv_len,
+                           (uint8_t*)&lv_len + 4);
+            payload.insert(payload.end(), levels.begin(), levels.end());
+          }
           if (validity)
-            for (int64_t i = 0; i < pn; i++) stats.null_count += validity[i] ? 0 : 1;
+            for (int64_t i = 0; i < pn; i++) page_nulls += validity[i] ? 0 : 1;
+          stats.null_count += page_nulls;
         }
         append_plain_values(cd, col, row_off + poff, pn, validity, payload, stats);
 
@@ -254,17 +266,32 @@ class ParquetWriter {
 
         // ---- page header + emit ----
         PageHeader ph;
-        ph.type = PAGE_DATA;
-        ph.uncompressed_size = (int32_t)payload.size();
-        ph.compressed_size = (int32_t)body_n;
+        ph.type = kPageV2 ? PAGE_DATA_V2 : PAGE_DATA;
         ph.num_values = (int32_t)pn;
         ph.encoding = ENC_PLAIN;
         ph.def_encoding = ENC_RLE;
         ph.rep_encoding = ENC_RLE;
+        if (kPageV2) {
+          // v2: levels sit uncompressed in front of the compressed values
+          ph.num_nulls = (int32_t)page_nulls;
+          ph.num_rows = (int32_t)pn;
+          ph.def_levels_byte_length = (int32_t)levels.size();
+          ph.rep_levels_byte_length = 0;
+          ph.v2_is_compressed = codec_ != CODEC_UNCOMPRESSED;
+          ph.uncompressed_size = (int32_t)(levels.size() + payload.size());
+          ph.compressed_size = (int32_t)(levels.size() + body_n);
+        } else {
+          ph.uncompressed_size = (int32_t)payload.size();
+          ph.compressed_size = (int32_t)body_n;
+        }
         auto ph_bytes = serialize_page_header(ph);
-        cm.total_uncompressed_size += (int64_t)(ph_bytes.size() + payload.size());
-        cm.total_compressed_size += (int64_t)(ph_bytes.size() + body_n);
+        cm.total_uncompressed_size +=
+            (int64_t)(ph_bytes.size() + ph.uncompressed_size);
+        cm.total_compressed_size +=
+            (int64_t)(ph_bytes.size() + ph.compressed_size);
         fwrite_all(ph_bytes.data(), ph_bytes.size());
+        if (kPageV2 && !levels.empty())
+          fwrite_all(levels.data(), levels.size());
         fwrite_all(body, body_n);
         if (n == 0) break;
       }

@@ -595,6 +595,19 @@ inline std::vector<uint8_t> serialize_page_header(const PageHeader& h) {
     w.field_i32(l2, 1, h.dict_num_values);
     w.field_i32(l2, 2, h.encoding);
     w.stop();
+  } else if (h.type == PAGE_DATA_V2) {
+    // DataPageHeaderV2 { 1:num_values 2:num_nulls 3:num_rows 4:encoding
+    //   5:def_levels_byte_length 6:rep_levels_byte_length 7:is_compressed }
+    w.field(last, 8, CType::STRUCT);
+    int16_t l2 = 0;
+    w.field_i32(l2, 1, h.num_values);
+    w.field_i32(l2, 2, h.num_nulls);
+    w.field_i32(l2, 3, h.num_rows);
+    w.field_i32(l2, 4, h.encoding);
+    w.field_i32(l2, 5, h.def_levels_byte_length);
+    w.field_i32(l2, 6, h.rep_levels_byte_length);
+    w.field_bool(l2, 7, h.v2_is_compressed);
+    w.stop();
   }
   w.stop();
   return std::move(w.buf);

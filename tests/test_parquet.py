@@ -5,6 +5,8 @@ This is the writer x reader consistency matrix idea from the reference's
 foreign engine.
 """
 
+import os
+
 import numpy as np
 import pyarrow as pa
 import pyarrow.parquet as pq
@@ -475,3 +477,53 @@ def test_map_read_as_parallel_lists(tmp_path):
     vvm = dv["validity"].numpy()
     assert vvm.tolist() == [1, 1, 0, 1]
     assert vals[0] == 1 and vals[3] == 4
+
+
+def test_datapage_v2_write_roundtrip(tmp_path, monkeypatch):
+    """LAKESOUL_PAGE_V2=1 emits DataPageV2; pyarrow and our reader agree.
+    NOTE: the v2 flag is read once per process (static init) — this test
+    spawns a subprocess with the env set."""
+    import subprocess
+    import sys
+    import textwrap
+
+    script = textwrap.dedent(f"""
+        import sys, numpy as np, torch
+        sys.path.insert(0, {str(os.path.dirname(os.path.dirname(os.path.abspath(__file__))))!r})
+        from lakesoul_amd.ops import cpp
+        n = 10000
+        rng = np.random.default_rng(0)
+        v = rng.normal(size=n)
+        validity = (rng.random(n) > 0.1).astype(np.uint8)
+        cpp().write_parquet(
+            {str(tmp_path / 'v2.parquet')!r}, ["a", "b"], ["int64", "float64"],
+            [torch.arange(n, dtype=torch.int64), torch.from_numpy(v)],
+            [None, None], [None, torch.from_numpy(validity)], [False, True],
+            4000, 6, 1)
+        print("WROTE")
+    """)
+    env = dict(os.environ, LAKESOUL_PAGE_V2="1")
+    r = subprocess.run([sys.executable, "-c", script], capture_output=True,
+                       text=True, env=env, timeout=300)
+    assert "WROTE" in r.stdout, r.stderr[-800:]
+
+    t = pq.read_table(str(tmp_path / "v2.parquet"))
+    assert t.num_rows == 10000
+    np.testing.assert_array_equal(t.column("a").to_numpy(), np.arange(10000))
+    # our reader handles the v2 pages too
+    h = cpp().open_parquet(str(tmp_path / "v2.parquet"))
+    try:
+        got_parts = []
+        meta = cpp().parquet_meta(h)
+        for rg in range(meta["num_row_groups"]):
+            d = cpp().read_chunk_cpu(h, rg, 0)
+            got_parts.append(d["data"].numpy().view(np.int64))
+        np.testing.assert_array_equal(np.concatenate(got_parts), np.arange(10000))
+        nulls = 0
+        for rg in range(meta["num_row_groups"]):
+            d = cpp().read_chunk_cpu(h, rg, 1)
+            vmask = d["validity"].numpy()
+            nulls += int((vmask == 0).sum()) if len(vmask) else 0
+        assert nulls == int((t.column("b").null_count))
+    finally:
+        cpp().close_parquet(h)
