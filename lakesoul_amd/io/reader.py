@@ -392,7 +392,7 @@ class LakeSoulScan:
         if not unit.files:
             return None
         oversized = not self._unit_fits(unit)
-        if oversized and self.device != "cuda":
+        if oversized and self.device != "cuda" and not self._chunkable():
             raise MemoryError(
                 f"scan unit bucket={unit.bucket_id} estimated decoded size "
                 f"exceeds LAKESOUL_MAX_UNIT_BYTES. Recreate the table with "
@@ -411,6 +411,8 @@ class LakeSoulScan:
         needs_merge = bool(self.pk) and (
             len(unit.files) > 1 or self.cdc_column is not None or bool(self.merge_ops)
         )
+        if oversized and self._chunkable():
+            return self._read_unit_chunked(unit)
         if self.device == "cuda" and (
             oversized or (needs_merge and not self._gpu_merge_supported())
         ):
@@ -440,6 +442,97 @@ class LakeSoulScan:
 
             scan_cache().put(cache_key, batch)
         return batch
+
+    def _chunkable(self) -> bool:
+        """Chunked (PK-range) merge needs a single integer primary key
+        whose row-group min/max stats order the key space."""
+        return (len(self.pk) == 1 and
+                self.schema.field(self.pk[0]).dtype in
+                ("int64", "int32", "int16", "int8"))
+
+    def _read_unit_chunked(self, unit: ScanUnit) -> Optional[Batch]:
+        """Chunked spill merge for buckets larger than the memory budget
+        (ROADMAP item): split the PK space at row-group stat boundaries,
+        decode+merge only the row groups overlapping each range, filter
+        the range's rows, concatenate. Works because bucket files are
+        globally PK-sorted, so ranges partition PK groups exactly."""
+        import warnings
+
+        from .batch import concat_batches
+
+        limit = int(os.environ.get("LAKESOUL_MAX_UNIT_BYTES", str(64 * 1024**3)))
+        pk0 = self.pk[0]
+        files = self._localize(unit.files)
+        # per-(file,rg): pk min/max + estimated decoded bytes
+        spans = []   # (path, rg, lo, hi, est_bytes)
+        row_bytes = sum(
+            8 if not f.is_fixed_width else max(1, 8) for f in self.eval_schema
+        ) or 8
+        for path in files:
+            h = cpp().open_parquet(path)
+            try:
+                meta = cpp().parquet_meta(h)
+                cols = meta["columns"]
+                pk_ci = next(i for i, c in enumerate(cols) if c["name"] == pk0)
+                dtype = cols[pk_ci]["dtype"]
+                for rg, g in enumerate(meta["row_groups"]):
+                    sd = g["columns"][pk_ci]
+                    if "min" not in sd:
+                        raise ValueError("missing pk stats")
+                    lo = decode_stat(sd["min"], dtype)
+                    hi = decode_stat(sd["max"], dtype)
+                    spans.append((path, rg, lo, hi, g["num_rows"] * row_bytes))
+            finally:
+                cpp().close_parquet(h)
+        # boundaries: accumulate estimated bytes in hi order; cut at ~1/4
+        # of the budget so decode intermediates stay well inside it
+        budget = max(1, limit // 4)
+        bounds = []
+        acc = 0
+        for _, _, _, hi, est in sorted(spans, key=lambda s: (s[3], s[2])):
+            acc += est
+            if acc >= budget:
+                bounds.append(hi)
+                acc = 0
+        warnings.warn(
+            f"scan unit bucket={unit.bucket_id} exceeds LAKESOUL_MAX_UNIT_BYTES;"
+            f" merging in {len(bounds) + 1} PK ranges (chunked spill merge)")
+        parts = []
+        prev = None
+        for bound in bounds + [None]:
+            rg_sets = {}
+            for path, rg, lo, hi, _ in spans:
+                if (bound is None or lo <= bound) and (prev is None or hi > prev):
+                    rg_sets.setdefault(path, []).append(rg)
+            if rg_sets:
+                file_cols, present = [], []
+                for path in files:
+                    if path not in rg_sets:
+                        continue
+                    cols, pres = self._read_file_cpu(path, self.read_cols,
+                                                     rg_subset=sorted(rg_sets[path]))
+                    file_cols.append(cols)
+                    present.append(pres)
+                merged = merge_sorted_files(
+                    file_cols, self.pk, self.merge_ops, self.cdc_column, present
+                )
+                batch = self._np_to_batch(merged, unit)
+                pkv = batch.columns[pk0].data
+                mask = torch.ones(batch.num_rows, dtype=torch.bool)
+                if prev is not None:
+                    mask &= pkv > prev
+                if bound is not None:
+                    mask &= pkv <= bound
+                idx = torch.nonzero(mask, as_tuple=True)[0]
+                if idx.numel():
+                    part = batch.take(idx)
+                    if self.device == "cuda":
+                        part = part.to_device("cuda")
+                    parts.append(part)
+            prev = bound
+        if not parts:
+            return None
+        return concat_batches(parts)
 
     def _read_unit_cpu(self, unit: ScanUnit) -> Optional[Batch]:
         file_cols: List[Dict[str, NpColumn]] = []
@@ -506,19 +599,22 @@ class LakeSoulScan:
         bys = np.frombuffer(enc * n, dtype=np.uint8).copy() if n else np.empty(0, np.uint8)
         return Column(f.dtype, offsets=torch.from_numpy(offs), bytes_=torch.from_numpy(bys))
 
-    def _read_file_cpu(self, path: str, names: Sequence[str]) -> Dict[str, NpColumn]:
+    def _read_file_cpu(self, path: str, names: Sequence[str],
+                       rg_subset: Optional[Sequence[int]] = None) -> Dict[str, NpColumn]:
         h = cpp().open_parquet(path)
         try:
             meta = cpp().parquet_meta(h)
             file_cols = {c["name"]: i for i, c in enumerate(meta["columns"])}
             nrg = meta["num_row_groups"]
-            total = meta["num_rows"]
+            rg_iter = list(rg_subset) if rg_subset is not None else list(range(nrg))
+            total = (meta["num_rows"] if rg_subset is None else
+                     sum(meta["row_groups"][rg]["num_rows"] for rg in rg_iter))
             out: Dict[str, NpColumn] = {}
             rc = []
             req = []
             for name in names:
                 if name in file_cols:
-                    for rg in range(nrg):
+                    for rg in rg_iter:
                         rc.append((rg, file_cols[name]))
                     req.append(name)
             chunks = cpp().read_chunks_cpu_batch(h, rc, 0) if rc else []
@@ -527,7 +623,7 @@ class LakeSoulScan:
                 f = self.schema.field(name)
                 parts, offs_parts, bytes_parts, masks = [], [], [], []
                 any_null = False
-                for rg in range(nrg):
+                for rg in rg_iter:
                     d = chunks[ci]
                     ci += 1
                     nv = d["num_values"]

@@ -394,8 +394,9 @@ def test_gpu_decimal_mor(dev, tmp_path):
 
 @pytest.mark.gpu
 def test_gpu_oversized_unit_fallback(dev, tmp_path, monkeypatch):
-    """A bucket estimated over LAKESOUL_MAX_UNIT_BYTES must fall back to
-    host decode+merge and still return correct HBM-resident results."""
+    """A bucket estimated over LAKESOUL_MAX_UNIT_BYTES must degrade
+    gracefully (chunked PK-range merge for int PKs) and still return
+    correct HBM-resident results."""
     import warnings
 
     from lakesoul_amd.io.schema import Field, Schema
@@ -414,7 +415,8 @@ def test_gpu_oversized_unit_fallback(dev, tmp_path, monkeypatch):
     with warnings.catch_warnings(record=True) as w:
         warnings.simplefilter("always")
         df = t.scan(device="cuda:0").to_arrow().to_pandas()
-    assert any("falling back" in str(x.message) for x in w)
+    assert any(("falling back" in str(x.message)) or ("chunked" in str(x.message))
+               for x in w)
     df = df.sort_values("id").reset_index(drop=True)
     assert len(df) == n
     expect = np.zeros(n)

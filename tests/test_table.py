@@ -300,9 +300,37 @@ def test_invalid_file_tolerance(catalog):
     assert any("invalid" in str(x.message) or "missing" in str(x.message) for x in w)
 
 
-def test_oversized_unit_guard(catalog, monkeypatch):
+def test_oversized_unit_chunked_merge(catalog, monkeypatch):
+    """Buckets over the memory budget merge in PK ranges (chunked spill
+    merge) and return exactly the same result as an unrestricted scan."""
+    import warnings
+
     t = _mk_pk_table(catalog, "huge", buckets=1)
-    t.upsert({"id": np.arange(1000, dtype=np.int64), "v": np.zeros(1000), "s": ["a"] * 1000})
+    n = 40000
+    rng = np.random.default_rng(3)
+    t.upsert({"id": np.arange(n, dtype=np.int64), "v": rng.normal(size=n),
+              "s": ["a"] * n})
+    for it in range(3):
+        ids = rng.choice(n, 8000, replace=False).astype(np.int64)
+        t.upsert({"id": ids, "v": rng.normal(size=8000), "s": [f"u{it}"] * 8000})
+    full = _df(t)
+    monkeypatch.setenv("LAKESOUL_MAX_UNIT_BYTES", "200000")
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        chunked = _df(t)
+    assert any("chunked" in str(x.message) for x in w)
+    pd.testing.assert_frame_equal(full, chunked)
+
+
+def test_oversized_unit_guard_string_pk(catalog, monkeypatch):
+    """Non-chunkable (string PK) oversized buckets still fail with the
+    actionable error on CPU scans."""
+    t = catalog.create_table(
+        "hugestr",
+        Schema([Field("k", "string", False), Field("v", "float64")]),
+        primary_keys=["k"], hash_bucket_num=1,
+    )
+    t.upsert({"k": [f"x{i}" for i in range(1000)], "v": np.zeros(1000)})
     monkeypatch.setenv("LAKESOUL_MAX_UNIT_BYTES", "1000")
     with pytest.raises(MemoryError, match="hash\\s*buckets|buckets"):
         t.to_pandas()
