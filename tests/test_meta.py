@@ -388,3 +388,25 @@ def test_compaction_replace_aborts_on_double_replace(catalog):
                 snapshot=[dci.commit_id], commit_op=CommitOp.CompactionCommit)],
             read_partition_info=[stale_read],
         ), CommitOp.CompactionCommit)
+
+
+def test_pg_store_interface_parity():
+    """PostgresMetaStore (gated: no PG server in CI) must implement every
+    public method the engine calls on SqliteMetaStore — static drift
+    check so the PG backend can't silently fall behind."""
+    import inspect
+
+    from lakesoul_amd.meta import pg_store, store
+
+    sqlite_api = {
+        n for n, m in inspect.getmembers(store.SqliteMetaStore,
+                                         predicate=inspect.isfunction)
+        if not n.startswith("_")
+    }
+    pg_api = {
+        n for n, m in inspect.getmembers(pg_store.PostgresMetaStore,
+                                         predicate=inspect.isfunction)
+        if not n.startswith("_")
+    }
+    missing = sqlite_api - pg_api
+    assert not missing, f"PostgresMetaStore missing: {sorted(missing)}"

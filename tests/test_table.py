@@ -420,3 +420,36 @@ def test_string_pk_gpu_merge_logic_cpu(catalog):
     for i in range(len(got) - 1):
         same_eq = all(bool(t[order[i]] == t[order[i + 1]]) for t in eq)
         assert same_eq == (got[i] == got[i + 1]), (got[i], got[i + 1])
+
+
+def test_chunked_merge_with_merge_ops_and_cdc(catalog, monkeypatch):
+    """Chunked spill merge composes with merge operators and CDC."""
+    import warnings
+
+    t = catalog.create_table(
+        "hugeops",
+        Schema([Field("id", "int64", False), Field("cnt", "int64"),
+                Field("x", "float64"), Field("rowKinds", "string")]),
+        primary_keys=["id"], hash_bucket_num=1,
+        properties={"merge_op.cnt": "SumAll",
+                    "lakesoul_cdc_change_column": "rowKinds"},
+    )
+    n = 20000
+    rng = np.random.default_rng(5)
+    t.upsert({"id": np.arange(n, dtype=np.int64),
+              "cnt": np.ones(n, dtype=np.int64),
+              "x": rng.normal(size=n), "rowKinds": ["insert"] * n})
+    t.upsert({"id": np.arange(0, n, 2, dtype=np.int64),
+              "cnt": np.full(n // 2, 10, dtype=np.int64),
+              "x": rng.normal(size=n // 2), "rowKinds": ["update"] * (n // 2)})
+    t.upsert({"id": np.arange(0, n, 1000, dtype=np.int64),
+              "cnt": np.zeros(n // 1000, dtype=np.int64),
+              "x": np.zeros(n // 1000), "rowKinds": ["delete"] * (n // 1000)})
+    full = _df(t)
+    monkeypatch.setenv("LAKESOUL_MAX_UNIT_BYTES", "150000")
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        chunked = _df(t)
+    assert any("chunked" in str(x.message) for x in w), [str(x.message) for x in w]
+    pd.testing.assert_frame_equal(full, chunked)
+    assert len(full) == n - n // 1000
