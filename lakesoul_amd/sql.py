@@ -35,7 +35,7 @@ _KEYWORDS = {
     "or", "not", "in", "is", "null", "between", "as", "asc", "desc",
     "show", "tables", "namespaces", "describe", "distinct", "version",
     "join", "inner", "left", "on", "insert", "into", "values",
-    "update", "set", "delete",
+    "update", "set", "delete", "offset", "having",
 }
 
 _AGGS = {"count", "sum", "min", "max", "avg"}
@@ -106,7 +106,9 @@ class Query:
     where: Optional[Expr] = None
     group_by: List[str] = field(default_factory=list)
     order_by: List[Tuple[str, bool]] = field(default_factory=list)  # (name, desc)
+    having: Optional[Expr] = None
     limit: Optional[int] = None
+    offset: int = 0
     distinct: bool = False
     version: Optional[int] = None
 
@@ -266,6 +268,8 @@ class _Parser:
             q.group_by = [self.qualified_id()]
             while self.accept("op", ","):
                 q.group_by.append(self.qualified_id())
+        if self.accept("kw", "having"):
+            q.having = self.or_expr()
         if self.accept("kw", "order"):
             self.expect("kw", "by")
             q.order_by = [self.order_item(q)]
@@ -273,6 +277,8 @@ class _Parser:
                 q.order_by.append(self.order_item(q))
         if self.accept("kw", "limit"):
             q.limit = int(self.expect("num"))
+        if self.accept("kw", "offset"):
+            q.offset = int(self.expect("num"))
         if self.peek()[0] != "eof":
             raise SqlError(f"trailing tokens: {self.peek()[1]!r}")
         return q
@@ -636,6 +642,17 @@ def _project_and_finish(q: Query, df, col, all_cols):
         if q.distinct:
             out = out.drop_duplicates().reset_index(drop=True)
 
+    if q.having is not None:
+        # HAVING references output columns (aggregate aliases / group cols)
+        def having_col(name):
+            if name in out.columns:
+                return name
+            base = name.split(".")[-1]
+            if base in out.columns:
+                return base
+            raise SqlError(f"HAVING references unknown output column {name!r}")
+
+        out = out[_pd_eval(q.having, out, having_col)].reset_index(drop=True)
     if q.order_by:
         names = []
         for n, _ in q.order_by:
@@ -644,6 +661,8 @@ def _project_and_finish(q: Query, df, col, all_cols):
                          else (col(n) if col(n) in out.columns else n))
         asc = [not d for _, d in q.order_by]
         out = out.sort_values(names, ascending=asc).reset_index(drop=True)
+    if q.offset:
+        out = out.iloc[q.offset:].reset_index(drop=True)
     if q.limit is not None:
         out = out.head(q.limit).reset_index(drop=True)
     return out.reset_index(drop=True)

@@ -197,3 +197,50 @@ def test_parquet_string_roundtrip_random(tmp_path_factory, items):
     )
     t = pq.read_table(path)
     assert t.column("s").to_pylist() == items
+
+
+# ---------------------------------------------------------------------- #
+# SQL parser/tokenizer robustness
+
+
+@settings(max_examples=150, deadline=None)
+@given(st.text(max_size=80))
+def test_sql_tokenizer_never_hangs(text):
+    """Arbitrary input either tokenizes or raises SqlError — never hangs
+    or throws anything else."""
+    from lakesoul_amd.sql import SqlError, tokenize
+
+    try:
+        tokenize(text)
+    except SqlError:
+        pass
+
+
+_ident = st.sampled_from(["id", "price", "qty", "region"])
+_lit = st.one_of(st.integers(-100, 100),
+                 st.sampled_from(["'east'", "'west'", "3.5"]))
+_cmp = st.sampled_from(["=", "!=", "<", "<=", ">", ">="])
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    st.lists(_ident, min_size=1, max_size=3, unique=True),
+    st.lists(st.tuples(_ident, _cmp, _lit), min_size=0, max_size=3),
+    st.sampled_from(["", "ORDER BY id", "ORDER BY id DESC"]),
+    st.integers(0, 20),
+)
+def test_generated_selects_parse(cols, preds, order, limit):
+    """Structured random SELECTs always parse into a well-formed AST."""
+    from lakesoul_amd.sql import parse_sql
+
+    sql = "SELECT " + ", ".join(cols) + " FROM orders"
+    if preds:
+        sql += " WHERE " + " AND ".join(f"{c} {o} {v}" for c, o, v in preds)
+    if order:
+        sql += " " + order
+    if limit:
+        sql += f" LIMIT {limit}"
+    kind, q = parse_sql(sql)
+    assert kind == "select"
+    assert [i.name for i in q.items] == cols
+    assert q.limit == (limit or None)

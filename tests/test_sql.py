@@ -281,3 +281,13 @@ def test_update_and_delete_statements(catalog):
     assert r["rows_deleted"].iloc[0] == 2
     df = execute_sql(catalog, "SELECT count(*) n FROM dml")
     assert df["n"].iloc[0] == 1
+
+
+def test_having_offset(sql_table):
+    cat, t = sql_table
+    df = execute_sql(cat,
+        "SELECT region, count(*) n FROM orders GROUP BY region "
+        "HAVING n > 333 ORDER BY region")
+    assert (df["n"] > 333).all()
+    df = execute_sql(cat, "SELECT id FROM orders ORDER BY id LIMIT 3 OFFSET 5")
+    assert df["id"].tolist() == [5, 6, 7]
