@@ -23,6 +23,15 @@ metadata commit semantics are kept compatible with the reference.
 __version__ = "0.1.0"
 
 from .config import IOConfig  # noqa: F401
+from .io.schema import Field, Schema  # noqa: F401
 from .meta.client import MetaClient  # noqa: F401
 from .tables.catalog import LakeSoulCatalog  # noqa: F401
 from .tables.table import LakeSoulTable  # noqa: F401
+
+
+def execute_sql(catalog, sql, device=None):
+    """Run a SQL statement against a catalog (console surface re-export,
+    see :mod:`lakesoul_amd.sql`)."""
+    from .sql import execute_sql as _run
+
+    return _run(catalog, sql, device=device)
