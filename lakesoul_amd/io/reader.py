@@ -412,7 +412,17 @@ class LakeSoulScan:
             len(unit.files) > 1 or self.cdc_column is not None or bool(self.merge_ops)
         )
         if oversized and self._chunkable():
-            return self._read_unit_chunked(unit)
+            try:
+                return self._read_unit_chunked(unit)
+            except MemoryError:
+                raise
+            except Exception as e:
+                # e.g. foreign files without PK stats: fall through to the
+                # non-chunked degraded paths below
+                import warnings
+
+                warnings.warn(f"chunked merge unavailable ({e}); "
+                              "falling back")
         if self.device == "cuda" and (
             oversized or (needs_merge and not self._gpu_merge_supported())
         ):
