@@ -77,6 +77,7 @@ class SelectItem:
     name: str = ""                 # column name (col/agg arg)
     fn: str = ""                   # aggregate fn
     alias: str = ""
+    distinct: bool = False         # count(DISTINCT col)
 
     @property
     def out_name(self) -> str:
@@ -291,6 +292,7 @@ class _Parser:
         if k == "id" and v.lower() in _AGGS and self.peek(1) == ("op", "("):
             fn = self.next()[1].lower()
             self.expect("op", "(")
+            distinct = self.accept("kw", "distinct")
             if self.accept("op", "*"):
                 if fn != "count":
                     raise SqlError(f"{fn}(*) not supported")
@@ -298,7 +300,9 @@ class _Parser:
             else:
                 arg = self.qualified_id()
             self.expect("op", ")")
-            item = SelectItem("agg", name=arg, fn=fn)
+            if distinct and fn != "count":
+                raise SqlError("DISTINCT only supported inside count()")
+            item = SelectItem("agg", name=arg, fn=fn, distinct=distinct)
         elif k == "id":
             item = SelectItem("col", name=self.qualified_id())
         else:
@@ -595,7 +599,10 @@ def _project_and_finish(q: Query, df, col, all_cols):
                     raise SqlError("SELECT * with aggregates is not valid")
                 s = sub[col(it.name)] if it.name else None
                 if it.fn == "count":
-                    row[it.out_name] = len(sub) if s is None else int(s.notna().sum())
+                    if it.distinct and s is not None:
+                        row[it.out_name] = int(s.dropna().nunique())
+                    else:
+                        row[it.out_name] = len(sub) if s is None else int(s.notna().sum())
                 elif it.fn == "sum":
                     row[it.out_name] = s.sum()
                 elif it.fn == "min":

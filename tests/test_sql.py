@@ -291,3 +291,12 @@ def test_having_offset(sql_table):
     assert (df["n"] > 333).all()
     df = execute_sql(cat, "SELECT id FROM orders ORDER BY id LIMIT 3 OFFSET 5")
     assert df["id"].tolist() == [5, 6, 7]
+
+
+def test_count_distinct(sql_table):
+    cat, t = sql_table
+    df = execute_sql(cat, "SELECT count(DISTINCT region) r FROM orders")
+    assert df["r"].iloc[0] == 3
+    df = execute_sql(cat,
+        "SELECT region, count(DISTINCT qty) q FROM orders GROUP BY region ORDER BY region")
+    assert (df["q"] <= 9).all() and (df["q"] >= 1).all()
