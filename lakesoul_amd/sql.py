@@ -631,7 +631,9 @@ def _project_and_finish(q: Query, df, col, all_cols):
                     cols.append(it.out_name)
             out = out[cols]
         else:
-            out = pd.DataFrame([agg_series(df)])
+            # dict-of-columns keeps count() integral next to float aggs
+            row = agg_series(df)
+            out = pd.DataFrame({k: [v] for k, v in row.items()})
     else:
         cols, ren = [], {}
         for it in q.items:
