@@ -587,7 +587,7 @@ def _project_and_finish(q: Query, df, col, all_cols):
 
     has_agg = any(it.kind == "agg" for it in q.items)
     if has_agg or q.group_by:
-        def agg_series(sub: pd.DataFrame):
+        def agg_row(sub: pd.DataFrame):
             row = {}
             for it in q.items:
                 if it.kind == "col":
@@ -611,7 +611,10 @@ def _project_and_finish(q: Query, df, col, all_cols):
                     row[it.out_name] = s.max()
                 elif it.fn == "avg":
                     row[it.out_name] = float(s.mean())
-            return pd.Series(row)
+            return row
+
+        def agg_series(sub: pd.DataFrame):
+            return pd.Series(agg_row(sub))
 
         if q.group_by:
             gcols = [col(g) for g in q.group_by]
@@ -632,8 +635,7 @@ def _project_and_finish(q: Query, df, col, all_cols):
             out = out[cols]
         else:
             # dict-of-columns keeps count() integral next to float aggs
-            row = agg_series(df)
-            out = pd.DataFrame({k: [v] for k, v in row.items()})
+            out = pd.DataFrame({k: [v] for k, v in agg_row(df).items()})
     else:
         cols, ren = [], {}
         for it in q.items:
