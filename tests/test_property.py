@@ -244,3 +244,57 @@ def test_generated_selects_parse(cols, preds, order, limit):
     assert kind == "select"
     assert [i.name for i in q.items] == cols
     assert q.limit == (limit or None)
+
+
+# ---------------------------------------------------------------------- #
+# SQL executor vs pandas oracle on generated queries
+
+
+@pytest.fixture(scope="module")
+def sql_oracle_table(tmp_path_factory):
+    from lakesoul_amd.meta.client import MetaClient
+    from lakesoul_amd.meta.store import SqliteMetaStore
+    from lakesoul_amd.tables.catalog import LakeSoulCatalog
+    from lakesoul_amd.io.schema import Field, Schema
+
+    d = tmp_path_factory.mktemp("sqlprop")
+    cat = LakeSoulCatalog(MetaClient(SqliteMetaStore(str(d / "m.db"))),
+                          warehouse=str(d / "wh"))
+    t = cat.create_table(
+        "props",
+        Schema([Field("id", "int64", False), Field("price", "float64"),
+                Field("qty", "int64"), Field("region", "string")]),
+        primary_keys=["id"], hash_bucket_num=2,
+    )
+    rng = np.random.default_rng(0)
+    n = 500
+    df = pd.DataFrame({
+        "id": np.arange(n, dtype=np.int64),
+        "price": rng.uniform(0, 100, n).round(2),
+        "qty": rng.integers(0, 10, n),
+        "region": [["east", "west"][i % 2] for i in range(n)],
+    })
+    t.upsert({c: df[c].to_numpy() if c != "region" else df[c].tolist()
+              for c in df.columns})
+    return cat, df
+
+
+_pcol = st.sampled_from(["id", "price", "qty"])
+_pop = st.sampled_from([("=", "=="), ("!=", "!="), ("<", "<"),
+                        ("<=", "<="), (">", ">"), (">=", ">=")])
+_pval = st.integers(0, 80)
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.lists(st.tuples(_pcol, _pop, _pval), min_size=1, max_size=3))
+def test_sql_where_matches_pandas(sql_oracle_table, preds):
+    from lakesoul_amd.sql import execute_sql
+
+    cat, df = sql_oracle_table
+    where = " AND ".join(f"{c} {sqlop} {v}" for c, (sqlop, _), v in preds)
+    got = execute_sql(cat, f"SELECT id FROM props WHERE {where} ORDER BY id")
+    mask = pd.Series(True, index=df.index)
+    for c, (_, pdop), v in preds:
+        mask &= eval(f"df[c] {pdop} v")
+    expect = df[mask]["id"].sort_values().tolist()
+    assert got["id"].tolist() == expect
