@@ -298,3 +298,54 @@ def test_sql_where_matches_pandas(sql_oracle_table, preds):
         mask &= eval(f"df[c] {pdop} v")
     expect = df[mask]["id"].sort_values().tolist()
     assert got["id"].tolist() == expect
+
+
+@settings(max_examples=25, deadline=None)
+@given(
+    st.lists(
+        st.tuples(
+            st.dictionaries(st.integers(0, 25), st.integers(-99, 99),
+                            min_size=1, max_size=15),
+            st.booleans(),   # include column b in this commit?
+        ),
+        min_size=1, max_size=5,
+    )
+)
+def test_merge_partial_columns_matches_model(history):
+    """Partial-column upserts (schema-evolution writes): UseLast must pick
+    the newest file that HAS the column — modeled per key/column."""
+    from lakesoul_amd.io.merge_cpu import NpColumn, merge_sorted_files
+
+    files, present = [], []
+    model_a, model_b = {}, {}
+    for h, with_b in history:
+        ids = np.array(sorted(h.keys()), dtype=np.int64)
+        a_vals = np.array([h[i] for i in ids], dtype=np.int64)
+        cols = {"id": NpColumn("int64", data=ids),
+                "a": NpColumn("int64", data=a_vals)}
+        pres = {"id", "a"}
+        if with_b:
+            cols["b"] = NpColumn("int64", data=a_vals * 2)
+            pres.add("b")
+        else:
+            # alignment requires null-filled placeholder
+            cols["b"] = NpColumn("int64", data=np.zeros(len(ids), np.int64),
+                                 validity=np.zeros(len(ids), np.uint8))
+        files.append(cols)
+        present.append(pres)
+        for i in ids:
+            model_a[int(i)] = h[int(i)]
+            if with_b:
+                model_b[int(i)] = h[int(i)] * 2
+    merged = merge_sorted_files(files, ["id"], present=present)
+    got_ids = merged["id"].data.tolist()
+    assert got_ids == sorted(model_a)
+    np.testing.assert_array_equal(
+        merged["a"].data, [model_a[i] for i in got_ids])
+    bv = merged["b"].validity
+    for pos, i in enumerate(got_ids):
+        if i in model_b:
+            assert bv is None or bv[pos]
+            assert merged["b"].data[pos] == model_b[i]
+        else:
+            assert bv is not None and not bv[pos]
