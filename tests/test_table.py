@@ -453,3 +453,36 @@ def test_chunked_merge_with_merge_ops_and_cdc(catalog, monkeypatch):
     assert any("chunked" in str(x.message) for x in w), [str(x.message) for x in w]
     pd.testing.assert_frame_equal(full, chunked)
     assert len(full) == n - n // 1000
+
+
+def test_composite_mixed_pk_merge_order_cpu():
+    """merge_key_order generic path: composite (int32, string, int64) PK
+    sorts lexicographically per column with stable snapshot ties."""
+    import torch
+
+    from lakesoul_amd.io.batch import Batch
+    from lakesoul_amd.io.merge_gpu import merge_key_order
+    from lakesoul_amd.io.schema import Field, Schema
+
+    sch = Schema([Field("a", "int32", False), Field("b", "string", False),
+                  Field("c", "int64", False)])
+    rows1 = [(1, "x", 5), (1, "y", 1), (2, "a", 9)]
+    rows2 = [(1, "x", 5), (1, "x", 7), (2, "a", 2)]
+
+    def mk(rows):
+        return Batch.from_dict({
+            "a": np.array([r[0] for r in rows], dtype=np.int32),
+            "b": [r[1] for r in rows],
+            "c": np.array([r[2] for r in rows], dtype=np.int64),
+        }, sch)
+
+    b1, b2 = mk(rows1), mk(rows2)
+    cols = [[b1.columns[k] for k in ("a", "b", "c")],
+            [b2.columns[k] for k in ("a", "b", "c")]]
+    order, keys, eq = merge_key_order(cols, [3, 3], torch.device("cpu"))
+    allrows = rows1 + rows2
+    got = [allrows[i] for i in order.tolist()]
+    assert got == sorted(got)  # python tuple order == our LSD order
+    # stability: equal (1,'x',5) rows keep file order (row 0 before row 3)
+    i0, i3 = order.tolist().index(0), order.tolist().index(3)
+    assert i0 < i3
