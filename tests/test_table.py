@@ -1,6 +1,8 @@
 """End-to-end table tests on the CPU path: write / upsert / MOR scan /
 merge operators / CDC / compaction / time travel / sharding."""
 
+import os
+
 import numpy as np
 import pandas as pd
 import pytest
@@ -486,3 +488,36 @@ def test_composite_mixed_pk_merge_order_cpu():
     # stability: equal (1,'x',5) rows keep file order (row 0 before row 3)
     i0, i3 = order.tolist().index(0), order.tolist().index(3)
     assert i0 < i3
+
+
+def test_mor_with_foreign_written_delta(catalog, tmp_path):
+    """A delta file written by pyarrow (dict-encoded, snappy) merges
+    correctly with our zstd PLAIN files — mixed-codec MOR units."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    from lakesoul_amd.meta.entities import (CommitOp, DataCommitInfo,
+                                            DataFileOp, FileOp)
+
+    t = _mk_pk_table(catalog, "foreign", buckets=1)
+    n = 5000
+    t.upsert({"id": np.arange(n, dtype=np.int64), "v": np.zeros(n),
+              "s": ["ours"] * n})
+    # foreign delta: ids 0..999 overwritten, written by pyarrow with
+    # dictionary + snappy (sorted by pk as the contract requires)
+    foreign = str(tmp_path / "part-foreign0000000_0000.parquet")
+    pq.write_table(pa.table({
+        "id": pa.array(np.arange(1000, dtype=np.int64)),
+        "v": pa.array(np.full(1000, 7.5)),
+        "s": pa.array(["theirs"] * 1000),
+    }), foreign, use_dictionary=True, compression="snappy")
+    t.client.commit_data_commit_info(DataCommitInfo(
+        table_id=t.table_id, partition_desc="-5",
+        file_ops=[DataFileOp(foreign, FileOp.add, os.path.getsize(foreign))],
+        commit_op=CommitOp.MergeCommit,
+    ))
+    df = _df(t)
+    assert len(df) == n
+    assert (df[df.id < 1000]["s"] == "theirs").all()
+    assert (df[df.id >= 1000]["s"] == "ours").all()
+    assert (df[df.id < 1000]["v"] == 7.5).all()
