@@ -35,7 +35,7 @@ _KEYWORDS = {
     "or", "not", "in", "is", "null", "between", "as", "asc", "desc",
     "show", "tables", "namespaces", "describe", "distinct", "version",
     "join", "inner", "left", "on", "insert", "into", "values",
-    "update", "set", "delete", "offset", "having",
+    "update", "set", "delete", "offset", "having", "timestamp",
 }
 
 _AGGS = {"count", "sum", "min", "max", "avg"}
@@ -112,6 +112,7 @@ class Query:
     offset: int = 0
     distinct: bool = False
     version: Optional[int] = None
+    timestamp_ms: Optional[int] = None
 
 
 class _Parser:
@@ -241,6 +242,8 @@ class _Parser:
             q.alias = self.next()[1]
         if self.accept("kw", "version"):   # time travel: FROM t VERSION 3
             q.version = int(self.expect("num"))
+        elif self.accept("kw", "timestamp"):  # FROM t TIMESTAMP <epoch_ms>
+            q.timestamp_ms = int(self.expect("num"))
         k, v = self.peek()
         if (k == "kw" and v in ("join", "inner", "left")):
             kind = "inner"
@@ -739,11 +742,13 @@ def _execute_select(catalog, q: Query, device=None):
     if (len(q.items) == 1 and q.items[0].kind == "agg"
             and q.items[0].fn == "count" and not q.items[0].name
             and q.where is None and not q.group_by and not q.distinct):
-        n = t.scan(version=q.version, device=device).count()
+        n = t.scan(version=q.version, timestamp_ms=q.timestamp_ms,
+                   device=device).count()
         return pd.DataFrame({q.items[0].out_name: [n]})
 
     scan = t.scan(columns=sorted(need) or None, filters=q.where,
-                  version=q.version, device=device)
+                  version=q.version, timestamp_ms=q.timestamp_ms,
+                  device=device)
     df = scan.to_arrow().to_pandas()
     return _project_and_finish(q, df, lambda n: n, all_cols=schema_cols)
 

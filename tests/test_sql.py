@@ -300,3 +300,18 @@ def test_count_distinct(sql_table):
     df = execute_sql(cat,
         "SELECT region, count(DISTINCT qty) q FROM orders GROUP BY region ORDER BY region")
     assert (df["q"] <= 9).all() and (df["q"] >= 1).all()
+
+
+def test_timestamp_time_travel(sql_table):
+    import time as _time
+
+    cat, t = sql_table
+    _time.sleep(0.01)
+    ts_between = int(_time.time() * 1000)
+    _time.sleep(0.01)
+    t.upsert({"id": np.array([0], dtype=np.int64), "price": np.array([-9.0]),
+              "qty": np.array([0], dtype=np.int64), "region": ["x"]})
+    old = execute_sql(cat, f"SELECT price FROM orders TIMESTAMP {ts_between} WHERE id = 0")
+    now = execute_sql(cat, "SELECT price FROM orders WHERE id = 0")
+    assert now["price"].iloc[0] == -9.0
+    assert old["price"].iloc[0] != -9.0
