@@ -18,7 +18,7 @@ import re
 from dataclasses import dataclass, field
 from typing import List, Optional, Tuple
 
-from .io.filters import And, Cmp, Expr, IsNull, Not, Or
+from .io.filters import And, Cmp, Expr, IsNull, Literal, Not, Or
 
 _TOKEN_RE = re.compile(
     r"""\s*(?:
@@ -43,6 +43,28 @@ _AGGS = {"count", "sum", "min", "max", "avg"}
 
 class SqlError(ValueError):
     pass
+
+
+@dataclass
+class SubqueryIn(Expr):
+    """col IN (SELECT ...) — non-correlated; resolved to an OR-of-eq /
+    isin before execution."""
+    col: str
+    query: "Query"
+
+    def evaluate(self, batch):  # pragma: no cover - resolved earlier
+        raise SqlError("unresolved subquery")
+
+
+@dataclass
+class SubqueryCmp(Expr):
+    """col <op> (SELECT scalar ...) — resolved to a Cmp literal."""
+    col: str
+    op: str
+    query: "Query"
+
+    def evaluate(self, batch):  # pragma: no cover - resolved earlier
+        raise SqlError("unresolved subquery")
 
 
 def tokenize(sql: str) -> List[Tuple[str, str]]:
@@ -229,7 +251,7 @@ class _Parser:
 
     # -- SELECT ---------------------------------------------------------- #
 
-    def select(self) -> Query:
+    def select(self, nested: bool = False) -> Query:
         self.expect("kw", "select")
         q = Query(table="")
         q.distinct = self.accept("kw", "distinct")
@@ -283,7 +305,7 @@ class _Parser:
             q.limit = int(self.expect("num"))
         if self.accept("kw", "offset"):
             q.offset = int(self.expect("num"))
-        if self.peek()[0] != "eof":
+        if not nested and self.peek()[0] != "eof":
             raise SqlError(f"trailing tokens: {self.peek()[1]!r}")
         return q
 
@@ -372,6 +394,11 @@ class _Parser:
             neg = self.accept("kw", "not")
             self.expect("kw", "in")
             self.expect("op", "(")
+            if self.peek() == ("kw", "select"):
+                sub = self.select(nested=True)
+                self.expect("op", ")")
+                e = SubqueryIn(col, sub)
+                return Not(e) if neg else e
             vals = [self.literal()]
             while self.accept("op", ","):
                 vals.append(self.literal())
@@ -393,6 +420,11 @@ class _Parser:
                "<=": "lteq", ">": "gt", ">=": "gteq"}
         if op not in ops:
             raise SqlError(f"bad comparison operator {op!r}")
+        if self.peek() == ("op", "(") and self.peek(1) == ("kw", "select"):
+            self.next()
+            sub = self.select(nested=True)
+            self.expect("op", ")")
+            return SubqueryCmp(col, ops[op], sub)
         return Cmp(col, ops[op], self.literal())
 
 
@@ -712,9 +744,41 @@ def _strip_quals(q: Query, valid_quals) -> None:
     walk(q.where)
 
 
+def _resolve_subqueries(catalog, expr, device):
+    """Execute non-correlated subqueries and substitute literal sets /
+    scalars (the reference delegates this to DataFusion's planner)."""
+    if expr is None:
+        return None
+    if isinstance(expr, (And, Or)):
+        expr.left = _resolve_subqueries(catalog, expr.left, device)
+        expr.right = _resolve_subqueries(catalog, expr.right, device)
+        return expr
+    if isinstance(expr, Not):
+        expr.inner = _resolve_subqueries(catalog, expr.inner, device)
+        return expr
+    if isinstance(expr, SubqueryIn):
+        sub = _execute_select(catalog, expr.query, device=device)
+        if len(sub.columns) != 1:
+            raise SqlError("IN subquery must return one column")
+        vals = [v for v in sub.iloc[:, 0].tolist() if v is not None]
+        if not vals:
+            return Literal(False)
+        e: Expr = Cmp(expr.col, "eq", vals[0])
+        for x in vals[1:]:
+            e = Or(e, Cmp(expr.col, "eq", x))
+        return e
+    if isinstance(expr, SubqueryCmp):
+        sub = _execute_select(catalog, expr.query, device=device)
+        if len(sub.columns) != 1 or len(sub) != 1:
+            raise SqlError("scalar subquery must return exactly one value")
+        return Cmp(expr.col, expr.op, sub.iloc[0, 0])
+    return expr
+
+
 def _execute_select(catalog, q: Query, device=None):
     import pandas as pd
 
+    q.where = _resolve_subqueries(catalog, q.where, device)
     if q.join is not None:
         return _execute_join_select(catalog, q, device=device)
     t = catalog.table(q.table, q.namespace)

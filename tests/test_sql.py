@@ -315,3 +315,18 @@ def test_timestamp_time_travel(sql_table):
     now = execute_sql(cat, "SELECT price FROM orders WHERE id = 0")
     assert now["price"].iloc[0] == -9.0
     assert old["price"].iloc[0] != -9.0
+
+
+def test_in_subquery_and_scalar_subquery(join_tables):
+    df = execute_sql(join_tables,
+        "SELECT oid FROM jorders WHERE cust IN "
+        "(SELECT cid FROM jcust WHERE name = 'cust1') ORDER BY oid")
+    assert df["oid"].tolist() == [1, 6, 11, 16]
+    df = execute_sql(join_tables,
+        "SELECT count(*) n FROM jorders WHERE amount > "
+        "(SELECT avg(amount) m FROM jorders)")
+    assert df["n"].iloc[0] == 10  # amounts 0..190, avg 95 -> 10 above
+    df = execute_sql(join_tables,
+        "SELECT oid FROM jorders WHERE cust NOT IN (SELECT cid FROM jcust)"
+        " ORDER BY oid")
+    assert df["oid"].tolist() == [4, 9, 14, 19]  # cust 4 has no row
