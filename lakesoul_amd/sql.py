@@ -35,7 +35,7 @@ _KEYWORDS = {
     "or", "not", "in", "is", "null", "between", "as", "asc", "desc",
     "show", "tables", "namespaces", "describe", "distinct", "version",
     "join", "inner", "left", "on", "insert", "into", "values",
-    "update", "set", "delete", "offset", "having", "timestamp",
+    "update", "set", "delete", "offset", "having", "timestamp", "explain",
 }
 
 _AGGS = {"count", "sum", "min", "max", "avg"}
@@ -178,6 +178,9 @@ class _Parser:
             self.next()
             ns, name = self.table_name()
             return ("describe", (ns, name))
+        if (k, v) == ("kw", "explain"):
+            self.next()
+            return ("explain", self.select())
         if (k, v) == ("kw", "select"):
             return ("select", self.select())
         if (k, v) == ("kw", "insert"):
@@ -443,6 +446,8 @@ def execute_sql(catalog, sql: str, device: Optional[str] = None):
     kind, payload = parse_sql(sql)
     if kind == "insert":
         return _execute_insert(catalog, payload, device=device)
+    if kind == "explain":
+        return _explain_select(catalog, payload, device=device)
     if kind in ("update", "delete"):
         import pandas as pd
 
@@ -742,6 +747,33 @@ def _strip_quals(q: Query, valid_quals) -> None:
             e.col = strip(e.col)
 
     walk(q.where)
+
+
+def _explain_select(catalog, q: Query, device=None):
+    """EXPLAIN: the physical scan plan — per-unit file counts after
+    partition/stats/bucket pruning, pushdown summary (the reference shows
+    DataFusion's plan; this shows ours)."""
+    import pandas as pd
+
+    q.where = _resolve_subqueries(catalog, q.where, device)
+    rows = []
+    for tbl_name, ns, flt in ((q.table, q.namespace, q.where),) + (
+        ((q.join.table, q.join.namespace, None),) if q.join else ()
+    ):
+        t = catalog.table(tbl_name, ns)
+        scan = t.scan(filters=flt, version=q.version,
+                      timestamp_ms=q.timestamp_ms, device=device)
+        units = scan.plan()
+        total_files = sum(len(u.files) for u in units)
+        rows.append({
+            "table": f"{ns}.{tbl_name}",
+            "scan_units": len(units),
+            "files": total_files,
+            "hash_buckets": t.hash_bucket_num,
+            "pushdown": str(flt) if flt is not None else "",
+            "device": scan.device,
+        })
+    return pd.DataFrame(rows)
 
 
 def _resolve_subqueries(catalog, expr, device):

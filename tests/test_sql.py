@@ -330,3 +330,12 @@ def test_in_subquery_and_scalar_subquery(join_tables):
         "SELECT oid FROM jorders WHERE cust NOT IN (SELECT cid FROM jcust)"
         " ORDER BY oid")
     assert df["oid"].tolist() == [4, 9, 14, 19]  # cust 4 has no row
+
+
+def test_explain(sql_table):
+    cat, t = sql_table
+    df = execute_sql(cat, "EXPLAIN SELECT id FROM orders WHERE id = 7")
+    assert df["scan_units"].iloc[0] == 1  # bucket pruning visible
+    assert "id" in df["pushdown"].iloc[0]
+    df2 = execute_sql(cat, "EXPLAIN SELECT * FROM orders")
+    assert df2["scan_units"].iloc[0] == 4
