@@ -36,6 +36,8 @@ _KEYWORDS = {
     "show", "tables", "namespaces", "describe", "distinct", "version",
     "join", "inner", "left", "on", "insert", "into", "values",
     "update", "set", "delete", "offset", "having", "timestamp", "explain",
+    "create", "table", "drop", "primary", "key", "hash", "buckets",
+    "partition", "if", "exists",
 }
 
 _AGGS = {"count", "sum", "min", "max", "avg"}
@@ -185,6 +187,18 @@ class _Parser:
             return ("select", self.select())
         if (k, v) == ("kw", "insert"):
             return ("insert", self.insert())
+        if (k, v) == ("kw", "create"):
+            return ("create", self.create_table())
+        if (k, v) == ("kw", "drop"):
+            self.next()
+            self.expect("kw", "table")
+            if_exists = False
+            if self.accept("kw", "if"):
+                self.expect("kw", "exists")
+                if_exists = True
+            ns, name = self.table_name()
+            return ("drop", {"namespace": ns, "table": name,
+                             "if_exists": if_exists})
         if (k, v) == ("kw", "update"):
             self.next()
             ns, name = self.table_name()
@@ -239,6 +253,75 @@ class _Parser:
             return {"namespace": ns, "table": name, "columns": cols,
                     "select": self.select()}
         raise SqlError("INSERT expects VALUES or SELECT")
+
+    _SQL_TYPES = {
+        "bigint": "int64", "long": "int64", "int": "int32",
+        "integer": "int32", "smallint": "int16", "tinyint": "int8",
+        "double": "float64", "float": "float32", "real": "float32",
+        "varchar": "string", "string": "string", "text": "string",
+        "binary": "binary", "boolean": "bool", "bool": "bool",
+        "date": "date32", "bigint_ts": "timestamp[us]",
+    }
+
+    def sql_type(self) -> str:
+        t = self.expect("id").lower()
+        if t == "decimal" and self.accept("op", "("):
+            p_ = self.expect("num")
+            self.expect("op", ",")
+            sc = self.expect("num")
+            self.expect("op", ")")
+            return f"decimal({p_},{sc})"
+        if t == "timestamp":
+            return "timestamp[us]"
+        if t == "varchar" and self.accept("op", "("):
+            self.expect("num")
+            self.expect("op", ")")
+            return "string"
+        if t not in self._SQL_TYPES:
+            raise SqlError(f"unknown SQL type {t!r}")
+        return self._SQL_TYPES[t]
+
+    def create_table(self):
+        self.expect("kw", "create")
+        self.expect("kw", "table")
+        ns, name = self.table_name()
+        self.expect("op", "(")
+        cols = []
+        while True:
+            cname = self.expect("id")
+            dtype = self.sql_type()
+            nullable = True
+            if self.accept("kw", "not"):
+                self.expect("kw", "null")
+                nullable = False
+            cols.append((cname, dtype, nullable))
+            if not self.accept("op", ","):
+                break
+        self.expect("op", ")")
+        pks, buckets, parts = [], 1, []
+        while self.peek()[0] != "eof":
+            if self.accept("kw", "primary"):
+                self.expect("kw", "key")
+                self.expect("op", "(")
+                pks.append(self.expect("id"))
+                while self.accept("op", ","):
+                    pks.append(self.expect("id"))
+                self.expect("op", ")")
+            elif self.accept("kw", "hash"):
+                self.expect("kw", "buckets")
+                buckets = int(self.expect("num"))
+            elif self.accept("kw", "partition"):
+                self.expect("kw", "by")
+                self.expect("op", "(")
+                parts.append(self.expect("id"))
+                while self.accept("op", ","):
+                    parts.append(self.expect("id"))
+                self.expect("op", ")")
+            else:
+                raise SqlError(f"unexpected token {self.peek()[1]!r}")
+        return {"namespace": ns, "table": name, "columns": cols,
+                "primary_keys": pks, "hash_buckets": buckets,
+                "range_partitions": parts}
 
     def table_name(self):
         name = self.expect("id")
@@ -446,6 +529,30 @@ def execute_sql(catalog, sql: str, device: Optional[str] = None):
     kind, payload = parse_sql(sql)
     if kind == "insert":
         return _execute_insert(catalog, payload, device=device)
+    if kind == "create":
+        import pandas as pd
+
+        from .io.schema import Field, Schema
+
+        t = catalog.create_table(
+            payload["table"],
+            Schema([Field(n, d, nu) for n, d, nu in payload["columns"]]),
+            primary_keys=payload["primary_keys"],
+            hash_bucket_num=payload["hash_buckets"],
+            range_partitions=payload["range_partitions"],
+            namespace=payload["namespace"],
+        )
+        return pd.DataFrame({"table_id": [t.table_id]})
+    if kind == "drop":
+        import pandas as pd
+
+        try:
+            catalog.drop_table(payload["table"], payload["namespace"],
+                               delete_data=True)
+        except Exception:
+            if not payload["if_exists"]:
+                raise
+        return pd.DataFrame({"dropped": [payload["table"]]})
     if kind == "explain":
         return _explain_select(catalog, payload, device=device)
     if kind in ("update", "delete"):

@@ -154,7 +154,7 @@ def test_errors(sql_table):
     with pytest.raises(SqlError):
         execute_sql(cat, "SELECT nosuch FROM orders")
     with pytest.raises(SqlError):
-        execute_sql(cat, "DROP TABLE orders")
+        execute_sql(cat, "ALTER TABLE orders ADD COLUMN x BIGINT")
     with pytest.raises(SqlError):
         execute_sql(cat, "SELECT id, sum(qty) FROM orders")  # id not grouped
     with pytest.raises(SqlError):
@@ -339,3 +339,22 @@ def test_explain(sql_table):
     assert "id" in df["pushdown"].iloc[0]
     df2 = execute_sql(cat, "EXPLAIN SELECT * FROM orders")
     assert df2["scan_units"].iloc[0] == 4
+
+
+def test_create_and_drop_table_sql(catalog):
+    execute_sql(catalog,
+        "CREATE TABLE sales (id BIGINT NOT NULL, amt DECIMAL(10,2), "
+        "region VARCHAR(16), d DATE) PRIMARY KEY (id) HASH BUCKETS 2 "
+        "PARTITION BY (region)")
+    t = catalog.table("sales")
+    assert t.primary_keys == ["id"]
+    assert t.hash_bucket_num == 2
+    assert t.range_keys == ["region"]
+    assert t.schema.field("amt").dtype == "decimal(10,2)"
+    execute_sql(catalog,
+        "INSERT INTO sales VALUES (1, 9.5, 'east', 19000), (2, 1.25, 'west', 19001)")
+    df = execute_sql(catalog, "SELECT count(*) n FROM sales")
+    assert df["n"].iloc[0] == 2
+    execute_sql(catalog, "DROP TABLE sales")
+    assert not catalog.table_exists("sales")
+    execute_sql(catalog, "DROP TABLE IF EXISTS sales")  # no error
