@@ -37,7 +37,7 @@ _KEYWORDS = {
     "join", "inner", "left", "on", "insert", "into", "values",
     "update", "set", "delete", "offset", "having", "timestamp", "explain",
     "create", "table", "drop", "primary", "key", "hash", "buckets",
-    "partition", "if", "exists",
+    "partition", "if", "exists", "alter", "add", "column",
 }
 
 _AGGS = {"count", "sum", "min", "max", "avg"}
@@ -189,6 +189,21 @@ class _Parser:
             return ("insert", self.insert())
         if (k, v) == ("kw", "create"):
             return ("create", self.create_table())
+        if (k, v) == ("kw", "alter"):
+            self.next()
+            self.expect("kw", "table")
+            ns, name = self.table_name()
+            self.expect("kw", "add")
+            self.accept("kw", "column")
+            cols = []
+            while True:
+                cname = self.expect("id")
+                dtype = self.sql_type()
+                cols.append((cname, dtype))
+                if not self.accept("op", ","):
+                    break
+            return ("alter_add", {"namespace": ns, "table": name,
+                                  "columns": cols})
         if (k, v) == ("kw", "drop"):
             self.next()
             self.expect("kw", "table")
@@ -543,6 +558,14 @@ def execute_sql(catalog, sql: str, device: Optional[str] = None):
             namespace=payload["namespace"],
         )
         return pd.DataFrame({"table_id": [t.table_id]})
+    if kind == "alter_add":
+        import pandas as pd
+
+        from .io.schema import Field
+
+        t = catalog.table(payload["table"], payload["namespace"])
+        t.add_columns([Field(n, d) for n, d in payload["columns"]])
+        return pd.DataFrame({"added": [len(payload["columns"])]})
     if kind == "drop":
         import pandas as pd
 

@@ -154,7 +154,7 @@ def test_errors(sql_table):
     with pytest.raises(SqlError):
         execute_sql(cat, "SELECT nosuch FROM orders")
     with pytest.raises(SqlError):
-        execute_sql(cat, "ALTER TABLE orders ADD COLUMN x BIGINT")
+        execute_sql(cat, "GRANT SELECT ON orders TO bob")
     with pytest.raises(SqlError):
         execute_sql(cat, "SELECT id, sum(qty) FROM orders")  # id not grouped
     with pytest.raises(SqlError):
@@ -358,3 +358,12 @@ def test_create_and_drop_table_sql(catalog):
     execute_sql(catalog, "DROP TABLE sales")
     assert not catalog.table_exists("sales")
     execute_sql(catalog, "DROP TABLE IF EXISTS sales")  # no error
+
+
+def test_alter_add_column_sql(sql_table):
+    cat, t = sql_table
+    execute_sql(cat, "ALTER TABLE orders ADD COLUMN note VARCHAR(20), score DOUBLE")
+    t2 = cat.table("orders")
+    assert "note" in t2.schema.names() and "score" in t2.schema.names()
+    df = execute_sql(cat, "SELECT note, score FROM orders WHERE id = 1")
+    assert df["note"].isna().iloc[0]
