@@ -37,7 +37,8 @@ _KEYWORDS = {
     "join", "inner", "left", "on", "insert", "into", "values",
     "update", "set", "delete", "offset", "having", "timestamp", "explain",
     "create", "table", "drop", "primary", "key", "hash", "buckets",
-    "partition", "if", "exists", "alter", "add", "column",
+    "partition", "if", "exists", "alter", "add", "column", "compact",
+    "vacuum", "keep",
 }
 
 _AGGS = {"count", "sum", "min", "max", "avg"}
@@ -189,6 +190,18 @@ class _Parser:
             return ("insert", self.insert())
         if (k, v) == ("kw", "create"):
             return ("create", self.create_table())
+        if (k, v) == ("kw", "compact"):
+            self.next()
+            self.accept("kw", "table")
+            ns, name = self.table_name()
+            return ("compact", {"namespace": ns, "table": name})
+        if (k, v) == ("kw", "vacuum"):
+            self.next()
+            ns, name = self.table_name()
+            keep = 1
+            if self.accept("kw", "keep"):
+                keep = int(self.expect("num"))
+            return ("vacuum", {"namespace": ns, "table": name, "keep": keep})
         if (k, v) == ("kw", "alter"):
             self.next()
             self.expect("kw", "table")
@@ -558,6 +571,18 @@ def execute_sql(catalog, sql: str, device: Optional[str] = None):
             namespace=payload["namespace"],
         )
         return pd.DataFrame({"table_id": [t.table_id]})
+    if kind == "compact":
+        import pandas as pd
+
+        t = catalog.table(payload["table"], payload["namespace"])
+        t.compaction()
+        return pd.DataFrame({"compacted": [payload["table"]]})
+    if kind == "vacuum":
+        import pandas as pd
+
+        t = catalog.table(payload["table"], payload["namespace"])
+        removed = t.cleanup_old_versions(keep_latest=payload["keep"])
+        return pd.DataFrame({"files_removed": [removed]})
     if kind == "alter_add":
         import pandas as pd
 

@@ -367,3 +367,18 @@ def test_alter_add_column_sql(sql_table):
     assert "note" in t2.schema.names() and "score" in t2.schema.names()
     df = execute_sql(cat, "SELECT note, score FROM orders WHERE id = 1")
     assert df["note"].isna().iloc[0]
+
+
+def test_compact_and_vacuum_sql(sql_table):
+    cat, t = sql_table
+    for it in range(3):
+        t.upsert({"id": np.arange(50, dtype=np.int64),
+                  "price": np.full(50, float(it)),
+                  "qty": np.ones(50, dtype=np.int64), "region": ["east"] * 50})
+    files_before = len(t.files())
+    execute_sql(cat, "COMPACT TABLE orders")
+    assert len(t.files()) < files_before
+    r = execute_sql(cat, "VACUUM orders KEEP 1")
+    assert r["files_removed"].iloc[0] >= 0
+    df = execute_sql(cat, "SELECT count(*) n FROM orders")
+    assert df["n"].iloc[0] == 1000
