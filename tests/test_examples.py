@@ -11,6 +11,7 @@ ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 @pytest.mark.parametrize("script", [
     "quickstart.py", "vector_search.py", "flight_client.py",
+    "cdc_ingest.py",
 ])
 def test_example_runs(script):
     if script == "flight_client.py":
