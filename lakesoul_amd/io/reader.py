@@ -345,8 +345,10 @@ class LakeSoulScan:
             for unit in self.plan():
                 for f in self._localize(unit.files):
                     h = cpp().open_parquet(f)
-                    total += cpp().parquet_meta(h)["num_rows"]
-                    cpp().close_parquet(h)
+                    try:
+                        total += cpp().parquet_meta(h)["num_rows"]
+                    finally:
+                        cpp().close_parquet(h)
             return total
         return sum(b.num_rows for b in self.iter_batches())
 
