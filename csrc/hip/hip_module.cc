@@ -56,6 +56,15 @@ static hipStream_t cur_stream() {
 #define CHECK_GPU(t) TORCH_CHECK((t).is_cuda(), #t " must be on GPU"); \
   TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
 
+// optional/empty tensors may be placeholder CPU empties; any tensor the
+// kernel will actually dereference must be device-resident (a CPU
+// pointer reaching a kernel = memory fault, see Column.take fix)
+#define CHECK_GPU_NONEMPTY(t)                                   \
+  if ((t).numel()) {                                            \
+    TORCH_CHECK((t).is_cuda(), #t " must be on GPU");           \
+    TORCH_CHECK((t).is_contiguous(), #t " must be contiguous"); \
+  }
+
 static const uint8_t* opt_u8(const torch::Tensor& t) {
   return t.numel() ? t.data_ptr<uint8_t>() : nullptr;
 }
@@ -64,6 +73,8 @@ static const uint8_t* opt_u8(const torch::Tensor& t) {
 
 static torch::Tensor hash_fixed_column(torch::Tensor data, torch::Tensor validity,
                                        torch::Tensor prev, bool first) {
+  CHECK_GPU_NONEMPTY(validity);
+  CHECK_GPU_NONEMPTY(prev);
   CHECK_GPU(data);
   int64_t n = data.numel();
   auto out = torch::empty({n}, data.options().dtype(torch::kInt64));
@@ -88,6 +99,9 @@ static torch::Tensor hash_fixed_column(torch::Tensor data, torch::Tensor validit
 static torch::Tensor hash_string_column(torch::Tensor offsets, torch::Tensor bytes,
                                         torch::Tensor validity, torch::Tensor prev,
                                         bool first) {
+  CHECK_GPU_NONEMPTY(bytes);
+  CHECK_GPU_NONEMPTY(validity);
+  CHECK_GPU_NONEMPTY(prev);
   CHECK_GPU(offsets);
   int64_t n = offsets.numel() - 1;
   auto out = torch::empty({n}, offsets.options().dtype(torch::kInt64));
@@ -120,6 +134,8 @@ static torch::Tensor rle_expand(torch::Tensor payload, torch::Tensor runs, int64
 static torch::Tensor dict_gather_scatter(torch::Tensor dict_vals, torch::Tensor idx,
                                          torch::Tensor validity, torch::Tensor positions,
                                          int64_t elem_size, int64_t n) {
+  CHECK_GPU_NONEMPTY(validity);
+  CHECK_GPU_NONEMPTY(positions);
   CHECK_GPU(dict_vals);
   CHECK_GPU(idx);
   auto out = torch::empty({n * elem_size}, dict_vals.options().dtype(torch::kUInt8));
@@ -146,6 +162,8 @@ static torch::Tensor dict_gather_scatter(torch::Tensor dict_vals, torch::Tensor 
 static torch::Tensor scatter_valid(torch::Tensor dense, torch::Tensor validity,
                                    torch::Tensor positions, int64_t elem_size,
                                    int64_t n) {
+  CHECK_GPU_NONEMPTY(validity);
+  CHECK_GPU_NONEMPTY(positions);
   CHECK_GPU(dense);
   auto out = torch::zeros({n * elem_size}, dense.options().dtype(torch::kUInt8));
   if (elem_size == 1)
@@ -163,6 +181,8 @@ static torch::Tensor scatter_valid(torch::Tensor dense, torch::Tensor validity,
 
 static std::vector<torch::Tensor> merge_pairs(torch::Tensor kA, torch::Tensor vA,
                                               torch::Tensor kB, torch::Tensor vB) {
+  CHECK_GPU_NONEMPTY(vA);
+  CHECK_GPU_NONEMPTY(vB);
   CHECK_GPU(kA);
   CHECK_GPU(kB);
   int64_t nA = kA.numel(), nB = kB.numel();
@@ -202,6 +222,7 @@ static torch::Tensor pack_key_i64(torch::Tensor x) {
 }
 
 static torch::Tensor pack_key_2xi32(torch::Tensor hi, torch::Tensor lo) {
+  CHECK_GPU_NONEMPTY(lo);
   CHECK_GPU(hi);
   auto out = torch::empty({hi.numel()}, hi.options().dtype(torch::kInt64));
   launch_pack_key_2xi32(hi.data_ptr<int32_t>(), lo.data_ptr<int32_t>(),
@@ -213,6 +234,7 @@ static torch::Tensor pack_key_2xi32(torch::Tensor hi, torch::Tensor lo) {
 
 static std::vector<torch::Tensor> gather_fixed_multi(std::vector<torch::Tensor> cols,
                                                      torch::Tensor idx) {
+  for (auto& c : cols) CHECK_GPU_NONEMPTY(c);
   CHECK_GPU(idx);
   int64_t n = idx.numel();
   std::vector<torch::Tensor> outs;
@@ -236,6 +258,9 @@ static std::vector<torch::Tensor> gather_fixed_multi(std::vector<torch::Tensor> 
 
 static torch::Tensor gather_strings(torch::Tensor src_bytes, torch::Tensor src_offsets,
                                     torch::Tensor idx, torch::Tensor dst_offsets) {
+  CHECK_GPU_NONEMPTY(src_offsets);
+  CHECK_GPU_NONEMPTY(idx);
+  CHECK_GPU_NONEMPTY(dst_offsets);
   CHECK_GPU(src_bytes);
   int64_t n = idx.numel();
   int64_t total = dst_offsets.numel() ? dst_offsets[n].item<int64_t>() : 0;
@@ -248,6 +273,8 @@ static torch::Tensor gather_strings(torch::Tensor src_bytes, torch::Tensor src_o
 
 static torch::Tensor bytes_ne_mask(torch::Tensor offsets, torch::Tensor bytes,
                                    torch::Tensor pattern) {
+  CHECK_GPU_NONEMPTY(bytes);
+  CHECK_GPU_NONEMPTY(pattern);
   CHECK_GPU(offsets);
   int64_t n = offsets.numel() - 1;
   auto out = torch::empty({n}, offsets.options().dtype(torch::kUInt8));
@@ -262,6 +289,9 @@ static torch::Tensor bytes_ne_mask(torch::Tensor offsets, torch::Tensor bytes,
 static std::vector<torch::Tensor> segmented_sum(torch::Tensor vals, torch::Tensor grp,
                                                 torch::Tensor contrib, torch::Tensor validity,
                                                 int64_t ngroups) {
+  CHECK_GPU_NONEMPTY(grp);
+  CHECK_GPU_NONEMPTY(contrib);
+  CHECK_GPU_NONEMPTY(validity);
   CHECK_GPU(vals);
   int64_t n = vals.numel();
   auto sums = torch::zeros({ngroups}, vals.options());
@@ -289,6 +319,8 @@ static std::vector<torch::Tensor> segmented_sum(torch::Tensor vals, torch::Tenso
 
 static torch::Tensor segmented_last(torch::Tensor grp, torch::Tensor contrib,
                                     torch::Tensor validity, int64_t ngroups, int64_t n) {
+  CHECK_GPU_NONEMPTY(contrib);
+  CHECK_GPU_NONEMPTY(validity);
   CHECK_GPU(grp);
   auto out = torch::zeros({ngroups}, grp.options());
   launch_segmented_last(grp.data_ptr<int64_t>(), opt_u8(contrib), opt_u8(validity),
@@ -350,6 +382,10 @@ static py::list scan_unit_uselast(torch::Tensor vals, torch::Tensor validity_buf
                                   torch::Tensor soffs, torch::Tensor desc,
                                   int64_t nfiles, int64_t ncols, int64_t pk_ci,
                                   int64_t pk_es) {
+  CHECK_GPU_NONEMPTY(validity_buf);
+  CHECK_GPU_NONEMPTY(dicts);
+  CHECK_GPU_NONEMPTY(runs);
+  CHECK_GPU_NONEMPTY(soffs);
   CHECK_GPU(vals);
   py::gil_scoped_release rel;  // long device-side section (incl. one sync)
   auto device = vals.device();
