@@ -173,10 +173,13 @@ class _Parser:
         k, v = self.peek()
         if (k, v) == ("kw", "show"):
             self.next()
-            kind = self.expect("kw")
-            if kind not in ("tables", "namespaces"):
-                raise SqlError(f"SHOW {kind}?")
-            return ("show", kind)
+            k2, v2 = self.next()
+            if (k2, v2) == ("kw", "tables") or (k2, v2) == ("kw", "namespaces"):
+                return ("show", v2)
+            if k2 == "id" and v2.lower() in ("partitions", "history"):
+                ns, name = self.table_name()
+                return ("show_" + v2.lower(), {"namespace": ns, "table": name})
+            raise SqlError(f"SHOW {v2}?")
         if (k, v) == ("kw", "describe"):
             self.next()
             ns, name = self.table_name()
@@ -571,6 +574,31 @@ def execute_sql(catalog, sql: str, device: Optional[str] = None):
             namespace=payload["namespace"],
         )
         return pd.DataFrame({"table_id": [t.table_id]})
+    if kind == "show_partitions":
+        import pandas as pd
+
+        t = catalog.table(payload["table"], payload["namespace"])
+        descs = t.partition_descs()
+        return pd.DataFrame({
+            "partition": descs,
+            "latest_version": [t.latest_version(d) for d in descs],
+        })
+    if kind == "show_history":
+        import pandas as pd
+
+        t = catalog.table(payload["table"], payload["namespace"])
+        rows = []
+        for desc in t.partition_descs():
+            cur = t.client.store.get_latest_partition_info(t.table_id, desc)
+            if cur is None:
+                continue
+            for p_ in t.client.store.get_partition_versions_in_range(
+                    t.table_id, desc, 0, cur.version):
+                rows.append({"partition": desc, "version": p_.version,
+                             "commit_op": p_.commit_op.name,
+                             "timestamp_ms": p_.timestamp,
+                             "commits": len(p_.snapshot)})
+        return pd.DataFrame(rows)
     if kind == "compact":
         import pandas as pd
 

@@ -382,3 +382,12 @@ def test_compact_and_vacuum_sql(sql_table):
     assert r["files_removed"].iloc[0] >= 0
     df = execute_sql(cat, "SELECT count(*) n FROM orders")
     assert df["n"].iloc[0] == 1000
+
+
+def test_show_partitions_and_history(sql_table):
+    cat, t = sql_table
+    df = execute_sql(cat, "SHOW PARTITIONS orders")
+    assert df["partition"].tolist() == ["-5"]
+    h = execute_sql(cat, "SHOW HISTORY orders")
+    assert h["commit_op"].iloc[0] == "MergeCommit"
+    assert (h["version"].diff().dropna() == 1).all()
