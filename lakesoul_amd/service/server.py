@@ -196,6 +196,28 @@ def create_app(catalog=None, secret: Optional[str] = None):
         t.compaction()
         return {"ok": True}
 
+    @app.post("/sql")
+    async def run_sql(request: Request, claims: dict = Depends(auth)):
+        """Run a SQL statement; returns rows as JSON (the Flight gateway
+        streams Arrow IPC for bulk — this endpoint is the console/BI
+        convenience surface)."""
+        from fastapi.responses import JSONResponse
+
+        from ..sql import SqlError, execute_sql
+
+        body = await request.json()
+        query = body.get("query", "")
+        if not query:
+            raise HTTPException(400, "query required")
+        try:
+            df = execute_sql(catalog, query)
+        except SqlError as e:
+            raise HTTPException(400, str(e))
+        return JSONResponse({
+            "columns": list(df.columns),
+            "rows": json.loads(df.to_json(orient="values")),
+        })
+
     @app.get("/metrics")
     def get_metrics(claims: dict = Depends(auth)):
         return metrics.snapshot()

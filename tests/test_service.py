@@ -176,3 +176,15 @@ def test_metrics_prometheus_format(app_client):
     body = r.text
     assert "# TYPE lakesoul_total_rows counter" in body
     assert "lakesoul_active_streams" in body
+
+
+def test_sql_endpoint(app_client):
+    hdr = _token(app_client)
+    r = app_client.post("/sql", json={"query": "SELECT count(*) n FROM served"},
+                        headers=hdr)
+    assert r.status_code == 200
+    body = r.json()
+    assert body["columns"] == ["n"]
+    assert body["rows"][0][0] >= 0
+    r = app_client.post("/sql", json={"query": "BOGUS"}, headers=hdr)
+    assert r.status_code == 400
