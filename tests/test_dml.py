@@ -91,3 +91,65 @@ def test_table_stream_incremental(catalog):
     stream2 = TableStream(t, start_versions=dict(stream.positions), device="cpu")
     batches, adv = stream2.poll()
     assert not adv
+
+
+def test_cdc_insert_after_delete(catalog):
+    """Key lifecycle insert → delete → re-insert across commits: the
+    newest row-kind wins at every read point (reference CDC semantics)."""
+    from lakesoul_amd.io.schema import Field, Schema
+
+    t = catalog.create_table(
+        "cdclife",
+        Schema([Field("id", "int64", False), Field("v", "float64"),
+                Field("rowKinds", "string")]),
+        primary_keys=["id"],
+        properties={"lakesoul_cdc_change_column": "rowKinds"},
+    )
+    t.upsert({"id": np.array([1], dtype=np.int64), "v": np.array([1.0]),
+              "rowKinds": ["insert"]})
+    t.upsert({"id": np.array([1], dtype=np.int64), "v": np.array([0.0]),
+              "rowKinds": ["delete"]})
+    assert len(t.to_pandas()) == 0
+    t.upsert({"id": np.array([1], dtype=np.int64), "v": np.array([2.0]),
+              "rowKinds": ["insert"]})
+    df = t.to_pandas()
+    assert df["v"].tolist() == [2.0]
+    # time travel sees each state
+    assert len(t.to_pandas(version=0)) == 1
+    assert len(t.to_pandas(version=1)) == 0
+    # compaction drops the tombstone permanently
+    t.compaction()
+    df = t.to_pandas()
+    assert df["v"].tolist() == [2.0]
+
+
+def test_cdc_multi_key_interleaved(catalog):
+    from lakesoul_amd.io.schema import Field, Schema
+
+    t = catalog.create_table(
+        "cdcmix",
+        Schema([Field("id", "int64", False), Field("v", "float64"),
+                Field("rowKinds", "string")]),
+        primary_keys=["id"], hash_bucket_num=2,
+        properties={"lakesoul_cdc_change_column": "rowKinds"},
+    )
+    n = 3000
+    rng = np.random.default_rng(9)
+    alive = {}
+    t.upsert({"id": np.arange(n, dtype=np.int64), "v": np.zeros(n),
+              "rowKinds": ["insert"] * n})
+    alive = {i: 0.0 for i in range(n)}
+    for it in range(6):
+        ids = rng.choice(n, 400, replace=False).astype(np.int64)
+        kinds = ["delete" if rng.random() < 0.4 else "update" for _ in ids]
+        vals = np.full(400, float(it + 1))
+        t.upsert({"id": ids, "v": vals, "rowKinds": kinds})
+        for i, k in zip(ids, kinds):
+            if k == "delete":
+                alive.pop(int(i), None)
+            else:
+                alive[int(i)] = float(it + 1)
+    df = t.to_pandas().sort_values("id")
+    assert df["id"].tolist() == sorted(alive)
+    np.testing.assert_allclose(df["v"].to_numpy(),
+                               [alive[i] for i in sorted(alive)])
