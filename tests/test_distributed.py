@@ -109,7 +109,35 @@ def _run_table_shard(rank, world, port, tmpdir, results):
         dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("fn", [_run_exchange, _run_table_shard])
+def _run_rebalance(rank, world, rdv_file, tmpdir, results):
+    dist = _init(rank, world, rdv_file)
+    try:
+        from lakesoul_amd.io.batch import Batch
+        from lakesoul_amd.io.schema import Field, Schema
+        from lakesoul_amd.parallel.shard import rebalance_by_pk
+        from lakesoul_amd.utils.murmur3_np import bucket_ids_np, hash_column
+
+        n = 500
+        ids = np.arange(rank * 10000, rank * 10000 + n, dtype=np.int64)
+        schema = Schema([Field("id", "int64", False), Field("v", "float64")])
+        batch = Batch.from_dict({"id": ids, "v": ids.astype(np.float64)}, schema)
+        out = rebalance_by_pk(batch, "id")
+        got = out.columns["id"].data.numpy()
+        # every received id murmur-hashes to this rank
+        h = hash_column(got, np.uint32(42))
+        assert (bucket_ids_np(h, world) == rank).all()
+        # payload stayed aligned
+        np.testing.assert_allclose(out.columns["v"].data.numpy(),
+                                   got.astype(np.float64))
+        t = torch.tensor([out.num_rows])
+        dist.all_reduce(t)
+        assert int(t.item()) == world * n
+        results[rank] = "ok"
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("fn", [_run_exchange, _run_table_shard, _run_rebalance])
 def test_multiprocess_gloo(fn, tmp_path):
     world = 2
     port = str(tmp_path / "rdv")
