@@ -208,3 +208,27 @@ def test_vacuum_is_atomic_under_concurrent_commit(catalog):
     df = t.to_pandas()
     assert len(df) >= 10
     assert (df[df.id < 10]["v"] == 5.0).all()  # newest base survives
+
+
+def test_timing_phases(monkeypatch):
+    from lakesoul_amd.utils import timing
+
+    monkeypatch.setattr(timing, "ENABLED", True)
+    timing.reset()
+    with timing.phase("unit_test_phase"):
+        pass
+    rep = timing.report()
+    assert "unit_test_phase" in rep
+    timing.reset()
+    assert "unit_test_phase" not in timing.report()
+
+
+def test_ioconfig_env_fallback(monkeypatch):
+    """Option map falls back to LAKESOUL_<KEY> env (reference
+    config/mod.rs:160-165 env fallback)."""
+    from lakesoul_amd.config import IOConfig
+
+    monkeypatch.setenv("LAKESOUL_MY_CUSTOM_OPT", "hello")
+    cfg = IOConfig()
+    assert cfg.option("my_custom_opt", "fallback") == "hello"
+    assert cfg.option("absent_opt", "fallback") == "fallback"
