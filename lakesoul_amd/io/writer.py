@@ -160,23 +160,43 @@ def _partition_descs(batch: Batch, range_cols: Sequence[str]):
         return [(constants.NON_PARTITION_TABLE_PART_DESC, "", None)]
     import numpy as _np
 
-    vals = []
+    # vectorized grouping: encode each range column to small int codes,
+    # combine, then one argsort — no python-per-row loop (large
+    # range-partitioned writes would otherwise crawl)
+    codes = _np.zeros(n, dtype=_np.int64)
+    uniques_per_col = []
     for name in range_cols:
         c = batch.columns[name]
         if c.is_string:
             b = c.bytes_.cpu().numpy().tobytes()
             o = c.offsets.cpu().numpy()
-            vals.append([b[o[i]:o[i + 1]].decode() for i in range(n)])
+            col_vals = _np.array(
+                [b[o[i]:o[i + 1]].decode() for i in range(n)], dtype=object)
         else:
-            vals.append([str(x) for x in c.data.cpu().numpy().tolist()])
-    groups: Dict[str, list] = {}
-    for i in range(n):
-        desc = ",".join(f"{c}={vals[k][i]}" for k, c in enumerate(range_cols))
-        groups.setdefault(desc, []).append(i)
+            col_vals = c.data.cpu().numpy()
+        uniq, inv = _np.unique(col_vals, return_inverse=True)
+        uniques_per_col.append(uniq)
+        codes = codes * (len(uniq) + 1) + inv
+    uniq_codes, inv_codes = _np.unique(codes, return_inverse=True)
+    order = _np.argsort(inv_codes, kind="stable")
+    bounds = _np.searchsorted(inv_codes[order], _np.arange(len(uniq_codes) + 1))
     out = []
-    for desc, rows in groups.items():
+    for g in range(len(uniq_codes)):
+        rows = order[bounds[g]:bounds[g + 1]]
+        i0 = int(rows[0])
+        parts = []
+        for k, cname in enumerate(range_cols):
+            c = batch.columns[cname]
+            if c.is_string:
+                b = c.bytes_.cpu().numpy().tobytes()
+                o = c.offsets.cpu().numpy()
+                v = b[o[i0]:o[i0 + 1]].decode()
+            else:
+                v = str(c.data.cpu().numpy()[i0])
+            parts.append(f"{cname}={v}")
+        desc = ",".join(parts)
         subdir = "/".join(desc.split(","))
-        out.append((desc, subdir, torch.tensor(rows, dtype=torch.int64)))
+        out.append((desc, subdir, torch.from_numpy(_np.ascontiguousarray(rows))))
     return out
 
 
