@@ -79,3 +79,30 @@ def test_partition_pruning_with_filter(catalog):
     assert len(units) == 1 and units[0].partition_desc == "region=eu"
     df = scan.to_arrow().to_pandas()
     assert sorted(df["id"].tolist()) == [5, 6, 7, 8, 9]
+
+
+def test_dsl_roundtrip_equivalence(catalog):
+    """DSL-parsed filters and tuple filters select identical rows."""
+    import numpy as np
+
+    from lakesoul_amd.io.schema import Field, Schema
+
+    t = catalog.create_table(
+        "dslrt", Schema([Field("id", "int64", False), Field("v", "float64")]),
+        primary_keys=["id"], hash_bucket_num=2,
+    )
+    n = 2000
+    rng = np.random.default_rng(0)
+    t.upsert({"id": np.arange(n, dtype=np.int64), "v": rng.uniform(0, 10, n)})
+    cases = [
+        ("and(gteq(id, 100), lt(id, 200))",
+         [("id", ">=", 100), ("id", "<", 200)]),
+        ("or(eq(id, 5), eq(id, 7))", None),
+        ("not(lt(v, 5.0))", [("v", ">=", 5.0)]),
+    ]
+    for dsl, tup in cases:
+        a = t.to_pandas(filters=dsl).sort_values("id")["id"].tolist()
+        if tup is not None:
+            b = t.to_pandas(filters=tup).sort_values("id")["id"].tolist()
+            assert a == b, dsl
+        assert len(a) > 0
