@@ -349,3 +349,39 @@ def test_merge_partial_columns_matches_model(history):
             assert merged["b"].data[pos] == model_b[i]
         else:
             assert bv is not None and not bv[pos]
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.lists(st.tuples(
+    st.sampled_from(["a", "b", "c", "d_1", "x"]),
+    st.sampled_from(["int64", "int32", "float64", "float32", "string",
+                     "binary", "date32", "timestamp[us]", "decimal(10,2)",
+                     "bool", "int8", "int16"]),
+    st.booleans(),
+), min_size=1, max_size=6, unique_by=lambda t: t[0]))
+def test_schema_json_roundtrip(fields):
+    """Spark-style schema JSON serde is a lossless roundtrip for every
+    supported dtype (table_schema compatibility contract)."""
+    from lakesoul_amd.io.schema import (Field, Schema, schema_from_json,
+                                        schema_to_json)
+
+    sch = Schema([Field(n, d, nu) for n, d, nu in fields])
+    back = schema_from_json(schema_to_json(sch))
+    assert back == sch
+
+
+def test_murmur3_edge_values():
+    """Spark hash contract edge cases: -0.0 folds to 0.0, NaN bit
+    patterns hash by their bits, int min/max, empty string."""
+    from lakesoul_amd.utils import murmur3 as m3
+    from lakesoul_amd.utils.murmur3_np import create_hashes_np
+
+    assert m3.hash_float64(-0.0) == m3.hash_float64(0.0)
+    assert m3.hash_float32(-0.0) == m3.hash_float32(0.0)
+    a = np.array([0.0, -0.0], dtype=np.float64)
+    h = create_hashes_np([a])
+    assert h[0] == h[1]
+    for v in (-(2**63), 2**63 - 1, 0, -1):
+        assert create_hashes_np([np.array([v], dtype=np.int64)])[0] == \
+            np.uint32(m3.hash_int64(v) & 0xFFFFFFFF)
+    assert m3.hash_str("") == m3.hash_bytes(b"")
