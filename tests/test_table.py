@@ -521,3 +521,23 @@ def test_mor_with_foreign_written_delta(catalog, tmp_path):
     assert (df[df.id < 1000]["s"] == "theirs").all()
     assert (df[df.id >= 1000]["s"] == "ours").all()
     assert (df[df.id < 1000]["v"] == 7.5).all()
+
+
+def test_merge_operator_sum_last(catalog):
+    """SumLast: last value per FILE per key, summed across files —
+    differs from SumAll exactly when one file holds duplicate PKs."""
+    t = catalog.create_table(
+        "sumlast",
+        Schema([Field("id", "int64", False), Field("cnt", "int64")]),
+        primary_keys=["id"], hash_bucket_num=1,
+        properties={"merge_op.cnt": "SumLast"},
+    )
+    # one batch with a duplicated key: file keeps both rows (stable sort)
+    t.upsert({"id": np.array([1, 1, 2], dtype=np.int64),
+              "cnt": np.array([10, 20, 5], dtype=np.int64)})
+    t.upsert({"id": np.array([1, 2], dtype=np.int64),
+              "cnt": np.array([100, 50], dtype=np.int64)})
+    df = _df(t)
+    got = dict(zip(df["id"], df["cnt"]))
+    # key 1: file1 last=20, file2 last=100 -> 120 (SumAll would give 130)
+    assert got == {1: 120, 2: 55}
