@@ -676,8 +676,6 @@ def _execute_insert(catalog, ins: dict, device=None):
     # cast to schema dtypes
     for f in t.schema:
         if f.name in df.columns and f.is_fixed_width and not f.dtype.startswith("decimal"):
-            import numpy as _np
-
             from .io.batch import np_dtype_for
 
             df[f.name] = df[f.name].astype(np_dtype_for(f.dtype))

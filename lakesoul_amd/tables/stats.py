@@ -5,7 +5,7 @@ and per partition, straight from commit metadata (no data scan)."""
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import Dict, List
+from typing import List
 
 
 @dataclass
