@@ -477,8 +477,12 @@ class LakeSoulScan:
         files = self._localize(unit.files)
         # per-(file,rg): pk min/max + estimated decoded bytes
         spans = []   # (path, rg, lo, hi, est_bytes)
+        from .batch import np_dtype_for
+
         row_bytes = sum(
-            8 if not f.is_fixed_width else max(1, 8) for f in self.eval_schema
+            np.dtype(np_dtype_for(f.dtype)).itemsize if f.is_fixed_width
+            else 16   # strings: offsets + typical payload
+            for f in self.eval_schema
         ) or 8
         for path in files:
             h = cpp().open_parquet(path)
