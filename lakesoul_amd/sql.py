@@ -673,9 +673,22 @@ def _execute_insert(catalog, ins: dict, device=None):
         if len(df.columns) != len(names):
             raise SqlError("INSERT SELECT column count mismatch")
         df.columns = names
-    # cast to schema dtypes
+    # cast to schema dtypes; decimal literals are LOGICAL values and
+    # scale to the unscaled int64 backing
     for f in t.schema:
-        if f.name in df.columns and f.is_fixed_width and not f.dtype.startswith("decimal"):
+        if f.name not in df.columns or not f.is_fixed_width:
+            continue
+        if f.dtype.startswith("decimal"):
+            import decimal as _dec
+
+            from .io.schema import decimal_params
+
+            _, sc = decimal_params(f.dtype)
+            df[f.name] = [
+                int(_dec.Decimal(str(v)).scaleb(sc).to_integral_value())
+                for v in df[f.name]
+            ]
+        else:
             from .io.batch import np_dtype_for
 
             df[f.name] = df[f.name].astype(np_dtype_for(f.dtype))

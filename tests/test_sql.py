@@ -355,6 +355,11 @@ def test_create_and_drop_table_sql(catalog):
         "INSERT INTO sales VALUES (1, 9.5, 'east', 19000), (2, 1.25, 'west', 19001)")
     df = execute_sql(catalog, "SELECT count(*) n FROM sales")
     assert df["n"].iloc[0] == 2
+    import decimal
+
+    amts = execute_sql(catalog, "SELECT id, amt FROM sales ORDER BY id")
+    assert amts["amt"].tolist() == [decimal.Decimal("9.50"),
+                                    decimal.Decimal("1.25")]
     execute_sql(catalog, "DROP TABLE sales")
     assert not catalog.table_exists("sales")
     execute_sql(catalog, "DROP TABLE IF EXISTS sales")  # no error
