@@ -190,6 +190,14 @@ class LakeSoulTable:
             if col in self.primary_keys:
                 raise ValueError("cannot update a primary-key column")
             if f.is_fixed_width:
+                if f.dtype.startswith("decimal") and not isinstance(val, int):
+                    # logical literal -> unscaled int64 backing
+                    import decimal as _dec
+
+                    from ..io.schema import decimal_params
+
+                    _, sc = decimal_params(f.dtype)
+                    val = int(_dec.Decimal(str(val)).scaleb(sc).to_integral_value())
                 dev = batch.columns[col].data.device if batch.columns[col].data is not None else "cpu"
                 batch.columns[col] = Column(
                     f.dtype,

@@ -396,3 +396,15 @@ def test_show_partitions_and_history(sql_table):
     h = execute_sql(cat, "SHOW HISTORY orders")
     assert h["commit_op"].iloc[0] == "MergeCommit"
     assert (h["version"].diff().dropna() == 1).all()
+
+
+def test_update_decimal_column(catalog):
+    import decimal
+
+    execute_sql(catalog,
+        "CREATE TABLE updec (id BIGINT NOT NULL, amt DECIMAL(8,2)) "
+        "PRIMARY KEY (id) HASH BUCKETS 1")
+    execute_sql(catalog, "INSERT INTO updec VALUES (1, 1.00), (2, 2.00)")
+    execute_sql(catalog, "UPDATE updec SET amt = 7.77 WHERE id = 2")
+    df = execute_sql(catalog, "SELECT id, amt FROM updec ORDER BY id")
+    assert df["amt"].tolist() == [decimal.Decimal("1.00"), decimal.Decimal("7.77")]
