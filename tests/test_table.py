@@ -541,3 +541,19 @@ def test_merge_operator_sum_last(catalog):
     got = dict(zip(df["id"], df["cnt"]))
     # key 1: file1 last=20, file2 last=100 -> 120 (SumAll would give 130)
     assert got == {1: 120, 2: 55}
+
+
+def test_scan_batch_size_slicing(catalog):
+    """scan(batch_size=n) yields bounded batches covering all rows once
+    (reference: LakeSoulIOConfig batch_size, config/mod.rs)."""
+    t = _mk_pk_table(catalog, "bsz", buckets=2)
+    n = 5000
+    t.upsert({"id": np.arange(n, dtype=np.int64), "v": np.zeros(n),
+              "s": [f"x{i}" for i in range(n)]})
+    got = []
+    sizes = []
+    for b in t.scan(batch_size=700).iter_batches():
+        sizes.append(b.num_rows)
+        got.extend(b.columns["id"].data.tolist())
+    assert max(sizes) <= 700
+    assert sorted(got) == list(range(n))
