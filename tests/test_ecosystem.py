@@ -64,3 +64,22 @@ def test_ray_daft_gated(small_table):
         lray.read_lakesoul(small_table)
     with pytest.raises(ImportError, match="daft"):
         ldaft.read_lakesoul(small_table)
+
+
+def test_torch_dataloader_workers(small_table):
+    """DataLoader num_workers=2: worker_info sharding covers every row
+    exactly once (reference arrow/dataset.py rank*worker sharding)."""
+    import torch
+
+    from lakesoul_amd.torch.dataset import LakeSoulIterableDataset
+
+    ds = LakeSoulIterableDataset(small_table, columns=["id"], device="cpu")
+    dl = torch.utils.data.DataLoader(
+        ds, batch_size=None, num_workers=2,
+        collate_fn=lambda x: x, persistent_workers=False,
+    )
+    ids = []
+    for item in dl:
+        ids.extend(item["id"].flatten().tolist())
+    assert sorted(ids) == sorted(
+        small_table.to_pandas()["id"].tolist())
