@@ -19,6 +19,33 @@ LAKESOUL_EMPTY_STRING = "__L@KE$OUL_EMPTY_STRING__"
 LAKESOUL_EQ = "__L@KE$OUL_EQ__"
 LAKESOUL_COMMA = "__L@KE$OUL_COMMA__"
 
+
+def encode_partition_value(v) -> str:
+    """Encode one range-partition value for a partition_desc.
+
+    Mirrors the reference's format_scalar_value (helpers/mod.rs:206-221):
+    NULL -> LAKESOUL_NULL_STRING, "" -> LAKESOUL_EMPTY_STRING, and '='/','
+    escaped so the "col=val,col2=val2" desc stays parseable. (The
+    reference's encoder at helpers/mod.rs:219 escapes ',' twice and '='
+    never — its decoder at :325 expects EQ/COMMA sentinels, so we encode
+    what the decoder expects.)"""
+    if v is None:
+        return LAKESOUL_NULL_STRING
+    s = str(v)
+    if s == "":
+        return LAKESOUL_EMPTY_STRING
+    return s.replace("=", LAKESOUL_EQ).replace(",", LAKESOUL_COMMA)
+
+
+def decode_partition_value(s: str):
+    """Inverse of encode_partition_value (reference helpers/mod.rs:325).
+    Returns None for the NULL sentinel."""
+    if s == LAKESOUL_NULL_STRING:
+        return None
+    if s == LAKESOUL_EMPTY_STRING:
+        return ""
+    return s.replace(LAKESOUL_EQ, "=").replace(LAKESOUL_COMMA, ",")
+
 # CDC row-kind column values (reference: flink LakeSoulRecordConvert RowKind -> cdc column)
 CDC_INSERT = "insert"
 CDC_UPDATE = "update"

@@ -166,7 +166,36 @@ class Batch:
         if mod.startswith("pyarrow"):
             return cls.from_arrow(data, schema)
         if mod.startswith("pandas"):
-            return cls.from_dict({c: data[c].to_numpy() for c in data.columns}, schema)
+            # nullable extension dtypes (Int32, boolean, string, …) carry an
+            # NA mask that plain .to_numpy() destroys (fills with a garbage
+            # sentinel); extract it as a validity mask. Plain float NaN stays
+            # NaN (not null) — unchanged semantics.
+            import pandas as _pd
+
+            d, masks = {}, {}
+            for c in data.columns:
+                s = data[c]
+                if _pd.api.types.is_extension_array_dtype(s.dtype):
+                    na = s.isna().to_numpy()
+                    if na.any():
+                        masks[c] = (~na).astype(np.uint8)
+                        fname_f = schema.field(c) if c in schema.names() else None
+                        if fname_f is not None and not fname_f.is_fixed_width:
+                            d[c] = [None if na[i] else s.iloc[i] for i in range(len(s))]
+                        else:
+                            d[c] = np.where(na, 0, s.to_numpy(dtype="object")).astype(
+                                np_dtype_for(fname_f.dtype) if fname_f is not None else "int64")
+                        continue
+                    d[c] = s.to_numpy(
+                        dtype=None if not c in schema.names() or not schema.field(c).is_fixed_width
+                        else np_dtype_for(schema.field(c).dtype))
+                else:
+                    d[c] = s.to_numpy()
+            b = cls.from_dict(d, schema)
+            for c, m in masks.items():
+                if b.columns[c].validity is None:
+                    b.columns[c].validity = torch.from_numpy(m)
+            return b
         if isinstance(data, dict):
             return cls.from_dict(data, schema)
         raise TypeError(f"unsupported data type {type(data)}")

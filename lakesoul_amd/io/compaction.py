@@ -56,7 +56,7 @@ def compact_partition(table, partition_desc: str, device: Optional[str] = None) 
         out_dir = os.path.join(table.table_path, subdir, constants.COMPACT_DIR)
         os.makedirs(out_dir, exist_ok=True)
         fpath = os.path.join(
-            out_dir, f"part-{random_str(16)}_{unit.bucket_id:04d}.parquet"
+            out_dir, f"part-{random_str(16)}_{max(unit.bucket_id, 0):04d}.parquet"
         )
         size = _write_batch_to_file(
             fpath, batch, cfg.compression, cfg.compression_level, cfg.max_row_group_size

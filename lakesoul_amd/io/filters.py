@@ -127,6 +127,12 @@ class IsNull(Expr):
             m = ~(c.validity.to(torch.bool))
         return ~m if self.negate else m
 
+    def partition_prune(self, pv):
+        if self.col not in pv:
+            return True
+        is_null = pv[self.col] is None
+        return (not is_null) if self.negate else is_null
+
 
 @dataclass
 class Cmp(Expr):
@@ -203,6 +209,10 @@ class Cmp(Expr):
     def partition_prune(self, pv):
         if self.col not in pv:
             return True
+        if pv[self.col] is None:
+            # NULL partition value: comparisons are unknown -> no row of
+            # this partition can satisfy the predicate
+            return False
         try:
             col_v = type(self.value)(pv[self.col]) if not isinstance(self.value, str) else pv[self.col]
         except (TypeError, ValueError):

@@ -173,7 +173,7 @@ class LakeSoulScan:
                 for kv in desc.split(","):
                     if "=" in kv:
                         k, v = kv.split("=", 1)
-                        pv[k] = v
+                        pv[k] = constants.decode_partition_value(v)
                 if self.filter_expr.partition_prune(pv):
                     kept.append(desc)
             descs = kept
@@ -204,7 +204,10 @@ class LakeSoulScan:
                         warnings.warn(f"skipping missing file {op.path}")
                         continue
                 b = extract_hash_bucket_id(op.path)
-                by_bucket.setdefault(b if b is not None else 0, []).append(op.path)
+                # bucket_id -1 = unknown (file name lacks the part-*_NNNN
+                # suffix, e.g. foreign/imported files): exempt from bucket
+                # pruning below
+                by_bucket.setdefault(b if b is not None else -1, []).append(op.path)
             for b, files in sorted(by_bucket.items()):
                 if self.filter_expr is not None:
                     files = self.prune_files_by_stats(files)
@@ -215,7 +218,8 @@ class LakeSoulScan:
         # PK point-filter bucket pruning (reader.rs:164-225)
         pruned_bucket = self._bucket_filter()
         if pruned_bucket is not None:
-            units = [u for u in units if u.bucket_id in pruned_bucket]
+            units = [u for u in units
+                     if u.bucket_id < 0 or u.bucket_id in pruned_bucket]
         if self._shard is not None:
             rank, ws = self._shard
             units = [u for i, u in enumerate(units) if i % ws == rank]
@@ -600,20 +604,28 @@ class LakeSoulScan:
         return Batch(self.eval_schema, cols)
 
     def _range_value_column(self, f, unit: ScanUnit, merged) -> Column:
-        """Materialize a range-partition column from the partition_desc."""
+        """Materialize a range-partition column from the partition_desc.
+
+        Values are sentinel-decoded (NULL/''/','/'=' round trip —
+        reference helpers/mod.rs:325); the NULL sentinel materializes as
+        an all-null column."""
         n = len(next(iter(merged.values())))
-        val = None
+        val, found = None, False
         for kv in unit.partition_desc.split(","):
             if "=" in kv and kv.split("=", 1)[0] == f.name:
-                val = kv.split("=", 1)[1]
+                val = constants.decode_partition_value(kv.split("=", 1)[1])
+                found = True
+        is_null = found and val is None
+        validity = (torch.zeros(n, dtype=torch.uint8) if is_null else None)
         if f.is_fixed_width:
             npdt = _np_target(f.dtype)
-            arr = np.full(n, npdt(val) if val is not None else 0, dtype=npdt)
-            return Column(f.dtype, data=torch.from_numpy(arr))
+            arr = np.full(n, npdt(val) if (found and not is_null) else 0, dtype=npdt)
+            return Column(f.dtype, data=torch.from_numpy(arr), validity=validity)
         enc = (val or "").encode()
         offs = np.arange(n + 1, dtype=np.int32) * len(enc)
         bys = np.frombuffer(enc * n, dtype=np.uint8).copy() if n else np.empty(0, np.uint8)
-        return Column(f.dtype, offsets=torch.from_numpy(offs), bytes_=torch.from_numpy(bys))
+        return Column(f.dtype, offsets=torch.from_numpy(offs),
+                      bytes_=torch.from_numpy(bys), validity=validity)
 
     def _read_file_cpu(self, path: str, names: Sequence[str],
                        rg_subset: Optional[Sequence[int]] = None) -> Dict[str, NpColumn]:

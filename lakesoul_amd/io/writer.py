@@ -161,21 +161,31 @@ def _partition_descs(batch: Batch, range_cols: Sequence[str]):
     import numpy as _np
 
     # vectorized grouping: encode each range column to small int codes,
-    # combine, then one argsort — no python-per-row loop (large
-    # range-partitioned writes would otherwise crawl)
+    # combine, then one argsort — no python-per-row loop on the fixed-width
+    # path (large range-partitioned writes would otherwise crawl). Values
+    # go through encode_partition_value so NULL/''/','/'=' survive the
+    # "col=val,col2=val2" desc round trip (reference helpers/mod.rs:206-221;
+    # ADVICE r1 high finding).
     codes = _np.zeros(n, dtype=_np.int64)
-    uniques_per_col = []
+    encoded_cols = []  # per range col: object ndarray of encoded strings
     for name in range_cols:
         c = batch.columns[name]
+        valid = None if c.validity is None else c.validity.cpu().numpy().astype(bool)
         if c.is_string:
             b = c.bytes_.cpu().numpy().tobytes()
             o = c.offsets.cpu().numpy()
-            col_vals = _np.array(
-                [b[o[i]:o[i + 1]].decode() for i in range(n)], dtype=object)
+            enc = _np.array(
+                [constants.encode_partition_value(
+                    None if (valid is not None and not valid[i])
+                    else b[o[i]:o[i + 1]].decode())
+                 for i in range(n)], dtype=object)
         else:
-            col_vals = c.data.cpu().numpy()
-        uniq, inv = _np.unique(col_vals, return_inverse=True)
-        uniques_per_col.append(uniq)
+            raw = c.data.cpu().numpy()
+            enc = raw.astype(str).astype(object)
+            if valid is not None and not valid.all():
+                enc[~valid] = constants.LAKESOUL_NULL_STRING
+        encoded_cols.append(enc)
+        uniq, inv = _np.unique(enc, return_inverse=True)
         codes = codes * (len(uniq) + 1) + inv
     uniq_codes, inv_codes = _np.unique(codes, return_inverse=True)
     order = _np.argsort(inv_codes, kind="stable")
@@ -184,18 +194,10 @@ def _partition_descs(batch: Batch, range_cols: Sequence[str]):
     for g in range(len(uniq_codes)):
         rows = order[bounds[g]:bounds[g + 1]]
         i0 = int(rows[0])
-        parts = []
-        for k, cname in enumerate(range_cols):
-            c = batch.columns[cname]
-            if c.is_string:
-                b = c.bytes_.cpu().numpy().tobytes()
-                o = c.offsets.cpu().numpy()
-                v = b[o[i0]:o[i0 + 1]].decode()
-            else:
-                v = str(c.data.cpu().numpy()[i0])
-            parts.append(f"{cname}={v}")
+        parts = [f"{cname}={encoded_cols[k][i0]}"
+                 for k, cname in enumerate(range_cols)]
         desc = ",".join(parts)
-        subdir = "/".join(desc.split(","))
+        subdir = "/".join(parts)
         out.append((desc, subdir, torch.from_numpy(_np.ascontiguousarray(rows))))
     return out
 

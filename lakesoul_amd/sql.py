@@ -493,6 +493,8 @@ class _Parser:
             return v
         if k == "id" and v.lower() in ("true", "false"):
             return v.lower() == "true"
+        if k == "kw" and v == "null":
+            return None
         raise SqlError(f"expected literal, got {v!r}")
 
     def predicate(self) -> Expr:
@@ -657,7 +659,19 @@ def execute_sql(catalog, sql: str, device: Optional[str] = None):
     return _execute_select(catalog, payload, device=device)
 
 
+def _is_null_cell(v) -> bool:
+    if v is None:
+        return True
+    try:
+        import math
+
+        return isinstance(v, float) and math.isnan(v)
+    except TypeError:
+        return False
+
+
 def _execute_insert(catalog, ins: dict, device=None):
+    import numpy as _np
     import pandas as pd
 
     t = catalog.table(ins["table"], ins["namespace"])
@@ -684,14 +698,29 @@ def _execute_insert(catalog, ins: dict, device=None):
             from .io.schema import decimal_params
 
             _, sc = decimal_params(f.dtype)
-            df[f.name] = [
-                int(_dec.Decimal(str(v)).scaleb(sc).to_integral_value())
+            vals = [
+                None if _is_null_cell(v)
+                else int(_dec.Decimal(str(v)).scaleb(sc).to_integral_value())
                 for v in df[f.name]
             ]
+            # nullable backing so NULL decimals survive
+            df[f.name] = (pd.array(vals, dtype="Int64")
+                          if any(v is None for v in vals) else vals)
         else:
             from .io.batch import np_dtype_for
 
-            df[f.name] = df[f.name].astype(np_dtype_for(f.dtype))
+            npdt = np_dtype_for(f.dtype)
+            if df[f.name].isna().any():
+                nullable = {"i": "Int64", "u": "UInt64", "b": "boolean"}.get(
+                    _np.dtype(npdt).kind)
+                if nullable and _np.dtype(npdt).itemsize <= 8:
+                    nullable = {"int8": "Int8", "int16": "Int16", "int32": "Int32",
+                                "int64": "Int64", "bool": "boolean",
+                                "uint8": "UInt8"}.get(_np.dtype(npdt).name, nullable)
+                df[f.name] = (df[f.name].astype(nullable) if nullable
+                              else df[f.name].astype(npdt))
+            else:
+                df[f.name] = df[f.name].astype(npdt)
     t.write(df, device=device)
     return pd.DataFrame({"rows_inserted": [len(df)]})
 

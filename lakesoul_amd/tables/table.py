@@ -281,7 +281,7 @@ class LakeSoulTable:
             )
             out_dir = _os.path.join(self.table_path, subdir, constants.COMPACT_DIR)
             _os.makedirs(out_dir, exist_ok=True)
-            fpath = _os.path.join(out_dir, f"part-{random_str(16)}_{unit.bucket_id:04d}.parquet")
+            fpath = _os.path.join(out_dir, f"part-{random_str(16)}_{max(unit.bucket_id, 0):04d}.parquet")
             size = _write_batch_to_file(
                 fpath, kept, cfg.compression, cfg.compression_level, cfg.max_row_group_size
             )

@@ -365,6 +365,31 @@ def test_create_and_drop_table_sql(catalog):
     execute_sql(catalog, "DROP TABLE IF EXISTS sales")  # no error
 
 
+def test_insert_null_values_and_decimal(catalog):
+    """NULL in INSERT VALUES and INSERT..SELECT from a nullable decimal
+    column (Decimal(str(None)) crash + literal() NULL)."""
+    import decimal
+
+    execute_sql(catalog,
+        "CREATE TABLE nv (id BIGINT NOT NULL, amt DECIMAL(10,2), "
+        "qty INT, note VARCHAR(16)) PRIMARY KEY (id) HASH BUCKETS 1")
+    execute_sql(catalog,
+        "INSERT INTO nv VALUES (1, 9.5, 10, 'a'), (2, NULL, NULL, NULL)")
+    df = execute_sql(catalog, "SELECT id, amt, qty, note FROM nv ORDER BY id")
+    assert df["amt"].tolist()[0] == decimal.Decimal("9.50")
+    assert pd.isna(df["amt"].iloc[1])
+    assert pd.isna(df["qty"].iloc[1])
+    assert df["note"].iloc[1] is None or pd.isna(df["note"].iloc[1])
+    # INSERT ... SELECT from the nullable decimal column must not abort
+    execute_sql(catalog,
+        "CREATE TABLE nv2 (id BIGINT NOT NULL, amt DECIMAL(10,2)) "
+        "PRIMARY KEY (id) HASH BUCKETS 1")
+    execute_sql(catalog, "INSERT INTO nv2 SELECT id, amt FROM nv")
+    df2 = execute_sql(catalog, "SELECT id, amt FROM nv2 ORDER BY id")
+    assert df2["amt"].tolist()[0] == decimal.Decimal("9.50")
+    assert pd.isna(df2["amt"].iloc[1])
+
+
 def test_alter_add_column_sql(sql_table):
     cat, t = sql_table
     execute_sql(cat, "ALTER TABLE orders ADD COLUMN note VARCHAR(20), score DOUBLE")
