@@ -72,8 +72,17 @@ def main():
         import torch.distributed as dist_mod
 
         dist = dist_mod
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29411")
         backend = "nccl" if device == "cuda" else "gloo"
-        dist.init_process_group(backend=backend)
+        if device == "cuda":
+            # bind the communicator to this rank's GPU (RCCL over xGMI)
+            dist.init_process_group(
+                backend=backend,
+                device_id=torch.device("cuda", local_rank % max(1, torch.cuda.device_count())),
+            )
+        else:
+            dist.init_process_group(backend=backend)
 
     def barrier():
         if dist is not None:
@@ -179,19 +188,24 @@ def main():
 
     # ---------------- timed: MOR scan steps ---------------- #
     def one_scan() -> int:
+        """Timed step: MOR scan of this rank's buckets; for N>1 each merged
+        batch is re-sharded across ranks by spark-murmur3 of the PK — the
+        real rebalance_by_pk exchange (one packed RCCL all-to-all over
+        xGMI per batch), enqueued async so it overlaps the next unit's
+        decode."""
+        from lakesoul_amd.parallel.shard import rebalance_by_pk_async
+
         scan = table.scan(device=device).shard(rank, world)
         total = 0
+        pending = None
         for batch in scan.iter_batches():
-            n = batch.num_rows
-            total += n
-            if dist is not None and device == "cuda":
-                # RCCL all-to-all shard exchange over xGMI: redistribute
-                # the merged rows round-robin to consumer ranks (variable
-                # per-peer splits handled by the shard exchange)
-                from lakesoul_amd.parallel.shard import exchange_batch_all_to_all
-
-                dest = torch.arange(n, device="cuda") % world
-                exchange_batch_all_to_all(batch, dest)
+            total += batch.num_rows
+            if dist is not None:
+                if pending is not None:
+                    pending.wait()
+                pending = rebalance_by_pk_async(batch, "id")
+        if pending is not None:
+            pending.wait()
         return total
 
     for w in range(args.warmup):

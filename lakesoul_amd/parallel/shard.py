@@ -3,14 +3,24 @@ reference's "each engine task reads its own buckets" distribution
 (SURVEY.md §2.5): decoded batches are redistributed across the node's
 GPUs with all-to-all collectives. xGMI is point-to-point (7 links/GPU),
 so pairwise all-to-all saturates aggregate bandwidth where a ring would
-bottleneck on one link.
+bottleneck on one link — and per-link efficiency wants FEW, LARGE
+collectives: all fixed-width columns (plus validity masks and string
+lengths) are packed into ONE byte buffer and exchanged with a single
+``all_to_all_single``; only string/binary payload bytes need a second
+round (their sizes are only known after the first).
+
+The exchange can run asynchronously (``exchange_batch_all_to_all_async``):
+collectives are enqueued with ``async_op=True`` so the caller can keep
+decoding the next scan unit while RCCL moves bytes over xGMI on its own
+internal streams; ``AsyncExchange.wait()`` completes the string round and
+returns the received Batch.
 
 Works with gloo on CPU for tests (world_size>1, one node).
 """
 
 from __future__ import annotations
 
-from typing import Dict, Optional
+from typing import Dict, List, Optional
 
 import torch
 
@@ -27,72 +37,221 @@ def shard_scan(scan):
     return scan
 
 
+def _col_row_bytes(f, c: Column) -> int:
+    """Bytes per row this field contributes to the packed buffer."""
+    if c.is_string:
+        b = 8  # int64 length
+    else:
+        b = c.data.element_size()
+    if f.nullable:
+        b += 1  # validity byte
+    return b
+
+
+class AsyncExchange:
+    """In-flight all-to-all exchange of one Batch.
+
+    Stage 1 (constructor): row counts exchanged (small, synchronous),
+    packed fixed-width buffer enqueued (async). Stage 2 (wait()): packed
+    buffer completion, string-byte rounds, unpack.
+    """
+
+    def __init__(self, batch: Batch, dest: torch.Tensor, group=None):
+        import torch.distributed as dist
+
+        self._dist = dist
+        self._group = group
+        rank, world = get_rank_world()
+        self.world = world
+        self.batch = batch
+        if world == 1:
+            self._done = batch
+            return
+        self._done = None
+
+        n = batch.num_rows
+        dest = dest.to(torch.int64)
+        order = torch.argsort(dest, stable=True)
+        dest_sorted = dest[order]
+        send_counts = torch.bincount(dest_sorted, minlength=world)
+        recv_counts = torch.empty_like(send_counts)
+        # row-count round: tiny, synchronous (sizes are needed on host to
+        # allocate receive buffers)
+        dist.all_to_all_single(recv_counts, send_counts, group=group)
+        self.send_rows = [int(x) for x in send_counts.cpu()]
+        self.recv_rows = [int(x) for x in recv_counts.cpu()]
+        self.n_recv = sum(self.recv_rows)
+
+        reordered = batch.take(order) if n else batch
+        self.reordered = reordered
+        dev = None
+        for f in batch.schema:
+            c = reordered.columns[f.name]
+            t = c.bytes_ if c.is_string else c.data
+            if t is not None:
+                dev = t.device
+                break
+        self.device = dev if dev is not None else torch.device("cpu")
+
+        # ---- pack: per-dest segment = [colA rows][colB rows]... ---- #
+        fields = list(batch.schema)
+        per_row = {f.name: _col_row_bytes(f, reordered.columns[f.name]) for f in fields}
+        stride = sum(per_row.values())
+        send_sizes_b = [r * stride for r in self.send_rows]
+        recv_sizes_b = [r * stride for r in self.recv_rows]
+        packed = torch.empty(n * stride, dtype=torch.uint8, device=self.device)
+
+        # byte views of every column (and aux arrays for strings)
+        self._lens: Dict[str, torch.Tensor] = {}
+        col_bytes: Dict[str, torch.Tensor] = {}
+        val_bytes: Dict[str, torch.Tensor] = {}
+        for f in fields:
+            c = reordered.columns[f.name]
+            if c.is_string:
+                lens = (c.offsets[1:] - c.offsets[:-1]).to(torch.int64).contiguous()
+                self._lens[f.name] = lens
+                col_bytes[f.name] = (lens.view(torch.uint8).view(n, 8) if n else
+                                     torch.empty(0, 8, dtype=torch.uint8, device=self.device))
+            else:
+                d = c.data.contiguous()
+                isz = d.element_size()
+                col_bytes[f.name] = (d.view(torch.uint8).view(n, isz) if n else
+                                     torch.empty(0, isz, dtype=torch.uint8, device=self.device))
+            if f.nullable:
+                if c.validity is not None:
+                    v = c.validity.contiguous().view(n, 1)
+                else:
+                    v = torch.ones(n, 1, dtype=torch.uint8, device=self.device)
+                val_bytes[f.name] = v
+
+        off_rows = 0
+        woff = 0
+        for d in range(world):
+            r = self.send_rows[d]
+            for f in fields:
+                cb = col_bytes[f.name]
+                nb = r * cb.shape[1]
+                if nb:
+                    packed[woff:woff + nb] = cb[off_rows:off_rows + r].reshape(-1)
+                woff += nb
+                if f.nullable:
+                    if r:
+                        packed[woff:woff + r] = val_bytes[f.name][off_rows:off_rows + r].reshape(-1)
+                    woff += r
+            off_rows += r
+
+        self._stride = stride
+        self._fields = fields
+        self._per_row = per_row
+        self.packed_recv = torch.empty(self.n_recv * stride, dtype=torch.uint8,
+                                       device=self.device)
+        self._work = dist.all_to_all_single(
+            self.packed_recv, packed, recv_sizes_b, send_sizes_b,
+            group=group, async_op=True,
+        )
+        # keep the send buffer alive until the collective completes
+        self._packed_send = packed
+
+    def wait(self) -> Batch:
+        if self._done is not None:
+            return self._done
+        dist = self._dist
+        if self._work is not None:
+            self._work.wait()
+        world, n_recv = self.world, self.n_recv
+        fields = self._fields
+        out_cols: Dict[str, Column] = {}
+
+        # ---- unpack fixed data / validity / string lens ---- #
+        data_out: Dict[str, torch.Tensor] = {}
+        val_out: Dict[str, Optional[torch.Tensor]] = {}
+        for f in fields:
+            c = self.reordered.columns[f.name]
+            isz = 8 if c.is_string else c.data.element_size()
+            data_out[f.name] = torch.empty(n_recv * isz, dtype=torch.uint8, device=self.device)
+            val_out[f.name] = (torch.empty(n_recv, dtype=torch.uint8, device=self.device)
+                               if f.nullable else None)
+        roff = 0
+        rrow = 0
+        for s in range(world):
+            r = self.recv_rows[s]
+            for f in fields:
+                c = self.reordered.columns[f.name]
+                isz = 8 if c.is_string else c.data.element_size()
+                nb = r * isz
+                if nb:
+                    data_out[f.name][rrow * isz: rrow * isz + nb] = self.packed_recv[roff:roff + nb]
+                roff += nb
+                if f.nullable:
+                    if r:
+                        val_out[f.name][rrow:rrow + r] = self.packed_recv[roff:roff + r]
+                    roff += r
+            rrow += r
+
+        # ---- second round: string payload bytes ---- #
+        for f in fields:
+            c = self.reordered.columns[f.name]
+            v = val_out[f.name]
+            if not c.is_string:
+                d = data_out[f.name].view(c.data.dtype)
+                out_cols[f.name] = Column(f.dtype, data=d, validity=v)
+                continue
+            recv_lens = data_out[f.name].view(torch.int64)
+            lens = self._lens[f.name]
+            byte_send, off = [], 0
+            for r in self.send_rows:
+                byte_send.append(int(lens[off:off + r].sum()) if r else 0)
+                off += r
+            byte_recv, off = [], 0
+            for r in self.recv_rows:
+                byte_recv.append(int(recv_lens[off:off + r].sum()) if r else 0)
+                off += r
+            recv_bytes = torch.empty(sum(byte_recv), dtype=torch.uint8, device=self.device)
+            dist.all_to_all_single(
+                recv_bytes, self.reordered.columns[f.name].bytes_.contiguous(),
+                byte_recv, byte_send, group=self._group,
+            )
+            new_offs = torch.zeros(n_recv + 1, dtype=torch.int64, device=self.device)
+            if n_recv:
+                torch.cumsum(recv_lens, 0, out=new_offs[1:].view(-1))
+            out_cols[f.name] = Column(f.dtype, offsets=new_offs, bytes_=recv_bytes,
+                                      validity=v)
+        self._done = Batch(self.batch.schema, out_cols)
+        # release references to in-flight buffers
+        self._packed_send = None
+        self.packed_recv = None
+        return self._done
+
+
+def exchange_batch_all_to_all_async(batch: Batch, dest: torch.Tensor,
+                                    group=None) -> AsyncExchange:
+    """Start an all-to-all row redistribution; returns an AsyncExchange
+    whose .wait() yields the received Batch. The packed fixed-width round
+    is enqueued async so decode of the next unit overlaps the transfer."""
+    return AsyncExchange(batch, dest, group=group)
+
+
 def exchange_batch_all_to_all(batch: Batch, dest: torch.Tensor, group=None) -> Batch:
     """Redistribute rows: row i goes to rank ``dest[i]``. Returns the rows
-    received by this rank (from all peers, peer-major order).
+    received by this rank (from all peers, peer-major order)."""
+    return AsyncExchange(batch, dest, group=group).wait()
 
-    Fixed-width columns exchange as one all_to_all_single each; string
-    columns exchange lengths then bytes.
-    """
-    import torch.distributed as dist
 
-    rank, world = get_rank_world()
-    if world == 1:
-        return batch
-    n = batch.num_rows
-    dest = dest.to(torch.int64)
+def pk_dest_ranks(batch: Batch, pk: str, world: int) -> torch.Tensor:
+    """Destination rank per row = spark-murmur3(pk) % world (bit-exact
+    with the table's bucket hash, utils/hash)."""
+    c = batch.columns[pk]
+    if c.data is not None and c.data.device.type == "cuda":
+        from ..ops import hip
 
-    # sort rows by destination so sends are contiguous
-    order = torch.argsort(dest, stable=True)
-    dest_sorted = dest[order]
-    send_counts = torch.bincount(dest_sorted, minlength=world)
-    recv_counts = torch.empty_like(send_counts)
-    dist.all_to_all_single(recv_counts, send_counts, group=group)
-    send_sizes = [int(x) for x in send_counts]
-    recv_sizes = [int(x) for x in recv_counts]
-    n_recv = sum(recv_sizes)
+        empty_prev = torch.empty(0, dtype=torch.int64, device=c.data.device)
+        empty_v = torch.empty(0, dtype=torch.uint8, device=c.data.device)
+        hashes = hip().hash_fixed_column(c.data, empty_v, empty_prev, True)
+        return hip().bucket_ids(hashes, world).to(torch.int64)
+    from ..utils.murmur3_np import bucket_ids_np, hash_column
 
-    out_cols: Dict[str, Column] = {}
-    reordered = batch.take(order)
-    for f in batch.schema:
-        c = reordered.columns[f.name]
-        if not c.is_string:
-            recv = torch.empty(n_recv, dtype=c.data.dtype, device=c.data.device)
-            dist.all_to_all_single(
-                recv, c.data.contiguous(), recv_sizes, send_sizes, group=group
-            )
-            v = None
-            if c.validity is not None:
-                v = torch.empty(n_recv, dtype=torch.uint8, device=c.validity.device)
-                dist.all_to_all_single(v, c.validity.contiguous(), recv_sizes, send_sizes, group=group)
-            out_cols[f.name] = Column(f.dtype, data=recv, validity=v)
-        else:
-            lens = (c.offsets[1:] - c.offsets[:-1]).to(torch.int64)
-            recv_lens = torch.empty(n_recv, dtype=torch.int64, device=lens.device)
-            dist.all_to_all_single(recv_lens, lens.contiguous(), recv_sizes, send_sizes, group=group)
-            # byte splits: sum of lens per destination segment
-            byte_send = []
-            off = 0
-            for s in send_sizes:
-                byte_send.append(int(lens[off : off + s].sum()))
-                off += s
-            byte_recv = []
-            off = 0
-            for s in recv_sizes:
-                byte_recv.append(int(recv_lens[off : off + s].sum()))
-                off += s
-            recv_bytes = torch.empty(sum(byte_recv), dtype=torch.uint8, device=c.bytes_.device)
-            dist.all_to_all_single(
-                recv_bytes, c.bytes_.contiguous(), byte_recv, byte_send, group=group
-            )
-            new_offs = torch.zeros(n_recv + 1, dtype=torch.int64, device=lens.device)
-            torch.cumsum(recv_lens, 0, out=new_offs[1:].view(-1))
-            v = None
-            if c.validity is not None:
-                v = torch.empty(n_recv, dtype=torch.uint8, device=c.validity.device)
-                dist.all_to_all_single(v, c.validity.contiguous(), recv_sizes, send_sizes, group=group)
-            out_cols[f.name] = Column(f.dtype, offsets=new_offs, bytes_=recv_bytes, validity=v)
-    return Batch(batch.schema, out_cols)
+    h = hash_column(c.data.cpu().numpy(), 42)
+    return torch.from_numpy(bucket_ids_np(h, world).astype("int64"))
 
 
 def rebalance_by_pk(batch: Batch, pk: str, group=None) -> Batch:
@@ -101,17 +260,12 @@ def rebalance_by_pk(batch: Batch, pk: str, group=None) -> Batch:
     rank, world = get_rank_world()
     if world == 1:
         return batch
-    c = batch.columns[pk]
-    if c.data is not None and c.data.device.type == "cuda":
-        from ..ops import hip
+    return exchange_batch_all_to_all(batch, pk_dest_ranks(batch, pk, world), group=group)
 
-        empty_prev = torch.empty(0, dtype=torch.int64, device=c.data.device)
-        empty_v = torch.empty(0, dtype=torch.uint8, device=c.data.device)
-        hashes = hip().hash_fixed_column(c.data, empty_v, empty_prev, True)
-        dest = hip().bucket_ids(hashes, world).to(torch.int64)
-    else:
-        from ..utils.murmur3_np import bucket_ids_np, hash_column
 
-        h = hash_column(c.data.cpu().numpy(), 42)
-        dest = torch.from_numpy(bucket_ids_np(h, world).astype("int64"))
-    return exchange_batch_all_to_all(batch, dest, group=group)
+def rebalance_by_pk_async(batch: Batch, pk: str, group=None) -> AsyncExchange:
+    """Async variant of rebalance_by_pk: returns an AsyncExchange handle."""
+    rank, world = get_rank_world()
+    if world == 1:
+        return AsyncExchange(batch, torch.zeros(batch.num_rows, dtype=torch.int64), group=group)
+    return exchange_batch_all_to_all_async(batch, pk_dest_ranks(batch, pk, world), group=group)

@@ -137,9 +137,85 @@ def _run_rebalance(rank, world, rdv_file, tmpdir, results):
         dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("fn", [_run_exchange, _run_table_shard, _run_rebalance])
-def test_multiprocess_gloo(fn, tmp_path):
-    world = 2
+def _run_exchange_hard(rank, world, rdv_file, tmpdir, results):
+    """The exact collective shapes bench.py uses at world=4/8: nullable
+    columns, strings, validity, skewed destinations (some peers receive
+    nothing), one entirely-empty rank, async overlap path."""
+    dist = _init(rank, world, rdv_file)
+    try:
+        from lakesoul_amd.io.batch import Batch
+        from lakesoul_amd.io.schema import Field, Schema
+        from lakesoul_amd.parallel.shard import (
+            exchange_batch_all_to_all, exchange_batch_all_to_all_async)
+
+        schema = Schema([
+            Field("id", "int64", False),
+            Field("v", "float64", True),
+            Field("k", "int32", False),
+            Field("s", "string", True),
+        ])
+
+        # case 1: skewed — every row goes to rank 0 (peers >0 receive 0)
+        n = 64 if rank != world - 1 else 0  # last rank sends nothing at all
+        ids = np.arange(rank * 1000, rank * 1000 + n, dtype=np.int64)
+        batch = Batch.from_dict({
+            "id": ids,
+            "v": ids.astype(np.float64),
+            "k": (ids % 7).astype(np.int32),
+            "s": [None if i % 5 == 0 else f"r{rank}:{i}" for i in range(n)],
+        }, schema)
+        if batch.columns["v"].validity is None and n:
+            batch.columns["v"].validity = torch.tensor(
+                [0 if i % 3 == 0 else 1 for i in range(n)], dtype=torch.uint8)
+        dest = torch.zeros(n, dtype=torch.int64)
+        out = exchange_batch_all_to_all(batch, dest)
+        t = torch.tensor([out.num_rows])
+        dist.all_reduce(t)
+        expect_total = 64 * (world - 1)
+        assert int(t.item()) == expect_total
+        if rank == 0:
+            assert out.num_rows == expect_total
+            got = out.columns["id"].data.numpy()
+            offs = out.columns["s"].offsets.numpy()
+            bys = out.columns["s"].bytes_.numpy().tobytes()
+            sval = out.columns["s"].validity
+            vval = out.columns["v"].validity.numpy()
+            for i, gid in enumerate(got):
+                src, idx = gid // 1000, gid % 1000
+                if idx % 5 == 0:
+                    assert sval is None or sval.numpy()[i] == 0
+                else:
+                    assert bys[offs[i]:offs[i + 1]].decode() == f"r{src}:{idx}"
+                assert vval[i] == (0 if idx % 3 == 0 else 1)
+        else:
+            assert out.num_rows == 0
+
+        # case 2: async murmur exchange, uniform scatter
+        n2 = 200
+        ids2 = np.arange(rank * 10_000, rank * 10_000 + n2, dtype=np.int64)
+        b2 = Batch.from_dict({
+            "id": ids2, "v": ids2.astype(np.float64),
+            "k": np.zeros(n2, np.int32), "s": [f"x{i}" for i in ids2],
+        }, schema)
+        dest2 = torch.from_numpy((ids2 % world).astype(np.int64))
+        h = exchange_batch_all_to_all_async(b2, dest2)
+        # caller could decode the next unit here; then:
+        out2 = h.wait()
+        got2 = out2.columns["id"].data.numpy()
+        assert (got2 % world == rank).all()
+        offs2 = out2.columns["s"].offsets.numpy()
+        bys2 = out2.columns["s"].bytes_.numpy().tobytes()
+        for i, gid in enumerate(got2):
+            assert bys2[offs2[i]:offs2[i + 1]].decode() == f"x{gid}"
+        t2 = torch.tensor([out2.num_rows])
+        dist.all_reduce(t2)
+        assert int(t2.item()) == world * n2
+        results[rank] = "ok"
+    finally:
+        dist.destroy_process_group()
+
+
+def _spawn(fn, world, tmp_path):
     port = str(tmp_path / "rdv")
     ctx = mp.get_context("spawn")
     with ctx.Manager() as mgr:
@@ -151,7 +227,24 @@ def test_multiprocess_gloo(fn, tmp_path):
         for p in procs:
             p.start()
         for p in procs:
-            p.join(timeout=180)
+            p.join(timeout=300)
         for p in procs:
             assert p.exitcode == 0, f"worker failed (exit {p.exitcode})"
-        assert results.get(0) == "ok" and results.get(1) == "ok"
+        for r in range(world):
+            assert results.get(r) == "ok", f"rank {r} did not finish"
+
+
+@pytest.mark.parametrize("fn", [_run_exchange, _run_table_shard, _run_rebalance])
+def test_multiprocess_gloo(fn, tmp_path):
+    _spawn(fn, 2, tmp_path)
+
+
+@pytest.mark.parametrize("world", [4, 8])
+def test_exchange_collectives_world_n(world, tmp_path):
+    """gloo matrix for the exact bench collectives at world=4 and 8."""
+    _spawn(_run_exchange_hard, world, tmp_path)
+
+
+@pytest.mark.parametrize("world", [4])
+def test_rebalance_world4(world, tmp_path):
+    _spawn(_run_rebalance, world, tmp_path)
