@@ -321,11 +321,17 @@ def _parse_literal(s: str, dtype: str):
 
 
 def resolve_filters(filters, schema: Schema) -> Optional[Expr]:
-    """Accept None | Expr | DSL string | list of (col,op,val) tuples."""
+    """Accept None | Expr | DSL string | list of (col,op,val) tuples |
+    Substrait protobuf bytes (Plan or ExtendedExpression — the encodings
+    engines push through the C ABI, reference parser.rs:44-60)."""
     if filters is None:
         return None
     if isinstance(filters, Expr):
         return filters
+    if isinstance(filters, (bytes, bytearray, memoryview)):
+        from .substrait import decode_substrait_filter
+
+        return decode_substrait_filter(bytes(filters), schema)
     if isinstance(filters, str):
         return parse_filter_dsl(filters, schema)
     if isinstance(filters, (list, tuple)):
