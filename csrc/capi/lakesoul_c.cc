@@ -1,26 +1,34 @@
 // lakesoul_amd C ABI implementation — torch-free; links only libzstd.
 // Build: g++ -O3 -std=c++17 -shared -fPIC lakesoul_c.cc -o
 //        liblakesoul_amd_c.so -l:libzstd.so.1 -pthread
+//
+// Functional parity with the reference's rust/lakesoul-io-c
+// (lib.rs:113-1324): config option map, merge operators, DSL + Substrait
+// filter pushdown, CDC rows, async callback reads, FlushResult.
 #include "lakesoul_c.h"
 
 #include <algorithm>
 #include <cstring>
+#include <map>
 #include <memory>
 #include <string>
+#include <thread>
 #include <vector>
 
 #include "../cpp/murmur3.h"
 #include "../cpp/parquet_file.h"
+#include "capi_engine.h"
 
 using namespace lakesoul;
+using namespace lakesoul_capi;
 
 static thread_local std::string g_err;
 
 extern "C" const char* lakesoul_c_last_error(void) { return g_err.c_str(); }
 
-#define C_TRY(body)                 \
+#define C_TRY(...)                  \
   try {                             \
-    body                            \
+    __VA_ARGS__                     \
   } catch (std::exception & e) {    \
     g_err = e.what();               \
     return -1;                      \
@@ -95,7 +103,93 @@ std::string dtype_to_arrow_fmt(const ColumnDesc& c) {
   }
 }
 
+int arrow_fmt_elem_size(const std::string& f) {
+  if (f == "l" || f == "g" || f.rfind("ts", 0) == 0) return 8;
+  if (f == "s") return 2;
+  if (f == "c") return 1;
+  return 4;  // i, f, tdD
+}
+
 }  // namespace
+
+// ===================================================================== //
+// config builder
+// ===================================================================== //
+
+struct LakesoulCConfig {
+  std::vector<std::string> files;
+  std::vector<std::string> columns;
+  std::vector<std::string> pks;
+  std::map<std::string, std::string> merge_ops;
+  std::vector<std::string> filters_dsl;
+  std::vector<uint8_t> filter_substrait;
+  std::map<std::string, std::string> options;
+  mutable std::string opt_scratch;
+
+  std::string get(const std::string& k, const std::string& dflt = "") const {
+    auto it = options.find(k);
+    if (it != options.end()) return it->second;
+    std::string env = "LAKESOUL_";
+    for (char c : k) env += (char)toupper((unsigned char)c);
+    const char* e = getenv(env.c_str());
+    return e ? std::string(e) : dflt;
+  }
+  bool has(const std::string& k) const {
+    if (options.count(k)) return true;
+    std::string env = "LAKESOUL_";
+    for (char c : k) env += (char)toupper((unsigned char)c);
+    return getenv(env.c_str()) != nullptr;
+  }
+};
+
+extern "C" LakesoulCConfig* lakesoul_c_config_create(void) {
+  return new LakesoulCConfig();
+}
+extern "C" int lakesoul_c_config_add_file(LakesoulCConfig* c, const char* p) {
+  c->files.push_back(p);
+  return 0;
+}
+extern "C" int lakesoul_c_config_add_column(LakesoulCConfig* c, const char* n) {
+  c->columns.push_back(n);
+  return 0;
+}
+extern "C" int lakesoul_c_config_add_primary_key(LakesoulCConfig* c, const char* n) {
+  c->pks.push_back(n);
+  return 0;
+}
+extern "C" int lakesoul_c_config_add_merge_op(LakesoulCConfig* c, const char* col,
+                                              const char* op) {
+  C_TRY({
+    parse_merge_op(op);  // validate
+    c->merge_ops[col] = op;
+    return 0;
+  })
+}
+extern "C" int lakesoul_c_config_add_filter(LakesoulCConfig* c, const char* dsl) {
+  C_TRY({
+    parse_dsl(dsl);  // validate eagerly so errors surface at config time
+    c->filters_dsl.push_back(dsl);
+    return 0;
+  })
+}
+extern "C" int lakesoul_c_config_set_filter_substrait(LakesoulCConfig* c,
+                                                      const uint8_t* buf,
+                                                      int64_t len) {
+  c->filter_substrait.assign(buf, buf + len);
+  return 0;
+}
+extern "C" int lakesoul_c_config_set_option(LakesoulCConfig* c, const char* k,
+                                            const char* v) {
+  c->options[k] = v;
+  return 0;
+}
+extern "C" const char* lakesoul_c_config_get_option(LakesoulCConfig* c,
+                                                    const char* k) {
+  if (!c->has(k)) return nullptr;
+  c->opt_scratch = c->get(k);
+  return c->opt_scratch.c_str();
+}
+extern "C" void lakesoul_c_config_free(LakesoulCConfig* c) { delete c; }
 
 // ===================================================================== //
 // reader
@@ -105,6 +199,11 @@ struct LakesoulCReader {
   std::vector<std::string> files;
   std::vector<std::string> columns;
   std::vector<std::string> pks;
+  std::map<std::string, std::string> merge_ops;
+  std::vector<std::string> filters_dsl;
+  std::vector<uint8_t> filter_substrait;
+  std::string cdc_column;
+  bool skip_merge = false;
   int64_t batch_size = 8192;
 
   // merged result
@@ -114,10 +213,32 @@ struct LakesoulCReader {
   int64_t total_rows = 0;
   int64_t cursor = 0;
   bool started = false;
+  std::thread async_thread;  // one in-flight async next
+
+  ~LakesoulCReader() {
+    if (async_thread.joinable()) async_thread.join();
+  }
 };
 
 extern "C" LakesoulCReader* lakesoul_c_reader_create(void) {
   return new LakesoulCReader();
+}
+
+extern "C" LakesoulCReader* lakesoul_c_reader_create_from_config(
+    const LakesoulCConfig* c) {
+  auto* r = new LakesoulCReader();
+  r->files = c->files;
+  r->columns = c->columns;
+  r->pks = c->pks;
+  r->merge_ops = c->merge_ops;
+  r->filters_dsl = c->filters_dsl;
+  r->filter_substrait = c->filter_substrait;
+  r->cdc_column = c->get("cdc_column");
+  std::string skip = c->get("skip_merge_on_read");
+  r->skip_merge = (skip == "1" || skip == "true");
+  std::string bs = c->get("batch_size");
+  if (!bs.empty()) r->batch_size = std::stoll(bs);
+  return r;
 }
 
 extern "C" int lakesoul_c_reader_add_file(LakesoulCReader* r, const char* p) {
@@ -140,12 +261,11 @@ extern "C" int lakesoul_c_reader_set_batch_size(LakesoulCReader* r, int64_t b) {
 namespace {
 
 struct FileCols {
-  std::vector<DecodedColumn> cols;   // per requested column
+  std::vector<DecodedColumn> cols;   // per read column
   std::vector<ColumnDesc> descs;
   int64_t rows = 0;
 };
 
-// key for merge ordering: (pk..., seq, row). integer pks only.
 struct RowRef {
   int64_t seq;
   int64_t row;
@@ -176,16 +296,224 @@ int cmp_value(const DecodedColumn& da, const ColumnDesc& ca, int64_t ra,
   if (ca.physical == PT_BYTE_ARRAY && cb.physical == PT_BYTE_ARRAY) {
     int32_t a0 = da.offsets[(size_t)ra], a1 = da.offsets[(size_t)ra + 1];
     int32_t b0 = db.offsets[(size_t)rb], b1 = db.offsets[(size_t)rb + 1];
-    size_t la = (size_t)(a1 - a0), lb = (size_t)(b1 - b0);
-    int c = std::memcmp(da.bytes.data() + a0, db.bytes.data() + b0,
-                        la < lb ? la : lb);
-    if (c != 0) return c < 0 ? -1 : 1;
-    if (la != lb) return la < lb ? -1 : 1;
-    return 0;
+    return cmp3_str(da.bytes.data() + a0, (size_t)(a1 - a0),
+                    db.bytes.data() + b0, (size_t)(b1 - b0));
   }
   int64_t va = int_value(da, ca, ra);
   int64_t vb = int_value(db, cb, rb);
   return va < vb ? -1 : (va > vb ? 1 : 0);
+}
+
+bool ref_valid(const DecodedColumn& dc, int64_t row) {
+  return dc.validity.empty() || dc.validity[(size_t)row] != 0;
+}
+
+void append_value(WorkCol& out, const DecodedColumn& dc, const ColumnDesc& cd,
+                  int64_t row, bool valid) {
+  if (cd.physical == PT_BYTE_ARRAY) {
+    if (valid) {
+      int32_t a = dc.offsets[(size_t)row], b = dc.offsets[(size_t)row + 1];
+      out.bytes.insert(out.bytes.end(), dc.bytes.data() + a, dc.bytes.data() + b);
+    }
+    out.offsets.push_back((int64_t)out.bytes.size());
+  } else {
+    int es = physical_elem_size(cd.physical);
+    if (valid) {
+      const uint8_t* src = dc.data.data() + row * es;
+      out.data.insert(out.data.end(), src, src + es);
+    } else {
+      out.data.insert(out.data.end(), (size_t)es, 0);
+    }
+  }
+  out.validity.push_back(valid ? 1 : 0);
+  out.n++;
+}
+
+void append_null(WorkCol& out, const ColumnDesc& cd) {
+  if (cd.physical == PT_BYTE_ARRAY) {
+    out.offsets.push_back(out.offsets.empty() ? 0 : out.offsets.back());
+  } else {
+    out.data.insert(out.data.end(), (size_t)physical_elem_size(cd.physical), 0);
+  }
+  out.validity.push_back(0);
+  out.n++;
+}
+
+void append_string(WorkCol& out, const std::string& s, bool valid) {
+  if (valid) out.bytes.insert(out.bytes.end(), s.begin(), s.end());
+  out.offsets.push_back((int64_t)out.bytes.size());
+  out.validity.push_back(valid ? 1 : 0);
+  out.n++;
+}
+
+// apply one merge op over one group [a, e) of sorted refs
+void apply_op_group(WorkCol& out, MergeOp op,
+                    const std::vector<FileCols>& fcs, size_t k,
+                    const std::vector<RowRef>& refs, size_t a, size_t e) {
+  const ColumnDesc& cd = fcs[refs[a].seq].descs[k];
+  switch (op) {
+    case OP_USE_LAST: {
+      const RowRef& r = refs[e - 1];
+      const DecodedColumn& dc = fcs[r.seq].cols[k];
+      append_value(out, dc, cd, r.row, ref_valid(dc, r.row));
+      return;
+    }
+    case OP_USE_LAST_NOT_NULL: {
+      for (size_t i = e; i-- > a;) {
+        const RowRef& r = refs[i];
+        const DecodedColumn& dc = fcs[r.seq].cols[k];
+        if (ref_valid(dc, r.row)) {
+          append_value(out, dc, cd, r.row, true);
+          return;
+        }
+      }
+      append_null(out, cd);
+      return;
+    }
+    case OP_SUM_ALL:
+    case OP_SUM_LAST: {
+      // SumLast: last row per file (stream) in the group
+      bool is_float = cd.physical == PT_FLOAT || cd.physical == PT_DOUBLE;
+      double fs = 0;
+      int64_t is = 0;
+      bool any_null = false;
+      for (size_t i = a; i < e; i++) {
+        if (op == OP_SUM_LAST) {
+          bool last_of_stream = (i + 1 == e) || (refs[i + 1].seq != refs[i].seq);
+          if (!last_of_stream) continue;
+        }
+        const RowRef& r = refs[i];
+        const DecodedColumn& dc = fcs[r.seq].cols[k];
+        if (!ref_valid(dc, r.row)) {
+          any_null = true;  // null-if-any-null (reference macro behavior)
+          continue;
+        }
+        if (is_float) {
+          if (cd.physical == PT_DOUBLE) { double v; std::memcpy(&v, dc.data.data() + r.row * 8, 8); fs += v; }
+          else { float v; std::memcpy(&v, dc.data.data() + r.row * 4, 4); fs += v; }
+        } else {
+          is += int_value(dc, cd, r.row);
+        }
+      }
+      if (any_null) {
+        append_null(out, cd);
+        return;
+      }
+      if (cd.physical == PT_DOUBLE) {
+        out.data.insert(out.data.end(), (uint8_t*)&fs, (uint8_t*)&fs + 8);
+      } else if (cd.physical == PT_FLOAT) {
+        float v = (float)fs;
+        out.data.insert(out.data.end(), (uint8_t*)&v, (uint8_t*)&v + 4);
+      } else if (cd.physical == PT_INT64) {
+        out.data.insert(out.data.end(), (uint8_t*)&is, (uint8_t*)&is + 8);
+      } else {
+        int32_t v = (int32_t)is;
+        out.data.insert(out.data.end(), (uint8_t*)&v, (uint8_t*)&v + 4);
+      }
+      out.validity.push_back(1);
+      out.n++;
+      return;
+    }
+    case OP_JOIN_ALL_COMMA:
+    case OP_JOIN_ALL_SEMI:
+    case OP_JOIN_LAST_COMMA:
+    case OP_JOIN_LAST_SEMI: {
+      if (cd.physical != PT_BYTE_ARRAY)
+        throw std::runtime_error("JoinedBy* merge ops require a string column");
+      char delim = (op == OP_JOIN_ALL_COMMA || op == OP_JOIN_LAST_COMMA) ? ',' : ';';
+      bool use_all = (op == OP_JOIN_ALL_COMMA || op == OP_JOIN_ALL_SEMI);
+      std::string acc;
+      bool first_part = true;
+      bool is_null = false;
+      for (size_t i = a; i < e; i++) {
+        if (!use_all) {
+          bool last_of_stream = (i + 1 == e) || (refs[i + 1].seq != refs[i].seq);
+          if (!last_of_stream) continue;
+        }
+        const RowRef& r = refs[i];
+        const DecodedColumn& dc = fcs[r.seq].cols[k];
+        if (!ref_valid(dc, r.row)) {
+          is_null = true;
+          break;
+        }
+        int32_t s0 = dc.offsets[(size_t)r.row], s1 = dc.offsets[(size_t)r.row + 1];
+        if (!first_part) acc += delim;
+        acc.append((const char*)dc.bytes.data() + s0, (size_t)(s1 - s0));
+        first_part = false;
+      }
+      append_string(out, acc, !is_null);
+      return;
+    }
+  }
+  throw std::runtime_error("unhandled merge op");
+}
+
+// WorkCol -> ExportedColumn (bit-pack validity/bools, int32 offsets,
+// narrow int8/int16 to their arrow widths)
+ExportedColumn export_workcol(const WorkCol& w, const std::string& name) {
+  ExportedColumn ec;
+  ec.name = name;
+  ec.format = dtype_to_arrow_fmt(w.desc);
+  ec.length = w.n;
+  bool any_null = false;
+  for (auto v : w.validity)
+    if (!v) { any_null = true; break; }
+  if (any_null) {
+    ec.validity_bits.assign(((size_t)w.n + 7) / 8, 0);
+    for (int64_t i = 0; i < w.n; i++) {
+      if (w.validity[(size_t)i])
+        ec.validity_bits[i >> 3] |= (uint8_t)(1 << (i & 7));
+      else
+        ec.null_count++;
+    }
+  }
+  if (w.is_string()) {
+    ec.offsets.reserve((size_t)w.n + 1);
+    ec.offsets.push_back(0);
+    for (int64_t i = 1; i <= w.n; i++) ec.offsets.push_back((int32_t)w.offsets[(size_t)i]);
+    ec.bytes = w.bytes;
+    return ec;
+  }
+  if (w.desc.physical == PT_BOOLEAN) {
+    ec.data.assign(((size_t)w.n + 7) / 8, 0);
+    for (int64_t i = 0; i < w.n; i++)
+      if (w.data[(size_t)i]) ec.data[i >> 3] |= (uint8_t)(1 << (i & 7));
+    return ec;
+  }
+  int src_es = physical_elem_size(w.desc.physical);
+  int dst_es = arrow_fmt_elem_size(ec.format);
+  if (src_es == dst_es) {
+    ec.data = w.data;
+  } else {
+    // narrow int8/int16 (stored as 4-byte physical INT32)
+    ec.data.resize((size_t)w.n * dst_es);
+    for (int64_t i = 0; i < w.n; i++)
+      std::memcpy(ec.data.data() + i * dst_es, w.data.data() + i * src_es, dst_es);
+  }
+  return ec;
+}
+
+WorkCol take_workcol(const WorkCol& w, const std::vector<int64_t>& idx) {
+  WorkCol out;
+  out.desc = w.desc;
+  if (w.is_string()) out.offsets.push_back(0);
+  for (int64_t i : idx) {
+    bool valid = w.valid(i);
+    if (w.is_string()) {
+      if (valid) {
+        auto [p, l] = w.as_str(i);
+        out.bytes.insert(out.bytes.end(), p, p + l);
+      }
+      out.offsets.push_back((int64_t)out.bytes.size());
+    } else {
+      int es = physical_elem_size(w.desc.physical);
+      out.data.insert(out.data.end(), w.data.data() + i * es,
+                      w.data.data() + (i + 1) * es);
+    }
+    out.validity.push_back(valid ? 1 : 0);
+    out.n++;
+  }
+  return out;
 }
 
 }  // namespace
@@ -193,19 +521,57 @@ int cmp_value(const DecodedColumn& da, const ColumnDesc& ca, int64_t ra,
 extern "C" int lakesoul_c_reader_start(LakesoulCReader* r) {
   C_TRY({
     if (r->files.empty()) throw std::runtime_error("no files configured");
-    std::vector<FileCols> fcs(r->files.size());
+
+    // parse filters up front
+    FilterPtr filter;
+    for (auto& dsl : r->filters_dsl) {
+      FilterPtr f = parse_dsl(dsl);
+      if (!filter) {
+        filter = std::move(f);
+      } else {
+        auto p = std::make_unique<FilterExpr>();
+        p->k = FilterExpr::AND_;
+        p->kids.push_back(std::move(filter));
+        p->kids.push_back(std::move(f));
+        filter = std::move(p);
+      }
+    }
+
     // resolve requested columns from the first file when unspecified
     {
       ParquetFile f0(r->files[0]);
       if (r->columns.empty())
         for (auto& c : f0.columns()) r->columns.push_back(c.name);
     }
-    // pk columns must be read
+
+    if (!r->filter_substrait.empty()) {
+      FilterPtr f = sub::decode_filter(r->filter_substrait.data(),
+                                       r->filter_substrait.size(), r->columns);
+      if (!filter) {
+        filter = std::move(f);
+      } else {
+        auto p = std::make_unique<FilterExpr>();
+        p->k = FilterExpr::AND_;
+        p->kids.push_back(std::move(filter));
+        p->kids.push_back(std::move(f));
+        filter = std::move(p);
+      }
+    }
+
+    std::vector<std::string> filter_cols;
+    if (filter) filter_columns(*filter, filter_cols);
+
+    // read set: pks + requested + filter cols + cdc col
     std::vector<std::string> read_cols = r->pks;
-    for (auto& c : r->columns)
+    auto add_col = [&](const std::string& c) {
       if (std::find(read_cols.begin(), read_cols.end(), c) == read_cols.end())
         read_cols.push_back(c);
+    };
+    for (auto& c : r->columns) add_col(c);
+    for (auto& c : filter_cols) add_col(c);
+    if (!r->cdc_column.empty()) add_col(r->cdc_column);
 
+    std::vector<FileCols> fcs(r->files.size());
     for (size_t fi = 0; fi < r->files.size(); fi++) {
       ParquetFile f(r->files[fi]);
       FileCols& fc = fcs[fi];
@@ -218,7 +584,6 @@ extern "C" int lakesoul_c_reader_start(LakesoulCReader* r) {
         for (size_t rg = 0; rg < f.num_row_groups(); rg++) {
           auto ch = f.read_chunk(rg, ci);
           DecodedColumn part = decode_chunk_cpu(ch);
-          // append
           if (dc.num_values == 0) {
             dc = std::move(part);
           } else {
@@ -244,7 +609,6 @@ extern "C" int lakesoul_c_reader_start(LakesoulCReader* r) {
       }
     }
 
-    // order refs
     std::vector<RowRef> refs;
     for (size_t fi = 0; fi < fcs.size(); fi++)
       for (int64_t i = 0; i < fcs[fi].rows; i++) refs.push_back({(int64_t)fi, i});
@@ -254,7 +618,17 @@ extern "C" int lakesoul_c_reader_start(LakesoulCReader* r) {
       auto it = std::find(read_cols.begin(), read_cols.end(), p);
       pk_idx.push_back((size_t)(it - read_cols.begin()));
     }
-    if (!r->pks.empty()) {
+
+    bool do_merge = !r->pks.empty() && !r->skip_merge;
+
+    // column index in read_cols
+    auto col_idx = [&](const std::string& n) -> size_t {
+      auto it = std::find(read_cols.begin(), read_cols.end(), n);
+      return (size_t)(it - read_cols.begin());
+    };
+
+    std::map<std::string, WorkCol> merged;
+    if (do_merge) {
       std::stable_sort(refs.begin(), refs.end(), [&](const RowRef& a, const RowRef& b) {
         for (size_t k : pk_idx) {
           int c = cmp_value(fcs[a.seq].cols[k], fcs[a.seq].descs[k], a.row,
@@ -264,71 +638,89 @@ extern "C" int lakesoul_c_reader_start(LakesoulCReader* r) {
         if (a.seq != b.seq) return a.seq < b.seq;
         return a.row < b.row;
       });
-      // dedup keep-last (UseLast)
-      std::vector<RowRef> dedup;
+      // group boundaries
+      std::vector<size_t> starts;
       for (size_t i = 0; i < refs.size(); i++) {
-        bool last = (i + 1 == refs.size());
-        if (!last) {
-          bool same = true;
-          for (size_t k : pk_idx) {
-            if (cmp_value(fcs[refs[i].seq].cols[k], fcs[refs[i].seq].descs[k],
-                          refs[i].row,
-                          fcs[refs[i + 1].seq].cols[k],
-                          fcs[refs[i + 1].seq].descs[k], refs[i + 1].row) != 0) {
-              same = false;
-              break;
-            }
-          }
-          if (same) continue;  // a newer row with the same PK follows
+        if (i == 0) {
+          starts.push_back(0);
+          continue;
         }
-        dedup.push_back(refs[i]);
+        for (size_t k : pk_idx) {
+          if (cmp_value(fcs[refs[i - 1].seq].cols[k], fcs[refs[i - 1].seq].descs[k],
+                        refs[i - 1].row, fcs[refs[i].seq].cols[k],
+                        fcs[refs[i].seq].descs[k], refs[i].row) != 0) {
+            starts.push_back(i);
+            break;
+          }
+        }
       }
-      refs = std::move(dedup);
+      starts.push_back(refs.size());
+
+      for (auto& name : read_cols) {
+        size_t k = col_idx(name);
+        MergeOp op = OP_USE_LAST;
+        auto mo = r->merge_ops.find(name);
+        bool is_pk = std::find(r->pks.begin(), r->pks.end(), name) != r->pks.end();
+        if (!is_pk && mo != r->merge_ops.end()) op = parse_merge_op(mo->second);
+        WorkCol w;
+        w.desc = fcs[0].descs[k];
+        if (w.is_string()) w.offsets.push_back(0);
+        for (size_t g = 0; g + 1 < starts.size(); g++)
+          apply_op_group(w, op, fcs, k, refs, starts[g], starts[g + 1]);
+        merged.emplace(name, std::move(w));
+      }
+    } else {
+      // pass-through concat in file order
+      for (auto& name : read_cols) {
+        size_t k = col_idx(name);
+        WorkCol w;
+        w.desc = fcs[0].descs[k];
+        if (w.is_string()) w.offsets.push_back(0);
+        for (auto& ref : refs) {
+          const DecodedColumn& dc = fcs[ref.seq].cols[k];
+          append_value(w, dc, fcs[ref.seq].descs[k], ref.row, ref_valid(dc, ref.row));
+        }
+        merged.emplace(name, std::move(w));
+      }
     }
 
-    // gather requested columns in ref order
-    for (auto& name : r->columns) {
-      auto it = std::find(read_cols.begin(), read_cols.end(), name);
-      size_t k = (size_t)(it - read_cols.begin());
-      const ColumnDesc& cd = fcs[0].descs[k];
-      ExportedColumn ec;
-      ec.name = name;
-      ec.format = dtype_to_arrow_fmt(cd);
-      ec.length = (int64_t)refs.size();
-      bool is_str = cd.physical == PT_BYTE_ARRAY;
-      bool is_bool = cd.physical == PT_BOOLEAN;
-      int es = physical_elem_size(cd.physical);
-      bool any_null = false;
-      for (auto& fc : fcs)
-        if (!fc.cols[k].validity.empty()) any_null = true;
-      if (any_null) ec.validity_bits.assign((refs.size() + 7) / 8, 0);
-      if (is_str) ec.offsets.push_back(0);
-      if (is_bool) ec.data.assign((refs.size() + 7) / 8, 0);
+    int64_t n_merged = merged.empty() ? 0 : merged.begin()->second.n;
 
-      for (size_t i = 0; i < refs.size(); i++) {
-        const DecodedColumn& dc = fcs[refs[i].seq].cols[k];
-        int64_t row = refs[i].row;
-        bool valid = dc.validity.empty() || dc.validity[row];
-        if (any_null) {
-          if (valid) ec.validity_bits[i >> 3] |= (uint8_t)(1 << (i & 7));
-          else ec.null_count++;
-        }
-        if (is_str) {
-          if (valid) {
-            ec.bytes.insert(ec.bytes.end(), dc.bytes.data() + dc.offsets[row],
-                            dc.bytes.data() + dc.offsets[row + 1]);
-          }
-          ec.offsets.push_back((int32_t)ec.bytes.size());
-        } else if (is_bool) {
-          if (valid && dc.data[row]) ec.data[i >> 3] |= (uint8_t)(1 << (i & 7));
-        } else {
-          const uint8_t* src = dc.data.data() + row * es;
-          ec.data.insert(ec.data.end(), src, src + es);
-        }
+    // row selection: CDC delete rows drop, then filter
+    std::vector<int64_t> keep;
+    bool need_select = false;
+    {
+      std::map<std::string, const WorkCol*> view;
+      for (auto& kv : merged) view[kv.first] = &kv.second;
+      const WorkCol* cdc = nullptr;
+      if (!r->cdc_column.empty()) {
+        auto it = merged.find(r->cdc_column);
+        if (it != merged.end()) cdc = &it->second;
       }
-      r->result.push_back(std::move(ec));
+      for (int64_t i = 0; i < n_merged; i++) {
+        if (cdc && cdc->valid(i)) {
+          auto [p, l] = cdc->as_str(i);
+          if (l == 6 && std::memcmp(p, "delete", 6) == 0) {
+            need_select = true;
+            continue;
+          }
+        }
+        if (filter && !filter_eval(*filter, view, i)) {
+          need_select = true;
+          continue;
+        }
+        keep.push_back(i);
+      }
+    }
+
+    for (auto& name : r->columns) {
+      auto& w = merged.at(name);
+      ExportedColumn ec = need_select
+                              ? export_workcol(take_workcol(w, keep), name)
+                              : export_workcol(w, name);
       r->out_names.push_back(name);
-      r->out_formats.push_back(dtype_to_arrow_fmt(cd));
+      r->out_formats.push_back(ec.format);
+      r->result.push_back(std::move(ec));
     }
     r->total_rows = r->result.empty() ? 0 : r->result[0].length;
     r->cursor = 0;
@@ -344,7 +736,6 @@ extern "C" int lakesoul_c_reader_schema(LakesoulCReader* r, struct ArrowSchema* 
     size_t n = r->out_names.size();
     priv->child_schemas.resize(n);
     priv->names = r->out_names;
-    static std::vector<std::string>* fmt_keep = nullptr;  // formats live in priv
     for (size_t i = 0; i < n; i++) {
       auto* cs = new ArrowSchema();
       std::memset(cs, 0, sizeof(ArrowSchema));
@@ -358,7 +749,6 @@ extern "C" int lakesoul_c_reader_schema(LakesoulCReader* r, struct ArrowSchema* 
       cs->release = release_child_schema;
       priv->child_schemas[i] = cs;
     }
-    (void)fmt_keep;
     std::memset(out, 0, sizeof(ArrowSchema));
     out->format = "+s";
     out->name = "";
@@ -410,11 +800,7 @@ extern "C" int lakesoul_c_reader_next(LakesoulCReader* r, struct ArrowArray* out
           if (full.data[(a + j) >> 3] & (1 << ((a + j) & 7)))
             ec->data[j >> 3] |= (uint8_t)(1 << (j & 7));
       } else {
-        int es = (full.format == "l" || full.format == "g" ||
-                  full.format.rfind("ts", 0) == 0) ? 8
-                 : (full.format == "s") ? 2
-                 : (full.format == "c") ? 1
-                 : 4;
+        int es = arrow_fmt_elem_size(full.format);
         ec->data.assign(full.data.begin() + a * es, full.data.begin() + (a + n) * es);
       }
 
@@ -450,6 +836,20 @@ extern "C" int lakesoul_c_reader_next(LakesoulCReader* r, struct ArrowArray* out
   })
 }
 
+extern "C" int lakesoul_c_reader_next_async(
+    LakesoulCReader* r, struct ArrowArray* out,
+    void (*cb)(void* user, int rc, const char* err), void* user) {
+  C_TRY({
+    if (!r->started) throw std::runtime_error("reader not started");
+    if (r->async_thread.joinable()) r->async_thread.join();
+    r->async_thread = std::thread([r, out, cb, user]() {
+      int rc = lakesoul_c_reader_next(r, out);
+      cb(user, rc, rc < 0 ? g_err.c_str() : nullptr);
+    });
+    return 0;
+  })
+}
+
 extern "C" void lakesoul_c_reader_close(LakesoulCReader* r) { delete r; }
 
 // ===================================================================== //
@@ -462,15 +862,28 @@ struct LakesoulCWriter {
   int level = 1;
   int64_t row_group = 250000;
   std::vector<ColumnDesc> descs;
-  // accumulated rows (row-major append of column buffers)
-  std::vector<ExportedColumn> acc;  // reuse as column accumulators (byte-validity in validity_bits)
+  std::vector<ExportedColumn> acc;
   std::unique_ptr<ParquetWriter> w;
   bool open = false;
+  int64_t rows_written = 0;
 };
 
 extern "C" LakesoulCWriter* lakesoul_c_writer_create(const char* path) {
   auto* w = new LakesoulCWriter();
   w->path = path;
+  return w;
+}
+
+extern "C" LakesoulCWriter* lakesoul_c_writer_create_from_config(
+    const LakesoulCConfig* c, const char* path) {
+  auto* w = new LakesoulCWriter();
+  w->path = path;
+  std::string codec = c->get("compression", "zstd");
+  if (codec == "none" || codec == "uncompressed") w->codec = CODEC_UNCOMPRESSED;
+  std::string lvl = c->get("compression_level");
+  if (!lvl.empty()) w->level = std::stoi(lvl);
+  std::string rg = c->get("max_row_group_size");
+  if (!rg.empty()) w->row_group = std::stoll(rg);
   return w;
 }
 
@@ -556,6 +969,7 @@ extern "C" int lakesoul_c_writer_write(LakesoulCWriter* w, struct ArrowArray* ba
       }
     }
     w->w->write_row_group(data, n);
+    w->rows_written += n;
     return 0;
   })
 }
@@ -569,6 +983,40 @@ extern "C" int64_t lakesoul_c_writer_close(LakesoulCWriter* w) {
   }
   delete w;
   return size;
+}
+
+extern "C" int lakesoul_c_writer_flush(LakesoulCWriter* w,
+                                       LakesoulCFlushResult* out) {
+  std::memset(out, 0, sizeof(LakesoulCFlushResult));
+  int64_t size = -1;
+  std::string path = w->path;
+  int64_t rows = w->rows_written;
+  std::string cols;
+  for (size_t i = 0; i < w->descs.size(); i++) {
+    if (i) cols += ",";
+    cols += w->descs[i].name;
+  }
+  try {
+    if (w->open) size = w->w->close();
+  } catch (std::exception& e) {
+    g_err = e.what();
+    delete w;
+    return -1;
+  }
+  delete w;
+  out->path = strdup(path.c_str());
+  out->size = size;
+  out->rows = rows;
+  out->exist_cols = strdup(cols.c_str());
+  return 0;
+}
+
+extern "C" void lakesoul_c_flush_result_free(LakesoulCFlushResult* r) {
+  if (!r) return;
+  free(r->path);
+  free(r->exist_cols);
+  r->path = nullptr;
+  r->exist_cols = nullptr;
 }
 
 extern "C" void lakesoul_c_writer_abort(LakesoulCWriter* w) {
