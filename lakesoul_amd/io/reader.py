@@ -99,6 +99,7 @@ class LakeSoulScan:
         batch_size: Optional[int] = None,
         incremental: Optional[Tuple[int, int]] = None,
         options: Optional[dict] = None,
+        vector_query: Optional[dict] = None,
     ):
         self.table = table
         self.schema: Schema = table.schema
@@ -117,6 +118,12 @@ class LakeSoulScan:
         self.version = version
         self.timestamp_ms = timestamp_ms
         self.filter_expr = resolve_filters(filters, self.schema)
+        # ANN-result injection: a vector_query turns into a PK id filter
+        # inside the normal scan (reference reader.rs:250-331
+        # inject_vector_search_filter)
+        self.vector_result = None
+        if vector_query is not None:
+            self._inject_vector_filter(dict(vector_query))
         # filter columns must be read (and materialized) even when not
         # selected; the final projection drops them (session.rs:650-730
         # projection computation analog)
@@ -153,6 +160,40 @@ class LakeSoulScan:
                 self.merge_ops[k[len("merge_op."):]] = v
 
     # ------------------------------------------------------------------ #
+
+    def _inject_vector_filter(self, vq: dict) -> None:
+        """Run the ANN search and AND `pk IN (top-k ids)` into the scan
+        filter. vq keys: column (str), query (array, 1-D or (nq, dim)),
+        k (int, default 10), nprobe, rescore, device."""
+        import numpy as _np
+
+        from ..vector.index import VectorIndex
+
+        column = vq.pop("column")
+        query = vq.pop("query")
+        k = int(vq.pop("k", 10))
+        if len(self.pk) != 1:
+            raise ValueError("vector_query needs a single-PK table")
+        root = os.path.join(self.table.table_path, "_vector_index", column)
+        if not os.path.exists(os.path.join(root, "manifest.json")):
+            raise ValueError(
+                f"no vector index for column {column!r}; call "
+                "build_vector_index first")
+        idx = VectorIndex.load(root)
+        kwargs = {}
+        for opt in ("nprobe", "rescore"):
+            if vq.get(opt) is not None:
+                kwargs[opt] = vq[opt]
+        dev = vq.get("device") or self.table.io_config().resolve_device()
+        ids, scores = idx.search(query, k=k, device=str(dev), **kwargs)
+        self.vector_result = (ids, scores)
+        flat = sorted(set(int(i) for i in _np.asarray(ids).ravel() if i >= 0))
+        from .filters import And as _And
+        from .filters import Cmp as _Cmp
+
+        id_filter = _Cmp(self.pk[0], "in", flat)
+        self.filter_expr = (id_filter if self.filter_expr is None
+                            else _And(self.filter_expr, id_filter))
 
     def shard(self, rank: int, world_size: int) -> "LakeSoulScan":
         """DP sharding: scan unit i -> rank i % world_size (reference:

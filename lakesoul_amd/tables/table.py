@@ -139,8 +139,13 @@ class LakeSoulTable:
         batch_size: Optional[int] = None,
         options: Optional[Dict[str, str]] = None,
         incremental: Optional[tuple] = None,
+        vector_query: Optional[dict] = None,
     ):
-        """Build a LakeSoulScan (reference: catalog.py:740 LakeSoulScan)."""
+        """Build a LakeSoulScan (reference: catalog.py:740 LakeSoulScan).
+
+        ``vector_query={"column": ..., "query": vec, "k": 10}`` runs an
+        ANN search on the column's vector index and restricts the scan to
+        the matching PK ids (reference reader.rs:250-331)."""
         from ..io.reader import LakeSoulScan
 
         return LakeSoulScan(
@@ -154,6 +159,7 @@ class LakeSoulTable:
             batch_size=batch_size,
             options=options,
             incremental=incremental,
+            vector_query=vector_query,
         )
 
     def to_arrow(self, **kwargs):

@@ -82,7 +82,8 @@ def kmeans(x: torch.Tensor, k: int, iters: int = 10, seed: int = 0) -> torch.Ten
 class VectorIndex:
     def __init__(self, root: str, column: str, dim: int, metric: str,
                  shards: List[ShardInfo], pk_dtype: str, version: int,
-                 ivf_clusters: int = 0, binary: bool = False):
+                 ivf_clusters: int = 0, binary: bool = False,
+                 rabitq_bits: int = 0):
         self.root = root
         self.column = column
         self.dim = dim
@@ -92,6 +93,7 @@ class VectorIndex:
         self.version = version
         self.ivf_clusters = ivf_clusters
         self.binary = binary
+        self.rabitq_bits = rabitq_bits
         self._gpu_cache: dict = {}
 
     # -- persistence ---------------------------------------------------- #
@@ -107,11 +109,13 @@ class VectorIndex:
             "pk_dtype": self.pk_dtype,
             "version": self.version,
             "created_ms": int(time.time() * 1000),
-            "engine": ("binary+mfma-rescore" if self.binary
+            "engine": ("ivf-rabitq+mfma-rescore" if self.rabitq_bits
+                       else "binary+mfma-rescore" if self.binary
                        else "mfma-exact-bf16" if not self.ivf_clusters
                        else "ivf+mfma-bf16"),
             "ivf_clusters": self.ivf_clusters,
             "binary": self.binary,
+            "rabitq_bits": self.rabitq_bits,
             "shards": [
                 {"bucket_id": s.bucket_id, "num_rows": s.num_rows, "path": s.path}
                 for s in self.shards
@@ -128,7 +132,7 @@ class VectorIndex:
         shards = [ShardInfo(s["bucket_id"], s["num_rows"], s["path"]) for s in m["shards"]]
         return cls(root, m["column"], m["dim"], m["metric"], shards,
                    m["pk_dtype"], m["version"], m.get("ivf_clusters", 0),
-                   m.get("binary", False))
+                   m.get("binary", False), m.get("rabitq_bits", 0))
 
     # -- shard data ----------------------------------------------------- #
 
@@ -148,7 +152,24 @@ class VectorIndex:
             codes = torch.from_numpy(
                 np.fromfile(s.path + ".bin", dtype=np.int64).reshape(s.num_rows, w)
             ).to(device)
-        entry = (vecs, ids, clu, codes)
+        rbq = None
+        if self.rabitq_bits:
+            from .rabitq import QuantizedBatch
+
+            eb = self.rabitq_bits - 1
+            n = s.num_rows
+            bits = torch.from_numpy(np.fromfile(
+                s.path + ".rbq.bits", dtype=np.uint8).reshape(n, (self.dim + 7) // 8))
+            if eb:
+                ex = torch.from_numpy(np.fromfile(
+                    s.path + ".rbq.ex", dtype=np.uint8).reshape(n, (self.dim + 1) // 2))
+            else:
+                ex = torch.empty(n, 0, dtype=torch.uint8)
+            fac = torch.from_numpy(np.fromfile(
+                s.path + ".rbq.fac", dtype=np.float32).reshape(7, n))
+            rbq = QuantizedBatch(self.dim, eb, bits, ex, fac[0], fac[1], fac[2],
+                                 fac[3], fac[4], fac[5], fac[6]).to(device)
+        entry = (vecs, ids, clu, codes, rbq)
         self._gpu_cache[key] = entry
         return entry
 
@@ -203,8 +224,10 @@ class VectorIndex:
             rot = self._rotation()
             qr = q.numpy() @ rot
             qcodes = torch.from_numpy(pack_sign_bits(qr)).to(device)
+        if self.rabitq_bits:
+            return self._search_rabitq(q_dev, k, device, nprobe, rescore)
         for s in self.shards:
-            vecs, ids, clu, codes = self._load_shard(s, device)
+            vecs, ids, clu, codes, _rbq = self._load_shard(s, device)
             if cents is not None and clu is not None:
                 # union of the nprobe nearest clusters over the query batch
                 cscores = q_dev.to(cents.dtype) @ cents.T  # (nq, k_c)
@@ -253,6 +276,137 @@ class VectorIndex:
             best_ids = torch.gather(cand_ids, 1, sel.indices)
         return best_ids.cpu().numpy(), best_scores.cpu().numpy()
 
+    def _search_rabitq(self, q_dev: torch.Tensor, k: int, device,
+                       nprobe: Optional[int], rescore: int):
+        """IVF + RaBitQ staged search (reference ivf/mod.rs
+        search_cluster_v2_batched): probe -> 1-bit FastScan estimate ->
+        ex-code refine -> exact bf16 MFMA rescore -> top-k.
+
+        On GPU the 1-bit pass runs as the LDS-LUT fastscan HIP kernel
+        when available; the torch path computes the same estimate
+        exactly (it is the CPU oracle the kernel is tested against)."""
+        from .rabitq import make_query, unpack_bits, unpack_nibbles
+
+        nq = q_dev.shape[0]
+        metric = "ip" if self.metric == "cosine" else "l2"
+        eb = self.rabitq_bits - 1
+        cents = self._centroids(device)
+        if cents is None:
+            raise ValueError("rabitq index requires IVF centroids")
+        if nprobe is None:
+            nprobe = max(4, self.ivf_clusters // 8)
+        nprobe = min(nprobe, self.ivf_clusters)
+        C = max(32 * k, rescore * k)     # stage-1 candidates per query
+        R = max(4 * k, rescore)          # exact-rescore budget per query
+
+        sum_q = q_dev.sum(dim=1)                      # (nq,)
+        c1_sum_q = -0.5 * sum_q
+        cb = -((1 << eb) - 0.5)
+        cb_sum_q = cb * sum_q
+        bscale = float(1 << eb)
+
+        best_scores = torch.full((nq, k), -float("inf"), device=device)
+        best_ids = torch.full((nq, k), -1, dtype=torch.int64, device=device)
+        for s in self.shards:
+            vecs, ids, clu, _codes, rbq = self._load_shard(s, device)
+            cscores = q_dev.to(cents.dtype) @ cents.T          # (nq, kc)
+            probe = torch.topk(cscores, nprobe, dim=1).indices  # (nq, npb)
+            # per-(query, cluster) g_add
+            if metric == "l2":
+                qn = (q_dev * q_dev).sum(1, keepdim=True)
+                cn = (cents * cents).sum(1)[None, :]
+                g_add_all = qn - 2.0 * (q_dev @ cents.T) + cn
+            else:
+                g_add_all = -(q_dev @ cents.T)
+            probe_mask = torch.zeros(nq, cents.shape[0], dtype=torch.bool,
+                                     device=device)
+            probe_mask.scatter_(1, probe, True)
+
+            cand_est = torch.full((nq, C), float("inf"), device=device)
+            cand_row = torch.full((nq, C), -1, dtype=torch.int64, device=device)
+            cand_ip = torch.zeros(nq, C, device=device)
+            wanted = torch.unique(probe.flatten()).tolist()
+            for cid in wanted:
+                a, b = int(clu[cid]), int(clu[cid + 1])
+                if b <= a:
+                    continue
+                m = b - a
+                use_hip = str(device).startswith("cuda")
+                if use_hip:
+                    try:
+                        from ..ops import hip
+
+                        ip = hip().fastscan_bit_dot(
+                            rbq.bits_packed[a:b], q_dev, self.dim)  # (m, nq)
+                    except (ImportError, AttributeError, RuntimeError):
+                        use_hip = False
+                if not use_hip:
+                    bits_f = unpack_bits(rbq.bits_packed[a:b], self.dim).to(
+                        torch.float32)
+                    ip = bits_f @ q_dev.T                       # (m, nq)
+                est = (rbq.f_add[a:b, None] + g_add_all[:, cid][None, :]
+                       + rbq.f_rescale[a:b, None] * (ip + c1_sum_q[None, :]))
+                est = torch.where(probe_mask[:, cid][None, :], est,
+                                  torch.full_like(est, float("inf")))
+                kk = min(C, m)
+                seg_top = torch.topk(-est, kk, dim=0)           # (kk, nq)
+                seg_est = -seg_top.values.T                     # (nq, kk)
+                seg_row = seg_top.indices.T + a                 # (nq, kk)
+                seg_ip = torch.gather(ip.T, 1, seg_top.indices.T)
+                # merge into running candidate set
+                all_est = torch.cat([cand_est, seg_est], dim=1)
+                all_row = torch.cat([cand_row, seg_row], dim=1)
+                all_ip = torch.cat([cand_ip, seg_ip], dim=1)
+                sel = torch.topk(-all_est, C, dim=1).indices
+                cand_est = torch.gather(all_est, 1, sel)
+                cand_row = torch.gather(all_row, 1, sel)
+                cand_ip = torch.gather(all_ip, 1, sel)
+
+            # stage 2: ex-code refinement of the C candidates
+            if eb > 0:
+                refined = torch.full_like(cand_est, float("inf"))
+                valid = cand_row >= 0
+                rows = cand_row.clamp_min(0)
+                flat_rows = torch.unique(rows.flatten())
+                ex_f = unpack_nibbles(rbq.ex_packed[flat_rows], self.dim).to(
+                    torch.float32)
+                ex_dots_flat = ex_f @ q_dev.T                   # (u, nq)
+                # map row -> position in flat_rows
+                pos = torch.searchsorted(flat_rows, rows.flatten()).view(rows.shape)
+                ex_dot = torch.gather(
+                    ex_dots_flat.T, 1,
+                    pos)                                        # (nq, C) via q-major
+                # per-candidate cluster id for g_add: recover from clu bounds
+                cl_of_row = torch.bucketize(rows.flatten().cpu(),
+                                            clu[1:], right=True).view(rows.shape)
+                g_add_cand = torch.gather(g_add_all, 1, cl_of_row.to(device))
+                total_term = (bscale * cand_ip + ex_dot + cb_sum_q[:, None])
+                refined = (rbq.f_add_ex[rows] + g_add_cand
+                           + rbq.f_rescale_ex[rows] * total_term)
+                refined = torch.where(valid, refined,
+                                      torch.full_like(refined, float("inf")))
+            else:
+                refined = cand_est
+
+            rr = min(R, refined.shape[1])
+            top_r = torch.topk(-refined, rr, dim=1).indices     # (nq, rr)
+            rescore_rows = torch.gather(cand_row, 1, top_r)
+            valid_r = rescore_rows >= 0
+            flat = torch.unique(rescore_rows.clamp_min(0).flatten())
+            sub = self._scores(vecs[flat], q_dev, device)       # (u, nq) exact
+            posr = torch.searchsorted(flat, rescore_rows.clamp_min(0).flatten()
+                                      ).view(rescore_rows.shape)
+            exact = torch.gather(sub.T, 1, posr)
+            exact = torch.where(valid_r, exact,
+                                torch.full_like(exact, -float("inf")))
+            shard_ids = ids[rescore_rows.clamp_min(0)]
+            cand_scores = torch.cat([best_scores, exact], dim=1)
+            cand_ids = torch.cat([best_ids, shard_ids], dim=1)
+            sel = torch.topk(cand_scores, k, dim=1)
+            best_scores = sel.values
+            best_ids = torch.gather(cand_ids, 1, sel.indices)
+        return best_ids.cpu().numpy(), best_scores.cpu().numpy()
+
     def _scores(self, vecs: torch.Tensor, q: torch.Tensor, device) -> torch.Tensor:
         n = vecs.shape[0]
         nq = q.shape[0]
@@ -288,6 +442,7 @@ def build_vector_index(
     device: Optional[str] = None,
     ivf_clusters: int = 0,
     binary: bool = False,
+    rabitq_bits: int = 0,
 ) -> VectorIndex:
     """Build per-bucket exact-search shards for a fixed-size-list float
     column stored as ``dim`` float32/float64 scalar columns or via numpy
@@ -330,6 +485,14 @@ def build_vector_index(
         raise ValueError("no data to index")
 
     dev = device or ("cuda" if torch.cuda.is_available() else "cpu")
+    if rabitq_bits:
+        # RaBitQ needs per-cluster centroids (reference: quantization is
+        # always relative to the IVF centroid, ivf/builder.rs)
+        if not ivf_clusters:
+            n_total = sum(len(p[2]) for p in _shard_payloads)
+            ivf_clusters = int(min(4096, max(8, n_total // 4096)))
+        if binary:
+            raise ValueError("choose either binary or rabitq_bits, not both")
     rot = None
     if binary:
         rot = random_rotation(dim)
@@ -360,11 +523,37 @@ def build_vector_index(
         ids_np[order].tofile(spath + ".ids")
         if rot is not None:
             pack_sign_bits(v_sorted @ rot).tofile(spath + ".bin")
+        if rabitq_bits:
+            from .rabitq import compute_const_scaling_factor, quantize_batch
+
+            eb = rabitq_bits - 1
+            if "_rbq_t" not in locals():
+                _rbq_t = (compute_const_scaling_factor(dim, eb) if eb else None)
+            qmetric = "ip" if metric == "cosine" else "l2"
+            vt = torch.from_numpy(v_sorted).to(dev)
+            ct = centroids.to(dev)
+            bits_parts, ex_parts, fac_parts = [], [], []
+            for cid in range(ivf_clusters):
+                a, b = int(clu[cid]), int(clu[cid + 1])
+                if b <= a:
+                    continue
+                qb = quantize_batch(vt[a:b], ct[cid], eb, _rbq_t, qmetric)
+                bits_parts.append(qb.bits_packed.cpu())
+                ex_parts.append(qb.ex_packed.cpu())
+                fac_parts.append(torch.stack([
+                    qb.f_add, qb.f_rescale, qb.f_error, qb.f_add_ex,
+                    qb.f_rescale_ex, qb.delta, qb.vl]).cpu())
+            torch.cat(bits_parts).numpy().tofile(spath + ".rbq.bits")
+            if eb:
+                torch.cat(ex_parts).numpy().tofile(spath + ".rbq.ex")
+            torch.cat(fac_parts, dim=1).numpy().astype(np.float32).tofile(
+                spath + ".rbq.fac")
         shards.append(ShardInfo(bucket_id, n, spath))
 
     version = table.latest_version() or 0
     idx = VectorIndex(root, column, int(dim), metric, shards, "int64", version,
-                      ivf_clusters=ivf_clusters, binary=binary)
+                      ivf_clusters=ivf_clusters, binary=binary,
+                      rabitq_bits=rabitq_bits)
     idx.save_manifest()
     return idx
 
