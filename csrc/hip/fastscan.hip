@@ -1,0 +1,107 @@
+// RaBitQ FastScan on gfx950: LDS-staged 16-entry lookup tables score the
+// 1-bit codes of a cluster against a query — the MI355X-native
+// replacement for the reference's AVX2 accumulate_batch
+// (rust/lakesoul-vector/src/rabitq/simd.rs:1012-1059, packed 4-bit LUT
+// gather over 32-vector batches).
+//
+// Design (cdna_hip_programming.md): the query's per-4-dim-group LUT
+// (g groups x 16 entries, f32) is built once on device with tensor ops
+// and staged in LDS by each workgroup (12 KB for 768-d — >10 workgroups
+// per CU of occupancy headroom against 160 KB LDS). Each thread owns one
+// vector: its packed sign bits stream from HBM once (w = dim/8 bytes per
+// vector — 16x less traffic than the bf16 exact path), every byte costs
+// two LDS lookups. Memory-bound by construction; the LDS lookups ride
+// under the HBM latency.
+//
+// Bit layout matches lakesoul_amd/vector/rabitq.py pack_bits: byte b of
+// a row covers dims 8b..8b+7 LSB-first, so the low nibble is group 2b
+// and the high nibble group 2b+1.
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+namespace lakesoul {
+
+// bits: [m][w] uint8; lut: [nq][g*16] f32; out: [m][nq] f32
+__global__ __launch_bounds__(256) void fastscan_lut_kernel(
+    const uint8_t* __restrict__ bits, const float* __restrict__ lut,
+    float* __restrict__ out, int64_t m, int32_t nq, int32_t w, int32_t g) {
+  extern __shared__ float slut[];  // g*16 floats
+  int q = (int)blockIdx.y;
+  const float* lq = lut + (int64_t)q * g * 16;
+  for (int i = (int)threadIdx.x; i < g * 16; i += (int)blockDim.x)
+    slut[i] = lq[i];
+  __syncthreads();
+  int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (row >= m) return;
+  const uint8_t* bp = bits + row * (int64_t)w;
+  float acc = 0.f;
+  int b = 0;
+  // 4-byte chunks through a single dword load
+  for (; b + 4 <= w; b += 4) {
+    uint32_t v4;
+    __builtin_memcpy(&v4, bp + b, 4);
+#pragma unroll
+    for (int j = 0; j < 4; j++) {
+      uint32_t byte = (v4 >> (8 * j)) & 0xFF;
+      acc += slut[(2 * (b + j)) * 16 + (byte & 0xF)];
+      int g2 = 2 * (b + j) + 1;
+      if (g2 < g) acc += slut[g2 * 16 + (byte >> 4)];
+    }
+  }
+  for (; b < w; b++) {
+    uint32_t byte = bp[b];
+    acc += slut[(2 * b) * 16 + (byte & 0xF)];
+    int g2 = 2 * b + 1;
+    if (g2 < g) acc += slut[g2 * 16 + (byte >> 4)];
+  }
+  out[row * (int64_t)nq + q] = acc;
+}
+
+extern "C" void launch_fastscan_lut(const uint8_t* bits, const float* lut,
+                                    float* out, int64_t m, int32_t nq,
+                                    int32_t w, int32_t g, hipStream_t s) {
+  dim3 grid((uint32_t)((m + 255) / 256), (uint32_t)nq);
+  size_t lds = (size_t)g * 16 * sizeof(float);
+  hipLaunchKernelGGL(fastscan_lut_kernel, grid, dim3(256), lds, s, bits, lut,
+                     out, m, nq, w, g);
+}
+
+// Ex-code refinement dot: ex nibbles [m][wn] uint8 (wn = ceil(dim/2),
+// low nibble = even dim), q: [nq][dim] f32 -> out [m][nq] f32 of
+// <ex_code, q>. Candidate sets are small (top-C per query), so a simple
+// one-thread-per-(row) loop with q staged in LDS suffices.
+__global__ __launch_bounds__(256) void fastscan_ex_dot_kernel(
+    const uint8_t* __restrict__ ex, const float* __restrict__ qv,
+    float* __restrict__ out, int64_t m, int32_t nq, int32_t wn,
+    int32_t dim) {
+  extern __shared__ float sq[];  // dim floats for this block's query
+  int q = (int)blockIdx.y;
+  const float* qp = qv + (int64_t)q * dim;
+  for (int i = (int)threadIdx.x; i < dim; i += (int)blockDim.x) sq[i] = qp[i];
+  __syncthreads();
+  int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (row >= m) return;
+  const uint8_t* ep = ex + row * (int64_t)wn;
+  float acc = 0.f;
+  for (int b = 0; b < wn; b++) {
+    uint32_t v = ep[b];
+    int d0 = 2 * b;
+    acc += (float)(v & 0xF) * sq[d0];
+    if (d0 + 1 < dim) acc += (float)(v >> 4) * sq[d0 + 1];
+  }
+  out[row * (int64_t)nq + q] = acc;
+}
+
+extern "C" void launch_fastscan_ex_dot(const uint8_t* ex, const float* qv,
+                                       float* out, int64_t m, int32_t nq,
+                                       int32_t wn, int32_t dim,
+                                       hipStream_t s) {
+  dim3 grid((uint32_t)((m + 255) / 256), (uint32_t)nq);
+  size_t lds = (size_t)dim * sizeof(float);
+  hipLaunchKernelGGL(fastscan_ex_dot_kernel, grid, dim3(256), lds, s, ex, qv,
+                     out, m, nq, wn, dim);
+}
+
+}  // namespace lakesoul

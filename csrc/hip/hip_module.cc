@@ -40,6 +40,10 @@ void launch_hamming_scores(const uint64_t*, const uint64_t*, int32_t*, int64_t,
 void launch_str_chunk_keys(const int64_t*, const uint8_t*, int64_t, int64_t*,
                            int64_t, hipStream_t);
 void launch_ann_scores(const short*, const short*, float*, int64_t, int32_t, int32_t, hipStream_t);
+void launch_fastscan_lut(const uint8_t*, const float*, float*, int64_t,
+                         int32_t, int32_t, int32_t, hipStream_t);
+void launch_fastscan_ex_dot(const uint8_t*, const float*, float*, int64_t,
+                            int32_t, int32_t, int32_t, hipStream_t);
 void launch_snappy_decompress(const uint8_t*, const int64_t*, int64_t, uint8_t*, int32_t*, hipStream_t);
 void launch_zstd_decompress(const uint8_t*, const int64_t*, int64_t, uint8_t*,
                             uint8_t*, int64_t, int32_t*, hipStream_t);
@@ -689,6 +693,56 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     launch_hamming_scores((const uint64_t*)codes.data_ptr<int64_t>(),
                           (const uint64_t*)qcodes.data_ptr<int64_t>(),
                           out.data_ptr<int32_t>(), n, nq, words, cur_stream());
+    return out;
+  });
+  m.def("fastscan_bit_dot", [](torch::Tensor bits, torch::Tensor q,
+                               int64_t dim) {
+    // bits (m, w) uint8 cuda; q (nq, dim) f32 cuda -> (m, nq) f32 of
+    // <sign_bits, q> (RaBitQ stage-1 FastScan, LDS-LUT kernel)
+    CHECK_GPU(bits);
+    CHECK_GPU(q);
+    TORCH_CHECK(bits.dtype() == torch::kUInt8 && bits.is_contiguous());
+    TORCH_CHECK(q.dtype() == torch::kFloat32);
+    int64_t m = bits.size(0);
+    int w = (int)bits.size(1);
+    int nq = (int)q.size(0);
+    int g = (int)((dim + 3) / 4);
+    TORCH_CHECK(w == (dim + 7) / 8, "bits width mismatch");
+    // per-query LUT: (nq, g, 16); entry v = sum_j q[4g+j]*((v>>j)&1)
+    auto qp = torch::zeros({nq, (int64_t)g * 4}, q.options());
+    qp.slice(1, 0, dim).copy_(q);
+    auto qg = qp.view({nq, g, 4});
+    static float pat_host[16 * 4];
+    static bool pat_init = false;
+    if (!pat_init) {
+      for (int v = 0; v < 16; v++)
+        for (int j = 0; j < 4; j++) pat_host[v * 4 + j] = (float)((v >> j) & 1);
+      pat_init = true;
+    }
+    auto pat = torch::from_blob(pat_host, {16, 4},
+                                torch::TensorOptions().dtype(torch::kFloat32))
+                   .to(q.device());
+    auto lut = torch::matmul(qg, pat.t()).contiguous();  // (nq, g, 16)
+    auto out = torch::empty({m, nq}, q.options());
+    launch_fastscan_lut(bits.data_ptr<uint8_t>(), lut.data_ptr<float>(),
+                        out.data_ptr<float>(), m, nq, w, g, cur_stream());
+    return out;
+  });
+  m.def("fastscan_ex_dot", [](torch::Tensor ex, torch::Tensor q, int64_t dim) {
+    // ex (m, wn) uint8 nibbles cuda; q (nq, dim) f32 -> (m, nq) f32
+    CHECK_GPU(ex);
+    CHECK_GPU(q);
+    TORCH_CHECK(ex.dtype() == torch::kUInt8 && ex.is_contiguous());
+    TORCH_CHECK(q.dtype() == torch::kFloat32);
+    int64_t m = ex.size(0);
+    int wn = (int)ex.size(1);
+    int nq = (int)q.size(0);
+    TORCH_CHECK(wn == (dim + 1) / 2, "ex width mismatch");
+    auto qc = q.contiguous();
+    auto out = torch::empty({m, nq}, q.options());
+    launch_fastscan_ex_dot(ex.data_ptr<uint8_t>(), qc.data_ptr<float>(),
+                           out.data_ptr<float>(), m, nq, wn, (int)dim,
+                           cur_stream());
     return out;
   });
   m.def("snappy_decompress", &snappy_decompress);

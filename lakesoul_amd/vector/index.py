@@ -368,9 +368,19 @@ class VectorIndex:
                 valid = cand_row >= 0
                 rows = cand_row.clamp_min(0)
                 flat_rows = torch.unique(rows.flatten())
-                ex_f = unpack_nibbles(rbq.ex_packed[flat_rows], self.dim).to(
-                    torch.float32)
-                ex_dots_flat = ex_f @ q_dev.T                   # (u, nq)
+                ex_rows = rbq.ex_packed[flat_rows]
+                use_hip_ex = str(device).startswith("cuda")
+                if use_hip_ex:
+                    try:
+                        from ..ops import hip
+
+                        ex_dots_flat = hip().fastscan_ex_dot(
+                            ex_rows.contiguous(), q_dev, self.dim)  # (u, nq)
+                    except (ImportError, AttributeError, RuntimeError):
+                        use_hip_ex = False
+                if not use_hip_ex:
+                    ex_f = unpack_nibbles(ex_rows, self.dim).to(torch.float32)
+                    ex_dots_flat = ex_f @ q_dev.T               # (u, nq)
                 # map row -> position in flat_rows
                 pos = torch.searchsorted(flat_rows, rows.flatten()).view(rows.shape)
                 ex_dot = torch.gather(

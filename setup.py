@@ -43,6 +43,7 @@ if _WITH_HIP:
                 os.path.join("csrc", "hip", "hip_module.cc"),
                 os.path.join("csrc", "hip", "kernels.hip"),
                 os.path.join("csrc", "hip", "ann.hip"),
+                os.path.join("csrc", "hip", "fastscan.hip"),
                 os.path.join("csrc", "hip", "snappy.hip"),
                 os.path.join("csrc", "hip", "zstd.hip"),
             ],
