@@ -59,7 +59,7 @@ __global__ __launch_bounds__(256) void fastscan_lut_kernel(
   out[row * (int64_t)nq + q] = acc;
 }
 
-extern "C" void launch_fastscan_lut(const uint8_t* bits, const float* lut,
+void launch_fastscan_lut(const uint8_t* bits, const float* lut,
                                     float* out, int64_t m, int32_t nq,
                                     int32_t w, int32_t g, hipStream_t s) {
   dim3 grid((uint32_t)((m + 255) / 256), (uint32_t)nq);
@@ -94,7 +94,7 @@ __global__ __launch_bounds__(256) void fastscan_ex_dot_kernel(
   out[row * (int64_t)nq + q] = acc;
 }
 
-extern "C" void launch_fastscan_ex_dot(const uint8_t* ex, const float* qv,
+void launch_fastscan_ex_dot(const uint8_t* ex, const float* qv,
                                        float* out, int64_t m, int32_t nq,
                                        int32_t wn, int32_t dim,
                                        hipStream_t s) {
