@@ -138,39 +138,34 @@ def main():
     bscale = float(1 << eb)
     g_add_all = -(qs @ centroids.T)  # ip metric
 
-    def staged_search(use_ex: bool, C: int, R: int):
-        cprobe = torch.topk(qs @ centroids.T, nprobe, dim=1).indices
-        probe_mask = torch.zeros(args.queries, clusters, dtype=torch.bool, device=dev)
-        probe_mask.scatter_(1, cprobe, True)
-        cand_est = torch.full((args.queries, C), float("inf"), device=dev)
-        cand_row = torch.full((args.queries, C), -1, dtype=torch.int64, device=dev)
-        cand_ip = torch.zeros(args.queries, C, device=dev)
-        wanted = torch.unique(cprobe.flatten()).tolist()
-        for cid in wanted:
-            a, b = int(clu[cid]), int(clu[cid + 1])
-            if b <= a:
-                continue
-            if str(dev).startswith("cuda"):
-                ip = hip_mod().fastscan_bit_dot(bits_all[a:b], qs, dim)
-            else:
-                from lakesoul_amd.vector.rabitq import unpack_bits
+    cl_of_row = torch.repeat_interleave(
+        torch.arange(clusters, dtype=torch.int64), counts.cpu()).to(dev)
 
-                ip = unpack_bits(bits_all[a:b], dim).to(torch.float32) @ qs.T
-            est = (fac_all[0, a:b, None] + g_add_all[:, cid][None, :]
-                   + fac_all[1, a:b, None] * (ip + c1_sum_q[None, :]))
-            est = torch.where(probe_mask[:, cid][None, :], est,
+    def staged_search(use_ex: bool, C: int, R: int, probe_frac: float = 1.0):
+        """Single-pass fastscan over ALL rows (96 B/vec HBM traffic), est
+        masked to the probed clusters when probe_frac < 1."""
+        if str(dev).startswith("cuda"):
+            ip = hip_mod().fastscan_bit_dot(bits_all, qs, dim)  # (n, nq)
+        else:
+            from lakesoul_amd.vector.rabitq import unpack_bits
+
+            ip = unpack_bits(bits_all, dim).to(torch.float32) @ qs.T
+        est = (fac_all[0][None, :] + g_add_all[:, cl_of_row]
+               + fac_all[1][None, :] * (ip.T + c1_sum_q[:, None]))
+        if probe_frac < 1.0:
+            npb = max(1, int(clusters * probe_frac))
+            cprobe = torch.topk(qs @ centroids.T, npb, dim=1).indices
+            probe_mask = torch.zeros(args.queries, clusters, dtype=torch.bool,
+                                     device=dev)
+            probe_mask.scatter_(1, cprobe, True)
+            est = torch.where(probe_mask[:, cl_of_row], est,
                               torch.full_like(est, float("inf")))
-            kk = min(C, b - a)
-            seg = torch.topk(-est, kk, dim=0)
-            all_est = torch.cat([cand_est, -seg.values.T], dim=1)
-            all_row = torch.cat([cand_row, seg.indices.T + a], dim=1)
-            all_ip = torch.cat([cand_ip, torch.gather(ip.T, 1, seg.indices.T)], dim=1)
-            sel = torch.topk(-all_est, C, dim=1).indices
-            cand_est = torch.gather(all_est, 1, sel)
-            cand_row = torch.gather(all_row, 1, sel)
-            cand_ip = torch.gather(all_ip, 1, sel)
+        top_c = torch.topk(-est, min(C, n), dim=1)
+        cand_est = -top_c.values
+        cand_row = top_c.indices
+        cand_ip = torch.gather(ip.T, 1, cand_row)
         if use_ex:
-            rows = cand_row.clamp_min(0)
+            rows = cand_row
             flat = torch.unique(rows.flatten())
             if str(dev).startswith("cuda"):
                 exd = hip_mod().fastscan_ex_dot(ex_all[flat].contiguous(), qs, dim)
@@ -180,9 +175,7 @@ def main():
                 exd = unpack_nibbles(ex_all[flat], dim).to(torch.float32) @ qs.T
             pos = torch.searchsorted(flat, rows.flatten()).view(rows.shape)
             ex_dot = torch.gather(exd.T, 1, pos)
-            cl_of = torch.bucketize(rows.flatten().cpu(), clu[1:], right=True
-                                    ).view(rows.shape).to(dev)
-            g_add_cand = torch.gather(g_add_all, 1, cl_of)
+            g_add_cand = torch.gather(g_add_all, 1, cl_of_row[rows])
             tt = bscale * cand_ip + ex_dot + cb_sum_q[:, None]
             refined = fac_all[3, rows] + g_add_cand + fac_all[4, rows] * tt
             refined = torch.where(cand_row >= 0, refined,
@@ -201,21 +194,25 @@ def main():
         final_rows = torch.gather(rrows, 1, top)
         return order[final_rows.flatten()].view(final_rows.shape)
 
-    for name, use_ex, C, R in [("ivf-1bit+rescore", False, 32 * k, 8 * k),
-                               ("ivf-rabitq4+rescore", True, 32 * k, 4 * k)]:
-        ids = staged_search(use_ex, C, R)
+    cases = [
+        ("fastscan-1bit+rescore", False, 128 * k, 16 * k, 1.0),
+        ("fastscan-rabitq4+rescore", True, 128 * k, 8 * k, 1.0),
+        ("ivf25-rabitq4+rescore", True, 128 * k, 8 * k, 0.25),
+    ]
+    for name, use_ex, C, R, pf in cases:
+        ids = staged_search(use_ex, C, R, pf)
         recall = np.mean([
             len(truth_sets[i] & set(ids[i].tolist())) / k
             for i in range(args.queries)])
         torch.cuda.synchronize() if dev == "cuda" else None
         t0 = time.time()
         for _ in range(reps):
-            staged_search(use_ex, C, R)
+            staged_search(use_ex, C, R, pf)
         torch.cuda.synchronize() if dev == "cuda" else None
         qps = args.queries * reps / (time.time() - t0)
         print(json.dumps({"engine": name, "recall_at_10": float(recall),
                           "qps": qps, "n": n, "dim": dim,
-                          "clusters": clusters, "nprobe": nprobe,
+                          "clusters": clusters, "probe_frac": pf,
                           "C": C, "R": R,
                           "ivf_s": ivf_s, "quant_s": quant_s}), flush=True)
 

@@ -167,7 +167,7 @@ def test_gpu_rabitq_index_recall(dev, tmp_path):
     idx = build_vector_index(t, "emb", metric="cosine", rabitq_bits=4,
                              ivf_clusters=64, device="cuda")
     q = vecs[rng.choice(n, 32, replace=False)]
-    ids_r, _ = idx.search(q, k=10, device="cuda", nprobe=16, rescore=40)
+    ids_r, _ = idx.search(q, k=10, device="cuda", rescore=40)
     ids_e, _ = exact.search(q, k=10, device="cuda")
     recall = np.mean([len(set(ids_r[i]) & set(ids_e[i])) / 10.0
                       for i in range(32)])
