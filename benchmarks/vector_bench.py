@@ -144,14 +144,19 @@ def main():
     def staged_search(use_ex: bool, C: int, R: int, probe_frac: float = 1.0):
         """Single-pass fastscan over ALL rows (96 B/vec HBM traffic), est
         masked to the probed clusters when probe_frac < 1."""
+        ip_T = None
         if str(dev).startswith("cuda"):
-            ip = hip_mod().fastscan_bit_dot(bits_all, qs, dim)  # (n, nq)
+            est = hip_mod().fastscan_est(
+                bits_all, qs, dim, fac_all[0].contiguous(),
+                fac_all[1].contiguous(), cl_of_row.to(torch.int32),
+                g_add_all.contiguous(), c1_sum_q.contiguous())   # (nq, n)
         else:
             from lakesoul_amd.vector.rabitq import unpack_bits
 
             ip = unpack_bits(bits_all, dim).to(torch.float32) @ qs.T
-        est = (fac_all[0][None, :] + g_add_all[:, cl_of_row]
-               + fac_all[1][None, :] * (ip.T + c1_sum_q[:, None]))
+            ip_T = ip.T
+            est = (fac_all[0][None, :] + g_add_all[:, cl_of_row]
+                   + fac_all[1][None, :] * (ip_T + c1_sum_q[:, None]))
         if probe_frac < 1.0:
             npb = max(1, int(clusters * probe_frac))
             cprobe = torch.topk(qs @ centroids.T, npb, dim=1).indices
@@ -163,7 +168,15 @@ def main():
         top_c = torch.topk(-est, min(C, n), dim=1)
         cand_est = -top_c.values
         cand_row = top_c.indices
-        cand_ip = torch.gather(ip.T, 1, cand_row)
+        if ip_T is not None:
+            cand_ip = torch.gather(ip_T, 1, cand_row)
+        else:
+            fr = fac_all[1][cand_row]
+            ga = torch.gather(g_add_all, 1, cl_of_row[cand_row])
+            cand_ip = torch.where(
+                fr != 0,
+                (cand_est - fac_all[0][cand_row] - ga) / fr - c1_sum_q[:, None],
+                torch.zeros_like(cand_est))
         if use_ex:
             rows = cand_row
             flat = torch.unique(rows.flatten())
@@ -195,9 +208,10 @@ def main():
         return order[final_rows.flatten()].view(final_rows.shape)
 
     cases = [
-        ("fastscan-1bit+rescore", False, 128 * k, 16 * k, 1.0),
-        ("fastscan-rabitq4+rescore", True, 128 * k, 8 * k, 1.0),
-        ("ivf25-rabitq4+rescore", True, 128 * k, 8 * k, 0.25),
+        ("fastscan-1bit+rescore", False, 256 * k, 32 * k, 1.0),
+        ("fastscan-rabitq4+rescore", True, 256 * k, 16 * k, 1.0),
+        ("fastscan-rabitq4-hiC", True, 1024 * k, 16 * k, 1.0),
+        ("ivf25-rabitq4+rescore", True, 256 * k, 16 * k, 0.25),
     ]
     for name, use_ex, C, R, pf in cases:
         ids = staged_search(use_ex, C, R, pf)

@@ -68,6 +68,64 @@ void launch_fastscan_lut(const uint8_t* bits, const float* lut,
                      out, m, nq, w, g);
 }
 
+// Fused stage-1 estimator: the LUT accumulate PLUS the RaBitQ
+// correction factors applied in-register, writing the estimated
+// distance directly (est[q][row] = f_add[row] + g_add[q][cl(row)] +
+// f_rescale[row] * (acc + c1_sum_q[q])) — saves two full (nq x n) f32
+// round trips through HBM vs computing est from a raw ip tensor.
+// Output is (nq, n) row-major so the per-query top-C reads contiguously.
+__global__ __launch_bounds__(256) void fastscan_est_kernel(
+    const uint8_t* __restrict__ bits, const float* __restrict__ lut,
+    const float* __restrict__ f_add, const float* __restrict__ f_rescale,
+    const int32_t* __restrict__ cl_of_row, const float* __restrict__ g_add,
+    const float* __restrict__ c1_sum_q, float* __restrict__ out, int64_t m,
+    int32_t nq, int32_t w, int32_t g, int32_t n_clusters) {
+  extern __shared__ float slut[];  // g*16 floats
+  int q = (int)blockIdx.y;
+  const float* lq = lut + (int64_t)q * g * 16;
+  for (int i = (int)threadIdx.x; i < g * 16; i += (int)blockDim.x)
+    slut[i] = lq[i];
+  __syncthreads();
+  int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (row >= m) return;
+  const uint8_t* bp = bits + row * (int64_t)w;
+  float acc = 0.f;
+  int b = 0;
+  for (; b + 4 <= w; b += 4) {
+    uint32_t v4;
+    __builtin_memcpy(&v4, bp + b, 4);
+#pragma unroll
+    for (int j = 0; j < 4; j++) {
+      uint32_t byte = (v4 >> (8 * j)) & 0xFF;
+      acc += slut[(2 * (b + j)) * 16 + (byte & 0xF)];
+      int g2 = 2 * (b + j) + 1;
+      if (g2 < g) acc += slut[g2 * 16 + (byte >> 4)];
+    }
+  }
+  for (; b < w; b++) {
+    uint32_t byte = bp[b];
+    acc += slut[(2 * b) * 16 + (byte & 0xF)];
+    int g2 = 2 * b + 1;
+    if (g2 < g) acc += slut[g2 * 16 + (byte >> 4)];
+  }
+  float ga = g_add[(int64_t)q * n_clusters + cl_of_row[row]];
+  out[(int64_t)q * m + row] =
+      f_add[row] + ga + f_rescale[row] * (acc + c1_sum_q[q]);
+}
+
+void launch_fastscan_est(const uint8_t* bits, const float* lut,
+                         const float* f_add, const float* f_rescale,
+                         const int32_t* cl_of_row, const float* g_add,
+                         const float* c1_sum_q, float* out, int64_t m,
+                         int32_t nq, int32_t w, int32_t g,
+                         int32_t n_clusters, hipStream_t s) {
+  dim3 grid((uint32_t)((m + 255) / 256), (uint32_t)nq);
+  size_t lds = (size_t)g * 16 * sizeof(float);
+  hipLaunchKernelGGL(fastscan_est_kernel, grid, dim3(256), lds, s, bits, lut,
+                     f_add, f_rescale, cl_of_row, g_add, c1_sum_q, out, m, nq,
+                     w, g, n_clusters);
+}
+
 // Ex-code refinement dot: ex nibbles [m][wn] uint8 (wn = ceil(dim/2),
 // low nibble = even dim), q: [nq][dim] f32 -> out [m][nq] f32 of
 // <ex_code, q>. Candidate sets are small (top-C per query), so a simple

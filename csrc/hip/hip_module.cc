@@ -44,6 +44,10 @@ void launch_fastscan_lut(const uint8_t*, const float*, float*, int64_t,
                          int32_t, int32_t, int32_t, hipStream_t);
 void launch_fastscan_ex_dot(const uint8_t*, const float*, float*, int64_t,
                             int32_t, int32_t, int32_t, hipStream_t);
+void launch_fastscan_est(const uint8_t*, const float*, const float*,
+                         const float*, const int32_t*, const float*,
+                         const float*, float*, int64_t, int32_t, int32_t,
+                         int32_t, int32_t, hipStream_t);
 void launch_snappy_decompress(const uint8_t*, const int64_t*, int64_t, uint8_t*, int32_t*, hipStream_t);
 void launch_zstd_decompress(const uint8_t*, const int64_t*, int64_t, uint8_t*,
                             uint8_t*, int64_t, int32_t*, hipStream_t);
@@ -726,6 +730,54 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     auto out = torch::empty({m, nq}, q.options());
     launch_fastscan_lut(bits.data_ptr<uint8_t>(), lut.data_ptr<float>(),
                         out.data_ptr<float>(), m, nq, w, g, cur_stream());
+    return out;
+  });
+  m.def("fastscan_est", [](torch::Tensor bits, torch::Tensor q, int64_t dim,
+                           torch::Tensor f_add, torch::Tensor f_rescale,
+                           torch::Tensor cl_of_row, torch::Tensor g_add,
+                           torch::Tensor c1_sum_q) {
+    // fused RaBitQ stage-1: (nq, n) estimated distances in one pass
+    CHECK_GPU(bits);
+    CHECK_GPU(q);
+    CHECK_GPU(f_add);
+    CHECK_GPU(f_rescale);
+    CHECK_GPU(cl_of_row);
+    CHECK_GPU(g_add);
+    CHECK_GPU(c1_sum_q);
+    TORCH_CHECK(bits.dtype() == torch::kUInt8 && bits.is_contiguous());
+    TORCH_CHECK(cl_of_row.dtype() == torch::kInt32);
+    int64_t m = bits.size(0);
+    int w = (int)bits.size(1);
+    int nq = (int)q.size(0);
+    int g = (int)((dim + 3) / 4);
+    int kc = (int)g_add.size(1);
+    TORCH_CHECK(w == (dim + 7) / 8, "bits width mismatch");
+    TORCH_CHECK(g_add.size(0) == nq && c1_sum_q.size(0) == nq);
+    TORCH_CHECK(f_add.size(0) == m && f_rescale.size(0) == m &&
+                cl_of_row.size(0) == m);
+    auto qp = torch::zeros({nq, (int64_t)g * 4}, q.options());
+    qp.slice(1, 0, dim).copy_(q);
+    auto qg = qp.view({nq, g, 4});
+    static float pat_host2[16 * 4];
+    static bool pat_init2 = false;
+    if (!pat_init2) {
+      for (int v = 0; v < 16; v++)
+        for (int j = 0; j < 4; j++) pat_host2[v * 4 + j] = (float)((v >> j) & 1);
+      pat_init2 = true;
+    }
+    auto pat = torch::from_blob(pat_host2, {16, 4},
+                                torch::TensorOptions().dtype(torch::kFloat32))
+                   .to(q.device());
+    auto lut = torch::matmul(qg, pat.t()).contiguous();
+    auto out = torch::empty({nq, m}, q.options());
+    launch_fastscan_est(
+        bits.data_ptr<uint8_t>(), lut.data_ptr<float>(),
+        f_add.contiguous().data_ptr<float>(),
+        f_rescale.contiguous().data_ptr<float>(),
+        cl_of_row.contiguous().data_ptr<int32_t>(),
+        g_add.contiguous().data_ptr<float>(),
+        c1_sum_q.contiguous().data_ptr<float>(), out.data_ptr<float>(), m, nq,
+        w, g, kc, cur_stream());
     return out;
   });
   m.def("fastscan_ex_dot", [](torch::Tensor ex, torch::Tensor q, int64_t dim) {

@@ -381,6 +381,31 @@ class LakeSoulScan:
             return schema_to_arrow(self.out_schema).empty_table()
         return pa.concat_tables(tables)
 
+    def to_batch(self) -> Batch:
+        """Whole scan as one Batch (tensor columns; stays on the scan
+        device — the query engine's input)."""
+        batches = list(self.iter_batches())
+        if not batches:
+            import numpy as _np
+
+            cols = {}
+            for f in self.out_schema:
+                if f.dtype in ("string", "binary"):
+                    cols[f.name] = Column(
+                        f.dtype, offsets=torch.zeros(1, dtype=torch.int64),
+                        bytes_=torch.empty(0, dtype=torch.uint8))
+                else:
+                    from .batch import torch_dtype_for
+
+                    cols[f.name] = Column(
+                        f.dtype, data=torch.empty(0, dtype=torch_dtype_for(f.dtype)))
+            return Batch(self.out_schema, cols)
+        if len(batches) == 1:
+            return batches[0]
+        from .batch import concat_batches
+
+        return concat_batches(batches)
+
     def count(self) -> int:
         """Count-only fast path (EmptyScanCountExec analog,
         physical_plan/empty_schema.rs:192): row counts come from parquet

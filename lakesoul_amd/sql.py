@@ -805,14 +805,51 @@ def _execute_join_select(catalog, q: Query, device=None):
     if star:
         need_l, need_r = set(lnames), set(rnames)
 
-    ldf = lt.scan(columns=sorted(need_l) or None, version=q.version,
-                  device=device).to_arrow().to_pandas()
-    rdf = rt.scan(columns=sorted(need_r) or None,
-                  device=device).to_arrow().to_pandas()
-    # disambiguate overlapping names: right-side dupes get "<qual>." prefix
+    lscan = lt.scan(columns=sorted(need_l) or None, version=q.version,
+                    device=device)
+    rscan = rt.scan(columns=sorted(need_r) or None, device=device)
     rqual = j.alias or j.table
-    overlap = set(ldf.columns) & set(rdf.columns)
-    rdf = rdf.rename(columns={c: f"{rqual}.{c}" for c in overlap})
+
+    # join key column names per side
+    key_l, key_r = [], []
+    for lcol, rcol in j.on:
+        sl, cl = side_of(lcol)
+        sr, cr = side_of(rcol)
+        if sl == sr:
+            raise SqlError("JOIN ON must reference both tables")
+        if sl == "R":
+            cl, cr = cr, cl
+        key_l.append(cl)
+        key_r.append(cr)
+
+    if _use_pandas_exec():
+        ldf = lscan.to_arrow().to_pandas()
+        rdf = rscan.to_arrow().to_pandas()
+        overlap = set(ldf.columns) & set(rdf.columns)
+        rdf = rdf.rename(columns={c: f"{rqual}.{c}" for c in overlap})
+
+        def df_col(name: str) -> str:
+            side, col = side_of(name)
+            if side == "R" and col in overlap:
+                return f"{rqual}.{col}"
+            return col
+
+        right_on = [f"{rqual}.{c}" if c in overlap else c for c in key_r]
+        out = ldf.merge(rdf, how=j.kind, left_on=key_l, right_on=right_on)
+        if q.where is not None:
+            out = out[_pd_eval(q.where, out, df_col)].reset_index(drop=True)
+        return _project_and_finish(q, out, df_col,
+                                   all_cols=list(ldf.columns) + list(rdf.columns))
+
+    # tensor engine path (vectorized hash join, query/engine.py)
+    import torch as _torch
+
+    from .query.engine import join_batches
+
+    lb = lscan.to_batch()
+    rb = rscan.to_batch()
+    overlap = set(lb.schema.names()) & set(rb.schema.names())
+    rename = {c: f"{rqual}.{c}" for c in overlap}
 
     def df_col(name: str) -> str:
         side, col = side_of(name)
@@ -820,26 +857,123 @@ def _execute_join_select(catalog, q: Query, device=None):
             return f"{rqual}.{col}"
         return col
 
-    left_on = []
-    right_on = []
-    for lcol, rcol in j.on:
-        sl, cl = side_of(lcol)
-        sr, cr = side_of(rcol)
-        if sl == sr:
-            raise SqlError("JOIN ON must reference both tables")
-        if sl == "R":
-            lcol, rcol = rcol, lcol
-            cl, cr = cr, cl
-        left_on.append(cl)
-        right_on.append(f"{rqual}.{cr}" if cr in overlap else cr)
-    out = ldf.merge(rdf, how=j.kind, left_on=left_on, right_on=right_on)
-
+    joined = join_batches(lb, rb, key_l, key_r, j.kind, rename_right=rename)
     if q.where is not None:
-        out = out[_pd_eval(q.where, out, df_col)].reset_index(drop=True)
+        view = _expr_view_batch(joined, q.where, df_col)
+        mask = q.where.evaluate(view)
+        joined = joined.take(_torch.nonzero(mask, as_tuple=False).flatten())
+    return _project_and_finish_tensor(
+        q, joined, df_col,
+        all_cols=list(joined.schema.names()))
 
-    # project / aggregate (shared tail with the single-table path)
-    return _project_and_finish(q, out, df_col,
-                               all_cols=list(ldf.columns) + list(rdf.columns))
+
+def _use_pandas_exec() -> bool:
+    """Escape hatch: LAKESOUL_SQL_PANDAS=1 selects the legacy pandas
+    execution (kept as the cross-check oracle; the default is the tensor
+    engine, lakesoul_amd/query/engine.py)."""
+    import os
+
+    return os.environ.get("LAKESOUL_SQL_PANDAS", "0") == "1"
+
+
+def _expr_view_batch(batch, expr: Expr, col):
+    """A Batch whose column names match the expression's references
+    (qualified names resolved through ``col``)."""
+    from .io.batch import Batch as _B
+    from .io.schema import Field as _F
+    from .io.schema import Schema as _S
+
+    fields, cols = [], {}
+    for name in sorted(expr.columns()):
+        src = col(name)
+        c = batch.columns[src]
+        f = batch.schema.field(src)
+        fields.append(_F(name, f.dtype, f.nullable))
+        cols[name] = c
+    return _B(_S(fields), cols)
+
+
+def _project_and_finish_tensor(q: Query, batch, col, all_cols):
+    """Tensor-engine SELECT tail: aggregation / projection / DISTINCT /
+    HAVING / ORDER BY / LIMIT over a Batch; pandas materializes only the
+    final (small) result."""
+    import torch as _torch
+
+    from .io.batch import Batch as _B
+    from .io.schema import Field as _F
+    from .io.schema import Schema as _S
+    from .query.engine import distinct_indices, groupby_agg, sort_indices
+
+    has_agg = any(it.kind == "agg" for it in q.items)
+    if has_agg or q.group_by:
+        gcols = [col(g) for g in q.group_by]
+        aggs = []
+        out_order = []
+        for it in q.items:
+            if it.kind == "col":
+                if it.name not in q.group_by:
+                    raise SqlError(f"column {it.name!r} must appear in GROUP BY")
+                out_order.append(("group", col(it.name),
+                                  it.alias or it.name.split(".")[-1]))
+                continue
+            if it.kind == "star":
+                raise SqlError("SELECT * with aggregates is not valid")
+            aggs.append((it.fn, col(it.name) if it.name else None,
+                         it.out_name, it.distinct))
+            out_order.append(("agg", it.out_name, it.out_name))
+        res = groupby_agg(batch, gcols, aggs)
+        fields, cols = [], {}
+        for kind, src, outn in out_order:
+            f = res.schema.field(src)
+            fields.append(_F(outn, f.dtype, f.nullable))
+            cols[outn] = res.columns[src]
+        out = _B(_S(fields), cols)
+    else:
+        fields, cols = [], {}
+        for it in q.items:
+            if it.kind == "star":
+                for c in all_cols:
+                    if c not in cols:
+                        fields.append(_F(c, batch.schema.field(c).dtype,
+                                         batch.schema.field(c).nullable))
+                        cols[c] = batch.columns[c]
+            else:
+                src = col(it.name)
+                outn = it.alias or it.name.split(".")[-1]
+                if outn not in cols:
+                    f = batch.schema.field(src)
+                    fields.append(_F(outn, f.dtype, f.nullable))
+                    cols[outn] = batch.columns[src]
+        out = _B(_S(fields), cols)
+        if q.distinct:
+            out = out.take(distinct_indices(out, [f.name for f in out.schema]))
+
+    if q.having is not None:
+        view = _expr_view_batch(out, q.having, lambda n: (
+            n if n in out.schema.names() else n.split(".")[-1]))
+        mask = q.having.evaluate(view)
+        out = out.take(_torch.nonzero(mask, as_tuple=False).flatten())
+    if q.order_by:
+        by = []
+        for n, d in q.order_by:
+            if n in out.schema.names():
+                by.append((n, not d))
+            else:
+                cn = col(n)
+                if cn in out.schema.names():
+                    by.append((cn, not d))
+                else:
+                    base = n.split(".")[-1]
+                    if base in out.schema.names():
+                        by.append((base, not d))
+                    else:
+                        raise SqlError(f"ORDER BY references unknown column {n!r}")
+        out = out.take(sort_indices(out, by))
+    a = q.offset or 0
+    b = out.num_rows if q.limit is None else min(out.num_rows, a + q.limit)
+    if a or b < out.num_rows:
+        out = out.slice(a, max(a, b))
+    return out.to_arrow().to_pandas()
 
 
 def _project_and_finish(q: Query, df, col, all_cols):
@@ -1068,8 +1202,13 @@ def _execute_select(catalog, q: Query, device=None):
     scan = t.scan(columns=sorted(need) or None, filters=q.where,
                   version=q.version, timestamp_ms=q.timestamp_ms,
                   device=device)
-    df = scan.to_arrow().to_pandas()
-    return _project_and_finish(q, df, lambda n: n, all_cols=schema_cols)
+    if _use_pandas_exec():
+        df = scan.to_arrow().to_pandas()
+        return _project_and_finish(q, df, lambda n: n, all_cols=schema_cols)
+    batch = scan.to_batch()
+    return _project_and_finish_tensor(q, batch, lambda n: n,
+                                      all_cols=[c for c in schema_cols
+                                                if c in batch.schema.names()])
 
 
 def repl(catalog, device=None, input_fn=input, print_fn=print):

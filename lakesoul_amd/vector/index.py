@@ -214,6 +214,11 @@ class VectorIndex:
             q = q / q.norm(dim=1, keepdim=True).clamp_min(1e-30)
         q_dev = q.to(device)
 
+        if self.rabitq_bits:
+            # dispatch BEFORE nprobe defaulting: the fastscan engine's
+            # default is a full scan (probing is a mask there, not a
+            # traffic saver)
+            return self._search_rabitq(q_dev, k, device, nprobe, rescore)
         best_scores = torch.full((nq, k), -float("inf"), device=device)
         best_ids = torch.full((nq, k), -1, dtype=torch.int64, device=device)
         cents = self._centroids(device)
@@ -224,8 +229,6 @@ class VectorIndex:
             rot = self._rotation()
             qr = q.numpy() @ rot
             qcodes = torch.from_numpy(pack_sign_bits(qr)).to(device)
-        if self.rabitq_bits:
-            return self._search_rabitq(q_dev, k, device, nprobe, rescore)
         for s in self.shards:
             vecs, ids, clu, codes, _rbq = self._load_shard(s, device)
             if cents is not None and clu is not None:
@@ -331,21 +334,28 @@ class VectorIndex:
 
             # stage 1: ONE fastscan pass over the whole shard (96 B/vec of
             # HBM traffic at 768-d — scanning everything beats per-cluster
-            # launches; probe pruning is a mask, not a loop)
+            # launches; probe pruning is a mask, not a loop). On GPU the
+            # fused kernel applies the correction factors in-register and
+            # writes est directly (no (nq, n) intermediate round trips).
             use_hip = str(device).startswith("cuda")
+            ip_T = None
             if use_hip:
                 try:
                     from ..ops import hip
 
-                    ip = hip().fastscan_bit_dot(rbq.bits_packed, q_dev,
-                                                self.dim)       # (n, nq)
+                    est = hip().fastscan_est(
+                        rbq.bits_packed, q_dev, self.dim, rbq.f_add,
+                        rbq.f_rescale, cl_of_row.to(torch.int32),
+                        g_add_all.contiguous(), c1_sum_q.contiguous()
+                    )                                           # (nq, n)
                 except (ImportError, AttributeError, RuntimeError):
                     use_hip = False
             if not use_hip:
                 bits_f = unpack_bits(rbq.bits_packed, self.dim).to(torch.float32)
                 ip = bits_f @ q_dev.T                           # (n, nq)
-            est = (rbq.f_add[None, :] + g_add_all[:, cl_of_row]
-                   + rbq.f_rescale[None, :] * (ip.T + c1_sum_q[:, None]))
+                ip_T = ip.T
+                est = (rbq.f_add[None, :] + g_add_all[:, cl_of_row]
+                       + rbq.f_rescale[None, :] * (ip_T + c1_sum_q[:, None]))
             if nprobe < self.ivf_clusters:
                 cscores = q_dev.to(cents.dtype) @ cents.T       # (nq, kc)
                 probe = torch.topk(cscores, nprobe, dim=1).indices
@@ -358,7 +368,17 @@ class VectorIndex:
             top_c = torch.topk(-est, kk, dim=1)
             cand_est = -top_c.values                            # (nq, kk)
             cand_row = top_c.indices
-            cand_ip = torch.gather(ip.T, 1, cand_row)
+            if ip_T is not None:
+                cand_ip = torch.gather(ip_T, 1, cand_row)
+            else:
+                # recover <bits, q> algebraically on the candidates only
+                fr = rbq.f_rescale[cand_row]
+                ga = torch.gather(g_add_all, 1, cl_of_row[cand_row])
+                cand_ip = torch.where(
+                    fr != 0,
+                    (cand_est - rbq.f_add[cand_row] - ga) / fr
+                    - c1_sum_q[:, None],
+                    torch.zeros_like(cand_est))
 
             # stage 2: ex-code refinement of the C candidates
             if eb > 0:

@@ -126,6 +126,30 @@ def test_fastscan_bit_dot_vs_oracle(dev):
         torch.testing.assert_close(got, ref, rtol=1e-4, atol=1e-3)
 
 
+def test_fastscan_est_fused_vs_oracle(dev):
+    """Fused estimator kernel vs the torch composition of the same
+    quantities."""
+    from lakesoul_amd.ops import hip
+    from lakesoul_amd.vector.rabitq import pack_bits, unpack_bits
+
+    rng = np.random.default_rng(9)
+    m, nq, dim, kc = 3000, 5, 256, 16
+    bits = torch.from_numpy(rng.integers(0, 2, (m, dim)).astype(np.uint8))
+    packed = pack_bits(bits.bool()).to(dev)
+    q = torch.from_numpy(rng.normal(size=(nq, dim)).astype(np.float32)).to(dev)
+    f_add = torch.from_numpy(rng.normal(size=m).astype(np.float32)).to(dev)
+    f_res = torch.from_numpy(rng.normal(size=m).astype(np.float32)).to(dev)
+    cl = torch.from_numpy(rng.integers(0, kc, m).astype(np.int32)).to(dev)
+    g_add = torch.from_numpy(rng.normal(size=(nq, kc)).astype(np.float32)).to(dev)
+    c1sq = (-0.5 * q.sum(dim=1)).contiguous()
+    got = hip().fastscan_est(packed, q, dim, f_add, f_res, cl, g_add, c1sq).cpu()
+    ip = (bits.to(torch.float32) @ q.cpu().T).T  # (nq, m)
+    ref = (f_add.cpu()[None, :] + torch.gather(
+        g_add.cpu(), 1, cl.cpu().to(torch.int64)[None, :].expand(nq, m))
+        + f_res.cpu()[None, :] * (ip + c1sq.cpu()[:, None]))
+    torch.testing.assert_close(got, ref, rtol=1e-4, atol=1e-3)
+
+
 def test_fastscan_ex_dot_vs_oracle(dev):
     from lakesoul_amd.ops import hip
     from lakesoul_amd.vector.rabitq import pack_nibbles, unpack_nibbles
