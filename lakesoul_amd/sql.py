@@ -22,9 +22,9 @@ from .io.filters import And, Cmp, Expr, IsNull, Literal, Not, Or
 
 _TOKEN_RE = re.compile(
     r"""\s*(?:
-        (?P<num>-?\d+\.\d+(?:[eE][+-]?\d+)?|-?\.\d+|-?\d+)
+        (?P<num>\d+\.\d+(?:[eE][+-]?\d+)?|\.\d+|\d+)
       | (?P<str>'(?:[^']|'')*')
-      | (?P<op><=|>=|<>|!=|=|<|>|\(|\)|,|\*|\.)
+      | (?P<op><=|>=|<>|!=|=|<|>|\(|\)|,|\*|\.|\+|-|/)
       | (?P<id>[A-Za-z_][A-Za-z_0-9]*)
     )""",
     re.VERBOSE,
@@ -96,6 +96,34 @@ def tokenize(sql: str) -> List[Tuple[str, str]]:
 
 
 @dataclass
+class Scalar:
+    """Arithmetic scalar expression over columns and literals (the
+    reference gets this from DataFusion's ProjectionExec;
+    session.rs:966-1036)."""
+    kind: str                   # "col" | "num" | "bin"
+    name: str = ""              # col
+    value: float = 0.0          # num
+    op: str = ""                # + - * /
+    left: Optional["Scalar"] = None
+    right: Optional["Scalar"] = None
+
+    def columns(self) -> set:
+        if self.kind == "col":
+            return {self.name}
+        if self.kind == "bin":
+            return self.left.columns() | self.right.columns()
+        return set()
+
+    def display(self) -> str:
+        if self.kind == "col":
+            return self.name
+        if self.kind == "num":
+            v = self.value
+            return str(int(v)) if float(v).is_integer() else str(v)
+        return f"({self.left.display()} {self.op} {self.right.display()})"
+
+
+@dataclass
 class SelectItem:
     # one of: column name | ("agg", fn, col_or_star) | ("lit", value)
     kind: str                      # "col" | "agg" | "star"
@@ -103,13 +131,17 @@ class SelectItem:
     fn: str = ""                   # aggregate fn
     alias: str = ""
     distinct: bool = False         # count(DISTINCT col)
+    expr: Optional[Scalar] = None  # computed scalar (name empty)
 
     @property
     def out_name(self) -> str:
         if self.alias:
             return self.alias
         if self.kind == "agg":
-            return f"{self.fn}({self.name or '*'})"
+            arg = self.expr.display() if self.expr is not None else (self.name or "*")
+            return f"{self.fn}({arg})"
+        if self.expr is not None:
+            return self.expr.display()
         return self.name
 
 
@@ -426,6 +458,49 @@ class _Parser:
             raise SqlError(f"trailing tokens: {self.peek()[1]!r}")
         return q
 
+    # ---- scalar expressions (q1/q6-style arithmetic) ---- #
+
+    def scalar_expr(self) -> Scalar:
+        e = self.scalar_term()
+        while self.peek() in (("op", "+"), ("op", "-")):
+            op = self.next()[1]
+            e = Scalar("bin", op=op, left=e, right=self.scalar_term())
+        return e
+
+    def scalar_term(self) -> Scalar:
+        e = self.scalar_factor()
+        while self.peek() in (("op", "*"), ("op", "/")):
+            op = self.next()[1]
+            e = Scalar("bin", op=op, left=e, right=self.scalar_factor())
+        return e
+
+    def scalar_factor(self) -> Scalar:
+        k, v = self.peek()
+        if (k, v) == ("op", "("):
+            self.next()
+            e = self.scalar_expr()
+            self.expect("op", ")")
+            return e
+        if (k, v) == ("op", "-"):
+            self.next()
+            inner = self.scalar_factor()
+            return Scalar("bin", op="-", left=Scalar("num", value=0.0),
+                          right=inner)
+        if k == "num":
+            self.next()
+            return Scalar("num", value=float(v))
+        if k == "id":
+            return Scalar("col", name=self.qualified_id())
+        raise SqlError(f"bad scalar expression near {v!r}")
+
+    def _maybe_scalar(self) -> Tuple[str, Optional[Scalar]]:
+        """Parse a scalar expression; pure column refs collapse to a
+        plain name (backward-compatible fast path)."""
+        e = self.scalar_expr()
+        if e.kind == "col":
+            return e.name, None
+        return "", e
+
     def select_item(self) -> SelectItem:
         k, v = self.peek()
         if (k, v) == ("op", "*"):
@@ -438,15 +513,17 @@ class _Parser:
             if self.accept("op", "*"):
                 if fn != "count":
                     raise SqlError(f"{fn}(*) not supported")
-                arg = ""
+                arg, expr = "", None
             else:
-                arg = self.qualified_id()
+                arg, expr = self._maybe_scalar()
             self.expect("op", ")")
             if distinct and fn != "count":
                 raise SqlError("DISTINCT only supported inside count()")
-            item = SelectItem("agg", name=arg, fn=fn, distinct=distinct)
-        elif k == "id":
-            item = SelectItem("col", name=self.qualified_id())
+            item = SelectItem("agg", name=arg, fn=fn, distinct=distinct,
+                              expr=expr)
+        elif k in ("id", "num") or (k, v) in (("op", "("), ("op", "-")):
+            arg, expr = self._maybe_scalar()
+            item = SelectItem("col", name=arg, expr=expr)
         else:
             raise SqlError(f"bad select item near {v!r}")
         if self.accept("kw", "as"):
@@ -487,6 +564,12 @@ class _Parser:
 
     def literal(self):
         k, v = self.next()
+        if (k, v) == ("op", "-"):
+            k2, v2 = self.next()
+            if k2 != "num":
+                raise SqlError(f"expected number after '-', got {v2!r}")
+            val = float(v2) if ("." in v2 or "e" in v2.lower()) else int(v2)
+            return -val
         if k == "num":
             return float(v) if ("." in v or "e" in v.lower()) else int(v)
         if k == "str":
@@ -790,10 +873,15 @@ def _execute_join_select(catalog, q: Query, device=None):
     # referenced columns per side
     refs = []
     star = any(it.kind == "star" for it in q.items)
+    out_aliases = {it.out_name for it in q.items} | {
+        it.alias for it in q.items if it.alias}
     for it in q.items:
         if it.kind in ("col", "agg") and it.name:
             refs.append(it.name)
-    refs += q.group_by + [n for n, _ in q.order_by]
+        if it.expr is not None:
+            refs += list(it.expr.columns())
+    refs += q.group_by + [n for n, _ in q.order_by
+                          if n not in out_aliases]
     if q.where is not None:
         refs += list(q.where.columns())
     for lcol, rcol in j.on:
@@ -867,6 +955,73 @@ def _execute_join_select(catalog, q: Query, device=None):
         all_cols=list(joined.schema.names()))
 
 
+def _eval_scalar_tensor(e: Scalar, batch, col):
+    """Scalar expression -> (values (n,) float64 tensor, validity or
+    None). Decimal columns are scaled to their logical value; nulls
+    propagate through arithmetic."""
+    import torch as _torch
+
+    n = batch.num_rows
+    if e.kind == "num":
+        return (_torch.full((n,), float(e.value), dtype=_torch.float64), None)
+    if e.kind == "col":
+        src = col(e.name)
+        c = batch.columns[src]
+        f = batch.schema.field(src)
+        if c.is_string:
+            raise SqlError(f"arithmetic on string column {e.name!r}")
+        v = c.data.to(_torch.float64)
+        if f.dtype.startswith("decimal"):
+            from .io.schema import decimal_params
+
+            _, sc = decimal_params(f.dtype)
+            v = v / (10 ** sc)
+        return v, c.validity
+    l, lv = _eval_scalar_tensor(e.left, batch, col)
+    r, rv = _eval_scalar_tensor(e.right, batch, col)
+    if e.op == "+":
+        out = l + r
+    elif e.op == "-":
+        out = l - r
+    elif e.op == "*":
+        out = l * r
+    elif e.op == "/":
+        out = l / r
+    else:
+        raise SqlError(f"unknown operator {e.op!r}")
+    if lv is None and rv is None:
+        return out, None
+    import torch as _torch
+
+    m = _torch.ones(len(out), dtype=_torch.bool, device=out.device)
+    if lv is not None:
+        m &= lv.to(_torch.bool)
+    if rv is not None:
+        m &= rv.to(_torch.bool)
+    return out, m.to(_torch.uint8)
+
+
+def _eval_scalar_pd(e: Scalar, df, col):
+    if e.kind == "num":
+        return e.value
+    if e.kind == "col":
+        import pandas as _pd
+
+        s = df[col(e.name)]
+        if len(s) and isinstance(s.iloc[0], __import__("decimal").Decimal):
+            s = s.astype(float)
+        return s
+    l = _eval_scalar_pd(e.left, df, col)
+    r = _eval_scalar_pd(e.right, df, col)
+    if e.op == "+":
+        return l + r
+    if e.op == "-":
+        return l - r
+    if e.op == "*":
+        return l * r
+    return l / r
+
+
 def _use_pandas_exec() -> bool:
     """Escape hatch: LAKESOUL_SQL_PANDAS=1 selects the legacy pandas
     execution (kept as the cross-check oracle; the default is the tensor
@@ -904,13 +1059,36 @@ def _project_and_finish_tensor(q: Query, batch, col, all_cols):
     from .io.schema import Schema as _S
     from .query.engine import distinct_indices, groupby_agg, sort_indices
 
+    # materialize computed scalar expressions as extra float64 columns so
+    # the aggregation/projection below sees plain columns
+    expr_names = {}
+    expr_i = 0
+    for it in q.items:
+        if it.expr is not None:
+            key = id(it)
+            name = f"__sx{expr_i}"
+            expr_i += 1
+            vals, validity = _eval_scalar_tensor(it.expr, batch, col)
+            from .io.batch import Column as _C
+
+            nf = list(batch.schema.fields) + [_F(name, "float64", True)]
+            nc = dict(batch.columns)
+            nc[name] = _C("float64", data=vals, validity=validity)
+            batch = _B(_S(nf), nc)
+            expr_names[key] = name
+
+    def _arg(it):
+        if it.expr is not None:
+            return expr_names[id(it)]
+        return col(it.name) if it.name else None
+
     has_agg = any(it.kind == "agg" for it in q.items)
     if has_agg or q.group_by:
         gcols = [col(g) for g in q.group_by]
         aggs = []
         out_order = []
         for it in q.items:
-            if it.kind == "col":
+            if it.kind == "col" and it.expr is None:
                 if it.name not in q.group_by:
                     raise SqlError(f"column {it.name!r} must appear in GROUP BY")
                 out_order.append(("group", col(it.name),
@@ -918,8 +1096,10 @@ def _project_and_finish_tensor(q: Query, batch, col, all_cols):
                 continue
             if it.kind == "star":
                 raise SqlError("SELECT * with aggregates is not valid")
-            aggs.append((it.fn, col(it.name) if it.name else None,
-                         it.out_name, it.distinct))
+            if it.kind == "col":
+                raise SqlError("computed columns must be aggregated "
+                               "when GROUP BY is present")
+            aggs.append((it.fn, _arg(it), it.out_name, it.distinct))
             out_order.append(("agg", it.out_name, it.out_name))
         res = groupby_agg(batch, gcols, aggs)
         fields, cols = [], {}
@@ -937,6 +1117,12 @@ def _project_and_finish_tensor(q: Query, batch, col, all_cols):
                         fields.append(_F(c, batch.schema.field(c).dtype,
                                          batch.schema.field(c).nullable))
                         cols[c] = batch.columns[c]
+            elif it.expr is not None:
+                outn = it.out_name
+                if outn not in cols:
+                    src = expr_names[id(it)]
+                    fields.append(_F(outn, "float64", True))
+                    cols[outn] = batch.columns[src]
             else:
                 src = col(it.name)
                 outn = it.alias or it.name.split(".")[-1]
@@ -994,7 +1180,10 @@ def _project_and_finish(q: Query, df, col, all_cols):
                     continue
                 if it.kind == "star":
                     raise SqlError("SELECT * with aggregates is not valid")
-                s = sub[col(it.name)] if it.name else None
+                if it.expr is not None:
+                    s = _eval_scalar_pd(it.expr, sub, col)
+                else:
+                    s = sub[col(it.name)] if it.name else None
                 if it.fn == "count":
                     if it.distinct and s is not None:
                         row[it.out_name] = int(s.dropna().nunique())
@@ -1035,11 +1224,14 @@ def _project_and_finish(q: Query, df, col, all_cols):
             out = pd.DataFrame({k: [v] for k, v in agg_row(df).items()})
     else:
         cols, ren = [], {}
+        computed = {}
         for it in q.items:
             if it.kind == "star":
                 for c in all_cols:
                     if c not in cols:
                         cols.append(c)
+            elif it.expr is not None:
+                computed[it.out_name] = _eval_scalar_pd(it.expr, df, col)
             else:
                 dfc = col(it.name)
                 cols.append(dfc)
@@ -1047,6 +1239,8 @@ def _project_and_finish(q: Query, df, col, all_cols):
                 if dfc != outn:
                     ren[dfc] = outn
         out = df[cols].rename(columns=ren)
+        for name, s in computed.items():
+            out[name] = s
         if q.distinct:
             out = out.drop_duplicates().reset_index(drop=True)
 
@@ -1086,9 +1280,19 @@ def _strip_quals(q: Query, valid_quals) -> None:
             raise SqlError(f"unknown table qualifier {qual!r}")
         return name
 
+    def strip_scalar(e):
+        if e is None:
+            return
+        if e.kind == "col":
+            e.name = strip(e.name)
+        elif e.kind == "bin":
+            strip_scalar(e.left)
+            strip_scalar(e.right)
+
     for it in q.items:
         if it.kind in ("col", "agg") and it.name:
             it.name = strip(it.name)
+        strip_scalar(it.expr)
     q.group_by = [strip(g) for g in q.group_by]
     q.order_by = [(strip(n) if "." in n else n, d) for n, d in q.order_by]
 
@@ -1181,6 +1385,8 @@ def _execute_select(catalog, q: Query, device=None):
     for it in q.items:
         if it.kind in ("col", "agg") and it.name:
             need.add(it.name)
+        if it.expr is not None:
+            need |= it.expr.columns()
     for name, _ in q.order_by:
         if name in schema_cols:
             need.add(name)

@@ -365,6 +365,100 @@ def test_create_and_drop_table_sql(catalog):
     execute_sql(catalog, "DROP TABLE IF EXISTS sales")  # no error
 
 
+def test_arithmetic_expressions(sql_table):
+    """Scalar arithmetic in SELECT / aggregates (TPC-H q1/q6 shapes)."""
+    cat, t = sql_table
+    ref = _ref_df(t)
+    # q6 shape: sum of a product under filters
+    df = execute_sql(cat, "SELECT sum(price * qty) rev FROM orders "
+                          "WHERE qty >= 3 AND price < 50")
+    m = (ref["qty"] >= 3) & (ref["price"] < 50)
+    assert abs(df["rev"].iloc[0] - (ref[m]["price"] * ref[m]["qty"]).sum()) < 1e-6
+    # q1 shape: grouped sums of compound expressions
+    df2 = execute_sql(cat, "SELECT region, sum(price * (1 - qty / 100.0)) s1, "
+                           "avg(price * 2) a2, count(*) n FROM orders "
+                           "GROUP BY region ORDER BY region")
+    g = ref.assign(s1=ref["price"] * (1 - ref["qty"] / 100.0),
+                   a2=ref["price"] * 2).groupby("region")
+    exp = g.agg(s1=("s1", "sum"), a2=("a2", "mean"), n=("region", "size")).reset_index()
+    assert df2["region"].tolist() == sorted(exp["region"].tolist())
+    np.testing.assert_allclose(df2["s1"].to_numpy(),
+                               exp.sort_values("region")["s1"].to_numpy())
+    np.testing.assert_allclose(df2["a2"].to_numpy(),
+                               exp.sort_values("region")["a2"].to_numpy())
+    # computed projection (no aggregates)
+    df3 = execute_sql(cat, "SELECT id, price * qty AS total FROM orders "
+                           "WHERE id < 5 ORDER BY id")
+    sub = ref[ref["id"] < 5]
+    np.testing.assert_allclose(df3["total"].to_numpy(),
+                               (sub["price"] * sub["qty"]).to_numpy())
+    # unary minus and negative literals
+    df4 = execute_sql(cat, "SELECT sum(-price) s FROM orders WHERE id = 1")
+    assert abs(df4["s"].iloc[0] + ref[ref["id"] == 1]["price"].iloc[0]) < 1e-9
+
+
+def test_tensor_vs_pandas_exec_equivalence(sql_table, monkeypatch):
+    """Every query class produces identical results on the tensor engine
+    and the legacy pandas path (LAKESOUL_SQL_PANDAS=1)."""
+    cat, t = sql_table
+    queries = [
+        "SELECT region, count(*) n, sum(price) s, avg(qty) a FROM orders "
+        "GROUP BY region ORDER BY region",
+        "SELECT count(DISTINCT region) d FROM orders",
+        "SELECT DISTINCT region FROM orders ORDER BY region",
+        "SELECT id, price FROM orders WHERE qty > 5 ORDER BY price DESC, id LIMIT 7",
+        "SELECT region, max(price) mx, min(qty) mn FROM orders "
+        "GROUP BY region HAVING mx > 50 ORDER BY region",
+        "SELECT sum(price * qty) v FROM orders",
+    ]
+    for sql in queries:
+        monkeypatch.delenv("LAKESOUL_SQL_PANDAS", raising=False)
+        a = execute_sql(cat, sql)
+        monkeypatch.setenv("LAKESOUL_SQL_PANDAS", "1")
+        b = execute_sql(cat, sql)
+        monkeypatch.delenv("LAKESOUL_SQL_PANDAS", raising=False)
+        assert list(a.columns) == list(b.columns), sql
+        assert len(a) == len(b), sql
+        for c in a.columns:
+            av, bv = a[c].to_numpy(), b[c].to_numpy()
+            if av.dtype.kind in "fc" or bv.dtype.kind in "fc":
+                np.testing.assert_allclose(
+                    av.astype(float), bv.astype(float), rtol=1e-9,
+                    err_msg=f"{sql} :: {c}")
+            else:
+                np.testing.assert_array_equal(av, bv, err_msg=f"{sql} :: {c}")
+
+
+def test_join_on_tensor_engine_matches_pandas(catalog, monkeypatch):
+    execute_sql(catalog,
+        "CREATE TABLE cust (cid BIGINT NOT NULL, name VARCHAR(16)) "
+        "PRIMARY KEY (cid) HASH BUCKETS 2")
+    execute_sql(catalog,
+        "CREATE TABLE ords (oid BIGINT NOT NULL, cid BIGINT, amt DOUBLE) "
+        "PRIMARY KEY (oid) HASH BUCKETS 2")
+    execute_sql(catalog,
+        "INSERT INTO cust VALUES (1, 'ann'), (2, 'bob'), (3, 'cyd')")
+    execute_sql(catalog,
+        "INSERT INTO ords VALUES (10, 1, 5.0), (11, 1, 7.0), (12, 2, 3.0), "
+        "(13, NULL, 9.0)")
+    sql = ("SELECT c.name, sum(o.amt) total FROM ords o "
+           "JOIN cust c ON o.cid = c.cid GROUP BY c.name ORDER BY c.name")
+    a = execute_sql(catalog, sql)
+    monkeypatch.setenv("LAKESOUL_SQL_PANDAS", "1")
+    b = execute_sql(catalog, sql)
+    monkeypatch.delenv("LAKESOUL_SQL_PANDAS", raising=False)
+    assert a["name"].tolist() == b["name"].tolist() == ["ann", "bob"]
+    np.testing.assert_allclose(a["total"].to_numpy(), b["total"].to_numpy())
+    assert a["total"].tolist() == [12.0, 3.0]
+    # left join keeps the null-cid order with null name
+    sql2 = ("SELECT o.oid, c.name FROM ords o LEFT JOIN cust c "
+            "ON o.cid = c.cid ORDER BY o.oid")
+    a2 = execute_sql(catalog, sql2)
+    assert a2["oid"].tolist() == [10, 11, 12, 13]
+    assert a2["name"].tolist()[:3] == ["ann", "ann", "bob"]
+    assert a2["name"].iloc[3] is None or pd.isna(a2["name"].iloc[3])
+
+
 def test_insert_null_values_and_decimal(catalog):
     """NULL in INSERT VALUES and INSERT..SELECT from a nullable decimal
     column (Decimal(str(None)) crash + literal() NULL)."""
