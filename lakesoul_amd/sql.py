@@ -963,7 +963,14 @@ def _eval_scalar_tensor(e: Scalar, batch, col):
 
     n = batch.num_rows
     if e.kind == "num":
-        return (_torch.full((n,), float(e.value), dtype=_torch.float64), None)
+        dev = None
+        for c in batch.columns.values():
+            t = c.offsets if c.is_string else c.data
+            if t is not None:
+                dev = t.device
+                break
+        return (_torch.full((n,), float(e.value), dtype=_torch.float64,
+                            device=dev or "cpu"), None)
     if e.kind == "col":
         src = col(e.name)
         c = batch.columns[src]
