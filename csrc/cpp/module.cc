@@ -124,7 +124,11 @@ static int64_t write_parquet(
   std::vector<torch::Tensor> keep;  // hold contiguous refs
 
   for (size_t i = 0; i < ncol; i++) {
-    DtypeInfo di = dtype_info(dtypes[i]);
+    bool is_list = dtypes[i].rfind("list<", 0) == 0;
+    std::string scalar_dtype = is_list
+        ? dtypes[i].substr(5, dtypes[i].size() - 6)
+        : dtypes[i];
+    DtypeInfo di = dtype_info(scalar_dtype);
     descs[i].name = names[i];
     descs[i].physical = di.physical;
     descs[i].converted = di.converted;
@@ -133,10 +137,27 @@ static int64_t write_parquet(
     descs[i].dec_precision = di.dec_precision;
     descs[i].dec_scale = di.dec_scale;
     descs[i].nullable = nullable[i];
+    descs[i].is_list = is_list;
+    if (is_list) {
+      descs[i].max_def = 2;  // optional list group + repeated, required elem
+    }
 
     torch::Tensor col = columns[i].contiguous().cpu();
     keep.push_back(col);
-    if (di.physical == PT_BYTE_ARRAY) {
+    if (is_list) {
+      TORCH_CHECK(di.physical != PT_BYTE_ARRAY,
+                  "list<string> write not supported");
+      TORCH_CHECK(offsets[i].has_value(), "list column needs offsets");
+      torch::Tensor off = offsets[i]->contiguous().cpu().to(torch::kInt64);
+      keep.push_back(off);
+      data[i].list_offsets = off.data_ptr<int64_t>();
+      data[i].data = (const uint8_t*)col.data_ptr();
+      int es = physical_elem_size(di.physical);
+      TORCH_CHECK(col.element_size() == es, "list element size mismatch");
+      int64_t n = off.numel() - 1;
+      TORCH_CHECK(num_rows < 0 || n == num_rows, "row count mismatch");
+      num_rows = n;
+    } else if (di.physical == PT_BYTE_ARRAY) {
       TORCH_CHECK(offsets[i].has_value(), "byte_array column needs offsets");
       torch::Tensor off = offsets[i]->contiguous().cpu().to(torch::kInt32);
       keep.push_back(off);

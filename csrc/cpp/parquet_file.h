@@ -112,6 +112,9 @@ struct ColumnData {
   const int32_t* offsets = nullptr;
   const uint8_t* bytes = nullptr;
   const uint8_t* validity = nullptr;  // byte per row, 1=valid; null=all valid
+  // LIST columns: `data` holds the ELEMENT values; list_offsets (n+1,
+  // row -> element range) defines rows; validity is per ROW (null list)
+  const int64_t* list_offsets = nullptr;
 };
 
 class ParquetWriter {
@@ -154,6 +157,34 @@ class ParquetWriter {
     root.num_children = (int32_t)cols_.size();
     fm.schema.push_back(root);
     for (auto& c : cols_) {
+      if (c.is_list) {
+        // standard 3-level LIST: optional group (LIST) { repeated group
+        // list { required <element>; } } — mirrors the reader's
+        // flatten_element LIST shape
+        SchemaElement g;
+        g.name = c.name;
+        g.type = -1;
+        g.repetition = REP_OPTIONAL;
+        g.converted = 3;  // LIST
+        g.num_children = 1;
+        fm.schema.push_back(g);
+        SchemaElement mid;
+        mid.name = "list";
+        mid.type = -1;
+        mid.repetition = REP_REPEATED;
+        mid.num_children = 1;
+        fm.schema.push_back(mid);
+        SchemaElement leaf;
+        leaf.name = "element";
+        leaf.type = c.physical;
+        leaf.repetition = REP_REQUIRED;
+        leaf.converted = c.converted;
+        leaf.logical = c.logical;
+        leaf.int_bit_width = c.int_bit_width;
+        leaf.int_signed = c.int_signed;
+        fm.schema.push_back(leaf);
+        continue;
+      }
       SchemaElement e;
       e.name = c.name;
       e.type = c.physical;
@@ -193,6 +224,10 @@ class ParquetWriter {
     for (size_t ci = 0; ci < cols_.size(); ci++) {
       const ColumnDesc& cd = cols_[ci];
       const ColumnData& col = data[ci];
+      if (cd.is_list) {
+        write_list_chunk(cd, col, row_off, n, rg);
+        continue;
+      }
 
       // page split: target ~LAKESOUL_PAGE_BYTES decompressed bytes per
       // page. Default 128 KB: measured best for the host zstd decode
@@ -319,6 +354,91 @@ class ParquetWriter {
     }
     s.min_value.assign((char*)&mn, sizeof(T));
     s.max_value.assign((char*)&mx, sizeof(T));
+  }
+
+  // One-page LIST chunk (3-level standard layout, v1 page:
+  // [u32][rep RLE][u32][def RLE][PLAIN element values]) — the exact
+  // shape the reader's list path decodes (read side ~line 655).
+  void write_list_chunk(const ColumnDesc& cd, const ColumnData& col,
+                        int64_t row_off, int64_t n, RowGroup& rg) {
+    if (cd.physical == PT_BYTE_ARRAY)
+      throw std::runtime_error("list<string> write not supported");
+    if (!col.list_offsets) throw std::runtime_error("list column without offsets");
+    const int64_t* offs = col.list_offsets;
+    std::vector<uint8_t> reps, defs;
+    int64_t lo = offs[row_off], hi = offs[row_off + n];
+    reps.reserve((size_t)(hi - lo + n));
+    defs.reserve((size_t)(hi - lo + n));
+    // optional list group + repeated: max_def = 2, elements required
+    for (int64_t i = 0; i < n; i++) {
+      bool valid = !col.validity || col.validity[row_off + i];
+      int64_t len = offs[row_off + i + 1] - offs[row_off + i];
+      if (!valid) {
+        reps.push_back(0);
+        defs.push_back(0);
+      } else if (len == 0) {
+        reps.push_back(0);
+        defs.push_back(1);
+      } else {
+        for (int64_t j = 0; j < len; j++) {
+          reps.push_back(j == 0 ? 0 : 1);
+          defs.push_back(2);
+        }
+      }
+    }
+    int64_t entries = (int64_t)reps.size();
+    auto rep_rle = encode_levels(reps.data(), entries);
+    auto def_rle = encode_levels(defs.data(), entries);
+
+    std::vector<uint8_t> payload;
+    uint32_t rl = (uint32_t)rep_rle.size(), dl = (uint32_t)def_rle.size();
+    payload.insert(payload.end(), (uint8_t*)&rl, (uint8_t*)&rl + 4);
+    payload.insert(payload.end(), rep_rle.begin(), rep_rle.end());
+    payload.insert(payload.end(), (uint8_t*)&dl, (uint8_t*)&dl + 4);
+    payload.insert(payload.end(), def_rle.begin(), def_rle.end());
+    int es = physical_elem_size(cd.physical);
+    payload.insert(payload.end(), col.data + lo * es, col.data + hi * es);
+
+    std::vector<uint8_t> compressed;
+    const uint8_t* body = payload.data();
+    size_t body_n = payload.size();
+    if (codec_ == CODEC_ZSTD) {
+      compressed = zstd_compress(payload.data(), payload.size(), level_);
+      body = compressed.data();
+      body_n = compressed.size();
+    } else if (codec_ != CODEC_UNCOMPRESSED) {
+      throw std::runtime_error("writer supports zstd/uncompressed only");
+    }
+
+    ColumnMeta cm;
+    cm.type = cd.physical;
+    cm.encodings = {ENC_PLAIN, ENC_RLE};
+    cm.path_in_schema = {cd.name, "list", "element"};
+    cm.codec = codec_;
+    cm.num_values = entries;
+    cm.data_page_offset = pos_;
+
+    PageHeader ph;
+    ph.type = PAGE_DATA;
+    ph.num_values = (int32_t)entries;
+    ph.encoding = ENC_PLAIN;
+    ph.def_encoding = ENC_RLE;
+    ph.rep_encoding = ENC_RLE;
+    ph.uncompressed_size = (int32_t)payload.size();
+    ph.compressed_size = (int32_t)body_n;
+    auto ph_bytes = serialize_page_header(ph);
+    cm.total_uncompressed_size = (int64_t)(ph_bytes.size() + payload.size());
+    cm.total_compressed_size = (int64_t)(ph_bytes.size() + body_n);
+    fwrite_all(ph_bytes.data(), ph_bytes.size());
+    fwrite_all(body, body_n);
+    Statistics stats;
+    int64_t nulls = 0;
+    if (col.validity)
+      for (int64_t i = 0; i < n; i++) nulls += col.validity[row_off + i] ? 0 : 1;
+    stats.null_count = nulls;
+    cm.stats = stats;
+    rg.columns.push_back(cm);
+    rg.total_byte_size += cm.total_uncompressed_size;
   }
 
   void append_plain_values(const ColumnDesc& cd, const ColumnData& col,

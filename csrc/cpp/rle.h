@@ -153,4 +153,28 @@ inline std::vector<uint8_t> encode_def_levels(const uint8_t* validity, int64_t n
   return out;
 }
 
+// Encode arbitrary small-int levels (values < 256) as pure RLE repeat
+// runs — always a legal RLE/bit-packed hybrid stream regardless of the
+// decoder's bit width (repeat-run values are stored byte-wide for
+// bit_width <= 8). Used for LIST rep/def levels.
+inline std::vector<uint8_t> encode_levels(const uint8_t* levels, int64_t n) {
+  std::vector<uint8_t> out;
+  auto put_varint = [&](uint64_t v) {
+    while (v >= 0x80) {
+      out.push_back((uint8_t)(v | 0x80));
+      v >>= 7;
+    }
+    out.push_back((uint8_t)v);
+  };
+  int64_t i = 0;
+  while (i < n) {
+    int64_t j = i;
+    while (j < n && levels[j] == levels[i]) j++;
+    put_varint(((uint64_t)(j - i)) << 1);
+    out.push_back(levels[i]);
+    i = j;
+  }
+  return out;
+}
+
 }  // namespace lakesoul

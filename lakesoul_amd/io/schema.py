@@ -91,6 +91,15 @@ def decimal_params(dtype: str):
     return (int(m.group(1)), int(m.group(2))) if m else None
 
 
+def list_element_dtype(dtype: str):
+    """Return the element dtype of a list dtype string, else None."""
+    s = dtype.strip().lower()
+    for prefix in ("list<", "array<"):
+        if s.startswith(prefix) and s.endswith(">"):
+            return s[len(prefix):-1].strip()
+    return None
+
+
 def canonical_dtype(dt: str) -> str:
     dp = decimal_params(dt)
     if dp is not None:
@@ -99,6 +108,14 @@ def canonical_dtype(dt: str) -> str:
             raise TypeError(
                 f"unsupported decimal precision/scale {dt} (int64-backed: p<=18)")
         return f"decimal({p},{sc})"
+    elem = list_element_dtype(dt)
+    if elem is not None:
+        inner = canonical_dtype(elem)
+        if inner not in FIXED_WIDTH_BYTES:
+            raise TypeError(
+                f"unsupported list element dtype {elem!r} "
+                "(fixed-width primitives only)")
+        return f"list<{inner}>"
     try:
         return _CANONICAL[dt.lower()]
     except KeyError:
@@ -199,6 +216,9 @@ def _arrow_type_to_dtype(t) -> str:
         return f"timestamp[{t.unit}]"
     if pt.is_decimal(t):
         return canonical_dtype(f"decimal({t.precision},{t.scale})")
+    if pt.is_list(t) or pt.is_large_list(t):
+        inner = _arrow_type_to_dtype(t.value_type)
+        return canonical_dtype(f"list<{inner}>")
     raise TypeError(f"unsupported arrow type {t}")
 
 
@@ -224,6 +244,9 @@ def dtype_to_arrow(dtype: str):
     dp = decimal_params(dtype)
     if dp is not None:
         return pa.decimal128(*dp)
+    elem = list_element_dtype(dtype)
+    if elem is not None:
+        return pa.list_(dtype_to_arrow(elem))
     return mapping[dtype]
 
 
@@ -235,6 +258,23 @@ def schema_to_arrow(schema: Schema):
     )
 
 
+def _dtype_to_spark(dtype: str):
+    elem = list_element_dtype(dtype)
+    if elem is not None:
+        # spark array type JSON (reference ser/arrow_java.rs array handling)
+        return {"type": "array",
+                "elementType": _TO_SPARK.get(elem, elem),
+                "containsNull": False}
+    return _TO_SPARK.get(dtype, dtype)
+
+
+def _dtype_from_spark(t):
+    if isinstance(t, dict) and t.get("type") == "array":
+        et = t.get("elementType")
+        return f"list<{_FROM_SPARK.get(et, et)}>"
+    return _FROM_SPARK.get(t, t)
+
+
 def schema_to_json(schema: Schema) -> str:
     return json.dumps(
         {
@@ -242,7 +282,7 @@ def schema_to_json(schema: Schema) -> str:
             "fields": [
                 {
                     "name": f.name,
-                    "type": _TO_SPARK.get(f.dtype, f.dtype),
+                    "type": _dtype_to_spark(f.dtype),
                     "nullable": f.nullable,
                     "metadata": {},
                 }
@@ -256,7 +296,6 @@ def schema_from_json(s: str) -> Schema:
     d = json.loads(s)
     fields = []
     for f in d.get("fields", []):
-        t = f["type"]
-        dtype = _FROM_SPARK.get(t, t)
-        fields.append(Field(f["name"], dtype, bool(f.get("nullable", True))))
+        fields.append(Field(f["name"], _dtype_from_spark(f["type"]),
+                            bool(f.get("nullable", True))))
     return Schema(fields)

@@ -171,7 +171,40 @@ def groupby_agg(batch: Batch, group_cols: Sequence[str],
         out_fields.append(Field(f.name, f.dtype, f.nullable))
         out_cols[c] = batch.columns[c].take(rep)
 
-    counts_all = torch.bincount(codes, minlength=g)
+    # few-group aggregations on GPU: scatter_reduce into a handful of
+    # slots serializes on atomics (every element CASes the same address);
+    # per-group masked reductions are g clean passes instead
+    small_g = (g <= 128 and g > 0 and codes.device.type == "cuda")
+    group_masks = (codes == torch.arange(g, device=codes.device)[:, None]
+                   ) if small_g else None  # (g, n) bool
+
+    def seg_count(valid_mask):
+        if small_g:
+            return (group_masks & valid_mask[None, :]).sum(1)
+        return torch.bincount(codes[valid_mask], minlength=g)
+
+    def seg_sum(v64, valid_mask, zero):
+        if small_g:
+            sel = torch.where(valid_mask, v64, zero)
+            return (group_masks.to(v64.dtype) @ sel)
+        s = torch.zeros(g, dtype=v64.dtype, device=codes.device)
+        s.index_add_(0, codes, torch.where(valid_mask, v64, zero))
+        return s
+
+    def seg_minmax(v64, valid_mask, sent, is_min):
+        sel = torch.where(valid_mask, v64, torch.full_like(v64, sent))
+        if small_g:
+            big = torch.where(group_masks, sel[None, :],
+                              torch.full((1, 1), sent, dtype=v64.dtype,
+                                         device=v64.device))
+            return big.amin(1) if is_min else big.amax(1)
+        red = torch.full((g,), sent, dtype=v64.dtype, device=codes.device)
+        red.scatter_reduce_(0, codes, sel, reduce="amin" if is_min else "amax",
+                            include_self=True)
+        return red
+
+    counts_all = (group_masks.sum(1) if small_g
+                  else torch.bincount(codes, minlength=g))
     for fn, cname, out_name, distinct in aggs:
         if fn == "count" and cname is None:
             out_fields.append(Field(out_name, "int64", False))
@@ -188,7 +221,7 @@ def groupby_agg(batch: Batch, group_cols: Sequence[str],
                 gsz = max(int(vc.max().item()) + 1 if n else 1, 1)
                 cnt = torch.bincount(upair_codes // gsz, minlength=g)
             else:
-                cnt = torch.bincount(codes[valid], minlength=g)
+                cnt = seg_count(valid)
             out_fields.append(Field(out_name, "int64", False))
             out_cols[out_name] = Column("int64", data=cnt.to(torch.int64))
             continue
@@ -201,12 +234,7 @@ def groupby_agg(batch: Batch, group_cols: Sequence[str],
                 ranks[order] = torch.arange(n, dtype=torch.int64,
                                             device=order.device)
                 sent = n if fn == "min" else -1
-                red = torch.full((g,), sent, dtype=torch.int64,
-                                 device=codes.device)
-                sel_ranks = torch.where(valid, ranks, torch.full_like(ranks, sent))
-                red.scatter_reduce_(0, codes, sel_ranks,
-                                    reduce="amin" if fn == "min" else "amax",
-                                    include_self=True)
+                red = seg_minmax(ranks, valid, sent, fn == "min")
                 has = (red != sent)
                 src_rows = torch.zeros(g, dtype=torch.int64, device=codes.device)
                 src_rows[has] = order[red[has]]  # rank r -> row order[r]
@@ -222,10 +250,9 @@ def groupby_agg(batch: Batch, group_cols: Sequence[str],
         acc_dtype = torch.float64 if is_float else torch.int64
         v64 = vals.to(acc_dtype)
         zero = torch.zeros_like(v64)
-        vcnt = torch.bincount(codes[valid], minlength=g)
+        vcnt = seg_count(valid)
         if fn in ("sum", "avg"):
-            s = torch.zeros(g, dtype=acc_dtype, device=codes.device)
-            s.index_add_(0, codes, torch.where(valid, v64, zero))
+            s = seg_sum(v64, valid, zero)
             if fn == "avg":
                 dt = "float64"
                 res_t = (s.to(torch.float64)
@@ -256,11 +283,7 @@ def groupby_agg(batch: Batch, group_cols: Sequence[str],
             else:
                 sent = (torch.iinfo(torch.int64).max if fn == "min"
                         else torch.iinfo(torch.int64).min)
-            red = torch.full((g,), sent, dtype=acc_dtype, device=codes.device)
-            sel = torch.where(valid, v64, torch.full_like(v64, sent))
-            red.scatter_reduce_(0, codes, sel,
-                                reduce="amin" if fn == "min" else "amax",
-                                include_self=True)
+            red = seg_minmax(v64, valid, sent, fn == "min")
             fdt = batch.schema.field(cname).dtype
             tdt = torch_dtype_of(fdt)
             out_fields.append(Field(out_name, fdt, True))
