@@ -312,6 +312,17 @@ static py::dict decoded_to_dict(DecodedColumn&& dc) {
   d["offsets"] = vec_to_tensor_i32(std::move(dc.offsets));
   d["bytes"] = vec_to_tensor_u8(std::move(dc.bytes));
   d["validity"] = vec_to_tensor_u8(std::move(dc.validity));
+  if (dc.is_list) {
+    auto lo = torch::empty({(int64_t)dc.list_offsets.size()}, torch::kInt64);
+    std::memcpy(lo.data_ptr(), dc.list_offsets.data(),
+                dc.list_offsets.size() * 8);
+    d["list_offsets"] = lo;
+    auto lv = torch::empty({(int64_t)dc.list_validity.size()}, torch::kUInt8);
+    if (!dc.list_validity.empty())
+      std::memcpy(lv.data_ptr(), dc.list_validity.data(),
+                  dc.list_validity.size());
+    d["list_validity"] = lv;
+  }
   return d;
 }
 

@@ -1025,12 +1025,21 @@ struct DecodedColumn {
   std::vector<uint8_t> bytes;     // byte_array
   std::vector<uint8_t> validity;  // empty if no nulls
   int64_t num_values = 0;
+  // LIST columns: rows -> element ranges (+1 sentinel) + per-row validity
+  bool is_list = false;
+  std::vector<int64_t> list_offsets;
+  std::vector<uint8_t> list_validity;
 };
 
 inline DecodedColumn decode_chunk_cpu(const ParquetFile::ChunkData& ch) {
   DecodedColumn out;
   out.num_values = ch.num_values;
   out.validity = ch.validity;
+  if (ch.is_list) {
+    out.is_list = true;
+    out.list_offsets = ch.list_offsets;
+    out.list_validity = ch.list_validity;
+  }
   const uint8_t* validity = ch.validity.empty() ? nullptr : ch.validity.data();
 
   // 1) materialize dense (non-null) values

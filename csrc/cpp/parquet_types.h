@@ -562,7 +562,9 @@ inline std::vector<uint8_t> serialize_file_meta(const FileMetaData& fm) {
   w.field_binary(last, 6, fm.created_by);
   // column_orders: TYPE_ORDER for every leaf — required for readers to
   // trust min_value/max_value statistics (parquet-format spec).
-  size_t nleaf = fm.schema.size() > 0 ? fm.schema.size() - 1 : 0;
+  size_t nleaf = 0;
+  for (size_t i = 1; i < fm.schema.size(); i++)
+    if (fm.schema[i].num_children == 0) nleaf++;
   w.field(last, 7, CType::LIST);
   w.list_header(CType::STRUCT, (uint32_t)nleaf);
   for (size_t i = 0; i < nleaf; i++) {

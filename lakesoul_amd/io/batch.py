@@ -71,8 +71,18 @@ class Column:
     def is_string(self) -> bool:
         return self.dtype in ("string", "binary")
 
+    @property
+    def is_list(self) -> bool:
+        # list<T>: `offsets` (int64 n+1, element ranges) + `data`
+        # (element values tensor) + per-row validity
+        return self.dtype.startswith("list<")
+
+    @property
+    def elem_dtype(self) -> str:
+        return self.dtype[5:-1]
+
     def __len__(self) -> int:
-        if self.is_string:
+        if self.is_string or self.is_list:
             return int(self.offsets.numel()) - 1
         return int(self.data.numel())
 
@@ -91,9 +101,26 @@ class Column:
         """Gather rows by index tensor (moved to this column's device —
         a CPU index against a CUDA column would reach the string gather
         kernel as a host pointer)."""
-        dev = self.offsets.device if self.is_string else self.data.device
+        dev = self.offsets.device if (self.is_string or self.is_list) else self.data.device
         if idx.device != dev:
             idx = idx.to(dev)
+        if self.is_list:
+            # gather element ranges through the byte-view of the values
+            # (the string gather path, offsets scaled by element size)
+            es = self.data.element_size()
+            byte_col = Column(
+                "binary",
+                offsets=self.offsets.to(torch.int64) * es,
+                bytes_=self.data.contiguous().view(torch.uint8)
+                if self.data.numel() else torch.empty(0, dtype=torch.uint8,
+                                                      device=dev),
+                validity=self.validity)
+            got = byte_col.take(idx)
+            new_offs = got.offsets.to(torch.int64) // es
+            vals = (got.bytes_.view(self.data.dtype) if got.bytes_.numel()
+                    else torch.empty(0, dtype=self.data.dtype, device=dev))
+            return Column(self.dtype, data=vals, offsets=new_offs,
+                          validity=got.validity)
         if self.is_string:
             offs = self.offsets
             lens = offs[1:] - offs[:-1]
@@ -145,7 +172,14 @@ class Batch:
         cols = {}
         for k, c in self.columns.items():
             v = None if c.validity is None else c.validity[a:b]
-            if c.is_string:
+            if c.is_list:
+                offs = c.offsets[a : b + 1]
+                base = offs[0]
+                lo, hi = int(base), int(c.offsets[b])
+                cols[k] = Column(c.dtype, data=c.data[lo:hi],
+                                 offsets=(offs - base).to(c.offsets.dtype),
+                                 validity=v)
+            elif c.is_string:
                 offs = c.offsets[a : b + 1]
                 base = offs[0]
                 lo = int(base)
@@ -207,6 +241,33 @@ class Batch:
             if f.name not in d:
                 raise KeyError(f"missing column {f.name}")
             v = d[f.name]
+            if f.dtype.startswith("list<"):
+                elem_dt = f.dtype[5:-1]
+                npdt = np_dtype_for(elem_dt)
+                if isinstance(v, tuple) and len(v) in (2, 3):
+                    offs = torch.as_tensor(v[0], dtype=torch.int64)
+                    vals = torch.as_tensor(np.asarray(v[1], dtype=npdt))
+                    val = (torch.as_tensor(v[2], dtype=torch.uint8)
+                           if len(v) == 3 and v[2] is not None else None)
+                else:
+                    items = list(v)
+                    validity_l = None
+                    if any(x is None for x in items):
+                        validity_l = torch.tensor(
+                            [0 if x is None else 1 for x in items],
+                            dtype=torch.uint8)
+                    arrs = [np.asarray([] if x is None else x, dtype=npdt)
+                            for x in items]
+                    offs = torch.zeros(len(arrs) + 1, dtype=torch.int64)
+                    torch.cumsum(torch.tensor([len(a) for a in arrs],
+                                              dtype=torch.int64), 0,
+                                 out=offs[1:].view(-1))
+                    vals = torch.from_numpy(
+                        np.concatenate(arrs) if arrs else np.empty(0, npdt))
+                    val = validity_l
+                cols[f.name] = Column(f.dtype, data=vals.to(
+                    torch_dtype_for(elem_dt)), offsets=offs, validity=val)
+                continue
             if f.is_fixed_width:
                 if isinstance(v, torch.Tensor):
                     t = v
@@ -314,6 +375,28 @@ class Batch:
         arrays = []
         for f in self.schema:
             c = self.columns[f.name]
+            if f.dtype.startswith("list<"):
+                elem_dt = f.dtype[5:-1]
+                offs_np = c.offsets.cpu().numpy().astype(np.int32, copy=False)
+                vals = pa.array(c.data.cpu().numpy(),
+                                type=dtype_to_arrow(elem_dt))
+                n = len(c)
+                validity_buf = None
+                null_count = 0
+                if c.validity is not None:
+                    vv = c.validity.cpu().numpy().astype(bool)
+                    null_count = int(n - vv.sum())
+                    validity_buf = pa.py_buffer(
+                        np.packbits(vv, bitorder="little").tobytes())
+                arrays.append(pa.ListArray.from_arrays(
+                    pa.Array.from_buffers(pa.int32(), n + 1,
+                                          [None, pa.py_buffer(offs_np.tobytes())]),
+                    vals) if validity_buf is None else
+                    pa.Array.from_buffers(
+                        dtype_to_arrow(f.dtype), n,
+                        [validity_buf, pa.py_buffer(offs_np.tobytes())],
+                        null_count=null_count, children=[vals]))
+                continue
             if f.dtype.startswith("decimal"):
                 v = c.data.cpu().numpy().astype(np.int64)
                 n = v.size
@@ -370,6 +453,24 @@ def concat_batches(batches: List[Batch]) -> Batch:
     for f in schema:
         cs = [b.columns[f.name] for b in batches]
         n_total = sum(len(c) for c in cs)
+        if cs and cs[0].is_list:
+            data = torch.cat([c.data for c in cs])
+            offs = torch.zeros(n_total + 1, dtype=torch.int64, device=data.device)
+            pos, base = 0, 0
+            for c in cs:
+                n = len(c)
+                offs[pos + 1: pos + n + 1] = c.offsets[1:].to(torch.int64) + base
+                base += int(c.offsets[-1])
+                pos += n
+            validity = None
+            if any(c.validity is not None for c in cs):
+                validity = torch.cat([
+                    c.validity if c.validity is not None
+                    else torch.ones(len(c), dtype=torch.uint8, device=data.device)
+                    for c in cs])
+            cols[f.name] = Column(f.dtype, data=data, offsets=offs,
+                                  validity=validity)
+            continue
         if f.is_fixed_width:
             data = torch.cat([c.data for c in cs])
             validity = None

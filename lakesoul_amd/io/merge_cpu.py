@@ -34,7 +34,9 @@ class NpColumn:
 
     @property
     def is_string(self) -> bool:
-        return self.dtype in ("string", "binary")
+        # list<T> columns ride the byte-string machinery (offsets in
+        # bytes over the raw element buffer; whole-value UseLast)
+        return self.dtype in ("string", "binary") or self.dtype.startswith("list<")
 
     def __len__(self):
         return len(self.offsets) - 1 if self.is_string else len(self.data)

@@ -495,9 +495,14 @@ class LakeSoulScan:
 
                 warnings.warn(f"chunked merge unavailable ({e}); "
                               "falling back")
+        has_list = any(self.schema.field(n).dtype.startswith("list<")
+                       for n in self.read_cols)
         if self.device == "cuda" and (
-            oversized or (needs_merge and not self._gpu_merge_supported())
+            oversized or has_list
+            or (needs_merge and not self._gpu_merge_supported())
         ):
+            # (list<T> columns decode on host for now; the merged batch
+            # ships to HBM — GPU list decode is a follow-up)
             # hybrid: CPU decode+merge (host RAM), then ship the merged
             # batch to HBM — for string-PK merges and for buckets whose
             # working set would not fit the HBM budget (the merged output
@@ -653,7 +658,18 @@ class LakeSoulScan:
                 cols[f.name] = self._range_value_column(f, unit, merged)
                 continue
             npc = merged[f.name]
-            if npc.is_string:
+            if f.dtype.startswith("list<"):
+                es = np.dtype(_np_phys(f.dtype[5:-1])).itemsize
+                offs = np.ascontiguousarray(npc.offsets, dtype=np.int64) // es
+                vals = np.ascontiguousarray(npc.bytes_).view(
+                    _np_phys(f.dtype[5:-1]))
+                cols[f.name] = Column(
+                    f.dtype,
+                    data=torch.from_numpy(vals.copy()),
+                    offsets=torch.from_numpy(offs),
+                    validity=None if npc.validity is None else torch.from_numpy(npc.validity),
+                )
+            elif npc.is_string:
                 cols[f.name] = Column(
                     f.dtype,
                     offsets=torch.from_numpy(np.ascontiguousarray(npc.offsets, dtype=np.int32)),
@@ -715,6 +731,39 @@ class LakeSoulScan:
             ci = 0
             for name in req:
                 f = self.schema.field(name)
+                if f.dtype.startswith("list<"):
+                    # list<T> rides the byte-string machinery downstream:
+                    # offsets in BYTES over the raw element buffer (merge/
+                    # gather treat the whole list value opaquely — UseLast
+                    # whole-value semantics, reference merge/mod.rs:65-89)
+                    es = np.dtype(_np_phys(f.dtype[5:-1])).itemsize
+                    offs_parts, bytes_parts, masks = [], [], []
+                    any_null = False
+                    for rg in rg_iter:
+                        d = chunks[ci]
+                        ci += 1
+                        offs_parts.append(d["list_offsets"].numpy() * es)
+                        bytes_parts.append(d["data"].numpy())
+                        lv = d["list_validity"].numpy()
+                        nrow = len(offs_parts[-1]) - 1
+                        if len(lv) and not lv.all():
+                            any_null = True
+                            masks.append(lv[:nrow])
+                        else:
+                            masks.append(np.ones(nrow, dtype=np.uint8))
+                    validity = np.concatenate(masks) if any_null else None
+                    total_rows = sum(len(o) - 1 for o in offs_parts)
+                    offs = np.zeros(total_rows + 1, dtype=np.int64)
+                    pos, base = 0, 0
+                    for o in offs_parts:
+                        nr = len(o) - 1
+                        offs[pos + 1: pos + nr + 1] = o[1:].astype(np.int64) + base
+                        base += int(o[-1]) if len(o) else 0
+                        pos += nr
+                    bys = (np.concatenate(bytes_parts) if bytes_parts
+                           else np.empty(0, np.uint8))
+                    out[name] = NpColumn(f.dtype, None, offs, bys, validity)
+                    continue
                 parts, offs_parts, bytes_parts, masks = [], [], [], []
                 any_null = False
                 for rg in rg_iter:
@@ -756,7 +805,7 @@ class LakeSoulScan:
                 if name in out:
                     continue
                 f = self.schema.field(name)
-                if f.dtype in ("string", "binary"):
+                if f.dtype in ("string", "binary") or f.dtype.startswith("list<"):
                     out[name] = NpColumn(
                         f.dtype,
                         None,

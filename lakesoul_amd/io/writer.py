@@ -228,7 +228,16 @@ def _write_batch_to_file_local(path: str, batch: Batch, compression: str, level:
         c = batch.columns[f.name]
         names.append(f.name)
         dtypes.append(f.dtype)
-        if c.is_string:
+        if c.is_list:
+            t = c.data.cpu()
+            ed = c.elem_dtype
+            if ed in ("int8", "int16"):
+                t = t.to(torch.int32)
+            if ed == "bool":
+                t = t.to(torch.uint8)
+            columns.append(t)
+            offsets.append(c.offsets.cpu().to(torch.int64))
+        elif c.is_string:
             columns.append(c.bytes_.cpu())
             offsets.append(c.offsets.cpu())
         else:
