@@ -5,6 +5,7 @@
 #include <atomic>
 #include <cstring>
 #include <memory>
+#include <map>
 #include <mutex>
 #include <thread>
 #include <unordered_map>
@@ -192,6 +193,143 @@ static int64_t write_parquet(
     size = w.close();
   }
   return size;
+}
+
+// ---------------------------------------------------------------------- //
+// incremental writer handles (streaming multipart upload support):
+// writer_open -> write batches one by one (each call appends row groups
+// and flushes) -> writer_bytes gives the stable part boundary ->
+// writer_finish emits the footer. The marshaling mirrors write_parquet.
+// ---------------------------------------------------------------------- //
+
+namespace {
+struct OpenWriter {
+  std::unique_ptr<ParquetWriter> w;
+  std::vector<ColumnDesc> descs;
+  std::vector<std::string> dtypes;
+  std::string path;
+};
+std::mutex g_wmu;
+std::map<int64_t, std::shared_ptr<OpenWriter>> g_writers;
+int64_t g_wnext = 1;
+
+std::shared_ptr<OpenWriter> get_writer(int64_t h) {
+  std::lock_guard<std::mutex> lk(g_wmu);
+  auto it = g_writers.find(h);
+  if (it == g_writers.end()) throw std::runtime_error("bad writer handle");
+  return it->second;
+}
+}  // namespace
+
+static int64_t writer_open(const std::string& path,
+                           const std::vector<std::string>& names,
+                           const std::vector<std::string>& dtypes,
+                           const std::vector<bool>& nullable,
+                           int64_t row_group_size, int64_t codec,
+                           int64_t level) {
+  auto ow = std::make_shared<OpenWriter>();
+  ow->path = path;
+  ow->dtypes = dtypes;
+  ow->descs.resize(names.size());
+  for (size_t i = 0; i < names.size(); i++) {
+    bool is_list = dtypes[i].rfind("list<", 0) == 0;
+    std::string sd = is_list ? dtypes[i].substr(5, dtypes[i].size() - 6)
+                             : dtypes[i];
+    DtypeInfo di = dtype_info(sd);
+    ColumnDesc& d = ow->descs[i];
+    d.name = names[i];
+    d.physical = di.physical;
+    d.converted = di.converted;
+    d.logical = di.logical;
+    d.int_bit_width = di.bit_width;
+    d.dec_precision = di.dec_precision;
+    d.dec_scale = di.dec_scale;
+    d.nullable = nullable[i];
+    d.is_list = is_list;
+    if (is_list) d.max_def = 2;
+  }
+  ow->w = std::make_unique<ParquetWriter>(path, ow->descs, (int)codec,
+                                          (int)level, row_group_size);
+  std::lock_guard<std::mutex> lk(g_wmu);
+  int64_t h = g_wnext++;
+  g_writers[h] = ow;
+  return h;
+}
+
+static void writer_write(int64_t h, const std::vector<torch::Tensor>& columns,
+                         const std::vector<c10::optional<torch::Tensor>>& offsets,
+                         const std::vector<c10::optional<torch::Tensor>>& validity) {
+  auto ow = get_writer(h);
+  size_t ncol = ow->descs.size();
+  TORCH_CHECK(columns.size() == ncol);
+  std::vector<ColumnData> data(ncol);
+  std::vector<torch::Tensor> keep;
+  int64_t num_rows = -1;
+  for (size_t i = 0; i < ncol; i++) {
+    const ColumnDesc& d = ow->descs[i];
+    torch::Tensor col = columns[i].contiguous().cpu();
+    keep.push_back(col);
+    if (d.is_list) {
+      TORCH_CHECK(offsets[i].has_value(), "list column needs offsets");
+      torch::Tensor off = offsets[i]->contiguous().cpu().to(torch::kInt64);
+      keep.push_back(off);
+      data[i].list_offsets = off.data_ptr<int64_t>();
+      data[i].data = (const uint8_t*)col.data_ptr();
+      int64_t n = off.numel() - 1;
+      TORCH_CHECK(num_rows < 0 || n == num_rows);
+      num_rows = n;
+    } else if (d.physical == PT_BYTE_ARRAY) {
+      TORCH_CHECK(offsets[i].has_value(), "byte_array column needs offsets");
+      torch::Tensor off = offsets[i]->contiguous().cpu().to(torch::kInt32);
+      keep.push_back(off);
+      data[i].offsets = off.data_ptr<int32_t>();
+      data[i].bytes = (const uint8_t*)col.data_ptr();
+      int64_t n = off.numel() - 1;
+      TORCH_CHECK(num_rows < 0 || n == num_rows);
+      num_rows = n;
+    } else {
+      data[i].data = (const uint8_t*)col.data_ptr();
+      int64_t n = col.numel();
+      TORCH_CHECK(num_rows < 0 || n == num_rows);
+      num_rows = n;
+    }
+    if (validity[i].has_value()) {
+      torch::Tensor v = validity[i]->contiguous().cpu().to(torch::kUInt8);
+      keep.push_back(v);
+      data[i].validity = v.data_ptr<uint8_t>();
+    }
+  }
+  if (num_rows < 0) num_rows = 0;
+  py::gil_scoped_release rel;
+  ow->w->write_row_group(data, num_rows);
+  ow->w->flush_os();
+}
+
+static int64_t writer_bytes(int64_t h) { return get_writer(h)->w->bytes_written(); }
+
+static int64_t writer_finish(int64_t h) {
+  auto ow = get_writer(h);
+  int64_t size;
+  {
+    py::gil_scoped_release rel;
+    size = ow->w->close();
+  }
+  std::lock_guard<std::mutex> lk(g_wmu);
+  g_writers.erase(h);
+  return size;
+}
+
+static void writer_abort(int64_t h) {
+  std::shared_ptr<OpenWriter> ow;
+  {
+    std::lock_guard<std::mutex> lk(g_wmu);
+    auto it = g_writers.find(h);
+    if (it == g_writers.end()) return;
+    ow = it->second;
+    g_writers.erase(it);
+  }
+  ow->w.reset();
+  std::remove(ow->path.c_str());
 }
 
 // ---------------------------------------------------------------------- //
@@ -686,6 +824,11 @@ static torch::Tensor bucket_ids_from_hashes(torch::Tensor hashes, int64_t nbucke
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "lakesoul_amd host-native core";
   m.def("write_parquet", &write_parquet, "write a parquet file");
+  m.def("writer_open", &writer_open);
+  m.def("writer_write", &writer_write);
+  m.def("writer_bytes", &writer_bytes);
+  m.def("writer_finish", &writer_finish);
+  m.def("writer_abort", &writer_abort);
   m.def("open_parquet", &open_parquet);
   m.def("close_parquet", &close_parquet);
   m.def("parquet_meta", &parquet_meta);

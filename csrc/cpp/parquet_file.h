@@ -146,6 +146,16 @@ class ParquetWriter {
     }
   }
 
+  // bytes emitted so far — a stable part boundary after every
+  // write_row_group (each call fully flushes its pages), so a streaming
+  // uploader can ship [uploaded, bytes_written) while the next row group
+  // encodes (reference multipart_writer.rs:43 overlap)
+  int64_t bytes_written() const { return pos_; }
+
+  void flush_os() {
+    if (f_) std::fflush(f_);
+  }
+
   int64_t close() {
     FileMetaData fm;
     fm.version = 2;

@@ -46,12 +46,19 @@ def _chunk_keys_cpu(offsets: np.ndarray, data: np.ndarray, chunk: int
     return out.view(np.int64)
 
 
+_SIGN64 = -0x8000000000000000
+
+
 def _str_chunk_keys(col: Column, chunk: int) -> torch.Tensor:
+    """RAW big-endian chunk keys (unsigned order = XOR sign bit before a
+    signed argsort). The HIP kernel returns keys pre-flipped for direct
+    signed sorts — undo that here so both paths agree."""
     offs = col.offsets
     if offs.device.type == "cuda":
         from ..ops import hip
 
-        return hip().str_chunk_keys(offs.to(torch.int64), col.bytes_, chunk)
+        k = hip().str_chunk_keys(offs.to(torch.int64), col.bytes_, chunk)
+        return k ^ torch.tensor(_SIGN64, dtype=torch.int64, device=k.device)
     o = offs.cpu().numpy().astype(np.int64)
     b = col.bytes_.cpu().numpy()
     return torch.from_numpy(_chunk_keys_cpu(o, b, chunk))
