@@ -190,6 +190,12 @@ class FileSystem:
             return
         raise ValueError(f"unsupported scheme {scheme}://")
 
+    def open_multipart(self, dest: str) -> "MultipartSink":
+        """Streaming upload sink: parts written incrementally become
+        visible only on complete(); abort() leaves no visible object
+        (reference multipart_writer.rs:43,239 abort semantics)."""
+        return MultipartSink(self, dest)
+
     def delete(self, path: str) -> None:
         scheme, rest = _split_scheme(path)
         if scheme == "file":
@@ -209,6 +215,76 @@ class FileSystem:
         elif scheme == "mock":
             os.makedirs(os.path.join(self._mock_root(), rest), exist_ok=True)
         # s3: directories are implicit
+
+
+class MultipartSink:
+    """Incremental writer to any scheme. The in-progress object is
+    invisible until complete():
+
+    - file:// and mock://: parts append to a hidden ``.__inprogress``
+      sibling; complete() atomically renames it into place; abort()
+      removes it (no final object ever exists).
+    - s3://: parts stream into pyarrow's S3 output stream (multipart
+      upload under the hood — S3 itself keeps parts invisible until the
+      upload completes); abort() closes and best-effort-deletes.
+    """
+
+    def __init__(self, fs: FileSystem, dest: str):
+        self.fs = fs
+        self.dest = dest
+        self.parts = 0
+        self.bytes = 0
+        self._done = False
+        scheme, rest = _split_scheme(dest)
+        self._scheme = scheme
+        if scheme == "file":
+            self._tmp = rest + ".__inprogress"
+            os.makedirs(os.path.dirname(rest), exist_ok=True)
+            self._f = open(self._tmp, "wb")
+        elif scheme == "mock":
+            target = os.path.join(fs._mock_root(), rest)
+            os.makedirs(os.path.dirname(target), exist_ok=True)
+            self._final = target
+            self._tmp = target + ".__inprogress"
+            self._f = open(self._tmp, "wb")
+        elif scheme in ("s3", "s3a"):
+            self._f = fs._s3fs().open_output_stream(rest)
+            self._tmp = None
+        else:
+            raise ValueError(f"unsupported scheme {scheme}://")
+
+    def write_part(self, data: bytes) -> None:
+        assert not self._done
+        self._f.write(data)
+        self.parts += 1
+        self.bytes += len(data)
+
+    def complete(self) -> None:
+        if self._done:
+            return
+        self._done = True
+        self._f.close()
+        if self._scheme == "file":
+            os.replace(self._tmp, _split_scheme(self.dest)[1])
+        elif self._scheme == "mock":
+            os.replace(self._tmp, self._final)
+
+    def abort(self) -> None:
+        """Cancel: no visible object remains."""
+        if self._done:
+            return
+        self._done = True
+        try:
+            self._f.close()
+        except Exception:
+            pass
+        if self._tmp is not None and os.path.exists(self._tmp):
+            os.remove(self._tmp)
+        if self._scheme in ("s3", "s3a"):
+            try:
+                self.fs.delete(self.dest)
+            except Exception:
+                pass
 
 
 _default_fs: Optional[FileSystem] = None

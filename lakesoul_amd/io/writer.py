@@ -204,20 +204,30 @@ def _partition_descs(batch: Batch, range_cols: Sequence[str]):
 
 def _write_batch_to_file(path: str, batch: Batch, compression: str, level: int,
                          row_group_size: int) -> int:
-    """Write one parquet file; remote destinations write locally then
-    upload (multipart-writer analog, multipart_writer.rs:43)."""
+    """Write one parquet file; remote destinations stream row-group parts
+    while the next row group encodes (multipart_writer.rs:43 overlap);
+    a failure aborts the upload so no partial object becomes visible."""
     if is_remote(path):
-        import tempfile
+        from .multipart import StreamingParquetUpload
 
-        fd, tmp = tempfile.mkstemp(suffix=".parquet")
-        os.close(fd)
+        up = StreamingParquetUpload(path, batch.schema, compression, level,
+                                    row_group_size)
         try:
-            size = _write_batch_to_file_local(tmp, batch, compression, level, row_group_size)
-            default_fs().upload(tmp, path)
-            return size
-        finally:
-            if os.path.exists(tmp):
-                os.remove(tmp)
+            n = batch.num_rows
+            cpu_batch = batch.to_device("cpu") if any(
+                (c.offsets if (c.is_string or c.is_list) else c.data).device.type
+                == "cuda" for c in batch.columns.values()) else batch
+            for a in range(0, max(n, 1), max(row_group_size, 1)):
+                b = min(a + row_group_size, n)
+                with timing.phase("w_encode_stream"):
+                    up.write_batch(cpu_batch.slice(a, b))
+                if b >= n:
+                    break
+            with timing.phase("w_upload_tail"):
+                return up.close()
+        except BaseException:
+            up.abort()
+            raise
     return _write_batch_to_file_local(path, batch, compression, level, row_group_size)
 
 

@@ -1,6 +1,9 @@
 """Staged GPU repro for the query-engine arrow corruption — run directly
 on a GPU box: python tests/gpu_repro_engine.py"""
 
+
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import numpy as np
 import torch
 
