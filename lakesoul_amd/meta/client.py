@@ -106,6 +106,7 @@ class MetaClient:
         for attempt in range(self.max_retry):
             try:
                 self._commit_data_once(meta_info, commit_op)
+                self._maybe_notify_compaction(meta_info, commit_op)
                 return
             except CommitConflictError as e:
                 last_err = e
@@ -114,6 +115,26 @@ class MetaClient:
         raise CommitConflictError(
             f"commit_data failed after {self.max_retry} retries: {last_err}"
         )
+
+    def _maybe_notify_compaction(self, meta_info: MetaInfo,
+                                 commit_op: CommitOp) -> None:
+        """Publisher half of the PG trigger (meta_init.sql:102-150): on a
+        delta commit that crosses the threshold, fire the compaction
+        channel. No-op unless a notify bus is attached."""
+        bus = getattr(self, "notify_bus", None)
+        if bus is None or commit_op not in (CommitOp.AppendCommit,
+                                            CommitOp.MergeCommit):
+            return
+        from .notify import CompactionEvent
+
+        ti = meta_info.table_info
+        for part in meta_info.list_partition:
+            if self.compaction_needed(ti.table_id, part.partition_desc):
+                cur = self.store.get_latest_partition_info(
+                    ti.table_id, part.partition_desc)
+                bus.publish(CompactionEvent(
+                    ti.table_id, part.partition_desc,
+                    cur.version if cur else 0))
 
     def _commit_data_once(self, meta_info: MetaInfo, commit_op: CommitOp) -> None:
         table_info = meta_info.table_info

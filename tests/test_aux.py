@@ -232,3 +232,50 @@ def test_ioconfig_env_fallback(monkeypatch):
     cfg = IOConfig()
     assert cfg.option("my_custom_opt", "fallback") == "hello"
     assert cfg.option("absent_opt", "fallback") == "fallback"
+
+
+def test_notify_driven_compaction(catalog):
+    """Notify-driven compaction pipeline (reference meta_init.sql:102-150
+    trigger -> pg_notify -> listener): the local bus fires when a
+    partition crosses 10 delta commits and the subscribed compactor
+    compacts exactly that partition."""
+    import numpy as np
+
+    from lakesoul_amd.io.schema import Field, Schema
+    from lakesoul_amd.meta.entities import CommitOp
+    from lakesoul_amd.meta.notify import LocalNotifyBus, NotifyDrivenCompactor
+
+    bus = LocalNotifyBus()
+    catalog.client.notify_bus = bus
+    try:
+        t = catalog.create_table(
+            "notif",
+            Schema([Field("id", "int64", False), Field("v", "float64")]),
+            primary_keys=["id"], hash_bucket_num=1,
+        )
+        comp = NotifyDrivenCompactor(catalog, bus)
+        for i in range(10):
+            t.upsert({"id": np.arange(10, dtype=np.int64),
+                      "v": np.full(10, float(i))})
+        assert not bus.published  # below the threshold: no event yet
+        assert comp.drain() == 0
+        t.upsert({"id": np.arange(10, dtype=np.int64), "v": np.full(10, 99.0)})
+        assert len(bus.published) == 1
+        ev = bus.published[0]
+        assert ev.table_id == t.table_id and ev.partition_desc == "-5"
+        # payload round trip (what pg_notify would carry)
+        from lakesoul_amd.meta.notify import CompactionEvent
+
+        assert CompactionEvent.from_payload(ev.payload()) == ev
+        assert comp.drain() == 1
+        # compaction committed: latest version is a CompactionCommit and
+        # data still reads correctly
+        cur = catalog.client.store.get_latest_partition_info(t.table_id, "-5")
+        assert cur.commit_op is CommitOp.CompactionCommit
+        df = t.to_pandas()
+        assert len(df) == 10 and (df["v"] == 99.0).all()
+        # further single delta does not re-fire
+        t.upsert({"id": np.arange(10, dtype=np.int64), "v": np.zeros(10)})
+        assert len(bus.published) == 1
+    finally:
+        catalog.client.notify_bus = None
