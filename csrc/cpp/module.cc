@@ -588,13 +588,13 @@ static py::tuple prep_rle_runs(torch::Tensor values, torch::Tensor idx_pages) {
 static py::dict read_unit_raw_py(const std::vector<std::string>& paths,
                                  const std::vector<std::string>& names,
                                  int64_t nthreads, bool pin, bool gpu_snappy,
-                                 bool gpu_zstd) {
+                                 bool gpu_zstd, double gpu_zstd_frac) {
   (void)nthreads;
   std::unique_ptr<UnitStage> st;
   auto t0 = std::chrono::steady_clock::now();
   {
     py::gil_scoped_release rel;
-    st = read_unit_stage1(paths, names, gpu_snappy, gpu_zstd);
+    st = read_unit_stage1(paths, names, gpu_snappy, gpu_zstd, gpu_zstd_frac);
   }
   auto t1 = std::chrono::steady_clock::now();
   UnitStage& ud = *st;
@@ -838,7 +838,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("prep_rle_runs", &prep_rle_runs);
   m.def("read_unit_raw", &read_unit_raw_py, py::arg("paths"), py::arg("names"),
         py::arg("nthreads") = 0, py::arg("pin") = true,
-        py::arg("gpu_snappy") = false, py::arg("gpu_zstd") = false);
+        py::arg("gpu_snappy") = false, py::arg("gpu_zstd") = false,
+        py::arg("gpu_zstd_frac") = 1.0);
   m.def("read_chunks_raw_batch", &read_chunks_raw_batch);
   m.def("zstd_compress_ref", [](py::bytes src, int64_t level) {
     std::string b = src;

@@ -140,7 +140,8 @@ struct UnitStage {
 
 inline std::unique_ptr<UnitStage> read_unit_stage1(
     const std::vector<std::string>& paths, const std::vector<std::string>& names,
-    bool gpu_snappy = false, bool gpu_zstd = false) {
+    bool gpu_snappy = false, bool gpu_zstd = false,
+    double gpu_zstd_frac = 1.0) {
   auto st = std::make_unique<UnitStage>();
   UnitStage& S = *st;
   auto now = [] { return std::chrono::steady_clock::now(); };
@@ -181,8 +182,15 @@ inline std::unique_ptr<UnitStage> read_unit_stage1(
       try {
         const Task& t = tasks[i];
         auto& fd = S.files[t.fi];
+        // hybrid host/GPU decompress split (profiles/r01_gpu_zstd.md
+        // round-2 plan item 3): route gpu_zstd_frac of the chunks to the
+        // GPU zstd kernel, the rest to the host pool — the prefetch
+        // pipeline overlaps them across units, so steady-state unit cost
+        // is max(host share, gpu share) instead of the loser's total
+        bool this_gpu_zstd =
+            gpu_zstd && ((double)((i * 2654435761u) % 1000) < gpu_zstd_frac * 1000.0);
         fd.chunks[t.c][t.rg] =
-            fd.f->read_chunk(t.rg, fd.col_idx[t.c], gpu_snappy, true, gpu_zstd);
+            fd.f->read_chunk(t.rg, fd.col_idx[t.c], gpu_snappy, true, this_gpu_zstd);
       } catch (std::exception& e) {
         std::lock_guard<std::mutex> lk(err_mu);
         err = e.what();

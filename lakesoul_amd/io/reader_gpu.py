@@ -50,16 +50,7 @@ _ESIZE = {torch.uint8: 1, torch.int32: 4, torch.int64: 8, torch.float32: 4, torc
 
 import os as _os
 
-def _default_gpu_zstd() -> bool:
-    """GPU zstd decode (wave-per-page kernel, ~6-13 GB/s measured) beats
-    the host pool only when host CPUs are scarce: the cgroup CPU quota is
-    shared by every rank on the node, so at 8 ranks each rank gets ~2
-    threads (~1.5 GB/s zstd) while the GPU path holds its rate. Default:
-    ON when fewer than 6 host threads are available per rank, OFF
-    otherwise (measured N=1 crossover, profiles/r01_gpu_zstd.md)."""
-    env = _os.environ.get("LAKESOUL_GPU_ZSTD")
-    if env is not None:
-        return env == "1"
+def _host_threads_per_rank() -> int:
     try:
         quota_s = open("/sys/fs/cgroup/cpu.max").read().split()
         ncpu = _os.cpu_count() or 16
@@ -68,17 +59,47 @@ def _default_gpu_zstd() -> bool:
     except OSError:
         quota = _os.cpu_count() or 16
     world = int(_os.environ.get("WORLD_SIZE", "1"))
-    return quota // max(1, world) < 6
+    return max(1, quota // max(1, world))
 
 
-_GPU_ZSTD = _default_gpu_zstd()
+def _default_gpu_zstd_frac() -> float:
+    """Fraction of zstd chunks decoded by the GPU kernel (the rest go to
+    the host pool; the prefetch pipeline overlaps the two across units).
+
+    Balanced split from measured single-source rates (profiles/
+    r01_gpu_zstd.md: host pool ~90 ms/step at 16 threads, GPU v2 kernel
+    ~136 ms/step): frac = host_rate/(host_rate+gpu_rate) scaled by the
+    rank's actual thread budget. Starved ranks (<3 threads) go full GPU;
+    LAKESOUL_GPU_ZSTD=0/1 still forces pure host / pure GPU and
+    LAKESOUL_GPU_ZSTD_FRAC pins an explicit split."""
+    env = _os.environ.get("LAKESOUL_GPU_ZSTD")
+    if env is not None and env != "":
+        if env == "0":
+            return 0.0
+        if env == "1":
+            return 1.0
+    envf = _os.environ.get("LAKESOUL_GPU_ZSTD_FRAC")
+    if envf:
+        return min(1.0, max(0.0, float(envf)))
+    threads = _host_threads_per_rank()
+    if threads < 3:
+        return 1.0
+    # host rate scales ~linearly with threads up to the pool cap (16)
+    host_rate = min(threads, 16) / 16.0 * (1.0 / 90.0)
+    gpu_rate = 1.0 / 136.0
+    return gpu_rate / (gpu_rate + host_rate)
+
+
+_GPU_ZSTD_FRAC = _default_gpu_zstd_frac()
+_GPU_ZSTD = _GPU_ZSTD_FRAC > 0.0
 
 
 def fetch_raw(files: List[str], names: List[str]) -> dict:
     """Host phase of a unit read (releases the GIL in C++) — safe to run
     on a prefetch thread while the GPU processes the previous unit."""
     with timing.phase("host_fetch"):
-        raw = cpp().read_unit_raw(files, names, 0, True, True, _GPU_ZSTD)
+        raw = cpp().read_unit_raw(files, names, 0, True, True, _GPU_ZSTD,
+                                  _GPU_ZSTD_FRAC)
     if timing.ENABLED:
         timing._acc["fetch.stage1"] += raw["t_stage1_us"] / 1e6
         timing._acc["fetch.s1_open"] += raw["t_open_us"] / 1e6
