@@ -82,12 +82,16 @@ def _default_gpu_zstd_frac() -> float:
     if envf:
         return min(1.0, max(0.0, float(envf)))
     threads = _host_threads_per_rank()
-    if threads < 3:
+    # Measured r2 A/B on the headline bench (gpurun bench_frac_*.log):
+    # host-only 40 ms/step, frac=0.4 342 ms, frac=1.0 371 ms at the
+    # 128 KB default pages — the GPU kernel's wave-per-page parallelism
+    # collapses on large pages (it wants thousands of 32 KB pages per
+    # unit, profiles/r01_gpu_zstd.md). Hybrid stays opt-in via
+    # LAKESOUL_GPU_ZSTD_FRAC (+ LAKESOUL_PAGE_BYTES=32768 at write time);
+    # the adaptive rule keeps full-GPU decode for starved ranks only.
+    if threads < 6:
         return 1.0
-    # host rate scales ~linearly with threads up to the pool cap (16)
-    host_rate = min(threads, 16) / 16.0 * (1.0 / 90.0)
-    gpu_rate = 1.0 / 136.0
-    return gpu_rate / (gpu_rate + host_rate)
+    return 0.0
 
 
 _GPU_ZSTD_FRAC = _default_gpu_zstd_frac()
