@@ -112,6 +112,25 @@ class DiskCache:
         }
 
 
+def retry_io(fn, attempts: int = 4, base_delay: float = 0.1,
+             retryable=(OSError, IOError)):
+    """Exponential-backoff retry for remote IO (reference
+    object_store.rs:22-82 configures the same via fs.s3a.* options).
+    Jittered delays base*2^k; the last failure propagates."""
+    import random as _random
+    import time as _time
+
+    last = None
+    for k in range(attempts):
+        try:
+            return fn()
+        except retryable as e:
+            last = e
+            if k + 1 < attempts:
+                _time.sleep(base_delay * (2 ** k) * (0.5 + _random.random()))
+    raise last
+
+
 class FileSystem:
     """Scheme-dispatching filesystem with read-through localization."""
 
@@ -119,6 +138,11 @@ class FileSystem:
         self.cache = cache
         self.options = options or {}
         self._s3 = None
+
+    def _retry_attempts(self) -> int:
+        # Hadoop-style option name, as the reference accepts
+        return int(self.options.get(
+            "fs.s3a.retry.limit", os.environ.get("LAKESOUL_S3_RETRIES", "4")))
 
     def _get_cache(self) -> DiskCache:
         if self.cache is None:
@@ -167,8 +191,11 @@ class FileSystem:
         if scheme == "mock":
             shutil.copyfile(os.path.join(self._mock_root(), rest), tmp)
         elif scheme in ("s3", "s3a"):
-            with self._s3fs().open_input_stream(rest) as src, open(tmp, "wb") as dst:
-                shutil.copyfileobj(src, dst)
+            def _dl():
+                with self._s3fs().open_input_stream(rest) as src, open(tmp, "wb") as dst:
+                    shutil.copyfileobj(src, dst)
+
+            retry_io(_dl, attempts=self._retry_attempts())
         else:
             raise ValueError(f"unsupported scheme {scheme}://")
         return cache.put_from(path, tmp)
@@ -185,8 +212,11 @@ class FileSystem:
             shutil.copyfile(local_path, target)
             return
         if scheme in ("s3", "s3a"):
-            with open(local_path, "rb") as src, self._s3fs().open_output_stream(rest) as dst:
-                shutil.copyfileobj(src, dst)
+            def _ul():
+                with open(local_path, "rb") as src, self._s3fs().open_output_stream(rest) as dst:
+                    shutil.copyfileobj(src, dst)
+
+            retry_io(_ul, attempts=self._retry_attempts())
             return
         raise ValueError(f"unsupported scheme {scheme}://")
 

@@ -70,3 +70,41 @@ def test_remote_table_roundtrip(mock_fs, tmp_path, meta_store, monkeypatch):
     # files live in the mock remote, not locally
     for f in t.files():
         assert f.path.startswith("mock://")
+
+
+def test_retry_io_backoff_and_raise():
+    """Remote IO retries with exponential backoff (reference
+    object_store.rs retry config)."""
+    from lakesoul_amd.io.fs import retry_io
+
+    calls = []
+
+    def flaky():
+        calls.append(1)
+        if len(calls) < 3:
+            raise OSError("transient")
+        return "ok"
+
+    assert retry_io(flaky, attempts=4, base_delay=0.001) == "ok"
+    assert len(calls) == 3
+    calls.clear()
+
+    def always():
+        calls.append(1)
+        raise OSError("down")
+
+    import pytest as _pytest
+
+    with _pytest.raises(OSError, match="down"):
+        retry_io(always, attempts=3, base_delay=0.001)
+    assert len(calls) == 3
+    # non-retryable errors propagate immediately
+    calls.clear()
+
+    def typeerr():
+        calls.append(1)
+        raise TypeError("bug")
+
+    with _pytest.raises(TypeError):
+        retry_io(typeerr, attempts=5, base_delay=0.001)
+    assert len(calls) == 1
