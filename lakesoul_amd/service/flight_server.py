@@ -88,6 +88,7 @@ class LakeSoulFlightServer(fl.FlightServerBase if _HAVE_FLIGHT else object):
         self.tokens = TokenService(secret)
         self.metrics = StreamWriteMetrics()
         self._lock = threading.Lock()
+        self._sql_results: dict = {}  # statement handle -> result table
         super().__init__(
             location,
             middleware={"auth": _AuthMiddlewareFactory(self.tokens)},
@@ -135,9 +136,68 @@ class LakeSoulFlightServer(fl.FlightServerBase if _HAVE_FLIGHT else object):
         n_rows = -1
         return fl.FlightInfo(schema_to_arrow(t.schema), desc, [ep], n_rows, -1)
 
+    # -- Flight SQL protocol (wire-level; service/flightsql.py) -------- #
+
+    def _try_flightsql(self, cmd: bytes):
+        """Return (short_type, payload) if cmd is a flight-sql Any."""
+        try:
+            from . import flightsql as fsql
+
+            name, payload = fsql.unpack_any(cmd)
+            return (name, payload) if name else (None, b"")
+        except Exception:
+            return (None, b"")
+
+    def _flightsql_info(self, context, name: str, payload: bytes, descriptor):
+        """GetFlightInfo for flight-sql commands: run/prepare the result,
+        stash it under a statement handle, return endpoints whose ticket
+        is an Any-packed TicketStatementQuery (flight_sql_service.rs:218
+        get_flight_info_statement analog)."""
+        import uuid
+
+        from . import flightsql as fsql
+
+        if name == "CommandStatementQuery":
+            from ..sql import execute_sql
+
+            query = fsql.parse_string_field(payload, 1)
+            df = execute_sql(self.catalog, query)
+            tbl = pa.Table.from_pandas(df, preserve_index=False)
+        elif name == "CommandGetCatalogs":
+            tbl = pa.table({"catalog_name": pa.array(
+                ["lakesoul"], pa.string())})
+        elif name == "CommandGetDbSchemas":
+            ns = self.catalog.list_namespaces() or ["default"]
+            tbl = pa.table({
+                "catalog_name": pa.array(["lakesoul"] * len(ns), pa.string()),
+                "db_schema_name": pa.array(ns, pa.string()),
+            })
+        elif name == "CommandGetTables":
+            rows = {"catalog_name": [], "db_schema_name": [],
+                    "table_name": [], "table_type": []}
+            for ns in self.catalog.list_namespaces() or ["default"]:
+                for tname in self.catalog.list_tables(ns):
+                    rows["catalog_name"].append("lakesoul")
+                    rows["db_schema_name"].append(ns)
+                    rows["table_name"].append(tname)
+                    rows["table_type"].append("TABLE")
+            tbl = pa.table({k: pa.array(v, pa.string())
+                            for k, v in rows.items()})
+        else:
+            raise fl.FlightServerError(f"unsupported flight-sql command {name}")
+        handle = uuid.uuid4().bytes
+        with self._lock:
+            self._sql_results[handle] = tbl
+        ticket = fl.Ticket(fsql.ticket_statement_query(handle))
+        ep = fl.FlightEndpoint(ticket, [])
+        return fl.FlightInfo(tbl.schema, descriptor, [ep], tbl.num_rows, -1)
+
     def get_flight_info(self, context, descriptor):
         claims = self._claims(context)
         if descriptor.descriptor_type == fl.DescriptorType.CMD:
+            name, payload = self._try_flightsql(descriptor.command)
+            if name:
+                return self._flightsql_info(context, name, payload, descriptor)
             d = json.loads(descriptor.command.decode())
         else:
             path = [p.decode() if isinstance(p, bytes) else p for p in descriptor.path]
@@ -154,6 +214,17 @@ class LakeSoulFlightServer(fl.FlightServerBase if _HAVE_FLIGHT else object):
 
     def do_get(self, context, ticket):
         claims = self._claims(context)
+        name, payload = self._try_flightsql(ticket.ticket)
+        if name == "TicketStatementQuery":
+            from . import flightsql as fsql
+
+            handle = fsql.parse_bytes_field(payload, 1)
+            with self._lock:
+                tbl = self._sql_results.pop(handle, None)
+            if tbl is None:
+                raise fl.FlightServerError("unknown statement handle")
+            self.metrics.requests += 1
+            return fl.RecordBatchStream(tbl)
         d = json.loads(ticket.ticket.decode())
         t = self._table(d["table"], d.get("namespace", "default"), claims)
         scan = t.scan(
@@ -175,6 +246,22 @@ class LakeSoulFlightServer(fl.FlightServerBase if _HAVE_FLIGHT else object):
     def do_put(self, context, descriptor, reader, writer):
         claims = self._claims(context)
         if descriptor.descriptor_type == fl.DescriptorType.CMD:
+            name, payload = self._try_flightsql(descriptor.command)
+            if name == "CommandStatementUpdate":
+                # flight-sql UPDATE/INSERT path: run the statement, reply
+                # with DoPutUpdateResult{record_count}
+                from . import flightsql as fsql
+                from ..sql import execute_sql
+
+                query = fsql.parse_string_field(payload, 1)
+                df = execute_sql(self.catalog, query)
+                count = 0
+                for c in ("rows_inserted", "rows_updated", "rows_deleted"):
+                    if c in getattr(df, "columns", []):
+                        count = int(df[c].iloc[0])
+                        break
+                writer.write(pa.py_buffer(fsql.do_put_update_result(count)))
+                return
             d = json.loads(descriptor.command.decode())
         else:
             path = [p.decode() if isinstance(p, bytes) else p for p in descriptor.path]

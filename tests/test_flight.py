@@ -157,3 +157,69 @@ def test_do_get_incremental(flight):
     ids = set(inc.column("id").to_pylist())
     assert {1000, 1001} <= ids
     assert len(ids) < 102  # not the full table
+
+
+def test_flightsql_statement_query_wire(flight):
+    """The standard Flight SQL GetFlightInfo/DoGet flow with wire-exact
+    Any-packed CommandStatementQuery / TicketStatementQuery envelopes
+    (reference flight_sql_service.rs:218)."""
+    from lakesoul_amd.service import flightsql as fsql
+
+    srv, catalog = flight
+    client, opts = _client(srv)
+    desc = fl.FlightDescriptor.for_command(
+        fsql.cmd_statement_query("SELECT id, v FROM ft WHERE id < 5 ORDER BY id"))
+    info = client.get_flight_info(desc, options=opts)
+    assert info.total_records == 5
+    ticket = info.endpoints[0].ticket
+    # the ticket must be an Any-packed TicketStatementQuery
+    name, payload = fsql.unpack_any(ticket.ticket)
+    assert name == "TicketStatementQuery"
+    assert fsql.parse_bytes_field(payload, 1)
+    t = client.do_get(ticket, options=opts).read_all()
+    assert t.column("id").to_pylist() == [0, 1, 2, 3, 4]
+    # a second do_get with the same handle fails (handle consumed)
+    with pytest.raises(fl.FlightError):
+        client.do_get(ticket, options=opts).read_all()
+
+
+def test_flightsql_catalog_metadata_commands(flight):
+    from lakesoul_amd.service import flightsql as fsql
+
+    srv, catalog = flight
+    client, opts = _client(srv)
+    for cmd, col, expect in [
+        (fsql.cmd_get_catalogs(), "catalog_name", ["lakesoul"]),
+        (fsql.cmd_get_db_schemas(), "db_schema_name", None),
+        (fsql.cmd_get_tables(), "table_name", None),
+    ]:
+        info = client.get_flight_info(fl.FlightDescriptor.for_command(cmd),
+                                      options=opts)
+        t = client.do_get(info.endpoints[0].ticket, options=opts).read_all()
+        assert col in t.schema.names
+        if expect is not None:
+            assert t.column(col).to_pylist() == expect
+    # tables listing includes ft
+    info = client.get_flight_info(
+        fl.FlightDescriptor.for_command(fsql.cmd_get_tables()), options=opts)
+    t = client.do_get(info.endpoints[0].ticket, options=opts).read_all()
+    assert "ft" in t.column("table_name").to_pylist()
+
+
+def test_flightsql_statement_update(flight):
+    from lakesoul_amd.service import flightsql as fsql
+
+    srv, catalog = flight
+    client, opts = _client(srv)
+    desc = fl.FlightDescriptor.for_command(
+        fsql.cmd_statement_update("INSERT INTO ft VALUES (1000, 42.0)"))
+    schema = pa.schema([])
+    writer, reader = client.do_put(desc, schema, options=opts)
+    writer.done_writing()
+    buf = reader.read()
+    assert fsql.parse_do_put_update_result(bytes(memoryview(buf))) == 1
+    writer.close()
+    # row landed
+    ticket = fl.Ticket(json.dumps({"table": "ft"}).encode())
+    t = client.do_get(ticket, options=opts).read_all()
+    assert 1000 in t.column("id").to_pylist()
