@@ -342,6 +342,19 @@ class LakeSoulScan:
             with _torch.cuda.stream(s):
                 batch = read_unit_gpu(self, unit, raw)
             s.synchronize()
+            # the batch's tensors were ALLOCATED on side stream s but are
+            # consumed (and eventually freed) on the default stream: tag
+            # the default stream as a user so the caching allocator does
+            # not hand their blocks back to s-stream work that may still
+            # overlap the consumer's reads (classic record_stream
+            # cross-stream lifetime rule; without this, later units
+            # reusing s corrupt still-referenced batches)
+            cur = _torch.cuda.current_stream()
+            if batch is not None:
+                for c in batch.columns.values():
+                    for t in (c.data, c.offsets, c.bytes_, c.validity):
+                        if t is not None and t.is_cuda:
+                            t.record_stream(cur)
             return batch
 
         with ThreadPoolExecutor(max_workers=depth) as fex, ThreadPoolExecutor(
