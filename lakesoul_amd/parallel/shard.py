@@ -39,8 +39,8 @@ def shard_scan(scan):
 
 def _col_row_bytes(f, c: Column) -> int:
     """Bytes per row this field contributes to the packed buffer."""
-    if c.is_string:
-        b = 8  # int64 length
+    if c.is_string or c.is_list:
+        b = 8  # int64 length (elements for lists, bytes for strings)
     else:
         b = c.data.element_size()
     if f.nullable:
@@ -88,6 +88,8 @@ class AsyncExchange:
         for f in batch.schema:
             c = reordered.columns[f.name]
             t = c.bytes_ if c.is_string else c.data
+            if t is None and c.is_list:
+                t = c.offsets
             if t is not None:
                 dev = t.device
                 break
@@ -107,7 +109,7 @@ class AsyncExchange:
         val_bytes: Dict[str, torch.Tensor] = {}
         for f in fields:
             c = reordered.columns[f.name]
-            if c.is_string:
+            if c.is_string or c.is_list:
                 lens = (c.offsets[1:] - c.offsets[:-1]).to(torch.int64).contiguous()
                 self._lens[f.name] = lens
                 col_bytes[f.name] = (lens.view(torch.uint8).view(n, 8) if n else
@@ -167,7 +169,7 @@ class AsyncExchange:
         val_out: Dict[str, Optional[torch.Tensor]] = {}
         for f in fields:
             c = self.reordered.columns[f.name]
-            isz = 8 if c.is_string else c.data.element_size()
+            isz = 8 if (c.is_string or c.is_list) else c.data.element_size()
             data_out[f.name] = torch.empty(n_recv * isz, dtype=torch.uint8, device=self.device)
             val_out[f.name] = (torch.empty(n_recv, dtype=torch.uint8, device=self.device)
                                if f.nullable else None)
@@ -177,7 +179,7 @@ class AsyncExchange:
             r = self.recv_rows[s]
             for f in fields:
                 c = self.reordered.columns[f.name]
-                isz = 8 if c.is_string else c.data.element_size()
+                isz = 8 if (c.is_string or c.is_list) else c.data.element_size()
                 nb = r * isz
                 if nb:
                     data_out[f.name][rrow * isz: rrow * isz + nb] = self.packed_recv[roff:roff + nb]
@@ -188,34 +190,46 @@ class AsyncExchange:
                     roff += r
             rrow += r
 
-        # ---- second round: string payload bytes ---- #
+        # ---- second round: string/list payloads ---- #
         for f in fields:
             c = self.reordered.columns[f.name]
             v = val_out[f.name]
-            if not c.is_string:
+            if not (c.is_string or c.is_list):
                 d = data_out[f.name].view(c.data.dtype)
                 out_cols[f.name] = Column(f.dtype, data=d, validity=v)
                 continue
             recv_lens = data_out[f.name].view(torch.int64)
             lens = self._lens[f.name]
+            es = 1 if c.is_string else c.data.element_size()
             byte_send, off = [], 0
             for r in self.send_rows:
-                byte_send.append(int(lens[off:off + r].sum()) if r else 0)
+                byte_send.append(int(lens[off:off + r].sum()) * es if r else 0)
                 off += r
             byte_recv, off = [], 0
             for r in self.recv_rows:
-                byte_recv.append(int(recv_lens[off:off + r].sum()) if r else 0)
+                byte_recv.append(int(recv_lens[off:off + r].sum()) * es if r else 0)
                 off += r
+            payload = (c.bytes_ if c.is_string else
+                       c.data.contiguous().view(torch.uint8)
+                       if c.data.numel() else
+                       torch.empty(0, dtype=torch.uint8, device=self.device))
             recv_bytes = torch.empty(sum(byte_recv), dtype=torch.uint8, device=self.device)
             dist.all_to_all_single(
-                recv_bytes, self.reordered.columns[f.name].bytes_.contiguous(),
+                recv_bytes, payload.contiguous(),
                 byte_recv, byte_send, group=self._group,
             )
             new_offs = torch.zeros(n_recv + 1, dtype=torch.int64, device=self.device)
             if n_recv:
                 torch.cumsum(recv_lens, 0, out=new_offs[1:].view(-1))
-            out_cols[f.name] = Column(f.dtype, offsets=new_offs, bytes_=recv_bytes,
-                                      validity=v)
+            if c.is_string:
+                out_cols[f.name] = Column(f.dtype, offsets=new_offs,
+                                          bytes_=recv_bytes, validity=v)
+            else:
+                vals = (recv_bytes.view(c.data.dtype) if recv_bytes.numel()
+                        else torch.empty(0, dtype=c.data.dtype,
+                                         device=self.device))
+                out_cols[f.name] = Column(f.dtype, data=vals,
+                                          offsets=new_offs, validity=v)
         self._done = Batch(self.batch.schema, out_cols)
         # release references to in-flight buffers
         self._packed_send = None

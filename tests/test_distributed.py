@@ -210,6 +210,35 @@ def _run_exchange_hard(rank, world, rdv_file, tmpdir, results):
         t2 = torch.tensor([out2.num_rows])
         dist.all_reduce(t2)
         assert int(t2.item()) == world * n2
+
+        # case 3: list<float32> column (embedding tables through the
+        # dataset exchange)
+        schema3 = Schema([Field("id", "int64", False),
+                          Field("emb", "list<float32>", True)])
+        n3 = 48
+        ids3 = np.arange(rank * 100, rank * 100 + n3, dtype=np.int64)
+        b3 = Batch.from_dict({
+            "id": ids3,
+            "emb": [None if i % 7 == 0 else
+                    [float(ids3[i]), float(i % 3)] * (i % 2 + 1)
+                    for i in range(n3)],
+        }, schema3)
+        dest3 = torch.from_numpy((ids3 % world).astype(np.int64))
+        out3 = exchange_batch_all_to_all(b3, dest3)
+        gid = out3.columns["id"].data.numpy()
+        assert (gid % world == rank).all()
+        emb = out3.columns["emb"]
+        offs3 = emb.offsets.numpy()
+        vals3 = emb.data.numpy()
+        assert int(offs3[-1]) == len(vals3)
+        for i, g in enumerate(gid):
+            idx_src = int(g % 100)
+            if idx_src % 7 == 0:
+                assert emb.validity is not None and emb.validity[i] == 0
+            else:
+                got_list = vals3[offs3[i]:offs3[i + 1]].tolist()
+                expect = [float(g), float(idx_src % 3)] * (idx_src % 2 + 1)
+                assert got_list == expect, (g, got_list, expect)
         results[rank] = "ok"
     finally:
         dist.destroy_process_group()
