@@ -470,6 +470,7 @@ def test_gpu_zstd_scan_path_active(dev, tmp_path, monkeypatch):
     import lakesoul_amd.io.reader_gpu as rgpu
 
     monkeypatch.setattr(rgpu, "_GPU_ZSTD", True)
+    monkeypatch.setattr(rgpu, "_GPU_ZSTD_FRAC", 1.0)
     from lakesoul_amd.io.reader_gpu import fetch_raw
 
     catalog = _mk_catalog(tmp_path)
