@@ -92,8 +92,15 @@ class Column:
         return self.data.cpu().numpy()
 
     def to_device(self, device) -> "Column":
+        # async only when moving TO the GPU: a non_blocking D2H copy into
+        # pageable host memory returns before the bytes land, and callers
+        # read the numpy view immediately (this was a real corruption —
+        # batches read stale offsets when earlier GPU work kept the copy
+        # queue busy; AMD_SERIALIZE_COPY=3 made it vanish)
+        nb = torch.device(device).type == "cuda"
+
         def mv(t):
-            return None if t is None else t.to(device, non_blocking=True)
+            return None if t is None else t.to(device, non_blocking=nb)
 
         return Column(self.dtype, mv(self.data), mv(self.offsets), mv(self.bytes_), mv(self.validity))
 
