@@ -93,6 +93,19 @@ def create_s3_proxy(catalog, backend_scheme: str = "mock", secret: Optional[str]
     def get_metrics():
         return metrics.snapshot()
 
+    @app.get("/__metrics/prometheus")
+    def get_metrics_prom():
+        """Prometheus exposition format (reference s3-proxy exports
+        IntCounters, main.rs:44-52)."""
+        from fastapi.responses import PlainTextResponse
+
+        lines = []
+        for k, v in metrics.snapshot().items():
+            name = f"lakesoul_s3proxy_{k}"
+            lines.append(f"# TYPE {name} counter")
+            lines.append(f"{name} {v}")
+        return PlainTextResponse("\n".join(lines) + "\n")
+
     @app.get("/{bucket}/{key:path}")
     def get_object(bucket: str, key: str, request: Request):
         obj = _check(request, bucket, key)

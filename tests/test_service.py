@@ -188,3 +188,17 @@ def test_sql_endpoint(app_client):
     assert body["rows"][0][0] >= 0
     r = app_client.post("/sql", json={"query": "BOGUS"}, headers=hdr)
     assert r.status_code == 400
+
+
+def test_s3_proxy_prometheus_metrics(tmp_path, monkeypatch, catalog):
+    from fastapi.testclient import TestClient
+
+    from lakesoul_amd.service.s3_proxy import create_s3_proxy
+
+    monkeypatch.setenv("LAKESOUL_MOCK_FS_ROOT", str(tmp_path / "store"))
+    (tmp_path / "store").mkdir()
+    app = create_s3_proxy(catalog)
+    c = TestClient(app)
+    r = c.get("/__metrics/prometheus")
+    assert r.status_code == 200
+    assert "# TYPE lakesoul_s3proxy_" in r.text
