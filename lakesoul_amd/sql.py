@@ -159,7 +159,7 @@ class Query:
     table: str
     namespace: str = "default"
     alias: str = ""
-    join: Optional["JoinSpec"] = None
+    joins: List["JoinSpec"] = field(default_factory=list)
     items: List[SelectItem] = field(default_factory=list)
     where: Optional[Expr] = None
     group_by: List[str] = field(default_factory=list)
@@ -415,8 +415,10 @@ class _Parser:
             q.version = int(self.expect("num"))
         elif self.accept("kw", "timestamp"):  # FROM t TIMESTAMP <epoch_ms>
             q.timestamp_ms = int(self.expect("num"))
-        k, v = self.peek()
-        if (k == "kw" and v in ("join", "inner", "left")):
+        while True:
+            k, v = self.peek()
+            if not (k == "kw" and v in ("join", "inner", "left")):
+                break
             kind = "inner"
             if self.accept("kw", "left"):
                 kind = "left"
@@ -435,7 +437,7 @@ class _Parser:
                 j.on.append((lcol, rcol))
                 if not self.accept("kw", "and"):
                     break
-            q.join = j
+            q.joins.append(j)
         if self.accept("kw", "where"):
             q.where = self.or_expr()
         if self.accept("kw", "group"):
@@ -843,34 +845,37 @@ def _pd_eval(expr: Expr, df, col):
 
 
 def _execute_join_select(catalog, q: Query, device=None):
-    """Two-table equi-join (reference: lakesoul-datafusion delegates joins
-    to DataFusion; here a pandas hash join over two MOR scans)."""
+    """N-way equi-join (reference: lakesoul-datafusion delegates joins to
+    DataFusion; here hash joins folded left-to-right over MOR scans —
+    tensor engine by default, pandas as the LAKESOUL_SQL_PANDAS oracle).
+
+    Column naming in the joined frame: the first table to contribute a
+    name keeps it bare; later tables' clashing names are prefixed with
+    their qualifier ("<alias_or_table>.<col>")."""
     import pandas as pd
 
-    lt = catalog.table(q.table, q.namespace)
-    j = q.join
-    rt = catalog.table(j.table, j.namespace)
-    lnames = set(lt.schema.names())
-    rnames = set(rt.schema.names())
-    lquals = {q.alias or q.table, q.table}
-    rquals = {j.alias or j.table, j.table}
+    specs = [(q.namespace, q.table, q.alias or q.table, None)] + [
+        (j.namespace, j.table, j.alias or j.table, j) for j in q.joins]
+    tables = [catalog.table(name, ns) for ns, name, _, _ in specs]
+    names_per = [set(t.schema.names()) for t in tables]
+    quals = [{alias, name} for (_, name, alias, _) in specs]
 
     def side_of(name: str):
+        """-> (table index, bare column)."""
         if "." in name:
             qual, col = name.split(".", 1)
-            if qual in lquals and col in lnames:
-                return "L", col
-            if qual in rquals and col in rnames:
-                return "R", col
+            for ti in range(len(specs)):
+                if qual in quals[ti] and col in names_per[ti]:
+                    return ti, col
             raise SqlError(f"cannot resolve {name!r}")
-        amb = (name in lnames) + (name in rnames)
-        if amb == 0:
+        hits = [ti for ti in range(len(specs)) if name in names_per[ti]]
+        if not hits:
             raise SqlError(f"unknown column {name!r}")
-        if amb == 2:
+        if len(hits) > 1:
             raise SqlError(f"ambiguous column {name!r}: qualify it")
-        return ("L", name) if name in lnames else ("R", name)
+        return hits[0], name
 
-    # referenced columns per side
+    # referenced columns per table
     refs = []
     star = any(it.kind == "star" for it in q.items)
     out_aliases = {it.out_name for it in q.items} | {
@@ -880,79 +885,99 @@ def _execute_join_select(catalog, q: Query, device=None):
             refs.append(it.name)
         if it.expr is not None:
             refs += list(it.expr.columns())
-    refs += q.group_by + [n for n, _ in q.order_by
-                          if n not in out_aliases]
+    refs += q.group_by + [n for n, _ in q.order_by if n not in out_aliases]
     if q.where is not None:
         refs += list(q.where.columns())
-    for lcol, rcol in j.on:
-        refs += [lcol, rcol]
-    need_l, need_r = set(), set()
+    for j in q.joins:
+        for a_col, b_col in j.on:
+            refs += [a_col, b_col]
+    need = [set() for _ in specs]
     for r in refs:
-        side, col = side_of(r)
-        (need_l if side == "L" else need_r).add(col)
+        ti, col = side_of(r)
+        need[ti].add(col)
     if star:
-        need_l, need_r = set(lnames), set(rnames)
+        need = [set(n) for n in names_per]
 
-    lscan = lt.scan(columns=sorted(need_l) or None, version=q.version,
-                    device=device)
-    rscan = rt.scan(columns=sorted(need_r) or None, device=device)
-    rqual = j.alias or j.table
+    # frame column mapping: first-come keeps the bare name
+    frame_name = {}   # (ti, col) -> column name in the joined frame
+    taken = set()
+    for ti in range(len(specs)):
+        for col in sorted(need[ti]):
+            if col in taken:
+                frame_name[(ti, col)] = f"{specs[ti][2]}.{col}"
+            else:
+                frame_name[(ti, col)] = col
+                taken.add(col)
 
-    # join key column names per side
-    key_l, key_r = [], []
-    for lcol, rcol in j.on:
-        sl, cl = side_of(lcol)
-        sr, cr = side_of(rcol)
-        if sl == sr:
-            raise SqlError("JOIN ON must reference both tables")
-        if sl == "R":
-            cl, cr = cr, cl
-        key_l.append(cl)
-        key_r.append(cr)
+    def df_col(name: str) -> str:
+        ti, col = side_of(name)
+        return frame_name[(ti, col)]
+
+    scans = [
+        tables[ti].scan(columns=sorted(need[ti]) or None,
+                        version=q.version if ti == 0 else None,
+                        device=device)
+        for ti in range(len(specs))
+    ]
+
+    def resolve_on(j: JoinSpec, new_ti: int):
+        """ON pairs -> (acc-frame key names, new-table bare key names)."""
+        acc_keys, new_keys = [], []
+        for a_col, b_col in j.on:
+            sa, ca = side_of(a_col)
+            sb, cb = side_of(b_col)
+            if sa == new_ti and sb != new_ti:
+                sa, ca, sb, cb = sb, cb, sa, ca
+            if sb != new_ti or sa == new_ti:
+                raise SqlError(
+                    "JOIN ON must link the new table to an earlier one")
+            acc_keys.append(frame_name[(sa, ca)])
+            new_keys.append(cb)
+        return acc_keys, new_keys
 
     if _use_pandas_exec():
-        ldf = lscan.to_arrow().to_pandas()
-        rdf = rscan.to_arrow().to_pandas()
-        overlap = set(ldf.columns) & set(rdf.columns)
-        rdf = rdf.rename(columns={c: f"{rqual}.{c}" for c in overlap})
-
-        def df_col(name: str) -> str:
-            side, col = side_of(name)
-            if side == "R" and col in overlap:
-                return f"{rqual}.{col}"
-            return col
-
-        right_on = [f"{rqual}.{c}" if c in overlap else c for c in key_r]
-        out = ldf.merge(rdf, how=j.kind, left_on=key_l, right_on=right_on)
+        acc = scans[0].to_arrow().to_pandas()
+        acc.columns = [frame_name[(0, c)] for c in acc.columns]
+        for ti in range(1, len(specs)):
+            rdf = scans[ti].to_arrow().to_pandas()
+            rdf.columns = [frame_name[(ti, c)] for c in rdf.columns]
+            acc_keys, new_keys = resolve_on(specs[ti][3], ti)
+            right_on = [frame_name[(ti, c)] for c in new_keys]
+            acc = acc.merge(rdf, how=specs[ti][3].kind,
+                            left_on=acc_keys, right_on=right_on)
         if q.where is not None:
-            out = out[_pd_eval(q.where, out, df_col)].reset_index(drop=True)
-        return _project_and_finish(q, out, df_col,
-                                   all_cols=list(ldf.columns) + list(rdf.columns))
+            acc = acc[_pd_eval(q.where, acc, df_col)].reset_index(drop=True)
+        return _project_and_finish(q, acc, df_col, all_cols=list(acc.columns))
 
     # tensor engine path (vectorized hash join, query/engine.py)
     import torch as _torch
 
+    from .io.batch import Batch as _B
+    from .io.schema import Field as _F
+    from .io.schema import Schema as _S
     from .query.engine import join_batches
 
-    lb = lscan.to_batch()
-    rb = rscan.to_batch()
-    overlap = set(lb.schema.names()) & set(rb.schema.names())
-    rename = {c: f"{rqual}.{c}" for c in overlap}
+    def renamed_batch(ti):
+        b = scans[ti].to_batch()
+        fields, cols = [], {}
+        for f in b.schema:
+            nn = frame_name[(ti, f.name)]
+            fields.append(_F(nn, f.dtype, f.nullable))
+            cols[nn] = b.columns[f.name]
+        return _B(_S(fields), cols)
 
-    def df_col(name: str) -> str:
-        side, col = side_of(name)
-        if side == "R" and col in overlap:
-            return f"{rqual}.{col}"
-        return col
-
-    joined = join_batches(lb, rb, key_l, key_r, j.kind, rename_right=rename)
+    acc = renamed_batch(0)
+    for ti in range(1, len(specs)):
+        rb = renamed_batch(ti)
+        acc_keys, new_keys = resolve_on(specs[ti][3], ti)
+        right_on = [frame_name[(ti, c)] for c in new_keys]
+        acc = join_batches(acc, rb, acc_keys, right_on, specs[ti][3].kind)
     if q.where is not None:
-        view = _expr_view_batch(joined, q.where, df_col)
+        view = _expr_view_batch(acc, q.where, df_col)
         mask = q.where.evaluate(view)
-        joined = joined.take(_torch.nonzero(mask, as_tuple=False).flatten())
+        acc = acc.take(_torch.nonzero(mask, as_tuple=False).flatten())
     return _project_and_finish_tensor(
-        q, joined, df_col,
-        all_cols=list(joined.schema.names()))
+        q, acc, df_col, all_cols=list(acc.schema.names()))
 
 
 def _eval_scalar_tensor(e: Scalar, batch, col):
@@ -1325,8 +1350,8 @@ def _explain_select(catalog, q: Query, device=None):
 
     q.where = _resolve_subqueries(catalog, q.where, device)
     rows = []
-    for tbl_name, ns, flt in ((q.table, q.namespace, q.where),) + (
-        ((q.join.table, q.join.namespace, None),) if q.join else ()
+    for tbl_name, ns, flt in ((q.table, q.namespace, q.where),) + tuple(
+        (j.table, j.namespace, None) for j in q.joins
     ):
         t = catalog.table(tbl_name, ns)
         scan = t.scan(filters=flt, version=q.version,
@@ -1379,7 +1404,7 @@ def _execute_select(catalog, q: Query, device=None):
     import pandas as pd
 
     q.where = _resolve_subqueries(catalog, q.where, device)
-    if q.join is not None:
+    if q.joins:
         return _execute_join_select(catalog, q, device=device)
     t = catalog.table(q.table, q.namespace)
     _strip_quals(q, {q.table, q.alias} - {""})

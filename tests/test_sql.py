@@ -459,6 +459,46 @@ def test_join_on_tensor_engine_matches_pandas(catalog, monkeypatch):
     assert a2["name"].iloc[3] is None or pd.isna(a2["name"].iloc[3])
 
 
+def test_three_way_join(catalog, monkeypatch):
+    """N-way join fold (the real TPC-H q3 shape: customer x orders x
+    lineitem), tensor engine vs pandas oracle."""
+    execute_sql(catalog,
+        "CREATE TABLE cust3 (cid BIGINT NOT NULL, seg VARCHAR(8)) "
+        "PRIMARY KEY (cid) HASH BUCKETS 2")
+    execute_sql(catalog,
+        "CREATE TABLE ord3 (oid BIGINT NOT NULL, cid BIGINT, odate BIGINT) "
+        "PRIMARY KEY (oid) HASH BUCKETS 2")
+    execute_sql(catalog,
+        "CREATE TABLE li3 (lid BIGINT NOT NULL, oid BIGINT, amt DOUBLE) "
+        "PRIMARY KEY (lid) HASH BUCKETS 2")
+    execute_sql(catalog, "INSERT INTO cust3 VALUES (1,'a'),(2,'b'),(3,'a')")
+    execute_sql(catalog,
+        "INSERT INTO ord3 VALUES (10,1,100),(11,2,200),(12,3,300),(13,1,400)")
+    execute_sql(catalog,
+        "INSERT INTO li3 VALUES (7,10,5.0),(8,10,6.0),(9,11,2.0),(14,13,1.0),"
+        "(15,12,9.0)")
+    sql = ("SELECT c.seg, sum(l.amt) total, count(*) n "
+           "FROM li3 l JOIN ord3 o ON l.oid = o.oid "
+           "JOIN cust3 c ON o.cid = c.cid "
+           "WHERE o.odate < 350 GROUP BY c.seg ORDER BY c.seg")
+    a = execute_sql(catalog, sql)
+    assert a["seg"].tolist() == ["a", "b"]
+    # seg 'a': orders 10 (amt 5+6) + 12 (9.0) -> 20.0 ; 13 excluded (odate 400)
+    np.testing.assert_allclose(a["total"].to_numpy(), [20.0, 2.0])
+    assert a["n"].tolist() == [3, 1]
+    monkeypatch.setenv("LAKESOUL_SQL_PANDAS", "1")
+    b = execute_sql(catalog, sql)
+    monkeypatch.delenv("LAKESOUL_SQL_PANDAS", raising=False)
+    assert a["seg"].tolist() == b["seg"].tolist()
+    np.testing.assert_allclose(a["total"].to_numpy(), b["total"].to_numpy())
+    # bare-name disambiguation across three tables
+    sql2 = ("SELECT lid, o.oid, o.cid FROM li3 JOIN ord3 o ON li3.oid = o.oid "
+            "JOIN cust3 ON o.cid = cust3.cid ORDER BY lid")
+    d2 = execute_sql(catalog, sql2)
+    assert d2["lid"].tolist() == [7, 8, 9, 14, 15]
+    assert d2["cid"].tolist() == [1, 1, 2, 1, 3]
+
+
 def test_insert_null_values_and_decimal(catalog):
     """NULL in INSERT VALUES and INSERT..SELECT from a nullable decimal
     column (Decimal(str(None)) crash + literal() NULL)."""
