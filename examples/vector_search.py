@@ -52,4 +52,19 @@ ids_bin, _ = bin_idx.search(q, k=5, rescore=32)
 print("1-bit top-5:", ids_bin.tolist())
 assert ids_bin[0, 0] == 7
 
+# IVF-RaBitQ (1 sign bit + 3 ex bits, the reference's quantizer) with
+# staged search: fastscan estimate -> ex refine -> exact MFMA rescore
+rbq = build_vector_index(docs, "emb", metric="cosine", rabitq_bits=4,
+                         ivf_clusters=32)
+ids_rbq, _ = rbq.search(q, k=5, rescore=40)
+print("rabitq top-5:", ids_rbq.tolist())
+assert ids_rbq[0, 0] == 7
+
+# ANN results as a table scan filter (reader.rs:250-331 analog)
+hits = docs.scan(columns=["doc_id"],
+                 vector_query={"column": "emb", "query": vecs[7], "k": 5})
+df = hits.to_arrow().to_pandas()
+print("scan(vector_query) rows:", sorted(df["doc_id"].tolist()))
+assert 7 in set(df["doc_id"].tolist())
+
 print("vector search OK —", workdir)
