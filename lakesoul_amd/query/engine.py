@@ -193,7 +193,9 @@ def groupby_agg(batch: Batch, group_cols: Sequence[str],
     def seg_sum(v64, valid_mask, zero):
         if small_g:
             sel = torch.where(valid_mask, v64, zero)
-            return (group_masks.to(v64.dtype) @ sel)
+            return torch.stack([
+                torch.where(group_masks[j], sel, zero).sum()
+                for j in range(g)])
         s = torch.zeros(g, dtype=v64.dtype, device=codes.device)
         s.index_add_(0, codes, torch.where(valid_mask, v64, zero))
         return s
@@ -237,9 +239,17 @@ def groupby_agg(batch: Batch, group_cols: Sequence[str],
         V = torch.stack(vstack, dim=1)           # (n, m)
         M = torch.stack(vmstack, dim=1)          # (n, m)
         if small_g:
-            gm = group_masks.to(torch.float64)   # (g, n)
-            S = gm @ V                           # (g, m)
-            Cn = gm @ M
+            # per-group masked column sums: g clean bandwidth passes over
+            # (n, m) — no atomics, and no skinny rocBLAS dgemm (a (g,n) x
+            # (n,m) f64 GEMM with g=6 measured 4x SLOWER than these
+            # passes on MI355X)
+            zero2 = torch.zeros_like(V)
+            S = torch.stack([
+                torch.where(group_masks[j].unsqueeze(1), V, zero2).sum(0)
+                for j in range(g)])
+            Cn = torch.stack([
+                torch.where(group_masks[j].unsqueeze(1), M, zero2).sum(0)
+                for j in range(g)])
         else:
             S = torch.zeros(g, V.shape[1], dtype=torch.float64,
                             device=codes.device)
