@@ -215,50 +215,8 @@ def groupby_agg(batch: Batch, group_cols: Sequence[str],
     counts_all = (group_masks.sum(1) if small_g
                   else torch.bincount(codes, minlength=g))
 
-    # ---- batched sum/avg: one stacked reduction instead of one pass per
-    # aggregate (q1-shape queries run 4-8 sums over the same codes) ---- #
     _batched_sums: Dict[int, torch.Tensor] = {}
     _batched_cnts: Dict[int, torch.Tensor] = {}
-    sum_jobs = []
-    for ai, (fn, cname, out_name, distinct) in enumerate(aggs):
-        if fn in ("sum", "avg") and cname is not None:
-            # float columns only: f64 accumulation matches the unbatched
-            # path exactly there; integer/decimal sums keep their exact
-            # int64 accumulators
-            if batch.schema.field(cname).dtype in _FLOAT_DTYPES:
-                sum_jobs.append(ai)
-    if len(sum_jobs) > 1 and n:
-        vstack, vmstack = [], []
-        for ai in sum_jobs:
-            _, cname, _, _ = aggs[ai]
-            colj = batch.columns[cname]
-            vj = colj.data.to(torch.float64)
-            mj = _valid_mask(colj, n)
-            vstack.append(torch.where(mj, vj, torch.zeros_like(vj)))
-            vmstack.append(mj.to(torch.float64))
-        V = torch.stack(vstack, dim=1)           # (n, m)
-        M = torch.stack(vmstack, dim=1)          # (n, m)
-        if small_g:
-            # per-group masked column sums: g clean bandwidth passes over
-            # (n, m) — no atomics, and no skinny rocBLAS dgemm (a (g,n) x
-            # (n,m) f64 GEMM with g=6 measured 4x SLOWER than these
-            # passes on MI355X)
-            zero2 = torch.zeros_like(V)
-            S = torch.stack([
-                torch.where(group_masks[j].unsqueeze(1), V, zero2).sum(0)
-                for j in range(g)])
-            Cn = torch.stack([
-                torch.where(group_masks[j].unsqueeze(1), M, zero2).sum(0)
-                for j in range(g)])
-        else:
-            S = torch.zeros(g, V.shape[1], dtype=torch.float64,
-                            device=codes.device)
-            S.index_add_(0, codes, V)
-            Cn = torch.zeros_like(S)
-            Cn.index_add_(0, codes, M)
-        for j, ai in enumerate(sum_jobs):
-            _batched_sums[ai] = S[:, j]
-            _batched_cnts[ai] = Cn[:, j].to(torch.int64)
 
     for agg_i, (fn, cname, out_name, distinct) in enumerate(aggs):
         if fn == "count" and cname is None:
