@@ -685,6 +685,7 @@ static py::dict read_unit_raw_py(const std::vector<std::string>& paths,
     cd["name"] = c.name;
     cd["present"] = c.present;
     cd["is_string"] = c.is_string;
+    cd["is_list"] = c.is_list;
     cd["is_dict"] = c.is_dict;
     cd["num_values"] = c.num_values;
     cd["null_count"] = c.null_count;

@@ -104,6 +104,8 @@ struct UnitColumn {
   int64_t dense_n = 0;
   int64_t soff_off = -1;
   int64_t sbytes_off = 0, sbytes_len = 0;
+  bool is_list = false;  // element values in the sbytes region, element
+                         // offsets in soffs, ROW validity in validity
 };
 
 inline int64_t ru_align8(int64_t x) { return (x + 7) & ~7LL; }
@@ -117,6 +119,7 @@ struct UnitStage {
   struct StrDecoded {
     std::vector<int64_t> offs;
     std::vector<uint8_t> bytes;
+    std::vector<uint8_t> row_valid;  // lists: per-row validity (may be empty)
   };
 
   int64_t t_open_us = 0, t_chunks_us = 0, t_layout_us = 0;
@@ -217,8 +220,37 @@ inline std::unique_ptr<UnitStage> read_unit_stage1(
       uc.present = true;
       const ColumnDesc& cd = fd.f->columns()[fd.col_idx[c]];
       uc.physical = cd.physical;
-      uc.is_string = cd.physical == PT_BYTE_ARRAY;
+      uc.is_list = cd.is_list;
+      uc.is_string = !cd.is_list && cd.physical == PT_BYTE_ARRAY;
       auto& chs = fd.chunks[c];
+      if (uc.is_list) {
+        // rows (not element slots) are the unit currency for lists;
+        // payload layout happens in the decoded pre-pass below (same
+        // host-decode lane as strings)
+        int64_t rows = 0;
+        for (auto& ch : chs)
+          rows += ch.list_offsets.empty() ? 0
+                  : (int64_t)ch.list_offsets.size() - 1;
+        uc.num_values = rows;
+        bool any_null_row = false;
+        for (auto& ch : chs)
+          for (auto v : ch.list_validity)
+            if (!v) any_null_row = true;
+        uc.null_count = 0;
+        for (auto& ch : chs) {
+          size_t rows_c = ch.list_offsets.empty() ? 0 : ch.list_offsets.size() - 1;
+          for (size_t i = 0; i < rows_c && i < ch.list_validity.size(); i++)
+            if (!ch.list_validity[i]) uc.null_count++;
+        }
+        if (any_null_row) {
+          uc.validity_off = vapos;
+          vapos += rows;
+        }
+        uc.soff_off = spos;
+        spos += rows + 1;
+        S.cols.push_back(uc);
+        continue;
+      }
       int64_t nv = 0, nulls = 0;
       bool any_dict = false, any_valid = false;
       for (auto& ch : chs) {
@@ -314,7 +346,8 @@ inline std::unique_ptr<UnitStage> read_unit_stage1(
   {
     std::vector<size_t> str_idx;
     for (size_t u = 0; u < S.cols.size(); u++)
-      if (S.cols[u].present && S.cols[u].is_string) str_idx.push_back(u);
+      if (S.cols[u].present && (S.cols[u].is_string || S.cols[u].is_list))
+        str_idx.push_back(u);
     std::string err;
     std::mutex err_mu;
     ThreadPool::instance().parallel_for((int64_t)str_idx.size(), [&](int64_t k) {
@@ -324,12 +357,31 @@ inline std::unique_ptr<UnitStage> read_unit_stage1(
         auto& fd = S.files[uc.file_idx];
         auto sd = std::make_unique<UnitStage::StrDecoded>();
         sd->offs.push_back(0);
-        for (auto& ch : fd.chunks[u % S.ncols]) {
-          DecodedColumn dc = decode_chunk_cpu(ch);
-          int64_t base = (int64_t)sd->bytes.size();
-          sd->bytes.insert(sd->bytes.end(), dc.bytes.begin(), dc.bytes.end());
-          for (size_t i = 1; i < dc.offsets.size(); i++)
-            sd->offs.push_back(base + dc.offsets[i]);
+        if (uc.is_list) {
+          int es = physical_elem_size(uc.physical);
+          for (auto& ch : fd.chunks[u % S.ncols]) {
+            DecodedColumn dc = decode_chunk_cpu(ch);
+            int64_t elem_base = (int64_t)sd->bytes.size() / es;
+            sd->bytes.insert(sd->bytes.end(), dc.data.begin(), dc.data.end());
+            for (size_t i = 1; i < dc.list_offsets.size(); i++)
+              sd->offs.push_back(elem_base + dc.list_offsets[i]);
+            size_t rows = dc.list_offsets.empty()
+                              ? 0 : dc.list_offsets.size() - 1;
+            if (!dc.list_validity.empty())
+              sd->row_valid.insert(sd->row_valid.end(),
+                                   dc.list_validity.begin(),
+                                   dc.list_validity.begin() + rows);
+            else
+              sd->row_valid.insert(sd->row_valid.end(), rows, 1);
+          }
+        } else {
+          for (auto& ch : fd.chunks[u % S.ncols]) {
+            DecodedColumn dc = decode_chunk_cpu(ch);
+            int64_t base = (int64_t)sd->bytes.size();
+            sd->bytes.insert(sd->bytes.end(), dc.bytes.begin(), dc.bytes.end());
+            for (size_t i = 1; i < dc.offsets.size(); i++)
+              sd->offs.push_back(base + dc.offsets[i]);
+          }
         }
         S.str_cols[u] = std::move(sd);
       } catch (std::exception& e) {
@@ -379,7 +431,7 @@ inline void read_unit_fill(UnitStage& S, uint8_t* values, uint8_t* validity,
     if (!uc.present) return;
     auto& fd = S.files[uc.file_idx];
     auto& chs = fd.chunks[u % S.ncols];
-    if (uc.validity_off >= 0) {
+    if (uc.validity_off >= 0 && !uc.is_list) {
       int64_t off = uc.validity_off;
       for (auto& ch : chs) {
         if (!ch.validity.empty())
@@ -387,11 +439,14 @@ inline void read_unit_fill(UnitStage& S, uint8_t* values, uint8_t* validity,
         off += ch.num_values;
       }
     }
-    if (uc.is_string) {
+    if (uc.is_string || uc.is_list) {
       auto& sd = *S.str_cols[u];
       std::memcpy(soffs + uc.soff_off, sd.offs.data(), sd.offs.size() * 8);
       if (!sd.bytes.empty())
         std::memcpy(values + uc.sbytes_off, sd.bytes.data(), sd.bytes.size());
+      if (uc.is_list && uc.validity_off >= 0 && !sd.row_valid.empty())
+        std::memcpy(validity + uc.validity_off, sd.row_valid.data(),
+                    sd.row_valid.size());
       return;
     }
     if (uc.is_dict) {

@@ -508,14 +508,10 @@ class LakeSoulScan:
 
                 warnings.warn(f"chunked merge unavailable ({e}); "
                               "falling back")
-        has_list = any(self.schema.field(n).dtype.startswith("list<")
-                       for n in self.read_cols)
         if self.device == "cuda" and (
-            oversized or has_list
+            oversized
             or (needs_merge and not self._gpu_merge_supported())
         ):
-            # (list<T> columns decode on host for now; the merged batch
-            # ships to HBM — GPU list decode is a follow-up)
             # hybrid: CPU decode+merge (host RAM), then ship the merged
             # batch to HBM — for string-PK merges and for buckets whose
             # working set would not fit the HBM budget (the merged output

@@ -222,6 +222,21 @@ def read_unit_gpu(scan, unit, raw: Optional[dict] = None,
             raise RuntimeError(f"GPU zstd decompression failed: status={int((status != 0).sum())} pages")
 
     read_schema = scan.schema.select(names)
+    # list<T> columns travel the unit as BYTE-offset binary (element
+    # payload in the string-bytes region) so the merge machinery treats
+    # whole lists opaquely (UseLast whole-value, merge/mod.rs:65-89);
+    # converted back to element-offset list columns at the end
+    has_list = any(f.dtype.startswith("list<") for f in read_schema)
+    if has_list:
+        from .schema import Field as _Fld
+        from .schema import Schema as _Sch
+
+        work_schema = _Sch([
+            _Fld(f.name, "binary" if f.dtype.startswith("list<") else f.dtype,
+                 f.nullable)
+            for f in read_schema])
+    else:
+        work_schema = read_schema
     ncols = len(names)
     empty_u8 = torch.empty(0, dtype=torch.uint8, device=device)
     empty_i64 = torch.empty(0, dtype=torch.int64, device=device)
@@ -232,6 +247,7 @@ def read_unit_gpu(scan, unit, raw: Optional[dict] = None,
     desc = raw.get("desc")
     if (
         desc is not None
+        and not has_list
         and nfiles > 1
         and len(scan.pk) == 1
         and not scan.merge_ops
@@ -287,10 +303,11 @@ def read_unit_gpu(scan, unit, raw: Optional[dict] = None,
         for ci, name in enumerate(names):
             cd = raw["cols"][fi * ncols + ci]
             f = scan.schema.field(name)
+            is_list_f = f.dtype.startswith("list<")
             if not cd["present"]:
-                if f.dtype in ("string", "binary"):
+                if f.dtype in ("string", "binary") or is_list_f:
                     cols[name] = Column(
-                        f.dtype,
+                        "binary" if is_list_f else f.dtype,
                         offsets=torch.zeros(nrows + 1, dtype=torch.int64, device=device),
                         bytes_=empty_u8,
                         validity=torch.zeros(nrows, dtype=torch.uint8, device=device),
@@ -308,6 +325,13 @@ def read_unit_gpu(scan, unit, raw: Optional[dict] = None,
             if cd["validity_off"] >= 0 and cd["null_count"] > 0:
                 vmask = validity_buf.narrow(0, cd["validity_off"], nv)
 
+            if cd.get("is_list"):
+                es = _ESIZE[_torch_view(f.dtype[5:-1])]
+                offs = soffs_buf.narrow(0, cd["soff_off"], nv + 1).to(torch.int64) * es
+                by = vals.narrow(0, cd["sbytes_off"], cd["sbytes_len"])
+                cols[name] = Column("binary", offsets=offs, bytes_=by,
+                                    validity=vmask)
+                continue
             if cd["is_string"]:
                 offs = soffs_buf.narrow(0, cd["soff_off"], nv + 1)
                 by = vals.narrow(0, cd["sbytes_off"], cd["sbytes_len"])
@@ -339,7 +363,7 @@ def read_unit_gpu(scan, unit, raw: Optional[dict] = None,
             if f.dtype in _TARGET:
                 data = data.to(_TARGET[f.dtype])
             cols[name] = Column(f.dtype, data=data, validity=vmask)
-        file_batches.append(Batch(read_schema, cols))
+        file_batches.append(Batch(work_schema, cols))
         present.append(pres)
     _dec.__exit__(None, None, None)
 
@@ -363,6 +387,16 @@ def read_unit_gpu(scan, unit, raw: Optional[dict] = None,
     for f in scan.eval_schema:
         if f.name in scan.range_cols:
             cols[f.name] = _range_col_gpu(scan, f, unit, nrows, device)
+        elif f.dtype.startswith("list<"):
+            c = merged.columns[f.name]
+            es = _ESIZE[_torch_view(f.dtype[5:-1])]
+            vals_t = (c.bytes_.view(torch_dtype_for(f.dtype[5:-1]))
+                      if c.bytes_.numel()
+                      else torch.empty(0, dtype=torch_dtype_for(f.dtype[5:-1]),
+                                       device=device))
+            cols[f.name] = Column(f.dtype, data=vals_t,
+                                  offsets=c.offsets.to(torch.int64) // es,
+                                  validity=c.validity)
         else:
             cols[f.name] = merged.columns[f.name]
     return Batch(scan.eval_schema, cols)

@@ -605,3 +605,40 @@ def test_gpu_range_partitioned_mor(dev, tmp_path):
     expect = np.zeros(n)
     expect[::4] = 1.0
     np.testing.assert_allclose(gpu["v"].to_numpy(), expect)
+
+
+def test_gpu_list_column_mor(dev, tmp_path):
+    """list<float32> columns scan through the GPU unit path (element
+    payload in the string lane, whole-value UseLast merge on device):
+    equal to the CPU scan, incl. null lists and updates."""
+    catalog = _mk_catalog(tmp_path)
+    from lakesoul_amd.io.schema import Field, Schema
+
+    t = catalog.create_table(
+        "gl",
+        Schema([Field("id", "int64", False), Field("emb", "list<float32>"),
+                Field("v", "float64")]),
+        primary_keys=["id"], hash_bucket_num=2,
+    )
+    n = 5000
+    base = [None if i % 11 == 0 else
+            list(np.float32(i) + np.arange(i % 4, dtype=np.float32))
+            for i in range(n)]
+    t.upsert({"id": np.arange(n, dtype=np.int64), "emb": base,
+              "v": np.arange(n, dtype=np.float64)})
+    t.upsert({"id": np.array([3, 44], dtype=np.int64),
+              "emb": [[9.0], None], "v": np.array([3.5, 44.5])})
+    cpu = t.scan(device="cpu").to_arrow().to_pandas().sort_values("id").reset_index(drop=True)
+    gpu = t.scan(device="cuda").to_arrow().to_pandas().sort_values("id").reset_index(drop=True)
+    assert len(gpu) == n
+    import pandas as pd
+
+    for i in range(n):
+        a, b = cpu["emb"].iloc[i], gpu["emb"].iloc[i]
+        a_null = a is None or (isinstance(a, float) and np.isnan(a))
+        b_null = b is None or (isinstance(b, float) and np.isnan(b))
+        assert a_null == b_null, i
+        if not a_null:
+            np.testing.assert_allclose(list(b), list(a), err_msg=str(i))
+    np.testing.assert_allclose(gpu["v"].to_numpy(), cpu["v"].to_numpy())
+    np.testing.assert_allclose(list(gpu["emb"].iloc[3]), [9.0])
