@@ -222,6 +222,14 @@ def create_app(catalog=None, secret: Optional[str] = None):
     def get_metrics(claims: dict = Depends(auth)):
         return metrics.snapshot()
 
+    @app.get("/metrics/memory")
+    def get_metrics_memory(claims: dict = Depends(auth)):
+        """Host RSS + HBM allocator snapshot (jemalloc-prof analog,
+        reference mem.rs:1-33)."""
+        from ..utils import memprof
+
+        return memprof.snapshot()
+
     @app.get("/metrics/prometheus")
     def get_metrics_prom(claims: dict = Depends(auth)):
         """Prometheus exposition format (reference: the s3-proxy and

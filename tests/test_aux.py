@@ -279,3 +279,18 @@ def test_notify_driven_compaction(catalog):
         assert len(bus.published) == 1
     finally:
         catalog.client.notify_bus = None
+
+
+def test_memprof_snapshot_and_track(capsys):
+    """Memory accounting hooks (reference mem.rs jemalloc prof +
+    LoggedMemoryPool analog)."""
+    from lakesoul_amd.utils import memprof
+
+    s = memprof.snapshot()
+    assert s["rss"] > 0 and s["peak_rss"] >= s["rss"]
+    assert "hbm_allocated" in s
+    msgs = []
+    with memprof.track("alloc-test", log=msgs.append):
+        buf = bytearray(50 * 1024 * 1024)
+        assert len(buf) == 50 * 1024 * 1024
+    assert len(msgs) == 1 and "alloc-test" in msgs[0]
