@@ -340,3 +340,199 @@ extern "C" int lakesoul_meta_commit_add_files(void* hp, const char* table_id,
     throw std::runtime_error("commit_add_files: CAS retries exhausted");
   });
 }
+
+
+// ===================================================================== //
+// extended DAO surface (reference lakesoul-metadata-c lib.rs DaoType
+// families: namespace/table listing, partition listing, versioned +
+// incremental snapshot queries, lookup by path)
+// ===================================================================== //
+
+namespace {
+
+std::string rows_to_json_array(sqlite3* db, const std::string& sql,
+                               const std::vector<std::string>& binds) {
+  sqlite3_stmt* st = nullptr;
+  if (sqlite3_prepare_v2(db, sql.c_str(), -1, &st, nullptr) != SQLITE_OK)
+    throw std::runtime_error("sqlite prepare failed: " + sql);
+  for (size_t i = 0; i < binds.size(); i++)
+    sqlite3_bind_text(st, (int)i + 1, binds[i].c_str(), -1, nullptr);
+  std::string out = "[";
+  bool first = true;
+  while (sqlite3_step(st) == SQLITE_ROW) {
+    const unsigned char* t = sqlite3_column_text(st, 0);
+    if (!first) out += ",";
+    first = false;
+    out += "\"" + jesc(t ? (const char*)t : "") + "\"";
+  }
+  sqlite3_finalize(st);
+  out += "]";
+  return out;
+}
+
+// resolve a snapshot (JSON list of commit ids) to surviving files
+std::string resolve_snapshot_files(sqlite3* db, const std::string& table_id,
+                                   const std::string& desc,
+                                   const std::string& snap) {
+  std::vector<std::pair<std::string, int64_t>> files;
+  for (auto& cid : parse_str_list(snap)) {
+    bool f2 = false;
+    std::string ops = q1(db,
+        "SELECT file_ops FROM data_commit_info WHERE table_id=? AND"
+        " partition_desc=? AND commit_id=?",
+        {table_id, desc, cid}, &f2);
+    if (!f2) continue;
+    size_t i = 0;
+    while ((i = ops.find("{", i)) != std::string::npos) {
+      size_t e = ops.find("}", i);
+      std::string obj = ops.substr(i, e - i);
+      auto grab = [&](const char* key) {
+        size_t k = obj.find(std::string("\"") + key + "\"");
+        if (k == std::string::npos) return std::string();
+        k = obj.find(':', k) + 1;
+        while (k < obj.size() && (obj[k] == ' ')) k++;
+        if (obj[k] == '"') {
+          size_t q = obj.find('"', k + 1);
+          return obj.substr(k + 1, q - k - 1);
+        }
+        size_t q = obj.find_first_of(",}", k);
+        return obj.substr(k, q - k);
+      };
+      std::string path = grab("path");
+      std::string op = grab("file_op");
+      std::string sz = grab("size");
+      if (op == "del") {
+        for (auto it = files.begin(); it != files.end(); ++it)
+          if (it->first == path) { files.erase(it); break; }
+      } else {
+        files.push_back({path, sz.empty() ? 0 : atoll(sz.c_str())});
+      }
+      i = e + 1;
+    }
+  }
+  std::string out = "[";
+  bool first = true;
+  for (auto& f : files) {
+    if (!first) out += ",";
+    first = false;
+    out += "{\"path\":\"" + jesc(f.first) +
+           "\",\"size\":" + std::to_string(f.second) + "}";
+  }
+  out += "]";
+  return out;
+}
+
+}  // namespace
+
+extern "C" char* lakesoul_meta_list_namespaces(void* hp) {
+  auto* h = (MetaHandle*)hp;
+  try {
+    return strdup(rows_to_json_array(
+        h->db, "SELECT namespace FROM namespace ORDER BY namespace", {})
+        .c_str());
+  } catch (std::exception& e) { g_meta_err = e.what(); return nullptr; }
+}
+
+extern "C" char* lakesoul_meta_list_tables(void* hp, const char* ns) {
+  auto* h = (MetaHandle*)hp;
+  try {
+    return strdup(rows_to_json_array(
+        h->db,
+        "SELECT table_name FROM table_info WHERE table_namespace=?"
+        " ORDER BY table_name",
+        {ns && *ns ? ns : "default"}).c_str());
+  } catch (std::exception& e) { g_meta_err = e.what(); return nullptr; }
+}
+
+extern "C" char* lakesoul_meta_partition_descs(void* hp, const char* table_id) {
+  auto* h = (MetaHandle*)hp;
+  try {
+    return strdup(rows_to_json_array(
+        h->db,
+        "SELECT DISTINCT partition_desc FROM partition_info WHERE table_id=?"
+        " ORDER BY partition_desc",
+        {table_id}).c_str());
+  } catch (std::exception& e) { g_meta_err = e.what(); return nullptr; }
+}
+
+extern "C" int64_t lakesoul_meta_latest_version(void* hp, const char* table_id,
+                                                const char* desc) {
+  auto* h = (MetaHandle*)hp;
+  try {
+    bool found = false;
+    std::string v = q1(h->db,
+        "SELECT version FROM partition_info WHERE table_id=? AND"
+        " partition_desc=? ORDER BY version DESC LIMIT 1",
+        {table_id, desc}, &found);
+    return found ? atoll(v.c_str()) : -1;
+  } catch (std::exception& e) { g_meta_err = e.what(); return -1; }
+}
+
+// files of a SPECIFIC partition version (time travel)
+extern "C" char* lakesoul_meta_files_for_version(void* hp,
+                                                 const char* table_id,
+                                                 const char* desc,
+                                                 int64_t version) {
+  auto* h = (MetaHandle*)hp;
+  try {
+    bool found = false;
+    std::string snap = q1(h->db,
+        "SELECT snapshot FROM partition_info WHERE table_id=? AND"
+        " partition_desc=? AND version=?",
+        {table_id, desc, std::to_string(version)}, &found);
+    if (!found) { g_meta_err = "version not found"; return nullptr; }
+    return strdup(resolve_snapshot_files(h->db, table_id, desc, snap).c_str());
+  } catch (std::exception& e) { g_meta_err = e.what(); return nullptr; }
+}
+
+// files ADDED in (from_version, to_version] minus later deletes — the
+// incremental-read query (reference metadata_client.rs:1052-1126)
+extern "C" char* lakesoul_meta_incremental_files(void* hp,
+                                                 const char* table_id,
+                                                 const char* desc,
+                                                 int64_t from_version,
+                                                 int64_t to_version) {
+  auto* h = (MetaHandle*)hp;
+  try {
+    bool f0 = false, f1 = false;
+    std::string snap_from =
+        from_version < 0 ? std::string("[]")
+        : q1(h->db,
+             "SELECT snapshot FROM partition_info WHERE table_id=? AND"
+             " partition_desc=? AND version=?",
+             {table_id, desc, std::to_string(from_version)}, &f0);
+    std::string snap_to = q1(h->db,
+        "SELECT snapshot FROM partition_info WHERE table_id=? AND"
+        " partition_desc=? AND version=?",
+        {table_id, desc, std::to_string(to_version)}, &f1);
+    if (!f1) { g_meta_err = "to_version not found"; return nullptr; }
+    auto from_ids = parse_str_list(snap_from);
+    std::string inc = "[";
+    bool first = true;
+    for (auto& cid : parse_str_list(snap_to)) {
+      bool seen = false;
+      for (auto& p : from_ids) if (p == cid) { seen = true; break; }
+      if (!seen) {
+        if (!first) inc += ",";
+        first = false;
+        inc += "\"" + jesc(cid) + "\"";
+      }
+    }
+    inc += "]";
+    return strdup(resolve_snapshot_files(h->db, table_id, desc, inc).c_str());
+  } catch (std::exception& e) { g_meta_err = e.what(); return nullptr; }
+}
+
+extern "C" char* lakesoul_meta_table_info_by_path(void* hp, const char* path) {
+  auto* h = (MetaHandle*)hp;
+  try {
+    bool f1 = false, f2 = false;
+    std::string name = q1(h->db,
+        "SELECT table_name FROM table_info WHERE table_path=?", {path}, &f1);
+    std::string ns = q1(h->db,
+        "SELECT table_namespace FROM table_info WHERE table_path=?", {path},
+        &f2);
+    if (!f1 || !f2) { g_meta_err = "path not found"; return nullptr; }
+    return lakesoul_meta_table_info(hp, name.c_str(), ns.c_str());
+  } catch (std::exception& e) { g_meta_err = e.what(); return nullptr; }
+}

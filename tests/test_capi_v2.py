@@ -255,3 +255,70 @@ def test_compiled_c_consumer(tmp_path):
                          text=True, timeout=120)
     assert out.returncode == 0, out.stdout + out.stderr
     assert "capi_smoke OK" in out.stdout
+
+
+def test_meta_c_abi_extended_daos(catalog):
+    """Extended DAO surface of the metadata C ABI: listing, versioned and
+    incremental snapshot queries, lookup by path."""
+    import json
+
+    if not os.path.exists(LIB):
+        pytest.skip("lib not built")
+    L = ctypes.CDLL(LIB)
+    vp, cp, i64 = ctypes.c_void_p, ctypes.c_char_p, ctypes.c_int64
+    L.lakesoul_meta_open.restype = vp
+    L.lakesoul_meta_open.argtypes = [cp]
+    for fn in ("lakesoul_meta_list_namespaces",):
+        getattr(L, fn).restype = vp
+        getattr(L, fn).argtypes = [vp]
+    for fn in ("lakesoul_meta_list_tables", "lakesoul_meta_partition_descs",
+               "lakesoul_meta_table_info_by_path"):
+        getattr(L, fn).restype = vp
+        getattr(L, fn).argtypes = [vp, cp]
+    L.lakesoul_meta_latest_version.restype = i64
+    L.lakesoul_meta_latest_version.argtypes = [vp, cp, cp]
+    L.lakesoul_meta_files_for_version.restype = vp
+    L.lakesoul_meta_files_for_version.argtypes = [vp, cp, cp, i64]
+    L.lakesoul_meta_incremental_files.restype = vp
+    L.lakesoul_meta_incremental_files.argtypes = [vp, cp, cp, i64, i64]
+    L.lakesoul_meta_free_string.argtypes = [vp]
+    L.lakesoul_meta_close.argtypes = [vp]
+    L.lakesoul_meta_last_error.restype = cp
+
+    from lakesoul_amd.io.schema import Field, Schema
+
+    t = catalog.create_table(
+        "cdao", Schema([Field("id", "int64", False), Field("v", "float64")]),
+        primary_keys=["id"], hash_bucket_num=1)
+    t.upsert({"id": np.arange(5, dtype=np.int64), "v": np.zeros(5)})
+    t.upsert({"id": np.arange(5, dtype=np.int64), "v": np.ones(5)})
+
+    def grab(p):
+        assert p, L.lakesoul_meta_last_error()
+        s = ctypes.cast(p, cp).value.decode()
+        L.lakesoul_meta_free_string(p)
+        return json.loads(s)
+
+    h = L.lakesoul_meta_open(t.client.store.path.encode())
+    assert h
+    try:
+        assert "default" in grab(L.lakesoul_meta_list_namespaces(h))
+        assert "cdao" in grab(L.lakesoul_meta_list_tables(h, b"default"))
+        descs = grab(L.lakesoul_meta_partition_descs(h, t.table_id.encode()))
+        assert descs == ["-5"]
+        latest = L.lakesoul_meta_latest_version(h, t.table_id.encode(), b"-5")
+        assert latest == 1
+        v0 = grab(L.lakesoul_meta_files_for_version(
+            h, t.table_id.encode(), b"-5", 0))
+        v1 = grab(L.lakesoul_meta_files_for_version(
+            h, t.table_id.encode(), b"-5", 1))
+        assert len(v0) == 1 and len(v1) == 2
+        inc = grab(L.lakesoul_meta_incremental_files(
+            h, t.table_id.encode(), b"-5", 0, 1))
+        assert len(inc) == 1
+        assert inc[0]["path"] == [f for f in v1 if f not in v0][0]["path"]
+        info = grab(L.lakesoul_meta_table_info_by_path(
+            h, t.table_path.encode()))
+        assert info["table_id"] == t.table_id
+    finally:
+        L.lakesoul_meta_close(h)
