@@ -116,7 +116,8 @@ static int64_t write_parquet(
     const std::vector<c10::optional<torch::Tensor>>& offsets,
     const std::vector<c10::optional<torch::Tensor>>& validity,
     const std::vector<bool>& nullable, int64_t row_group_size, int64_t codec,
-    int64_t level) {
+    int64_t level,
+    const std::vector<c10::optional<torch::Tensor>>& elem_offsets = {}) {
   size_t ncol = names.size();
   TORCH_CHECK(dtypes.size() == ncol && columns.size() == ncol);
   std::vector<ColumnDesc> descs(ncol);
@@ -146,15 +147,24 @@ static int64_t write_parquet(
     torch::Tensor col = columns[i].contiguous().cpu();
     keep.push_back(col);
     if (is_list) {
-      TORCH_CHECK(di.physical != PT_BYTE_ARRAY,
-                  "list<string> write not supported");
       TORCH_CHECK(offsets[i].has_value(), "list column needs offsets");
       torch::Tensor off = offsets[i]->contiguous().cpu().to(torch::kInt64);
       keep.push_back(off);
       data[i].list_offsets = off.data_ptr<int64_t>();
-      data[i].data = (const uint8_t*)col.data_ptr();
-      int es = physical_elem_size(di.physical);
-      TORCH_CHECK(col.element_size() == es, "list element size mismatch");
+      if (di.physical == PT_BYTE_ARRAY) {
+        // list<string>: col = payload bytes, elem_offsets = element
+        // byte offsets (int32)
+        TORCH_CHECK(i < elem_offsets.size() && elem_offsets[i].has_value(),
+                    "list<string> column needs elem_offsets");
+        torch::Tensor eo = elem_offsets[i]->contiguous().cpu().to(torch::kInt32);
+        keep.push_back(eo);
+        data[i].offsets = eo.data_ptr<int32_t>();
+        data[i].bytes = (const uint8_t*)col.data_ptr();
+      } else {
+        data[i].data = (const uint8_t*)col.data_ptr();
+        int es = physical_elem_size(di.physical);
+        TORCH_CHECK(col.element_size() == es, "list element size mismatch");
+      }
       int64_t n = off.numel() - 1;
       TORCH_CHECK(num_rows < 0 || n == num_rows, "row count mismatch");
       num_rows = n;
@@ -258,7 +268,8 @@ static int64_t writer_open(const std::string& path,
 
 static void writer_write(int64_t h, const std::vector<torch::Tensor>& columns,
                          const std::vector<c10::optional<torch::Tensor>>& offsets,
-                         const std::vector<c10::optional<torch::Tensor>>& validity) {
+                         const std::vector<c10::optional<torch::Tensor>>& validity,
+                         const std::vector<c10::optional<torch::Tensor>>& elem_offsets = {}) {
   auto ow = get_writer(h);
   size_t ncol = ow->descs.size();
   TORCH_CHECK(columns.size() == ncol);
@@ -274,7 +285,16 @@ static void writer_write(int64_t h, const std::vector<torch::Tensor>& columns,
       torch::Tensor off = offsets[i]->contiguous().cpu().to(torch::kInt64);
       keep.push_back(off);
       data[i].list_offsets = off.data_ptr<int64_t>();
-      data[i].data = (const uint8_t*)col.data_ptr();
+      if (d.physical == PT_BYTE_ARRAY) {
+        TORCH_CHECK(i < elem_offsets.size() && elem_offsets[i].has_value(),
+                    "list<string> column needs elem_offsets");
+        torch::Tensor eo = elem_offsets[i]->contiguous().cpu().to(torch::kInt32);
+        keep.push_back(eo);
+        data[i].offsets = eo.data_ptr<int32_t>();
+        data[i].bytes = (const uint8_t*)col.data_ptr();
+      } else {
+        data[i].data = (const uint8_t*)col.data_ptr();
+      }
       int64_t n = off.numel() - 1;
       TORCH_CHECK(num_rows < 0 || n == num_rows);
       num_rows = n;
@@ -822,11 +842,73 @@ static torch::Tensor bucket_ids_from_hashes(torch::Tensor hashes, int64_t nbucke
   return out;
 }
 
+// list<string> columns ride the MOR merge as opaque per-row byte blobs
+// whose payload is the PLAIN parquet representation ([u32 len][bytes]
+// per element). After merge this parses the stream back into row ->
+// element and element -> byte offsets (the chain walk is sequential, so
+// it lives here at memcpy speed instead of a python loop).
+static py::dict split_len_prefixed(const torch::Tensor& bytes,
+                                   const torch::Tensor& row_byte_offsets) {
+  torch::Tensor b = bytes.contiguous().cpu();
+  torch::Tensor ro = row_byte_offsets.contiguous().cpu().to(torch::kInt64);
+  const uint8_t* bp = b.data_ptr<uint8_t>();
+  const int64_t* rp = ro.data_ptr<int64_t>();
+  int64_t nrows = ro.numel() - 1;
+  int64_t total = b.numel();
+  // count elements first (u32 walk)
+  int64_t m = 0;
+  {
+    int64_t p = 0;
+    while (p + 4 <= total) {
+      uint32_t len;
+      std::memcpy(&len, bp + p, 4);
+      p += 4 + len;
+      m++;
+    }
+    TORCH_CHECK(p == total, "corrupt len-prefixed stream");
+  }
+  auto row_eoffs = torch::empty({nrows + 1}, torch::kInt64);
+  auto eoffs = torch::empty({m + 1}, torch::kInt64);
+  int64_t out_bytes = total - 4 * m;
+  auto payload = torch::empty({out_bytes}, torch::kUInt8);
+  int64_t* rep = row_eoffs.data_ptr<int64_t>();
+  int64_t* ep = eoffs.data_ptr<int64_t>();
+  uint8_t* pp = payload.data_ptr<uint8_t>();
+  int64_t p = 0, e = 0, w = 0, r = 0;
+  ep[0] = 0;
+  rep[0] = 0;
+  while (p + 4 <= total) {
+    while (r < nrows && rp[r] == p) rep[r++] = e;  // rows ending here
+    uint32_t len;
+    std::memcpy(&len, bp + p, 4);
+    p += 4;
+    std::memcpy(pp + w, bp + p, len);
+    p += len;
+    w += len;
+    ep[++e] = w;
+  }
+  while (r <= nrows) rep[r++] = e;  // trailing (incl. empty) rows
+  py::dict d;
+  d["row_offsets"] = row_eoffs;
+  d["elem_offsets"] = eoffs;
+  d["bytes"] = payload;
+  return d;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "lakesoul_amd host-native core";
-  m.def("write_parquet", &write_parquet, "write a parquet file");
+  m.def("write_parquet", &write_parquet, "write a parquet file",
+        py::arg("path"), py::arg("names"), py::arg("dtypes"),
+        py::arg("columns"), py::arg("offsets"), py::arg("validity"),
+        py::arg("nullable"), py::arg("row_group_size"), py::arg("codec"),
+        py::arg("level"),
+        py::arg("elem_offsets") = std::vector<c10::optional<torch::Tensor>>{});
   m.def("writer_open", &writer_open);
-  m.def("writer_write", &writer_write);
+  m.def("writer_write", &writer_write, py::arg("h"), py::arg("columns"),
+        py::arg("offsets"), py::arg("validity"),
+        py::arg("elem_offsets") = std::vector<c10::optional<torch::Tensor>>{});
+  m.def("split_len_prefixed", &split_len_prefixed,
+        "parse a PLAIN len-prefixed byte-array stream back into row/element offsets");
   m.def("writer_bytes", &writer_bytes);
   m.def("writer_finish", &writer_finish);
   m.def("writer_abort", &writer_abort);

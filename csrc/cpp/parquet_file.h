@@ -371,8 +371,6 @@ class ParquetWriter {
   // shape the reader's list path decodes (read side ~line 655).
   void write_list_chunk(const ColumnDesc& cd, const ColumnData& col,
                         int64_t row_off, int64_t n, RowGroup& rg) {
-    if (cd.physical == PT_BYTE_ARRAY)
-      throw std::runtime_error("list<string> write not supported");
     if (!col.list_offsets) throw std::runtime_error("list column without offsets");
     const int64_t* offs = col.list_offsets;
     std::vector<uint8_t> reps, defs;
@@ -406,8 +404,20 @@ class ParquetWriter {
     payload.insert(payload.end(), rep_rle.begin(), rep_rle.end());
     payload.insert(payload.end(), (uint8_t*)&dl, (uint8_t*)&dl + 4);
     payload.insert(payload.end(), def_rle.begin(), def_rle.end());
-    int es = physical_elem_size(cd.physical);
-    payload.insert(payload.end(), col.data + lo * es, col.data + hi * es);
+    if (cd.physical == PT_BYTE_ARRAY) {
+      // list<string>: elements as PLAIN len-prefixed byte arrays
+      // (col.offsets = element byte offsets, col.bytes = payload)
+      if (!col.offsets) throw std::runtime_error("list<string> without element offsets");
+      for (int64_t e = lo; e < hi; e++) {
+        uint32_t len = (uint32_t)(col.offsets[e + 1] - col.offsets[e]);
+        payload.insert(payload.end(), (uint8_t*)&len, (uint8_t*)&len + 4);
+        payload.insert(payload.end(), col.bytes + col.offsets[e],
+                       col.bytes + col.offsets[e + 1]);
+      }
+    } else {
+      int es = physical_elem_size(cd.physical);
+      payload.insert(payload.end(), col.data + lo * es, col.data + hi * es);
+    }
 
     std::vector<uint8_t> compressed;
     const uint8_t* body = payload.data();

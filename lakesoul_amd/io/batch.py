@@ -66,6 +66,9 @@ class Column:
     offsets: Optional[torch.Tensor] = None   # int32 [n+1], string/binary
     bytes_: Optional[torch.Tensor] = None    # uint8, string/binary
     validity: Optional[torch.Tensor] = None  # uint8 [n], 1=valid; None=all valid
+    # list<string> only: element -> byte offsets (int64 m+1); then
+    # `offsets` is row -> element and `bytes_` the element payload
+    elem_offsets: Optional[torch.Tensor] = None
 
     @property
     def is_string(self) -> bool:
@@ -76,6 +79,10 @@ class Column:
         # list<T>: `offsets` (int64 n+1, element ranges) + `data`
         # (element values tensor) + per-row validity
         return self.dtype.startswith("list<")
+
+    @property
+    def is_list_str(self) -> bool:
+        return self.dtype == "list<string>"
 
     @property
     def elem_dtype(self) -> str:
@@ -102,7 +109,8 @@ class Column:
         def mv(t):
             return None if t is None else t.to(device, non_blocking=nb)
 
-        return Column(self.dtype, mv(self.data), mv(self.offsets), mv(self.bytes_), mv(self.validity))
+        return Column(self.dtype, mv(self.data), mv(self.offsets), mv(self.bytes_),
+                      mv(self.validity), mv(self.elem_offsets))
 
     def take(self, idx: torch.Tensor) -> "Column":
         """Gather rows by index tensor (moved to this column's device —
@@ -111,6 +119,24 @@ class Column:
         dev = self.offsets.device if (self.is_string or self.is_list) else self.data.device
         if idx.device != dev:
             idx = idx.to(dev)
+        if self.is_list_str:
+            # two gathers: per-row payload bytes (string path over the
+            # row's contiguous byte range) + per-row element byte lens
+            # (numeric-list path over elem_offsets diffs), then cumsum
+            # rebuilds element offsets in the new row order
+            row_boffs = self.elem_offsets[self.offsets]
+            byte_col = Column("binary", offsets=row_boffs, bytes_=self.bytes_,
+                              validity=self.validity)
+            got = byte_col.take(idx)
+            elem_lens = self.elem_offsets[1:] - self.elem_offsets[:-1]
+            lens_col = Column("list<int64>", data=elem_lens, offsets=self.offsets)
+            lg = lens_col.take(idx)
+            new_eoffs = torch.zeros(lg.data.numel() + 1, dtype=torch.int64,
+                                    device=dev)
+            torch.cumsum(lg.data, 0, out=new_eoffs[1:].view(-1))
+            return Column(self.dtype, offsets=lg.offsets.to(torch.int64),
+                          bytes_=got.bytes_, validity=got.validity,
+                          elem_offsets=new_eoffs)
         if self.is_list:
             # gather element ranges through the byte-view of the values
             # (the string gather path, offsets scaled by element size)
@@ -179,7 +205,18 @@ class Batch:
         cols = {}
         for k, c in self.columns.items():
             v = None if c.validity is None else c.validity[a:b]
-            if c.is_list:
+            if c.is_list_str:
+                offs = c.offsets[a : b + 1]
+                base = offs[0]
+                lo, hi = int(base), int(c.offsets[b])
+                eoffs = c.elem_offsets[lo : hi + 1]
+                eb = int(eoffs[0]) if eoffs.numel() else 0
+                cols[k] = Column(
+                    c.dtype, offsets=(offs - base).to(torch.int64),
+                    bytes_=c.bytes_[eb:int(c.elem_offsets[hi])] if eoffs.numel()
+                    else c.bytes_[:0],
+                    validity=v, elem_offsets=(eoffs - eb).to(torch.int64))
+            elif c.is_list:
                 offs = c.offsets[a : b + 1]
                 base = offs[0]
                 lo, hi = int(base), int(c.offsets[b])
@@ -248,6 +285,36 @@ class Batch:
             if f.name not in d:
                 raise KeyError(f"missing column {f.name}")
             v = d[f.name]
+            if f.dtype == "list<string>":
+                if isinstance(v, tuple) and len(v) in (3, 4):
+                    offs = torch.as_tensor(v[0], dtype=torch.int64)
+                    eoffs = torch.as_tensor(v[1], dtype=torch.int64)
+                    bys = torch.as_tensor(v[2], dtype=torch.uint8)
+                    val = (torch.as_tensor(v[3], dtype=torch.uint8)
+                           if len(v) == 4 and v[3] is not None else None)
+                else:
+                    items = list(v)
+                    val = None
+                    if any(x is None for x in items):
+                        val = torch.tensor(
+                            [0 if x is None else 1 for x in items],
+                            dtype=torch.uint8)
+                    offs = torch.zeros(len(items) + 1, dtype=torch.int64)
+                    torch.cumsum(torch.tensor(
+                        [0 if x is None else len(x) for x in items],
+                        dtype=torch.int64), 0, out=offs[1:].view(-1))
+                    enc = [s.encode() if isinstance(s, str) else bytes(s)
+                           for x in items if x is not None for s in x]
+                    eoffs = torch.zeros(len(enc) + 1, dtype=torch.int64)
+                    torch.cumsum(torch.tensor([len(e) for e in enc],
+                                              dtype=torch.int64), 0,
+                                 out=eoffs[1:].view(-1))
+                    bys = torch.from_numpy(np.frombuffer(
+                        b"".join(enc), dtype=np.uint8).copy()) if enc else \
+                        torch.empty(0, dtype=torch.uint8)
+                cols[f.name] = Column(f.dtype, offsets=offs, bytes_=bys,
+                                      validity=val, elem_offsets=eoffs)
+                continue
             if f.dtype.startswith("list<"):
                 elem_dt = f.dtype[5:-1]
                 npdt = np_dtype_for(elem_dt)
@@ -382,6 +449,27 @@ class Batch:
         arrays = []
         for f in self.schema:
             c = self.columns[f.name]
+            if f.dtype == "list<string>":
+                offs_np = c.offsets.cpu().numpy().astype(np.int32, copy=False)
+                eoffs_np = c.elem_offsets.cpu().numpy().astype(np.int32, copy=False)
+                m = len(eoffs_np) - 1
+                vals = pa.Array.from_buffers(
+                    pa.string(), m,
+                    [None, pa.py_buffer(eoffs_np.tobytes()),
+                     pa.py_buffer(c.bytes_.cpu().numpy().tobytes())])
+                n = len(c)
+                validity_buf = None
+                null_count = 0
+                if c.validity is not None:
+                    vv = c.validity.cpu().numpy().astype(bool)
+                    null_count = int(n - vv.sum())
+                    validity_buf = pa.py_buffer(
+                        np.packbits(vv, bitorder="little").tobytes())
+                arrays.append(pa.Array.from_buffers(
+                    dtype_to_arrow(f.dtype), n,
+                    [validity_buf, pa.py_buffer(offs_np.tobytes())],
+                    null_count=null_count, children=[vals]))
+                continue
             if f.dtype.startswith("list<"):
                 elem_dt = f.dtype[5:-1]
                 offs_np = c.offsets.cpu().numpy().astype(np.int32, copy=False)
@@ -460,6 +548,30 @@ def concat_batches(batches: List[Batch]) -> Batch:
     for f in schema:
         cs = [b.columns[f.name] for b in batches]
         n_total = sum(len(c) for c in cs)
+        if cs and cs[0].is_list_str:
+            bytes_ = torch.cat([c.bytes_ for c in cs])
+            m_total = sum(int(c.offsets[-1]) for c in cs)
+            offs = torch.zeros(n_total + 1, dtype=torch.int64, device=bytes_.device)
+            eoffs = torch.zeros(m_total + 1, dtype=torch.int64, device=bytes_.device)
+            pos, ebase, epos, bbase = 0, 0, 0, 0
+            for c in cs:
+                n = len(c)
+                m = int(c.offsets[-1])
+                offs[pos + 1: pos + n + 1] = c.offsets[1:].to(torch.int64) + ebase
+                eoffs[epos + 1: epos + m + 1] = c.elem_offsets[1:].to(torch.int64) + bbase
+                ebase += m
+                bbase += int(c.elem_offsets[-1])
+                pos += n
+                epos += m
+            validity = None
+            if any(c.validity is not None for c in cs):
+                validity = torch.cat([
+                    c.validity if c.validity is not None
+                    else torch.ones(len(c), dtype=torch.uint8, device=bytes_.device)
+                    for c in cs])
+            cols[f.name] = Column(f.dtype, offsets=offs, bytes_=bytes_,
+                                  validity=validity, elem_offsets=eoffs)
+            continue
         if cs and cs[0].is_list:
             data = torch.cat([c.data for c in cs])
             offs = torch.zeros(n_total + 1, dtype=torch.int64, device=data.device)

@@ -86,9 +86,15 @@ class StreamingParquetUpload:
         the uploader (overlaps with the caller's next chunk prep)."""
         self._check()
         cols, offs, vals = [], [], []
+        eoffs = []
         for f in self.schema:
             c = batch.columns[f.name]
-            if c.is_list:
+            eoffs.append(None)
+            if c.is_list_str:
+                cols.append(c.bytes_.cpu())
+                offs.append(c.offsets.cpu().to(torch.int64))
+                eoffs[-1] = c.elem_offsets.cpu().to(torch.int32)
+            elif c.is_list:
                 t = c.data.cpu()
                 if f.dtype[5:-1] in ("int8", "int16"):
                     t = t.to(torch.int32)
@@ -106,7 +112,7 @@ class StreamingParquetUpload:
                 cols.append(t)
                 offs.append(None)
             vals.append(None if c.validity is None else c.validity.cpu())
-        cpp().writer_write(self._h, cols, offs, vals)
+        cpp().writer_write(self._h, cols, offs, vals, eoffs)
         self._rows += batch.num_rows
         self._q.put(cpp().writer_bytes(self._h))
 

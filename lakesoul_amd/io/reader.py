@@ -508,8 +508,12 @@ class LakeSoulScan:
 
                 warnings.warn(f"chunked merge unavailable ({e}); "
                               "falling back")
+        has_list_str = any(
+            self.schema.field(n).dtype == "list<string>"
+            for n in self.read_cols if n not in self.range_cols)
         if self.device == "cuda" and (
             oversized
+            or has_list_str
             or (needs_merge and not self._gpu_merge_supported())
         ):
             # hybrid: CPU decode+merge (host RAM), then ship the merged
@@ -667,6 +671,18 @@ class LakeSoulScan:
                 cols[f.name] = self._range_value_column(f, unit, merged)
                 continue
             npc = merged[f.name]
+            if f.dtype == "list<string>":
+                d = cpp().split_len_prefixed(
+                    torch.from_numpy(np.ascontiguousarray(npc.bytes_)),
+                    torch.from_numpy(np.ascontiguousarray(npc.offsets,
+                                                          dtype=np.int64)))
+                cols[f.name] = Column(
+                    f.dtype, offsets=d["row_offsets"],
+                    bytes_=d["bytes"], elem_offsets=d["elem_offsets"],
+                    validity=None if npc.validity is None
+                    else torch.from_numpy(npc.validity),
+                )
+                continue
             if f.dtype.startswith("list<"):
                 es = np.dtype(_np_phys(f.dtype[5:-1])).itemsize
                 offs = np.ascontiguousarray(npc.offsets, dtype=np.int64) // es
@@ -740,6 +756,55 @@ class LakeSoulScan:
             ci = 0
             for name in req:
                 f = self.schema.field(name)
+                if f.dtype == "list<string>":
+                    # rides the byte-string machinery as the PLAIN
+                    # parquet stream ([u32 len][bytes] per element) so the
+                    # opaque whole-row gather keeps element boundaries;
+                    # split_len_prefixed() parses it back at the end
+                    offs_parts, bytes_parts, masks = [], [], []
+                    any_null = False
+                    for rg in rg_iter:
+                        d = chunks[ci]
+                        ci += 1
+                        eoffs = d["offsets"].numpy().astype(np.int64)
+                        by = d["bytes"].numpy()
+                        lo_ = d["list_offsets"].numpy()
+                        lens = np.diff(eoffs)
+                        m = len(lens)
+                        new_eoffs = np.zeros(m + 1, np.int64)
+                        np.cumsum(lens + 4, out=new_eoffs[1:])
+                        out_b = np.empty(int(new_eoffs[-1]), np.uint8)
+                        if m:
+                            pref = lens.astype("<u4").view(np.uint8).reshape(m, 4)
+                            out_b[(new_eoffs[:-1, None] +
+                                   np.arange(4)).reshape(-1)] = pref.reshape(-1)
+                            nb_ = int(eoffs[-1])
+                            if nb_:
+                                dst = (np.repeat(new_eoffs[:-1] + 4, lens) +
+                                       np.arange(nb_) - np.repeat(eoffs[:-1], lens))
+                                out_b[dst] = by[:nb_]
+                        offs_parts.append(new_eoffs[lo_])
+                        bytes_parts.append(out_b)
+                        lv = d["list_validity"].numpy()
+                        nrow = len(offs_parts[-1]) - 1
+                        if len(lv) and not lv.all():
+                            any_null = True
+                            masks.append(lv[:nrow])
+                        else:
+                            masks.append(np.ones(nrow, dtype=np.uint8))
+                    validity = np.concatenate(masks) if any_null else None
+                    total_rows = sum(len(o) - 1 for o in offs_parts)
+                    offs = np.zeros(total_rows + 1, dtype=np.int64)
+                    pos, base = 0, 0
+                    for o in offs_parts:
+                        nr = len(o) - 1
+                        offs[pos + 1: pos + nr + 1] = o[1:] + base
+                        base += int(o[-1]) if len(o) else 0
+                        pos += nr
+                    bys = (np.concatenate(bytes_parts) if bytes_parts
+                           else np.empty(0, np.uint8))
+                    out[name] = NpColumn(f.dtype, None, offs, bys, validity)
+                    continue
                 if f.dtype.startswith("list<"):
                     # list<T> rides the byte-string machinery downstream:
                     # offsets in BYTES over the raw element buffer (merge/

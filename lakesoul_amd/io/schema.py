@@ -111,10 +111,10 @@ def canonical_dtype(dt: str) -> str:
     elem = list_element_dtype(dt)
     if elem is not None:
         inner = canonical_dtype(elem)
-        if inner not in FIXED_WIDTH_BYTES:
+        if inner not in FIXED_WIDTH_BYTES and inner != "string":
             raise TypeError(
                 f"unsupported list element dtype {elem!r} "
-                "(fixed-width primitives only)")
+                "(fixed-width primitives or string)")
         return f"list<{inner}>"
     try:
         return _CANONICAL[dt.lower()]

@@ -234,11 +234,17 @@ def _write_batch_to_file(path: str, batch: Batch, compression: str, level: int,
 def _write_batch_to_file_local(path: str, batch: Batch, compression: str, level: int,
                                row_group_size: int) -> int:
     names, dtypes, columns, offsets, validity, nullable = [], [], [], [], [], []
+    elem_offs = []
     for f in batch.schema:
         c = batch.columns[f.name]
         names.append(f.name)
         dtypes.append(f.dtype)
-        if c.is_list:
+        elem_offs.append(None)
+        if c.is_list_str:
+            columns.append(c.bytes_.cpu())
+            offsets.append(c.offsets.cpu().to(torch.int64))
+            elem_offs[-1] = c.elem_offsets.cpu().to(torch.int32)
+        elif c.is_list:
             t = c.data.cpu()
             ed = c.elem_dtype
             if ed in ("int8", "int16"):
@@ -262,7 +268,7 @@ def _write_batch_to_file_local(path: str, batch: Batch, compression: str, level:
         nullable.append(f.nullable)
     return cpp().write_parquet(
         path, names, dtypes, columns, offsets, validity, nullable,
-        row_group_size, _CODEC_ID.get(compression, 6), level,
+        row_group_size, _CODEC_ID.get(compression, 6), level, elem_offs,
     )
 
 

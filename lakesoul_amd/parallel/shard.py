@@ -37,6 +37,52 @@ def shard_scan(scan):
     return scan
 
 
+def _pack_list_str(c: Column):
+    """Flatten a list<string> column into (row byte offsets, prefixed
+    payload): per element [u32 len][bytes], rows contiguous. Pure tensor
+    ops, so it runs on whichever device the column lives on."""
+    dev = c.offsets.device
+    eoffs = c.elem_offsets.to(torch.int64)
+    lens = (eoffs[1:] - eoffs[:-1]).contiguous()
+    m = lens.numel()
+    new_eoffs = torch.zeros(m + 1, dtype=torch.int64, device=dev)
+    if m:
+        torch.cumsum(lens + 4, 0, out=new_eoffs[1:].view(-1))
+    out = torch.zeros(int(new_eoffs[-1]) if m else 0, dtype=torch.uint8,
+                      device=dev)
+    if m:
+        pref = lens.to(torch.int32).contiguous().view(torch.uint8).view(m, 4)
+        idx = (new_eoffs[:-1].unsqueeze(1) +
+               torch.arange(4, device=dev)).reshape(-1)
+        out[idx] = pref.reshape(-1)
+        nb = int(eoffs[-1])
+        if nb:
+            dst = (torch.repeat_interleave(new_eoffs[:-1] + 4, lens) +
+                   torch.arange(nb, device=dev) -
+                   torch.repeat_interleave(eoffs[:-1], lens))
+            out[dst] = c.bytes_
+    row_boffs = new_eoffs[c.offsets.to(torch.int64)]
+    return row_boffs, out
+
+
+def _unpack_list_str(dtype: str, offsets: torch.Tensor, bytes_: torch.Tensor,
+                     validity) -> Column:
+    """Inverse of _pack_list_str after the exchange: parse the prefixed
+    stream back into row/element offsets (C++ walk; a GPU-resident batch
+    takes one D2H/H2D hop — exchange of list<string> is not a hot path)."""
+    from ..ops import cpp
+
+    dev = offsets.device
+    d = cpp().split_len_prefixed(bytes_.cpu(), offsets.cpu())
+    return Column(
+        dtype,
+        offsets=d["row_offsets"].to(dev),
+        bytes_=d["bytes"].to(dev),
+        elem_offsets=d["elem_offsets"].to(dev),
+        validity=validity,
+    )
+
+
 def _col_row_bytes(f, c: Column) -> int:
     """Bytes per row this field contributes to the packed buffer."""
     if c.is_string or c.is_list:
@@ -82,7 +128,19 @@ class AsyncExchange:
         self.recv_rows = [int(x) for x in recv_counts.cpu()]
         self.n_recv = sum(self.recv_rows)
 
-        reordered = batch.take(order) if n else batch
+        reordered = (batch.take(order) if n else
+                     Batch(batch.schema, dict(batch.columns)))
+        # list<string> travels the wire as one opaque blob per row: the
+        # PLAIN parquet stream ([u32 len][bytes] per element) keeps the
+        # element boundaries inside the single byte payload, so it rides
+        # the existing string round unchanged and is parsed back on
+        # arrival (wait())
+        for f in batch.schema:
+            c = reordered.columns[f.name]
+            if c.is_list_str:
+                ro, ob = _pack_list_str(c)
+                reordered.columns[f.name] = Column(
+                    "binary", offsets=ro, bytes_=ob, validity=c.validity)
         self.reordered = reordered
         dev = None
         for f in batch.schema:
@@ -222,8 +280,12 @@ class AsyncExchange:
             if n_recv:
                 torch.cumsum(recv_lens, 0, out=new_offs[1:].view(-1))
             if c.is_string:
-                out_cols[f.name] = Column(f.dtype, offsets=new_offs,
-                                          bytes_=recv_bytes, validity=v)
+                if f.dtype == "list<string>":
+                    out_cols[f.name] = _unpack_list_str(
+                        f.dtype, new_offs, recv_bytes, v)
+                else:
+                    out_cols[f.name] = Column(f.dtype, offsets=new_offs,
+                                              bytes_=recv_bytes, validity=v)
             else:
                 vals = (recv_bytes.view(c.data.dtype) if recv_bytes.numel()
                         else torch.empty(0, dtype=c.data.dtype,

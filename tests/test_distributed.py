@@ -239,6 +239,33 @@ def _run_exchange_hard(rank, world, rdv_file, tmpdir, results):
                 got_list = vals3[offs3[i]:offs3[i + 1]].tolist()
                 expect = [float(g), float(idx_src % 3)] * (idx_src % 2 + 1)
                 assert got_list == expect, (g, got_list, expect)
+
+        # case 4: list<string> column (opaque prefixed-blob round)
+        schema4 = Schema([Field("id", "int64", False),
+                          Field("tags", "list<string>", True)])
+        n4 = 40
+        ids4 = np.arange(rank * 100, rank * 100 + n4, dtype=np.int64)
+        b4 = Batch.from_dict({
+            "id": ids4,
+            "tags": [None if i % 6 == 0 else
+                     [f"t{ids4[i]}", ""] [: i % 3] for i in range(n4)],
+        }, schema4)
+        dest4 = torch.from_numpy((ids4 % world).astype(np.int64))
+        out4 = exchange_batch_all_to_all(b4, dest4)
+        gid4 = out4.columns["id"].data.numpy()
+        assert (gid4 % world == rank).all()
+        tg = out4.columns["tags"]
+        ro = tg.offsets.numpy()
+        eo = tg.elem_offsets.numpy()
+        bb = tg.bytes_.numpy().tobytes()
+        for i, g in enumerate(gid4):
+            idx_src = int(g % 100)
+            if idx_src % 6 == 0:
+                assert tg.validity is not None and tg.validity[i] == 0
+            else:
+                elems = [bb[eo[e]:eo[e + 1]].decode()
+                         for e in range(ro[i], ro[i + 1])]
+                assert elems == [f"t{g}", ""][: idx_src % 3], (g, elems)
         results[rank] = "ok"
     finally:
         dist.destroy_process_group()

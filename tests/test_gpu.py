@@ -642,3 +642,34 @@ def test_gpu_list_column_mor(dev, tmp_path):
             np.testing.assert_allclose(list(b), list(a), err_msg=str(i))
     np.testing.assert_allclose(gpu["v"].to_numpy(), cpu["v"].to_numpy())
     np.testing.assert_allclose(list(gpu["emb"].iloc[3]), [9.0])
+
+
+def test_gpu_list_string_scan(dev, tmp_path):
+    """list<string> scans on device='cuda' route through the host
+    decode+merge gate and ship the parsed column to HBM; values match
+    the CPU scan and the tensors land on the GPU."""
+    catalog = _mk_catalog(tmp_path)
+    from lakesoul_amd.io.schema import Field, Schema
+
+    t = catalog.create_table(
+        "gls",
+        Schema([Field("id", "int64", False), Field("tags", "list<string>"),
+                Field("v", "float64")]),
+        primary_keys=["id"], hash_bucket_num=2,
+    )
+    n = 2000
+    base = [None if i % 11 == 0 else
+            [f"t{i}", ""] [: i % 3] for i in range(n)]
+    t.upsert({"id": np.arange(n, dtype=np.int64), "tags": base,
+              "v": np.arange(n, dtype=np.float64)})
+    t.upsert({"id": np.array([3, 44], dtype=np.int64),
+              "tags": [["zz"], None], "v": np.array([3.5, 44.5])})
+    scan = t.scan(device="cuda")
+    for b in scan.iter_batches():
+        c = b.columns["tags"]
+        assert c.offsets.device.type == "cuda"
+        assert c.elem_offsets.device.type == "cuda"
+    cpu = t.scan(device="cpu").to_arrow().sort_by("id")
+    gpu = t.scan(device="cuda").to_arrow().sort_by("id")
+    assert gpu.column("tags").to_pylist() == cpu.column("tags").to_pylist()
+    assert gpu.column("tags").to_pylist()[3] == ["zz"]

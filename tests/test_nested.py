@@ -17,8 +17,10 @@ def test_list_dtype_canonical():
     assert not f.is_fixed_width
     f2 = Field("xs", "array<long>")
     assert f2.dtype == "list<int64>"
+    f3 = Field("tags", "list<string>")
+    assert f3.dtype == "list<string>"
     with pytest.raises(TypeError):
-        Field("bad", "list<string>")
+        Field("bad", "list<list<int32>>")
 
 
 def test_schema_json_roundtrip_with_list():
@@ -134,3 +136,103 @@ def test_list_schema_evolution_missing_column(catalog):
     e = df["emb"].tolist()
     np.testing.assert_allclose(e[3], [1.0, 2.0])
     assert e[0] is None or (isinstance(e[0], float) and np.isnan(e[0]))
+
+
+# ------------------------------------------------------------------ #
+# list<string>
+# ------------------------------------------------------------------ #
+
+LS_LISTS = [["a", "bb"], [], None, ["ccc"], ["d", "", "ee"]]
+
+
+def _ls_schema():
+    return Schema([Field("id", "int64", False), Field("tags", "list<string>")])
+
+
+def test_batch_list_string_ops():
+    s = _ls_schema()
+    b = Batch.from_dict({"id": np.arange(5, dtype=np.int64),
+                         "tags": LS_LISTS}, s)
+    c = b.columns["tags"]
+    assert c.is_list and c.is_list_str and len(c) == 5
+    assert c.offsets.tolist() == [0, 2, 2, 2, 3, 6]
+    assert c.validity.tolist() == [1, 1, 0, 1, 1]
+    # take
+    t = b.take(torch.tensor([4, 2, 0]))
+    assert t.to_arrow().column("tags").to_pylist() == \
+        [["d", "", "ee"], None, ["a", "bb"]]
+    # slice
+    assert b.slice(1, 4).to_arrow().column("tags").to_pylist() == \
+        [[], None, ["ccc"]]
+    # concat
+    from lakesoul_amd.io.batch import concat_batches
+
+    cc = concat_batches([b, t])
+    assert cc.to_arrow().column("tags").to_pylist() == \
+        b.to_arrow().column("tags").to_pylist() + \
+        t.to_arrow().column("tags").to_pylist()
+    # arrow round trip
+    b2 = Batch.from_arrow(b.to_arrow(), s)
+    assert b2.to_arrow().equals(b.to_arrow())
+
+
+def test_split_len_prefixed_roundtrip():
+    """The C++ parser inverts the PLAIN prefixed stream the CPU scan
+    builds for opaque list<string> merging."""
+    from lakesoul_amd.ops import cpp
+
+    # rows: ["ab","c"] | [] | ["defg"]
+    stream = (b"\x02\x00\x00\x00ab" + b"\x01\x00\x00\x00c" +
+              b"" + b"\x04\x00\x00\x00defg")
+    row_offs = torch.tensor([0, 11, 11, 19], dtype=torch.int64)
+    bys = torch.frombuffer(bytearray(stream), dtype=torch.uint8)
+    d = cpp().split_len_prefixed(bys, row_offs)
+    assert d["row_offsets"].tolist() == [0, 2, 2, 3]
+    assert d["elem_offsets"].tolist() == [0, 2, 3, 7]
+    assert bytes(d["bytes"].numpy().tobytes()) == b"abcdefg"
+
+
+def test_list_string_table_mor(catalog):
+    t = catalog.create_table("lstr", _ls_schema(), primary_keys=["id"],
+                             hash_bucket_num=2)
+    t.upsert({"id": np.arange(5, dtype=np.int64), "tags": LS_LISTS})
+    t.upsert({"id": np.array([1, 3], dtype=np.int64),
+              "tags": [["x"], ["y", "zz"]]})
+    exp = [["a", "bb"], ["x"], None, ["y", "zz"], ["d", "", "ee"]]
+    got = t.scan().to_arrow().sort_by("id").column("tags").to_pylist()
+    assert got == exp
+    # compaction rewrites and re-reads the 3-level BYTE_ARRAY pages
+    t.compaction()
+    got2 = t.scan().to_arrow().sort_by("id").column("tags").to_pylist()
+    assert got2 == exp
+
+
+def test_list_string_pyarrow_cross_read(catalog):
+    """Files we write are standard 3-level LIST of UTF8 — pyarrow reads
+    them without us in the loop."""
+    import glob
+
+    import pyarrow.parquet as pq
+
+    t = catalog.create_table("lstr_pa", _ls_schema(), primary_keys=["id"],
+                             hash_bucket_num=1)
+    t.upsert({"id": np.arange(5, dtype=np.int64), "tags": LS_LISTS})
+    files = sorted(glob.glob(t.table_path + "/**/*.parquet", recursive=True))
+    pt = pq.read_table(files[0]).sort_by("id")
+    assert str(pt.schema.field("tags").type).startswith("list")
+    assert pt.column("tags").to_pylist() == LS_LISTS
+
+
+def test_list_string_schema_evolution(catalog):
+    t = catalog.create_table(
+        "lstr_ev",
+        Schema([Field("id", "int64", False), Field("v", "float64")]),
+        primary_keys=["id"], hash_bucket_num=1)
+    t.upsert({"id": np.arange(4, dtype=np.int64), "v": np.zeros(4)})
+    t.add_columns([Field("tags", "list<string>")])
+    t2 = catalog.table("lstr_ev")
+    t2.upsert({"id": np.array([2], dtype=np.int64), "v": np.array([1.0]),
+               "tags": [["new"]]})
+    got = t2.scan().to_arrow().sort_by("id").column("tags").to_pylist()
+    assert got[2] == ["new"]
+    assert got[0] is None and got[3] is None
