@@ -289,9 +289,16 @@ class LakeSoulScan:
 
     # ------------------------------------------------------------------ #
 
+    def _has_list_str(self) -> bool:
+        """list<string> columns decode on the host (prefixed-stream MOR)
+        and ship to HBM — they bypass the GPU unit path."""
+        return any(self.schema.field(n).dtype == "list<string>"
+                   for n in self.read_cols if n not in self.range_cols)
+
     def iter_batches(self) -> Iterator[Batch]:
         units = self.plan()
         if (self.device == "cuda" and len(units) > 1
+                and not self._has_list_str()
                 and self._gpu_merge_supported() and not self.use_cache
                 and all(self._unit_fits(u) for u in units)
                 and os.environ.get("LAKESOUL_SCAN_PIPELINE", "1") != "0"):
@@ -508,12 +515,9 @@ class LakeSoulScan:
 
                 warnings.warn(f"chunked merge unavailable ({e}); "
                               "falling back")
-        has_list_str = any(
-            self.schema.field(n).dtype == "list<string>"
-            for n in self.read_cols if n not in self.range_cols)
         if self.device == "cuda" and (
             oversized
-            or has_list_str
+            or self._has_list_str()
             or (needs_merge and not self._gpu_merge_supported())
         ):
             # hybrid: CPU decode+merge (host RAM), then ship the merged
