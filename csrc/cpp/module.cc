@@ -117,7 +117,8 @@ static int64_t write_parquet(
     const std::vector<c10::optional<torch::Tensor>>& validity,
     const std::vector<bool>& nullable, int64_t row_group_size, int64_t codec,
     int64_t level,
-    const std::vector<c10::optional<torch::Tensor>>& elem_offsets = {}) {
+    const std::vector<c10::optional<torch::Tensor>>& elem_offsets = {},
+    const std::vector<std::string>& parents = {}) {
   size_t ncol = names.size();
   TORCH_CHECK(dtypes.size() == ncol && columns.size() == ncol);
   std::vector<ColumnDesc> descs(ncol);
@@ -142,6 +143,14 @@ static int64_t write_parquet(
     descs[i].is_list = is_list;
     if (is_list) {
       descs[i].max_def = 2;  // optional list group + repeated, required elem
+    }
+    if (i < parents.size() && !parents[i].empty()) {
+      // "struct:NAME" / "map:NAME" — the enclosing group
+      const std::string& p = parents[i];
+      auto colon = p.find(':');
+      TORCH_CHECK(colon != std::string::npos, "bad parent spec ", p);
+      descs[i].parent = p.substr(colon + 1);
+      descs[i].parent_kind = p.rfind("map", 0) == 0 ? 2 : 1;
     }
 
     torch::Tensor col = columns[i].contiguous().cpu();
@@ -236,7 +245,8 @@ static int64_t writer_open(const std::string& path,
                            const std::vector<std::string>& dtypes,
                            const std::vector<bool>& nullable,
                            int64_t row_group_size, int64_t codec,
-                           int64_t level) {
+                           int64_t level,
+                           const std::vector<std::string>& parents = {}) {
   auto ow = std::make_shared<OpenWriter>();
   ow->path = path;
   ow->dtypes = dtypes;
@@ -257,6 +267,13 @@ static int64_t writer_open(const std::string& path,
     d.nullable = nullable[i];
     d.is_list = is_list;
     if (is_list) d.max_def = 2;
+    if (i < parents.size() && !parents[i].empty()) {
+      const std::string& p = parents[i];
+      auto colon = p.find(':');
+      TORCH_CHECK(colon != std::string::npos, "bad parent spec ", p);
+      d.parent = p.substr(colon + 1);
+      d.parent_kind = p.rfind("map", 0) == 0 ? 2 : 1;
+    }
   }
   ow->w = std::make_unique<ParquetWriter>(path, ow->descs, (int)codec,
                                           (int)level, row_group_size);
@@ -902,8 +919,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("columns"), py::arg("offsets"), py::arg("validity"),
         py::arg("nullable"), py::arg("row_group_size"), py::arg("codec"),
         py::arg("level"),
-        py::arg("elem_offsets") = std::vector<c10::optional<torch::Tensor>>{});
-  m.def("writer_open", &writer_open);
+        py::arg("elem_offsets") = std::vector<c10::optional<torch::Tensor>>{},
+        py::arg("parents") = std::vector<std::string>{});
+  m.def("writer_open", &writer_open, py::arg("path"), py::arg("names"),
+        py::arg("dtypes"), py::arg("nullable"), py::arg("row_group_size"),
+        py::arg("codec"), py::arg("level"),
+        py::arg("parents") = std::vector<std::string>{});
   m.def("writer_write", &writer_write, py::arg("h"), py::arg("columns"),
         py::arg("offsets"), py::arg("validity"),
         py::arg("elem_offsets") = std::vector<c10::optional<torch::Tensor>>{});

@@ -84,6 +84,12 @@ struct ColumnDesc {
   bool is_list = false;
   bool elem_nullable = false;
   int32_t max_def = 0;      // 0 for flat columns (unused)
+  // struct/map member leaves (writer): the enclosing group's name and
+  // kind. Struct members write like flat nullable columns (optional
+  // group + required leaf = max_def 1); map key/value leaves write like
+  // list elements under group{MAP}/key_value.
+  std::string parent;       // empty = top-level column
+  int parent_kind = 0;      // 0 none, 1 struct, 2 map
 };
 
 inline int physical_elem_size(int32_t pt) {
@@ -164,9 +170,74 @@ class ParquetWriter {
     root.name = "schema";
     root.type = -1;
     root.repetition = REP_REQUIRED;
-    root.num_children = (int32_t)cols_.size();
+    // top-level children: each group (struct/map) counts once
+    int32_t top = 0;
+    for (size_t i = 0; i < cols_.size(); i++)
+      if (cols_[i].parent.empty() || i == 0 ||
+          cols_[i - 1].parent != cols_[i].parent)
+        top++;
+    root.num_children = top;
     fm.schema.push_back(root);
-    for (auto& c : cols_) {
+    for (size_t ci = 0; ci < cols_.size(); ci++) {
+      auto& c = cols_[ci];
+      if (c.parent_kind == 1) {
+        // struct: optional group { required members... } — consecutive
+        // descs sharing the parent are the members
+        if (ci == 0 || cols_[ci - 1].parent != c.parent) {
+          int32_t nmem = 0;
+          for (size_t j = ci; j < cols_.size() &&
+                              cols_[j].parent == c.parent; j++)
+            nmem++;
+          SchemaElement g;
+          g.name = c.parent;
+          g.type = -1;
+          g.repetition = c.nullable ? REP_OPTIONAL : REP_REQUIRED;
+          g.num_children = nmem;
+          fm.schema.push_back(g);
+        }
+        SchemaElement leaf;
+        leaf.name = c.name;
+        leaf.type = c.physical;
+        leaf.repetition = REP_REQUIRED;
+        leaf.converted = c.converted;
+        leaf.logical = c.logical;
+        leaf.int_bit_width = c.int_bit_width;
+        leaf.int_signed = c.int_signed;
+        leaf.dec_precision = c.dec_precision;
+        leaf.dec_scale = c.dec_scale;
+        fm.schema.push_back(leaf);
+        continue;
+      }
+      if (c.parent_kind == 2) {
+        // map: optional group (MAP) { repeated key_value { key; value } }
+        if (ci == 0 || cols_[ci - 1].parent != c.parent) {
+          SchemaElement g;
+          g.name = c.parent;
+          g.type = -1;
+          g.repetition = c.nullable ? REP_OPTIONAL : REP_REQUIRED;
+          g.converted = 1;  // MAP
+          g.num_children = 1;
+          fm.schema.push_back(g);
+          SchemaElement kv;
+          kv.name = "key_value";
+          kv.type = -1;
+          kv.repetition = REP_REPEATED;
+          kv.num_children = 2;
+          fm.schema.push_back(kv);
+        }
+        SchemaElement leaf;
+        leaf.name = c.name;
+        leaf.type = c.physical;
+        leaf.repetition = REP_REQUIRED;
+        leaf.converted = c.converted;
+        leaf.logical = c.logical;
+        leaf.int_bit_width = c.int_bit_width;
+        leaf.int_signed = c.int_signed;
+        leaf.dec_precision = c.dec_precision;
+        leaf.dec_scale = c.dec_scale;
+        fm.schema.push_back(leaf);
+        continue;
+      }
       if (c.is_list) {
         // standard 3-level LIST: optional group (LIST) { repeated group
         // list { required <element>; } } — mirrors the reader's
@@ -264,7 +335,9 @@ class ParquetWriter {
       ColumnMeta cm;
       cm.type = cd.physical;
       cm.encodings = {ENC_PLAIN, ENC_RLE};
-      cm.path_in_schema = {cd.name};
+      cm.path_in_schema = cd.parent.empty()
+          ? std::vector<std::string>{cd.name}
+          : std::vector<std::string>{cd.parent, cd.name};
       cm.codec = codec_;
       cm.num_values = n;
       cm.data_page_offset = pos_;
@@ -433,7 +506,9 @@ class ParquetWriter {
     ColumnMeta cm;
     cm.type = cd.physical;
     cm.encodings = {ENC_PLAIN, ENC_RLE};
-    cm.path_in_schema = {cd.name, "list", "element"};
+    cm.path_in_schema = cd.parent_kind == 2
+        ? std::vector<std::string>{cd.parent, "key_value", cd.name}
+        : std::vector<std::string>{cd.name, "list", "element"};
     cm.codec = codec_;
     cm.num_values = entries;
     cm.data_page_offset = pos_;

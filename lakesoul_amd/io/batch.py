@@ -69,10 +69,23 @@ class Column:
     # list<string> only: element -> byte offsets (int64 m+1); then
     # `offsets` is row -> element and `bytes_` the element payload
     elem_offsets: Optional[torch.Tensor] = None
+    # struct<...>: child Columns per member (members are scalar, carry no
+    # validity of their own — the struct's `validity` governs the row).
+    # map<K,V>: children {"key": list<K> Column, "value": list<V> Column}
+    # sharing row structure (same offsets).
+    children: Optional[Dict[str, "Column"]] = None
 
     @property
     def is_string(self) -> bool:
         return self.dtype in ("string", "binary")
+
+    @property
+    def is_struct(self) -> bool:
+        return self.dtype.startswith("struct<")
+
+    @property
+    def is_map(self) -> bool:
+        return self.dtype.startswith("map<")
 
     @property
     def is_list(self) -> bool:
@@ -89,8 +102,11 @@ class Column:
         return self.dtype[5:-1]
 
     def __len__(self) -> int:
-        if self.is_string or self.is_list:
-            return int(self.offsets.numel()) - 1
+        if self.is_struct:
+            return len(next(iter(self.children.values())))
+        if self.is_string or self.is_list or self.is_map:
+            return int(self.offsets.numel()) - 1 if self.offsets is not None \
+                else len(self.children["key"])
         return int(self.data.numel())
 
     def to_numpy(self):
@@ -109,13 +125,20 @@ class Column:
         def mv(t):
             return None if t is None else t.to(device, non_blocking=nb)
 
+        kids = (None if self.children is None else
+                {k: v.to_device(device) for k, v in self.children.items()})
         return Column(self.dtype, mv(self.data), mv(self.offsets), mv(self.bytes_),
-                      mv(self.validity), mv(self.elem_offsets))
+                      mv(self.validity), mv(self.elem_offsets), kids)
 
     def take(self, idx: torch.Tensor) -> "Column":
         """Gather rows by index tensor (moved to this column's device —
         a CPU index against a CUDA column would reach the string gather
         kernel as a host pointer)."""
+        if self.is_struct or self.is_map:
+            kids = {k: v.take(idx) for k, v in self.children.items()}
+            v = None if self.validity is None else self.validity[idx.to(
+                self.validity.device)]
+            return Column(self.dtype, validity=v, children=kids)
         dev = self.offsets.device if (self.is_string or self.is_list) else self.data.device
         if idx.device != dev:
             idx = idx.to(dev)
@@ -180,6 +203,42 @@ class Column:
         return Column(self.dtype, self.data[idx], None, None, v)
 
 
+def _slice_col(c: Column, a: int, b: int) -> Column:
+    """Zero-copy row slice [a, b) of one column (views; variable-width
+    columns rebase offsets; struct/map recurse into children)."""
+    v = None if c.validity is None else c.validity[a:b]
+    if c.is_struct or c.is_map:
+        return Column(c.dtype, validity=v,
+                      children={k: _slice_col(ch, a, b)
+                                for k, ch in c.children.items()})
+    if c.is_list_str:
+        offs = c.offsets[a : b + 1]
+        base = offs[0]
+        lo, hi = int(base), int(c.offsets[b])
+        eoffs = c.elem_offsets[lo : hi + 1]
+        eb = int(eoffs[0]) if eoffs.numel() else 0
+        return Column(
+            c.dtype, offsets=(offs - base).to(torch.int64),
+            bytes_=c.bytes_[eb:int(c.elem_offsets[hi])] if eoffs.numel()
+            else c.bytes_[:0],
+            validity=v, elem_offsets=(eoffs - eb).to(torch.int64))
+    if c.is_list:
+        offs = c.offsets[a : b + 1]
+        base = offs[0]
+        lo, hi = int(base), int(c.offsets[b])
+        return Column(c.dtype, data=c.data[lo:hi],
+                      offsets=(offs - base).to(c.offsets.dtype),
+                      validity=v)
+    if c.is_string:
+        offs = c.offsets[a : b + 1]
+        base = offs[0]
+        lo = int(base)
+        hi = int(c.offsets[b])
+        return Column(c.dtype, None, (offs - base).to(c.offsets.dtype),
+                      c.bytes_[lo:hi], v)
+    return Column(c.dtype, c.data[a:b], None, None, v)
+
+
 @dataclass
 class Batch:
     schema: Schema
@@ -202,37 +261,8 @@ class Batch:
 
     def slice(self, a: int, b: int) -> "Batch":
         """Zero-copy row slice [a, b) (views; strings rebase offsets)."""
-        cols = {}
-        for k, c in self.columns.items():
-            v = None if c.validity is None else c.validity[a:b]
-            if c.is_list_str:
-                offs = c.offsets[a : b + 1]
-                base = offs[0]
-                lo, hi = int(base), int(c.offsets[b])
-                eoffs = c.elem_offsets[lo : hi + 1]
-                eb = int(eoffs[0]) if eoffs.numel() else 0
-                cols[k] = Column(
-                    c.dtype, offsets=(offs - base).to(torch.int64),
-                    bytes_=c.bytes_[eb:int(c.elem_offsets[hi])] if eoffs.numel()
-                    else c.bytes_[:0],
-                    validity=v, elem_offsets=(eoffs - eb).to(torch.int64))
-            elif c.is_list:
-                offs = c.offsets[a : b + 1]
-                base = offs[0]
-                lo, hi = int(base), int(c.offsets[b])
-                cols[k] = Column(c.dtype, data=c.data[lo:hi],
-                                 offsets=(offs - base).to(c.offsets.dtype),
-                                 validity=v)
-            elif c.is_string:
-                offs = c.offsets[a : b + 1]
-                base = offs[0]
-                lo = int(base)
-                hi = int(c.offsets[b])
-                cols[k] = Column(c.dtype, None, (offs - base).to(c.offsets.dtype),
-                                 c.bytes_[lo:hi], v)
-            else:
-                cols[k] = Column(c.dtype, c.data[a:b], None, None, v)
-        return Batch(self.schema, cols)
+        return Batch(self.schema,
+                     {k: _slice_col(c, a, b) for k, c in self.columns.items()})
 
     # ------------------------------------------------------------------ #
 
@@ -285,6 +315,59 @@ class Batch:
             if f.name not in d:
                 raise KeyError(f"missing column {f.name}")
             v = d[f.name]
+            if f.dtype.startswith("struct<"):
+                from .schema import struct_members
+
+                members = struct_members(f.dtype)
+                if isinstance(v, tuple) and len(v) in (1, 2):
+                    kids = dict(v[0])
+                    val = (torch.as_tensor(v[1], dtype=torch.uint8)
+                           if len(v) == 2 and v[1] is not None else None)
+                else:
+                    items = list(v)
+                    val = None
+                    if any(x is None for x in items):
+                        val = torch.tensor(
+                            [0 if x is None else 1 for x in items],
+                            dtype=torch.uint8)
+                    child_schema = Schema([LsField(n, t, False)
+                                           for n, t in members])
+                    per = {}
+                    for n, t in members:
+                        dflt = "" if t in ("string", "binary") else 0
+                        per[n] = [dflt if x is None else x[n] for x in items]
+                    kids = cls.from_dict(per, child_schema).columns
+                cols[f.name] = Column(f.dtype, validity=val, children=kids)
+                continue
+            if f.dtype.startswith("map<"):
+                from .schema import map_params
+
+                kt, vt = map_params(f.dtype)
+                if isinstance(v, tuple) and len(v) in (1, 2):
+                    kids = dict(v[0])
+                    val = (torch.as_tensor(v[1], dtype=torch.uint8)
+                           if len(v) == 2 and v[1] is not None else None)
+                else:
+                    items = list(v)
+                    val = None
+                    if any(x is None for x in items):
+                        val = torch.tensor(
+                            [0 if x is None else 1 for x in items],
+                            dtype=torch.uint8)
+
+                    def pairs(x):
+                        if x is None:
+                            return []
+                        return list(x.items()) if isinstance(x, dict) else list(x)
+
+                    keys = [[p[0] for p in pairs(x)] for x in items]
+                    vals_ = [[p[1] for p in pairs(x)] for x in items]
+                    kv_schema = Schema([LsField("key", f"list<{kt}>", False),
+                                        LsField("value", f"list<{vt}>", False)])
+                    kids = cls.from_dict({"key": keys, "value": vals_},
+                                         kv_schema).columns
+                cols[f.name] = Column(f.dtype, validity=val, children=kids)
+                continue
             if f.dtype == "list<string>":
                 if isinstance(v, tuple) and len(v) in (3, 4):
                     offs = torch.as_tensor(v[0], dtype=torch.int64)
@@ -449,6 +532,52 @@ class Batch:
         arrays = []
         for f in self.schema:
             c = self.columns[f.name]
+            if f.dtype.startswith("struct<"):
+                from .schema import struct_members
+
+                members = struct_members(f.dtype)
+                child_b = Batch(
+                    Schema([LsField(n, t, False) for n, t in members]),
+                    c.children)
+                ct = child_b.to_arrow()
+                mask = None
+                if c.validity is not None:
+                    mask = pa.array(
+                        ~c.validity.cpu().numpy().astype(bool))
+                arrays.append(pa.StructArray.from_arrays(
+                    [ct.column(n).combine_chunks() for n, _ in members],
+                    [n for n, _ in members], mask=mask))
+                continue
+            if f.dtype.startswith("map<"):
+                from .schema import map_params
+
+                kt, vt = map_params(f.dtype)
+                kv_b = Batch(
+                    Schema([LsField("key", f"list<{kt}>", False),
+                            LsField("value", f"list<{vt}>", False)]),
+                    c.children)
+                kvt = kv_b.to_arrow()
+                karr = kvt.column("key").combine_chunks()
+                varr = kvt.column("value").combine_chunks()
+                n = len(c)
+                offs_np = (c.children["key"].offsets.cpu().numpy()
+                           .astype(np.int32, copy=False))
+                mt = dtype_to_arrow(f.dtype)
+                entries = pa.StructArray.from_arrays(
+                    [karr.values, varr.values],
+                    fields=[mt.key_field, mt.item_field])
+                validity_buf = None
+                null_count = 0
+                if c.validity is not None:
+                    vv = c.validity.cpu().numpy().astype(bool)
+                    null_count = int(n - vv.sum())
+                    validity_buf = pa.py_buffer(
+                        np.packbits(vv, bitorder="little").tobytes())
+                arrays.append(pa.Array.from_buffers(
+                    dtype_to_arrow(f.dtype), n,
+                    [validity_buf, pa.py_buffer(offs_np.tobytes())],
+                    null_count=null_count, children=[entries]))
+                continue
             if f.dtype == "list<string>":
                 offs_np = c.offsets.cpu().numpy().astype(np.int32, copy=False)
                 eoffs_np = c.elem_offsets.cpu().numpy().astype(np.int32, copy=False)
@@ -548,6 +677,32 @@ def concat_batches(batches: List[Batch]) -> Batch:
     for f in schema:
         cs = [b.columns[f.name] for b in batches]
         n_total = sum(len(c) for c in cs)
+        if cs and (cs[0].is_struct or cs[0].is_map):
+            if cs[0].is_struct:
+                from .schema import struct_members
+
+                kid_fields = [LsField(n, t, False)
+                              for n, t in struct_members(f.dtype)]
+            else:
+                from .schema import map_params
+
+                kt, vt = map_params(f.dtype)
+                kid_fields = [LsField("key", f"list<{kt}>", False),
+                              LsField("value", f"list<{vt}>", False)]
+            ks = Schema(kid_fields)
+            kids = concat_batches(
+                [Batch(ks, c.children) for c in cs]).columns
+            validity = None
+            if any(c.validity is not None for c in cs):
+                dev0 = next(iter(kids.values()))
+                dev0 = (dev0.offsets.device if dev0.offsets is not None
+                        else dev0.data.device)
+                validity = torch.cat([
+                    c.validity if c.validity is not None
+                    else torch.ones(len(c), dtype=torch.uint8, device=dev0)
+                    for c in cs])
+            cols[f.name] = Column(f.dtype, validity=validity, children=kids)
+            continue
         if cs and cs[0].is_list_str:
             bytes_ = torch.cat([c.bytes_ for c in cs])
             m_total = sum(int(c.offsets[-1]) for c in cs)

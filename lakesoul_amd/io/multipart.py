@@ -38,12 +38,13 @@ class StreamingParquetUpload:
         self.part_bytes = part_bytes
         fd, self._spool = tempfile.mkstemp(suffix=".parquet.spool")
         os.close(fd)
-        names = [f.name for f in schema]
-        dtypes = [f.dtype for f in schema]
-        nullable = [f.nullable for f in schema]
+        from .writer import expand_schema_leaves
+
+        names, dtypes, nullable, parents = expand_schema_leaves(schema)
         self._h = cpp().writer_open(self._spool, names, dtypes, nullable,
                                     row_group_size,
-                                    _CODEC_ID.get(compression, 6), level)
+                                    _CODEC_ID.get(compression, 6), level,
+                                    parents)
         self._sink = default_fs().open_multipart(dest)
         self._uploaded = 0
         self._rows = 0
@@ -85,33 +86,9 @@ class StreamingParquetUpload:
         """Encode one chunk as row group(s); finished bytes are handed to
         the uploader (overlaps with the caller's next chunk prep)."""
         self._check()
-        cols, offs, vals = [], [], []
-        eoffs = []
-        for f in self.schema:
-            c = batch.columns[f.name]
-            eoffs.append(None)
-            if c.is_list_str:
-                cols.append(c.bytes_.cpu())
-                offs.append(c.offsets.cpu().to(torch.int64))
-                eoffs[-1] = c.elem_offsets.cpu().to(torch.int32)
-            elif c.is_list:
-                t = c.data.cpu()
-                if f.dtype[5:-1] in ("int8", "int16"):
-                    t = t.to(torch.int32)
-                cols.append(t)
-                offs.append(c.offsets.cpu().to(torch.int64))
-            elif c.is_string:
-                cols.append(c.bytes_.cpu())
-                offs.append(c.offsets.cpu())
-            else:
-                t = c.data.cpu()
-                if f.dtype in ("int8", "int16"):
-                    t = t.to(torch.int32)
-                if f.dtype == "bool":
-                    t = t.to(torch.uint8)
-                cols.append(t)
-                offs.append(None)
-            vals.append(None if c.validity is None else c.validity.cpu())
+        from .writer import marshal_batch
+
+        (_, _, cols, offs, vals, _, eoffs, _) = marshal_batch(batch)
         cpp().writer_write(self._h, cols, offs, vals, eoffs)
         self._rows += batch.num_rows
         self._q.put(cpp().writer_bytes(self._h))

@@ -131,6 +131,33 @@ class LakeSoulScan:
             for c in sorted(self.filter_expr.columns()):
                 if c not in self.read_cols and c not in self.range_cols:
                     self.read_cols.append(c)
+        # struct/map columns read as their parquet leaves (struct s ->
+        # s.a / s.b; map m -> m.key / m.value parallel lists — the same
+        # dotted names flatten_element surfaces); the MOR merge runs per
+        # leaf and _np_to_batch reassembles the top-level column
+        from .schema import Field as _LsField, map_params, struct_members
+
+        self._leaf_fields: Dict[str, "object"] = {}
+        expanded: List[str] = []
+        for n in self.read_cols:
+            fld = self.schema.field(n)
+            sm = struct_members(fld.dtype)
+            mp = map_params(fld.dtype)
+            if sm is not None:
+                for mn, mt in sm:
+                    ln = f"{n}.{mn}"
+                    self._leaf_fields[ln] = _LsField(ln, mt, fld.nullable)
+                    expanded.append(ln)
+            elif mp is not None:
+                kt, vt = mp
+                for role, t in (("key", f"list<{kt}>"),
+                                ("value", f"list<{vt}>")):
+                    ln = f"{n}.{role}"
+                    self._leaf_fields[ln] = _LsField(ln, t, fld.nullable)
+                    expanded.append(ln)
+            else:
+                expanded.append(n)
+        self.read_cols = expanded
         self.eval_fields = list(self.out_schema.fields)
         have = {f.name for f in self.eval_fields}
         if self.filter_expr is not None:
@@ -289,10 +316,20 @@ class LakeSoulScan:
 
     # ------------------------------------------------------------------ #
 
+    def _field_for(self, name: str):
+        """Resolve a read column: schema field, or a synthesized leaf
+        field for struct members / map key-value lists."""
+        lf = self._leaf_fields.get(name)
+        return lf if lf is not None else self.schema.field(name)
+
     def _has_list_str(self) -> bool:
-        """list<string> columns decode on the host (prefixed-stream MOR)
-        and ship to HBM — they bypass the GPU unit path."""
-        return any(self.schema.field(n).dtype == "list<string>"
+        """list<string> (incl. struct/map leaves that expand to it)
+        columns decode on the host (prefixed-stream MOR) and ship to
+        HBM — they bypass the GPU unit path. Struct members and map
+        leaves likewise host-decode for now."""
+        if self._leaf_fields:
+            return True
+        return any(self._field_for(n).dtype == "list<string>"
                    for n in self.read_cols if n not in self.range_cols)
 
     def iter_batches(self) -> Iterator[Batch]:
@@ -668,50 +705,69 @@ class LakeSoulScan:
                 }
         return self._np_to_batch(merged, unit)
 
+    @staticmethod
+    def _np_col_to_column(dtype: str, npc: NpColumn) -> Column:
+        """Convert one merged NpColumn into a Batch Column."""
+        validity = (None if npc.validity is None
+                    else torch.from_numpy(npc.validity))
+        if dtype == "list<string>":
+            d = cpp().split_len_prefixed(
+                torch.from_numpy(np.ascontiguousarray(npc.bytes_)),
+                torch.from_numpy(np.ascontiguousarray(npc.offsets,
+                                                      dtype=np.int64)))
+            return Column(dtype, offsets=d["row_offsets"], bytes_=d["bytes"],
+                          elem_offsets=d["elem_offsets"], validity=validity)
+        if dtype.startswith("list<"):
+            es = np.dtype(_np_phys(dtype[5:-1])).itemsize
+            offs = np.ascontiguousarray(npc.offsets, dtype=np.int64) // es
+            vals = np.ascontiguousarray(npc.bytes_).view(_np_phys(dtype[5:-1]))
+            return Column(dtype, data=torch.from_numpy(vals.copy()),
+                          offsets=torch.from_numpy(offs), validity=validity)
+        if npc.is_string:
+            return Column(
+                dtype,
+                offsets=torch.from_numpy(
+                    np.ascontiguousarray(npc.offsets, dtype=np.int32)),
+                bytes_=torch.from_numpy(np.ascontiguousarray(npc.bytes_)),
+                validity=validity)
+        data = npc.data.astype(_np_target(dtype), copy=False)
+        return Column(dtype, data=torch.from_numpy(np.ascontiguousarray(data)),
+                      validity=validity)
+
     def _np_to_batch(self, merged: Dict[str, NpColumn], unit: ScanUnit) -> Batch:
+        from .schema import map_params, struct_members
+
         cols: Dict[str, Column] = {}
         for f in self.eval_schema:
             if f.name in self.range_cols:
                 cols[f.name] = self._range_value_column(f, unit, merged)
                 continue
-            npc = merged[f.name]
-            if f.dtype == "list<string>":
-                d = cpp().split_len_prefixed(
-                    torch.from_numpy(np.ascontiguousarray(npc.bytes_)),
-                    torch.from_numpy(np.ascontiguousarray(npc.offsets,
-                                                          dtype=np.int64)))
-                cols[f.name] = Column(
-                    f.dtype, offsets=d["row_offsets"],
-                    bytes_=d["bytes"], elem_offsets=d["elem_offsets"],
-                    validity=None if npc.validity is None
-                    else torch.from_numpy(npc.validity),
-                )
+            sm = struct_members(f.dtype)
+            mp = map_params(f.dtype)
+            if sm is not None:
+                # reassemble from the dotted leaves (merge ran per leaf;
+                # row alignment is shared, so the struct validity is any
+                # member's validity)
+                kids, validity = {}, None
+                for mn, mt in sm:
+                    mc = self._np_col_to_column(mt, merged[f"{f.name}.{mn}"])
+                    validity = mc.validity if validity is None else validity
+                    mc.validity = None
+                    kids[mn] = mc
+                cols[f.name] = Column(f.dtype, validity=validity, children=kids)
                 continue
-            if f.dtype.startswith("list<"):
-                es = np.dtype(_np_phys(f.dtype[5:-1])).itemsize
-                offs = np.ascontiguousarray(npc.offsets, dtype=np.int64) // es
-                vals = np.ascontiguousarray(npc.bytes_).view(
-                    _np_phys(f.dtype[5:-1]))
-                cols[f.name] = Column(
-                    f.dtype,
-                    data=torch.from_numpy(vals.copy()),
-                    offsets=torch.from_numpy(offs),
-                    validity=None if npc.validity is None else torch.from_numpy(npc.validity),
-                )
-            elif npc.is_string:
-                cols[f.name] = Column(
-                    f.dtype,
-                    offsets=torch.from_numpy(np.ascontiguousarray(npc.offsets, dtype=np.int32)),
-                    bytes_=torch.from_numpy(np.ascontiguousarray(npc.bytes_)),
-                    validity=None if npc.validity is None else torch.from_numpy(npc.validity),
-                )
-            else:
-                data = npc.data.astype(_np_target(f.dtype), copy=False)
-                cols[f.name] = Column(
-                    f.dtype,
-                    data=torch.from_numpy(np.ascontiguousarray(data)),
-                    validity=None if npc.validity is None else torch.from_numpy(npc.validity),
-                )
+            if mp is not None:
+                kt, vt = mp
+                kc = self._np_col_to_column(f"list<{kt}>",
+                                            merged[f"{f.name}.key"])
+                vc = self._np_col_to_column(f"list<{vt}>",
+                                            merged[f"{f.name}.value"])
+                validity = kc.validity
+                kc.validity = vc.validity = None
+                cols[f.name] = Column(f.dtype, validity=validity,
+                                      children={"key": kc, "value": vc})
+                continue
+            cols[f.name] = self._np_col_to_column(f.dtype, merged[f.name])
         return Batch(self.eval_schema, cols)
 
     def _range_value_column(self, f, unit: ScanUnit, merged) -> Column:
@@ -759,7 +815,7 @@ class LakeSoulScan:
             chunks = cpp().read_chunks_cpu_batch(h, rc, 0) if rc else []
             ci = 0
             for name in req:
-                f = self.schema.field(name)
+                f = self._field_for(name)
                 if f.dtype == "list<string>":
                     # rides the byte-string machinery as the PLAIN
                     # parquet stream ([u32 len][bytes] per element) so the
@@ -882,7 +938,7 @@ class LakeSoulScan:
             for name in names:
                 if name in out:
                     continue
-                f = self.schema.field(name)
+                f = self._field_for(name)
                 if f.dtype in ("string", "binary") or f.dtype.startswith("list<"):
                     out[name] = NpColumn(
                         f.dtype,

@@ -100,7 +100,83 @@ def list_element_dtype(dtype: str):
     return None
 
 
+def struct_members(dtype: str):
+    """Parse ``struct<a:T1,b:T2,...>`` into [(name, dtype), ...], else
+    None. Member dtypes may themselves contain commas (decimal(p,s));
+    split at depth 0 of <> and () only."""
+    s = dtype.strip()
+    if not (s.lower().startswith("struct<") and s.endswith(">")):
+        return None
+    body = s[7:-1]
+    parts, depth, cur = [], 0, []
+    for ch in body:
+        if ch in "<(":
+            depth += 1
+        elif ch in ">)":
+            depth -= 1
+        if ch == "," and depth == 0:
+            parts.append("".join(cur))
+            cur = []
+        else:
+            cur.append(ch)
+    if cur:
+        parts.append("".join(cur))
+    out = []
+    for p in parts:
+        if ":" not in p:
+            raise TypeError(f"bad struct member {p!r} in {dtype!r}")
+        name, mdt = p.split(":", 1)
+        out.append((name.strip(), mdt.strip()))
+    if not out:
+        raise TypeError(f"empty struct {dtype!r}")
+    return out
+
+
+def map_params(dtype: str):
+    """Parse ``map<K,V>`` into (key_dtype, value_dtype), else None."""
+    s = dtype.strip()
+    if not (s.lower().startswith("map<") and s.endswith(">")):
+        return None
+    body = s[4:-1]
+    depth, cur, parts = 0, [], []
+    for ch in body:
+        if ch in "<(":
+            depth += 1
+        elif ch in ">)":
+            depth -= 1
+        if ch == "," and depth == 0:
+            parts.append("".join(cur))
+            cur = []
+        else:
+            cur.append(ch)
+    parts.append("".join(cur))
+    if len(parts) != 2:
+        raise TypeError(f"map needs exactly key,value: {dtype!r}")
+    return parts[0].strip(), parts[1].strip()
+
+
 def canonical_dtype(dt: str) -> str:
+    sm = struct_members(dt)
+    if sm is not None:
+        mem = []
+        for name, mdt in sm:
+            inner = canonical_dtype(mdt)
+            if inner not in FIXED_WIDTH_BYTES and inner != "string" \
+                    and not inner.startswith("decimal"):
+                raise TypeError(
+                    f"unsupported struct member dtype {mdt!r} "
+                    "(scalar members only)")
+            mem.append(f"{name}:{inner}")
+        return "struct<" + ",".join(mem) + ">"
+    mp = map_params(dt)
+    if mp is not None:
+        k, v = canonical_dtype(mp[0]), canonical_dtype(mp[1])
+        for inner in (k, v):
+            if inner not in FIXED_WIDTH_BYTES and inner != "string":
+                raise TypeError(
+                    f"unsupported map key/value dtype {inner!r} "
+                    "(scalar keys/values only)")
+        return f"map<{k},{v}>"
     dp = decimal_params(dt)
     if dp is not None:
         p, sc = dp
@@ -219,6 +295,14 @@ def _arrow_type_to_dtype(t) -> str:
     if pt.is_list(t) or pt.is_large_list(t):
         inner = _arrow_type_to_dtype(t.value_type)
         return canonical_dtype(f"list<{inner}>")
+    if pt.is_struct(t):
+        mem = ",".join(f"{t.field(i).name}:{_arrow_type_to_dtype(t.field(i).type)}"
+                       for i in range(t.num_fields))
+        return canonical_dtype(f"struct<{mem}>")
+    if pt.is_map(t):
+        return canonical_dtype(
+            f"map<{_arrow_type_to_dtype(t.key_type)},"
+            f"{_arrow_type_to_dtype(t.item_type)}>")
     raise TypeError(f"unsupported arrow type {t}")
 
 
@@ -247,6 +331,13 @@ def dtype_to_arrow(dtype: str):
     elem = list_element_dtype(dtype)
     if elem is not None:
         return pa.list_(dtype_to_arrow(elem))
+    sm = struct_members(dtype)
+    if sm is not None:
+        return pa.struct([pa.field(n, dtype_to_arrow(t), nullable=False)
+                          for n, t in sm])
+    mp = map_params(dtype)
+    if mp is not None:
+        return pa.map_(dtype_to_arrow(mp[0]), dtype_to_arrow(mp[1]))
     return mapping[dtype]
 
 
@@ -265,6 +356,18 @@ def _dtype_to_spark(dtype: str):
         return {"type": "array",
                 "elementType": _TO_SPARK.get(elem, elem),
                 "containsNull": False}
+    sm = struct_members(dtype)
+    if sm is not None:
+        return {"type": "struct",
+                "fields": [{"name": n, "type": _dtype_to_spark(t),
+                            "nullable": False, "metadata": {}}
+                           for n, t in sm]}
+    mp = map_params(dtype)
+    if mp is not None:
+        return {"type": "map",
+                "keyType": _TO_SPARK.get(mp[0], mp[0]),
+                "valueType": _TO_SPARK.get(mp[1], mp[1]),
+                "valueContainsNull": False}
     return _TO_SPARK.get(dtype, dtype)
 
 
@@ -272,6 +375,15 @@ def _dtype_from_spark(t):
     if isinstance(t, dict) and t.get("type") == "array":
         et = t.get("elementType")
         return f"list<{_FROM_SPARK.get(et, et)}>"
+    if isinstance(t, dict) and t.get("type") == "struct":
+        mem = ",".join(
+            f"{f['name']}:{_dtype_from_spark(f['type'])}"
+            for f in t.get("fields", []))
+        return f"struct<{mem}>"
+    if isinstance(t, dict) and t.get("type") == "map":
+        kt, vt = t.get("keyType"), t.get("valueType")
+        return (f"map<{_FROM_SPARK.get(kt, kt)},"
+                f"{_FROM_SPARK.get(vt, vt)}>")
     return _FROM_SPARK.get(t, t)
 
 

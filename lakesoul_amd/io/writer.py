@@ -231,44 +231,103 @@ def _write_batch_to_file(path: str, batch: Batch, compression: str, level: int,
     return _write_batch_to_file_local(path, batch, compression, level, row_group_size)
 
 
-def _write_batch_to_file_local(path: str, batch: Batch, compression: str, level: int,
-                               row_group_size: int) -> int:
+def marshal_leaf(dtype: str, c) -> tuple:
+    """Flatten one leaf column for the C++ writer: (data, offsets,
+    elem_offsets)."""
+    if c.is_list_str:
+        return (c.bytes_.cpu(), c.offsets.cpu().to(torch.int64),
+                c.elem_offsets.cpu().to(torch.int32))
+    if c.is_list:
+        t = c.data.cpu()
+        ed = c.elem_dtype
+        if ed in ("int8", "int16"):
+            t = t.to(torch.int32)
+        if ed == "bool":
+            t = t.to(torch.uint8)
+        return (t, c.offsets.cpu().to(torch.int64), None)
+    if c.is_string:
+        return (c.bytes_.cpu(), c.offsets.cpu(), None)
+    t = c.data.cpu()
+    if dtype in ("int8", "int16"):
+        t = t.to(torch.int32)  # physical INT32
+    if dtype == "bool":
+        t = t.to(torch.uint8)
+    return (t, None, None)
+
+
+def expand_schema_leaves(schema):
+    """Schema-level leaf expansion matching marshal_batch's ordering:
+    (names, dtypes, nullable, parents) per physical parquet leaf."""
+    from .schema import map_params, struct_members
+
+    names, dtypes, nullable, parents = [], [], [], []
+    for f in schema:
+        if f.dtype.startswith("struct<"):
+            for n, t in struct_members(f.dtype):
+                names.append(n)
+                dtypes.append(t)
+                nullable.append(f.nullable)
+                parents.append(f"struct:{f.name}")
+        elif f.dtype.startswith("map<"):
+            kt, vt = map_params(f.dtype)
+            for role, t in (("key", f"list<{kt}>"), ("value", f"list<{vt}>")):
+                names.append(role)
+                dtypes.append(t)
+                nullable.append(f.nullable)
+                parents.append(f"map:{f.name}")
+        else:
+            names.append(f.name)
+            dtypes.append(f.dtype)
+            nullable.append(f.nullable)
+            parents.append("")
+    return names, dtypes, nullable, parents
+
+
+def marshal_batch(batch: Batch):
+    """Expand a batch into the C++ writer's per-leaf arrays. Struct
+    members become group leaves (parent 'struct:NAME'); map key/value
+    become two list leaves under 'map:NAME' (group validity on both)."""
+    from .schema import struct_members
+
     names, dtypes, columns, offsets, validity, nullable = [], [], [], [], [], []
-    elem_offs = []
+    elem_offs, parents = [], []
+
+    def emit(name, dtype, c, parent, null_ok, vmask):
+        d, o, eo = marshal_leaf(dtype, c)
+        names.append(name)
+        dtypes.append(dtype)
+        columns.append(d)
+        offsets.append(o)
+        elem_offs.append(eo)
+        parents.append(parent)
+        nullable.append(null_ok)
+        validity.append(None if vmask is None else vmask.cpu())
+
     for f in batch.schema:
         c = batch.columns[f.name]
-        names.append(f.name)
-        dtypes.append(f.dtype)
-        elem_offs.append(None)
-        if c.is_list_str:
-            columns.append(c.bytes_.cpu())
-            offsets.append(c.offsets.cpu().to(torch.int64))
-            elem_offs[-1] = c.elem_offsets.cpu().to(torch.int32)
-        elif c.is_list:
-            t = c.data.cpu()
-            ed = c.elem_dtype
-            if ed in ("int8", "int16"):
-                t = t.to(torch.int32)
-            if ed == "bool":
-                t = t.to(torch.uint8)
-            columns.append(t)
-            offsets.append(c.offsets.cpu().to(torch.int64))
-        elif c.is_string:
-            columns.append(c.bytes_.cpu())
-            offsets.append(c.offsets.cpu())
-        else:
-            t = c.data.cpu()
-            if f.dtype in ("int8", "int16"):
-                t = t.to(torch.int32)  # physical INT32
-            if f.dtype == "bool":
-                t = t.to(torch.uint8)
-            columns.append(t)
-            offsets.append(None)
-        validity.append(None if c.validity is None else c.validity.cpu())
-        nullable.append(f.nullable)
+        if c.is_struct:
+            for mname, _ in struct_members(f.dtype):
+                ch = c.children[mname]
+                emit(mname, ch.dtype, ch, f"struct:{f.name}", f.nullable,
+                     c.validity)
+            continue
+        if c.is_map:
+            for role in ("key", "value"):
+                emit(role, c.children[role].dtype, c.children[role],
+                     f"map:{f.name}", f.nullable, c.validity)
+            continue
+        emit(f.name, f.dtype, c, "", f.nullable, c.validity)
+    return names, dtypes, columns, offsets, validity, nullable, elem_offs, parents
+
+
+def _write_batch_to_file_local(path: str, batch: Batch, compression: str, level: int,
+                               row_group_size: int) -> int:
+    (names, dtypes, columns, offsets, validity, nullable, elem_offs,
+     parents) = marshal_batch(batch)
     return cpp().write_parquet(
         path, names, dtypes, columns, offsets, validity, nullable,
         row_group_size, _CODEC_ID.get(compression, 6), level, elem_offs,
+        parents,
     )
 
 
