@@ -130,12 +130,47 @@ class AsyncExchange:
 
         reordered = (batch.take(order) if n else
                      Batch(batch.schema, dict(batch.columns)))
+        # struct/map columns travel as their leaf columns (struct s ->
+        # s.a/s.b flat leaves; map m -> m.key/m.value list leaves); the
+        # group validity rides on the FIRST leaf and wait() reassembles
+        from ..io.schema import (Field as _LsField, Schema as _LsSchema,
+                                 map_params, struct_members)
+
+        self._tops: List[tuple] = []  # (field, kind, [leaf names])
+        xfields: List = []
+        for f in batch.schema:
+            c = reordered.columns[f.name]
+            if c.is_struct or c.is_map:
+                sm = struct_members(f.dtype)
+                if sm is not None:
+                    leaves = [(f"{f.name}.{mn}", mt) for mn, mt in sm]
+                    kind = "struct"
+                else:
+                    kt, vt = map_params(f.dtype)
+                    leaves = [(f"{f.name}.key", f"list<{kt}>"),
+                              (f"{f.name}.value", f"list<{vt}>")]
+                    kind = "map"
+                names = []
+                for i, (ln, lt) in enumerate(leaves):
+                    child = c.children[ln.rsplit(".", 1)[1]]
+                    lc = Column(lt, data=child.data, offsets=child.offsets,
+                                bytes_=child.bytes_,
+                                elem_offsets=child.elem_offsets,
+                                validity=c.validity if i == 0 else None)
+                    reordered.columns[ln] = lc
+                    xfields.append(_LsField(ln, lt, i == 0 and f.nullable))
+                    names.append(ln)
+                del reordered.columns[f.name]
+                self._tops.append((f, kind, names))
+            else:
+                xfields.append(f)
+        self._xschema = batch.schema if not self._tops else _LsSchema(xfields)
         # list<string> travels the wire as one opaque blob per row: the
         # PLAIN parquet stream ([u32 len][bytes] per element) keeps the
         # element boundaries inside the single byte payload, so it rides
         # the existing string round unchanged and is parsed back on
         # arrival (wait())
-        for f in batch.schema:
+        for f in self._xschema:
             c = reordered.columns[f.name]
             if c.is_list_str:
                 ro, ob = _pack_list_str(c)
@@ -143,7 +178,7 @@ class AsyncExchange:
                     "binary", offsets=ro, bytes_=ob, validity=c.validity)
         self.reordered = reordered
         dev = None
-        for f in batch.schema:
+        for f in self._xschema:
             c = reordered.columns[f.name]
             t = c.bytes_ if c.is_string else c.data
             if t is None and c.is_list:
@@ -154,7 +189,7 @@ class AsyncExchange:
         self.device = dev if dev is not None else torch.device("cpu")
 
         # ---- pack: per-dest segment = [colA rows][colB rows]... ---- #
-        fields = list(batch.schema)
+        fields = list(self._xschema)
         per_row = {f.name: _col_row_bytes(f, reordered.columns[f.name]) for f in fields}
         stride = sum(per_row.values())
         send_sizes_b = [r * stride for r in self.send_rows]
@@ -292,6 +327,20 @@ class AsyncExchange:
                                          device=self.device))
                 out_cols[f.name] = Column(f.dtype, data=vals,
                                           offsets=new_offs, validity=v)
+        # reassemble struct/map tops from their exchanged leaves
+        for f, kind, names in self._tops:
+            if kind == "struct":
+                kids = {ln.rsplit(".", 1)[1]: out_cols.pop(ln) for ln in names}
+            else:
+                kids = {"key": out_cols.pop(names[0]),
+                        "value": out_cols.pop(names[1])}
+            validity = None
+            for i, ch in enumerate(kids.values()):
+                if i == 0:
+                    validity = ch.validity
+                ch.validity = None
+            out_cols[f.name] = Column(f.dtype, validity=validity,
+                                      children=kids)
         self._done = Batch(self.batch.schema, out_cols)
         # release references to in-flight buffers
         self._packed_send = None

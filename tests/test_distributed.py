@@ -266,6 +266,35 @@ def _run_exchange_hard(rank, world, rdv_file, tmpdir, results):
                 elems = [bb[eo[e]:eo[e + 1]].decode()
                          for e in range(ro[i], ro[i + 1])]
                 assert elems == [f"t{g}", ""][: idx_src % 3], (g, elems)
+
+        # case 5: struct + map columns (leaf-expanded exchange)
+        schema5 = Schema([Field("id", "int64", False),
+                          Field("st", "struct<a:int64,b:string>", True),
+                          Field("mp", "map<string,int64>", True)])
+        n5 = 30
+        ids5 = np.arange(rank * 100, rank * 100 + n5, dtype=np.int64)
+        st_rows = [None if i % 4 == 0 else
+                   {"a": int(ids5[i]), "b": f"b{ids5[i]}"} for i in range(n5)]
+        mp_rows = [None if i % 5 == 0 else
+                   {f"k{j}": int(ids5[i]) + j for j in range(i % 3)}
+                   for i in range(n5)]
+        b5 = Batch.from_dict({"id": ids5, "st": st_rows, "mp": mp_rows},
+                             schema5)
+        dest5 = torch.from_numpy((ids5 % world).astype(np.int64))
+        out5 = exchange_batch_all_to_all(b5, dest5)
+        t5 = out5.to_arrow()
+        for i, g in enumerate(t5.column("id").to_pylist()):
+            idx_src = int(g % 100)
+            st = t5.column("st").to_pylist()[i]
+            mp_ = t5.column("mp").to_pylist()[i]
+            if idx_src % 4 == 0:
+                assert st is None
+            else:
+                assert st == {"a": g, "b": f"b{g}"}, (g, st)
+            if idx_src % 5 == 0:
+                assert mp_ is None
+            else:
+                assert mp_ == [(f"k{j}", g + j) for j in range(idx_src % 3)]
         results[rank] = "ok"
     finally:
         dist.destroy_process_group()
