@@ -323,12 +323,11 @@ class LakeSoulScan:
         return lf if lf is not None else self.schema.field(name)
 
     def _has_list_str(self) -> bool:
-        """list<string> (incl. struct/map leaves that expand to it)
-        columns decode on the host (prefixed-stream MOR) and ship to
-        HBM — they bypass the GPU unit path. Struct members and map
-        leaves likewise host-decode for now."""
-        if self._leaf_fields:
-            return True
+        """list<string> (incl. map<string,*>/map<*,string> leaves that
+        expand to it) decodes on the host (prefixed-stream MOR) and
+        ships to HBM — it bypasses the GPU unit path. Struct members and
+        primitive map leaves are flat/list columns the GPU path handles
+        natively."""
         return any(self._field_for(n).dtype == "list<string>"
                    for n in self.read_cols if n not in self.range_cols)
 

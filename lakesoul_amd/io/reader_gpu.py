@@ -186,6 +186,59 @@ class UnitTransfer:
         self.event = None
 
 
+
+def _project_eval(scan, merged, unit, device):
+    """Project the merged (leaf-level) batch to scan.eval_schema:
+    materialize range columns, convert byte-offset list lanes back to
+    element-offset list columns, reassemble struct/map columns from
+    their dotted leaves."""
+    from .schema import map_params, struct_members
+
+    def conv_list(lt, c):
+        es = _ESIZE[_torch_view(lt[5:-1])]
+        vals_t = (c.bytes_.view(torch_dtype_for(lt[5:-1]))
+                  if c.bytes_.numel()
+                  else torch.empty(0, dtype=torch_dtype_for(lt[5:-1]),
+                                   device=device))
+        return Column(lt, data=vals_t,
+                      offsets=c.offsets.to(torch.int64) // es,
+                      validity=c.validity)
+
+    cols: Dict[str, Column] = {}
+    nrows = merged.num_rows
+    for f in scan.eval_schema:
+        if f.name in scan.range_cols:
+            cols[f.name] = _range_col_gpu(scan, f, unit, nrows, device)
+            continue
+        sm = struct_members(f.dtype)
+        mp = map_params(f.dtype)
+        if sm is not None:
+            kids, validity = {}, None
+            for mn, mt in sm:
+                mc = merged.columns[f"{f.name}.{mn}"]
+                if validity is None:
+                    validity = mc.validity
+                kids[mn] = Column(mt, data=mc.data, offsets=mc.offsets,
+                                  bytes_=mc.bytes_,
+                                  elem_offsets=mc.elem_offsets)
+            cols[f.name] = Column(f.dtype, validity=validity, children=kids)
+            continue
+        if mp is not None:
+            kt, vt = mp
+            kc = conv_list(f"list<{kt}>", merged.columns[f"{f.name}.key"])
+            vc = conv_list(f"list<{vt}>", merged.columns[f"{f.name}.value"])
+            validity = kc.validity
+            kc.validity = vc.validity = None
+            cols[f.name] = Column(f.dtype, validity=validity,
+                                  children={"key": kc, "value": vc})
+            continue
+        if f.dtype.startswith("list<"):
+            cols[f.name] = conv_list(f.dtype, merged.columns[f.name])
+            continue
+        cols[f.name] = merged.columns[f.name]
+    return Batch(scan.eval_schema, cols)
+
+
 def read_unit_gpu(scan, unit, raw: Optional[dict] = None,
                   transfer: Optional[UnitTransfer] = None) -> Optional[Batch]:
     from .merge_gpu import merge_sorted_files_gpu
@@ -221,7 +274,9 @@ def read_unit_gpu(scan, unit, raw: Optional[dict] = None,
         if bool((status != 0).any()):
             raise RuntimeError(f"GPU zstd decompression failed: status={int((status != 0).sum())} pages")
 
-    read_schema = scan.schema.select(names)
+    from .schema import Schema as _RSch
+
+    read_schema = _RSch([scan._field_for(n) for n in names])
     # list<T> columns travel the unit as BYTE-offset binary (element
     # payload in the string-bytes region) so the merge machinery treats
     # whole lists opaquely (UseLast whole-value, merge/mod.rs:65-89);
@@ -273,7 +328,7 @@ def read_unit_gpu(scan, unit, raw: Optional[dict] = None,
             )
         cols: Dict[str, Column] = {}
         for ci, name in enumerate(names):
-            f = scan.schema.field(name)
+            f = scan._field_for(name)
             entry = outs[ci]
             if f.dtype in ("string", "binary"):
                 offs, by, vmask = entry[0], entry[1], entry[2]
@@ -285,13 +340,7 @@ def read_unit_gpu(scan, unit, raw: Optional[dict] = None,
                     data = data.to(_TARGET[f.dtype])
                 cols[name] = Column(f.dtype, data=data, validity=vmask)
         merged = Batch(read_schema, cols)
-        out_cols: Dict[str, Column] = {}
-        for f in scan.eval_schema:
-            if f.name in scan.range_cols:
-                out_cols[f.name] = _range_col_gpu(scan, f, unit, merged.num_rows, device)
-            else:
-                out_cols[f.name] = merged.columns[f.name]
-        return Batch(scan.eval_schema, out_cols)
+        return _project_eval(scan, merged, unit, device)
 
     file_batches: List[Batch] = []
     present: List[set] = []
@@ -302,7 +351,7 @@ def read_unit_gpu(scan, unit, raw: Optional[dict] = None,
         pres = set()
         for ci, name in enumerate(names):
             cd = raw["cols"][fi * ncols + ci]
-            f = scan.schema.field(name)
+            f = scan._field_for(name)
             is_list_f = f.dtype.startswith("list<")
             if not cd["present"]:
                 if f.dtype in ("string", "binary") or is_list_f:
@@ -382,24 +431,7 @@ def read_unit_gpu(scan, unit, raw: Optional[dict] = None,
 
     # project to eval schema (+ materialize range-partition columns);
     # filter-only columns are dropped after filter evaluation
-    cols: Dict[str, Column] = {}
-    nrows = merged.num_rows
-    for f in scan.eval_schema:
-        if f.name in scan.range_cols:
-            cols[f.name] = _range_col_gpu(scan, f, unit, nrows, device)
-        elif f.dtype.startswith("list<"):
-            c = merged.columns[f.name]
-            es = _ESIZE[_torch_view(f.dtype[5:-1])]
-            vals_t = (c.bytes_.view(torch_dtype_for(f.dtype[5:-1]))
-                      if c.bytes_.numel()
-                      else torch.empty(0, dtype=torch_dtype_for(f.dtype[5:-1]),
-                                       device=device))
-            cols[f.name] = Column(f.dtype, data=vals_t,
-                                  offsets=c.offsets.to(torch.int64) // es,
-                                  validity=c.validity)
-        else:
-            cols[f.name] = merged.columns[f.name]
-    return Batch(scan.eval_schema, cols)
+    return _project_eval(scan, merged, unit, device)
 
 
 def _decode_fixed_chunk_gpu(d: dict, dtype: str, device) -> Column:

@@ -704,3 +704,38 @@ def test_gpu_struct_map_scan(dev, tmp_path):
     assert gpu.column("st").to_pylist() == cpu.column("st").to_pylist()
     assert gpu.column("m").to_pylist() == cpu.column("m").to_pylist()
     assert gpu.column("st").to_pylist()[1] == {"a": 100, "b": "upd"}
+
+
+def test_gpu_struct_native_unit_path(dev, tmp_path):
+    """struct-of-scalars (and map of primitives) scans stay on the GPU
+    unit path — leaves decode as flat/list device columns and reassemble
+    at projection (no host decode gate)."""
+    catalog = _mk_catalog(tmp_path)
+    from lakesoul_amd.io.schema import Field, Schema
+
+    t = catalog.create_table(
+        "gsn",
+        Schema([Field("id", "int64", False),
+                Field("st", "struct<a:int64,b:string>"),
+                Field("mi", "map<int32,float64>")]),
+        primary_keys=["id"], hash_bucket_num=2,
+    )
+    n = 4000
+    st0 = [{"a": i * 2, "b": f"s{i}"} if i % 5 else None for i in range(n)]
+    mi0 = [None if i % 7 == 0 else
+           {j: float(i + j) for j in range(i % 3)} for i in range(n)]
+    t.upsert({"id": np.arange(n, dtype=np.int64), "st": st0, "mi": mi0})
+    t.upsert({"id": np.array([3, 44], dtype=np.int64),
+              "st": [{"a": -1, "b": "upd"}, None],
+              "mi": [{9: 9.5}, None]})
+    scan = t.scan(device="cuda")
+    assert not scan._has_list_str()  # native path engaged
+    for b in scan.iter_batches():
+        assert b.columns["st"].children["a"].data.device.type == "cuda"
+        assert b.columns["st"].children["b"].offsets.device.type == "cuda"
+        assert b.columns["mi"].children["key"].offsets.device.type == "cuda"
+    cpu = t.scan(device="cpu").to_arrow().sort_by("id")
+    gpu = t.scan(device="cuda").to_arrow().sort_by("id")
+    assert gpu.column("st").to_pylist() == cpu.column("st").to_pylist()
+    assert gpu.column("mi").to_pylist() == cpu.column("mi").to_pylist()
+    assert gpu.column("st").to_pylist()[3] == {"a": -1, "b": "upd"}
