@@ -584,11 +584,13 @@ class LakeSoulScan:
         return batch
 
     def _chunkable(self) -> bool:
-        """Chunked (PK-range) merge needs a single integer primary key
-        whose row-group min/max stats order the key space."""
+        """Chunked (PK-range) merge needs a single primary key whose
+        row-group min/max stats order the key space: integers, or
+        strings (lexicographic — trusts exact, untruncated byte-array
+        stats, which our writer always records)."""
         return (len(self.pk) == 1 and
                 self.schema.field(self.pk[0]).dtype in
-                ("int64", "int32", "int16", "int8"))
+                ("int64", "int32", "int16", "int8", "string"))
 
     def _read_unit_chunked(self, unit: ScanUnit) -> Optional[Batch]:
         """Chunked spill merge for buckets larger than the memory budget
@@ -661,12 +663,29 @@ class LakeSoulScan:
                     file_cols, self.pk, self.merge_ops, self.cdc_column, present
                 )
                 batch = self._np_to_batch(merged, unit)
-                pkv = batch.columns[pk0].data
-                mask = torch.ones(batch.num_rows, dtype=torch.bool)
-                if prev is not None:
-                    mask &= pkv > prev
-                if bound is not None:
-                    mask &= pkv <= bound
+                pkc = batch.columns[pk0]
+                if pkc.is_string:
+                    # string PK: compare decoded values the same way the
+                    # stats were decoded (utf-8/replace) so cut points
+                    # and row filters share one total order
+                    offs = pkc.offsets.numpy()
+                    by = pkc.bytes_.numpy().tobytes()
+                    vals = np.array(
+                        [by[offs[i]:offs[i + 1]].decode("utf-8", "replace")
+                         for i in range(batch.num_rows)], dtype=object)
+                    m_np = np.ones(batch.num_rows, dtype=bool)
+                    if prev is not None:
+                        m_np &= vals > prev
+                    if bound is not None:
+                        m_np &= vals <= bound
+                    mask = torch.from_numpy(m_np)
+                else:
+                    pkv = pkc.data
+                    mask = torch.ones(batch.num_rows, dtype=torch.bool)
+                    if prev is not None:
+                        mask &= pkv > prev
+                    if bound is not None:
+                        mask &= pkv <= bound
                 idx = torch.nonzero(mask, as_tuple=True)[0]
                 if idx.numel():
                     part = batch.take(idx)

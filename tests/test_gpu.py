@@ -739,3 +739,31 @@ def test_gpu_struct_native_unit_path(dev, tmp_path):
     assert gpu.column("st").to_pylist() == cpu.column("st").to_pylist()
     assert gpu.column("mi").to_pylist() == cpu.column("mi").to_pylist()
     assert gpu.column("st").to_pylist()[3] == {"a": -1, "b": "upd"}
+
+
+def test_gpu_oversized_string_pk_chunked(dev, tmp_path, monkeypatch):
+    """String-PK oversized buckets chunk-merge by lexicographic PK
+    ranges and ship each part to HBM — same rows as the CPU scan."""
+    import warnings
+
+    from lakesoul_amd.io.schema import Field, Schema
+
+    gpu_catalog = _mk_catalog(tmp_path)
+    t = gpu_catalog.create_table(
+        "bigstr",
+        Schema([Field("k", "string", False), Field("v", "float64")]),
+        primary_keys=["k"], hash_bucket_num=1,
+    )
+    n = 50000
+    keys = [f"x{i:06d}" for i in range(n)]
+    t.upsert({"k": keys, "v": np.zeros(n)})
+    t.upsert({"k": keys[::2], "v": np.ones(n // 2)})
+    full = t.scan(device="cpu").to_arrow().to_pandas().sort_values("k").reset_index(drop=True)
+    monkeypatch.setenv("LAKESOUL_MAX_UNIT_BYTES", "300000")
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        df = t.scan(device="cuda").to_arrow().to_pandas().sort_values("k").reset_index(drop=True)
+    assert any("chunked" in str(x.message) for x in w)
+    import pandas as pd
+
+    pd.testing.assert_frame_equal(full, df)

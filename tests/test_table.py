@@ -402,15 +402,46 @@ def test_oversized_unit_chunked_merge(catalog, monkeypatch):
     pd.testing.assert_frame_equal(full, chunked)
 
 
-def test_oversized_unit_guard_string_pk(catalog, monkeypatch):
-    """Non-chunkable (string PK) oversized buckets still fail with the
-    actionable error on CPU scans."""
+def test_oversized_unit_chunked_merge_string_pk(catalog, monkeypatch):
+    """String-PK oversized buckets also merge in PK ranges (lexicographic
+    cut points from byte-array row-group stats) — same result as an
+    unrestricted scan."""
+    import warnings
+
     t = catalog.create_table(
         "hugestr",
         Schema([Field("k", "string", False), Field("v", "float64")]),
         primary_keys=["k"], hash_bucket_num=1,
     )
-    t.upsert({"k": [f"x{i}" for i in range(1000)], "v": np.zeros(1000)})
+    n = 20000
+    rng = np.random.default_rng(5)
+    keys = [f"x{i:06d}" for i in range(n)]
+    t.upsert({"k": keys, "v": rng.normal(size=n)})
+    for it in range(3):
+        sel = rng.choice(n, 4000, replace=False)
+        t.upsert({"k": [keys[i] for i in sel],
+                  "v": np.full(4000, float(it + 10))})
+    full = t.to_pandas().sort_values("k").reset_index(drop=True)
+    monkeypatch.setenv("LAKESOUL_MAX_UNIT_BYTES", "150000")
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        chunked = t.to_pandas().sort_values("k").reset_index(drop=True)
+    assert any("chunked" in str(x.message) for x in w), \
+        [str(x.message) for x in w]
+    pd.testing.assert_frame_equal(full, chunked)
+
+
+def test_oversized_unit_guard_composite_pk(catalog, monkeypatch):
+    """Non-chunkable (composite PK) oversized buckets still fail with the
+    actionable error on CPU scans."""
+    t = catalog.create_table(
+        "hugecomp",
+        Schema([Field("k", "string", False), Field("k2", "int64", False),
+                Field("v", "float64")]),
+        primary_keys=["k", "k2"], hash_bucket_num=1,
+    )
+    t.upsert({"k": [f"x{i}" for i in range(1000)],
+              "k2": np.arange(1000, dtype=np.int64), "v": np.zeros(1000)})
     monkeypatch.setenv("LAKESOUL_MAX_UNIT_BYTES", "1000")
     with pytest.raises(MemoryError, match="hash\\s*buckets|buckets"):
         t.to_pandas()
