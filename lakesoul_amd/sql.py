@@ -1100,7 +1100,10 @@ def _project_and_finish_tensor(q: Query, batch, col, all_cols):
             key = id(it)
             name = f"__sx{expr_i}"
             expr_i += 1
-            vals, validity = _eval_scalar_tensor(it.expr, batch, col)
+            from .utils import timing as _tm0
+
+            with _tm0.phase("sql_expr_eval", sync_gpu=True):
+                vals, validity = _eval_scalar_tensor(it.expr, batch, col)
             from .io.batch import Column as _C
 
             nf = list(batch.schema.fields) + [_F(name, "float64", True)]
@@ -1133,7 +1136,10 @@ def _project_and_finish_tensor(q: Query, batch, col, all_cols):
                                "when GROUP BY is present")
             aggs.append((it.fn, _arg(it), it.out_name, it.distinct))
             out_order.append(("agg", it.out_name, it.out_name))
-        res = groupby_agg(batch, gcols, aggs)
+        from .utils import timing as _tm1
+
+        with _tm1.phase("sql_groupby_agg", sync_gpu=True):
+            res = groupby_agg(batch, gcols, aggs)
         fields, cols = [], {}
         for kind, src, outn in out_order:
             f = res.schema.field(src)
@@ -1443,7 +1449,10 @@ def _execute_select(catalog, q: Query, device=None):
     if _use_pandas_exec():
         df = scan.to_arrow().to_pandas()
         return _project_and_finish(q, df, lambda n: n, all_cols=schema_cols)
-    batch = scan.to_batch()
+    from .utils import timing as _tm
+
+    with _tm.phase("sql_scan_to_batch", sync_gpu=True):
+        batch = scan.to_batch()
     return _project_and_finish_tensor(q, batch, lambda n: n,
                                       all_cols=[c for c in schema_cols
                                                 if c in batch.schema.names()])
