@@ -181,6 +181,11 @@ class LakeSoulScan:
         self._shard: Optional[Tuple[int, int]] = None
         props = table.info.get_properties()
         self.cdc_column = props.get("lakesoul_cdc_change_column", None)
+        # CDC filtering needs the change column even when the projection
+        # excludes it (the merge drops 'delete' rows); eval_schema keeps
+        # the user's projection, so it never reaches the output
+        if self.cdc_column and self.cdc_column not in self.read_cols:
+            self.read_cols.append(self.cdc_column)
         self.merge_ops = dict(table.io_config().merge_operators)
         for k, v in props.items():
             if k.startswith("merge_op."):
