@@ -365,7 +365,10 @@ class LakeSoulScan:
             else:
                 yield batch
 
-    def _iter_units_pipelined(self, units: List[ScanUnit], depth: int = 3):
+    def _iter_units_pipelined(self, units: List[ScanUnit],
+                              depth: Optional[int] = None):
+        if depth is None:
+            depth = int(os.environ.get("LAKESOUL_SCAN_DEPTH", "3"))
         """GPU path, two-level pipeline (overlap engineering, SURVEY.md
         §7.2 item 5):
         - host stage: prefetch IO/decompress of upcoming units on
