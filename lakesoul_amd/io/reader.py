@@ -368,7 +368,12 @@ class LakeSoulScan:
     def _iter_units_pipelined(self, units: List[ScanUnit],
                               depth: Optional[int] = None):
         if depth is None:
-            depth = int(os.environ.get("LAKESOUL_SCAN_DEPTH", "3"))
+            # A/B on MI355X (benchmarks/scan_depth_ab.py, 2026-09-12):
+            # the step is host-fetch bound and deeper prefetch keeps the
+            # 16-thread pool busy across unit boundaries — 3: 47.1,
+            # 8: 36.6, 10: 36.4, 16: 37.8 ms/step; 2 HIP streams beat
+            # 3-4 (extra streams fragment the copy queue)
+            depth = int(os.environ.get("LAKESOUL_SCAN_DEPTH", "10"))
         """GPU path, two-level pipeline (overlap engineering, SURVEY.md
         §7.2 item 5):
         - host stage: prefetch IO/decompress of upcoming units on
