@@ -65,7 +65,13 @@ void release_schema(ArrowSchema* s) {
     if (s->children[i] && s->children[i]->release)
       s->children[i]->release(s->children[i]);
   }
-  if (s->private_data) delete (ExportPrivate*)s->private_data;
+  if (s->private_data) {
+    auto* p = (ExportPrivate*)s->private_data;
+    // the child structs are heap objects owned by this parent (their
+    // own release callbacks are no-ops per the C Data contract)
+    for (ArrowSchema* c : p->child_schemas) delete c;
+    delete p;
+  }
   s->release = nullptr;
 }
 
@@ -77,7 +83,11 @@ void release_array(ArrowArray* a) {
     if (a->children[i] && a->children[i]->release)
       a->children[i]->release(a->children[i]);
   }
-  if (a->private_data) delete (ExportPrivate*)a->private_data;
+  if (a->private_data) {
+    auto* p = (ExportPrivate*)a->private_data;
+    for (ArrowArray* c : p->child_arrays) delete c;
+    delete p;
+  }
   a->release = nullptr;
 }
 
