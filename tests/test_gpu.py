@@ -759,7 +759,7 @@ def test_gpu_oversized_string_pk_chunked(dev, tmp_path, monkeypatch):
     t.upsert({"k": keys, "v": np.zeros(n)})
     t.upsert({"k": keys[::2], "v": np.ones(n // 2)})
     full = t.scan(device="cpu").to_arrow().to_pandas().sort_values("k").reset_index(drop=True)
-    monkeypatch.setenv("LAKESOUL_MAX_UNIT_BYTES", "50000")
+    monkeypatch.setenv("LAKESOUL_MAX_UNIT_BYTES", "1000")
     with warnings.catch_warnings(record=True) as w:
         warnings.simplefilter("always")
         df = t.scan(device="cuda").to_arrow().to_pandas().sort_values("k").reset_index(drop=True)
