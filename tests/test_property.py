@@ -385,3 +385,62 @@ def test_murmur3_edge_values():
         assert create_hashes_np([np.array([v], dtype=np.int64)])[0] == \
             np.uint32(m3.hash_int64(v) & 0xFFFFFFFF)
     assert m3.hash_str("") == m3.hash_bytes(b"")
+
+
+# ---------------------------------------------------------------------- #
+# nested types: arbitrary list<string>/struct/map data survives
+# batch -> parquet -> pyarrow AND batch -> arrow -> batch round trips
+
+_str_elem = st.text(
+    alphabet=st.characters(min_codepoint=32, max_codepoint=0x2FF),
+    max_size=8)
+
+
+@settings(max_examples=25, deadline=None)
+@given(
+    tags=st.lists(st.one_of(st.none(), st.lists(_str_elem, max_size=4)),
+                  min_size=1, max_size=20),
+    structs=st.lists(
+        st.one_of(st.none(),
+                  st.fixed_dictionaries({"a": st.integers(-2**40, 2**40),
+                                         "b": _str_elem})),
+        min_size=1, max_size=20),
+    maps=st.lists(
+        st.one_of(st.none(),
+                  st.dictionaries(_str_elem, st.integers(-2**40, 2**40),
+                                  max_size=3)),
+        min_size=1, max_size=20),
+)
+def test_nested_types_roundtrip_random(tmp_path_factory, tags, structs, maps):
+    import pyarrow.parquet as pq
+
+    from lakesoul_amd.io.batch import Batch
+    from lakesoul_amd.io.schema import Field, Schema
+    from lakesoul_amd.io.writer import _write_batch_to_file_local
+
+    n = min(len(tags), len(structs), len(maps))
+    tags, structs, maps = tags[:n], structs[:n], maps[:n]
+    s = Schema([Field("id", "int64", False), Field("tags", "list<string>"),
+                Field("st", "struct<a:int64,b:string>"),
+                Field("mp", "map<string,int64>")])
+    b = Batch.from_dict({"id": np.arange(n, dtype=np.int64), "tags": tags,
+                         "st": structs, "mp": maps}, s)
+    # arrow round trip
+    t = b.to_arrow()
+    b2 = Batch.from_arrow(t, s)
+    assert b2.to_arrow().equals(t)
+    # parquet -> pyarrow golden reader
+    path = str(tmp_path_factory.mktemp("nested") / "n.parquet")
+    _write_batch_to_file_local(path, b, "zstd", 1, max(1, n // 2))
+    got = pq.read_table(path)
+    assert got.column("tags").to_pylist() == t.column("tags").to_pylist()
+    assert got.column("st").to_pylist() == t.column("st").to_pylist()
+    assert got.column("mp").to_pylist() == t.column("mp").to_pylist()
+    # take-permutation preserves rows
+    import torch as _t
+
+    perm = np.random.default_rng(0).permutation(n)
+    tk = b.take(_t.from_numpy(perm)).to_arrow()
+    for cname in ("tags", "st", "mp"):
+        ref = t.column(cname).to_pylist()
+        assert tk.column(cname).to_pylist() == [ref[i] for i in perm]
