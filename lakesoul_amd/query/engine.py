@@ -179,7 +179,9 @@ def groupby_agg(batch: Batch, group_cols: Sequence[str],
     else:
         dev = None
         for c in batch.columns.values():
-            t = c.offsets if c.is_string else c.data
+            t = c.offsets if c.offsets is not None else c.data
+            if t is None:  # struct/map columns carry tensors in children
+                continue
             dev = t.device
             break
         codes = torch.zeros(n, dtype=torch.int64, device=dev or "cpu")

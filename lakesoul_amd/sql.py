@@ -1308,13 +1308,17 @@ def _project_and_finish(q: Query, df, col, all_cols):
     return out.reset_index(drop=True)
 
 
-def _strip_quals(q: Query, valid_quals) -> None:
-    """Single-table queries: rewrite `t.col`/`alias.col` to `col`."""
+def _strip_quals(q: Query, valid_quals, passthrough=()) -> None:
+    """Single-table queries: rewrite `t.col`/`alias.col` to `col`.
+    Dotted names whose first part is a STRUCT column (``st.a``) pass
+    through untouched — they resolve as member access downstream."""
     def strip(name: str) -> str:
         if "." in name:
             qual, col = name.split(".", 1)
             if qual in valid_quals:
                 return col
+            if qual in passthrough:
+                return name
             raise SqlError(f"unknown table qualifier {qual!r}")
         return name
 
@@ -1413,7 +1417,9 @@ def _execute_select(catalog, q: Query, device=None):
     if q.joins:
         return _execute_join_select(catalog, q, device=device)
     t = catalog.table(q.table, q.namespace)
-    _strip_quals(q, {q.table, q.alias} - {""})
+    _strip_quals(q, {q.table, q.alias} - {""},
+                 passthrough={f.name for f in t.schema
+                              if f.dtype.startswith("struct<")})
     schema_cols = t.schema.names()
 
     # columns actually needed from storage
@@ -1430,9 +1436,24 @@ def _execute_select(catalog, q: Query, device=None):
             need.add(name)
     if star:
         need = set(schema_cols)
-    for c in need:
-        if c not in schema_cols:
-            raise SqlError(f"unknown column {c!r} in {q.namespace}.{q.table}")
+    # struct member access: SELECT st.a resolves to the struct column's
+    # child (Spark-style dotted access; the scan reads the struct and the
+    # member materializes as a flat column named "st.a")
+    from .io.schema import struct_members as _smem
+
+    member_refs = set()
+    for c in list(need):
+        if c in schema_cols:
+            continue
+        base, _, mem = c.partition(".")
+        sm = (_smem(t.schema.field(base).dtype)
+              if base in schema_cols else None)
+        if sm is not None and any(mn == mem for mn, _ in sm):
+            member_refs.add(c)
+            need.discard(c)
+            need.add(base)
+            continue
+        raise SqlError(f"unknown column {c!r} in {q.namespace}.{q.table}")
 
     # count(*) with no WHERE/grouping: metadata-free count-only scan
     # (reference: EmptyScanCountExec, physical_plan/empty_schema.rs:192)
@@ -1448,14 +1469,40 @@ def _execute_select(catalog, q: Query, device=None):
                   device=device)
     if _use_pandas_exec():
         df = scan.to_arrow().to_pandas()
+        for r in member_refs:
+            base, _, mem = r.partition(".")
+            df[r] = df[base].apply(
+                lambda d: None if d is None or (not isinstance(d, dict)
+                                                and pd.isna(d)) else d[mem])
         return _project_and_finish(q, df, lambda n: n, all_cols=schema_cols)
     from .utils import timing as _tm
 
     with _tm.phase("sql_scan_to_batch", sync_gpu=True):
         batch = scan.to_batch()
+    if member_refs:
+        batch = _materialize_struct_members(batch, member_refs)
     return _project_and_finish_tensor(q, batch, lambda n: n,
                                       all_cols=[c for c in schema_cols
                                                 if c in batch.schema.names()])
+
+
+def _materialize_struct_members(batch, refs):
+    """Lift struct children referenced as ``base.member`` into flat
+    columns of the batch (the struct's validity governs nullability)."""
+    from .io.batch import Batch as _B, Column as _C
+    from .io.schema import Field as _F, Schema as _S
+
+    fields = list(batch.schema.fields)
+    cols = dict(batch.columns)
+    for r in sorted(refs):
+        base, _, mem = r.partition(".")
+        c = cols[base]
+        ch = c.children[mem]
+        cols[r] = _C(ch.dtype, data=ch.data, offsets=ch.offsets,
+                     bytes_=ch.bytes_, elem_offsets=ch.elem_offsets,
+                     validity=c.validity)
+        fields.append(_F(r, ch.dtype, True))
+    return _B(_S(fields), cols)
 
 
 def repl(catalog, device=None, input_fn=input, print_fn=print):
