@@ -311,6 +311,8 @@ class VectorIndex:
         cb_sum_q = cb * sum_q
         bscale = float(1 << eb)
 
+        from ..utils import timing as _tm
+
         best_scores = torch.full((nq, k), -float("inf"), device=device)
         best_ids = torch.full((nq, k), -1, dtype=torch.int64, device=device)
         for s in self.shards:
@@ -343,11 +345,12 @@ class VectorIndex:
                 try:
                     from ..ops import hip
 
-                    est = hip().fastscan_est(
-                        rbq.bits_packed, q_dev, self.dim, rbq.f_add,
-                        rbq.f_rescale, cl_of_row.to(torch.int32),
-                        g_add_all.contiguous(), c1_sum_q.contiguous()
-                    )                                           # (nq, n)
+                    with _tm.phase("vq_fastscan_est", sync_gpu=True):
+                        est = hip().fastscan_est(
+                            rbq.bits_packed, q_dev, self.dim, rbq.f_add,
+                            rbq.f_rescale, cl_of_row.to(torch.int32),
+                            g_add_all.contiguous(), c1_sum_q.contiguous(),
+                        )                                       # (nq, n)
                 except (ImportError, AttributeError, RuntimeError):
                     use_hip = False
             if not use_hip:
@@ -365,7 +368,8 @@ class VectorIndex:
                 est = torch.where(probe_mask[:, cl_of_row], est,
                                   torch.full_like(est, float("inf")))
             kk = min(C, n)
-            top_c = torch.topk(-est, kk, dim=1)
+            with _tm.phase("vq_stage1_topk", sync_gpu=True):
+                top_c = torch.topk(-est, kk, dim=1)
             cand_est = -top_c.values                            # (nq, kk)
             cand_row = top_c.indices
             if ip_T is not None:
@@ -381,6 +385,8 @@ class VectorIndex:
                     torch.zeros_like(cand_est))
 
             # stage 2: ex-code refinement of the C candidates
+            _ex_t = _tm.phase("vq_ex_refine", sync_gpu=True)
+            _ex_t.__enter__()
             if eb > 0:
                 refined = torch.full_like(cand_est, float("inf"))
                 valid = cand_row >= 0
@@ -413,8 +419,11 @@ class VectorIndex:
                                       torch.full_like(refined, float("inf")))
             else:
                 refined = cand_est
+            _ex_t.__exit__(None, None, None)
 
             rr = min(R, refined.shape[1])
+            _rs_t = _tm.phase("vq_exact_rescore", sync_gpu=True)
+            _rs_t.__enter__()
             top_r = torch.topk(-refined, rr, dim=1).indices     # (nq, rr)
             rescore_rows = torch.gather(cand_row, 1, top_r)
             valid_r = rescore_rows >= 0
@@ -431,6 +440,7 @@ class VectorIndex:
             sel = torch.topk(cand_scores, k, dim=1)
             best_scores = sel.values
             best_ids = torch.gather(cand_ids, 1, sel.indices)
+            _rs_t.__exit__(None, None, None)
         return best_ids.cpu().numpy(), best_scores.cpu().numpy()
 
     def _scores(self, vecs: torch.Tensor, q: torch.Tensor, device) -> torch.Tensor:
