@@ -144,12 +144,15 @@ def main():
     def staged_search(use_ex: bool, C: int, R: int, probe_frac: float = 1.0):
         """Single-pass fastscan over ALL rows (96 B/vec HBM traffic), est
         masked to the probed clusters when probe_frac < 1."""
+        from lakesoul_amd.utils import timing as _tm
+
         ip_T = None
         if str(dev).startswith("cuda"):
-            est = hip_mod().fastscan_est(
-                bits_all, qs, dim, fac_all[0].contiguous(),
-                fac_all[1].contiguous(), cl_of_row.to(torch.int32),
-                g_add_all.contiguous(), c1_sum_q.contiguous())   # (nq, n)
+            with _tm.phase("vb_fastscan_est", sync_gpu=True):
+                est = hip_mod().fastscan_est(
+                    bits_all, qs, dim, fac_all[0].contiguous(),
+                    fac_all[1].contiguous(), cl_of_row.to(torch.int32),
+                    g_add_all.contiguous(), c1_sum_q.contiguous())  # (nq, n)
         else:
             from lakesoul_amd.vector.rabitq import unpack_bits
 
@@ -165,7 +168,8 @@ def main():
             probe_mask.scatter_(1, cprobe, True)
             est = torch.where(probe_mask[:, cl_of_row], est,
                               torch.full_like(est, float("inf")))
-        top_c = torch.topk(-est, min(C, n), dim=1)
+        with _tm.phase("vb_stage1_topk", sync_gpu=True):
+            top_c = torch.topk(-est, min(C, n), dim=1)
         cand_est = -top_c.values
         cand_row = top_c.indices
         if ip_T is not None:
@@ -177,6 +181,8 @@ def main():
                 fr != 0,
                 (cand_est - fac_all[0][cand_row] - ga) / fr - c1_sum_q[:, None],
                 torch.zeros_like(cand_est))
+        _ex = _tm.phase("vb_ex_refine", sync_gpu=True)
+        _ex.__enter__()
         if use_ex:
             rows = cand_row
             flat = torch.unique(rows.flatten())
@@ -195,7 +201,10 @@ def main():
                                   torch.full_like(refined, float("inf")))
         else:
             refined = cand_est
+        _ex.__exit__(None, None, None)
         rr = min(R, refined.shape[1])
+        _rs = _tm.phase("vb_rescore", sync_gpu=True)
+        _rs.__enter__()
         topr = torch.topk(-refined, rr, dim=1).indices
         rrows = torch.gather(cand_row, 1, topr).clamp_min(0)
         # exact rescore of R rows per query
@@ -205,6 +214,7 @@ def main():
         exact = torch.gather(sub.T, 1, pos)
         top = torch.topk(exact, k, dim=1).indices
         final_rows = torch.gather(rrows, 1, top)
+        _rs.__exit__(None, None, None)
         return order[final_rows.flatten()].view(final_rows.shape)
 
     cases = [
@@ -229,6 +239,12 @@ def main():
                           "clusters": clusters, "probe_frac": pf,
                           "C": C, "R": R,
                           "ivf_s": ivf_s, "quant_s": quant_s}), flush=True)
+        from lakesoul_amd.utils import timing as _tmr
+
+        if _tmr.ENABLED:
+            print(f"-- stage timing ({name}):\n{_tmr.report()}", flush=True)
+            _tmr._acc.clear()
+            _tmr._cnt.clear()
 
 
 if __name__ == "__main__":
