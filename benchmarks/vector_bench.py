@@ -185,15 +185,17 @@ def main():
         _ex.__enter__()
         if use_ex:
             rows = cand_row
-            flat = torch.unique(rows.flatten())
             if str(dev).startswith("cuda"):
-                exd = hip_mod().fastscan_ex_dot(ex_all[flat].contiguous(), qs, dim)
+                # pair-wise: one dot per (query, candidate) — no unique /
+                # gather round trip, no nq-x redundant dots
+                ex_dot = hip_mod().fastscan_ex_dot_pairs(ex_all, rows, qs, dim)
             else:
                 from lakesoul_amd.vector.rabitq import unpack_nibbles
 
+                flat = torch.unique(rows.flatten())
                 exd = unpack_nibbles(ex_all[flat], dim).to(torch.float32) @ qs.T
-            pos = torch.searchsorted(flat, rows.flatten()).view(rows.shape)
-            ex_dot = torch.gather(exd.T, 1, pos)
+                pos = torch.searchsorted(flat, rows.flatten()).view(rows.shape)
+                ex_dot = torch.gather(exd.T, 1, pos)
             g_add_cand = torch.gather(g_add_all, 1, cl_of_row[rows])
             tt = bscale * cand_ip + ex_dot + cb_sum_q[:, None]
             refined = fac_all[3, rows] + g_add_cand + fac_all[4, rows] * tt

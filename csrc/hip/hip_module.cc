@@ -44,6 +44,9 @@ void launch_fastscan_lut(const uint8_t*, const float*, float*, int64_t,
                          int32_t, int32_t, int32_t, hipStream_t);
 void launch_fastscan_ex_dot(const uint8_t*, const float*, float*, int64_t,
                             int32_t, int32_t, int32_t, hipStream_t);
+void launch_fastscan_ex_dot_pairs(const uint8_t*, const int64_t*,
+                                  const float*, float*, int32_t, int32_t,
+                                  int32_t, int32_t, hipStream_t);
 void launch_fastscan_est(const uint8_t*, const float*, const float*,
                          const float*, const int32_t*, const float*,
                          const float*, float*, int64_t, int32_t, int32_t,
@@ -795,6 +798,31 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     launch_fastscan_ex_dot(ex.data_ptr<uint8_t>(), qc.data_ptr<float>(),
                            out.data_ptr<float>(), m, nq, wn, (int)dim,
                            cur_stream());
+    return out;
+  });
+  m.def("fastscan_ex_dot_pairs",
+        [](torch::Tensor ex, torch::Tensor cand, torch::Tensor q,
+           int64_t dim) {
+    // ex (n, wn) uint8 (FULL table); cand (nq, C) int64 row indices
+    // (-1 = missing); q (nq, dim) f32 -> (nq, C) f32 per-pair dots
+    CHECK_GPU(ex);
+    CHECK_GPU(cand);
+    CHECK_GPU(q);
+    TORCH_CHECK(ex.dtype() == torch::kUInt8 && ex.is_contiguous());
+    TORCH_CHECK(cand.dtype() == torch::kInt64);
+    TORCH_CHECK(q.dtype() == torch::kFloat32);
+    int wn = (int)ex.size(1);
+    int nq = (int)cand.size(0);
+    int C = (int)cand.size(1);
+    TORCH_CHECK(wn == (dim + 1) / 2, "ex width mismatch");
+    TORCH_CHECK(q.size(0) == nq, "query count mismatch");
+    auto qc = q.contiguous();
+    auto cc = cand.contiguous();
+    auto out = torch::empty({nq, C}, q.options());
+    launch_fastscan_ex_dot_pairs(ex.data_ptr<uint8_t>(),
+                                 cc.data_ptr<int64_t>(),
+                                 qc.data_ptr<float>(), out.data_ptr<float>(),
+                                 C, nq, wn, (int)dim, cur_stream());
     return out;
   });
   m.def("snappy_decompress", &snappy_decompress);

@@ -391,25 +391,28 @@ class VectorIndex:
                 refined = torch.full_like(cand_est, float("inf"))
                 valid = cand_row >= 0
                 rows = cand_row.clamp_min(0)
-                flat_rows = torch.unique(rows.flatten())
-                ex_rows = rbq.ex_packed[flat_rows]
                 use_hip_ex = str(device).startswith("cuda")
+                ex_dot = None
                 if use_hip_ex:
                     try:
                         from ..ops import hip
 
-                        ex_dots_flat = hip().fastscan_ex_dot(
-                            ex_rows.contiguous(), q_dev, self.dim)  # (u, nq)
+                        # pair-wise kernel: one dot per (query, candidate)
+                        # — the unique-rows variant recomputed each row's
+                        # dot against EVERY query (nq-x waste, 75% of the
+                        # staged search; profiles/r02_vector_recall_qps.md)
+                        ex_dot = hip().fastscan_ex_dot_pairs(
+                            rbq.ex_packed, rows, q_dev, self.dim)  # (nq, C)
                     except (ImportError, AttributeError, RuntimeError):
                         use_hip_ex = False
-                if not use_hip_ex:
+                if ex_dot is None:
+                    flat_rows = torch.unique(rows.flatten())
+                    ex_rows = rbq.ex_packed[flat_rows]
                     ex_f = unpack_nibbles(ex_rows, self.dim).to(torch.float32)
                     ex_dots_flat = ex_f @ q_dev.T               # (u, nq)
-                # map row -> position in flat_rows
-                pos = torch.searchsorted(flat_rows, rows.flatten()).view(rows.shape)
-                ex_dot = torch.gather(
-                    ex_dots_flat.T, 1,
-                    pos)                                        # (nq, C) via q-major
+                    pos = torch.searchsorted(flat_rows, rows.flatten()
+                                             ).view(rows.shape)
+                    ex_dot = torch.gather(ex_dots_flat.T, 1, pos)  # (nq, C)
                 # per-candidate cluster id for g_add
                 g_add_cand = torch.gather(g_add_all, 1, cl_of_row[rows])
                 total_term = (bscale * cand_ip + ex_dot + cb_sum_q[:, None])

@@ -164,6 +164,32 @@ def test_fastscan_ex_dot_vs_oracle(dev):
         torch.testing.assert_close(got, ref, rtol=1e-4, atol=1e-3)
 
 
+def test_fastscan_ex_dot_pairs_vs_oracle(dev):
+    """Pair-wise ex dots (one dot per (query, candidate)) match the
+    dense oracle on the selected pairs; -1 rows produce 0."""
+    from lakesoul_amd.ops import hip
+    from lakesoul_amd.vector.rabitq import pack_nibbles
+
+    rng = np.random.default_rng(9)
+    for m, nq, C, dim in [(500, 3, 17, 64), (2049, 5, 333, 768),
+                          (100, 2, 7, 99)]:
+        codes = torch.from_numpy(rng.integers(0, 8, (m, dim)).astype(np.uint8))
+        packed = pack_nibbles(codes).to(dev)
+        q = torch.from_numpy(rng.normal(size=(nq, dim)).astype(np.float32)).to(dev)
+        cand = torch.from_numpy(
+            rng.integers(0, m, (nq, C)).astype(np.int64)).to(dev)
+        cand[0, 0] = -1
+        got = hip().fastscan_ex_dot_pairs(packed, cand, q, dim).cpu()
+        dense = codes.to(torch.float32) @ q.cpu().T   # (m, nq)
+        cc = cand.cpu()
+        for qi in range(nq):
+            for ci in range(C):
+                r = int(cc[qi, ci])
+                exp = 0.0 if r < 0 else float(dense[r, qi])
+                assert abs(float(got[qi, ci]) - exp) <= 1e-2 + 1e-4 * abs(exp), \
+                    (qi, ci, r)
+
+
 def test_gpu_rabitq_index_recall(dev, tmp_path):
     """IVF-RaBitQ staged search on GPU (fastscan kernels + MFMA rescore):
     recall@10 vs exact on the same device."""
