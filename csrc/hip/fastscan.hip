@@ -20,6 +20,7 @@
 #include <hip/hip_runtime.h>
 
 #include <cstdint>
+#include <cstdlib>
 
 namespace lakesoul {
 
@@ -120,7 +121,7 @@ __global__ __launch_bounds__(256) void fastscan_est_kernel(
 // LDS (8 x g x 16 f32 = 98 KB at 768-d, inside the 160 KB CU budget),
 // each thread keeps 8 accumulators in registers. HBM traffic drops
 // ~8x; the extra LDS reads (16/byte) ride far under LDS bandwidth.
-#define FS_QB 8
+template <int FS_QB>
 __global__ __launch_bounds__(256) void fastscan_est_qb_kernel(
     const uint8_t* __restrict__ bits, const float* __restrict__ lut,
     const float* __restrict__ f_add, const float* __restrict__ f_rescale,
@@ -189,11 +190,22 @@ void launch_fastscan_est(const uint8_t* bits, const float* lut,
                          const float* c1_sum_q, float* out, int64_t m,
                          int32_t nq, int32_t w, int32_t g,
                          int32_t n_clusters, hipStream_t s) {
-  size_t qb_lds = (size_t)FS_QB * g * 16 * sizeof(float);
-  if (nq >= 4 && qb_lds <= 120 * 1024) {
+  // query-block factor: trade HBM traffic (bits re-reads) against LDS
+  // occupancy (QB LUTs resident). Sweep via LAKESOUL_FS_QB; 0 = off.
+  static const int kQB = []() {
+    const char* e = getenv("LAKESOUL_FS_QB");
+    int v = e ? atoi(e) : 4;
+    return v == 1 || v == 2 || v == 4 || v == 8 ? v : (v <= 0 ? 0 : 4);
+  }();
+  size_t qb_lds = (size_t)kQB * g * 16 * sizeof(float);
+  if (kQB > 0 && nq >= kQB && qb_lds <= 120 * 1024) {
     dim3 grid((uint32_t)((m + 255) / 256),
-              (uint32_t)((nq + FS_QB - 1) / FS_QB));
-    hipLaunchKernelGGL(fastscan_est_qb_kernel, grid, dim3(256), qb_lds, s,
+              (uint32_t)((nq + kQB - 1) / kQB));
+    auto* fn = kQB == 8 ? fastscan_est_qb_kernel<8>
+               : kQB == 4 ? fastscan_est_qb_kernel<4>
+               : kQB == 2 ? fastscan_est_qb_kernel<2>
+                          : fastscan_est_qb_kernel<1>;
+    hipLaunchKernelGGL(fn, grid, dim3(256), qb_lds, s,
                        bits, lut, f_add, f_rescale, cl_of_row, g_add,
                        c1_sum_q, out, m, nq, w, g, n_clusters);
     return;
