@@ -194,8 +194,12 @@ void launch_fastscan_est(const uint8_t* bits, const float* lut,
   // occupancy (QB LUTs resident). Sweep via LAKESOUL_FS_QB; 0 = off.
   static const int kQB = []() {
     const char* e = getenv("LAKESOUL_FS_QB");
-    int v = e ? atoi(e) : 4;
-    return v == 1 || v == 2 || v == 4 || v == 8 ? v : (v <= 0 ? 0 : 4);
+    // measured on MI355X (benchmarks/fs_qb_ab.py, 5Mx768, 64 queries):
+    // QB=0 (per-query) 25.7 ms/batch, 2: 15.3, 4: 16.9, 8: 43.6 — the
+    // sweet spot trades a 2x bits-traffic cut against LDS occupancy
+    // (24.6 KB/block keeps ~6 blocks/CU; 98 KB at QB=8 collapses to 1)
+    int v = e ? atoi(e) : 2;
+    return v == 1 || v == 2 || v == 4 || v == 8 ? v : (v <= 0 ? 0 : 2);
   }();
   size_t qb_lds = (size_t)kQB * g * 16 * sizeof(float);
   if (kQB > 0 && nq >= kQB && qb_lds <= 120 * 1024) {
