@@ -72,13 +72,13 @@ def main():
             nq_pad = (nq + 15) // 16 * 16
             qb = torch.zeros(nq_pad, dim, dtype=torch.bfloat16, device=dev)
             qb[:nq] = q.to(torch.bfloat16)
-            return hip_mod().ann_scores(vb, qb)[:, :nq]
-        return vecs_t @ q.T
+            return hip_mod().ann_scores_t(vb, qb)[:nq]   # (nq, n)
+        return (vecs_t @ q.T).T.contiguous()
 
     torch.cuda.synchronize() if dev == "cuda" else None
     t0 = time.time()
     sc = exact_scores(qs)
-    truth = torch.topk(sc, k, dim=0).indices.T  # (nq, k)
+    truth = torch.topk(sc, k, dim=1).indices    # (nq, k)
     torch.cuda.synchronize() if dev == "cuda" else None
     exact_s = time.time() - t0
     truth_sets = [set(truth[i].tolist()) for i in range(args.queries)]
@@ -88,7 +88,7 @@ def main():
     t0 = time.time()
     for _ in range(reps):
         s = exact_scores(qs)
-        torch.topk(s, k, dim=0)
+        torch.topk(s, k, dim=1)
     torch.cuda.synchronize() if dev == "cuda" else None
     qps_exact = args.queries * reps / (time.time() - t0)
     print(json.dumps({"engine": "exact-mfma-bf16", "recall_at_10": 1.0,

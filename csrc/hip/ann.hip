@@ -20,8 +20,13 @@ namespace lakesoul {
 using bf16x8 = __attribute__((ext_vector_type(8))) short;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
-// X: [n][K] bf16 row-major; Q: [nq][K] bf16 row-major; out: [n][nq] f32.
+// X: [n][K] bf16 row-major; Q: [nq][K] bf16 row-major.
+// out: [n][nq] f32, or [nq][n] when TRANSPOSED (query-major output lets
+// the downstream per-query top-k read contiguously — torch.topk over
+// the strided dim of (n, nq) measured 28 ms vs 2.5 ms contiguous on
+// 5M x 64; benchmarks/topk_micro.py).
 // K must be a multiple of 32; nq a multiple of 16 (caller pads).
+template <bool TRANSPOSED>
 __global__ __launch_bounds__(256) void ann_scores_kernel(
     const short* __restrict__ X, const short* __restrict__ Q,
     float* __restrict__ out, int64_t n, int32_t nq, int32_t K) {
@@ -65,7 +70,12 @@ __global__ __launch_bounds__(256) void ann_scores_kernel(
 #pragma unroll
       for (int reg = 0; reg < 4; reg++) {
         int64_t orow = row0 + khalf * 4 + reg;
-        if (orow < n) out[orow * nq + q0 + t * 16 + r] = acc[t][reg];
+        if (orow < n) {
+          if (TRANSPOSED)
+            out[(int64_t)(q0 + t * 16 + r) * n + orow] = acc[t][reg];
+          else
+            out[orow * nq + q0 + t * 16 + r] = acc[t][reg];
+        }
       }
     }
   }
@@ -74,8 +84,15 @@ __global__ __launch_bounds__(256) void ann_scores_kernel(
 void launch_ann_scores(const short* X, const short* Q, float* out, int64_t n,
                        int32_t nq, int32_t K, hipStream_t s) {
   int64_t blocks = (n + 63) / 64;
-  hipLaunchKernelGGL(ann_scores_kernel, dim3((uint32_t)blocks), dim3(256), 0, s,
-                     X, Q, out, n, nq, K);
+  hipLaunchKernelGGL(ann_scores_kernel<false>, dim3((uint32_t)blocks),
+                     dim3(256), 0, s, X, Q, out, n, nq, K);
+}
+
+void launch_ann_scores_t(const short* X, const short* Q, float* out,
+                         int64_t n, int32_t nq, int32_t K, hipStream_t s) {
+  int64_t blocks = (n + 63) / 64;
+  hipLaunchKernelGGL(ann_scores_kernel<true>, dim3((uint32_t)blocks),
+                     dim3(256), 0, s, X, Q, out, n, nq, K);
 }
 
 }  // namespace lakesoul

@@ -264,16 +264,16 @@ class VectorIndex:
                         .sum(axis=2).astype(np.int32))
                 cand = torch.topk(-ham.to(torch.float32), nc, dim=0).indices  # (nc, nq)
                 flat = torch.unique(cand.flatten())
-                sub_scores = self._scores(vecs[flat], q_dev, device)  # (m, nq)
-                scores = torch.full((vecs.shape[0], nq), -float("inf"),
-                                    device=sub_scores.device)
-                scores[flat] = sub_scores
+                sub_t = self._scores_t(vecs[flat], q_dev, device)  # (nq, m)
+                scores_t = torch.full((nq, vecs.shape[0]), -float("inf"),
+                                      device=sub_t.device)
+                scores_t[:, flat] = sub_t
             else:
-                scores = self._scores(vecs, q_dev, device)  # (n, nq) f32
-            kk = min(k, scores.shape[0])
-            top = torch.topk(scores, kk, dim=0)  # (kk, nq)
-            cand_scores = torch.cat([best_scores, top.values.T], dim=1)
-            cand_ids = torch.cat([best_ids, ids[top.indices].T], dim=1)
+                scores_t = self._scores_t(vecs, q_dev, device)  # (nq, n) f32
+            kk = min(k, scores_t.shape[1])
+            top = torch.topk(scores_t, kk, dim=1)  # (nq, kk)
+            cand_scores = torch.cat([best_scores, top.values], dim=1)
+            cand_ids = torch.cat([best_ids, ids[top.indices]], dim=1)
             sel = torch.topk(cand_scores, k, dim=1)
             best_scores = sel.values
             best_ids = torch.gather(cand_ids, 1, sel.indices)
@@ -431,10 +431,10 @@ class VectorIndex:
             rescore_rows = torch.gather(cand_row, 1, top_r)
             valid_r = rescore_rows >= 0
             flat = torch.unique(rescore_rows.clamp_min(0).flatten())
-            sub = self._scores(vecs[flat], q_dev, device)       # (u, nq) exact
+            sub_t = self._scores_t(vecs[flat], q_dev, device)   # (nq, u) exact
             posr = torch.searchsorted(flat, rescore_rows.clamp_min(0).flatten()
                                       ).view(rescore_rows.shape)
-            exact = torch.gather(sub.T, 1, posr)
+            exact = torch.gather(sub_t, 1, posr)
             exact = torch.where(valid_r, exact,
                                 torch.full_like(exact, -float("inf")))
             shard_ids = ids[rescore_rows.clamp_min(0)]
@@ -445,6 +445,26 @@ class VectorIndex:
             best_ids = torch.gather(cand_ids, 1, sel.indices)
             _rs_t.__exit__(None, None, None)
         return best_ids.cpu().numpy(), best_scores.cpu().numpy()
+
+    def _scores_t(self, vecs: torch.Tensor, q: torch.Tensor, device) -> torch.Tensor:
+        """Query-major exact scores (nq, n): the MFMA kernel writes the
+        transposed layout directly so per-query top-k reads contiguous
+        rows (torch.topk over the strided dim of (n, nq) costs 11x —
+        benchmarks/topk_micro.py)."""
+        nq = q.shape[0]
+        if str(device).startswith("cuda"):
+            from ..ops import hip
+
+            nq_pad = (nq + 15) // 16 * 16
+            qb = torch.zeros(nq_pad, self.dim, dtype=torch.bfloat16, device=device)
+            qb[:nq] = q.to(torch.bfloat16)
+            if self.metric == "cosine":
+                return hip().ann_scores_t(vecs, qb)[:nq]
+            dots = hip().ann_scores_t(vecs, qb)[:nq]
+            xn = vecs.to(torch.float32).pow(2).sum(1)
+            qn = q.pow(2).sum(1)
+            return 2 * dots - xn[None, :] - qn[:, None]
+        return self._scores(vecs, q, device).T.contiguous()
 
     def _scores(self, vecs: torch.Tensor, q: torch.Tensor, device) -> torch.Tensor:
         n = vecs.shape[0]

@@ -223,3 +223,18 @@ def test_gpu_rabitq_index_recall(dev, tmp_path):
                       for i in range(32)])
     assert recall >= 0.85, recall
     assert (ids_r[:, 0] == ids_e[:, 0]).mean() >= 0.9
+
+
+def test_ann_scores_t_matches_untransposed(dev):
+    """Query-major MFMA output equals the row-major kernel transposed."""
+    from lakesoul_amd.ops import hip
+
+    rng = np.random.default_rng(11)
+    for n, nq, k_ in [(1000, 16, 128), (4097, 32, 768)]:
+        X = torch.from_numpy(rng.normal(size=(n, k_)).astype(np.float32)
+                             ).to(torch.bfloat16).to(dev)
+        Q = torch.from_numpy(rng.normal(size=(nq, k_)).astype(np.float32)
+                             ).to(torch.bfloat16).to(dev)
+        a = hip().ann_scores(X, Q)
+        b = hip().ann_scores_t(X, Q)
+        torch.testing.assert_close(b, a.T.contiguous())
