@@ -262,7 +262,10 @@ class VectorIndex:
                     ham = torch.from_numpy(
                         np.bitwise_count(x[:, None, :] ^ qq[None, :, :])
                         .sum(axis=2).astype(np.int32))
-                cand = torch.topk(-ham.to(torch.float32), nc, dim=0).indices  # (nc, nq)
+                # transpose first: topk over the strided dim of (n, nq)
+                # costs 11x (benchmarks/topk_micro.py)
+                cand = torch.topk(-ham.T.contiguous().to(torch.float32),
+                                  nc, dim=1).indices            # (nq, nc)
                 flat = torch.unique(cand.flatten())
                 sub_t = self._scores_t(vecs[flat], q_dev, device)  # (nq, m)
                 scores_t = torch.full((nq, vecs.shape[0]), -float("inf"),
