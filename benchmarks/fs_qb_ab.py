@@ -3,7 +3,7 @@ import os
 import subprocess
 import sys
 
-for qb in ("0", "2", "4", "8"):
+for qb in ("0", "2", "4", "8", "4"):
     env = dict(os.environ, LAKESOUL_FS_QB=qb, LAKESOUL_TIMING="1")
     p = subprocess.run([sys.executable, "benchmarks/vector_bench.py",
                         "--n", "5000000", "--dim", "768"],

@@ -128,7 +128,11 @@ __global__ __launch_bounds__(256) void fastscan_est_qb_kernel(
     const int32_t* __restrict__ cl_of_row, const float* __restrict__ g_add,
     const float* __restrict__ c1_sum_q, float* __restrict__ out, int64_t m,
     int32_t nq, int32_t w, int32_t g, int32_t n_clusters) {
-  extern __shared__ float slut[];  // FS_QB * g * 16 floats
+  // f16 LDS LUTs: half the footprint doubles the usable query-block at
+  // equal occupancy. Precision is a non-issue here — the estimate's own
+  // 1-bit quantization error dwarfs f16's ~5e-4 relative, and the top-C
+  // candidates are exactness-rescored downstream anyway.
+  extern __shared__ _Float16 slut_h[];  // FS_QB * g * 16 halves
   int qlo = (int)blockIdx.y * FS_QB;
   int qn = nq - qlo < FS_QB ? nq - qlo : FS_QB;
   int total = FS_QB * g * 16;
@@ -136,8 +140,9 @@ __global__ __launch_bounds__(256) void fastscan_est_qb_kernel(
     int qi = i / (g * 16);
     // zero-fill LUT slots beyond the live queries so the unrolled
     // accumulate below needs no per-iteration bound checks
-    slut[i] = qi < qn ? lut[(int64_t)(qlo + qi) * g * 16 + (i % (g * 16))]
-                      : 0.f;
+    slut_h[i] = qi < qn
+        ? (_Float16)lut[(int64_t)(qlo + qi) * g * 16 + (i % (g * 16))]
+        : (_Float16)0.f;
   }
   __syncthreads();
   int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -158,8 +163,8 @@ __global__ __launch_bounds__(256) void fastscan_est_qb_kernel(
       int hi = g2 * 16 + (int)(byte >> 4);
 #pragma unroll
       for (int qi = 0; qi < FS_QB; qi++) {
-        acc[qi] += slut[qi * g * 16 + lo];
-        if (g2 < g) acc[qi] += slut[qi * g * 16 + hi];
+        acc[qi] += (float)slut_h[qi * g * 16 + lo];
+        if (g2 < g) acc[qi] += (float)slut_h[qi * g * 16 + hi];
       }
     }
   }
@@ -170,8 +175,8 @@ __global__ __launch_bounds__(256) void fastscan_est_qb_kernel(
     int hi = g2 * 16 + (int)(byte >> 4);
 #pragma unroll
     for (int qi = 0; qi < FS_QB; qi++) {
-      acc[qi] += slut[qi * g * 16 + lo];
-      if (g2 < g) acc[qi] += slut[qi * g * 16 + hi];
+      acc[qi] += (float)slut_h[qi * g * 16 + lo];
+      if (g2 < g) acc[qi] += (float)slut_h[qi * g * 16 + hi];
     }
   }
   float fa = f_add[row];
@@ -198,10 +203,10 @@ void launch_fastscan_est(const uint8_t* bits, const float* lut,
     // QB=0 (per-query) 25.7 ms/batch, 2: 15.3, 4: 16.9, 8: 43.6 — the
     // sweet spot trades a 2x bits-traffic cut against LDS occupancy
     // (24.6 KB/block keeps ~6 blocks/CU; 98 KB at QB=8 collapses to 1)
-    int v = e ? atoi(e) : 2;
-    return v == 1 || v == 2 || v == 4 || v == 8 ? v : (v <= 0 ? 0 : 2);
+    int v = e ? atoi(e) : 4;
+    return v == 1 || v == 2 || v == 4 || v == 8 ? v : (v <= 0 ? 0 : 4);
   }();
-  size_t qb_lds = (size_t)kQB * g * 16 * sizeof(float);
+  size_t qb_lds = (size_t)kQB * g * 16 * sizeof(_Float16);
   if (kQB > 0 && nq >= kQB && qb_lds <= 120 * 1024) {
     dim3 grid((uint32_t)((m + 255) / 256),
               (uint32_t)((nq + kQB - 1) / kQB));

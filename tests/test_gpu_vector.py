@@ -147,7 +147,9 @@ def test_fastscan_est_fused_vs_oracle(dev):
     ref = (f_add.cpu()[None, :] + torch.gather(
         g_add.cpu(), 1, cl.cpu().to(torch.int64)[None, :].expand(nq, m))
         + f_res.cpu()[None, :] * (ip + c1sq.cpu()[:, None]))
-    torch.testing.assert_close(got, ref, rtol=1e-4, atol=1e-3)
+    # f16 LDS LUTs in the query-blocked kernel: ~5e-4 relative per LUT
+    # entry — far under the estimate's own 1-bit quantization error
+    torch.testing.assert_close(got, ref, rtol=5e-3, atol=5e-2)
 
 
 def test_fastscan_ex_dot_vs_oracle(dev):
