@@ -56,8 +56,8 @@ def main():
     print(f"[ds] wrote {n} rows in {time.time()-t0:.1f}s", file=sys.stderr)
 
     ds = LakeSoulIterableDataset(t, batch_size=args.batch_size, device=dev)
-    # warm epoch
-    rows = sum(b.num_rows for b in ds)
+    # warm epoch (dataset yields dicts of tensors)
+    rows = sum(b["id"].numel() for b in ds)
     assert rows == n, rows
     torch.cuda.synchronize() if dev == "cuda" else None
     t0 = time.time()
@@ -65,7 +65,7 @@ def main():
         s = 0.0
         for b in ds:
             # touch the tensors like a training step would
-            s += float(b.columns["x"].data.sum())
+            s += float(b["x"].sum())
     torch.cuda.synchronize() if dev == "cuda" else None
     dt = (time.time() - t0) / args.epochs
     print(json.dumps({"metric": "iterable_dataset_rows_per_sec",
