@@ -358,10 +358,8 @@ class LakeSoulScan:
             if self.batch_size:
                 n = batch.num_rows
                 for off in range(0, n, self.batch_size):
-                    idx = torch.arange(
-                        off, min(off + self.batch_size, n), dtype=torch.int64
-                    )
-                    yield batch.take(idx)
+                    # contiguous rows: zero-copy slice views, not a gather
+                    yield batch.slice(off, min(off + self.batch_size, n))
             else:
                 yield batch
 
