@@ -17,6 +17,7 @@
 #include <cstdlib>
 #include <stdexcept>
 #include <random>
+#include <map>
 #include <string>
 #include <vector>
 
@@ -534,5 +535,104 @@ extern "C" char* lakesoul_meta_table_info_by_path(void* hp, const char* path) {
         &f2);
     if (!f1 || !f2) { g_meta_err = "path not found"; return nullptr; }
     return lakesoul_meta_table_info(hp, name.c_str(), ns.c_str());
+  } catch (std::exception& e) { g_meta_err = e.what(); return nullptr; }
+}
+
+// SplitDesc array — the scan-planning unit engine connectors consume
+// (reference lakesoul-metadata-c lib.rs:560 create_split_desc_array +
+// transfusion.rs:316,403 SplitDesc{file_paths, primary_keys,
+// range_partition_desc, table_schema}): one entry per (partition desc,
+// hash bucket) over the LATEST snapshot, files in commit order.
+extern "C" char* lakesoul_meta_split_descs(void* hp, const char* table_name,
+                                           const char* ns) {
+  auto* h = (MetaHandle*)hp;
+  try {
+    bool f1 = false, f2 = false, f3 = false;
+    std::string tid = q1(h->db,
+        "SELECT table_id FROM table_info WHERE table_name=? AND"
+        " table_namespace=?", {table_name, ns}, &f1);
+    if (!f1) { g_meta_err = "table not found"; return nullptr; }
+    std::string schema = q1(h->db,
+        "SELECT table_schema FROM table_info WHERE table_id=?", {tid}, &f2);
+    std::string parts = q1(h->db,
+        "SELECT partitions FROM table_info WHERE table_id=?", {tid}, &f3);
+    // partitions = "rangeKeys;hashKeys" (comma-split)
+    std::string pks;
+    auto semi = parts.find(';');
+    if (semi != std::string::npos) pks = parts.substr(semi + 1);
+    std::string pks_json = "[";
+    {
+      bool first = true;
+      size_t i = 0;
+      while (i < pks.size()) {
+        size_t e = pks.find(',', i);
+        std::string k = pks.substr(i, e == std::string::npos ? e : e - i);
+        if (!k.empty()) {
+          if (!first) pks_json += ",";
+          first = false;
+          pks_json += "\"" + jesc(k) + "\"";
+        }
+        if (e == std::string::npos) break;
+        i = e + 1;
+      }
+      pks_json += "]";
+    }
+
+    std::string out = "[";
+    bool first_sd = true;
+    // latest version per partition desc
+    sqlite3_stmt* st = nullptr;
+    if (sqlite3_prepare_v2(h->db,
+            "SELECT partition_desc, MAX(version) FROM partition_info WHERE"
+            " table_id=? GROUP BY partition_desc",
+            -1, &st, nullptr) != SQLITE_OK)
+      throw std::runtime_error("prepare failed");
+    sqlite3_bind_text(st, 1, tid.c_str(), -1, nullptr);
+    std::vector<std::pair<std::string, int64_t>> descs;
+    while (sqlite3_step(st) == SQLITE_ROW) {
+      descs.push_back({(const char*)sqlite3_column_text(st, 0),
+                       sqlite3_column_int64(st, 1)});
+    }
+    sqlite3_finalize(st);
+    for (auto& dv : descs) {
+      bool fs = false;
+      std::string snap = q1(h->db,
+          "SELECT snapshot FROM partition_info WHERE table_id=? AND"
+          " partition_desc=? AND version=?",
+          {tid, dv.first, std::to_string(dv.second)}, &fs);
+      if (!fs) continue;
+      std::string files_json =
+          resolve_snapshot_files(h->db, tid, dv.first, snap);
+      // group file paths by bucket id parsed from part-..._NNNN.parquet
+      std::map<int, std::vector<std::string>> by_bucket;
+      size_t i = 0;
+      while ((i = files_json.find("\"path\":\"", i)) != std::string::npos) {
+        i += 8;
+        size_t e = files_json.find('"', i);
+        std::string p = files_json.substr(i, e - i);
+        int bucket = -1;
+        auto us = p.rfind('_');
+        auto dot = p.rfind('.');
+        if (us != std::string::npos && dot != std::string::npos && dot > us)
+          bucket = atoi(p.substr(us + 1, dot - us - 1).c_str());
+        by_bucket[bucket].push_back(p);
+        i = e;
+      }
+      for (auto& bb : by_bucket) {
+        if (!first_sd) out += ",";
+        first_sd = false;
+        out += "{\"file_paths\":[";
+        for (size_t k = 0; k < bb.second.size(); k++) {
+          if (k) out += ",";
+          out += "\"" + jesc(bb.second[k]) + "\"";
+        }
+        out += "],\"primary_keys\":" + pks_json;
+        out += ",\"partition_desc\":\"" + jesc(dv.first) + "\"";
+        out += ",\"hash_bucket\":" + std::to_string(bb.first);
+        out += ",\"table_schema\":\"" + jesc(schema) + "\"}";
+      }
+    }
+    out += "]";
+    return strdup(out.c_str());
   } catch (std::exception& e) { g_meta_err = e.what(); return nullptr; }
 }
