@@ -636,3 +636,88 @@ extern "C" char* lakesoul_meta_split_descs(void* hp, const char* table_name,
     return strdup(out.c_str());
   } catch (std::exception& e) { g_meta_err = e.what(); return nullptr; }
 }
+
+// JWT-shaped tokens (reference lakesoul-metadata-c lib.rs:465-521 /
+// jwt.rs): HMAC-SHA256 over a base64url JSON payload, wire-compatible
+// with the python gateway's TokenService (service/server.py) so a C
+// consumer can mint/verify the same bearer tokens.
+#include <openssl/hmac.h>
+#include <ctime>
+
+namespace {
+std::string b64url(const unsigned char* d, size_t n) {
+  static const char* tbl =
+      "ABCDEFGHIJKLMNOPQRSTUVWXYZabcdefghijklmnopqrstuvwxyz0123456789-_";
+  std::string out;
+  for (size_t i = 0; i < n; i += 3) {
+    uint32_t v = d[i] << 16;
+    if (i + 1 < n) v |= d[i + 1] << 8;
+    if (i + 2 < n) v |= d[i + 2];
+    out += tbl[(v >> 18) & 63];
+    out += tbl[(v >> 12) & 63];
+    if (i + 1 < n) out += tbl[(v >> 6) & 63];
+    if (i + 2 < n) out += tbl[v & 63];
+  }
+  return out;  // no padding (rstrip'd '=' on the python side)
+}
+
+std::string hmac_b64(const std::string& key, const std::string& msg) {
+  unsigned char mac[32];
+  unsigned int maclen = 0;
+  HMAC(EVP_sha256(), key.data(), (int)key.size(),
+       (const unsigned char*)msg.data(), msg.size(), mac, &maclen);
+  return b64url(mac, maclen);
+}
+}  // namespace
+
+extern "C" char* lakesoul_meta_jwt_encode(const char* sub, const char* domain,
+                                          int64_t ttl_s, const char* secret) {
+  try {
+    std::string payload = std::string("{\"sub\": \"") + jesc(sub) +
+        "\", \"domain\": \"" + jesc(domain) + "\", \"exp\": " +
+        std::to_string((int64_t)time(nullptr) + ttl_s) + "}";
+    std::string b = b64url((const unsigned char*)payload.data(),
+                           payload.size());
+    return strdup((b + "." + hmac_b64(secret, b)).c_str());
+  } catch (std::exception& e) { g_meta_err = e.what(); return nullptr; }
+}
+
+extern "C" char* lakesoul_meta_jwt_decode(const char* token,
+                                          const char* secret) {
+  try {
+    std::string t = token;
+    auto dot = t.find('.');
+    if (dot == std::string::npos) { g_meta_err = "malformed token"; return nullptr; }
+    std::string b = t.substr(0, dot), sig = t.substr(dot + 1);
+    if (hmac_b64(secret, b) != sig) { g_meta_err = "bad signature"; return nullptr; }
+    // base64url decode the payload
+    auto inv = [](char c) -> int {
+      if (c >= 'A' && c <= 'Z') return c - 'A';
+      if (c >= 'a' && c <= 'z') return c - 'a' + 26;
+      if (c >= '0' && c <= '9') return c - '0' + 52;
+      if (c == '-') return 62;
+      if (c == '_') return 63;
+      return -1;
+    };
+    std::string out;
+    uint32_t buf = 0;
+    int bits = 0;
+    for (char c : b) {
+      int v = inv(c);
+      if (v < 0) { g_meta_err = "bad base64"; return nullptr; }
+      buf = (buf << 6) | (uint32_t)v;
+      bits += 6;
+      if (bits >= 8) {
+        bits -= 8;
+        out += (char)((buf >> bits) & 0xFF);
+      }
+    }
+    // expiry check
+    auto k = out.find("\"exp\"");
+    if (k != std::string::npos) {
+      int64_t exp = atoll(out.c_str() + out.find(':', k) + 1);
+      if (exp < (int64_t)time(nullptr)) { g_meta_err = "expired"; return nullptr; }
+    }
+    return strdup(out.c_str());
+  } catch (std::exception& e) { g_meta_err = e.what(); return nullptr; }
+}

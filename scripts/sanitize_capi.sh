@@ -13,7 +13,7 @@ SAN="-fsanitize=address,undefined -fno-sanitize-recover=undefined -g -O1"
 g++ $SAN -std=c++17 -shared -fPIC \
     csrc/capi/lakesoul_c.cc csrc/capi/lakesoul_meta_c.cc \
     -o "$OUT/liblakesoul_amd_c_asan.so" \
-    -l:libzstd.so.1 -l:libsqlite3.so.0 -pthread
+    -l:libzstd.so.1 -l:libsqlite3.so.0 -l:libcrypto.so.3 -pthread
 gcc $SAN -O1 -g csrc/capi/tests/capi_smoke.c -o "$OUT/capi_smoke_asan" \
     -ldl -lpthread
 ASAN_OPTIONS=detect_leaks=1 UBSAN_OPTIONS=print_stacktrace=1 \

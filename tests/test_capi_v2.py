@@ -371,3 +371,43 @@ def test_meta_c_abi_split_descs(catalog):
         units = {u.bucket_id: u.files for u in t.scan().plan()}
         assert [os.path.basename(f) for f in d["file_paths"]] == \
             [os.path.basename(f) for f in units[d["hash_bucket"]]]
+
+
+def test_meta_c_abi_jwt_interop():
+    """C ABI JWT mint/verify interoperates with the gateway's python
+    TokenService (same HMAC-SHA256 payload.sig scheme)."""
+    import json
+
+    if not os.path.exists(LIB):
+        pytest.skip("lib not built")
+    L = ctypes.CDLL(LIB)
+    vp, cp, i64 = ctypes.c_void_p, ctypes.c_char_p, ctypes.c_int64
+    L.lakesoul_meta_jwt_encode.restype = vp
+    L.lakesoul_meta_jwt_encode.argtypes = [cp, cp, i64, cp]
+    L.lakesoul_meta_jwt_decode.restype = vp
+    L.lakesoul_meta_jwt_decode.argtypes = [cp, cp]
+    L.lakesoul_meta_free_string.argtypes = [vp]
+    L.lakesoul_meta_last_error.restype = cp
+
+    from lakesoul_amd.service.server import TokenService
+
+    ts = TokenService(secret="xsecret")
+    # C-minted -> python-verified
+    p = L.lakesoul_meta_jwt_encode(b"alice", b"teamA", 3600, b"xsecret")
+    assert p, L.lakesoul_meta_last_error()
+    tok = ctypes.cast(p, cp).value.decode()
+    L.lakesoul_meta_free_string(p)
+    claims = ts.verify(tok)
+    assert claims["sub"] == "alice" and claims["domain"] == "teamA"
+    # python-minted -> C-verified
+    tok2 = ts.issue("bob", "teamB")
+    p2 = L.lakesoul_meta_jwt_decode(tok2.encode(), b"xsecret")
+    assert p2, L.lakesoul_meta_last_error()
+    payload = json.loads(ctypes.cast(p2, cp).value.decode())
+    L.lakesoul_meta_free_string(p2)
+    assert payload["sub"] == "bob" and payload["domain"] == "teamB"
+    # tampered signature rejected
+    bad = tok[:-2] + ("AA" if not tok.endswith("AA") else "BB")
+    assert L.lakesoul_meta_jwt_decode(bad.encode(), b"xsecret") is None
+    # wrong secret rejected
+    assert L.lakesoul_meta_jwt_decode(tok2.encode(), b"other") is None
