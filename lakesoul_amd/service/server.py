@@ -189,6 +189,19 @@ def create_app(catalog=None, secret: Optional[str] = None):
         metrics.total_bytes += len(body)
         return {"rows": tbl.num_rows}
 
+    @app.get("/table/{name}/splits")
+    def table_splits(name: str, namespace: str = "default",
+                     claims: dict = Depends(auth)):
+        """Scan-plan units (SplitDesc analog, transfusion.rs:316): one
+        entry per (partition, hash bucket) with the ordered file list —
+        what an external engine needs to plan a distributed read."""
+        t = catalog.table(name, namespace)
+        _check_domain(t, claims)
+        return {"splits": [
+            {"partition_desc": u.partition_desc, "hash_bucket": u.bucket_id,
+             "file_paths": u.files, "primary_keys": t.primary_keys}
+            for u in t.scan().plan()]}
+
     @app.post("/table/{name}/compaction")
     def table_compaction(name: str, namespace: str = "default", claims: dict = Depends(auth)):
         t = catalog.table(name, namespace)

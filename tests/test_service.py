@@ -202,3 +202,22 @@ def test_s3_proxy_prometheus_metrics(tmp_path, monkeypatch, catalog):
     r = c.get("/__metrics/prometheus")
     assert r.status_code == 200
     assert "# TYPE lakesoul_s3proxy_" in r.text
+
+
+def test_table_splits_endpoint(app_client, catalog):
+    import numpy as np
+
+    from lakesoul_amd.io.schema import Field, Schema
+
+    t = catalog.create_table(
+        "gsp", Schema([Field("id", "int64", False), Field("v", "float64")]),
+        primary_keys=["id"], hash_bucket_num=2)
+    t.upsert({"id": np.arange(8, dtype=np.int64), "v": np.zeros(8)})
+    tok = app_client.post("/handshake", json={"username": "u"}).json()["token"]
+    r = app_client.get("/table/gsp/splits",
+                       headers={"Authorization": f"Bearer {tok}"})
+    assert r.status_code == 200, r.text
+    splits = r.json()["splits"]
+    assert sorted(s["hash_bucket"] for s in splits) == [0, 1]
+    for s in splits:
+        assert s["primary_keys"] == ["id"] and len(s["file_paths"]) == 1
