@@ -24,7 +24,7 @@ _TOKEN_RE = re.compile(
     r"""\s*(?:
         (?P<num>\d+\.\d+(?:[eE][+-]?\d+)?|\.\d+|\d+)
       | (?P<str>'(?:[^']|'')*')
-      | (?P<op><=|>=|<>|!=|=|<|>|\(|\)|,|\*|\.|\+|-|/)
+      | (?P<op><=|>=|<>|!=|=|<|>|\(|\)|,|\*|\.|\+|-|/|:)
       | (?P<id>[A-Za-z_][A-Za-z_0-9]*)
     )""",
     re.VERBOSE,
@@ -211,6 +211,10 @@ class _Parser:
             if k2 == "id" and v2.lower() in ("partitions", "history"):
                 ns, name = self.table_name()
                 return ("show_" + v2.lower(), {"namespace": ns, "table": name})
+            if (k2, v2) == ("kw", "create"):
+                self.expect("kw", "table")
+                ns, name = self.table_name()
+                return ("show_create", {"namespace": ns, "table": name})
             raise SqlError(f"SHOW {v2}?")
         if (k, v) == ("kw", "describe"):
             self.next()
@@ -340,9 +344,27 @@ class _Parser:
             self.expect("num")
             self.expect("op", ")")
             return "string"
-        if t not in self._SQL_TYPES:
+        if t in self._SQL_TYPES:
+            return self._SQL_TYPES[t]
+        # canonical engine dtype names round-trip (SHOW CREATE TABLE
+        # emits them; nested types like struct<...> arrive as one token
+        # stream — re-join the <...> args)
+        if t in ("struct", "map", "list", "array") and self.accept("op", "<"):
+            depth, parts = 1, [t, "<"]
+            while depth > 0:
+                k, v = self.next()
+                if (k, v) == ("op", "<"):
+                    depth += 1
+                elif (k, v) == ("op", ">"):
+                    depth -= 1
+                parts.append(v if k != "str" else f"'{v}'")
+            t = "".join(parts)
+        try:
+            from .io.schema import canonical_dtype
+
+            return canonical_dtype(t)
+        except TypeError:
             raise SqlError(f"unknown SQL type {t!r}")
-        return self._SQL_TYPES[t]
 
     def create_table(self):
         self.expect("kw", "create")
@@ -731,6 +753,20 @@ def execute_sql(catalog, sql: str, device: Optional[str] = None):
         if payload == "namespaces":
             return pd.DataFrame({"namespace": catalog.list_namespaces()})
         return pd.DataFrame({"table": catalog.list_tables()})
+    if kind == "show_create":
+        t = catalog.table(payload["table"], payload["namespace"])
+        sch = t.schema
+        cols = ", ".join(
+            f"{f.name} {f.dtype.upper() if f.dtype in ('string',) else f.dtype}"
+            + ("" if f.nullable else " NOT NULL") for f in sch)
+        stmt = f"CREATE TABLE {payload['table']} ({cols})"
+        if t.primary_keys:
+            stmt += f" PRIMARY KEY ({', '.join(t.primary_keys)})"
+        stmt += f" HASH BUCKETS {t.hash_bucket_num}"
+        if t.range_keys:
+            stmt += f" PARTITION BY ({', '.join(t.range_keys)})"
+        return pd.DataFrame({"table": [payload["table"]],
+                             "create_statement": [stmt]})
     if kind == "describe":
         ns, name = payload
         t = catalog.table(name, ns)

@@ -567,3 +567,47 @@ def test_update_decimal_column(catalog):
     execute_sql(catalog, "UPDATE updec SET amt = 7.77 WHERE id = 2")
     df = execute_sql(catalog, "SELECT id, amt FROM updec ORDER BY id")
     assert df["amt"].tolist() == [decimal.Decimal("1.00"), decimal.Decimal("7.77")]
+
+
+def test_show_create_table(catalog):
+    execute_sql(catalog,
+        "CREATE TABLE sct (id BIGINT NOT NULL, v DOUBLE, r VARCHAR(8)) "
+        "PRIMARY KEY (id) HASH BUCKETS 4 PARTITION BY (r)")
+    df = execute_sql(catalog, "SHOW CREATE TABLE sct")
+    stmt = df["create_statement"].iloc[0]
+    assert "id int64 NOT NULL" in stmt
+    assert "PRIMARY KEY (id)" in stmt
+    assert "HASH BUCKETS 4" in stmt
+    assert "PARTITION BY (r)" in stmt
+    # round trip: the emitted statement recreates an equivalent table
+    execute_sql(catalog, stmt.replace("TABLE sct", "TABLE sct2"))
+    t2 = catalog.table("sct2")
+    assert t2.primary_keys == ["id"] and t2.hash_bucket_num == 4
+    assert t2.range_keys == ["r"]
+
+
+def test_create_table_nested_types_sql(catalog):
+    """CREATE TABLE with canonical nested dtypes (list<string>,
+    struct<...>, map<K,V>) — the names SHOW CREATE TABLE emits."""
+    import numpy as np
+
+    execute_sql(catalog,
+        "CREATE TABLE nst (id BIGINT NOT NULL, tags list<string>, "
+        "st struct<a:int64,b:string>, mp map<string,int64>) "
+        "PRIMARY KEY (id) HASH BUCKETS 1")
+    t = catalog.table("nst")
+    assert t.schema.field("tags").dtype == "list<string>"
+    assert t.schema.field("st").dtype == "struct<a:int64,b:string>"
+    assert t.schema.field("mp").dtype == "map<string,int64>"
+    t.upsert({"id": np.arange(2, dtype=np.int64),
+              "tags": [["x"], None],
+              "st": [{"a": 1, "b": "y"}, None],
+              "mp": [{"k": 7}, None]})
+    df = execute_sql(catalog, "SELECT id, st.a FROM nst ORDER BY id")
+    assert df["a"].iloc[0] == 1
+    # SHOW CREATE round-trips the nested types too
+    stmt = execute_sql(catalog, "SHOW CREATE TABLE nst")["create_statement"].iloc[0]
+    assert "list<string>" in stmt and "struct<a:int64,b:string>" in stmt
+    execute_sql(catalog, stmt.replace("TABLE nst", "TABLE nst2"))
+    assert catalog.table("nst2").schema.field("st").dtype == \
+        "struct<a:int64,b:string>"
